@@ -1,0 +1,155 @@
+"""AMG hierarchy: setup loop + V/W/F/CG cycles.
+
+Reference: src/amg.cu (AMG_Setup::setup :147-804, AMG_Solve :1086-1120),
+src/cycles/fixed_cycle.cu:59-230 (the recursive cycle engine), src/amg_level.cu.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+from ..config import ConfigScope
+from ..solvers.base import create_solver
+from .level import AMGLevel, create_level
+
+
+class AMGHierarchy:
+    """Owns the level chain, smoothers and the coarse solver."""
+
+    def __init__(self, scope: ConfigScope, resources):
+        self.scope = scope
+        self.res = resources
+        self.levels: List[AMGLevel] = []
+        self.coarse_solver = None
+        self.algorithm = scope.get("algorithm")
+        self.cycle_type = scope.get("cycle")
+        self.presweeps = scope.get("presweeps")
+        self.postsweeps = scope.get("postsweeps")
+        self.coarsest_sweeps = scope.get("coarsest_sweeps")
+        self.max_levels = scope.get("max_levels")
+        self.min_coarse_rows = scope.get("min_coarse_rows")
+        self.coarsen_threshold = scope.get("coarsen_threshold")
+        self.cycle_iters = scope.get("cycle_iters")
+        self.print_grid_stats = bool(scope.get("print_grid_stats"))
+        self.setup_time = 0.0
+
+    # ------------------------------------------------------------------ setup
+    def setup(self, A):
+        """Reference AMG_Setup::setup loop (src/amg.cu:201-415): coarsen until
+        min_coarse_rows / max_levels / insufficient coarsening, then build the
+        coarse solver and per-level smoothers."""
+        t0 = time.perf_counter()
+        self.levels = []
+        dist = getattr(A, "manager", None)
+        level = create_level(self.algorithm, A, self.scope, 0)
+        self.levels.append(level)
+        while True:
+            A_l = level.A
+            n = A_l.n_rows
+            n_global = dist.global_rows(n) if dist is not None else n
+            if (len(self.levels) >= self.max_levels
+                    or n_global <= max(self.min_coarse_rows, 2)):
+                break
+            nc = level.create_coarse_vertices()
+            nc_global = dist.global_rows(nc) if dist is not None else nc
+            # insufficient coarsening -> stop (reference src/amg.cu:365-367)
+            if nc_global >= n_global * self.coarsen_threshold or nc_global == 0 \
+                    or nc_global == n_global:
+                break
+            Ac = level.create_coarse_matrix()
+            level.alloc_coarse_vectors(Ac)
+            nxt = create_level(self.algorithm, Ac, self.scope,
+                               len(self.levels))
+            level.next = nxt
+            self.levels.append(nxt)
+            level = nxt
+        # smoothers on all but the coarsest level; coarse solver on the last
+        for lvl in self.levels[:-1]:
+            lvl.smoother = self._make_smoother()
+            lvl.smoother.setup(lvl.A)
+        self._setup_coarse_solver()
+        if self.res.is_cuda:
+            torch.cuda.synchronize()
+        self.setup_time = time.perf_counter() - t0
+        if self.print_grid_stats and self.res.rank == 0:
+            print(self.grid_stats())
+
+    def _make_smoother(self):
+        name, sub = self.scope.sub_solver("smoother", "BLOCK_JACOBI")
+        # smoother inherits the amg scope's relaxation_factor unless it sets one
+        if not sub.has("relaxation_factor") and self.scope.has("relaxation_factor"):
+            sub = sub.child(dict(sub.node,
+                                 relaxation_factor=self.scope.get("relaxation_factor")))
+        return create_solver(name, sub, self.res)
+
+    def _setup_coarse_solver(self):
+        coarsest = self.levels[-1]
+        name, sub = self.scope.sub_solver("coarse_solver", "DENSE_LU_SOLVER")
+        if name == "DENSE_LU_SOLVER":
+            maxr = self.scope.get("dense_lu_max_rows")
+            if maxr and coarsest.A.n_rows > maxr:
+                name = None
+        if name and name != "NOSOLVER":
+            self.coarse_solver = create_solver(name, sub, self.res)
+            self.coarse_solver.setup(coarsest.A)
+        else:
+            # coarsest_sweeps smoothing instead (reference fixed_cycle.cu:146)
+            self.coarse_solver = None
+            coarsest.smoother = self._make_smoother()
+            coarsest.smoother.setup(coarsest.A)
+
+    # ------------------------------------------------------------------ cycles
+    def cycle(self, b: torch.Tensor, x: torch.Tensor,
+              zero_initial_guess: bool = True):
+        """One AMG cycle on the finest level."""
+        self._cycle(0, b, x, zero_initial_guess, self.cycle_type)
+
+    def _cycle(self, li: int, b, x, zero_guess: bool, ctype: str):
+        """Reference FixedCycle::cycle (src/cycles/fixed_cycle.cu:59-230)."""
+        level = self.levels[li]
+        if li == len(self.levels) - 1:
+            if self.coarse_solver is not None:
+                self.coarse_solver.solve(b, x, zero_initial_guess=True)
+            else:
+                if zero_guess:
+                    x.zero_()
+                level.smoother.sweep(b, x, self.coarsest_sweeps)
+            return
+        if zero_guess:
+            x.zero_()
+        if self.presweeps > 0:
+            level.smoother.sweep(b, x, self.presweeps)
+        ops.residual(level.A, x, b, level.r)
+        level.restrict_residual(level.r, level.bc)
+        # recurse (W recurses twice; F = W then V; reference src/cycles/)
+        repeats = 2 if ctype == "W" and li + 2 < len(self.levels) else 1
+        sub_type = ctype if ctype != "F" else "F"
+        level.xc.zero_()
+        for rep in range(repeats):
+            self._cycle(li + 1, level.bc, level.xc, rep == 0,
+                        "V" if ctype == "F" and rep > 0 else sub_type)
+        level.prolongate_and_apply(level.xc, x)
+        if self.postsweeps > 0:
+            level.smoother.sweep(b, x, self.postsweeps)
+
+    # ------------------------------------------------------------------ stats
+    def grid_stats(self) -> str:
+        """Reference AMG::getGridStatisticsString (src/amg.cu:1230-1330)."""
+        lines = ["AMG Grid:", "   Number of Levels: %d" % len(self.levels),
+                 "      LVL        ROWS         NNZ    SPRSTY"]
+        total_rows = total_nnz = 0
+        fine_rows = self.levels[0].A.n_rows
+        fine_nnz = self.levels[0].A.nnz
+        for i, l in enumerate(self.levels):
+            n, nnz = l.A.n_rows, l.A.nnz
+            total_rows += n
+            total_nnz += nnz
+            sp = nnz / (n * n) if n else 0.0
+            lines.append(f"      {i:3d} {n:11d} {nnz:11d}  {sp:8.3g}")
+        lines.append(f"      Grid Complexity: {total_rows / max(fine_rows, 1):.5g}")
+        lines.append(f"      Operator Complexity: {total_nnz / max(fine_nnz, 1):.5g}")
+        return "\n".join(lines)

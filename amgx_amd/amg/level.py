@@ -1,0 +1,145 @@
+"""AMG level implementations: aggregation and classical Ruge-Stueben.
+
+Reference: include/amg_level.h:51-316, src/aggregation/aggregation_amg_level.cu,
+src/classical/classical_amg_level.cu. A level owns its fine matrix A, builds
+the coarse operator, and implements restrict/prolongate.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..matrix import CSRMatrix
+
+
+class AMGLevel:
+    """Linked-list node of the hierarchy (reference include/amg_level.h)."""
+
+    def __init__(self, A: CSRMatrix, scope, index: int):
+        self.A = A
+        self.scope = scope
+        self.index = index
+        self.next: Optional[AMGLevel] = None
+        self.smoother = None
+        # work vectors allocated at setup
+        n = A.n_rows * A.block_dim
+        self.r = torch.zeros(n, dtype=A.dtype, device=A.device)
+        self.bc = None   # coarse rhs
+        self.xc = None   # coarse correction
+
+    # -- to be implemented by subclasses -------------------------------------
+    def create_coarse_vertices(self) -> int:
+        raise NotImplementedError
+
+    def create_coarse_matrix(self) -> CSRMatrix:
+        raise NotImplementedError
+
+    def restrict_residual(self, r, bc):
+        raise NotImplementedError
+
+    def prolongate_and_apply(self, xc, x):
+        raise NotImplementedError
+
+    def alloc_coarse_vectors(self, Ac: CSRMatrix):
+        nc = Ac.n_rows * Ac.block_dim
+        self.bc = torch.zeros(nc, dtype=Ac.dtype, device=Ac.device)
+        self.xc = torch.zeros(nc, dtype=Ac.dtype, device=Ac.device)
+
+
+class AggregationLevel(AMGLevel):
+    """Unsmoothed aggregation with piecewise-constant P (reference
+    src/aggregation/aggregation_amg_level.cu). SIZE_4/SIZE_8 compose repeated
+    SIZE_2 pairwise matchings (reference src/aggregation/selectors/size4... is
+    'two SIZE_2 passes')."""
+
+    def __init__(self, A, scope, index):
+        super().__init__(A, scope, index)
+        self.aggregates = None
+        self.num_aggregates = 0
+
+    def create_coarse_vertices(self) -> int:
+        selector = self.scope.get("selector")
+        passes = {"SIZE_2": 1, "SIZE_4": 2, "SIZE_8": 3,
+                  "MULTI_PAIRWISE": 2, "DUMMY": 1}.get(selector, 1)
+        maxit = self.scope.get("max_matching_iterations")
+        agg, num = ops.size2_matching(self.A, max_iterations=maxit)
+        work = self.A
+        for _ in range(passes - 1):
+            if num <= self.scope.get("min_coarse_rows"):
+                break
+            work = ops.galerkin_aggregation(work, agg.to(work.row_offsets.device),
+                                            num)
+            agg2, num2 = ops.size2_matching(work, max_iterations=maxit)
+            agg = agg2.to(agg.device)[agg.long()]
+            num = num2
+        self.aggregates = agg.to(self.A.row_offsets.device)
+        self.num_aggregates = num
+        return num
+
+    def create_coarse_matrix(self) -> CSRMatrix:
+        Ac = ops.galerkin_aggregation(self.A, self.aggregates,
+                                      self.num_aggregates)
+        return Ac
+
+    def restrict_residual(self, r, bc):
+        out = ops.restrict_agg(r, self.aggregates, self.num_aggregates,
+                               self.A.block_dim)
+        bc.copy_(out.reshape(bc.shape))
+
+    def prolongate_and_apply(self, xc, x):
+        ops.prolongate_agg(x, xc, self.aggregates, self.A.block_dim)
+
+
+class ClassicalLevel(AMGLevel):
+    """Classical Ruge-Stueben level (reference
+    src/classical/classical_amg_level.cu): AHAT strength -> PMIS C/F split ->
+    distance-1 (direct) interpolation -> R = P^T -> RAP Galerkin product."""
+
+    def __init__(self, A, scope, index):
+        super().__init__(A, scope, index)
+        self.P = None
+        self.R = None
+        self.cf_map = None
+        self.num_coarse = 0
+
+    def create_coarse_vertices(self) -> int:
+        S = ops._backend(self.A).strength_ahat(
+            self.A, float(self.scope.get("strength_threshold")),
+            float(self.scope.get("max_row_sum")))
+        cf, nc = ops._backend(self.A).pmis_select(self.A, S)
+        self.S = S
+        self.cf_map = cf
+        self.num_coarse = nc
+        return nc
+
+    def create_coarse_matrix(self) -> CSRMatrix:
+        interp = self.scope.get("interpolator")
+        self.P = ops._backend(self.A).interp_d1(self.A, self.S, self.cf_map,
+                                                self.num_coarse)
+        tf = float(self.scope.get("interp_truncation_factor"))
+        me = int(self.scope.get("interp_max_elements"))
+        if tf > 0.0 or me >= 0:
+            self.P = ops.truncate_rows(self.P, tf, me)
+        self.R = ops.transpose(self.P)
+        Ac = ops.galerkin_rap(self.R, self.A, self.P)
+        return Ac
+
+    def restrict_residual(self, r, bc):
+        ops.spmv(self.R, r, bc)
+
+    def prolongate_and_apply(self, xc, x):
+        tmp = torch.zeros_like(x)
+        ops.spmv(self.P, xc, tmp)
+        ops.axpy(x, tmp, 1.0)
+
+
+def create_level(algorithm: str, A, scope, index) -> AMGLevel:
+    if algorithm == "AGGREGATION":
+        return AggregationLevel(A, scope, index)
+    if algorithm in ("CLASSICAL", "ENERGYMIN"):
+        return ClassicalLevel(A, scope, index)
+    raise KeyError(f"unknown AMG algorithm {algorithm!r}")

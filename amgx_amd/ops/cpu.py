@@ -1,0 +1,684 @@
+"""CPU reference backend (scipy/torch/numpy).
+
+These are the host-path implementations: numerically defining references for
+the gfx950 kernels in ``amgx_amd/csrc`` (GPU numerics tests compare against
+these) and the engine of the host memory-space solve path (driver config #1).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import scipy.sparse as sp
+import torch
+
+
+# ---------------------------------------------------------------------- helpers
+def _np(x: torch.Tensor) -> np.ndarray:
+    return x.detach().numpy()
+
+
+def _csr(A) -> sp.csr_matrix:
+    m = A.to_scipy()
+    if not sp.issparse(m):
+        raise TypeError("expected sparse")
+    return m
+
+
+# ---------------------------------------------------------------------- structure
+def compute_diag_index(A) -> torch.Tensor:
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    n = A.n_rows
+    out = np.full(n, -1, dtype=np.int32)
+    for i in range(n):
+        s, e = ro[i], ro[i + 1]
+        hits = np.nonzero(ci[s:e] == i)[0]
+        if hits.size:
+            out[i] = s + hits[0]
+    return torch.from_numpy(out)
+
+
+def extract_diagonal(A) -> torch.Tensor:
+    di = _np(A.diag_index()).astype(np.int64)
+    if A.block_dim == 1:
+        vals = _np(A.values)
+        out = np.where(di >= 0, vals[np.maximum(di, 0)], 0.0)
+        return torch.from_numpy(out.astype(vals.dtype))
+    vals = _np(A.values)
+    b = A.block_dim
+    out = np.zeros((A.n_rows, b, b), dtype=vals.dtype)
+    mask = di >= 0
+    out[mask] = vals[di[mask]]
+    return torch.from_numpy(out)
+
+
+# ---------------------------------------------------------------------- SpMV
+def spmv(A, x, y=None, alpha=1.0, beta=0.0, row_begin=0, row_end=-1):
+    m = A.to_scipy()
+    b = A.block_dim
+    if row_end < 0:
+        row_end = A.n_rows
+    xv = _np(x).reshape(-1)
+    prod = m @ xv
+    if y is None:
+        y = torch.zeros(A.n_rows * b, dtype=x.dtype)
+        beta = 0.0
+    yv = _np(y).reshape(-1)
+    lo, hi = row_begin * b, row_end * b
+    yv[lo:hi] = alpha * prod[lo:hi] + (beta * yv[lo:hi] if beta != 0.0 else 0.0)
+    return y
+
+
+def residual(A, x, b, r=None):
+    m = A.to_scipy()
+    rv = _np(b).reshape(-1) - m @ _np(x).reshape(-1)
+    if r is None:
+        return torch.from_numpy(rv.astype(_np(b).dtype))
+    _np(r).reshape(-1)[:] = rv
+    return r
+
+
+# ---------------------------------------------------------------------- BLAS-1
+def dot(x, y):
+    return float(torch.dot(x.reshape(-1).double(), y.reshape(-1).double()))
+
+
+def nrm2(x):
+    return float(torch.linalg.vector_norm(x.reshape(-1).double()))
+
+
+def nrm1(x):
+    return float(x.reshape(-1).double().abs().sum())
+
+
+def nrmmax(x):
+    return float(x.reshape(-1).double().abs().max()) if x.numel() else 0.0
+
+
+def axpy(y, x, alpha):
+    y.reshape(-1).add_(x.reshape(-1), alpha=alpha)
+    return y
+
+
+def axpby(y, x, alpha, beta):
+    yv = y.reshape(-1)
+    yv.mul_(beta).add_(x.reshape(-1), alpha=alpha)
+    return y
+
+
+def scal(x, alpha):
+    x.reshape(-1).mul_(alpha)
+    return x
+
+
+# ---------------------------------------------------------------------- smoothers
+def jacobi_dinv(A, l1: bool = False) -> torch.Tensor:
+    """Scalar: 1/(d_i [+ l1 off-row-sum]); block: inverse of each diagonal block
+    (l1 adds the off-diagonal block-row L1 norms to the block diagonal,
+    reference src/solvers/jacobi_l1_solver.cu)."""
+    if A.block_dim == 1:
+        d = extract_diagonal(A).double().clone()
+        if l1:
+            m = _csr(A)
+            abssum = np.abs(m).sum(axis=1).A1 - np.abs(_np(extract_diagonal(A)))
+            d += torch.from_numpy(np.sign(_np(d)) * abssum)
+            # reference uses d_i + sign(d_i)*sum|offdiag|
+        d = torch.where(d.abs() > 0, d, torch.ones_like(d))
+        return (1.0 / d).to(A.dtype)
+    # block
+    b = A.block_dim
+    D = extract_diagonal(A).double().clone()  # (n, b, b)
+    if l1:
+        vals = _np(A.values)
+        ro = _np(A.row_offsets).astype(np.int64)
+        ci = _np(A.col_indices).astype(np.int64)
+        add = np.zeros((A.n_rows, b), dtype=np.float64)
+        for i in range(A.n_rows):
+            for k in range(ro[i], ro[i + 1]):
+                if ci[k] != i:
+                    add[i] += np.abs(vals[k]).sum(axis=1)
+        D += torch.diag_embed(torch.from_numpy(add))
+    # guard singular blocks
+    eye = torch.eye(b, dtype=torch.float64).expand_as(D)
+    sing = torch.linalg.matrix_rank(D) < b
+    D = torch.where(sing.view(-1, 1, 1), eye, D)
+    return torch.linalg.inv(D).to(A.dtype)
+
+
+def jacobi_smooth(A, dinv, b, x, omega: float):
+    r = residual(A, x, b)
+    if A.block_dim == 1:
+        x.add_(dinv.reshape(-1) * r.reshape(-1), alpha=omega)
+    else:
+        bd = A.block_dim
+        rb = r.reshape(-1, bd, 1).to(dinv.dtype)
+        upd = torch.bmm(dinv, rb).reshape(-1)
+        x.reshape(-1).add_(upd.to(x.dtype), alpha=omega)
+    return x
+
+
+def gs_smooth_color(A, dinv, b, x, color_rows, omega: float):
+    m = _csr(A) if A.block_dim == 1 else A.to_scipy()
+    rows = _np(color_rows).astype(np.int64)
+    bd = A.block_dim
+    if bd == 1:
+        sub = m[rows, :] @ _np(x).reshape(-1)
+        upd = omega * _np(dinv)[rows] * (_np(b).reshape(-1)[rows] - sub)
+        _np(x).reshape(-1)[rows] += upd
+    else:
+        xv = _np(x).reshape(-1)
+        rowsb = (rows[:, None] * bd + np.arange(bd)[None, :]).reshape(-1)
+        sub = (m @ xv)[rowsb]
+        res = (_np(b).reshape(-1)[rowsb] - sub).reshape(-1, bd, 1)
+        upd = np.matmul(_np(dinv)[rows].astype(res.dtype), res).reshape(-1)
+        xv[rowsb] += omega * upd
+    return x
+
+
+# ---------------------------------------------------------------------- coloring
+def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0):
+    """Sequential greedy distance-1 coloring (host reference;
+    GPU uses the parallel min-max kernels — colors differ, validity tested)."""
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    n = A.n_rows
+    colors = np.full(n, -1, dtype=np.int32)
+    for i in range(n):
+        nb = ci[ro[i]:ro[i + 1]]
+        used = set(colors[j] for j in nb if j != i and j < n and colors[j] >= 0)
+        c = 0
+        while c in used:
+            c += 1
+        colors[i] = c
+    num = int(colors.max()) + 1 if n else 0
+    return torch.from_numpy(colors), num
+
+
+# ---------------------------------------------------------------------- aggregation
+def _strength_weights(A):
+    """Symmetric edge weights 0.5*(|a_ij|+|a_ji|) scaled by diagonal
+    (reference: size2 selector edge weights,
+    src/aggregation/selectors/size2_selector.cu)."""
+    m = _csr(A) if A.block_dim == 1 else _block_norm_csr(A)
+    ad = np.abs(m.diagonal())
+    ad = np.where(ad > 0, ad, 1.0)
+    absm = abs(m.tocsr())
+    w = 0.5 * (absm + absm.T).tocsr()
+    # scale w_ij / sqrt(d_i d_j)
+    dinv = 1.0 / np.sqrt(ad)
+    w = sp.diags(dinv) @ w @ sp.diags(dinv)
+    w = w.tocsr()
+    w.setdiag(0.0)
+    w.eliminate_zeros()
+    return w
+
+
+def _block_norm_csr(A) -> sp.csr_matrix:
+    """Frobenius norm of each block -> scalar CSR (for selectors/coloring on
+    block matrices)."""
+    vals = _np(A.values)
+    norms = np.sqrt((vals.astype(np.float64) ** 2).sum(axis=(1, 2)))
+    # keep signs of diagonal-ish entries irrelevant; selectors use |.| anyway
+    return sp.csr_matrix((norms, _np(A.col_indices).astype(np.int64),
+                          _np(A.row_offsets).astype(np.int64)),
+                         shape=(A.n_rows, A.n_cols))
+
+
+def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
+                   seed: int = 0):
+    w = _strength_weights(A)
+    n = A.n_rows
+    agg = np.full(n, -1, dtype=np.int64)
+    rng = np.random.RandomState(seed)
+    tie = rng.rand(n)
+    next_id = 0
+    for _ in range(max_iterations):
+        un = np.nonzero(agg < 0)[0]
+        if un.size == 0:
+            break
+        # each unaggregated node proposes to its strongest unaggregated neighbor
+        prop = np.full(n, -1, dtype=np.int64)
+        for i in un:
+            s, e = w.indptr[i], w.indptr[i + 1]
+            best, bw = -1, 0.0
+            for k in range(s, e):
+                j = w.indices[k]
+                if agg[j] >= 0 or j == i:
+                    continue
+                wk = w.data[k]
+                if (wk > bw) or (wk == bw and best >= 0 and tie[j] > tie[best]):
+                    best, bw = j, wk
+            prop[i] = best
+        merged = False
+        for i in un:
+            j = prop[i]
+            if j >= 0 and prop[j] == i and i < j:
+                agg[i] = agg[j] = next_id
+                next_id += 1
+                merged = True
+        if not merged:
+            break
+    # leftovers: join strongest aggregated neighbor, else singleton
+    for i in np.nonzero(agg < 0)[0]:
+        s, e = w.indptr[i], w.indptr[i + 1]
+        best, bw = -1, 0.0
+        for k in range(s, e):
+            j = w.indices[k]
+            if agg[j] >= 0 and w.data[k] >= bw:
+                best, bw = j, w.data[k]
+        if best >= 0:
+            agg[i] = agg[best]
+        else:
+            agg[i] = next_id
+            next_id += 1
+    return torch.from_numpy(agg.astype(np.int32)), int(next_id)
+
+
+def galerkin_aggregation(A, aggregates, num_aggregates):
+    from ..matrix import CSRMatrix
+    agg = _np(aggregates).astype(np.int64)
+    n = A.n_rows
+    bd = A.block_dim
+    if bd == 1:
+        m = _csr(A)
+        P = sp.csr_matrix((np.ones(n), (np.arange(n), agg)),
+                          shape=(n, num_aggregates))
+        Ac = (P.T @ m @ P).tocsr()
+        Ac.sort_indices()
+        return CSRMatrix.from_scipy(Ac, dtype=A.dtype)
+    # block: accumulate block sums by (aggI, aggJ)
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    vals = _np(A.values)
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    I, J = agg[rows], agg[ci]
+    key = I * num_aggregates + J
+    order = np.argsort(key, kind="stable")
+    key_s = key[order]
+    uniq, start = np.unique(key_s, return_index=True)
+    out = np.add.reduceat(vals[order], start, axis=0)
+    ro_c = np.zeros(num_aggregates + 1, dtype=np.int32)
+    np.add.at(ro_c, (uniq // num_aggregates) + 1, 1)
+    ro_c = np.cumsum(ro_c).astype(np.int32)
+    return type(A).from_bsr(ro_c, (uniq % num_aggregates).astype(np.int32), out,
+                            n_cols=num_aggregates, dtype=A.dtype)
+
+
+def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):
+    agg = aggregates.to(torch.int64)
+    rc = torch.zeros(num_aggregates * block_dim, dtype=r.dtype)
+    if block_dim == 1:
+        rc.index_add_(0, agg, r.reshape(-1))
+    else:
+        idx = (agg[:, None] * block_dim +
+               torch.arange(block_dim, dtype=torch.int64)[None, :]).reshape(-1)
+        rc.index_add_(0, idx, r.reshape(-1))
+    return rc
+
+
+def prolongate_agg(x, xc, aggregates, block_dim: int = 1):
+    agg = aggregates.to(torch.int64)
+    if block_dim == 1:
+        x.reshape(-1).add_(xc.reshape(-1)[agg])
+    else:
+        idx = (agg[:, None] * block_dim +
+               torch.arange(block_dim, dtype=torch.int64)[None, :]).reshape(-1)
+        x.reshape(-1).add_(xc.reshape(-1)[idx])
+    return x
+
+
+# ---------------------------------------------------------------------- SpGEMM etc.
+def spgemm(A, B):
+    from ..matrix import CSRMatrix
+    C = (_csr(A) @ _csr(B)).tocsr()
+    C.sum_duplicates()
+    C.sort_indices()
+    return CSRMatrix.from_scipy(C, dtype=A.dtype)
+
+
+def transpose(A):
+    from ..matrix import CSRMatrix
+    return CSRMatrix.from_scipy(_csr(A).T.tocsr(), dtype=A.dtype)
+
+
+def galerkin_rap(R, A, P):
+    from ..matrix import CSRMatrix
+    C = (_csr(R) @ _csr(A) @ _csr(P)).tocsr()
+    C.sum_duplicates()
+    C.sort_indices()
+    return CSRMatrix.from_scipy(C, dtype=A.dtype)
+
+
+def truncate_rows(P, trunc_factor: float = 0.0, max_elements: int = -1):
+    """Drop |p| < factor*rowmax (and cap per-row count), rescale to preserve the
+    row sum (reference src/truncate.cu truncateAndScale_kernel)."""
+    from ..matrix import CSRMatrix
+    if trunc_factor <= 0.0 and max_elements < 0:
+        return P
+    m = _csr(P).copy()
+    n = m.shape[0]
+    data, indices, indptr = m.data, m.indices, m.indptr
+    new_data, new_idx, new_ptr = [], [], [0]
+    for i in range(n):
+        s, e = indptr[i], indptr[i + 1]
+        d, c = data[s:e], indices[s:e]
+        if d.size:
+            rs_old = d.sum()
+            keep = np.abs(d) >= trunc_factor * np.abs(d).max()
+            d2, c2 = d[keep], c[keep]
+            if max_elements >= 0 and d2.size > max_elements:
+                top = np.argsort(-np.abs(d2), kind="stable")[:max_elements]
+                top.sort()
+                d2, c2 = d2[top], c2[top]
+            rs_new = d2.sum()
+            if rs_new != 0 and rs_old != 0:
+                d2 = d2 * (rs_old / rs_new)
+            new_data.append(d2)
+            new_idx.append(c2)
+        new_ptr.append(new_ptr[-1] + (len(new_data[-1]) if d.size else 0))
+    out = sp.csr_matrix((np.concatenate(new_data) if new_data else np.zeros(0),
+                         np.concatenate(new_idx) if new_idx else np.zeros(0, dtype=int),
+                         np.asarray(new_ptr)), shape=m.shape)
+    return CSRMatrix.from_scipy(out, dtype=P.dtype)
+
+
+# ---------------------------------------------------------------------- DILU
+def dilu_setup(A, coloring):
+    """E_i = A_ii - sum_{color(j)<color(i)} A_ij Einv_j A_ji, color by color."""
+    bd = A.block_dim
+    colors = _np(coloring.colors).astype(np.int64)
+    if bd == 1:
+        m = _csr(A)
+        d = m.diagonal().astype(np.float64)
+        # B_ij = a_ij * a_ji where both exist
+        B = m.multiply(m.T).tocsr()
+        einv = np.zeros(A.n_rows, dtype=np.float64)
+        done = np.zeros(A.n_rows, dtype=np.float64)
+        for c in range(coloring.num_colors):
+            rows = np.nonzero(colors == c)[0]
+            e = d[rows] - (B[rows, :] @ (einv * done))
+            e = np.where(e != 0.0, e, 1.0)
+            einv[rows] = 1.0 / e
+            done[rows] = 1.0
+        return torch.from_numpy(einv).to(A.dtype)
+    # block version (host reference, O(nnz*b^3) python loop on small tests)
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    vals = _np(A.values).astype(np.float64)
+    D = _np(extract_diagonal(A)).astype(np.float64).copy()
+    # transpose value lookup
+    AT = transpose_block_index(A)
+    einv = np.zeros_like(D)
+    for c in range(coloring.num_colors):
+        rows = np.nonzero(colors == c)[0]
+        for i in rows:
+            E = D[i].copy()
+            for k in range(ro[i], ro[i + 1]):
+                j = ci[k]
+                if j != i and colors[j] < c and AT[k] >= 0:
+                    E -= vals[k] @ einv[j] @ vals[AT[k]]
+            if abs(np.linalg.det(E)) < 1e-300:
+                E = np.eye(A.block_dim)
+            einv[i] = np.linalg.inv(E)
+    return torch.from_numpy(einv).to(A.dtype)
+
+
+def transpose_block_index(A) -> np.ndarray:
+    """For each nz k=(i,j), index of the (j,i) entry in values (-1 if absent)."""
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    n = A.n_rows
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    out = np.full(ci.size, -1, dtype=np.int64)
+    pos = {}
+    for k in range(ci.size):
+        pos[(rows[k], ci[k])] = k
+    for k in range(ci.size):
+        out[k] = pos.get((ci[k], rows[k]), -1)
+    return out
+
+
+def dilu_solve(A, Einv, coloring, r, relaxation, x):
+    bd = A.block_dim
+    colors = _np(coloring.colors).astype(np.int64)
+    m = A.to_scipy()
+    rv = _np(r).reshape(-1).astype(np.float64)
+    w = np.zeros_like(rv)
+    ei = _np(Einv).astype(np.float64)
+    nc = coloring.num_colors
+    def rowsb(rows):
+        return (rows[:, None] * bd + np.arange(bd)[None, :]).reshape(-1)
+    # forward: valid coloring => same-color off-diagonals absent, and the
+    # diagonal contributes 0 while w[rows] == 0
+    for c in range(nc):
+        rows = np.nonzero(colors == c)[0]
+        rb = rowsb(rows)
+        tmp = rv[rb] - (m @ w)[rb]
+        if bd == 1:
+            w[rb] = ei[rows] * tmp
+        else:
+            w[rb] = np.matmul(ei[rows], tmp.reshape(-1, bd, 1)).reshape(-1)
+    # backward: z_i = w_i - Einv_i * sum_{color(j)>c} A_ij z_j
+    z = w.copy()
+    later = np.zeros_like(w)
+    for c in range(nc - 1, -1, -1):
+        rows = np.nonzero(colors == c)[0]
+        rb = rowsb(rows)
+        s = (m @ later)[rb]
+        if bd == 1:
+            z[rb] = w[rb] - ei[rows] * s
+        else:
+            z[rb] = w[rb] - np.matmul(ei[rows], s.reshape(-1, bd, 1)).reshape(-1)
+        later[rb] = z[rb]
+    _np(x).reshape(-1)[:] += relaxation * z
+    return x
+
+
+# ---------------------------------------------------------------------- dense
+def dense_solve(Ainv, b, x):
+    x.reshape(-1).copy_(
+        (Ainv.to(torch.float64) @ b.reshape(-1).to(torch.float64)).to(x.dtype))
+    return x
+
+
+# ---------------------------------------------------------------------- ILU(0)
+def ilu0_setup(A, coloring):
+    """ILU(0) factorization in color order (scalar). Returns factored values
+    aligned with A's CSR structure. Host reference for the per-color GPU
+    kernels (reference src/solvers/multicolor_ilu_solver.cu)."""
+    assert A.block_dim == 1, "block ILU: use DILU or scalar path"
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    vals = _np(A.values).astype(np.float64).copy()
+    colors = _np(coloring.colors).astype(np.int64)
+    n = A.n_rows
+    # color-order position of each row
+    order = np.lexsort((np.arange(n), colors))
+    pos = np.empty(n, dtype=np.int64)
+    pos[order] = np.arange(n)
+    # pattern lookup
+    lut = {}
+    for i in range(n):
+        for k in range(ro[i], ro[i + 1]):
+            lut[(i, ci[k])] = k
+    for i in order:
+        # eliminate with pivots k of smaller color-position, ascending
+        row_ks = sorted((pos[ci[k]], k) for k in range(ro[i], ro[i + 1]))
+        for pk, kidx in row_ks:
+            k = ci[kidx]
+            if pk >= pos[i]:
+                continue
+            ukk = vals[lut[(k, k)]]
+            if ukk == 0.0:
+                ukk = 1.0
+            vals[kidx] /= ukk
+            lik = vals[kidx]
+            for k2 in range(ro[k], ro[k + 1]):
+                j = ci[k2]
+                if pos[j] > pk and (i, j) in lut:
+                    vals[lut[(i, j)]] -= lik * vals[k2]
+    return torch.from_numpy(vals).to(A.dtype)
+
+
+def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    vals = _np(factors).astype(np.float64)
+    colors = _np(coloring.colors).astype(np.int64)
+    n = A.n_rows
+    order = np.lexsort((np.arange(n), colors))
+    pos = np.empty(n, dtype=np.int64)
+    pos[order] = np.arange(n)
+    rv = _np(r).reshape(-1).astype(np.float64)
+    y = np.zeros(n)
+    for i in order:
+        s = rv[i]
+        for k in range(ro[i], ro[i + 1]):
+            j = ci[k]
+            if pos[j] < pos[i]:
+                s -= vals[k] * y[j]
+        y[i] = s
+    z = np.zeros(n)
+    for i in order[::-1]:
+        s = y[i]
+        d = 1.0
+        for k in range(ro[i], ro[i + 1]):
+            j = ci[k]
+            if pos[j] > pos[i]:
+                s -= vals[k] * z[j]
+            elif j == i:
+                d = vals[k]
+        z[i] = s / (d if d != 0.0 else 1.0)
+    _np(x).reshape(-1)[:] += relaxation * z
+    return x
+
+
+# ---------------------------------------------------------------------- classical
+def strength_ahat(A, theta: float = 0.25, max_row_sum: float = 1.1):
+    """AHAT strength mask aligned with A's CSR entries: strong iff
+    |a_ij| >= theta * max_{k!=i}|a_ik| (reference
+    src/classical/strength/strength_base.cu; SURVEY.md 'classic |a_ij|>=th*max').
+    Rows whose |row sum| exceeds max_row_sum*|a_ii| are made all-weak."""
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    v = _np(A.values).astype(np.float64)
+    n = A.n_rows
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    off = rows != ci
+    absv = np.abs(v)
+    rowmax = np.zeros(n)
+    np.maximum.at(rowmax, rows[off], absv[off])
+    strong = off & (absv >= theta * rowmax[rows]) & (rowmax[rows] > 0)
+    if max_row_sum < 1.0:
+        rs = np.zeros(n)
+        np.add.at(rs, rows, v)
+        d = np.abs(_np(extract_diagonal(A)))
+        weak_rows = np.abs(rs) > max_row_sum * np.where(d > 0, d, 1.0)
+        strong &= ~weak_rows[rows]
+    return torch.from_numpy(strong)
+
+
+def pmis_select(A, S):
+    """PMIS C/F splitting (reference src/classical/selectors/pmis.cu:
+    random-weight independent-set rounds). Returns (cf_map int32: >=0 coarse
+    index for C, -1 for F, and num_coarse)."""
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    n = A.n_rows
+    strong = _np(S)
+    # symmetrized strong adjacency (S union S^T) as CSR of indices
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    sr, sc = rows[strong], ci[strong]
+    adj = sp.csr_matrix((np.ones(2 * sr.size), (np.concatenate([sr, sc]),
+                                                np.concatenate([sc, sr]))),
+                        shape=(n, n))
+    adj.sum_duplicates()
+    # lambda_i = number of points strongly influenced by i (= S^T row count)
+    lam = np.zeros(n)
+    np.add.at(lam, sc, 1.0)
+    rng = np.random.RandomState(10007)
+    w = lam + rng.rand(n)
+    state = np.zeros(n, dtype=np.int8)  # 0 undecided, 1 C, -1 F
+    # isolated points (no strong edges at all): F (smoother-only rows)
+    deg = np.diff(adj.indptr)
+    state[deg == 0] = -1
+    while (state == 0).any():
+        und = state == 0
+        # local max among undecided neighbors
+        newC = []
+        for i in np.nonzero(und)[0]:
+            nb = adj.indices[adj.indptr[i]:adj.indptr[i + 1]]
+            nb = nb[state[nb] == 0]
+            if all(w[i] > w[j] for j in nb if j != i):
+                newC.append(i)
+        if not newC:
+            newC = [np.nonzero(und)[0][0]]
+        state[np.asarray(newC, dtype=np.int64)] = 1
+        # neighbors of new C become F
+        for i in newC:
+            nb = adj.indices[adj.indptr[i]:adj.indptr[i + 1]]
+            nb = nb[state[nb] == 0]
+            state[nb] = -1
+    cf = np.full(n, -1, dtype=np.int32)
+    cpts = np.nonzero(state == 1)[0]
+    cf[cpts] = np.arange(cpts.size, dtype=np.int32)
+    return torch.from_numpy(cf), int(cpts.size)
+
+
+def interp_d1(A, S, cf_map, num_coarse):
+    """Distance-1 (direct) interpolation with pos/neg splitting (reference
+    src/classical/interpolators/distance1.cu):
+      w_ij = -alpha_i * a_ij / a_ii,  alpha = sum(neg a) / sum(neg a over C),
+    likewise beta for positive entries; positives lumped into the diagonal
+    when no positive C connection exists."""
+    from ..matrix import CSRMatrix
+    ro = _np(A.row_offsets).astype(np.int64)
+    ci = _np(A.col_indices).astype(np.int64)
+    v = _np(A.values).astype(np.float64)
+    strong = _np(S)
+    cf = _np(cf_map).astype(np.int64)
+    n = A.n_rows
+    Pr, Pc, Pv = [], [], []
+    indptr = [0]
+    for i in range(n):
+        if cf[i] >= 0:
+            Pc.append(cf[i]); Pv.append(1.0)
+            indptr.append(indptr[-1] + 1)
+            continue
+        s, e = ro[i], ro[i + 1]
+        diag = 0.0
+        neg_all = pos_all = neg_c = pos_c = 0.0
+        entries = []
+        for k in range(s, e):
+            j, a = ci[k], v[k]
+            if j == i:
+                diag = a
+                continue
+            if a < 0:
+                neg_all += a
+            else:
+                pos_all += a
+            if strong[k] and cf[j] >= 0:
+                if a < 0:
+                    neg_c += a
+                else:
+                    pos_c += a
+                entries.append((cf[j], a))
+        if not entries or diag == 0.0:
+            indptr.append(indptr[-1])
+            continue
+        if pos_c == 0.0:
+            diag += pos_all
+            pos_all = 0.0
+        alpha = neg_all / neg_c if neg_c != 0.0 else 0.0
+        beta = pos_all / pos_c if pos_c != 0.0 else 0.0
+        acc = {}
+        for jc, a in entries:
+            wgt = -(alpha if a < 0 else beta) * a / diag
+            acc[jc] = acc.get(jc, 0.0) + wgt
+        for jc in sorted(acc):
+            Pc.append(jc); Pv.append(acc[jc])
+        indptr.append(indptr[-1] + len(acc))
+    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
+                       np.asarray(indptr)), shape=(n, num_coarse))
+    return CSRMatrix.from_scipy(P, dtype=A.dtype)

@@ -1,0 +1,102 @@
+"""Synthetic problem generators (reference: embedded CUSP
+``gallery::poisson5/7/9/27pt`` used throughout the tests, and
+``AMGX_generate_distributed_poisson_7pt``, src/amgx_c.cu:4566-4731)."""
+
+from __future__ import annotations
+
+import numpy as np
+import scipy.sparse as sp
+import torch
+
+from .matrix import CSRMatrix
+
+
+def poisson_2d(nx: int, ny: int, stencil: int = 5, device="cpu",
+               dtype=torch.float64) -> CSRMatrix:
+    ex = np.ones(nx)
+    ey = np.ones(ny)
+    Tx = sp.diags([-ex[:-1], 2 * ex, -ex[:-1]], [-1, 0, 1])
+    Ty = sp.diags([-ey[:-1], 2 * ey, -ey[:-1]], [-1, 0, 1])
+    Ix, Iy = sp.identity(nx), sp.identity(ny)
+    A = sp.kron(Iy, Tx) + sp.kron(Ty, Ix)
+    if stencil == 9:
+        D = sp.diags([-ex[:-1], np.zeros(nx), -ex[:-1]], [-1, 0, 1])
+        Dy = sp.diags([-ey[:-1], np.zeros(ny), -ey[:-1]], [-1, 0, 1])
+        A = A + 0.5 * sp.kron(Dy, D)
+        A = A + sp.identity(A.shape[0]) * 0  # keep structure
+    return CSRMatrix.from_scipy(A.tocsr(), device=device, dtype=dtype)
+
+
+def poisson_3d(nx: int, ny: int, nz: int, device="cpu",
+               dtype=torch.float64) -> CSRMatrix:
+    """7-point 3D Poisson: diag 6, neighbors -1 (matches
+    AMGX_generate_distributed_poisson_7pt semantics, src/amgx_c.cu:4566)."""
+    def lap1(n):
+        e = np.ones(n)
+        return sp.diags([-e[:-1], 2 * e, -e[:-1]], [-1, 0, 1])
+    Ix, Iy, Iz = sp.identity(nx), sp.identity(ny), sp.identity(nz)
+    A = (sp.kron(sp.kron(Iz, Iy), lap1(nx))
+         + sp.kron(sp.kron(Iz, lap1(ny)), Ix)
+         + sp.kron(sp.kron(lap1(nz), Iy), Ix))
+    return CSRMatrix.from_scipy(A.tocsr(), device=device, dtype=dtype)
+
+
+def poisson_3d_local(nx: int, ny: int, nz: int, rank: int, world: int):
+    """Rank-local slab rows (z-partition) of a global nx*ny*(nz*world) 7-pt
+    Poisson, as (row_offsets, col_indices GLOBAL, values, row_start) numpy.
+    One process per GPU partitioning for the distributed path
+    (reference AMGX_generate_distributed_poisson_7pt px*py*pz grid)."""
+    NZg = nz * world
+    n_local = nx * ny * nz
+    row_start = rank * n_local
+    ro = [0]
+    cols = []
+    vals = []
+    for z in range(nz):
+        gz = rank * nz + z
+        for y in range(ny):
+            for x in range(nx):
+                gid = (gz * ny + y) * nx + x
+                row = []
+                if gz > 0:
+                    row.append((gid - nx * ny, -1.0))
+                if y > 0:
+                    row.append((gid - nx, -1.0))
+                if x > 0:
+                    row.append((gid - 1, -1.0))
+                row.append((gid, 6.0))
+                if x < nx - 1:
+                    row.append((gid + 1, -1.0))
+                if y < ny - 1:
+                    row.append((gid + nx, -1.0))
+                if gz < NZg - 1:
+                    row.append((gid + nx * ny, -1.0))
+                for c, v in row:
+                    cols.append(c)
+                    vals.append(v)
+                ro.append(len(cols))
+    return (np.asarray(ro, dtype=np.int64), np.asarray(cols, dtype=np.int64),
+            np.asarray(vals, dtype=np.float64), row_start)
+
+
+def block_laplacian(nx: int, ny: int, block_dim: int = 4, device="cpu",
+                    dtype=torch.float64, seed: int = 0) -> CSRMatrix:
+    """Block-CSR test system: 2D Poisson coupling with SPD random blocks on
+    the diagonal (driver config #4 class: block-4 coupled system)."""
+    A = poisson_2d(nx, ny).to_scipy().tocsr()
+    rng = np.random.RandomState(seed)
+    n = A.shape[0]
+    nnz = A.nnz
+    blocks = np.zeros((nnz, block_dim, block_dim))
+    rows = np.repeat(np.arange(n), np.diff(A.indptr))
+    for k in range(nnz):
+        i, j, v = rows[k], A.indices[k], A.data[k]
+        if i == j:
+            Q = rng.randn(block_dim, block_dim) * 0.1
+            blocks[k] = v * (np.eye(block_dim) + Q @ Q.T)
+        else:
+            blocks[k] = v * (np.eye(block_dim)
+                             + 0.05 * rng.randn(block_dim, block_dim))
+    return CSRMatrix.from_bsr(A.indptr.astype(np.int32),
+                              A.indices.astype(np.int32), blocks,
+                              n_cols=n, device=device, dtype=dtype)

@@ -1,0 +1,255 @@
+"""Solver base class, convergence objects and the solver factory registry.
+
+Mirrors the reference ``Solver<TConfig>`` services (include/solvers/solver.h:22,
+src/solvers/solver.cu:333,586): setup/solve lifecycle, residual monitoring,
+norm machinery, convergence delegation, nested preconditioner composition, and
+the name->factory registry (reference src/core.cu:596-628) that makes every
+algorithm addressable from JSON configs.
+"""
+
+from __future__ import annotations
+
+import math
+import time
+from typing import Dict, Optional, Type
+
+import torch
+
+from .. import ops
+from ..config import ConfigScope
+from ..resources import Resources, default_resources
+
+SOLVER_REGISTRY: Dict[str, Type["Solver"]] = {}
+
+
+def register_solver(name: str):
+    def deco(cls):
+        SOLVER_REGISTRY[name] = cls
+        cls.solver_name = name
+        return cls
+    return deco
+
+
+def create_solver(name_or_scope, scope: Optional[ConfigScope] = None,
+                  resources: Optional[Resources] = None) -> "Solver":
+    """Create a solver by registry name (reference SolverFactory::allocate)."""
+    if isinstance(name_or_scope, ConfigScope):
+        scope = name_or_scope
+        name = scope.get("solver")
+    else:
+        name = name_or_scope
+        if scope is None:
+            scope = ConfigScope(None, {})
+    if resources is None:
+        resources = default_resources()
+    cls = SOLVER_REGISTRY.get(name)
+    if cls is None:
+        raise KeyError(f"unknown solver {name!r}; known: {sorted(SOLVER_REGISTRY)}")
+    return cls(scope, resources)
+
+
+# ------------------------------------------------------------------- convergence
+class Convergence:
+    """Reference include/convergence/convergence.h:64-101 + src/convergence/."""
+
+    def __init__(self, kind: str, tolerance: float, alt_rel_tol: float = -1.0):
+        self.kind = kind
+        self.tol = tolerance
+        self.alt_rel_tol = alt_rel_tol
+        self.ini_norm = None
+
+    def set_initial(self, nrm: float):
+        self.ini_norm = nrm
+
+    def converged(self, nrm: float) -> bool:
+        if self.kind == "ABSOLUTE":
+            return nrm <= self.tol
+        if self.kind in ("RELATIVE_INI", "RELATIVE_INI_CORE"):
+            return nrm <= self.tol * (self.ini_norm if self.ini_norm else 1.0)
+        if self.kind == "RELATIVE_MAX":
+            ref = max(self.ini_norm or 0.0, 1e-300)
+            return nrm <= self.tol * ref
+        if self.kind == "COMBINED_REL_INI_ABS":
+            rel = nrm <= self.tol * (self.ini_norm if self.ini_norm else 1.0)
+            ab = self.alt_rel_tol > 0 and nrm <= self.alt_rel_tol
+            return rel or ab
+        raise ValueError(f"unknown convergence {self.kind}")
+
+
+class SolveStatus:
+    SUCCESS = 0
+    DIVERGED = 1
+    NOT_CONVERGED = 2
+
+    def __init__(self):
+        self.status = SolveStatus.NOT_CONVERGED
+        self.iterations = 0
+        self.residuals = []
+        self.setup_time = 0.0
+        self.solve_time = 0.0
+
+    @property
+    def converged(self):
+        return self.status == SolveStatus.SUCCESS
+
+    def __repr__(self):  # pragma: no cover
+        return (f"SolveStatus(conv={self.converged}, it={self.iterations}, "
+                f"res={self.residuals[-1] if self.residuals else None})")
+
+
+# ------------------------------------------------------------------------ solver
+class Solver:
+    """Base solver. Subclasses implement solver_setup / solve_init /
+    solve_iteration / solve_finalize (reference include/solvers/solver.h:152-156)."""
+
+    solver_name = "?"
+    is_smoother = False   # smoothers skip convergence monitoring by default
+
+    def __init__(self, scope: ConfigScope, resources: Resources):
+        self.scope = scope
+        self.res = resources
+        self.A = None
+        self.max_iters = scope.get("max_iters")
+        self.monitor_residual = bool(scope.get("monitor_residual"))
+        self.store_res_history = bool(scope.get("store_res_history"))
+        self.print_solve_stats = bool(scope.get("print_solve_stats"))
+        self.norm = scope.get("norm")
+        self.convergence = Convergence(scope.get("convergence"),
+                                       scope.get("tolerance"),
+                                       scope.get("alt_rel_tolerance"))
+        self.relaxation_factor = scope.get("relaxation_factor")
+        self.status = SolveStatus()
+
+    # -- lifecycle -----------------------------------------------------------
+    def setup(self, A):
+        t0 = time.perf_counter()
+        self.A = A
+        self.solver_setup()
+        self.res.synchronize() if self.res.is_cuda else None
+        self.status.setup_time = time.perf_counter() - t0
+
+    def resetup(self, A):
+        """Values changed, structure identical (reference AMGX_solver_resetup)."""
+        self.setup(A)
+
+    def solver_setup(self):
+        pass
+
+    # -- norms ---------------------------------------------------------------
+    def compute_norm(self, r: torch.Tensor) -> float:
+        if self.norm == "L2":
+            nrm = ops.nrm2(r)
+        elif self.norm == "L1":
+            nrm = ops.nrm1(r)
+        elif self.norm == "LMAX":
+            nrm = ops.nrmmax(r)
+        else:
+            nrm = ops.nrm2(r)
+        if self.A is not None and getattr(self.A, "manager", None) is not None:
+            nrm = self.A.manager.global_norm(nrm, self.norm)
+        return nrm
+
+    def dot(self, x, y) -> float:
+        d = ops.dot(x, y)
+        if self.A is not None and getattr(self.A, "manager", None) is not None:
+            d = self.A.manager.global_sum(d)
+        return d
+
+    # -- solve ---------------------------------------------------------------
+    def solve(self, b: torch.Tensor, x: torch.Tensor,
+              zero_initial_guess: bool = False) -> SolveStatus:
+        """Reference Solver::solve (src/solvers/solver.cu:586): initial
+        residual + norm, solve_init, iterate until converged/max_iters."""
+        st = self.status = SolveStatus()
+        t0 = time.perf_counter()
+        if zero_initial_guess:
+            x.zero_()
+        monitoring = self.monitor_residual
+        if monitoring:
+            r = ops.residual(self.A, x, b)
+            nrm = self.compute_norm(r)
+            self.convergence.set_initial(nrm)
+            st.residuals.append(nrm)
+            if self.print_solve_stats and self.res.rank == 0:
+                print(f"           iter      residual   rate")
+                print(f"           ----------------------------")
+                print(f"            Ini {nrm:14.6e}")
+            if self.convergence.converged(nrm) and self.convergence.kind != "RELATIVE_INI":
+                st.status = SolveStatus.SUCCESS
+                st.solve_time = time.perf_counter() - t0
+                return st
+            if nrm == 0.0:
+                st.status = SolveStatus.SUCCESS
+                st.solve_time = time.perf_counter() - t0
+                return st
+        self.solve_init(b, x, zero_initial_guess)
+        for it in range(self.max_iters):
+            done = self.solve_iteration(b, x)
+            st.iterations = it + 1
+            if monitoring:
+                nrm = self.last_residual_norm(b, x)
+                st.residuals.append(nrm)
+                if self.print_solve_stats and self.res.rank == 0:
+                    rate = (st.residuals[-1] / st.residuals[-2]
+                            if st.residuals[-2] else 0.0)
+                    print(f"           {it:4d} {nrm:14.6e}  {rate:6.4f}")
+                if self.convergence.converged(nrm):
+                    st.status = SolveStatus.SUCCESS
+                    break
+                if not math.isfinite(nrm):
+                    st.status = SolveStatus.DIVERGED
+                    break
+            if done:
+                if not monitoring:
+                    st.status = SolveStatus.SUCCESS
+                break
+        self.solve_finalize(b, x)
+        if self.res.is_cuda:
+            self.res.synchronize()
+        st.solve_time = time.perf_counter() - t0
+        return st
+
+    # default residual-norm recomputation; Krylov solvers override with their
+    # internally tracked norm to avoid an extra SpMV
+    def last_residual_norm(self, b, x) -> float:
+        r = ops.residual(self.A, x, b)
+        return self.compute_norm(r)
+
+    def solve_init(self, b, x, zero_initial_guess):
+        pass
+
+    def solve_iteration(self, b, x) -> bool:
+        raise NotImplementedError
+
+    def solve_finalize(self, b, x):
+        pass
+
+    # -- helpers --------------------------------------------------------------
+    def make_preconditioner(self, role: str = "preconditioner",
+                            default_name: str = "NOSOLVER"):
+        name, sub = self.scope.sub_solver(role, default_name)
+        if name is None or name == "NOSOLVER":
+            return None
+        solver = create_solver(name, sub, self.res)
+        return solver
+
+    def new_vec(self, like: torch.Tensor) -> torch.Tensor:
+        return torch.zeros_like(like)
+
+
+@register_solver("NOSOLVER")
+@register_solver("DUMMY")
+class DummySolver(Solver):
+    """Identity preconditioner (reference src/solvers/dummy_solver.cu)."""
+    is_smoother = True
+
+    def solve(self, b, x, zero_initial_guess=False):
+        if zero_initial_guess:
+            x.zero_()
+        x.copy_(b.reshape(x.shape))
+        st = SolveStatus()
+        st.status = SolveStatus.SUCCESS
+        return st
+
+    def solve_iteration(self, b, x):
+        return True

@@ -1,0 +1,340 @@
+"""Krylov solvers: CG/PCG/PCGF, BiCGStab/PBiCGStab, GMRES/FGMRES, IDR(s).
+
+Reference: src/solvers/{cg,pcg,pcgf,bicgstab,pbicgstab,gmres,fgmres,idr}_solver.cu.
+Each solve_iteration matches the reference's operation order so iteration
+counts are comparable (iteration parity is the primary
+correctness-of-algorithm signal).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+
+from .. import ops
+from .base import Solver, register_solver
+
+
+@register_solver("PCG")
+@register_solver("CG")
+class PCGSolver(Solver):
+    """Preconditioned CG (reference src/solvers/pcg_solver.cu:109-189).
+    Plain CG is PCG with no preconditioner (NOSOLVER)."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.flexible = bool(scope.get("pcg_flexible"))
+
+    def solver_setup(self):
+        self.precond = self.make_preconditioner()
+        if self.precond is not None:
+            self.precond.setup(self.A)
+
+    def solve_init(self, b, x, zero_initial_guess):
+        self.r = ops.residual(self.A, x, b)
+        if self.precond is not None:
+            self.z = self.new_vec(self.r)
+            self.precond.solve(self.r, self.z, zero_initial_guess=True)
+        else:
+            self.z = self.r.clone()
+        self.p = self.z.clone()
+        self.Ap = self.new_vec(self.r)
+        self.rz = self.dot(self.r, self.z)
+        self._rnorm = None
+
+    def solve_iteration(self, b, x):
+        ops.spmv(self.A, self.p, self.Ap)
+        pAp = self.dot(self.p, self.Ap)
+        if pAp == 0.0:
+            return True
+        alpha = self.rz / pAp
+        ops.axpy(x, self.p, alpha)
+        ops.axpy(self.r, self.Ap, -alpha)
+        self._rnorm = self.compute_norm(self.r)
+        if self.precond is not None:
+            if self.flexible:
+                z_old = self.z.clone()
+            self.precond.solve(self.r, self.z, zero_initial_guess=True)
+            if self.flexible:
+                # Polak-Ribiere beta (reference src/solvers/pcgf_solver.cu)
+                rz_new = self.dot(self.r, self.z)
+                beta = (rz_new - self.dot(self.r, z_old)) / self.rz if self.rz else 0.0
+                self.rz = rz_new
+            else:
+                rz_new = self.dot(self.r, self.z)
+                beta = rz_new / self.rz if self.rz else 0.0
+                self.rz = rz_new
+        else:
+            self.z.copy_(self.r)
+            rz_new = self.dot(self.r, self.z)
+            beta = rz_new / self.rz if self.rz else 0.0
+            self.rz = rz_new
+        ops.axpby(self.p, self.z, 1.0, beta)
+        return False
+
+    def last_residual_norm(self, b, x):
+        return self._rnorm if self._rnorm is not None else super().last_residual_norm(b, x)
+
+
+@register_solver("PCGF")
+class PCGFSolver(PCGSolver):
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.flexible = True
+
+
+@register_solver("PBICGSTAB")
+@register_solver("BICGSTAB")
+class BiCGStabSolver(Solver):
+    """(Preconditioned) BiCGStab (reference src/solvers/pbicgstab_solver.cu)."""
+
+    def solver_setup(self):
+        self.precond = self.make_preconditioner()
+        if self.precond is not None:
+            self.precond.setup(self.A)
+
+    def _apply_M(self, v, out):
+        if self.precond is None:
+            out.copy_(v)
+        else:
+            self.precond.solve(v, out, zero_initial_guess=True)
+
+    def solve_init(self, b, x, zero_initial_guess):
+        self.r = ops.residual(self.A, x, b)
+        self.r_tld = self.r.clone()
+        self.p = self.new_vec(self.r)
+        self.v = self.new_vec(self.r)
+        self.ph = self.new_vec(self.r)
+        self.sh = self.new_vec(self.r)
+        self.t = self.new_vec(self.r)
+        self.rho_old = self.alpha = self.omega = 1.0
+        self._rnorm = None
+
+    def solve_iteration(self, b, x):
+        rho = self.dot(self.r_tld, self.r)
+        if rho == 0.0:
+            return True
+        if self.omega == 0.0:
+            return True
+        beta = (rho / self.rho_old) * (self.alpha / self.omega)
+        # p = r + beta*(p - omega*v)
+        ops.axpy(self.p, self.v, -self.omega)
+        ops.axpby(self.p, self.r, 1.0, beta)
+        self._apply_M(self.p, self.ph)
+        ops.spmv(self.A, self.ph, self.v)
+        denom = self.dot(self.r_tld, self.v)
+        if denom == 0.0:
+            return True
+        self.alpha = rho / denom
+        ops.axpy(self.r, self.v, -self.alpha)  # s = r - alpha v (in r)
+        snorm = self.compute_norm(self.r)
+        if self.convergence.converged(snorm):
+            ops.axpy(x, self.ph, self.alpha)
+            self._rnorm = snorm
+            self.rho_old = rho
+            return True
+        self._apply_M(self.r, self.sh)
+        ops.spmv(self.A, self.sh, self.t)
+        tt = self.dot(self.t, self.t)
+        self.omega = self.dot(self.t, self.r) / tt if tt != 0.0 else 0.0
+        ops.axpy(x, self.ph, self.alpha)
+        ops.axpy(x, self.sh, self.omega)
+        ops.axpy(self.r, self.t, -self.omega)
+        self._rnorm = self.compute_norm(self.r)
+        self.rho_old = rho
+        return False
+
+    def last_residual_norm(self, b, x):
+        return self._rnorm if self._rnorm is not None else super().last_residual_norm(b, x)
+
+
+@register_solver("FGMRES")
+@register_solver("GMRES")
+class FGMRESSolver(Solver):
+    """Right-preconditioned flexible GMRES with restart (reference
+    src/solvers/fgmres_solver.cu: Arnoldi + host-side Givens on the small
+    Hessenberg; gmres_n_restart default 20, src/core.cu:390)."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.restart = scope.get("gmres_n_restart")
+
+    def solver_setup(self):
+        self.precond = self.make_preconditioner()
+        if self.precond is not None:
+            self.precond.setup(self.A)
+
+    def _apply_M(self, v, out):
+        if self.precond is None:
+            out.copy_(v)
+        else:
+            self.precond.solve(v, out, zero_initial_guess=True)
+
+    def solve_init(self, b, x, zero_initial_guess):
+        m = self.restart
+        self.V = []          # Krylov basis
+        self.Z = []          # preconditioned vectors (flexible)
+        self.H = torch.zeros(m + 1, m, dtype=torch.float64)
+        self.cs = torch.zeros(m, dtype=torch.float64)
+        self.sn = torch.zeros(m, dtype=torch.float64)
+        self.g = torch.zeros(m + 1, dtype=torch.float64)
+        self._restart_init(b, x)
+
+    def _restart_init(self, b, x):
+        r = ops.residual(self.A, x, b)
+        beta = self.compute_norm(r) if self.norm == "L2" else math.sqrt(max(self.dot(r, r), 0.0))
+        beta = math.sqrt(max(self.dot(r, r), 0.0))
+        self.beta = beta
+        self.j = 0
+        self.V = [r / beta if beta > 0 else r]
+        self.Z = []
+        self.g.zero_()
+        self.g[0] = beta
+        self._rnorm = beta
+
+    def solve_iteration(self, b, x):
+        j = self.j
+        m = self.restart
+        vj = self.V[j]
+        z = self.new_vec(vj)
+        self._apply_M(vj, z)
+        self.Z.append(z)
+        w = self.new_vec(vj)
+        ops.spmv(self.A, z, w)
+        # modified Gram-Schmidt
+        for i in range(j + 1):
+            hij = self.dot(w, self.V[i])
+            self.H[i, j] = hij
+            ops.axpy(w, self.V[i], -hij)
+        hnext = math.sqrt(max(self.dot(w, w), 0.0))
+        self.H[j + 1, j] = hnext
+        # apply stored Givens rotations to column j
+        for i in range(j):
+            t = self.cs[i] * self.H[i, j] + self.sn[i] * self.H[i + 1, j]
+            self.H[i + 1, j] = -self.sn[i] * self.H[i, j] + self.cs[i] * self.H[i + 1, j]
+            self.H[i, j] = t
+        # new rotation
+        denom = math.hypot(float(self.H[j, j]), hnext)
+        if denom == 0.0:
+            self._update_x(x, j)
+            return True
+        self.cs[j] = self.H[j, j] / denom
+        self.sn[j] = hnext / denom
+        self.H[j, j] = denom
+        self.H[j + 1, j] = 0.0
+        self.g[j + 1] = -self.sn[j] * self.g[j]
+        self.g[j] = self.cs[j] * self.g[j]
+        self._rnorm = abs(float(self.g[j + 1]))
+        lucky = hnext == 0.0
+        converged = self.convergence.converged(self._rnorm)
+        if converged or lucky or j + 1 == m:
+            self._update_x(x, j)
+            if not (converged or lucky):
+                self._restart_init(b, x)   # restart
+            return converged or lucky
+        self.V.append(w / hnext)
+        self.j += 1
+        return False
+
+    def _update_x(self, x, j):
+        # back-substitute y from the j+1 x j+1 triangular system
+        y = torch.zeros(j + 1, dtype=torch.float64)
+        for i in range(j, -1, -1):
+            s = float(self.g[i]) - float(self.H[i, i + 1:j + 1] @ y[i + 1:j + 1])
+            y[i] = s / float(self.H[i, i]) if float(self.H[i, i]) != 0.0 else 0.0
+        for i in range(j + 1):
+            ops.axpy(x, self.Z[i], float(y[i]))
+
+    def last_residual_norm(self, b, x):
+        return self._rnorm
+
+
+@register_solver("IDR")
+@register_solver("IDRMSYNC")
+class IDRSolver(Solver):
+    """IDR(s) (reference src/solvers/idr_solver.cu; subspace_dim_s=8 default,
+    src/core.cu:393). Implemented as the standard IDR(s) biorthogonal variant."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.s = scope.get("subspace_dim_s")
+
+    def solver_setup(self):
+        self.precond = self.make_preconditioner()
+        if self.precond is not None:
+            self.precond.setup(self.A)
+
+    def _apply_M(self, v, out):
+        if self.precond is None:
+            out.copy_(v)
+        else:
+            self.precond.solve(v, out, zero_initial_guess=True)
+
+    def solve_init(self, b, x, zero_initial_guess):
+        s = self.s
+        n = b.numel()
+        self.r = ops.residual(self.A, x, b)
+        g = torch.Generator().manual_seed(1234)
+        P = torch.randn(s, n, generator=g, dtype=torch.float64)
+        # orthonormalize rows (host QR on small s x n is overkill; MGS)
+        for i in range(s):
+            for k in range(i):
+                P[i] -= (P[i] @ P[k]) * P[k]
+            P[i] /= torch.linalg.vector_norm(P[i])
+        self.P = P.to(self.r.dtype).to(self.r.device)
+        self.G = [self.new_vec(self.r) for _ in range(s)]
+        self.U = [self.new_vec(self.r) for _ in range(s)]
+        self.M = torch.eye(s, dtype=torch.float64)
+        self.omega = 1.0
+        self._rnorm = None
+
+    def solve_iteration(self, b, x):
+        s = self.s
+        f = torch.tensor([self.dot(self.P[i], self.r) for i in range(s)],
+                         dtype=torch.float64)
+        for k in range(s):
+            # solve lower-triangular M[k:,k:] c = f[k:]
+            c = torch.linalg.solve_triangular(self.M[k:, k:], f[k:].unsqueeze(1),
+                                              upper=False).squeeze(1)
+            v = self.r.clone()
+            for i in range(k, s):
+                ops.axpy(v, self.G[i], -float(c[i - k]))
+            vh = self.new_vec(v)
+            self._apply_M(v, vh)
+            # U[k] = omega*vh + sum c_i U[i]
+            uk = vh.mul_(self.omega)
+            for i in range(k, s):
+                ops.axpy(uk, self.U[i], float(c[i - k]))
+            self.U[k] = uk.clone()
+            ops.spmv(self.A, self.U[k], self.G[k])
+            # biorthogonalize G[k] against P[0..k-1]
+            for i in range(k):
+                alpha = self.dot(self.P[i], self.G[k]) / self.M[i, i]
+                ops.axpy(self.G[k], self.G[i], -float(alpha))
+                ops.axpy(self.U[k], self.U[i], -float(alpha))
+            for i in range(k, s):
+                self.M[i, k] = self.dot(self.P[i], self.G[k])
+            if self.M[k, k] == 0.0:
+                return True
+            beta = float(f[k] / self.M[k, k])
+            ops.axpy(self.r, self.G[k], -beta)
+            ops.axpy(x, self.U[k], beta)
+            if k + 1 < s:
+                for i in range(k + 1, s):
+                    f[i] -= beta * self.M[i, k]
+        # dimension-reduction step
+        v = self.r.clone()
+        vh = self.new_vec(v)
+        self._apply_M(v, vh)
+        t = self.new_vec(v)
+        ops.spmv(self.A, vh, t)
+        tt = self.dot(t, t)
+        self.omega = self.dot(t, self.r) / tt if tt != 0.0 else 0.0
+        ops.axpy(self.r, t, -self.omega)
+        ops.axpy(x, vh, self.omega)
+        self._rnorm = self.compute_norm(self.r)
+        return False
+
+    def last_residual_norm(self, b, x):
+        return self._rnorm if self._rnorm is not None else super().last_residual_norm(b, x)

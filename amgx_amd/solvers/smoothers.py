@@ -1,0 +1,186 @@
+"""Smoothers: Block-Jacobi, L1-Jacobi, (multicolor) Gauss-Seidel, Chebyshev.
+
+Reference: src/solvers/{block_jacobi,jacobi_l1,multicolor_gauss_seidel,
+gauss_seidel,cheb,chebyshev_poly}_solver.cu. All are Solvers (usable
+standalone or composed as AMG smoothers); one ``solve_iteration`` = one sweep.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .. import ops
+from .base import Solver, register_solver
+
+
+class _SmootherBase(Solver):
+    is_smoother = True
+
+    def sweep(self, b, x, n: int = 1):
+        for _ in range(n):
+            self.solve_iteration(b, x)
+
+
+@register_solver("BLOCK_JACOBI")
+class BlockJacobiSolver(_SmootherBase):
+    """Damped (block-)Jacobi; relaxation_factor default 0.9
+    (reference src/solvers/block_jacobi_solver.cu, Dinv per-block inversion)."""
+
+    l1 = False
+
+    def solver_setup(self):
+        self.dinv = ops.jacobi_dinv(self.A, l1=self.l1)
+
+    def solve_iteration(self, b, x):
+        ops.jacobi_smooth(self.A, self.dinv, b, x, self.relaxation_factor)
+        return False
+
+
+@register_solver("JACOBI_L1")
+class JacobiL1Solver(_SmootherBase):
+    """L1-Jacobi (reference src/solvers/jacobi_l1_solver.cu): diag gets the
+    off-diagonal L1 row sum added; undamped by default in the reference."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        if not scope.has("relaxation_factor"):
+            self.relaxation_factor = 1.0
+
+    def solver_setup(self):
+        self.dinv = ops.jacobi_dinv(self.A, l1=True)
+
+    def solve_iteration(self, b, x):
+        ops.jacobi_smooth(self.A, self.dinv, b, x, self.relaxation_factor)
+        return False
+
+
+@register_solver("MULTICOLOR_GS")
+class MulticolorGSSolver(_SmootherBase):
+    """Multicolor Gauss-Seidel (reference
+    src/solvers/multicolor_gauss_seidel_solver.cu): rows of one color update
+    in parallel; colors sweep ascending, then descending when symmetric_GS."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.symmetric = bool(scope.get("symmetric_GS"))
+        if not scope.has("relaxation_factor"):
+            self.relaxation_factor = 1.0
+
+    def solver_setup(self):
+        A = self.A
+        if A.coloring is None:
+            from ..amg.coloring import MatrixColoring
+            A.coloring = MatrixColoring.create(A, self.scope)
+        self.dinv = ops.jacobi_dinv(A, l1=bool(self.scope.get("GS_L1_variant")))
+
+    def solve_iteration(self, b, x):
+        col = self.A.coloring
+        for c in range(col.num_colors):
+            ops.gs_smooth_color(self.A, self.dinv, b, x, col.rows_of(c),
+                                self.relaxation_factor)
+        if self.symmetric:
+            for c in range(col.num_colors - 1, -1, -1):
+                ops.gs_smooth_color(self.A, self.dinv, b, x, col.rows_of(c),
+                                    self.relaxation_factor)
+        return False
+
+
+@register_solver("GS")
+class GSSolver(MulticolorGSSolver):
+    """Sequential-flavor GS; on device we use the multicolor schedule
+    (reference src/solvers/gauss_seidel_solver.cu is serial-ish; a serial
+    sweep has no MI355X-native expression, color-parallel is the idiom)."""
+
+
+@register_solver("CHEBYSHEV")
+class ChebyshevSolver(_SmootherBase):
+    """Chebyshev iteration preconditioned by the (L1-)diagonal; lambda_max
+    estimated by power iteration on D^-1 A when
+    chebyshev_lambda_estimate_mode=0 (reference src/solvers/cheb_solver.cu,
+    src/core.cu:409-412)."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.est_mode = scope.get("chebyshev_lambda_estimate_mode")
+
+    def solver_setup(self):
+        self.dinv = ops.jacobi_dinv(self.A, l1=True)
+        if self.est_mode in (0, 1):
+            self.lmax = self._power_iteration(16)
+        else:
+            self.lmax = self.scope.get("cheby_max_lambda")
+        self.lmin = self.lmax / 8.0
+        self._init_cheb()
+
+    def _power_iteration(self, iters: int) -> float:
+        n = self.A.n_rows * self.A.block_dim
+        g = torch.Generator().manual_seed(7177)
+        v = torch.rand(n, generator=g, dtype=torch.float64) \
+            .to(self.A.dtype).to(self.A.device)
+        Av = torch.zeros_like(v)
+        lam = 1.0
+        for _ in range(iters):
+            nv = ops.nrm2(v)
+            if nv == 0:
+                break
+            ops.scal(v, 1.0 / nv)
+            ops.spmv(self.A, v, Av)
+            self._apply_dinv(Av)
+            lam = ops.dot(v, Av)
+            v, Av = Av, v
+        return abs(lam) * 1.05   # safety factor
+
+    def _apply_dinv(self, v):
+        if self.A.block_dim == 1:
+            v.mul_(self.dinv.reshape(-1))
+        else:
+            bd = self.A.block_dim
+            v.copy_(torch.bmm(self.dinv.to(v.dtype),
+                              v.reshape(-1, bd, 1)).reshape(v.shape))
+
+    def _init_cheb(self):
+        self.theta = 0.5 * (self.lmax + self.lmin)
+        self.delta = 0.5 * (self.lmax - self.lmin)
+
+    def solve_init(self, b, x, zero_initial_guess):
+        self._sigma = self.theta / self.delta if self.delta else 1.0
+        self._rho = 1.0 / self._sigma if self._sigma else 1.0
+        self._d = None
+
+    def solve_iteration(self, b, x):
+        r = ops.residual(self.A, x, b)
+        self._apply_dinv(r)
+        if self._d is None:
+            self._d = r.mul_(1.0 / self.theta)
+        else:
+            rho_new = 1.0 / (2.0 * self._sigma - self._rho)
+            a = rho_new * self._rho
+            c = 2.0 * rho_new / self.delta
+            ops.axpby(self._d, r, c, a)
+            self._rho = rho_new
+        ops.axpy(x, self._d, 1.0)
+        return False
+
+
+@register_solver("CHEBYSHEV_POLY")
+class ChebyshevPolySolver(ChebyshevSolver):
+    """Chebyshev polynomial smoother of fixed order (reference
+    src/solvers/chebyshev_poly.cu, order 5 default): one solve_iteration runs
+    the full order-k polynomial."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.order = scope.get("chebyshev_polynomial_order")
+
+    def solve_iteration(self, b, x):
+        self.solve_init(b, x, False)
+        for _ in range(self.order):
+            super().solve_iteration(b, x)
+        return False
+
+
+@register_solver("POLYNOMIAL")
+@register_solver("KPZ_POLYNOMIAL")
+class PolynomialSolver(ChebyshevPolySolver):
+    """Polynomial smoother family; maps to the Chebyshev polynomial engine
+    (reference src/solvers/polynomial_solver.cu)."""

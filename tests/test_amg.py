@@ -1,0 +1,132 @@
+import numpy as np
+import pytest
+import torch
+
+from amgx_amd import AMGConfig, create_solver, ops
+from amgx_amd.resources import Resources
+from amgx_amd.problems import poisson_2d, poisson_3d
+
+FGMRES_AGG = {
+    "solver": {
+        "preconditioner": {
+            "algorithm": "AGGREGATION",
+            "solver": "AMG",
+            "smoother": "MULTICOLOR_DILU",
+            "presweeps": 0,
+            "postsweeps": 3,
+            "selector": "SIZE_2",
+            "coarse_solver": "DENSE_LU_SOLVER",
+            "max_iters": 1,
+            "min_coarse_rows": 32,
+            "relaxation_factor": 0.75,
+            "scope": "amg",
+            "max_levels": 50,
+            "cycle": "V",
+        },
+        "solver": "FGMRES",
+        "max_iters": 100,
+        "gmres_n_restart": 10,
+        "monitor_residual": 1,
+        "convergence": "RELATIVE_INI",
+        "tolerance": 1e-6,
+        "norm": "L2",
+    }
+}
+
+PCG_CLASSICAL = {
+    "solver": {
+        "preconditioner": {
+            "solver": "AMG",
+            "algorithm": "CLASSICAL",
+            "smoother": {"solver": "BLOCK_JACOBI", "relaxation_factor": 0.8},
+            "presweeps": 1,
+            "postsweeps": 1,
+            "max_iters": 1,
+            "min_coarse_rows": 10,
+            "max_levels": 20,
+            "cycle": "V",
+        },
+        "solver": "PCG",
+        "max_iters": 100,
+        "monitor_residual": 1,
+        "convergence": "RELATIVE_INI",
+        "tolerance": 1e-6,
+    }
+}
+
+
+def run(cfg_dict, A, max_expected_iters, tol=1e-6):
+    cfg = AMGConfig.from_dict(cfg_dict)
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    r = ops.residual(A, x, b)
+    rel = ops.nrm2(r) / ops.nrm2(b)
+    assert st.converged, f"not converged: {st}"
+    assert rel < 10 * tol
+    assert st.iterations <= max_expected_iters, st.iterations
+    return st
+
+
+def test_amg_aggregation_standalone():
+    """Plain AMG V-cycles as the solver (aggregation)."""
+    A = poisson_2d(24, 24)
+    cfg = {
+        "solver": {
+            "solver": "AMG",
+            "algorithm": "AGGREGATION",
+            "selector": "SIZE_2",
+            "smoother": "MULTICOLOR_DILU",
+            "presweeps": 1, "postsweeps": 1,
+            "min_coarse_rows": 16,
+            "max_iters": 60,
+            "monitor_residual": 1,
+            "convergence": "RELATIVE_INI",
+            "tolerance": 1e-8,
+        }
+    }
+    st = run(cfg, A, max_expected_iters=60, tol=1e-8)
+
+
+def test_fgmres_aggregation_poisson2d():
+    """The flagship FGMRES_AGGREGATION config (driver config class #2)."""
+    A = poisson_2d(24, 24)
+    st = run(FGMRES_AGG, A, max_expected_iters=30)
+
+
+def test_fgmres_aggregation_poisson3d():
+    A = poisson_3d(8, 8, 8)
+    st = run(FGMRES_AGG, A, max_expected_iters=30)
+
+
+def test_pcg_classical_poisson2d():
+    A = poisson_2d(24, 24)
+    st = run(PCG_CLASSICAL, A, max_expected_iters=30)
+
+
+def test_pcg_classical_poisson3d():
+    A = poisson_3d(8, 8, 8)
+    st = run(PCG_CLASSICAL, A, max_expected_iters=30)
+
+
+def test_w_cycle():
+    A = poisson_2d(20, 20)
+    cfg = {"solver": dict(FGMRES_AGG["solver"])}
+    cfg["solver"]["preconditioner"] = dict(FGMRES_AGG["solver"]["preconditioner"],
+                                           cycle="W")
+    run(cfg, A, max_expected_iters=30)
+
+
+def test_grid_stats():
+    A = poisson_2d(24, 24)
+    from amgx_amd.amg.amg import AMGHierarchy
+    cfg = AMGConfig.from_dict(FGMRES_AGG)
+    _, sub = cfg.root_scope().sub_solver("preconditioner")
+    h = AMGHierarchy(sub, Resources("cpu"))
+    h.setup(A)
+    stats = h.grid_stats()
+    assert "Number of Levels" in stats and len(h.levels) >= 3
+    # aggregation should roughly halve rows per level
+    assert h.levels[1].A.n_rows < 0.7 * A.n_rows

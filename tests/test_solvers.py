@@ -1,0 +1,146 @@
+import numpy as np
+import pytest
+import torch
+
+from amgx_amd import AMGConfig, create_solver
+from amgx_amd.config import ConfigScope
+from amgx_amd.resources import Resources
+from amgx_amd import ops
+from amgx_amd.problems import poisson_2d, poisson_3d, block_laplacian
+
+
+def make(cfg_dict):
+    cfg = AMGConfig.from_dict(cfg_dict)
+    return create_solver(cfg.root_scope(), resources=Resources("cpu"))
+
+
+def solve_poisson(solver, n=None, A=None, tol_check=1e-6):
+    if A is None:
+        A = poisson_2d(16, 16)
+    b = torch.ones(A.n_rows * A.block_dim, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    solver.setup(A)
+    st = solver.solve(b, x, zero_initial_guess=True)
+    r = ops.residual(A, x, b)
+    rel = ops.nrm2(r) / ops.nrm2(b)
+    return st, rel
+
+
+def test_cg_poisson():
+    s = make({"solver": "CG", "max_iters": 400, "monitor_residual": 1,
+              "tolerance": 1e-8, "convergence": "RELATIVE_INI"})
+    st, rel = solve_poisson(s)
+    assert st.converged and rel < 1e-7
+
+
+def test_pcg_jacobi_poisson():
+    s = make({"solver": "PCG", "preconditioner": "BLOCK_JACOBI",
+              "max_iters": 400, "monitor_residual": 1, "tolerance": 1e-8,
+              "convergence": "RELATIVE_INI"})
+    st, rel = solve_poisson(s)
+    assert st.converged and rel < 1e-7
+
+
+def test_bicgstab_poisson():
+    s = make({"solver": "PBICGSTAB", "preconditioner": "BLOCK_JACOBI",
+              "max_iters": 400, "monitor_residual": 1, "tolerance": 1e-8,
+              "convergence": "RELATIVE_INI"})
+    st, rel = solve_poisson(s)
+    assert st.converged and rel < 1e-6
+
+
+def test_fgmres_poisson():
+    s = make({"solver": "FGMRES", "preconditioner": "BLOCK_JACOBI",
+              "gmres_n_restart": 20, "max_iters": 300, "monitor_residual": 1,
+              "tolerance": 1e-8, "convergence": "RELATIVE_INI"})
+    st, rel = solve_poisson(s)
+    assert st.converged and rel < 1e-6
+
+
+def test_idr_poisson():
+    s = make({"solver": "IDR", "subspace_dim_s": 4, "max_iters": 300,
+              "monitor_residual": 1, "tolerance": 1e-8,
+              "convergence": "RELATIVE_INI"})
+    st, rel = solve_poisson(s)
+    assert st.converged and rel < 1e-6
+
+
+def test_jacobi_smoother_reduces_residual():
+    A = poisson_2d(12, 12)
+    s = make({"solver": "BLOCK_JACOBI", "max_iters": 20})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    r0 = ops.nrm2(ops.residual(A, x, b))
+    s.solve(b, x, zero_initial_guess=False)
+    r1 = ops.nrm2(ops.residual(A, x, b))
+    assert r1 < 0.7 * r0
+
+
+def test_multicolor_gs_smoother():
+    A = poisson_2d(12, 12)
+    s = make({"solver": "MULTICOLOR_GS", "max_iters": 10})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    r0 = ops.nrm2(ops.residual(A, x, b))
+    s.solve(b, x)
+    r1 = ops.nrm2(ops.residual(A, x, b))
+    assert r1 < 0.75 * r0
+
+
+def test_chebyshev_smoother():
+    A = poisson_2d(12, 12)
+    s = make({"solver": "CHEBYSHEV", "max_iters": 10})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    r0 = ops.nrm2(ops.residual(A, x, b))
+    s.solve(b, x)
+    r1 = ops.nrm2(ops.residual(A, x, b))
+    assert r1 < 0.75 * r0
+
+
+def test_dilu_smoother_scalar():
+    A = poisson_2d(12, 12)
+    s = make({"solver": "MULTICOLOR_DILU", "max_iters": 10,
+              "relaxation_factor": 1.0})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    r0 = ops.nrm2(ops.residual(A, x, b))
+    s.solve(b, x)
+    r1 = ops.nrm2(ops.residual(A, x, b))
+    assert r1 < 0.6 * r0
+
+
+def test_ilu0_smoother():
+    A = poisson_2d(10, 10)
+    s = make({"solver": "MULTICOLOR_ILU", "max_iters": 6})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    r0 = ops.nrm2(ops.residual(A, x, b))
+    s.solve(b, x)
+    r1 = ops.nrm2(ops.residual(A, x, b))
+    assert r1 < 0.7 * r0
+
+
+def test_block_jacobi_on_block_matrix():
+    A = block_laplacian(8, 8, block_dim=4)
+    s = make({"solver": "PCG", "preconditioner": "BLOCK_JACOBI",
+              "max_iters": 500, "monitor_residual": 1, "tolerance": 1e-8,
+              "convergence": "RELATIVE_INI"})
+    st, rel = solve_poisson(s, A=A)
+    assert st.converged and rel < 1e-6
+
+
+def test_dense_lu_direct():
+    A = poisson_2d(6, 6)
+    s = make({"solver": "DENSE_LU_SOLVER"})
+    b = torch.rand(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    s.solve(b, x)
+    r = ops.residual(A, x, b)
+    assert ops.nrm2(r) < 1e-10
