@@ -4,8 +4,11 @@ A coloring partitions rows so no two adjacent rows (distance-1) share a
 color; multicolor smoothers (GS/DILU/ILU/Kaczmarz) sweep color by color with
 full parallelism inside a color. The GPU implements the parallel MIN_MAX
 hash-based local-maximum scheme (reference src/matrix_coloring/min_max.cu);
-the host reference is sequential greedy. ``rows_of(c)`` gives the row index
-tensor of a color, precomputed and kept on the matrix's device.
+the host reference is sequential greedy.
+
+The attachment precomputes ``rows_sorted`` (row ids stably sorted by color,
+device-resident) and host-side ``bounds`` so every per-color kernel slice is
+a view — no per-sweep index rebuilds.
 """
 
 from __future__ import annotations
@@ -19,7 +22,13 @@ class MatrixColoring:
     def __init__(self, colors: torch.Tensor, num_colors: int):
         self.colors = colors
         self.num_colors = num_colors
-        self._rows = None
+        counts = torch.bincount(colors.to(torch.int64), minlength=num_colors)
+        order = torch.argsort(colors.to(torch.int64), stable=True)
+        self.rows_sorted = order.to(torch.int32).contiguous()
+        b = [0]
+        for c in counts.cpu().tolist():
+            b.append(b[-1] + int(c))
+        self.bounds = b
 
     @classmethod
     def create(cls, A, scope=None) -> "MatrixColoring":
@@ -28,22 +37,7 @@ class MatrixColoring:
         return cls(colors.to(A.row_offsets.device), num)
 
     def rows_of(self, c: int) -> torch.Tensor:
-        if self._rows is None:
-            device = self.colors.device
-            if device.type == "cuda":
-                # single sort on device; avoids num_colors nonzero() syncs
-                colors = self.colors
-                order = torch.argsort(colors.to(torch.int64), stable=True)
-                counts = torch.bincount(colors.to(torch.int64),
-                                        minlength=self.num_colors)
-                bounds = torch.zeros(self.num_colors + 1, dtype=torch.int64)
-                torch.cumsum(counts.cpu(), 0, out=bounds[1:])
-                self._rows = [order[bounds[i]:bounds[i + 1]].to(torch.int32)
-                              for i in range(self.num_colors)]
-            else:
-                self._rows = [(self.colors == i).nonzero(as_tuple=True)[0]
-                              .to(torch.int32) for i in range(self.num_colors)]
-        return self._rows[c]
+        return self.rows_sorted[self.bounds[c]:self.bounds[c + 1]]
 
     def validate(self, A) -> bool:
         """Distance-1 validity check (reference src/tests/valid_coloring.cu)."""

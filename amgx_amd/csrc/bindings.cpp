@@ -1,0 +1,444 @@
+// Python bindings of the gfx950 kernel library (amgx_amd._core).
+//
+// Besides 1:1 kernel wrappers this TU hosts the C++-side orchestration loops
+// whose per-step launches would otherwise pay a Python round trip per color /
+// per round: the whole DILU apply, the MIN_MAX coloring loop and the SIZE_2
+// matching loop each run as ONE call from Python.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <vector>
+
+#include "core_api.h"
+
+namespace {
+
+using torch::Tensor;
+
+hipStream_t cur_stream() {
+    return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+#define DISPATCH_FT(TENSOR, NAME, ...)                                  \
+    AT_DISPATCH_FLOATING_TYPES(TENSOR.scalar_type(), NAME, __VA_ARGS__)
+
+inline void check_dev(const Tensor& t) {
+    TORCH_CHECK(t.is_cuda(), "amgx_amd._core requires device tensors");
+}
+
+// ---------------------------------------------------------------- SpMV
+void csrmv(Tensor ro, Tensor ci, Tensor va, int64_t block_dim, Tensor x,
+           Tensor y, c10::optional<Tensor> bvec, double alpha, double beta,
+           double gamma, int64_t r0, int64_t r1) {
+    check_dev(va);
+    double avg = ci.numel() / std::max<double>(1.0, ro.numel() - 1);
+    DISPATCH_FT(va, "csrmv", [&] {
+        const scalar_t* bp =
+            bvec.has_value() ? bvec->data_ptr<scalar_t>() : nullptr;
+        if (block_dim == 1) {
+            amgx_hip::csrmv<scalar_t>(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                                      va.data_ptr<scalar_t>(),
+                                      x.data_ptr<scalar_t>(),
+                                      y.data_ptr<scalar_t>(), bp,
+                                      (scalar_t)alpha, (scalar_t)beta,
+                                      (scalar_t)gamma, (int)r0, (int)r1, avg,
+                                      cur_stream());
+        } else {
+            amgx_hip::bsrmv<scalar_t>(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                                      va.data_ptr<scalar_t>(), (int)block_dim,
+                                      x.data_ptr<scalar_t>(),
+                                      y.data_ptr<scalar_t>(), bp,
+                                      (scalar_t)alpha, (scalar_t)beta,
+                                      (scalar_t)gamma, (int)r0, (int)r1,
+                                      cur_stream());
+        }
+    });
+}
+
+// ---------------------------------------------------------------- BLAS
+Tensor reduce_op(Tensor x, c10::optional<Tensor> y, int64_t op) {
+    check_dev(x);
+    auto out = torch::empty({1}, x.options());
+    auto ws = torch::empty({2048}, x.options());
+    DISPATCH_FT(x, "reduce", [&] {
+        const scalar_t* yp = y.has_value() ? y->data_ptr<scalar_t>()
+                                           : x.data_ptr<scalar_t>();
+        amgx_hip::reduce<scalar_t>(x.data_ptr<scalar_t>(), yp, x.numel(),
+                                   (int)op, ws.data_ptr<scalar_t>(),
+                                   out.data_ptr<scalar_t>(), cur_stream());
+    });
+    return out;
+}
+
+void axpy(Tensor y, Tensor x, double a) {
+    DISPATCH_FT(x, "axpy", [&] {
+        amgx_hip::axpy<scalar_t>(y.data_ptr<scalar_t>(),
+                                 x.data_ptr<scalar_t>(), (scalar_t)a,
+                                 x.numel(), cur_stream());
+    });
+}
+
+void axpby(Tensor y, Tensor x, double a, double b) {
+    DISPATCH_FT(x, "axpby", [&] {
+        amgx_hip::axpby<scalar_t>(y.data_ptr<scalar_t>(),
+                                  x.data_ptr<scalar_t>(), (scalar_t)a,
+                                  (scalar_t)b, x.numel(), cur_stream());
+    });
+}
+
+void scal(Tensor x, double a) {
+    DISPATCH_FT(x, "scal", [&] {
+        amgx_hip::scal<scalar_t>(x.data_ptr<scalar_t>(), (scalar_t)a,
+                                 x.numel(), cur_stream());
+    });
+}
+
+// ---------------------------------------------------------------- structure
+Tensor diag_index(Tensor ro, Tensor ci, int64_t n) {
+    auto out = torch::empty({n}, ro.options());
+    amgx_hip::diag_index(ro.data_ptr<int>(), ci.data_ptr<int>(), (int)n,
+                         out.data_ptr<int>(), cur_stream());
+    return out;
+}
+
+Tensor extract_diag(Tensor ro, Tensor ci, Tensor va, Tensor didx, int64_t n,
+                    int64_t b) {
+    auto out = b == 1 ? torch::empty({n}, va.options())
+                      : torch::empty({n, b, b}, va.options());
+    DISPATCH_FT(va, "extract_diag", [&] {
+        amgx_hip::extract_diag<scalar_t>(ro.data_ptr<int>(),
+                                         ci.data_ptr<int>(),
+                                         va.data_ptr<scalar_t>(),
+                                         didx.data_ptr<int>(), (int)n, (int)b,
+                                         out.data_ptr<scalar_t>(),
+                                         cur_stream());
+    });
+    return out;
+}
+
+Tensor trans_index(Tensor ro, Tensor ci, int64_t n) {
+    auto out = torch::empty({ci.numel()}, ro.options());
+    amgx_hip::trans_index(ro.data_ptr<int>(), ci.data_ptr<int>(), (int)n,
+                          (int)ci.numel(), out.data_ptr<int>(), cur_stream());
+    return out;
+}
+
+// ---------------------------------------------------------------- smoothers
+Tensor jacobi_dinv(Tensor ro, Tensor ci, Tensor va, Tensor didx, int64_t n,
+                   int64_t b, bool l1) {
+    auto out = b == 1 ? torch::empty({n}, va.options())
+                      : torch::empty({n, b, b}, va.options());
+    DISPATCH_FT(va, "jacobi_dinv", [&] {
+        amgx_hip::jacobi_dinv<scalar_t>(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                                        va.data_ptr<scalar_t>(),
+                                        didx.data_ptr<int>(), (int)n, (int)b,
+                                        l1, out.data_ptr<scalar_t>(),
+                                        cur_stream());
+    });
+    return out;
+}
+
+void jacobi_smooth(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
+                   Tensor bvec, Tensor xi, Tensor xo, double omega) {
+    int n = (int)(ro.numel() - 1);
+    double avg = ci.numel() / std::max<double>(1.0, n);
+    DISPATCH_FT(va, "jacobi_smooth", [&] {
+        amgx_hip::jacobi_smooth<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            dinv.data_ptr<scalar_t>(), bvec.data_ptr<scalar_t>(),
+            xi.data_ptr<scalar_t>(), xo.data_ptr<scalar_t>(), (scalar_t)omega,
+            n, (int)b, avg, cur_stream());
+    });
+}
+
+void gs_smooth_rows(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
+                    Tensor bvec, Tensor x, Tensor rows, double omega) {
+    int n = (int)(ro.numel() - 1);
+    DISPATCH_FT(va, "gs_smooth_rows", [&] {
+        amgx_hip::gs_smooth_rows<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            dinv.data_ptr<scalar_t>(), bvec.data_ptr<scalar_t>(),
+            x.data_ptr<scalar_t>(), rows.data_ptr<int>(), (int)rows.numel(),
+            (scalar_t)omega, n, (int)b, cur_stream());
+    });
+}
+
+// multicolor GS full sweep: colors ascending (and descending if symmetric)
+void gs_sweep(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
+              Tensor bvec, Tensor x, Tensor rows_sorted,
+              std::vector<int64_t> bounds, double omega, bool symmetric) {
+    int n = (int)(ro.numel() - 1);
+    int nc = (int)bounds.size() - 1;
+    DISPATCH_FT(va, "gs_sweep", [&] {
+        auto run = [&](int c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) return;
+            amgx_hip::gs_smooth_rows<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_t>(), dinv.data_ptr<scalar_t>(),
+                bvec.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                (scalar_t)omega, n, (int)b, cur_stream());
+        };
+        for (int c = 0; c < nc; ++c) run(c);
+        if (symmetric)
+            for (int c = nc - 1; c >= 0; --c) run(c);
+    });
+}
+
+// ---------------------------------------------------------------- DILU
+Tensor dilu_setup(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor didx,
+                  Tensor tidx, Tensor colors, Tensor rows_sorted,
+                  std::vector<int64_t> bounds) {
+    int n = (int)(ro.numel() - 1);
+    auto einv = b == 1 ? torch::zeros({n}, va.options())
+                       : torch::zeros({n, b, b}, va.options());
+    int nc = (int)bounds.size() - 1;
+    DISPATCH_FT(va, "dilu_setup", [&] {
+        for (int c = 0; c < nc; ++c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::dilu_setup_color<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_t>(), didx.data_ptr<int>(),
+                tidx.data_ptr<int>(), colors.data_ptr<int>(),
+                rows_sorted.data_ptr<int>() + s, (int)(e - s), c,
+                einv.data_ptr<scalar_t>(), (int)b, cur_stream());
+        }
+    });
+    return einv;
+}
+
+// full M^-1 apply: r given; w,z scratch (pre-allocated); x += relax*z
+void dilu_apply(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor einv,
+                Tensor colors, Tensor rows_sorted, std::vector<int64_t> bounds,
+                Tensor r, Tensor w, Tensor z, Tensor x, double relax) {
+    int n = (int)(ro.numel() - 1);
+    int nc = (int)bounds.size() - 1;
+    w.zero_();
+    z.zero_();
+    DISPATCH_FT(va, "dilu_apply", [&] {
+        hipStream_t st = cur_stream();
+        for (int c = 0; c < nc; ++c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::dilu_fwd_color<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_t>(), einv.data_ptr<scalar_t>(),
+                colors.data_ptr<int>(), rows_sorted.data_ptr<int>() + s,
+                (int)(e - s), c, r.data_ptr<scalar_t>(),
+                w.data_ptr<scalar_t>(), (int)b, st);
+        }
+        for (int c = nc - 1; c >= 0; --c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::dilu_bwd_color<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_t>(), einv.data_ptr<scalar_t>(),
+                colors.data_ptr<int>(), rows_sorted.data_ptr<int>() + s,
+                (int)(e - s), c, w.data_ptr<scalar_t>(),
+                z.data_ptr<scalar_t>(), (int)b, st);
+        }
+        amgx_hip::axpy<scalar_t>(x.data_ptr<scalar_t>(),
+                                 z.data_ptr<scalar_t>(), (scalar_t)relax,
+                                 x.numel(), st);
+    });
+}
+
+// ---------------------------------------------------------------- coloring
+std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
+                                         int64_t max_rounds, int64_t seed) {
+    auto colors = torch::full({n}, -1,
+                              ro.options().dtype(torch::kInt32));
+    auto counter = torch::zeros({1}, ro.options().dtype(torch::kInt32));
+    hipStream_t st = cur_stream();
+    int rounds = 0;
+    for (; rounds < max_rounds; ++rounds) {
+        counter.zero_();
+        amgx_hip::color_minmax_round(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                                     (int)n, colors.data_ptr<int>(), rounds,
+                                     (int)(seed + rounds * 7919),
+                                     counter.data_ptr<int>(), st);
+        int left = counter.cpu().item<int>();
+        if (left == 0) break;
+    }
+    int64_t ncolors = (colors.max().cpu().item<int>()) + 1;
+    return {colors, ncolors};
+}
+
+// ---------------------------------------------------------------- aggregation
+Tensor size2_match(Tensor ro, Tensor ci, Tensor va, Tensor tidx, Tensor diag,
+                   int64_t n, int64_t max_iters) {
+    auto agg = torch::full({n}, -1, ro.options().dtype(torch::kInt32));
+    auto prop = torch::empty({n}, ro.options().dtype(torch::kInt32));
+    auto changed = torch::zeros({1}, ro.options().dtype(torch::kInt32));
+    hipStream_t st = cur_stream();
+    DISPATCH_FT(va, "size2_match", [&] {
+        for (int it = 0; it < max_iters; ++it) {
+            amgx_hip::agg_propose<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_t>(), tidx.data_ptr<int>(),
+                diag.data_ptr<scalar_t>(), (int)n, agg.data_ptr<int>(),
+                prop.data_ptr<int>(), st);
+            changed.zero_();
+            amgx_hip::agg_match(prop.data_ptr<int>(), (int)n,
+                                agg.data_ptr<int>(), changed.data_ptr<int>(),
+                                st);
+            if (changed.cpu().item<int>() == 0) break;
+        }
+        auto agg_out = torch::empty_like(agg);
+        amgx_hip::agg_merge_singletons<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            tidx.data_ptr<int>(), diag.data_ptr<scalar_t>(), (int)n,
+            agg.data_ptr<int>(), agg_out.data_ptr<int>(), st);
+        agg = agg_out;
+    });
+    return agg;   // root fine-node ids; Python renumbers with torch.unique
+}
+
+// ---------------------------------------------------------------- transfers
+void restrict_agg(Tensor r, Tensor agg, int64_t b, Tensor rc) {
+    rc.zero_();
+    DISPATCH_FT(r, "restrict_agg", [&] {
+        amgx_hip::restrict_agg<scalar_t>(r.data_ptr<scalar_t>(),
+                                         agg.data_ptr<int>(),
+                                         (int)agg.numel(), (int)b,
+                                         rc.data_ptr<scalar_t>(),
+                                         cur_stream());
+    });
+}
+
+void prolongate_agg(Tensor x, Tensor xc, Tensor agg, int64_t b) {
+    DISPATCH_FT(x, "prolongate_agg", [&] {
+        amgx_hip::prolongate_agg<scalar_t>(x.data_ptr<scalar_t>(),
+                                           xc.data_ptr<scalar_t>(),
+                                           agg.data_ptr<int>(),
+                                           (int)agg.numel(), (int)b,
+                                           cur_stream());
+    });
+}
+
+// ---------------------------------------------------------------- dense
+void dense_gemv(Tensor Ainv, Tensor b, Tensor x) {
+    DISPATCH_FT(b, "dense_gemv", [&] {
+        amgx_hip::dense_gemv<scalar_t>(Ainv.data_ptr<scalar_t>(),
+                                       b.data_ptr<scalar_t>(),
+                                       x.data_ptr<scalar_t>(),
+                                       (int)b.numel(), cur_stream());
+    });
+}
+
+// ---------------------------------------------------------------- pack
+void gather(Tensor src, Tensor idx, int64_t b, Tensor dst) {
+    DISPATCH_FT(src, "gather", [&] {
+        amgx_hip::gather<scalar_t>(src.data_ptr<scalar_t>(),
+                                   idx.data_ptr<int>(), (int)idx.numel(),
+                                   (int)b, dst.data_ptr<scalar_t>(),
+                                   cur_stream());
+    });
+}
+
+void scatter(Tensor src, Tensor idx, int64_t b, Tensor dst, bool add) {
+    DISPATCH_FT(src, "scatter", [&] {
+        if (add)
+            amgx_hip::scatter_add<scalar_t>(src.data_ptr<scalar_t>(),
+                                            idx.data_ptr<int>(),
+                                            (int)idx.numel(), (int)b,
+                                            dst.data_ptr<scalar_t>(),
+                                            cur_stream());
+        else
+            amgx_hip::scatter<scalar_t>(src.data_ptr<scalar_t>(),
+                                        idx.data_ptr<int>(), (int)idx.numel(),
+                                        (int)b, dst.data_ptr<scalar_t>(),
+                                        cur_stream());
+    });
+}
+
+// ---------------------------------------------------------------- SpGEMM
+std::tuple<Tensor, Tensor, Tensor> galerkin_agg(Tensor ro, Tensor ci,
+                                                Tensor va, Tensor agg,
+                                                int64_t nc, int64_t block_dim) {
+    int n = (int)(ro.numel() - 1);
+    long long nnz = ci.numel();
+    int bb = (int)(block_dim * block_dim);
+    auto ro_c = torch::empty({nc + 1}, ro.options());
+    auto ci_c = torch::empty({nnz}, ro.options());
+    auto va_c = block_dim == 1
+                    ? torch::empty({nnz}, va.options())
+                    : torch::empty({nnz, block_dim, block_dim}, va.options());
+    long long nnz_c = 0;
+    DISPATCH_FT(va, "galerkin_agg", [&] {
+        nnz_c = amgx_hip::galerkin_agg<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            n, nnz, agg.data_ptr<int>(), (int)nc, ro_c.data_ptr<int>(),
+            ci_c.data_ptr<int>(), va_c.data_ptr<scalar_t>(), bb,
+            cur_stream());
+    });
+    return {ro_c, ci_c.narrow(0, 0, nnz_c), va_c.narrow(0, 0, nnz_c)};
+}
+
+std::tuple<Tensor, Tensor, Tensor> spgemm(Tensor roA, Tensor ciA, Tensor vaA,
+                                          Tensor roB, Tensor ciB, Tensor vaB,
+                                          int64_t n_cols_B,
+                                          int64_t capacity) {
+    int m = (int)(roA.numel() - 1);
+    int k = (int)(roB.numel() - 1);
+    auto ro_c = torch::empty({m + 1}, roA.options());
+    auto ci_c = torch::empty({capacity}, roA.options());
+    auto va_c = torch::empty({capacity}, vaA.options());
+    long long nnz_c = 0;
+    DISPATCH_FT(vaA, "spgemm", [&] {
+        nnz_c = amgx_hip::spgemm_esc<scalar_t>(
+            roA.data_ptr<int>(), ciA.data_ptr<int>(), vaA.data_ptr<scalar_t>(),
+            m, ciA.numel(), roB.data_ptr<int>(), ciB.data_ptr<int>(),
+            vaB.data_ptr<scalar_t>(), k, (int)n_cols_B, ro_c.data_ptr<int>(),
+            ci_c.data_ptr<int>(), va_c.data_ptr<scalar_t>(), cur_stream());
+    });
+    return {ro_c, ci_c.narrow(0, 0, nnz_c), va_c.narrow(0, 0, nnz_c)};
+}
+
+std::tuple<Tensor, Tensor, Tensor> transpose(Tensor ro, Tensor ci, Tensor va,
+                                             int64_t n_cols) {
+    int m = (int)(ro.numel() - 1);
+    long long nnz = ci.numel();
+    auto ro_t = torch::empty({n_cols + 1}, ro.options());
+    auto ci_t = torch::empty({nnz}, ro.options());
+    auto va_t = torch::empty({nnz}, va.options());
+    DISPATCH_FT(va, "transpose", [&] {
+        amgx_hip::transpose_csr<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            m, (int)n_cols, nnz, ro_t.data_ptr<int>(), ci_t.data_ptr<int>(),
+            va_t.data_ptr<scalar_t>(), cur_stream());
+    });
+    return {ro_t, ci_t, va_t};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("csrmv", &csrmv);
+    m.def("reduce_op", &reduce_op);
+    m.def("axpy", &axpy);
+    m.def("axpby", &axpby);
+    m.def("scal", &scal);
+    m.def("diag_index", &diag_index);
+    m.def("extract_diag", &extract_diag);
+    m.def("trans_index", &trans_index);
+    m.def("jacobi_dinv", &jacobi_dinv);
+    m.def("jacobi_smooth", &jacobi_smooth);
+    m.def("gs_smooth_rows", &gs_smooth_rows);
+    m.def("gs_sweep", &gs_sweep);
+    m.def("dilu_setup", &dilu_setup);
+    m.def("dilu_apply", &dilu_apply);
+    m.def("color_minmax", &color_minmax);
+    m.def("size2_match", &size2_match);
+    m.def("restrict_agg", &restrict_agg);
+    m.def("prolongate_agg", &prolongate_agg);
+    m.def("dense_gemv", &dense_gemv);
+    m.def("gather", &gather);
+    m.def("scatter", &scatter);
+    m.def("galerkin_agg", &galerkin_agg);
+    m.def("spgemm", &spgemm);
+    m.def("transpose", &transpose);
+}

@@ -1,0 +1,143 @@
+// Launcher API of the gfx950 kernel library (implemented in *.hip TUs,
+// called from bindings.cpp). Raw pointers + hipStream_t so the bindings TU
+// compiles with the host compiler and the kernels with hipcc only.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace amgx_hip {
+
+// ---- SpMV family (csrmv.hip) ------------------------------------------------
+// y[i] = alpha * (A x)[i] + beta * y[i] + gamma * b[i]  over rows [r0, r1)
+template <typename T>
+void csrmv(const int* ro, const int* ci, const T* va, const T* x, T* y,
+           const T* bvec, T alpha, T beta, T gamma, int r0, int r1,
+           double avg_deg, hipStream_t s);
+
+// block-CSR variant, block_dim b in [2,8]; values (nnz, b, b) row-major
+template <typename T>
+void bsrmv(const int* ro, const int* ci, const T* va, int b, const T* x, T* y,
+           const T* bvec, T alpha, T beta, T gamma, int r0, int r1,
+           hipStream_t s);
+
+// ---- BLAS-1 (blas.hip) ------------------------------------------------------
+// op: 0 = dot(x,y), 1 = sum|x| (L1), 2 = max|x| (Lmax). Deterministic
+// two-stage reduction; out is a device scalar, ws a device scratch of
+// >= 2048 T.
+template <typename T>
+void reduce(const T* x, const T* y, long long n, int op, T* ws, T* out,
+            hipStream_t s);
+
+template <typename T>
+void axpy(T* y, const T* x, T a, long long n, hipStream_t s);
+template <typename T>
+void axpby(T* y, const T* x, T a, T b, long long n, hipStream_t s);
+template <typename T>
+void scal(T* x, T a, long long n, hipStream_t s);
+
+// ---- structure (misc.hip) ---------------------------------------------------
+void diag_index(const int* ro, const int* ci, int n, int* out, hipStream_t s);
+template <typename T>
+void extract_diag(const int* ro, const int* ci, const T* va, const int* didx,
+                  int n, int b, T* out, hipStream_t s);
+void trans_index(const int* ro, const int* ci, int n, int nnz, int* out,
+                 hipStream_t s);
+
+// ---- smoothers (smoothers.hip) ---------------------------------------------
+template <typename T>
+void jacobi_dinv(const int* ro, const int* ci, const T* va, const int* didx,
+                 int n, int b, bool l1, T* dinv, hipStream_t s);
+// xo = xi + omega * dinv * (b - A xi)   (block-aware; b=1 scalar fast path)
+template <typename T>
+void jacobi_smooth(const int* ro, const int* ci, const T* va, const T* dinv,
+                   const T* bvec, const T* xi, T* xo, T omega, int n, int b,
+                   double avg_deg, hipStream_t s);
+// in-place GS update of rows[count]: x[r] += omega*dinv[r]*(b - A x)[r]
+template <typename T>
+void gs_smooth_rows(const int* ro, const int* ci, const T* va, const T* dinv,
+                    const T* bvec, T* x, const int* rows, int count, T omega,
+                    int n, int b, hipStream_t s);
+
+// ---- DILU (dilu.hip) --------------------------------------------------------
+template <typename T>
+void dilu_setup_color(const int* ro, const int* ci, const T* va,
+                      const int* didx, const int* tidx, const int* colors,
+                      const int* rows, int count, int color, T* einv, int b,
+                      hipStream_t s);
+template <typename T>
+void dilu_fwd_color(const int* ro, const int* ci, const T* va, const T* einv,
+                    const int* colors, const int* rows, int count, int color,
+                    const T* r, T* w, int b, hipStream_t s);
+template <typename T>
+void dilu_bwd_color(const int* ro, const int* ci, const T* va, const T* einv,
+                    const int* colors, const int* rows, int count, int color,
+                    const T* w, T* z, int b, hipStream_t s);
+
+// ---- coloring (setup.hip) ---------------------------------------------------
+// one min-max hash round; returns (via counter) number newly colored.
+void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
+                        int iter, int seed, int* n_uncolored, hipStream_t s);
+
+// ---- aggregation (setup.hip) ------------------------------------------------
+template <typename T>
+void agg_propose(const int* ro, const int* ci, const T* va, const int* tidx,
+                 const T* diag, int n, const int* agg, int* prop,
+                 hipStream_t s);
+void agg_match(const int* prop, int n, int* agg, int* changed, hipStream_t s);
+template <typename T>
+void agg_merge_singletons(const int* ro, const int* ci, const T* va,
+                          const int* tidx, const T* diag, int n,
+                          const int* agg_in, int* agg_out, hipStream_t s);
+
+// ---- transfer operators (misc.hip) ------------------------------------------
+template <typename T>
+void restrict_agg(const T* r, const int* agg, int n, int b, T* rc,
+                  hipStream_t s);
+template <typename T>
+void prolongate_agg(T* x, const T* xc, const int* agg, int n, int b,
+                    hipStream_t s);
+
+// ---- dense coarse solve (misc.hip) -------------------------------------------
+template <typename T>
+void dense_gemv(const T* Ainv, const T* b, T* x, int n, hipStream_t s);
+
+// ---- gather/scatter for halo pack (misc.hip) ---------------------------------
+template <typename T>
+void gather(const T* src, const int* idx, int count, int b, T* dst,
+            hipStream_t s);
+template <typename T>
+void scatter(const T* src, const int* idx, int count, int b, T* dst,
+             hipStream_t s);
+template <typename T>
+void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
+                 hipStream_t s);
+
+// ---- SpGEMM / Galerkin (kernels_setup.hip) ----------------------------------
+// Aggregation Galerkin: COO keys agg[i]*nc+agg[j] -> radix sort ->
+// reduce_by_key -> CSR. Returns nnz_c; fills ro_c (nc+1); ci_c/va_c are
+// caller-allocated at worst case nnz (nnz_c <= nnz) and trimmed after.
+// bb = block_dim^2 (1 for scalar). Temps allocated internally (setup-time).
+template <typename T>
+long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
+                       long long nnz, const int* agg, int nc, int* ro_c,
+                       int* ci_c, T* va_c, int bb, hipStream_t s);
+
+// General ESC SpGEMM: C = A(m x k) @ B(k x n). Outputs sized by caller at
+// the expansion worst case is impractical; instead ci_c/va_c must be sized
+// >= nnz(C); pass capacity = expansion bound from Python (sum deg). Returns
+// nnz_c. Temps allocated internally.
+template <typename T>
+long long spgemm_esc(const int* roA, const int* ciA, const T* vaA, int m,
+                     long long nnzA, const int* roB, const int* ciB,
+                     const T* vaB, int k, int n, int* ro_c, int* ci_c, T* va_c,
+                     hipStream_t s);
+
+// CSR transpose via stable radix sort on column ids.
+template <typename T>
+void transpose_csr(const int* ro, const int* ci, const T* va, int m, int n,
+                   long long nnz, int* ro_t, int* ci_t, T* va_t,
+                   hipStream_t s);
+
+}  // namespace amgx_hip

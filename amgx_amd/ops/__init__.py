@@ -87,15 +87,23 @@ def jacobi_dinv(A, l1: bool = False):
     return _backend(A).jacobi_dinv(A, l1)
 
 
-def jacobi_smooth(A, dinv, b, x, omega: float):
-    """x += omega * dinv * (b - A x) — one damped-Jacobi sweep, fused."""
-    return _backend(A).jacobi_smooth(A, dinv, b, x, omega)
+def jacobi_smooth(A, dinv, b, x_in, x_out, omega: float):
+    """x_out = x_in + omega * dinv * (b - A x_in) — one damped-Jacobi sweep,
+    fused single pass over A (Jacobi needs the OLD x, so in/out are separate;
+    solvers ping-pong)."""
+    return _backend(A).jacobi_smooth(A, dinv, b, x_in, x_out, omega)
 
 
 def gs_smooth_color(A, dinv, b, x, color_rows, omega: float):
     """In-place Gauss-Seidel update of the rows in ``color_rows``
     (reference src/solvers/multicolor_gauss_seidel_solver.cu)."""
     return _backend(A).gs_smooth_color(A, dinv, b, x, color_rows, omega)
+
+
+def gs_sweep(A, dinv, b, x, coloring, omega: float, symmetric: bool = False):
+    """Full multicolor GS sweep (ascending colors, + descending when
+    symmetric). On GPU this is ONE extension call (per-color loop in C++)."""
+    return _backend(A).gs_sweep(A, dinv, b, x, coloring, omega, symmetric)
 
 
 # ---------------------------------------------------------------------- coloring

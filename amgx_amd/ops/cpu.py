@@ -145,15 +145,24 @@ def jacobi_dinv(A, l1: bool = False) -> torch.Tensor:
     return torch.linalg.inv(D).to(A.dtype)
 
 
-def jacobi_smooth(A, dinv, b, x, omega: float):
-    r = residual(A, x, b)
+def jacobi_smooth(A, dinv, b, x_in, x_out, omega: float):
+    r = residual(A, x_in, b)
     if A.block_dim == 1:
-        x.add_(dinv.reshape(-1) * r.reshape(-1), alpha=omega)
+        upd = dinv.reshape(-1) * r.reshape(-1)
     else:
         bd = A.block_dim
         rb = r.reshape(-1, bd, 1).to(dinv.dtype)
-        upd = torch.bmm(dinv, rb).reshape(-1)
-        x.reshape(-1).add_(upd.to(x.dtype), alpha=omega)
+        upd = torch.bmm(dinv, rb).reshape(-1).to(x_in.dtype)
+    x_out.reshape(-1).copy_(x_in.reshape(-1) + omega * upd)
+    return x_out
+
+
+def gs_sweep(A, dinv, b, x, coloring, omega, symmetric=False):
+    for c in range(coloring.num_colors):
+        gs_smooth_color(A, dinv, b, x, coloring.rows_of(c), omega)
+    if symmetric:
+        for c in range(coloring.num_colors - 1, -1, -1):
+            gs_smooth_color(A, dinv, b, x, coloring.rows_of(c), omega)
     return x
 
 

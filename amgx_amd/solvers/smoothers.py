@@ -20,6 +20,12 @@ class _SmootherBase(Solver):
         for _ in range(n):
             self.solve_iteration(b, x)
 
+    def _get_scratch(self, like):
+        sc = getattr(self, "_scratch", None)
+        if sc is None or sc.numel() != like.numel() or sc.device != like.device:
+            sc = self._scratch = torch.zeros_like(like.reshape(-1))
+        return sc
+
 
 @register_solver("BLOCK_JACOBI")
 class BlockJacobiSolver(_SmootherBase):
@@ -32,8 +38,22 @@ class BlockJacobiSolver(_SmootherBase):
         self.dinv = ops.jacobi_dinv(self.A, l1=self.l1)
 
     def solve_iteration(self, b, x):
-        ops.jacobi_smooth(self.A, self.dinv, b, x, self.relaxation_factor)
+        # fused single-pass sweep with ping-pong scratch, copied back once
+        sc = self._get_scratch(x)
+        ops.jacobi_smooth(self.A, self.dinv, b, x, sc, self.relaxation_factor)
+        x.reshape(-1).copy_(sc)
         return False
+
+    def sweep(self, b, x, n: int = 1):
+        # ping-pong across sweeps: one copy at the end only when n is odd
+        sc = self._get_scratch(x)
+        cur, other = x.reshape(-1), sc
+        for _ in range(n):
+            ops.jacobi_smooth(self.A, self.dinv, b, cur, other,
+                              self.relaxation_factor)
+            cur, other = other, cur
+        if cur.data_ptr() != x.reshape(-1).data_ptr():
+            x.reshape(-1).copy_(cur)
 
 
 @register_solver("JACOBI_L1")
@@ -50,7 +70,9 @@ class JacobiL1Solver(_SmootherBase):
         self.dinv = ops.jacobi_dinv(self.A, l1=True)
 
     def solve_iteration(self, b, x):
-        ops.jacobi_smooth(self.A, self.dinv, b, x, self.relaxation_factor)
+        sc = self._get_scratch(x)
+        ops.jacobi_smooth(self.A, self.dinv, b, x, sc, self.relaxation_factor)
+        x.reshape(-1).copy_(sc)
         return False
 
 
@@ -74,14 +96,8 @@ class MulticolorGSSolver(_SmootherBase):
         self.dinv = ops.jacobi_dinv(A, l1=bool(self.scope.get("GS_L1_variant")))
 
     def solve_iteration(self, b, x):
-        col = self.A.coloring
-        for c in range(col.num_colors):
-            ops.gs_smooth_color(self.A, self.dinv, b, x, col.rows_of(c),
-                                self.relaxation_factor)
-        if self.symmetric:
-            for c in range(col.num_colors - 1, -1, -1):
-                ops.gs_smooth_color(self.A, self.dinv, b, x, col.rows_of(c),
-                                    self.relaxation_factor)
+        ops.gs_sweep(self.A, self.dinv, b, x, self.A.coloring,
+                     self.relaxation_factor, self.symmetric)
         return False
 
 

@@ -1,0 +1,183 @@
+#!/usr/bin/env python3
+"""Driver benchmark contract: AMG setup+solve on 3D 7-pt Poisson fp64 with
+FGMRES + aggregation-AMG (the reference FGMRES_AGGREGATION.json config;
+BASELINE.json metric "AMG setup+solve sec & iters, 3D Poisson 256^3 fp64,
+FGMRES+AGG, 1/2/4/8 MI355X").
+
+Weak scaling: every rank owns a 256^3 subdomain (z-slab of a global
+256 x 256 x 256*N cube). One step = one full setup + solve-to-convergence
+(tolerance 1e-6, RELATIVE_INI) from scratch. Rank 0 prints ONE JSON line.
+
+    python bench.py --gpus N --steps K --warmup W [--size 256] [--config ...]
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from amgx_amd import AMGConfig, CSRMatrix, create_solver, ops  # noqa: E402
+from amgx_amd.resources import Resources  # noqa: E402
+
+FGMRES_AGG = {
+    "config_version": 2,
+    "solver": {
+        "preconditioner": {
+            "algorithm": "AGGREGATION",
+            "solver": "AMG",
+            "smoother": "MULTICOLOR_DILU",
+            "presweeps": 0,
+            "postsweeps": 3,
+            "selector": "SIZE_2",
+            "coarse_solver": "DENSE_LU_SOLVER",
+            "max_iters": 1,
+            "min_coarse_rows": 32,
+            "relaxation_factor": 0.75,
+            "scope": "amg",
+            "max_levels": 50,
+            "cycle": "V",
+        },
+        "solver": "FGMRES",
+        "max_iters": 100,
+        "gmres_n_restart": 10,
+        "monitor_residual": 1,
+        "convergence": "RELATIVE_INI",
+        "tolerance": 1e-6,
+        "norm": "L2",
+    },
+}
+
+
+def build_local_matrix(size, rank, world, device):
+    """Rank-local z-slab of the global size x size x (size*world) 7-pt
+    Poisson. For world=1 this is exactly the 256^3 single-GPU config."""
+    if world == 1:
+        from amgx_amd.problems import poisson_3d
+        return poisson_3d(size, size, size, device=device), None
+    from amgx_amd.distributed.manager import DistributedManager
+    from amgx_amd.problems import poisson_3d_local
+    ro, cols, vals, row_start = poisson_3d_local(size, size, size, rank, world)
+    n_local = len(ro) - 1
+    A = DistributedManager.upload_global_csr(
+        ro, cols, vals, n_local, row_start, size * size * size * world,
+        device=device)
+    return A, A.manager
+
+
+def run_step(A, cfg, res, b):
+    x = torch.zeros_like(b)
+    solver = create_solver(cfg.root_scope(), resources=res)
+    t0 = time.perf_counter()
+    solver.setup(A)
+    t1 = time.perf_counter()
+    st = solver.solve(b, x, zero_initial_guess=True)
+    t2 = time.perf_counter()
+    return {
+        "setup_s": t1 - t0,
+        "solve_s": t2 - t1,
+        "total_s": t2 - t0,
+        "iterations": st.iterations,
+        "converged": bool(st.converged),
+        "residual": st.residuals[-1] / st.residuals[0] if st.residuals else None,
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--size", type=int, default=256,
+                    help="per-rank subdomain edge (size^3 rows per GPU)")
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world)
+
+    use_cuda = torch.cuda.is_available() if args.device is None \
+        else args.device.startswith("cuda")
+    device = args.device or (f"cuda:{local_rank}" if use_cuda else "cpu")
+    distributed = world > 1
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+    res = Resources(device, distributed=distributed)
+    dist = None
+    if distributed:
+        import torch.distributed as tdist
+        dist = tdist
+
+    cfg = AMGConfig.from_dict(FGMRES_AGG)
+    A, manager = build_local_matrix(args.size, rank, world, device)
+    n_local = args.size ** 3
+    g = torch.Generator().manual_seed(1234 + rank)
+    b = torch.rand(n_local, generator=g, dtype=torch.float64).to(device)
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        run_step(A, cfg, res, b)
+
+    barrier_sync()
+    t_start = time.perf_counter()
+    stats = [run_step(A, cfg, res, b) for _ in range(args.steps)]
+    barrier_sync()
+    t_total = time.perf_counter() - t_start
+
+    # max over ranks of the timed window
+    if dist is not None:
+        tt = torch.tensor([t_total], dtype=torch.float64,
+                          device=device if res.is_cuda else "cpu")
+        dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+        t_total = float(tt.item())
+
+    sec_per_step = t_total / args.steps
+    last = stats[-1]
+    dof = args.size ** 3 * n_gpus
+    if rank == 0:
+        line = {
+            "metric": "amg_setup_solve_seconds_poisson256_fgmres_agg",
+            "value": sec_per_step,
+            "unit": "s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": sec_per_step * 1e3,
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "3D 7-pt Poisson, FGMRES + aggregation-AMG V-cycle "
+                         "(FGMRES_AGGREGATION.json), tol 1e-6 RELATIVE_INI",
+                "global_batch": dof,
+                "seq_len": args.size,
+                "parallelism": f"dd{n_gpus}",
+                "rows_per_gpu": args.size ** 3,
+                "iterations": last["iterations"],
+                "converged": last["converged"],
+                "setup_s": last["setup_s"],
+                "solve_s": last["solve_s"],
+                "final_rel_residual": last["residual"],
+            },
+        }
+        print(json.dumps(line))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,287 @@
+"""GPU numerics tests: every gfx950 kernel vs the plain CPU/scipy fp64
+reference (same op, same inputs)."""
+
+import numpy as np
+import pytest
+import scipy.sparse as sp
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from amgx_amd import AMGConfig, CSRMatrix, create_solver, ops
+from amgx_amd.amg.coloring import MatrixColoring
+from amgx_amd.resources import Resources
+from amgx_amd.problems import poisson_2d, poisson_3d, block_laplacian
+
+
+def to_gpu(A):
+    return A.to("cuda:0")
+
+
+def rand_csr(n=500, density=0.02, seed=3):
+    rng = np.random.RandomState(seed)
+    m = sp.random(n, n, density=density, random_state=rng, format="csr")
+    m = m + m.T + sp.identity(n) * 4.0
+    m = m.tocsr()
+    m.sum_duplicates()
+    m.sort_indices()
+    return CSRMatrix.from_scipy(m)
+
+
+def test_native_extension_loaded():
+    """The HIP extension must be the running path on GPU machines."""
+    from amgx_amd import _core
+    assert hasattr(_core, "csrmv")
+
+
+def test_spmv_scalar():
+    A = rand_csr()
+    x = torch.rand(A.n_rows, dtype=torch.float64)
+    ref = ops.spmv(A, x)
+    Ag = to_gpu(A)
+    y = ops.spmv(Ag, x.cuda())
+    assert torch.allclose(y.cpu(), ref, rtol=1e-12, atol=1e-12)
+
+
+def test_spmv_window_and_beta():
+    A = rand_csr()
+    x = torch.rand(A.n_rows, dtype=torch.float64)
+    y0 = torch.rand(A.n_rows, dtype=torch.float64)
+    ref = y0.clone()
+    ops.spmv(A, x, ref, alpha=2.0, beta=-0.5, row_begin=100, row_end=300)
+    yg = y0.cuda()
+    ops.spmv(to_gpu(A), x.cuda(), yg, alpha=2.0, beta=-0.5,
+             row_begin=100, row_end=300)
+    assert torch.allclose(yg.cpu(), ref, rtol=1e-12, atol=1e-12)
+
+
+def test_spmv_block():
+    A = block_laplacian(10, 8, block_dim=4)
+    x = torch.rand(A.n_rows * 4, dtype=torch.float64)
+    ref = ops.spmv(A, x)
+    y = ops.spmv(to_gpu(A), x.cuda())
+    assert torch.allclose(y.cpu(), ref, rtol=1e-12, atol=1e-12)
+
+
+def test_residual_and_reductions():
+    A = rand_csr()
+    x = torch.rand(A.n_rows, dtype=torch.float64)
+    b = torch.rand(A.n_rows, dtype=torch.float64)
+    r_ref = ops.residual(A, x, b)
+    Ag = to_gpu(A)
+    r = ops.residual(Ag, x.cuda(), b.cuda())
+    assert torch.allclose(r.cpu(), r_ref, rtol=1e-12, atol=1e-12)
+    assert np.isclose(ops.nrm2(r), ops.nrm2(r_ref), rtol=1e-12)
+    assert np.isclose(ops.nrm1(r), ops.nrm1(r_ref), rtol=1e-12)
+    assert np.isclose(ops.nrmmax(r), ops.nrmmax(r_ref), rtol=1e-12)
+    assert np.isclose(ops.dot(r, r), ops.dot(r_ref, r_ref), rtol=1e-12)
+
+
+def test_blas1():
+    x = torch.rand(10000, dtype=torch.float64)
+    y = torch.rand(10000, dtype=torch.float64)
+    xg, yg = x.cuda(), y.cuda()
+    ops.axpy(y, x, 0.7)
+    ops.axpy(yg, xg, 0.7)
+    assert torch.allclose(yg.cpu(), y, rtol=1e-14)
+    ops.axpby(y, x, 1.3, -0.2)
+    ops.axpby(yg, xg, 1.3, -0.2)
+    assert torch.allclose(yg.cpu(), y, rtol=1e-13)
+
+
+def test_diag_and_jacobi():
+    A = rand_csr()
+    Ag = to_gpu(A)
+    assert torch.equal(Ag.diag_index().cpu(), A.diag_index())
+    assert torch.allclose(Ag.diagonal().cpu(), A.diagonal())
+    for l1 in (False, True):
+        dref = ops.jacobi_dinv(A, l1=l1)
+        dg = ops.jacobi_dinv(Ag, l1=l1)
+        assert torch.allclose(dg.cpu(), dref, rtol=1e-12), f"l1={l1}"
+    b = torch.rand(A.n_rows, dtype=torch.float64)
+    x = torch.rand(A.n_rows, dtype=torch.float64)
+    out_ref = torch.zeros_like(x)
+    out_g = torch.zeros_like(x).cuda()
+    ops.jacobi_smooth(A, dref, b, x, out_ref, 0.8)
+    ops.jacobi_smooth(Ag, dg, b.cuda(), x.cuda(), out_g, 0.8)
+    assert torch.allclose(out_g.cpu(), out_ref, rtol=1e-12, atol=1e-13)
+
+
+def test_jacobi_block():
+    A = block_laplacian(8, 8, block_dim=4)
+    Ag = to_gpu(A)
+    dref = ops.jacobi_dinv(A)
+    dg = ops.jacobi_dinv(Ag)
+    assert torch.allclose(dg.cpu(), dref.to(dg.dtype), rtol=1e-10, atol=1e-12)
+    n = A.n_rows * 4
+    b = torch.rand(n, dtype=torch.float64)
+    x = torch.rand(n, dtype=torch.float64)
+    o1, o2 = torch.zeros_like(x), torch.zeros_like(x).cuda()
+    ops.jacobi_smooth(A, dref, b, x, o1, 0.8)
+    ops.jacobi_smooth(Ag, dg, b.cuda(), x.cuda(), o2, 0.8)
+    assert torch.allclose(o2.cpu(), o1, rtol=1e-10, atol=1e-12)
+
+
+def test_coloring_valid_gpu():
+    A = to_gpu(poisson_2d(50, 40))
+    col = MatrixColoring.create(A)
+    assert col.validate(A)
+    assert col.num_colors <= 8
+    assert sum(col.bounds[c + 1] - col.bounds[c]
+               for c in range(col.num_colors)) == A.n_rows
+
+
+def test_gs_sweep_matches_cpu():
+    # same coloring on both devices => identical sweep order => same numbers
+    A = poisson_2d(20, 20)
+    Ag = to_gpu(A)
+    colg = MatrixColoring.create(Ag)
+    col_cpu = MatrixColoring(colg.colors.cpu(), colg.num_colors)
+    dinv = ops.jacobi_dinv(A)
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x1 = torch.zeros(A.n_rows, dtype=torch.float64)
+    x2 = x1.clone().cuda()
+    ops.gs_sweep(A, dinv, b, x1, col_cpu, 1.0)
+    ops.gs_sweep(Ag, ops.jacobi_dinv(Ag), b.cuda(), x2, colg, 1.0)
+    assert torch.allclose(x2.cpu(), x1, rtol=1e-12, atol=1e-13)
+
+
+def test_dilu_matches_cpu():
+    A = poisson_2d(16, 16)
+    Ag = to_gpu(A)
+    colg = MatrixColoring.create(Ag)
+    col_cpu = MatrixColoring(colg.colors.cpu(), colg.num_colors)
+    e_ref = ops.dilu_setup(A, col_cpu)
+    e_gpu = ops.dilu_setup(Ag, colg)
+    assert torch.allclose(e_gpu.cpu(), e_ref, rtol=1e-12, atol=1e-13)
+    r = torch.rand(A.n_rows, dtype=torch.float64)
+    x1 = torch.zeros(A.n_rows, dtype=torch.float64)
+    x2 = x1.clone().cuda()
+    ops.dilu_solve(A, e_ref, col_cpu, r, 0.9, x1)
+    ops.dilu_solve(Ag, e_gpu, colg, r.cuda(), 0.9, x2)
+    assert torch.allclose(x2.cpu(), x1, rtol=1e-11, atol=1e-12)
+
+
+def test_dilu_block_gpu():
+    A = block_laplacian(8, 8, block_dim=4)
+    Ag = to_gpu(A)
+    colg = MatrixColoring.create(Ag)
+    col_cpu = MatrixColoring(colg.colors.cpu(), colg.num_colors)
+    e_ref = ops.dilu_setup(A, col_cpu)
+    e_gpu = ops.dilu_setup(Ag, colg)
+    assert torch.allclose(e_gpu.cpu(), e_ref, rtol=1e-9, atol=1e-11)
+    r = torch.rand(A.n_rows * 4, dtype=torch.float64)
+    x1 = torch.zeros_like(r)
+    x2 = x1.clone().cuda()
+    ops.dilu_solve(A, e_ref, col_cpu, r, 1.0, x1)
+    ops.dilu_solve(Ag, e_gpu, colg, r.cuda(), 1.0, x2)
+    assert torch.allclose(x2.cpu(), x1, rtol=1e-9, atol=1e-11)
+
+
+def test_size2_and_galerkin():
+    A = poisson_2d(30, 30)
+    Ag = to_gpu(A)
+    agg, nc = ops.size2_matching(Ag)
+    agg_c = agg.cpu().numpy()
+    assert agg_c.min() >= 0 and agg_c.max() == nc - 1
+    assert A.n_rows // 4 <= nc <= A.n_rows * 3 // 4
+    Ac = ops.galerkin_aggregation(Ag, agg, nc)
+    n = A.n_rows
+    P = sp.csr_matrix((np.ones(n), (np.arange(n), agg_c)), shape=(n, nc))
+    ref = (P.T @ A.to_scipy() @ P).toarray()
+    assert np.allclose(Ac.to_scipy().toarray(), ref, rtol=1e-12, atol=1e-12)
+
+
+def test_galerkin_block_gpu():
+    A = block_laplacian(10, 10, block_dim=3)
+    Ag = to_gpu(A)
+    agg, nc = ops.size2_matching(Ag)
+    Ac = ops.galerkin_aggregation(Ag, agg, nc)
+    ref = (__import__("amgx_amd.ops.cpu", fromlist=["cpu"])
+           .galerkin_aggregation(A, agg.cpu(), nc))
+    assert np.allclose(Ac.to_scipy().toarray(), ref.to_scipy().toarray(),
+                       rtol=1e-12, atol=1e-12)
+
+
+def test_restrict_prolongate_gpu():
+    A = to_gpu(poisson_2d(20, 20))
+    agg, nc = ops.size2_matching(A)
+    r = torch.rand(A.n_rows, dtype=torch.float64, device="cuda")
+    rc = ops.restrict_agg(r, agg, nc)
+    P = sp.csr_matrix((np.ones(A.n_rows),
+                       (np.arange(A.n_rows), agg.cpu().numpy())),
+                      shape=(A.n_rows, nc))
+    assert np.allclose(rc.cpu().numpy(), P.T @ r.cpu().numpy())
+    x = torch.zeros(A.n_rows, dtype=torch.float64, device="cuda")
+    xc = torch.rand(nc, dtype=torch.float64, device="cuda")
+    ops.prolongate_agg(x, xc, agg)
+    assert np.allclose(x.cpu().numpy(), P @ xc.cpu().numpy())
+
+
+def test_spgemm_transpose_gpu():
+    A = rand_csr(300, seed=1)
+    B = rand_csr(300, seed=2)
+    Ag, Bg = to_gpu(A), to_gpu(B)
+    C = ops.spgemm(Ag, Bg)
+    ref = (A.to_scipy() @ B.to_scipy()).toarray()
+    assert np.allclose(C.to_scipy().toarray(), ref, rtol=1e-12, atol=1e-12)
+    At = ops.transpose(Ag)
+    assert np.allclose(At.to_scipy().toarray(), A.to_scipy().T.toarray())
+
+
+def test_dense_lu_gpu():
+    A = poisson_2d(8, 8)
+    cfg = AMGConfig.from_dict({"solver": "DENSE_LU_SOLVER"})
+    s = create_solver(cfg.root_scope(), resources=Resources("cuda:0"))
+    Ag = to_gpu(A)
+    s.setup(Ag)
+    b = torch.rand(A.n_rows, dtype=torch.float64, device="cuda")
+    x = torch.zeros_like(b)
+    s.solve(b, x)
+    r = ops.residual(Ag, x, b)
+    assert ops.nrm2(r) < 1e-10
+
+
+def _solve_gpu(cfg_dict, A, tol=1e-6):
+    cfg = AMGConfig.from_dict(cfg_dict)
+    s = create_solver(cfg.root_scope(), resources=Resources("cuda:0"))
+    b = torch.ones(A.n_rows * A.block_dim, dtype=A.dtype, device=A.device)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
+    return st, rel
+
+
+def test_fgmres_agg_poisson3d_gpu():
+    from tests.test_amg import FGMRES_AGG
+    A = to_gpu(poisson_3d(24, 24, 24))
+    st, rel = _solve_gpu(FGMRES_AGG, A)
+    assert st.converged and rel < 1e-5
+    assert st.iterations <= 30
+
+
+def test_pcg_amg_gpu():
+    cfg = {
+        "solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "AGGREGATION",
+                "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+                "max_iters": 1, "min_coarse_rows": 32, "cycle": "V",
+            },
+            "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+            "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+        }
+    }
+    A = to_gpu(poisson_3d(16, 16, 16))
+    st, rel = _solve_gpu(cfg, A, tol=1e-8)
+    assert st.converged and rel < 1e-7
+
+
+def test_block_jacobi_pcg_gpu():
+    A = to_gpu(block_laplacian(12, 12, block_dim=4))
+    cfg = {"solver": {"solver": "PCG", "preconditioner": "BLOCK_JACOBI",
+                      "max_iters": 500, "monitor_residual": 1,
+                      "tolerance": 1e-8, "convergence": "RELATIVE_INI"}}
+    st, rel = _solve_gpu(cfg, A)
+    assert st.converged and rel < 1e-6
