@@ -88,6 +88,11 @@ class AMGHierarchy:
 
     def _setup_coarse_solver(self):
         coarsest = self.levels[-1]
+        if getattr(coarsest.A, "manager", None) is not None:
+            from ..solvers.dense_lu import GatheredDenseLU
+            self.coarse_solver = GatheredDenseLU(self.res)
+            self.coarse_solver.setup(coarsest.A)
+            return
         name, sub = self.scope.sub_solver("coarse_solver", "DENSE_LU_SOLVER")
         if name == "DENSE_LU_SOLVER":
             maxr = self.scope.get("dense_lu_max_rows")

@@ -25,8 +25,9 @@ class AMGLevel:
         self.index = index
         self.next: Optional[AMGLevel] = None
         self.smoother = None
-        # work vectors allocated at setup
-        n = A.n_rows * A.block_dim
+        # work vectors allocated at setup (halo-extended when distributed)
+        mgr = getattr(A, "manager", None)
+        n = mgr.ext_size if mgr is not None else A.n_rows * A.block_dim
         self.r = torch.zeros(n, dtype=A.dtype, device=A.device)
         self.bc = None   # coarse rhs
         self.xc = None   # coarse correction
@@ -45,7 +46,8 @@ class AMGLevel:
         raise NotImplementedError
 
     def alloc_coarse_vectors(self, Ac: CSRMatrix):
-        nc = Ac.n_rows * Ac.block_dim
+        mgr = getattr(Ac, "manager", None)
+        nc = mgr.ext_size if mgr is not None else Ac.n_rows * Ac.block_dim
         self.bc = torch.zeros(nc, dtype=Ac.dtype, device=Ac.device)
         self.xc = torch.zeros(nc, dtype=Ac.dtype, device=Ac.device)
 
@@ -65,6 +67,9 @@ class AggregationLevel(AMGLevel):
         selector = self.scope.get("selector")
         passes = {"SIZE_2": 1, "SIZE_4": 2, "SIZE_8": 3,
                   "MULTI_PAIRWISE": 2, "DUMMY": 1}.get(selector, 1)
+        mgr = getattr(self.A, "manager", None)
+        if mgr is not None:
+            passes = 1   # distributed multi-pass selectors: later round
         maxit = self.scope.get("max_matching_iterations")
         agg, num = ops.size2_matching(self.A, max_iterations=maxit)
         work = self.A
@@ -81,14 +86,52 @@ class AggregationLevel(AMGLevel):
         return num
 
     def create_coarse_matrix(self) -> CSRMatrix:
-        Ac = ops.galerkin_aggregation(self.A, self.aggregates,
-                                      self.num_aggregates)
+        mgr = getattr(self.A, "manager", None)
+        if mgr is None:
+            return ops.galerkin_aggregation(self.A, self.aggregates,
+                                            self.num_aggregates)
+        return self._create_coarse_matrix_distributed(mgr)
+
+    def _create_coarse_matrix_distributed(self, mgr) -> CSRMatrix:
+        """Distributed Galerkin (reference prepareNextLevelMatrix +
+        setNeighborAggregates, src/aggregation/aggregation_amg_level.cu:
+        1221-1560): aggregates are rank-local; halo columns map to REMOTE
+        coarse aggregates, whose global ids arrive by one halo exchange; the
+        product is a local CSR with global coarse columns that is then
+        re-uploaded to build the coarse level's own halo structure."""
+        import torch.distributed as tdist
+        from ..distributed.manager import DistributedManager
+        A = self.A
+        nc_local = self.num_aggregates
+        counts = [None] * mgr.world
+        tdist.all_gather_object(counts, nc_local)
+        coarse_start = int(sum(counts[:mgr.rank]))
+        n_global_c = int(sum(counts))
+        # global coarse id of every column (owned + halo)
+        aggv = torch.zeros((A.n_cols,), dtype=torch.float64, device=A.device)
+        aggv[:mgr.n_local] = self.aggregates.to(torch.float64) + coarse_start
+        mgr.exchange_halo(aggv, block_override=1)
+        agg_col = aggv.round().to(torch.int32)
+        Ac_local = ops.galerkin_aggregation(A, self.aggregates, nc_local,
+                                            agg_col, n_global_c)
+        # rebuild distributed structure from global column ids (host pass)
+        ro = Ac_local.row_offsets.cpu().numpy()
+        ci = Ac_local.col_indices.cpu().numpy()
+        va = Ac_local.values.cpu().numpy().reshape(Ac_local.nnz, -1) \
+            if A.block_dim > 1 else Ac_local.values.cpu().numpy()
+        Ac = DistributedManager.upload_global_csr(
+            ro, ci, va, nc_local, coarse_start, n_global_c,
+            device=A.device, block_dim=A.block_dim, dtype=A.dtype)
+        # coarse rows were renumbered interior-first by the upload: compose
+        # the fine->coarse map with that renumbering
+        iperm = Ac.manager.row_iperm.to(self.aggregates.device)
+        self.aggregates = iperm[self.aggregates.long()].to(torch.int32)
         return Ac
 
     def restrict_residual(self, r, bc):
         out = ops.restrict_agg(r, self.aggregates, self.num_aggregates,
                                self.A.block_dim)
-        bc.copy_(out.reshape(bc.shape))
+        bc.reshape(-1)[:out.numel()].copy_(out.reshape(-1))
 
     def prolongate_and_apply(self, xc, x):
         ops.prolongate_agg(x, xc, self.aggregates, self.A.block_dim)

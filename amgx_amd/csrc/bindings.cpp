@@ -358,7 +358,9 @@ void scatter(Tensor src, Tensor idx, int64_t b, Tensor dst, bool add) {
 // ---------------------------------------------------------------- SpGEMM
 std::tuple<Tensor, Tensor, Tensor> galerkin_agg(Tensor ro, Tensor ci,
                                                 Tensor va, Tensor agg,
-                                                int64_t nc, int64_t block_dim) {
+                                                Tensor agg_col, int64_t nc,
+                                                int64_t ncmod,
+                                                int64_t block_dim) {
     int n = (int)(ro.numel() - 1);
     long long nnz = ci.numel();
     int bb = (int)(block_dim * block_dim);
@@ -371,9 +373,9 @@ std::tuple<Tensor, Tensor, Tensor> galerkin_agg(Tensor ro, Tensor ci,
     DISPATCH_FT(va, "galerkin_agg", [&] {
         nnz_c = amgx_hip::galerkin_agg<scalar_t>(
             ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
-            n, nnz, agg.data_ptr<int>(), (int)nc, ro_c.data_ptr<int>(),
-            ci_c.data_ptr<int>(), va_c.data_ptr<scalar_t>(), bb,
-            cur_stream());
+            n, nnz, agg.data_ptr<int>(), agg_col.data_ptr<int>(), (int)nc,
+            (long long)ncmod, ro_c.data_ptr<int>(), ci_c.data_ptr<int>(),
+            va_c.data_ptr<scalar_t>(), bb, cur_stream());
     });
     return {ro_c, ci_c.narrow(0, 0, nnz_c), va_c.narrow(0, 0, nnz_c)};
 }

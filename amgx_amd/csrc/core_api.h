@@ -119,10 +119,15 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
 // reduce_by_key -> CSR. Returns nnz_c; fills ro_c (nc+1); ci_c/va_c are
 // caller-allocated at worst case nnz (nnz_c <= nnz) and trimmed after.
 // bb = block_dim^2 (1 for scalar). Temps allocated internally (setup-time).
+// agg = per-ROW local coarse id (n entries); agg_col = per-COLUMN coarse id
+// (n_cols entries; equals agg for single-process, GLOBAL coarse ids for the
+// distributed path); nc = local coarse rows (ro_c size); ncmod = column-id
+// modulus (global coarse columns).
 template <typename T>
 long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
-                       long long nnz, const int* agg, int nc, int* ro_c,
-                       int* ci_c, T* va_c, int bb, hipStream_t s);
+                       long long nnz, const int* agg, const int* agg_col,
+                       int nc, long long ncmod, int* ro_c, int* ci_c, T* va_c,
+                       int bb, hipStream_t s);
 
 // General ESC SpGEMM: C = A(m x k) @ B(k x n). Outputs sized by caller at
 // the expansion worst case is impractical; instead ci_c/va_c must be sized

@@ -37,16 +37,53 @@ def extract_diagonal(A):
     return _backend(A).extract_diagonal(A)
 
 
+def _mgr(A):
+    m = getattr(A, "manager", None)
+    return m if (m is not None and m.neighbors) else None
+
+
 # ---------------------------------------------------------------------- SpMV
 def spmv(A, x, y=None, alpha=1.0, beta=0.0, row_begin=0, row_end=-1):
-    """y = alpha*A@x + beta*y over rows [row_begin, row_end). The row window is
-    the latency-hiding interior/boundary split (reference src/multiply.cu:95-111)."""
-    return _backend(A).spmv(A, x, y, alpha, beta, row_begin, row_end)
+    """y = alpha*A@x + beta*y over rows [row_begin, row_end).
+
+    Distributed: halo exchange (grouped RCCL send/recv into the vector tail)
+    overlapped with the interior-row SpMV, then boundary rows — the reference
+    latency-hiding split (src/multiply.cu:95-111) without host staging."""
+    B = _backend(A)
+    mgr = _mgr(A)
+    if mgr is None:
+        return B.spmv(A, x, y, alpha, beta, row_begin, row_end)
+    assert row_begin == 0 and row_end < 0, "windowed distributed spmv: internal"
+    if y is None:
+        y = _new_dist_vec(A, x)
+    reqs = mgr.exchange_halo(x, async_start=True)
+    B.spmv(A, x, y, alpha, beta, 0, mgr.boundary_start)
+    for rq in reqs:
+        rq.wait()
+    B.spmv(A, x, y, alpha, beta, mgr.boundary_start, A.n_rows)
+    return y
+
+
+def _new_dist_vec(A, like):
+    import torch
+    return torch.zeros(A.manager.ext_size, dtype=like.dtype,
+                       device=like.device)
 
 
 def residual(A, x, b, r=None):
     """r = b - A@x (reference: axmb, src/solvers/solver.cu compute_residual)."""
-    return _backend(A).residual(A, x, b, r)
+    B = _backend(A)
+    mgr = _mgr(A)
+    if mgr is None:
+        return B.residual(A, x, b, r)
+    if r is None:
+        r = _new_dist_vec(A, b)
+    reqs = mgr.exchange_halo(x, async_start=True)
+    B.residual(A, x, b, r, 0, mgr.boundary_start)
+    for rq in reqs:
+        rq.wait()
+    B.residual(A, x, b, r, mgr.boundary_start, A.n_rows)
+    return r
 
 
 # ---------------------------------------------------------------------- BLAS-1
@@ -91,6 +128,9 @@ def jacobi_smooth(A, dinv, b, x_in, x_out, omega: float):
     """x_out = x_in + omega * dinv * (b - A x_in) — one damped-Jacobi sweep,
     fused single pass over A (Jacobi needs the OLD x, so in/out are separate;
     solvers ping-pong)."""
+    mgr = _mgr(A)
+    if mgr is not None:
+        mgr.exchange_halo(x_in)
     return _backend(A).jacobi_smooth(A, dinv, b, x_in, x_out, omega)
 
 
@@ -102,7 +142,12 @@ def gs_smooth_color(A, dinv, b, x, color_rows, omega: float):
 
 def gs_sweep(A, dinv, b, x, coloring, omega: float, symmetric: bool = False):
     """Full multicolor GS sweep (ascending colors, + descending when
-    symmetric). On GPU this is ONE extension call (per-color loop in C++)."""
+    symmetric). On GPU this is ONE extension call (per-color loop in C++).
+    Distributed: halo of x refreshed once per sweep (processor-lagged GS,
+    reference halo_coloring semantics, src/core.cu:354)."""
+    mgr = _mgr(A)
+    if mgr is not None:
+        mgr.exchange_halo(x)
     return _backend(A).gs_sweep(A, dinv, b, x, coloring, omega, symmetric)
 
 
@@ -121,11 +166,14 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
     return _backend(A).size2_matching(A, max_iterations, deterministic, seed)
 
 
-def galerkin_aggregation(A, aggregates, num_aggregates):
+def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
+                         ncols_mod=None):
     """Coarse A for piecewise-constant aggregation P:
     Ac[I,J] = sum_{i in I, j in J} A[i,j].
-    Reference: src/aggregation/coarseAgenerators/ (LOW_DEG / THRUST)."""
-    return _backend(A).galerkin_aggregation(A, aggregates, num_aggregates)
+    Reference: src/aggregation/coarseAgenerators/ (LOW_DEG / THRUST).
+    agg_col/ncols_mod: distributed variant (GLOBAL coarse column ids)."""
+    return _backend(A).galerkin_aggregation(A, aggregates, num_aggregates,
+                                            agg_col, ncols_mod)
 
 
 def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):

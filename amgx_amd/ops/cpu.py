@@ -69,12 +69,18 @@ def spmv(A, x, y=None, alpha=1.0, beta=0.0, row_begin=0, row_end=-1):
     return y
 
 
-def residual(A, x, b, r=None):
+def residual(A, x, b, r=None, row_begin=0, row_end=-1):
     m = A.to_scipy()
-    rv = _np(b).reshape(-1) - m @ _np(x).reshape(-1)
+    bd = A.block_dim
+    if row_end < 0:
+        row_end = A.n_rows
+    rv = _np(b).reshape(-1)[:A.n_rows * bd] - m @ _np(x).reshape(-1)
     if r is None:
-        return torch.from_numpy(rv.astype(_np(b).dtype))
-    _np(r).reshape(-1)[:] = rv
+        out = torch.zeros_like(b)
+        _np(out).reshape(-1)[:A.n_rows * bd] = rv
+        return out
+    _np(r).reshape(-1)[row_begin * bd:row_end * bd] = \
+        rv[row_begin * bd:row_end * bd]
     return r
 
 
@@ -209,6 +215,8 @@ def _strength_weights(A):
     (reference: size2 selector edge weights,
     src/aggregation/selectors/size2_selector.cu)."""
     m = _csr(A) if A.block_dim == 1 else _block_norm_csr(A)
+    if m.shape[1] != m.shape[0]:
+        m = m[:, :m.shape[0]]  # distributed: ignore halo columns in matching
     ad = np.abs(m.diagonal())
     ad = np.where(ad > 0, ad, 1.0)
     absm = abs(m.tocsr())
@@ -283,11 +291,43 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
     return torch.from_numpy(agg.astype(np.int32)), int(next_id)
 
 
-def galerkin_aggregation(A, aggregates, num_aggregates):
+def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
+                         ncols_mod=None):
     from ..matrix import CSRMatrix
     agg = _np(aggregates).astype(np.int64)
     n = A.n_rows
     bd = A.block_dim
+    if agg_col is not None:
+        # distributed: rows -> local coarse ids, cols -> global coarse ids
+        aggc = _np(agg_col).astype(np.int64)
+        ro = _np(A.row_offsets).astype(np.int64)
+        ci = _np(A.col_indices).astype(np.int64)
+        vals = _np(A.values)
+        rows = np.repeat(np.arange(n), np.diff(ro))
+        I, J = agg[rows], aggc[ci]
+        key = I * int(ncols_mod) + J
+        order = np.argsort(key, kind="stable")
+        key_s = key[order]
+        uniq, start = np.unique(key_s, return_index=True)
+        if bd == 1:
+            sums = np.add.reduceat(vals[order], start)
+        else:
+            sums = np.add.reduceat(vals[order], start, axis=0)
+        ro_c = np.zeros(num_aggregates + 1, dtype=np.int64)
+        np.add.at(ro_c, (uniq // int(ncols_mod)) + 1, 1)
+        ro_c = np.cumsum(ro_c)
+        ci_c = (uniq % int(ncols_mod)).astype(np.int32)
+        if bd == 1:
+            out = CSRMatrix(torch.from_numpy(ro_c.astype(np.int32)),
+                            torch.from_numpy(ci_c),
+                            torch.from_numpy(sums).to(A.dtype),
+                            n_cols=int(ncols_mod))
+        else:
+            out = CSRMatrix(torch.from_numpy(ro_c.astype(np.int32)),
+                            torch.from_numpy(ci_c),
+                            torch.from_numpy(sums).to(A.dtype),
+                            n_cols=int(ncols_mod), block_dim=bd)
+        return out
     if bd == 1:
         m = _csr(A)
         P = sp.csr_matrix((np.ones(n), (np.arange(n), agg)),
@@ -315,6 +355,7 @@ def galerkin_aggregation(A, aggregates, num_aggregates):
 
 def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):
     agg = aggregates.to(torch.int64)
+    r = r.reshape(-1)[:agg.numel() * block_dim]   # owned prefix (distributed)
     rc = torch.zeros(num_aggregates * block_dim, dtype=r.dtype)
     if block_dim == 1:
         rc.index_add_(0, agg, r.reshape(-1))
@@ -327,12 +368,13 @@ def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):
 
 def prolongate_agg(x, xc, aggregates, block_dim: int = 1):
     agg = aggregates.to(torch.int64)
+    xo = x.reshape(-1)[:agg.numel() * block_dim]   # owned prefix (distributed)
     if block_dim == 1:
-        x.reshape(-1).add_(xc.reshape(-1)[agg])
+        xo.add_(xc.reshape(-1)[agg])
     else:
         idx = (agg[:, None] * block_dim +
                torch.arange(block_dim, dtype=torch.int64)[None, :]).reshape(-1)
-        x.reshape(-1).add_(xc.reshape(-1)[idx])
+        xo.add_(xc.reshape(-1)[idx])
     return x
 
 
@@ -398,6 +440,8 @@ def dilu_setup(A, coloring):
     colors = _np(coloring.colors).astype(np.int64)
     if bd == 1:
         m = _csr(A)
+        if m.shape[1] != m.shape[0]:
+            m = m[:, :m.shape[0]].tocsr()  # distributed: local couplings only
         d = m.diagonal().astype(np.float64)
         # B_ij = a_ij * a_ji where both exist
         B = m.multiply(m.T).tocsr()

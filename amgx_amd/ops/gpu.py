@@ -69,12 +69,14 @@ def spmv(A, x, y=None, alpha=1.0, beta=0.0, row_begin=0, row_end=-1):
     return y
 
 
-def residual(A, x, b, r=None):
+def residual(A, x, b, r=None, row_begin=0, row_end=-1):
     if r is None:
         r = torch.empty_like(b)
+    if row_end < 0:
+        row_end = A.n_rows
     _core.csrmv(A.row_offsets, A.col_indices, A.values, A.block_dim,
                 x.reshape(-1), r.reshape(-1), b.reshape(-1), -1.0, 0.0, 1.0,
-                0, A.n_rows)
+                row_begin, row_end)
     return r
 
 
@@ -158,7 +160,7 @@ def dilu_setup(A, coloring):
 
 
 def dilu_solve(A, Einv, coloring, r, relaxation, x):
-    n = A.n_rows * A.block_dim
+    n = A.n_cols * A.block_dim   # ext size: halo tails stay zero in the sweeps
     w = _scratch(A, "dilu_w", n)
     z = _scratch(A, "dilu_z", n)
     _core.dilu_apply(A.row_offsets, A.col_indices, A.values, A.block_dim,
@@ -186,13 +188,21 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
     return agg.to(torch.int32), int(uniq.numel())
 
 
-def galerkin_aggregation(A, aggregates, num_aggregates):
+def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
+                         ncols_mod=None):
+    """agg_col/ncols_mod: distributed variant — per-column coarse ids (GLOBAL)
+    and the global coarse column count; defaults to the single-process case."""
     from ..matrix import CSRMatrix
+    if agg_col is None:
+        agg_col = aggregates
+    if ncols_mod is None:
+        ncols_mod = num_aggregates
     ro_c, ci_c, va_c = _core.galerkin_agg(A.row_offsets, A.col_indices,
-                                          A.values, aggregates,
-                                          num_aggregates, A.block_dim)
+                                          A.values, aggregates, agg_col,
+                                          num_aggregates, int(ncols_mod),
+                                          A.block_dim)
     out = CSRMatrix(ro_c, ci_c.contiguous(), va_c.contiguous(),
-                    n_cols=num_aggregates, block_dim=A.block_dim)
+                    n_cols=int(ncols_mod), block_dim=A.block_dim)
     return out
 
 

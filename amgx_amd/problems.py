@@ -49,34 +49,28 @@ def poisson_3d_local(nx: int, ny: int, nz: int, rank: int, world: int):
     NZg = nz * world
     n_local = nx * ny * nz
     row_start = rank * n_local
-    ro = [0]
-    cols = []
-    vals = []
-    for z in range(nz):
-        gz = rank * nz + z
-        for y in range(ny):
-            for x in range(nx):
-                gid = (gz * ny + y) * nx + x
-                row = []
-                if gz > 0:
-                    row.append((gid - nx * ny, -1.0))
-                if y > 0:
-                    row.append((gid - nx, -1.0))
-                if x > 0:
-                    row.append((gid - 1, -1.0))
-                row.append((gid, 6.0))
-                if x < nx - 1:
-                    row.append((gid + 1, -1.0))
-                if y < ny - 1:
-                    row.append((gid + nx, -1.0))
-                if gz < NZg - 1:
-                    row.append((gid + nx * ny, -1.0))
-                for c, v in row:
-                    cols.append(c)
-                    vals.append(v)
-                ro.append(len(cols))
-    return (np.asarray(ro, dtype=np.int64), np.asarray(cols, dtype=np.int64),
-            np.asarray(vals, dtype=np.float64), row_start)
+    # vectorized stencil assembly
+    z = np.arange(nz).repeat(ny * nx)
+    y = np.tile(np.arange(ny).repeat(nx), nz)
+    x = np.tile(np.arange(nx), nz * ny)
+    gz = z + rank * nz
+    gid = (gz * ny + y) * nx + x
+    neigh = [
+        (gz > 0, -nx * ny), (y > 0, -nx), (x > 0, -1),
+        (np.ones(n_local, dtype=bool), 0),
+        (x < nx - 1, 1), (y < ny - 1, nx), (gz < NZg - 1, nx * ny),
+    ]
+    mask = np.stack([m for m, _ in neigh])            # (7, n)
+    offs = np.asarray([o for _, o in neigh])
+    deg = mask.sum(axis=0)
+    ro = np.zeros(n_local + 1, dtype=np.int64)
+    np.cumsum(deg, out=ro[1:])
+    colmat = gid[None, :] + offs[:, None]             # (7, n)
+    valmat = np.where(offs[:, None] == 0, 6.0, -1.0) * np.ones((7, n_local))
+    sel = mask.T.reshape(-1)                          # row-major per row
+    cols = colmat.T.reshape(-1)[sel]
+    vals = valmat.T.reshape(-1)[sel]
+    return ro, cols.astype(np.int64), vals.astype(np.float64), row_start
 
 
 def block_laplacian(nx: int, ny: int, block_dim: int = 4, device="cpu",

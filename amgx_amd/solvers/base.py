@@ -136,7 +136,16 @@ class Solver:
         pass
 
     # -- norms ---------------------------------------------------------------
+    def _owned(self, x: torch.Tensor) -> torch.Tensor:
+        """Owned prefix of a (possibly halo-extended) distributed vector
+        (reference ViewType OWNED, include/vector.h:18-27)."""
+        mgr = getattr(self.A, "manager", None) if self.A is not None else None
+        if mgr is not None:
+            return x.reshape(-1)[:mgr.owned_size]
+        return x
+
     def compute_norm(self, r: torch.Tensor) -> float:
+        r = self._owned(r)
         if self.norm == "L2":
             nrm = ops.nrm2(r)
         elif self.norm == "L1":
@@ -150,7 +159,7 @@ class Solver:
         return nrm
 
     def dot(self, x, y) -> float:
-        d = ops.dot(x, y)
+        d = ops.dot(self._owned(x), self._owned(y))
         if self.A is not None and getattr(self.A, "manager", None) is not None:
             d = self.A.manager.global_sum(d)
         return d

@@ -7,6 +7,8 @@ standalone or composed as AMG smoothers); one ``solve_iteration`` = one sweep.
 
 from __future__ import annotations
 
+import math
+
 import torch
 
 from .. import ops
@@ -129,25 +131,32 @@ class ChebyshevSolver(_SmootherBase):
         self._init_cheb()
 
     def _power_iteration(self, iters: int) -> float:
-        n = self.A.n_rows * self.A.block_dim
+        mgr = getattr(self.A, "manager", None)
+        n = mgr.ext_size if mgr is not None \
+            else self.A.n_rows * self.A.block_dim
         g = torch.Generator().manual_seed(7177)
         v = torch.rand(n, generator=g, dtype=torch.float64) \
             .to(self.A.dtype).to(self.A.device)
         Av = torch.zeros_like(v)
         lam = 1.0
         for _ in range(iters):
-            nv = ops.nrm2(v)
+            nv = math.sqrt(max(self.dot(v, v), 0.0))
             if nv == 0:
                 break
             ops.scal(v, 1.0 / nv)
             ops.spmv(self.A, v, Av)
             self._apply_dinv(Av)
-            lam = ops.dot(v, Av)
+            lam = self.dot(v, Av)
             v, Av = Av, v
         return abs(lam) * 1.05   # safety factor
 
     def _apply_dinv(self, v):
         if self.A.block_dim == 1:
+            mgr = getattr(self.A, "manager", None)
+            if mgr is not None:
+                vo = v.reshape(-1)[:mgr.owned_size]
+                vo.mul_(self.dinv.reshape(-1))
+                return
             v.mul_(self.dinv.reshape(-1))
         else:
             bd = self.A.block_dim

@@ -1,0 +1,168 @@
+"""Distributed-layer tests: world_size=2 over gloo on CPU (the same code path
+drives RCCL on MI355X). Reference test strategy: in-process multi-partition
+checks upgraded to REAL multi-process coverage."""
+
+import json
+import os
+import tempfile
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _run_dist(fn, world=2, args=()):
+    ctx = mp.get_context("spawn")
+    port = str(29600 + abs(hash(fn.__name__)) % 200)
+    with tempfile.TemporaryDirectory() as tmp:
+        procs = [ctx.Process(target=_entry,
+                             args=(fn.__name__, r, world, port, tmp, args))
+                 for r in range(world)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(180)
+        for p in procs:
+            assert p.exitcode == 0, f"rank failed: exitcode={p.exitcode}"
+
+
+def _entry(fn_name, rank, world, port, tmp, args):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = port
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        globals()["_impl_" + fn_name](rank, world, tmp, *args)
+    finally:
+        dist.destroy_process_group()
+
+
+def _make_dist_A(rank, world, n=4):
+    from amgx_amd.distributed.manager import DistributedManager
+    from amgx_amd.problems import poisson_3d_local
+    ro, cols, vals, rs = poisson_3d_local(n, n, n, rank, world)
+    return DistributedManager.upload_global_csr(
+        ro, cols, vals, n * n * n, rs, n * n * n * world, device="cpu")
+
+
+# --------------------------------------------------------------------- spmv
+def test_dist_spmv():
+    _run_dist(test_dist_spmv)
+
+
+def _impl_test_dist_spmv(rank, world, tmp):
+    from amgx_amd import ops
+    from amgx_amd.problems import poisson_3d
+    n = 4
+    A = _make_dist_A(rank, world, n)
+    mgr = A.manager
+    n_local = n * n * n
+    n_global = n_local * world
+    # global x; my owned slice permuted into internal order
+    g = torch.Generator().manual_seed(99)
+    xg = torch.rand(n_global, generator=g, dtype=torch.float64)
+    x = mgr.new_ext_vec(torch.float64)
+    mine = xg[mgr.row_start:mgr.row_start + n_local]
+    x[:mgr.owned_size] = mine[mgr.row_perm]
+    y = ops.spmv(A, x)
+    # reference
+    ref = (poisson_3d(n, n, n * world).to_scipy()
+           @ xg.numpy())[mgr.row_start:mgr.row_start + n_local]
+    y_user = mgr.permute_out(y).numpy()
+    assert np.allclose(y_user, ref, rtol=1e-13, atol=1e-13)
+    # residual + norms
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    r = ops.residual(A, x, b)
+    ref_r = 1.0 - ref
+    assert np.allclose(mgr.permute_out(r).numpy(), ref_r, atol=1e-13)
+
+
+# ---------------------------------------------------------------- reductions
+def test_dist_reductions():
+    _run_dist(test_dist_reductions)
+
+
+def _impl_test_dist_reductions(rank, world, tmp):
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.resources import Resources
+    A = _make_dist_A(rank, world)
+    mgr = A.manager
+    cfg = AMGConfig.from_dict({"solver": "CG"})
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu", distributed=True))
+    s.A = A
+    x = mgr.new_ext_vec(torch.float64)
+    x[:mgr.owned_size] = float(rank + 1)
+    n_local = mgr.n_local
+    expect = sum((r + 1) ** 2 * n_local for r in range(world))
+    assert abs(s.dot(x, x) - expect) < 1e-10
+    assert abs(s.compute_norm(x) - np.sqrt(expect)) < 1e-10
+
+
+# --------------------------------------------------------------------- solve
+def test_dist_fgmres_agg():
+    _run_dist(test_dist_fgmres_agg)
+
+
+def _impl_test_dist_fgmres_agg(rank, world, tmp):
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    from tests.test_amg import FGMRES_AGG
+    n = 8
+    A = _make_dist_A(rank, world, n)
+    mgr = A.manager
+    cfg = AMGConfig.from_dict(FGMRES_AGG)
+    res = Resources("cpu", distributed=True)
+    s = create_solver(cfg.root_scope(), resources=res)
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    r = ops.residual(A, x, b)
+    nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+        r[:mgr.owned_size])), "L2")
+    bn = mgr.global_norm(float(torch.linalg.vector_norm(
+        b[:mgr.owned_size])), "L2")
+    assert st.converged, f"not converged: {st}"
+    assert nrm / bn < 1e-5
+    assert st.iterations <= 40
+    if rank == 0:
+        with open(os.path.join(tmp, "iters.json"), "w") as f:
+            json.dump({"iters": st.iterations}, f)
+
+
+# ------------------------------------------------------------------ pcg + gs
+def test_dist_pcg_amg_gs():
+    _run_dist(test_dist_pcg_amg_gs)
+
+
+def _impl_test_dist_pcg_amg_gs(rank, world, tmp):
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({
+        "solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "AGGREGATION",
+                "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+                "max_iters": 1, "min_coarse_rows": 16, "cycle": "V",
+            },
+            "solver": "PCG", "max_iters": 120, "monitor_residual": 1,
+            "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+        }
+    })
+    A = _make_dist_A(rank, world, 6)
+    mgr = A.manager
+    res = Resources("cpu", distributed=True)
+    s = create_solver(cfg.root_scope(), resources=res)
+    b = mgr.new_ext_vec(torch.float64)
+    g = torch.Generator().manual_seed(3 + rank)
+    b[:mgr.owned_size] = torch.rand(mgr.owned_size, generator=g,
+                                    dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged, f"rank {rank}: {st}"
