@@ -416,6 +416,129 @@ std::tuple<Tensor, Tensor, Tensor> transpose(Tensor ro, Tensor ci, Tensor va,
     return {ro_t, ci_t, va_t};
 }
 
+// ---------------------------------------------------------------- classical
+Tensor strength_ahat(Tensor ro, Tensor ci, Tensor va, Tensor didx,
+                     double theta, double max_row_sum) {
+    int n = (int)(ro.numel() - 1);
+    auto strong = torch::empty({ci.numel()}, ro.options().dtype(torch::kUInt8));
+    DISPATCH_FT(va, "strength_ahat", [&] {
+        amgx_hip::strength_ahat<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            didx.data_ptr<int>(), n, theta, max_row_sum,
+            strong.data_ptr<unsigned char>(), cur_stream());
+    });
+    return strong;
+}
+
+Tensor pmis_select(Tensor ro, Tensor ci, Tensor tidx, Tensor strong,
+                   int64_t max_rounds) {
+    int n = (int)(ro.numel() - 1);
+    auto w = torch::empty({n}, ro.options().dtype(torch::kFloat32));
+    auto state = torch::zeros({n}, ro.options().dtype(torch::kChar));
+    auto state_mid = torch::zeros_like(state);
+    auto state_out = torch::zeros_like(state);
+    auto counter = torch::zeros({1}, ro.options().dtype(torch::kInt32));
+    hipStream_t st = cur_stream();
+    amgx_hip::pmis_lambda(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                          tidx.data_ptr<int>(),
+                          strong.data_ptr<unsigned char>(), n,
+                          w.data_ptr<float>(), st);
+    amgx_hip::pmis_mark_isolated(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                                 tidx.data_ptr<int>(),
+                                 strong.data_ptr<unsigned char>(), n,
+                                 (signed char*)state.data_ptr(), st);
+    for (int round = 0; round < max_rounds; ++round) {
+        counter.zero_();
+        amgx_hip::pmis_one_round(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), tidx.data_ptr<int>(),
+            strong.data_ptr<unsigned char>(), n, w.data_ptr<float>(),
+            (signed char*)state.data_ptr(), (signed char*)state_mid.data_ptr(),
+            (signed char*)state_out.data_ptr(), counter.data_ptr<int>(), st);
+        std::swap(state, state_out);
+        if (counter.cpu().item<int>() == 0) break;
+    }
+    return state;   // 1 = C, -1 = F; Python builds cf_map via cumsum
+}
+
+std::tuple<Tensor, Tensor, Tensor> interp_d1(Tensor ro, Tensor ci, Tensor va,
+                                             Tensor strong, Tensor cf,
+                                             Tensor didx, Tensor p_ro,
+                                             int64_t p_nnz) {
+    int n = (int)(ro.numel() - 1);
+    auto p_ci = torch::empty({p_nnz}, ro.options());
+    auto p_va = torch::empty({p_nnz}, va.options());
+    DISPATCH_FT(va, "interp_d1", [&] {
+        amgx_hip::interp_d1<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            strong.data_ptr<unsigned char>(), cf.data_ptr<int>(),
+            didx.data_ptr<int>(), p_ro.data_ptr<int>(), n,
+            p_ci.data_ptr<int>(), p_va.data_ptr<scalar_t>(), cur_stream());
+    });
+    return {p_ro, p_ci, p_va};
+}
+
+Tensor interp_d1_count(Tensor ro, Tensor ci, Tensor strong, Tensor cf) {
+    int n = (int)(ro.numel() - 1);
+    auto counts = torch::empty({n}, ro.options());
+    amgx_hip::interp_d1_count(ro.data_ptr<int>(), ci.data_ptr<int>(),
+                              strong.data_ptr<unsigned char>(),
+                              cf.data_ptr<int>(), n, counts.data_ptr<int>(),
+                              cur_stream());
+    return counts;
+}
+
+// ---------------------------------------------------------------- ILU(0)
+Tensor ilu0_setup(Tensor ro, Tensor ci, Tensor va, Tensor didx, Tensor pos,
+                  Tensor rows_sorted, std::vector<int64_t> bounds) {
+    int n = (int)(ro.numel() - 1);
+    auto lu = va.clone();
+    int nc = (int)bounds.size() - 1;
+    DISPATCH_FT(va, "ilu0_setup", [&] {
+        for (int c = 0; c < nc; ++c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::ilu0_factor_color_launch<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(), pos.data_ptr<int>(),
+                didx.data_ptr<int>(), rows_sorted.data_ptr<int>() + s,
+                (int)(e - s), lu.data_ptr<scalar_t>(), n, cur_stream());
+        }
+    });
+    return lu;
+}
+
+void ilu0_apply(Tensor ro, Tensor ci, Tensor lu, Tensor didx, Tensor pos,
+                Tensor rows_sorted, std::vector<int64_t> bounds, Tensor r,
+                Tensor y, Tensor z, Tensor x, double relax) {
+    int n = (int)(ro.numel() - 1);
+    int nc = (int)bounds.size() - 1;
+    y.zero_();
+    z.zero_();
+    DISPATCH_FT(lu, "ilu0_apply", [&] {
+        hipStream_t st = cur_stream();
+        for (int c = 0; c < nc; ++c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::ilu0_fwd_launch<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(), pos.data_ptr<int>(),
+                lu.data_ptr<scalar_t>(), rows_sorted.data_ptr<int>() + s,
+                (int)(e - s), r.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                n, st);
+        }
+        for (int c = nc - 1; c >= 0; --c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::ilu0_bwd_launch<scalar_t>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(), pos.data_ptr<int>(),
+                lu.data_ptr<scalar_t>(), didx.data_ptr<int>(),
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                y.data_ptr<scalar_t>(), z.data_ptr<scalar_t>(), n, st);
+        }
+        amgx_hip::axpy<scalar_t>(x.data_ptr<scalar_t>(),
+                                 z.data_ptr<scalar_t>(), (scalar_t)relax,
+                                 x.numel(), st);
+    });
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -443,4 +566,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("galerkin_agg", &galerkin_agg);
     m.def("spgemm", &spgemm);
     m.def("transpose", &transpose);
+    m.def("strength_ahat", &strength_ahat);
+    m.def("pmis_select", &pmis_select);
+    m.def("interp_d1", &interp_d1);
+    m.def("interp_d1_count", &interp_d1_count);
+    m.def("ilu0_setup", &ilu0_setup);
+    m.def("ilu0_apply", &ilu0_apply);
 }

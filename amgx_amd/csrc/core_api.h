@@ -145,4 +145,39 @@ void transpose_csr(const int* ro, const int* ci, const T* va, int m, int n,
                    long long nnz, int* ro_t, int* ci_t, T* va_t,
                    hipStream_t s);
 
+// ---- classical setup (kernels_classical.hip) --------------------------------
+template <typename T>
+void strength_ahat(const int* ro, const int* ci, const T* va, const int* didx,
+                   int n, double theta, double max_row_sum,
+                   unsigned char* strong, hipStream_t s);
+void pmis_lambda(const int* ro, const int* ci, const int* tidx,
+                 const unsigned char* strong, int n, float* w, hipStream_t s);
+void pmis_mark_isolated(const int* ro, const int* ci, const int* tidx,
+                        const unsigned char* strong, int n, signed char* state,
+                        hipStream_t s);
+void pmis_one_round(const int* ro, const int* ci, const int* tidx,
+                    const unsigned char* strong, int n, const float* w,
+                    const signed char* state, signed char* state_mid,
+                    signed char* state_out, int* n_undecided, hipStream_t s);
+void interp_d1_count(const int* ro, const int* ci, const unsigned char* strong,
+                     const int* cf, int n, int* counts, hipStream_t s);
+template <typename T>
+void interp_d1(const int* ro, const int* ci, const T* va,
+               const unsigned char* strong, const int* cf, const int* didx,
+               const int* p_ro, int n, int* p_ci, T* p_va, hipStream_t s);
+
+// ---- ILU(0) (kernels_classical.hip) ------------------------------------------
+template <typename T>
+void ilu0_factor_color_launch(const int* ro, const int* ci, const int* pos,
+                              const int* didx, const int* rows, int count,
+                              T* lu, int n, hipStream_t s);
+template <typename T>
+void ilu0_fwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
+                     const int* rows, int count, const T* r, T* y, int n,
+                     hipStream_t s);
+template <typename T>
+void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
+                     const int* didx, const int* rows, int count, const T* y,
+                     T* z, int n, hipStream_t s);
+
 }  // namespace amgx_hip

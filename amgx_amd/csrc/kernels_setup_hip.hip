@@ -193,46 +193,48 @@ template <typename T>
 __global__ void make_agg_keys(const int* __restrict__ rows,
                               const int* __restrict__ ci,
                               const T* __restrict__ va,
-                              const int* __restrict__ agg, long long nnz,
-                              long long nc, unsigned long long* __restrict__ keys,
+                              const int* __restrict__ agg_row,
+                              const int* __restrict__ agg_col, long long nnz,
+                              long long ncmod, unsigned long long* __restrict__ keys,
                               T* __restrict__ vals) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long k = (long long)blockIdx.x * blockDim.x + threadIdx.x;
          k < nnz; k += stride) {
-        keys[k] = (unsigned long long)agg[rows[k]] * nc + agg[ci[k]];
+        keys[k] = (unsigned long long)agg_row[rows[k]] * ncmod + agg_col[ci[k]];
         vals[k] = va[k];
     }
 }
 
 // decompose sorted unique keys -> CSR of the coarse matrix
 __global__ void keys_to_csr(const unsigned long long* __restrict__ keys,
-                            long long nnz_c, long long nc,
+                            long long nnz_c, long long ncmod, long long nrows,
                             int* __restrict__ ro_c, int* __restrict__ ci_c) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
          t < nnz_c; t += stride) {
         unsigned long long key = keys[t];
-        int r = (int)(key / nc);
-        ci_c[t] = (int)(key % nc);
+        int r = (int)(key / ncmod);
+        ci_c[t] = (int)(key % ncmod);
         // row start: first t whose row is r
-        int rprev = (t == 0) ? -1 : (int)(keys[t - 1] / nc);
+        int rprev = (t == 0) ? -1 : (int)(keys[t - 1] / ncmod);
         if (r != rprev)
             for (int q = rprev + 1; q <= r; ++q) ro_c[q] = (int)t;
         if (t == nnz_c - 1)
-            for (long long q = r + 1; q <= nc; ++q) ro_c[q] = (int)nnz_c;
+            for (long long q = r + 1; q <= nrows; ++q) ro_c[q] = (int)nnz_c;
     }
 }
 
 __global__ void make_agg_keys_perm(const int* __restrict__ rows,
                                    const int* __restrict__ ci,
-                                   const int* __restrict__ agg, long long nnz,
-                                   long long nc,
+                                   const int* __restrict__ agg_row,
+                                   const int* __restrict__ agg_col,
+                                   long long nnz, long long ncmod,
                                    unsigned long long* __restrict__ keys,
                                    int* __restrict__ perm) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long k = (long long)blockIdx.x * blockDim.x + threadIdx.x;
          k < nnz; k += stride) {
-        keys[k] = (unsigned long long)agg[rows[k]] * nc + agg[ci[k]];
+        keys[k] = (unsigned long long)agg_row[rows[k]] * ncmod + agg_col[ci[k]];
         perm[k] = (int)k;
     }
 }
@@ -280,7 +282,8 @@ __global__ void sum_blocks_by_run(const int* __restrict__ run_starts,
 // ============================================================ galerkin (agg)
 template <typename T>
 long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
-                       long long nnz, const int* agg, int nc, int* ro_c,
+                       long long nnz, const int* agg, const int* agg_col,
+                       int nc, long long ncmod, int* ro_c,
                        int* ci_c, T* va_c, int bb, hipStream_t s) {
     using K = unsigned long long;
     int* rows = (int*)dev_alloc(nnz * sizeof(int), s);
@@ -288,14 +291,14 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
                        ro, n, rows);
     K* keys = (K*)dev_alloc(nnz * sizeof(K) * 2, s);
     K* keys_alt = keys + nnz;
-    long long nc_ll = nc;
+    long long nc_ll = ncmod;
 
     if (bb == 1) {
         T* vals = (T*)dev_alloc(nnz * sizeof(T) * 2, s);
         T* vals_alt = vals + nnz;
         hipLaunchKernelGGL((make_agg_keys<T>), dim3(grid_1d(nnz, AMGX_BLOCK, 4096)),
-                           dim3(AMGX_BLOCK), 0, s, rows, ci, va, agg, nnz,
-                           nc_ll, keys, vals);
+                           dim3(AMGX_BLOCK), 0, s, rows, ci, va, agg, agg_col,
+                           nnz, nc_ll, keys, vals);
         size_t tmp_bytes = 0;
         rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys, keys_alt, vals,
                                   vals_alt, nnz, 0, 64, s);
@@ -319,7 +322,7 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
         HIP_CHECK(hipStreamSynchronize(s));
         hipLaunchKernelGGL(keys_to_csr, dim3(grid_1d(h_runs, AMGX_BLOCK, 4096)),
                            dim3(AMGX_BLOCK), 0, s, ukeys, (long long)h_runs,
-                           nc_ll, ro_c, ci_c);
+                           nc_ll, (long long)nc, ro_c, ci_c);
         dev_free(tmp, s); dev_free(tmpb, s); dev_free(vals, s);
         dev_free(nruns, s); dev_free(keys, s); dev_free(rows, s);
         return (long long)h_runs;
@@ -328,8 +331,8 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
     int* perm = (int*)dev_alloc(nnz * sizeof(int) * 2, s);
     int* perm_alt = perm + nnz;
     hipLaunchKernelGGL(make_agg_keys_perm, dim3(grid_1d(nnz, AMGX_BLOCK, 4096)),
-                       dim3(AMGX_BLOCK), 0, s, rows, ci, agg, nnz, nc_ll,
-                       keys, perm);
+                       dim3(AMGX_BLOCK), 0, s, rows, ci, agg, agg_col, nnz,
+                       nc_ll, keys, perm);
     size_t tmp_bytes = 0;
     rocprim::radix_sort_pairs(nullptr, tmp_bytes, keys, keys_alt, perm,
                               perm_alt, nnz, 0, 64, s);
@@ -360,7 +363,7 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
                        (long long)h_runs, ukeys);
     hipLaunchKernelGGL(keys_to_csr, dim3(grid_1d(h_runs, AMGX_BLOCK, 4096)),
                        dim3(AMGX_BLOCK), 0, s, ukeys, (long long)h_runs, nc_ll,
-                       ro_c, ci_c);
+                       (long long)nc, ro_c, ci_c);
     hipLaunchKernelGGL((sum_blocks_by_run<T>),
                        dim3(grid_1d((long long)h_runs * bb, AMGX_BLOCK, 4096)),
                        dim3(AMGX_BLOCK), 0, s, rs_alt, perm_alt, va, h_runs,
@@ -466,7 +469,7 @@ long long spgemm_esc(const int* roA, const int* ciA, const T* vaA, int m,
     HIP_CHECK(hipStreamSynchronize(s));
     hipLaunchKernelGGL(keys_to_csr, dim3(grid_1d(h_runs, AMGX_BLOCK, 4096)),
                        dim3(AMGX_BLOCK), 0, s, keys, (long long)h_runs,
-                       (long long)n, ro_c, ci_c);
+                       (long long)n, (long long)m, ro_c, ci_c);
     dev_free(tmpb, s); dev_free(tmpc, s); dev_free(nruns, s);
     dev_free(vals, s); dev_free(keys, s); dev_free(rowsA, s); dev_free(deg, s);
     return (long long)h_runs;
@@ -550,8 +553,9 @@ void transpose_csr(const int* ro, const int* ci, const T* va, int m, int n,
                                           const int*, const T*, int,           \
                                           const int*, int*, hipStream_t);      \
     template long long galerkin_agg<T>(const int*, const int*, const T*, int,  \
-                                       long long, const int*, int, int*, int*, \
-                                       T*, int, hipStream_t);                  \
+                                       long long, const int*, const int*, int, \
+                                       long long, int*, int*, T*, int,         \
+                                       hipStream_t);                  \
     template long long spgemm_esc<T>(const int*, const int*, const T*, int,    \
                                      long long, const int*, const int*,        \
                                      const T*, int, int, int*, int*, T*,       \

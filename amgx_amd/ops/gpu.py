@@ -260,32 +260,58 @@ def dense_solve(Ainv, b, x):
 
 
 # ---------------------------------------------------------------------- classical
-# Classical-AMG setup ops run on host for now (setup-time roundtrip; the GPU
-# kernels for strength/PMIS/D1 are on the roadmap — solve path is fully GPU).
-def _host_roundtrip(fn_name, A, *args):
-    from . import cpu
-    host = A.to("cpu")
-    out = getattr(cpu, fn_name)(host, *args)
-    return out
-
-
 def strength_ahat(A, theta=0.25, max_row_sum=1.1):
-    return _host_roundtrip("strength_ahat", A, theta, max_row_sum)
+    return _core.strength_ahat(A.row_offsets, A.col_indices, A.values,
+                               _didx(A), float(theta), float(max_row_sum))
 
 
 def pmis_select(A, S):
-    cf, nc = _host_roundtrip("pmis_select", A, S)
-    return cf.to(A.device), nc
+    state = _core.pmis_select(A.row_offsets, A.col_indices, _tidx(A), S, 64)
+    cmask = state == 1
+    n = A.n_rows
+    cf = torch.full((n,), -1, dtype=torch.int32, device=A.device)
+    csum = torch.cumsum(cmask.to(torch.int64), 0)
+    cf[cmask] = (csum[cmask] - 1).to(torch.int32)
+    return cf, int(cmask.sum().item())
 
 
 def interp_d1(A, S, cf_map, num_coarse):
-    P = _host_roundtrip("interp_d1", A, S.cpu(), cf_map.cpu(), num_coarse)
-    return P.to(A.device)
+    from ..matrix import CSRMatrix
+    counts = _core.interp_d1_count(A.row_offsets, A.col_indices, S, cf_map)
+    p_ro = torch.zeros(A.n_rows + 1, dtype=torch.int32, device=A.device)
+    p_ro[1:] = torch.cumsum(counts.to(torch.int64), 0).to(torch.int32)
+    p_nnz = int(p_ro[-1].item())
+    p_ro, p_ci, p_va = _core.interp_d1(A.row_offsets, A.col_indices, A.values,
+                                       S, cf_map, _didx(A), p_ro, p_nnz)
+    return CSRMatrix(p_ro, p_ci, p_va, n_cols=num_coarse)
+
+
+# ---------------------------------------------------------------------- ILU(0)
+def _color_pos(A, coloring):
+    key = "ilu_pos"
+    pos = A._cache.get(key)
+    if pos is None:
+        n = A.n_rows
+        pos = torch.empty(n, dtype=torch.int32, device=A.device)
+        pos[coloring.rows_sorted.to(torch.int64)] = \
+            torch.arange(n, dtype=torch.int32, device=A.device)
+        A._cache[key] = pos
+    return pos
 
 
 def ilu0_setup(A, coloring):
-    raise NotImplementedError("GPU ILU(0) kernels: planned (use DILU)")
+    pos = _color_pos(A, coloring)
+    lu = _core.ilu0_setup(A.row_offsets, A.col_indices, A.values, _didx(A),
+                          pos, coloring.rows_sorted, coloring.bounds)
+    return lu
 
 
 def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
-    raise NotImplementedError("GPU ILU(0) kernels: planned (use DILU)")
+    n = A.n_cols * A.block_dim
+    y = _scratch(A, "ilu_y", n)
+    z = _scratch(A, "ilu_z", n)
+    pos = _color_pos(A, coloring)
+    _core.ilu0_apply(A.row_offsets, A.col_indices, factors, _didx(A), pos,
+                     coloring.rows_sorted, coloring.bounds, r.reshape(-1),
+                     y, z, x.reshape(-1), float(relaxation))
+    return x

@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(csrc, "bindings.cpp"),
         os.path.join(csrc, "kernels_solve.hip"),
         os.path.join(csrc, "kernels_setup.hip"),
+        os.path.join(csrc, "kernels_classical.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
