@@ -220,7 +220,17 @@ __global__ void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
             if (a < 0) neg_c += a; else pos_c += a;
         }
     }
-    if (diag == 0.0) return;   // empty row already counted 0
+    if (diag == 0.0) {
+        // degenerate row: emit zero weights for its counted slots
+        for (int k = s; k < e; ++k) {
+            if (ci[k] == i || !strong[k] || ci[k] >= n || cf[ci[k]] < 0)
+                continue;
+            p_ci[out] = cf[ci[k]];
+            p_va[out] = T(0);
+            ++out;
+        }
+        return;
+    }
     if (pos_c == 0.0) { diag += pos_all; pos_all = 0.0; }
     double alpha = neg_c != 0.0 ? neg_all / neg_c : 0.0;
     double beta = pos_c != 0.0 ? pos_all / pos_c : 0.0;

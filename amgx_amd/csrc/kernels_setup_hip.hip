@@ -36,9 +36,12 @@ __device__ __forceinline__ unsigned int hash_u32(unsigned int a,
     return a;
 }
 
-// One MIN_MAX round (reference src/matrix_coloring/min_max.cu): an uncolored
-// row whose hash is the strict max (min) among uncolored neighbors takes
-// color 2*iter (2*iter+1). Ties broken by row id => deterministic.
+// One greedy min-max round (reference src/matrix_coloring/
+// greedy_min_max_2ring.cu class): an uncolored row whose hash is the strict
+// max among uncolored neighbors takes the SMALLEST color not used by its
+// already-colored neighbors — greedy-quality color counts (~max degree) with
+// Jones-Plassmann parallel rounds. Hash fixed across rounds, tie-break on
+// row id => deterministic. Colors >= 64 fall back past the bitmask (rare).
 __global__ void color_round_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci, int n,
                                    int* __restrict__ colors, int iter,
@@ -49,23 +52,30 @@ __global__ void color_round_kernel(const int* __restrict__ ro,
     unsigned long long mine =
         ((unsigned long long)hash_u32((unsigned)i, (unsigned)seed) << 32) |
         (unsigned)i;
-    bool is_max = true, is_min = true;
+    bool is_max = true;
+    unsigned long long used = 0ull;
+    int big_used_max = -1;
     for (int k = ro[i]; k < ro[i + 1]; ++k) {
         int j = ci[k];
         if (j == i || j >= n) continue;
-        if (colors[j] >= 0) continue;
+        int cj = colors[j];
+        if (cj >= 0) {
+            if (cj < 64) used |= (1ull << cj);
+            else big_used_max = max(big_used_max, cj);
+            continue;
+        }
         unsigned long long h =
             ((unsigned long long)hash_u32((unsigned)j, (unsigned)seed) << 32) |
             (unsigned)j;
-        if (h > mine) is_max = false;
-        if (h < mine) is_min = false;
+        if (h > mine) { is_max = false; break; }
     }
-    if (is_max)
-        colors[i] = 2 * iter;
-    else if (is_min)
-        colors[i] = 2 * iter + 1;
-    else
+    if (is_max) {
+        int c = (int)(__builtin_ffsll((long long)~used) - 1);
+        if (c < 0 || c >= 64) c = big_used_max + 1;   // beyond-bitmask fallback
+        colors[i] = c;
+    } else {
         atomicAdd(n_uncolored, 1);
+    }
 }
 
 void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
@@ -320,6 +330,8 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
         HIP_CHECK(hipMemcpyAsync(&h_runs, nruns, sizeof(unsigned int),
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
+        if (h_runs == 0)
+            HIP_CHECK(hipMemsetAsync(ro_c, 0, (nc + 1) * sizeof(int), s));
         hipLaunchKernelGGL(keys_to_csr, dim3(grid_1d(h_runs, AMGX_BLOCK, 4096)),
                            dim3(AMGX_BLOCK), 0, s, ukeys, (long long)h_runs,
                            nc_ll, (long long)nc, ro_c, ci_c);
@@ -350,6 +362,12 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
     HIP_CHECK(hipMemcpyAsync(&h_runs, nruns, sizeof(unsigned int),
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
+    if (h_runs == 0) {
+        HIP_CHECK(hipMemsetAsync(ro_c, 0, (nc + 1) * sizeof(int), s));
+        dev_free(run_starts, s); dev_free(nruns, s); dev_free(perm, s);
+        dev_free(tmp, s); dev_free(keys, s); dev_free(rows, s);
+        return 0;
+    }
     // sort run starts (they were written unordered via atomic)
     size_t tmp3 = 0;
     int* rs_alt = (int*)dev_alloc(h_runs * sizeof(int), s);
@@ -467,6 +485,8 @@ long long spgemm_esc(const int* roA, const int* ciA, const T* vaA, int m,
     HIP_CHECK(hipMemcpyAsync(&h_runs, nruns, sizeof(unsigned int),
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
+    if (h_runs == 0)
+        HIP_CHECK(hipMemsetAsync(ro_c, 0, (m + 1) * sizeof(int), s));
     hipLaunchKernelGGL(keys_to_csr, dim3(grid_1d(h_runs, AMGX_BLOCK, 4096)),
                        dim3(AMGX_BLOCK), 0, s, keys, (long long)h_runs,
                        (long long)n, (long long)m, ro_c, ci_c);

@@ -223,11 +223,14 @@ class DistributedManager:
                     ii = (idx[:, None] * b
                           + torch.arange(b, dtype=torch.int64)[None, :])
                     buf.copy_(xf[ii.reshape(-1)])
-            ops.append(dist.P2POp(dist.isend, buf, r))
+            if buf.numel() > 0:
+                ops.append(dist.P2POp(dist.isend, buf, r))
             lo, hi = self.halo_slices[i]
             if hi > lo:
                 view = xf[(self.n_local + lo) * b:(self.n_local + hi) * b]
                 ops.append(dist.P2POp(dist.irecv, view, r))
+        if not ops:
+            return [] if async_start else None
         reqs = dist.batch_isend_irecv(ops)
         if async_start:
             return reqs
