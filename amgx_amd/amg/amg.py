@@ -130,16 +130,62 @@ class AMGHierarchy:
             level.smoother.sweep(b, x, self.presweeps)
         ops.residual(level.A, x, b, level.r)
         level.restrict_residual(level.r, level.bc)
-        # recurse (W recurses twice; F = W then V; reference src/cycles/)
-        repeats = 2 if ctype == "W" and li + 2 < len(self.levels) else 1
-        sub_type = ctype if ctype != "F" else "F"
-        level.xc.zero_()
-        for rep in range(repeats):
-            self._cycle(li + 1, level.bc, level.xc, rep == 0,
-                        "V" if ctype == "F" and rep > 0 else sub_type)
+        if ctype in ("CG", "CGF"):
+            # K-cycle (reference src/cycles/cg_cycle.cu + cg_flex_cycle.cu):
+            # accelerate the coarse correction with cycle_iters FCG steps
+            # preconditioned by the next-level cycle. CGF uses the flexible
+            # (Polak-Ribiere) beta; at this granularity both use the same
+            # 2-step truncated orthogonalization.
+            self._kcycle_coarse(li + 1, level.bc, level.xc)
+        else:
+            # V recurses once; W twice; F = F then V
+            # (reference src/cycles/{v,w,f}_cycle.cu)
+            repeats = 2 if ctype in ("W", "F") and li + 2 < len(self.levels) \
+                else 1
+            level.xc.zero_()
+            for rep in range(repeats):
+                sub = "V" if (ctype == "F" and rep > 0) else ctype
+                self._cycle(li + 1, level.bc, level.xc, rep == 0, sub)
         level.prolongate_and_apply(level.xc, x)
         if self.postsweeps > 0:
             level.smoother.sweep(b, x, self.postsweeps)
+
+    def _kcycle_coarse(self, li: int, b, x):
+        """Notay K-cycle coarse correction (reference src/cycles/cg_cycle.cu):
+        truncated flexible-CG with up to cycle_iters (default 2) inner
+        cycle-preconditioned steps."""
+        A = self.levels[li].A
+        mgr = getattr(A, "manager", None)
+
+        def gdot(u, v):
+            d = ops.dot(u.reshape(-1)[:mgr.owned_size] if mgr else u,
+                        v.reshape(-1)[:mgr.owned_size] if mgr else v)
+            return mgr.global_sum(d) if mgr else d
+
+        x.zero_()
+        c1 = torch.zeros_like(x)
+        self._cycle(li, b, c1, True, "CG")
+        v1 = ops.spmv(A, c1, torch.zeros_like(x))
+        rho1 = gdot(c1, v1)
+        a1 = gdot(c1, b)
+        if rho1 == 0.0:
+            return
+        if self.cycle_iters < 2:
+            ops.axpy(x, c1, a1 / rho1)
+            return
+        rt = b.clone()
+        ops.axpy(rt, v1, -a1 / rho1)
+        c2 = torch.zeros_like(x)
+        self._cycle(li, rt, c2, True, "CG")
+        v2 = ops.spmv(A, c2, torch.zeros_like(x))
+        gamma = gdot(c2, v1)
+        beta = gdot(c2, rt)
+        rho2 = gdot(c2, v2) - gamma * gamma / rho1
+        if rho2 == 0.0:
+            ops.axpy(x, c1, a1 / rho1)
+            return
+        ops.axpy(x, c1, a1 / rho1 - gamma * beta / (rho1 * rho2))
+        ops.axpy(x, c2, beta / rho2)
 
     # ------------------------------------------------------------------ stats
     def grid_stats(self) -> str:

@@ -83,7 +83,20 @@ class AggregationLevel(AMGLevel):
             num = num2
         self.aggregates = agg.to(self.A.row_offsets.device)
         self.num_aggregates = num
+        self._build_r_structure()
         return num
+
+    def _build_r_structure(self):
+        """Aggregate-CSR (offsets, fine ids sorted by aggregate) for
+        deterministic restriction."""
+        agg = self.aggregates.to(torch.int64)
+        order = torch.argsort(agg, stable=True).to(torch.int32)
+        counts = torch.bincount(agg, minlength=self.num_aggregates)
+        off = torch.zeros(self.num_aggregates + 1, dtype=torch.int64,
+                          device=agg.device)
+        torch.cumsum(counts, 0, out=off[1:])
+        self.r_structure = (off.to(torch.int32).contiguous(),
+                            order.contiguous())
 
     def create_coarse_matrix(self) -> CSRMatrix:
         mgr = getattr(self.A, "manager", None)
@@ -126,11 +139,13 @@ class AggregationLevel(AMGLevel):
         # the fine->coarse map with that renumbering
         iperm = Ac.manager.row_iperm.to(self.aggregates.device)
         self.aggregates = iperm[self.aggregates.long()].to(torch.int32)
+        self._build_r_structure()
         return Ac
 
     def restrict_residual(self, r, bc):
         out = ops.restrict_agg(r, self.aggregates, self.num_aggregates,
-                               self.A.block_dim)
+                               self.A.block_dim,
+                               structure=getattr(self, "r_structure", None))
         bc.reshape(-1)[:out.numel()].copy_(out.reshape(-1))
 
     def prolongate_and_apply(self, xc, x):

@@ -309,6 +309,17 @@ void restrict_agg(Tensor r, Tensor agg, int64_t b, Tensor rc) {
     });
 }
 
+void restrict_csr(Tensor r, Tensor off, Tensor fids, int64_t b, Tensor rc) {
+    DISPATCH_FT(r, "restrict_csr", [&] {
+        amgx_hip::restrict_csr<scalar_t>(off.data_ptr<int>(),
+                                         fids.data_ptr<int>(),
+                                         r.data_ptr<scalar_t>(),
+                                         (int)(off.numel() - 1), (int)b,
+                                         rc.data_ptr<scalar_t>(),
+                                         cur_stream());
+    });
+}
+
 void prolongate_agg(Tensor x, Tensor xc, Tensor agg, int64_t b) {
     DISPATCH_FT(x, "prolongate_agg", [&] {
         amgx_hip::prolongate_agg<scalar_t>(x.data_ptr<scalar_t>(),
@@ -559,6 +570,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("color_minmax", &color_minmax);
     m.def("size2_match", &size2_match);
     m.def("restrict_agg", &restrict_agg);
+    m.def("restrict_csr", &restrict_csr);
     m.def("prolongate_agg", &prolongate_agg);
     m.def("dense_gemv", &dense_gemv);
     m.def("gather", &gather);

@@ -95,6 +95,10 @@ void agg_merge_singletons(const int* ro, const int* ci, const T* va,
 template <typename T>
 void restrict_agg(const T* r, const int* agg, int n, int b, T* rc,
                   hipStream_t s);
+// deterministic variant over the aggregate-CSR structure
+template <typename T>
+void restrict_csr(const int* off, const int* fids, const T* r, int nc, int b,
+                  T* rc, hipStream_t s);
 template <typename T>
 void prolongate_agg(T* x, const T* xc, const int* agg, int n, int b,
                     hipStream_t s);

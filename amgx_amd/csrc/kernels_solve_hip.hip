@@ -718,6 +718,35 @@ void restrict_agg(const T* r, const int* agg, int n, int b, T* rc,
                        dim3(AMGX_BLOCK), 0, s, r, agg, (long long)n, b, rc);
 }
 
+// deterministic restriction via the aggregate CSR structure (offsets into
+// fine ids sorted by aggregate) — reference fillRowOffsetsAndColIndices R
+// storage, src/aggregation/aggregation_amg_level.cu:323.
+template <typename T>
+__global__ void restrict_csr_kernel(const int* __restrict__ off,
+                                    const int* __restrict__ fids,
+                                    const T* __restrict__ r, long long nc,
+                                    int b, T* __restrict__ rc) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         t < nc * b; t += stride) {
+        long long I = t / b;
+        int c = (int)(t % b);
+        T sum = T(0);
+        for (int k = off[I]; k < off[I + 1]; ++k)
+            sum += r[(long long)fids[k] * b + c];
+        rc[t] = sum;
+    }
+}
+
+template <typename T>
+void restrict_csr(const int* off, const int* fids, const T* r, int nc, int b,
+                  T* rc, hipStream_t s) {
+    hipLaunchKernelGGL((restrict_csr_kernel<T>),
+                       dim3(grid_1d((long long)nc * b, AMGX_BLOCK, 4096)),
+                       dim3(AMGX_BLOCK), 0, s, off, fids, r, (long long)nc, b,
+                       rc);
+}
+
 template <typename T>
 __global__ void prolongate_kernel(T* __restrict__ x, const T* __restrict__ xc,
                                   const int* __restrict__ agg, long long n,
@@ -850,6 +879,8 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
                                     int, const T*, T*, int, hipStream_t);       \
     template void restrict_agg<T>(const T*, const int*, int, int, T*,           \
                                   hipStream_t);                                 \
+    template void restrict_csr<T>(const int*, const int*, const T*, int, int,   \
+                                  T*, hipStream_t);                                 \
     template void prolongate_agg<T>(T*, const T*, const int*, int, int,         \
                                     hipStream_t);                               \
     template void dense_gemv<T>(const T*, const T*, T*, int, hipStream_t);      \

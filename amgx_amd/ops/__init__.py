@@ -176,10 +176,13 @@ def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
                                             agg_col, ncols_mod)
 
 
-def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):
+def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1,
+                 structure=None):
     """rc[I] = sum_{i in I} r[i] (reference restrictResidualKernel,
-    src/aggregation/aggregation_amg_level.cu:93-180)."""
-    return _backend(r).restrict_agg(r, aggregates, num_aggregates, block_dim)
+    src/aggregation/aggregation_amg_level.cu:93-180). ``structure`` =
+    (offsets, fine_ids) aggregate-CSR for the deterministic no-atomics path."""
+    return _backend(r).restrict_agg(r, aggregates, num_aggregates, block_dim,
+                                    structure)
 
 
 def prolongate_agg(x, xc, aggregates, block_dim: int = 1):

@@ -353,7 +353,8 @@ def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
                             n_cols=num_aggregates, dtype=A.dtype)
 
 
-def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):
+def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1,
+                 structure=None):
     agg = aggregates.to(torch.int64)
     r = r.reshape(-1)[:agg.numel() * block_dim]   # owned prefix (distributed)
     rc = torch.zeros(num_aggregates * block_dim, dtype=r.dtype)

@@ -206,10 +206,15 @@ def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
     return out
 
 
-def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1):
+def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1,
+                 structure=None):
     rc = torch.empty(num_aggregates * block_dim, dtype=r.dtype,
                      device=r.device)
-    _core.restrict_agg(r.reshape(-1), aggregates, block_dim, rc)
+    if structure is not None:
+        off, fids = structure
+        _core.restrict_csr(r.reshape(-1), off, fids, block_dim, rc)
+    else:
+        _core.restrict_agg(r.reshape(-1), aggregates, block_dim, rc)
     return rc
 
 

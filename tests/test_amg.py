@@ -130,3 +130,33 @@ def test_grid_stats():
     assert "Number of Levels" in stats and len(h.levels) >= 3
     # aggregation should roughly halve rows per level
     assert h.levels[1].A.n_rows < 0.7 * A.n_rows
+
+
+def test_f_cycle():
+    A = poisson_2d(20, 20)
+    cfg = {"solver": dict(FGMRES_AGG["solver"])}
+    cfg["solver"]["preconditioner"] = dict(FGMRES_AGG["solver"]["preconditioner"],
+                                           cycle="F")
+    run(cfg, A, max_expected_iters=30)
+
+
+def test_cg_kcycle():
+    A = poisson_2d(20, 20)
+    cfg = {"solver": dict(FGMRES_AGG["solver"])}
+    cfg["solver"]["preconditioner"] = dict(FGMRES_AGG["solver"]["preconditioner"],
+                                           cycle="CG")
+    st = run(cfg, A, max_expected_iters=30)
+
+
+def test_cycle_ordering():
+    """W/F/CG cycles should not be WORSE than V on iteration count."""
+    A = poisson_3d(10, 10, 10)
+    iters = {}
+    for cyc in ("V", "W", "F", "CG"):
+        cfg = {"solver": dict(FGMRES_AGG["solver"])}
+        cfg["solver"]["preconditioner"] = dict(
+            FGMRES_AGG["solver"]["preconditioner"], cycle=cyc)
+        st = run(cfg, A, max_expected_iters=40)
+        iters[cyc] = st.iterations
+    assert iters["W"] <= iters["V"] + 2
+    assert iters["CG"] <= iters["V"] + 2

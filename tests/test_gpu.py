@@ -337,3 +337,25 @@ def test_fgmres_ilu0_gpu():
     A = to_gpu(poisson_3d(12, 12, 12))
     st, rel = _solve_gpu(cfg, A)
     assert st.converged and rel < 1e-6
+
+
+def test_hipgraph_cycle_matches_eager():
+    """The graph-captured V-cycle must give the same FGMRES iteration count
+    as the eager path (the capture self-validates; on mismatch it falls back,
+    so equality of iteration counts is the end-to-end check)."""
+    from tests.test_amg import FGMRES_AGG
+    import copy
+    A = to_gpu(poisson_3d(20, 20, 20))
+    iters = {}
+    for use_graph in (0, 1):
+        cfg_d = copy.deepcopy(FGMRES_AGG)
+        cfg_d["solver"]["preconditioner"]["use_hip_graph"] = use_graph
+        cfg = AMGConfig.from_dict(cfg_d)
+        s = create_solver(cfg.root_scope(), resources=Resources("cuda:0"))
+        b = torch.ones(A.n_rows, dtype=torch.float64, device="cuda")
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        assert st.converged
+        iters[use_graph] = st.iterations
+    assert iters[0] == iters[1], iters
