@@ -1,0 +1,634 @@
+"""AMGX_* API parity layer (reference include/amgx_c.h:150-603,
+src/amgx_c.cu).
+
+The reference exposes ~75 flat C entry points over opaque handles. This
+module reproduces that surface 1:1 in Python (same names, same handle
+lifecycle, same RC error-code discipline: every call returns RC_OK or an
+error code, never raises across the boundary — reference AMGX_TRIES/
+AMGX_CATCHES, src/amgx_c.cu). Outputs are returned after the RC
+(Pythonic substitute for C out-pointers).
+
+Mode strings follow the reference bit-packed modes (include/amgx_config.h:
+79-120): e.g. "dDDI" = device memory space, double vector, double matrix,
+int index; "hDDI" = host. Mixed precision "dDFI" = double vector x float
+matrix.
+"""
+
+from __future__ import annotations
+
+import threading
+from typing import Any, Dict, Optional
+
+import numpy as np
+import torch
+
+from . import ops
+from .config import AMGConfig, write_parameters_description
+from .matrix import CSRMatrix
+from .resources import Resources
+from .solvers import create_solver
+
+# ----------------------------------------------------------------- RC codes
+RC_OK = 0
+RC_BAD_PARAMETERS = 1
+RC_UNKNOWN = 2
+RC_NOT_SUPPORTED_TARGET = 3
+RC_NOT_SUPPORTED_BLOCKSIZE = 4
+RC_CUDA_FAILURE = 5
+RC_IO_ERROR = 6
+RC_BAD_MODE = 7
+RC_CORE = 8
+RC_PLUGIN = 9
+RC_BAD_CONFIGURATION = 10
+RC_NOT_IMPLEMENTED = 11
+RC_LICENSE_NOT_FOUND = 12
+RC_INTERNAL = 13
+
+AMGX_DIST_PARTITION_VECTOR = 0
+AMGX_DIST_PARTITION_OFFSETS = 1
+
+_API_VERSION = (2, 4)
+
+_lock = threading.RLock()
+_initialized = False
+_print_callback = None
+
+
+def _amgx_try(fn):
+    def wrapper(*args, **kwargs):
+        global _print_callback
+        with _lock:
+            try:
+                return fn(*args, **kwargs)
+            except Exception as e:  # noqa: BLE001 - API boundary
+                msg = f"AMGX error in {fn.__name__}: {e}\n"
+                if _print_callback is not None:
+                    try:
+                        _print_callback(msg)
+                    except Exception:
+                        pass
+                rc = RC_BAD_PARAMETERS if isinstance(e, (TypeError, ValueError,
+                                                         KeyError, AssertionError)) \
+                    else RC_UNKNOWN
+                return rc
+    wrapper.__name__ = fn.__name__
+    return wrapper
+
+
+def _parse_mode(mode: str):
+    if not isinstance(mode, str) or len(mode) != 4:
+        raise ValueError(f"bad mode {mode!r}")
+    mem = {"d": "cuda", "h": "cpu"}[mode[0]]
+    vec = {"D": torch.float64, "F": torch.float32}[mode[1]]
+    mat = {"D": torch.float64, "F": torch.float32}[mode[2]]
+    assert mode[3] == "I"
+    return mem, vec, mat
+
+
+# ----------------------------------------------------------------- handles
+class _ConfigHandle:
+    def __init__(self, cfg: AMGConfig):
+        self.cfg = cfg
+
+
+class _ResourcesHandle:
+    def __init__(self, res: Resources):
+        self.res = res
+
+
+class _DistributionHandle:
+    def __init__(self):
+        self.kind = AMGX_DIST_PARTITION_OFFSETS
+        self.data = None
+
+
+class _MatrixHandle:
+    def __init__(self, res: Resources, mode: str):
+        self.res = res
+        self.mode = mode
+        self.A: Optional[CSRMatrix] = None
+
+
+class _VectorHandle:
+    def __init__(self, res: Resources, mode: str):
+        self.res = res
+        self.mode = mode
+        self.v: Optional[torch.Tensor] = None
+        self.bound: Optional[_MatrixHandle] = None
+        self.block_dim = 1
+        self.n = 0
+
+
+class _SolverHandle:
+    def __init__(self, res: Resources, mode: str, cfg: AMGConfig):
+        self.res = res
+        self.mode = mode
+        self.cfg = cfg
+        self.solver = None
+        self.status = None
+
+
+# ----------------------------------------------------------------- lifecycle
+@_amgx_try
+def AMGX_initialize():
+    global _initialized
+    _initialized = True
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_initialize_plugins():
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_finalize():
+    global _initialized
+    _initialized = False
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_finalize_plugins():
+    return RC_OK
+
+
+def AMGX_abort(res, err):  # pragma: no cover - mirrors reference abort
+    raise SystemExit(err)
+
+
+@_amgx_try
+def AMGX_get_api_version():
+    return RC_OK, _API_VERSION[0], _API_VERSION[1]
+
+
+@_amgx_try
+def AMGX_get_build_info_strings():
+    from . import __version__
+    return RC_OK, f"amgx_amd {__version__}", "MI355X gfx950", "rocm"
+
+
+@_amgx_try
+def AMGX_install_signal_handler():
+    import faulthandler
+    faulthandler.enable()
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_reset_signal_handler():
+    import faulthandler
+    faulthandler.disable()
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_register_print_callback(cb):
+    global _print_callback
+    _print_callback = cb
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_write_parameters_description(path: str):
+    with open(path, "w") as f:
+        f.write(write_parameters_description())
+    return RC_OK
+
+
+# ----------------------------------------------------------------- config
+@_amgx_try
+def AMGX_config_create(options: str):
+    return RC_OK, _ConfigHandle(AMGConfig.parse(options))
+
+
+@_amgx_try
+def AMGX_config_create_from_file(path: str):
+    return RC_OK, _ConfigHandle(AMGConfig.from_file(path))
+
+
+@_amgx_try
+def AMGX_config_add_parameters(cfg: _ConfigHandle, options: str):
+    extra = AMGConfig.parse(options)
+    node = cfg.cfg.tree
+    node.update(extra.tree)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_config_get_default_number_of_rings(cfg: _ConfigHandle):
+    # reference: 1 ring unless aggregation-style configs request 2
+    return RC_OK, 1
+
+
+@_amgx_try
+def AMGX_config_destroy(cfg: _ConfigHandle):
+    cfg.cfg = None
+    return RC_OK
+
+
+# ----------------------------------------------------------------- resources
+@_amgx_try
+def AMGX_resources_create_simple(cfg: _ConfigHandle):
+    return RC_OK, _ResourcesHandle(Resources())
+
+
+@_amgx_try
+def AMGX_resources_create(cfg: _ConfigHandle, comm=None, device_num: int = 0,
+                          devices=None):
+    dev = f"cuda:{device_num}" if torch.cuda.is_available() else "cpu"
+    distributed = comm is not None
+    return RC_OK, _ResourcesHandle(Resources(dev, distributed=distributed))
+
+
+@_amgx_try
+def AMGX_resources_destroy(res: _ResourcesHandle):
+    res.res = None
+    return RC_OK
+
+
+# ----------------------------------------------------------------- distribution
+@_amgx_try
+def AMGX_distribution_create(cfg: _ConfigHandle = None):
+    return RC_OK, _DistributionHandle()
+
+
+@_amgx_try
+def AMGX_distribution_set_partition_data(dist: _DistributionHandle, kind,
+                                         data):
+    dist.kind = kind
+    dist.data = np.asarray(data, dtype=np.int64)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_distribution_destroy(dist: _DistributionHandle):
+    dist.data = None
+    return RC_OK
+
+
+# ----------------------------------------------------------------- matrix
+@_amgx_try
+def AMGX_matrix_create(res: _ResourcesHandle, mode: str):
+    _parse_mode(mode)
+    return RC_OK, _MatrixHandle(res.res, mode)
+
+
+@_amgx_try
+def AMGX_matrix_destroy(m: _MatrixHandle):
+    m.A = None
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_upload_all(m: _MatrixHandle, n, nnz, block_dimx, block_dimy,
+                           row_ptrs, col_indices, data, diag_data=None):
+    mem, _, matprec = _parse_mode(m.mode)
+    assert block_dimx == block_dimy, "rectangular blocks unsupported"
+    device = mem if mem == "cpu" else m.res.device
+    ro = torch.as_tensor(np.asarray(row_ptrs), dtype=torch.int32)
+    ci = torch.as_tensor(np.asarray(col_indices), dtype=torch.int32)
+    b = int(block_dimx)
+    va = torch.as_tensor(np.asarray(data)).to(matprec)
+    if b > 1:
+        va = va.reshape(nnz, b, b)
+    diag = None
+    if diag_data is not None:
+        diag = torch.as_tensor(np.asarray(diag_data)).to(matprec)
+        diag = diag.reshape(n, b, b) if b > 1 else diag.reshape(n)
+        diag = diag.to(device)
+    m.A = CSRMatrix(ro.to(device), ci.to(device), va.to(device),
+                    n_cols=n, block_dim=b, diag=diag)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_upload_all_global(m: _MatrixHandle, n_global, n, nnz,
+                                  block_dimx, block_dimy, row_ptrs,
+                                  col_indices_global, data, diag_data=None,
+                                  allocated_halo_depth=1, num_import_rings=1,
+                                  partition_vector=None):
+    """Distributed upload with GLOBAL column indices (reference
+    src/amgx_c.cu:4615 / matrix_upload_distributed:1739)."""
+    import torch.distributed as tdist
+    from .distributed.manager import DistributedManager
+    mem, _, matprec = _parse_mode(m.mode)
+    assert block_dimx == block_dimy
+    assert tdist.is_initialized(), "upload_all_global needs torch.distributed"
+    rank = tdist.get_rank()
+    if partition_vector is not None:
+        pv = np.asarray(partition_vector)
+        row_start = int(np.sum([(pv[:] == r).sum() for r in range(rank)]))
+        assert (np.sort(np.nonzero(pv == rank)[0])
+                == np.arange(row_start, row_start + n)).all(), \
+            "non-contiguous partition vectors unsupported"
+    else:
+        counts = [None] * tdist.get_world_size()
+        tdist.all_gather_object(counts, int(n))
+        row_start = int(sum(counts[:rank]))
+    device = mem if mem == "cpu" else m.res.device
+    vals = np.asarray(data)
+    if block_dimx > 1:
+        vals = vals.reshape(nnz, block_dimx * block_dimx)
+    m.A = DistributedManager.upload_global_csr(
+        np.asarray(row_ptrs), np.asarray(col_indices_global), vals,
+        n, row_start, n_global, device=device, block_dim=int(block_dimx),
+        dtype=matprec)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_upload_distributed(m, n_global, n, nnz, block_dimx,
+                                   block_dimy, row_ptrs, col_indices_global,
+                                   data, diag_data, dist: _DistributionHandle):
+    pv = None
+    if dist is not None and dist.kind == AMGX_DIST_PARTITION_VECTOR:
+        pv = dist.data
+    return AMGX_matrix_upload_all_global(
+        m, n_global, n, nnz, block_dimx, block_dimy, row_ptrs,
+        col_indices_global, data, diag_data, partition_vector=pv)
+
+
+@_amgx_try
+def AMGX_matrix_replace_coefficients(m: _MatrixHandle, n, nnz, data,
+                                     diag_data=None):
+    _, _, matprec = _parse_mode(m.mode)
+    va = torch.as_tensor(np.asarray(data)).to(matprec)
+    if m.A.block_dim > 1:
+        va = va.reshape(-1, m.A.block_dim, m.A.block_dim)
+    diag = None
+    if diag_data is not None:
+        diag = torch.as_tensor(np.asarray(diag_data)).to(matprec)
+    m.A.replace_coefficients(va, diag)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_get_size(m: _MatrixHandle):
+    return RC_OK, m.A.n_rows, m.A.block_dim, m.A.block_dim
+
+
+@_amgx_try
+def AMGX_matrix_download_all(m: _MatrixHandle):
+    A = m.A
+    return (RC_OK, A.row_offsets.cpu().numpy(), A.col_indices.cpu().numpy(),
+            A.values.cpu().numpy(),
+            A.diag.cpu().numpy() if A.diag is not None else None)
+
+
+@_amgx_try
+def AMGX_matrix_check_symmetry(m: _MatrixHandle):
+    s = m.A.to_scipy()
+    diff = abs(s - s.T)
+    structurally = (s != 0).multiply((s.T != 0) != (s != 0)).nnz == 0
+    sym = diff.nnz == 0 or diff.max() < 1e-12
+    return RC_OK, bool(structurally or sym), bool(sym)
+
+
+@_amgx_try
+def AMGX_matrix_check_diag_dominant(m: _MatrixHandle):
+    s = m.A.to_scipy().tocsr()
+    d = np.abs(s.diagonal())
+    off = np.asarray(abs(s).sum(axis=1)).ravel() - d
+    return RC_OK, bool((d >= off).all())
+
+
+# ----------------------------------------------------------------- vector
+@_amgx_try
+def AMGX_vector_create(res: _ResourcesHandle, mode: str):
+    _parse_mode(mode)
+    return RC_OK, _VectorHandle(res.res, mode)
+
+
+@_amgx_try
+def AMGX_vector_destroy(v: _VectorHandle):
+    v.v = None
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_vector_bind(v: _VectorHandle, m: _MatrixHandle):
+    """Attach to a (possibly distributed) matrix (reference
+    AMGX_vector_bind: vector inherits the matrix's comm pattern)."""
+    v.bound = m
+    return RC_OK
+
+
+def _vec_device(v: _VectorHandle):
+    mem, vecprec, _ = _parse_mode(v.mode)
+    dev = mem if mem == "cpu" else v.res.device
+    return dev, vecprec
+
+
+@_amgx_try
+def AMGX_vector_upload(v: _VectorHandle, n, block_dim, data):
+    dev, prec = _vec_device(v)
+    t = torch.as_tensor(np.asarray(data)).to(prec).reshape(-1)
+    v.n, v.block_dim = int(n), int(block_dim)
+    A = v.bound.A if v.bound is not None else None
+    mgr = getattr(A, "manager", None) if A is not None else None
+    if mgr is not None:
+        full = mgr.new_ext_vec(prec)
+        full[:mgr.owned_size] = t.to(full.device)[
+            (mgr.row_perm[:, None] * block_dim
+             + torch.arange(block_dim, device=full.device)[None, :]).reshape(-1)
+            if block_dim > 1 else mgr.row_perm]
+        v.v = full
+    else:
+        v.v = t.to(dev)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_vector_set_zero(v: _VectorHandle, n=None, block_dim=None):
+    if v.v is None:
+        dev, prec = _vec_device(v)
+        A = v.bound.A if v.bound is not None else None
+        mgr = getattr(A, "manager", None) if A is not None else None
+        if mgr is not None:
+            v.v = mgr.new_ext_vec(prec)
+            v.n = mgr.n_local
+        else:
+            assert n is not None
+            v.n = int(n)
+            v.block_dim = int(block_dim or 1)
+            v.v = torch.zeros(v.n * v.block_dim, dtype=prec, device=dev)
+    else:
+        v.v.zero_()
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_vector_set_random(v: _VectorHandle, n):
+    dev, prec = _vec_device(v)
+    v.n = int(n)
+    v.v = torch.rand(v.n * v.block_dim, dtype=prec, device=dev)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_vector_download(v: _VectorHandle):
+    A = v.bound.A if v.bound is not None else None
+    mgr = getattr(A, "manager", None) if A is not None else None
+    if mgr is not None:
+        return RC_OK, mgr.permute_out(v.v).cpu().numpy()
+    return RC_OK, v.v.detach().cpu().numpy()
+
+
+@_amgx_try
+def AMGX_vector_get_size(v: _VectorHandle):
+    return RC_OK, v.n, v.block_dim
+
+
+# ----------------------------------------------------------------- solver
+@_amgx_try
+def AMGX_solver_create(res: _ResourcesHandle, mode: str, cfg: _ConfigHandle):
+    _parse_mode(mode)
+    return RC_OK, _SolverHandle(res.res, mode, cfg.cfg)
+
+
+@_amgx_try
+def AMGX_solver_destroy(s: _SolverHandle):
+    s.solver = None
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_solver_setup(s: _SolverHandle, m: _MatrixHandle):
+    mem, _, _ = _parse_mode(s.mode)
+    res = s.res if mem != "cpu" else Resources("cpu")
+    s.solver = create_solver(s.cfg.root_scope(), resources=res)
+    s.solver.setup(m.A)
+    s.matrix = m
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_solver_resetup(s: _SolverHandle, m: _MatrixHandle):
+    s.solver.resetup(m.A)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_solver_solve(s: _SolverHandle, rhs: _VectorHandle,
+                      sol: _VectorHandle, zero_initial_guess=False):
+    if sol.v is None or sol.v.numel() != rhs.v.numel():
+        sol.v = torch.zeros_like(rhs.v)
+        sol.n = rhs.n
+        sol.block_dim = rhs.block_dim
+    s.status = s.solver.solve(rhs.v, sol.v,
+                              zero_initial_guess=zero_initial_guess)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_solver_solve_with_0_initial_guess(s, rhs, sol):
+    return AMGX_solver_solve(s, rhs, sol, zero_initial_guess=True)
+
+
+@_amgx_try
+def AMGX_solver_get_status(s: _SolverHandle):
+    # reference AMGX_SOLVE_SUCCESS=0, FAILED=1, DIVERGED=2
+    st = s.status
+    code = 0 if st.converged else (2 if st.status == st.DIVERGED else 1)
+    return RC_OK, code
+
+
+@_amgx_try
+def AMGX_solver_get_iterations_number(s: _SolverHandle):
+    return RC_OK, s.status.iterations
+
+
+@_amgx_try
+def AMGX_solver_get_iteration_residual(s: _SolverHandle, it: int = -1,
+                                       idx: int = 0):
+    res = s.status.residuals
+    return RC_OK, float(res[it] if res else float("nan"))
+
+
+# ----------------------------------------------------------------- IO
+@_amgx_try
+def AMGX_read_system(m: _MatrixHandle, rhs: _VectorHandle,
+                     sol: _VectorHandle, path: str):
+    from .io.matrix_market import read_system
+    mem, vecprec, matprec = _parse_mode(m.mode)
+    dev = mem if mem == "cpu" else m.res.device
+    A, b, x0 = read_system(path, device=dev, dtype=matprec)
+    m.A = A
+    n = A.n_rows
+    if rhs is not None:
+        rhs.v = b.to(vecprec) if b is not None else \
+            torch.ones(n, dtype=vecprec, device=dev)
+        rhs.n = n
+    if sol is not None:
+        sol.v = x0.to(vecprec) if x0 is not None else \
+            torch.zeros(n, dtype=vecprec, device=dev)
+        sol.n = n
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_write_system(m: _MatrixHandle, rhs: _VectorHandle,
+                      sol: _VectorHandle, path: str):
+    from .io.matrix_market import write_system
+    write_system(path, m.A,
+                 rhs.v if rhs is not None else None,
+                 sol.v if sol is not None else None)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_generate_distributed_poisson_7pt(m: _MatrixHandle,
+                                          rhs: _VectorHandle,
+                                          sol: _VectorHandle,
+                                          allocated_halo_depth, num_import_rings,
+                                          nx, ny, nz, px=1, py=1, pz=None):
+    """Built-in 3D Poisson generator (reference src/amgx_c.cu:4566-4731).
+    Current partitioner: z-slabs (px=py=1)."""
+    import torch.distributed as tdist
+    from .problems import poisson_3d, poisson_3d_local
+    from .distributed.manager import DistributedManager
+    mem, vecprec, matprec = _parse_mode(m.mode)
+    dev = mem if mem == "cpu" else m.res.device
+    world = tdist.get_world_size() if tdist.is_initialized() else 1
+    if world == 1:
+        m.A = poisson_3d(nx, ny, nz, device=dev, dtype=matprec)
+        n = m.A.n_rows
+        if rhs is not None:
+            rhs.v = torch.ones(n, dtype=vecprec, device=dev)
+            rhs.n = n
+        if sol is not None:
+            sol.v = torch.zeros(n, dtype=vecprec, device=dev)
+            sol.n = n
+        return RC_OK
+    assert px == 1 and py == 1, "x/y process grids: z-slab partitioner only"
+    rank = tdist.get_rank()
+    ro, cols, vals, rs = poisson_3d_local(nx, ny, nz, rank, world)
+    m.A = DistributedManager.upload_global_csr(
+        ro, cols, vals, nx * ny * nz, rs, nx * ny * nz * world,
+        device=dev, dtype=matprec)
+    mgr = m.A.manager
+    if rhs is not None:
+        rhs.v = mgr.new_ext_vec(vecprec)
+        rhs.v[:mgr.owned_size] = 1.0
+        rhs.n = mgr.n_local
+        rhs.bound = m
+    if sol is not None:
+        sol.v = mgr.new_ext_vec(vecprec)
+        sol.n = mgr.n_local
+        sol.bound = m
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_pin_memory(ptr, bytes_=0):
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_unpin_memory(ptr):
+    return RC_OK
+
+
+def AMGX_SOLVE_SUCCESS():
+    return 0

@@ -67,3 +67,29 @@ def read_system(path: str, device="cpu", dtype=torch.float64
 
 def write_matrix_market(path: str, A: CSRMatrix):
     scipy.io.mmwrite(path, A.to_scipy())
+
+
+def write_system(path: str, A: CSRMatrix, b=None, x=None):
+    """Write matrix (+ optional rhs/solution blocks with the AmgX extension
+    header) — reference AMGX_write_system / src/matrix_io.cu."""
+    m = A.to_scipy().tocoo()
+    flags = []
+    if b is not None:
+        flags.append("rhs")
+    if x is not None:
+        flags.append("solution")
+    with open(path, "w") as f:
+        f.write("%%MatrixMarket matrix coordinate real general\n")
+        if flags:
+            f.write("%%NVAMG " + " ".join(flags) + "\n")
+        f.write(f"{m.shape[0]} {m.shape[1]} {m.nnz}\n")
+        for r, c, v in zip(m.row, m.col, m.data):
+            f.write(f"{r + 1} {c + 1} {v:.17g}\n")
+        import torch as _torch
+        for vec in (b, x):
+            if vec is None:
+                continue
+            arr = vec.detach().cpu().numpy().reshape(-1) \
+                if isinstance(vec, _torch.Tensor) else np.asarray(vec).reshape(-1)
+            for val in arr[:m.shape[0]]:
+                f.write(f"{val:.17g}\n")
