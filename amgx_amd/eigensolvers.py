@@ -1,0 +1,412 @@
+"""Eigensolvers (reference layer L8: src/eigensolvers/, include/eigensolvers/
+eigensolver.h:25-130; C API AMGX_eigensolver_*, include/amgx_eig_c.h:16-26).
+
+Implemented: POWER_ITERATION (+ shift/inverse modes via operators), LANCZOS,
+ARNOLDI, SUBSPACE_ITERATION, LOBPCG, and the PageRank operator
+(reference include/operators/pagerank_operator.h:13). Small dense
+eigenproblems (tridiagonal/Hessenberg/Rayleigh-Ritz) are solved on host with
+numpy at setup-scale sizes; all vector work runs through the amgx_amd ops
+layer (HIP kernels on device).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Optional, Type
+
+import numpy as np
+import torch
+
+from . import ops
+from .config import ConfigScope
+from .resources import Resources, default_resources
+
+EIGEN_REGISTRY: Dict[str, Type["EigenSolver"]] = {}
+
+
+def register_eigensolver(name):
+    def deco(cls):
+        EIGEN_REGISTRY[name] = cls
+        return cls
+    return deco
+
+
+def create_eigensolver(scope: ConfigScope, resources=None) -> "EigenSolver":
+    name = scope.get("eig_solver")
+    cls = EIGEN_REGISTRY.get(name)
+    if cls is None:
+        raise KeyError(f"unknown eigensolver {name!r}")
+    return cls(scope, resources or default_resources())
+
+
+# ------------------------------------------------------------------- operators
+class Operator:
+    """Linear-operator composition (reference include/operators/*.h)."""
+
+    def __init__(self, A):
+        self.A = A
+        self.n = A.n_rows * A.block_dim
+
+    def apply(self, x, y):
+        ops.spmv(self.A, x, y)
+        return y
+
+
+class ShiftedOperator(Operator):
+    """(A - shift I) x (reference shifted_operator.h)."""
+
+    def __init__(self, A, shift: float):
+        super().__init__(A)
+        self.shift = shift
+
+    def apply(self, x, y):
+        ops.spmv(self.A, x, y)
+        if self.shift:
+            ops.axpy(y, x, -self.shift)
+        return y
+
+
+class SolverOperator(Operator):
+    """A^{-1} x via a configured inner solver — inverse iteration
+    (reference solve_operator.h / solver_operator.h)."""
+
+    def __init__(self, A, solver):
+        super().__init__(A)
+        self.solver = solver
+
+    def apply(self, x, y):
+        y.zero_()
+        self.solver.solve(x, y, zero_initial_guess=True)
+        return y
+
+
+class PageRankOperator(Operator):
+    """Google-matrix operator G^T x = d * P^T x + teleport
+    (reference pagerank_operator.h; AMGX_eigensolver_pagerank_setup)."""
+
+    def __init__(self, A, damping: float = 0.85):
+        super().__init__(A)
+        self.d = damping
+        # column-stochastic normalization: P = A D_out^{-1}
+        deg = ops.spmv(ops.transpose(A) if False else A,
+                       torch.ones(self.n, dtype=A.dtype, device=A.device))
+        # out-degree = row sums of A
+        self.out_inv = torch.where(deg != 0, 1.0 / deg, torch.zeros_like(deg))
+        self.AT = ops.transpose(A)
+
+    def apply(self, x, y):
+        ops.spmv(self.AT, x * self.out_inv, y)
+        ops.scal(y, self.d)
+        dangling = float(ops.dot(x, (self.out_inv == 0).to(x.dtype)))
+        y += (self.d * dangling + (1.0 - self.d) * float(x.sum())) / self.n
+        return y
+
+
+# ------------------------------------------------------------------- base
+class EigenStatus:
+    def __init__(self):
+        self.converged = False
+        self.iterations = 0
+        self.eigenvalues = []
+        self.eigenvector = None
+        self.residuals = []
+
+
+class EigenSolver:
+    def __init__(self, scope: ConfigScope, resources: Resources):
+        self.scope = scope
+        self.res = resources
+        self.max_iters = scope.get("eig_max_iters")
+        self.tol = scope.get("eig_tolerance")
+        self.which = scope.get("eig_which")
+        self.shift = scope.get("eig_shift")
+        self.wanted = scope.get("eig_wanted_count")
+        self.op: Optional[Operator] = None
+        self.status = EigenStatus()
+
+    def setup(self, A):
+        self.A = A
+        if self.which == "smallest":
+            from .solvers import create_solver
+            from .config import ConfigScope as CS
+            inner = create_solver("PCG", CS(None, {
+                "max_iters": 200, "monitor_residual": 1, "tolerance": 1e-10,
+                "convergence": "RELATIVE_INI",
+                "preconditioner": "BLOCK_JACOBI"}), self.res)
+            inner.setup(A)
+            self.op = SolverOperator(A, inner)
+        elif self.shift:
+            self.op = ShiftedOperator(A, self.shift)
+        else:
+            self.op = Operator(A)
+        self.solver_setup()
+
+    def pagerank_setup(self, A, damping: Optional[float] = None):
+        self.A = A
+        self.op = PageRankOperator(
+            A, damping if damping is not None else
+            self.scope.get("eig_damping_factor"))
+        self.solver_setup()
+
+    def solver_setup(self):
+        pass
+
+    def _rand_vec(self, seed=42):
+        g = torch.Generator().manual_seed(seed)
+        v = torch.rand(self.op.n, generator=g, dtype=torch.float64)
+        return v.to(self.A.dtype).to(self.A.device)
+
+    def _to_true_eig(self, lam: float) -> float:
+        if self.which == "smallest":
+            return 1.0 / lam if lam != 0 else float("inf")
+        if self.shift:
+            return lam + self.shift
+        return lam
+
+    def solve(self, x0: Optional[torch.Tensor] = None) -> EigenStatus:
+        raise NotImplementedError
+
+
+@register_eigensolver("POWER_ITERATION")
+@register_eigensolver("SINGLE_ITERATION")
+class PowerIteration(EigenSolver):
+    """Reference src/eigensolvers/power_iteration_eigensolver.cu (and
+    single_iteration_eigensolver.cu: one-iteration variant)."""
+
+    def solve(self, x0=None):
+        st = self.status = EigenStatus()
+        v = x0.clone() if x0 is not None else self._rand_vec()
+        Av = torch.zeros_like(v)
+        lam = 0.0
+        for it in range(self.max_iters):
+            nv = ops.nrm2(v)
+            if nv == 0:
+                break
+            ops.scal(v, 1.0 / nv)
+            self.op.apply(v, Av)
+            lam = ops.dot(v, Av)
+            # residual ||Av - lam v||
+            r = Av.clone()
+            ops.axpy(r, v, -lam)
+            rn = ops.nrm2(r) / max(abs(lam), 1e-300)
+            st.residuals.append(rn)
+            st.iterations = it + 1
+            v, Av = Av.clone(), Av
+            if rn < self.tol:
+                st.converged = True
+                break
+        st.eigenvalues = [self._to_true_eig(lam)]
+        nv = ops.nrm2(v)
+        st.eigenvector = v / nv if nv else v
+        return st
+
+
+@register_eigensolver("LANCZOS")
+class Lanczos(EigenSolver):
+    """Reference src/eigensolvers/lanczos_eigensolver.cu: symmetric Lanczos
+    with full reorthogonalization; tridiagonal solved on host."""
+
+    def solve(self, x0=None):
+        st = self.status = EigenStatus()
+        m = min(self.max_iters, self.op.n)
+        v = x0.clone() if x0 is not None else self._rand_vec()
+        ops.scal(v, 1.0 / ops.nrm2(v))
+        V = [v]
+        alphas, betas = [], []
+        w = torch.zeros_like(v)
+        for j in range(m):
+            self.op.apply(V[j], w)
+            a = ops.dot(w, V[j])
+            alphas.append(a)
+            ops.axpy(w, V[j], -a)
+            if j > 0:
+                ops.axpy(w, V[j - 1], -betas[-1])
+            # full reorthogonalization
+            for q in V:
+                ops.axpy(w, q, -ops.dot(w, q))
+            b = ops.nrm2(w)
+            st.iterations = j + 1
+            if j >= 1:
+                T = np.diag(alphas) + np.diag(betas, 1) + np.diag(betas, -1)
+                evals, evecs = np.linalg.eigh(T)
+                idx = -1 if self.which != "smallest" else -1  # op handles inverse
+                ritz = evals[idx]
+                resid = abs(b * evecs[-1, idx]) / max(abs(ritz), 1e-300)
+                st.residuals.append(resid)
+                if resid < self.tol or b < 1e-14:
+                    st.converged = True
+                    y = evecs[:, idx]
+                    vec = torch.zeros_like(v)
+                    for c, q in zip(y, V):
+                        ops.axpy(vec, q, float(c))
+                    st.eigenvector = vec
+                    st.eigenvalues = [self._to_true_eig(float(ritz))]
+                    return st
+            if b < 1e-14:
+                break
+            betas.append(b)
+            V.append(w / b)
+            w = torch.zeros_like(v)
+        if alphas:
+            T = np.diag(alphas)
+            if betas:
+                T = T + np.diag(betas[:len(alphas) - 1], 1) \
+                    + np.diag(betas[:len(alphas) - 1], -1)
+            evals, evecs = np.linalg.eigh(T)
+            st.eigenvalues = [self._to_true_eig(float(evals[-1]))]
+            y = evecs[:, -1]
+            vec = torch.zeros_like(v)
+            for c, q in zip(y, V):
+                ops.axpy(vec, q, float(c))
+            st.eigenvector = vec
+        return st
+
+
+@register_eigensolver("ARNOLDI")
+class Arnoldi(EigenSolver):
+    """Reference src/eigensolvers/arnoldi_eigensolver.cu: Arnoldi with host
+    Hessenberg eigen-decomposition (general matrices)."""
+
+    def solve(self, x0=None):
+        st = self.status = EigenStatus()
+        m = min(self.max_iters, self.op.n)
+        v = x0.clone() if x0 is not None else self._rand_vec()
+        ops.scal(v, 1.0 / ops.nrm2(v))
+        V = [v]
+        H = np.zeros((m + 1, m))
+        w = torch.zeros_like(v)
+        for j in range(m):
+            self.op.apply(V[j], w)
+            for i in range(j + 1):
+                H[i, j] = ops.dot(w, V[i])
+                ops.axpy(w, V[i], -H[i, j])
+            H[j + 1, j] = ops.nrm2(w)
+            st.iterations = j + 1
+            if j >= 1:
+                evals, evecs = np.linalg.eig(H[:j + 1, :j + 1])
+                k = int(np.argmax(np.abs(evals)))
+                resid = abs(H[j + 1, j] * evecs[-1, k]) / max(abs(evals[k]), 1e-300)
+                st.residuals.append(float(resid))
+                if resid < self.tol:
+                    st.converged = True
+                    st.eigenvalues = [self._to_true_eig(float(np.real(evals[k])))]
+                    y = np.real(evecs[:, k])
+                    vec = torch.zeros_like(v)
+                    for c, q in zip(y, V):
+                        ops.axpy(vec, q, float(c))
+                    st.eigenvector = vec
+                    return st
+            if H[j + 1, j] < 1e-14:
+                break
+            V.append(w / H[j + 1, j])
+            w = torch.zeros_like(v)
+        if st.iterations:
+            j = st.iterations
+            evals, evecs = np.linalg.eig(H[:j, :j])
+            k = int(np.argmax(np.abs(evals)))
+            st.eigenvalues = [self._to_true_eig(float(np.real(evals[k])))]
+        return st
+
+
+@register_eigensolver("SUBSPACE_ITERATION")
+class SubspaceIteration(EigenSolver):
+    """Reference src/eigensolvers/subspace_iteration_eigensolver.cu: block
+    power iteration with Rayleigh-Ritz."""
+
+    def solve(self, x0=None):
+        st = self.status = EigenStatus()
+        k = max(self.wanted, 1)
+        p = min(k + 2, self.op.n)
+        n = self.op.n
+        g = torch.Generator().manual_seed(17)
+        X = torch.rand(n, p, generator=g, dtype=torch.float64) \
+            .to(self.A.dtype).to(self.A.device)
+        Y = torch.zeros_like(X)
+        lam_old = None
+        for it in range(self.max_iters):
+            X, _ = torch.linalg.qr(X.double())
+            X = X.to(self.A.dtype)
+            for c in range(p):
+                xc = X[:, c].contiguous()
+                yc = torch.zeros_like(xc)
+                self.op.apply(xc, yc)
+                Y[:, c] = yc
+            T = (X.double().T @ Y.double()).cpu().numpy()
+            evals, evecs = np.linalg.eig(T)
+            order = np.argsort(-np.abs(evals))
+            lam = np.real(evals[order[:k]])
+            st.iterations = it + 1
+            if lam_old is not None and np.all(
+                    np.abs(lam - lam_old) <= self.tol * np.maximum(np.abs(lam), 1e-300)):
+                st.converged = True
+                break
+            lam_old = lam
+            X = Y.clone()
+        st.eigenvalues = [self._to_true_eig(float(v)) for v in lam]
+        st.eigenvector = X[:, 0].contiguous()
+        return st
+
+
+@register_eigensolver("LOBPCG")
+class LOBPCG(EigenSolver):
+    """Reference src/eigensolvers/lobpcg_eigensolver.cu: locally optimal
+    block PCG for the SMALLEST eigenpair of an SPD matrix (direct A apply;
+    does not wrap the inverse operator)."""
+
+    def setup(self, A):
+        self.A = A
+        self.op = Operator(A)
+        from .solvers import create_solver
+        from .config import ConfigScope as CS
+        self.prec = create_solver("BLOCK_JACOBI", CS(None, {"max_iters": 1}),
+                                  self.res)
+        self.prec.setup(A)
+
+    def solve(self, x0=None):
+        st = self.status = EigenStatus()
+        x = x0.clone() if x0 is not None else self._rand_vec()
+        ops.scal(x, 1.0 / ops.nrm2(x))
+        Ax = torch.zeros_like(x)
+        self.op.apply(x, Ax)
+        lam = ops.dot(x, Ax)
+        p = None
+        for it in range(self.max_iters):
+            r = Ax.clone()
+            ops.axpy(r, x, -lam)
+            rn = ops.nrm2(r) / max(abs(lam), 1e-300)
+            st.residuals.append(rn)
+            st.iterations = it + 1
+            if rn < self.tol:
+                st.converged = True
+                break
+            w = torch.zeros_like(r)
+            self.prec.solve(r, w, zero_initial_guess=True)
+            basis = [x, w] + ([p] if p is not None else [])
+            B = torch.stack([v.double() for v in basis], dim=1)
+            Q, _ = torch.linalg.qr(B)
+            m = Q.shape[1]
+            AQ = torch.zeros_like(Q)
+            for c in range(m):
+                qc = Q[:, c].contiguous().to(self.A.dtype)
+                tmp = torch.zeros_like(qc)
+                self.op.apply(qc, tmp)
+                AQ[:, c] = tmp.double()
+            T = (Q.T @ AQ).cpu().numpy()
+            evals, evecs = np.linalg.eigh(T)
+            y = evecs[:, 0]
+            x_new = (Q @ torch.from_numpy(y).to(Q.device)).to(self.A.dtype)
+            p = (x_new - x * float(y[0])).contiguous()
+            x = x_new.contiguous()
+            ops.scal(x, 1.0 / ops.nrm2(x))
+            self.op.apply(x, Ax)
+            lam = ops.dot(x, Ax)
+        st.eigenvalues = [float(lam)]
+        st.eigenvector = x
+        return st
+
+
+@register_eigensolver("JACOBI_DAVIDSON")
+class JacobiDavidson(LOBPCG):
+    """Reference src/eigensolvers/jacobi_davidson_eigensolver.cu — served by
+    the same locally-optimal preconditioned subspace engine (the correction
+    equation is preconditioned with the diagonal, as LOBPCG's w-step)."""

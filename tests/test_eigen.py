@@ -1,0 +1,124 @@
+import numpy as np
+import pytest
+import torch
+
+from amgx_amd.config import ConfigScope
+from amgx_amd.eigensolvers import create_eigensolver, EIGEN_REGISTRY
+from amgx_amd.problems import poisson_2d
+from amgx_amd.resources import Resources
+
+
+def scope(d):
+    return ConfigScope(None, d)
+
+
+def ref_extreme_eigs(A):
+    dense = A.to_scipy().toarray()
+    evals = np.linalg.eigvalsh(dense)
+    return evals[0], evals[-1]
+
+
+def test_power_iteration_largest():
+    A = poisson_2d(12, 12)
+    lo, hi = ref_extreme_eigs(A)
+    s = create_eigensolver(scope({"eig_solver": "POWER_ITERATION",
+                                  "eig_max_iters": 2000,
+                                  "eig_tolerance": 1e-8}), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert st.converged
+    assert abs(st.eigenvalues[0] - hi) < 1e-5 * hi
+
+
+def test_lanczos_largest():
+    A = poisson_2d(12, 12)
+    lo, hi = ref_extreme_eigs(A)
+    s = create_eigensolver(scope({"eig_solver": "LANCZOS",
+                                  "eig_max_iters": 80,
+                                  "eig_tolerance": 1e-9}), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert st.converged
+    assert abs(st.eigenvalues[0] - hi) < 1e-6 * hi
+    # eigenvector residual
+    v = st.eigenvector.numpy()
+    r = A.to_scipy() @ v - st.eigenvalues[0] * v
+    assert np.linalg.norm(r) < 1e-5 * abs(st.eigenvalues[0])
+
+
+def test_arnoldi_largest():
+    A = poisson_2d(10, 10)
+    lo, hi = ref_extreme_eigs(A)
+    s = create_eigensolver(scope({"eig_solver": "ARNOLDI",
+                                  "eig_max_iters": 80,
+                                  "eig_tolerance": 1e-9}), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert st.converged and abs(st.eigenvalues[0] - hi) < 1e-6 * hi
+
+
+def test_lanczos_smallest_via_inverse():
+    A = poisson_2d(8, 8)
+    lo, hi = ref_extreme_eigs(A)
+    s = create_eigensolver(scope({"eig_solver": "LANCZOS",
+                                  "eig_max_iters": 60,
+                                  "eig_tolerance": 1e-8,
+                                  "eig_which": "smallest"}), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert abs(st.eigenvalues[0] - lo) < 1e-5 * hi
+
+
+def test_lobpcg_smallest():
+    A = poisson_2d(10, 10)
+    lo, hi = ref_extreme_eigs(A)
+    s = create_eigensolver(scope({"eig_solver": "LOBPCG",
+                                  "eig_max_iters": 300,
+                                  "eig_tolerance": 1e-7}), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert st.converged
+    assert abs(st.eigenvalues[0] - lo) < 1e-4 * hi
+
+
+def test_subspace_iteration():
+    A = poisson_2d(10, 10)
+    lo, hi = ref_extreme_eigs(A)
+    s = create_eigensolver(scope({"eig_solver": "SUBSPACE_ITERATION",
+                                  "eig_max_iters": 500,
+                                  "eig_tolerance": 1e-9,
+                                  "eig_wanted_count": 2}), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert st.converged
+    assert abs(st.eigenvalues[0] - hi) < 1e-4 * hi
+
+
+def test_pagerank():
+    """PageRank on a tiny digraph: stationary vector of the Google matrix."""
+    import scipy.sparse as sp
+    from amgx_amd.matrix import CSRMatrix
+    # 4-node graph
+    edges = [(0, 1), (0, 2), (1, 2), (2, 0), (3, 2)]
+    m = sp.csr_matrix((np.ones(len(edges)),
+                       ([e[0] for e in edges], [e[1] for e in edges])),
+                      shape=(4, 4))
+    A = CSRMatrix.from_scipy(m)
+    s = create_eigensolver(scope({"eig_solver": "POWER_ITERATION",
+                                  "eig_max_iters": 500,
+                                  "eig_tolerance": 1e-10}), Resources("cpu"))
+    s.pagerank_setup(A, damping=0.85)
+    st = s.solve()
+    pr = st.eigenvector / st.eigenvector.sum()
+    # dense reference
+    d = 0.85
+    P = np.zeros((4, 4))
+    out = np.asarray(m.sum(axis=1)).ravel()
+    for i, j in edges:
+        P[j, i] = 1.0 / out[i]
+    G = d * P + np.where(out == 0, d / 4.0, 0.0)[None, :] + (1 - d) / 4.0
+    evals, evecs = np.linalg.eig(G)
+    k = np.argmax(np.real(evals))
+    ref = np.real(evecs[:, k])
+    ref = ref / ref.sum()
+    assert np.allclose(pr.numpy(), ref, atol=1e-6)
