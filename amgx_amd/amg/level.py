@@ -172,6 +172,7 @@ class ClassicalLevel(AMGLevel):
         self.S = S
         self.cf_map = cf
         self.num_coarse = nc
+        self.A._cache["cf_map"] = cf   # CF_JACOBI smoother reads this
         return nc
 
     def create_coarse_matrix(self) -> CSRMatrix:

@@ -9,6 +9,7 @@ from . import dilu              # noqa: F401  (registers MULTICOLOR_DILU)
 from . import dense_lu          # noqa: F401  (registers DENSE_LU_SOLVER)
 from . import amg_solver        # noqa: F401  (registers AMG)
 from . import ilu               # noqa: F401  (registers MULTICOLOR_ILU)
+from . import kaczmarz          # noqa: F401  (registers KACZMARZ/CF_JACOBI)
 
 __all__ = ["Solver", "SolveStatus", "Convergence", "create_solver",
            "register_solver", "SOLVER_REGISTRY"]

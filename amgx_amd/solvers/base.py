@@ -124,6 +124,19 @@ class Solver:
     def setup(self, A):
         t0 = time.perf_counter()
         self.A = A
+        self.scaler = None
+        scaling = self.scope.get("scaling")
+        if scaling and scaling != "NONE":
+            from ..scalers import create_scaler
+            prev = getattr(A, "_scaled_by", None)
+            if prev is not None:      # re-setup: undo the previous scaling
+                prev.d = 1.0 / prev.d
+                prev.scale_matrix(A)
+                A._scaled_by = None
+            self.scaler = create_scaler(scaling)
+            self.scaler.setup(A)
+            self.scaler.scale_matrix(A)
+            A._scaled_by = self.scaler
         self.solver_setup()
         self.res.synchronize() if self.res.is_cuda else None
         self.status.setup_time = time.perf_counter() - t0
@@ -167,8 +180,20 @@ class Solver:
     # -- solve ---------------------------------------------------------------
     def solve(self, b: torch.Tensor, x: torch.Tensor,
               zero_initial_guess: bool = False) -> SolveStatus:
-        """Reference Solver::solve (src/solvers/solver.cu:586): initial
-        residual + norm, solve_init, iterate until converged/max_iters."""
+        """Reference Solver::solve (src/solvers/solver.cu:586): scaler hooks,
+        initial residual + norm, solve_init, iterate until
+        converged/max_iters."""
+        if getattr(self, "scaler", None) is not None:
+            bs = self.scaler.scale_rhs(b)
+            if not zero_initial_guess:
+                self.scaler.scale_guess(x)
+            st = self._solve_inner(bs, x, zero_initial_guess)
+            self.scaler.unscale_solution(x)
+            return st
+        return self._solve_inner(b, x, zero_initial_guess)
+
+    def _solve_inner(self, b: torch.Tensor, x: torch.Tensor,
+                     zero_initial_guess: bool = False) -> SolveStatus:
         st = self.status = SolveStatus()
         t0 = time.perf_counter()
         if zero_initial_guess:
