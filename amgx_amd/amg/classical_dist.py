@@ -1,0 +1,291 @@
+"""Distributed classical (Ruge-Stueben) AMG setup: PMIS C/F splitting with
+halo-state rounds, distance-1 interpolation onto GLOBAL coarse columns, and
+the distributed Galerkin triple product RAP = P^T A P with external-row
+accumulation.
+
+Reference behavior: src/classical/classical_amg_level.cu:657-850
+(computeAOperator_1x1_distributed: exchange_halo_rows_P, RAP_int +
+exchange_RAP_ext + csr_RAP_sparse_add) and the distributed PMIS in
+src/classical/selectors/pmis.cu. The MI355X redesign keeps the same
+capability on top of torch.distributed (RCCL over xGMI / gloo on host):
+
+* PMIS rounds exchange halo weights/states as plain halo-vector exchanges
+  (no 2-ring matrix exchange; strength of the incoming edge j->i is taken as
+  |a_ij| >= theta*rowmax_j with the owner's rowmax fetched by one halo
+  exchange — exact for symmetric-value matrices, the common AMG case).
+* The fine-level weight is lam + hash(global row id), a total order that is
+  invariant to the partitioning, so C/F splits agree with the single-rank
+  split up to the strength approximation above.
+* P's coarse columns get their own HaloExchange (coarse-column halo), which
+  also drives prolongation gather and restriction scatter-add at solve time.
+* RAP is computed rank-locally over P_ext (own rows + exchanged halo rows of
+  P) and external coarse-row contributions are shipped to their owners.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import scipy.sparse as sp
+import torch
+import torch.distributed as tdist
+
+from ..distributed.manager import (DistributedManager, HaloExchange,
+                                   exchange_csr_rows)
+
+
+def _hash01(gid: np.ndarray) -> np.ndarray:
+    """Deterministic partition-invariant pseudo-random weight in [0,1) from a
+    global row id (reference: hashed random weights in pmis.cu)."""
+    x = gid.astype(np.uint64, copy=True)
+    x = (x ^ (x >> np.uint64(33))) * np.uint64(0xFF51AFD7ED558CCD)
+    x = (x ^ (x >> np.uint64(33))) * np.uint64(0xC4CEB9FE1A85EC53)
+    x = x ^ (x >> np.uint64(33))
+    return (x >> np.uint64(11)).astype(np.float64) / float(1 << 53)
+
+
+def strength_dist(A, mgr: DistributedManager, theta: float,
+                  max_row_sum: float):
+    """Return (strong_out, strong_union) boolean masks over A's entries.
+    strong_out: |a_ij| >= theta * rowmax_i (the serial AHAT criterion on
+    owned rows). strong_union additionally includes incoming strength
+    |a_ij| >= theta * rowmax_j using the exchanged halo rowmax."""
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    v = A.values.cpu().numpy().astype(np.float64).reshape(-1)
+    n = A.n_rows
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    off = rows != ci
+    absv = np.abs(v)
+    rowmax = np.zeros(n)
+    np.maximum.at(rowmax, rows[off], absv[off])
+    # halo rowmax by one exchange
+    rm_ext = mgr.new_ext_vec(torch.float64)
+    rm_ext[:n] = torch.from_numpy(rowmax)
+    mgr.exchange_halo(rm_ext, block_override=1)
+    rm_all = rm_ext.cpu().numpy()
+    strong_out = off & (absv >= theta * rowmax[rows]) & (rowmax[rows] > 0)
+    strong_in = off & (absv >= theta * rm_all[ci]) & (rm_all[ci] > 0)
+    if max_row_sum < 1.0:
+        rs = np.zeros(n)
+        np.add.at(rs, rows, v)
+        d = np.zeros(n)
+        diag_mask = rows == ci
+        d[rows[diag_mask]] = np.abs(v[diag_mask])
+        weak_rows = np.abs(rs) > max_row_sum * np.where(d > 0, d, 1.0)
+        strong_out &= ~weak_rows[rows]
+        strong_in &= ~weak_rows[rows]
+    return strong_out, strong_out | strong_in
+
+
+def pmis_dist(A, mgr: DistributedManager, strong_union: np.ndarray):
+    """Distributed PMIS: Luby-style independent-set rounds over the strong
+    graph with per-round halo state exchange. Returns (cf_local int32 with
+    the local coarse index for C rows / -1 for F, nc_local)."""
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    n = A.n_rows
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    er, ec = rows[strong_union], ci[strong_union]   # edges incident to my rows
+
+    # lam_i = number of strong incoming edges (S^T row count); contributions
+    # to halo columns are summed into their owners
+    lam_ext = mgr.new_ext_vec(torch.float64)
+    lam_host = np.zeros(mgr.n_local + mgr.n_halo)
+    np.add.at(lam_host, ec, 1.0)
+    lam_ext.copy_(torch.from_numpy(lam_host).to(lam_ext.device))
+    mgr.add_from_halo(lam_ext, block_override=1)
+
+    # partition-invariant tie-broken weights
+    gid_own = (mgr.row_perm.cpu().numpy() + mgr.row_start)
+    w_ext = mgr.new_ext_vec(torch.float64)
+    w_ext[:n] = lam_ext[:n] + torch.from_numpy(_hash01(gid_own)).to(w_ext.dtype)
+    mgr.exchange_halo(w_ext, block_override=1)
+    w = w_ext.cpu().numpy()
+    gid_ext = (np.concatenate([gid_own, mgr.halo_global]) if mgr.n_halo
+               else gid_own)
+
+    state_ext = mgr.new_ext_vec(torch.float64)   # 0 undec, 1 C, -1 F
+    st = np.zeros(mgr.n_local + mgr.n_halo)
+    # isolated rows (no strong edges): F
+    has_edge = np.zeros(n, dtype=bool)
+    has_edge[er] = True
+    st[:n][~has_edge] = -1.0
+
+    def _sync_state():
+        state_ext[:mgr.n_local + mgr.n_halo] = torch.from_numpy(st) \
+            .to(state_ext.dtype)
+        mgr.exchange_halo(state_ext, block_override=1)
+        st[mgr.n_local:] = state_ext[mgr.n_local:].cpu().numpy()
+
+    _sync_state()
+    guard = 0
+    while True:
+        undec_local = int((st[:n] == 0).sum())
+        total = mgr.global_sum(float(undec_local))
+        if total == 0:
+            break
+        guard += 1
+        if guard > 10 * max(1, int(np.log2(mgr.n_global + 2)) + 8):
+            raise RuntimeError("distributed PMIS failed to converge")
+        # a row is new-C if undecided and a strict (w, gid) local max among
+        # its undecided strong neighbors (total order -> guaranteed progress)
+        und_row = st[:n] == 0
+        e_act = und_row[er] & (st[ec] == 0) & (er != ec)
+        beaten = np.zeros(n, dtype=bool)
+        if e_act.any():
+            ea_r, ea_c = er[e_act], ec[e_act]
+            wr, wc = w[ea_r], w[ea_c]
+            gr, gc = gid_ext[ea_r], gid_ext[ea_c]
+            loses = (wc > wr) | ((wc == wr) & (gc > gr))
+            np.logical_or.at(beaten, ea_r, loses)
+        new_c = und_row & ~beaten
+        st[:n][new_c] = 1.0
+        _sync_state()
+        # undecided rows with a C strong neighbor -> F
+        e_f = (st[:n][er] == 0) & (st[ec] == 1.0)
+        st[:n][np.unique(er[e_f])] = -1.0
+        _sync_state()
+
+    cf = np.full(n, -1, dtype=np.int32)
+    c_rows = np.nonzero(st[:n] == 1.0)[0]
+    cf[c_rows] = np.arange(c_rows.size, dtype=np.int32)
+    return cf, int(c_rows.size)
+
+
+def coarse_numbering(mgr: DistributedManager, cf: np.ndarray, nc_local: int):
+    """Global coarse ids: owned C rows numbered coarse_start + local index;
+    halo columns' coarse ids fetched by one halo exchange. Returns
+    (cf_ext int64 len n_cols with global coarse id or -1, coarse_offsets)."""
+    counts = [None] * mgr.world
+    tdist.all_gather_object(counts, nc_local)
+    coarse_offs = np.zeros(mgr.world + 1, dtype=np.int64)
+    coarse_offs[1:] = np.cumsum([int(c) for c in counts])
+    cs = int(coarse_offs[mgr.rank])
+    gc_ext = mgr.new_ext_vec(torch.float64)
+    own = np.where(cf >= 0, cf.astype(np.float64) + cs, -1.0)
+    gc_ext[:mgr.n_local] = torch.from_numpy(own).to(gc_ext.dtype)
+    mgr.exchange_halo(gc_ext, block_override=1)
+    cf_ext = gc_ext.cpu().numpy().round().astype(np.int64)
+    return cf_ext, coarse_offs
+
+
+def rap_dist(A, mgr: DistributedManager, P_own: sp.csr_matrix,
+             coarse_offs: np.ndarray):
+    """Distributed Galerkin RAP. P_own: owned fine rows x GLOBAL coarse cols.
+    Exchanges halo rows of P, forms P_own^T (A_loc P_ext) and ships external
+    coarse-row contributions to their owners. Returns the owned coarse rows
+    as a scipy CSR with global coarse columns."""
+    n, n_ext = mgr.n_local, mgr.n_local + mgr.n_halo
+    ngc = int(coarse_offs[-1])
+    # halo rows of P (reference exchange_halo_rows_P)
+    halo_rows = exchange_csr_rows(mgr, P_own.indptr, P_own.indices,
+                                  P_own.data)
+    hr, hc, hv = [], [], []
+    for pos, rowdat in enumerate(halo_rows):
+        if rowdat is None:
+            continue
+        cols, vals = rowdat
+        hr.append(np.full(cols.size, n + pos, dtype=np.int64))
+        hc.append(cols)
+        hv.append(vals)
+    if hr:
+        halo_part = sp.csr_matrix(
+            (np.concatenate(hv), (np.concatenate(hr), np.concatenate(hc))),
+            shape=(n_ext, ngc))
+    else:
+        halo_part = sp.csr_matrix((n_ext, ngc))
+    P_ext = sp.vstack([P_own, sp.csr_matrix((mgr.n_halo, ngc))]).tocsr() \
+        + halo_part
+    A_loc = sp.csr_matrix(
+        (A.values.cpu().numpy().astype(np.float64).reshape(-1),
+         A.col_indices.cpu().numpy().astype(np.int64),
+         A.row_offsets.cpu().numpy().astype(np.int64)), shape=(n, n_ext))
+    T = A_loc @ P_ext                      # owned fine x global coarse
+    C = (P_own.T @ T).tocsr()              # global coarse x global coarse
+    C.sum_duplicates()
+    cs, ce = int(coarse_offs[mgr.rank]), int(coarse_offs[mgr.rank + 1])
+    mine = C[cs:ce]
+    # external rows -> owners (reference exchange_RAP_ext + sparse_add)
+    frags = []
+    for r in range(mgr.world):
+        if r == mgr.rank:
+            frags.append(None)
+            continue
+        lo, hi = int(coarse_offs[r]), int(coarse_offs[r + 1])
+        block = C[lo:hi]
+        frags.append((block.indptr, block.indices, block.data)
+                     if block.nnz else None)
+    gathered = [None] * mgr.world
+    tdist.all_gather_object(gathered, frags)
+    acc = mine
+    add = sp.csr_matrix(acc.shape)
+    for r in range(mgr.world):
+        if r == mgr.rank or gathered[r] is None:
+            continue
+        frag = gathered[r][mgr.rank]
+        if frag is None:
+            continue
+        indptr, indices, data = frag
+        add = add + sp.csr_matrix((data, indices, indptr), shape=acc.shape)
+    out = (acc + add).tocsr()
+    out.sum_duplicates()
+    out.eliminate_zeros()
+    return out
+
+
+class ClassicalDistOperators:
+    """Solve-time restriction/prolongation operators for one distributed
+    classical level: local P/R CSR with coarse-ext columns plus the coarse
+    HaloExchange that completes P^T r (scatter-add to owners) and P xc
+    (gather of halo coarse values)."""
+
+    def __init__(self, A, mgr, P_own: sp.csr_matrix, Ac, coarse_offs):
+        from ..matrix import CSRMatrix
+        mgr_c = Ac.manager
+        cs, ce = int(coarse_offs[mgr.rank]), int(coarse_offs[mgr.rank + 1])
+        nc_local = ce - cs
+        cols = P_own.indices.astype(np.int64)
+        own_mask = (cols >= cs) & (cols < ce)
+        halo_gids = np.unique(cols[~own_mask])
+        iperm_c = mgr_c.row_iperm.cpu().numpy().astype(np.int64)
+        self.halomap = HaloExchange(halo_gids, coarse_offs,
+                                    device=A.device,
+                                    owner_local_map=iperm_c)
+        new_cols = np.empty_like(cols)
+        new_cols[own_mask] = iperm_c[cols[own_mask] - cs]
+        new_cols[~own_mask] = nc_local + np.searchsorted(halo_gids,
+                                                         cols[~own_mask])
+        self.nc_local = nc_local
+        self.n_halo_p = int(halo_gids.size)
+        dev = A.device
+        self.P = CSRMatrix(
+            torch.from_numpy(P_own.indptr.astype(np.int32)).to(dev),
+            torch.from_numpy(new_cols.astype(np.int32)).to(dev),
+            torch.from_numpy(P_own.data.astype(np.float64)).to(A.dtype)
+            .to(dev),
+            n_cols=nc_local + self.n_halo_p)
+        from .. import ops
+        self.R = ops.transpose(self.P)
+        self.owned_c = mgr_c.owned_size
+
+    def restrict(self, r, bc):
+        from .. import ops
+        n_fine_owned = self.P.n_rows
+        t = torch.zeros(self.nc_local + self.n_halo_p, dtype=r.dtype,
+                        device=r.device)
+        ops.spmv(self.R, r.reshape(-1)[:n_fine_owned], t)
+        bc.zero_()
+        bc_owned = bc.reshape(-1)[:self.owned_c]
+        bc_owned.copy_(t[:self.nc_local])
+        self.halomap.reverse_add(t[self.nc_local:], bc_owned)
+
+    def prolongate_add(self, xc, x):
+        from .. import ops
+        xc_ext = torch.empty(self.nc_local + self.n_halo_p, dtype=xc.dtype,
+                             device=xc.device)
+        xc_ext[:self.nc_local] = xc.reshape(-1)[:self.nc_local]
+        self.halomap.forward(xc.reshape(-1)[:self.owned_c],
+                             xc_ext[self.nc_local:])
+        tmp = torch.zeros(self.P.n_rows, dtype=x.dtype, device=x.device)
+        ops.spmv(self.P, xc_ext, tmp)
+        x.reshape(-1)[:self.P.n_rows] += tmp

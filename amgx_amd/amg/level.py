@@ -165,6 +165,18 @@ class ClassicalLevel(AMGLevel):
         self.num_coarse = 0
 
     def create_coarse_vertices(self) -> int:
+        mgr = getattr(self.A, "manager", None)
+        if mgr is not None:
+            from .classical_dist import pmis_dist, strength_dist
+            theta = float(self.scope.get("strength_threshold"))
+            mrs = float(self.scope.get("max_row_sum"))
+            self._strong_out, strong_union = strength_dist(
+                self.A, mgr, theta, mrs)
+            cf, nc = pmis_dist(self.A, mgr, strong_union)
+            self.cf_map = torch.from_numpy(cf)
+            self.num_coarse = nc
+            self.A._cache["cf_map"] = self.cf_map
+            return nc
         S = ops._backend(self.A).strength_ahat(
             self.A, float(self.scope.get("strength_threshold")),
             float(self.scope.get("max_row_sum")))
@@ -176,7 +188,9 @@ class ClassicalLevel(AMGLevel):
         return nc
 
     def create_coarse_matrix(self) -> CSRMatrix:
-        interp = self.scope.get("interpolator")
+        mgr = getattr(self.A, "manager", None)
+        if mgr is not None:
+            return self._create_coarse_matrix_distributed(mgr)
         self.P = ops._backend(self.A).interp_d1(self.A, self.S, self.cf_map,
                                                 self.num_coarse)
         tf = float(self.scope.get("interp_truncation_factor"))
@@ -187,10 +201,54 @@ class ClassicalLevel(AMGLevel):
         Ac = ops.galerkin_rap(self.R, self.A, self.P)
         return Ac
 
+    def _create_coarse_matrix_distributed(self, mgr) -> CSRMatrix:
+        """Distributed classical coarsening (reference
+        computeAOperator_1x1_distributed, src/classical/
+        classical_amg_level.cu:657-850): D1 interpolation onto global coarse
+        columns, halo-row exchange of P, RAP with external-row shipping."""
+        import numpy as np
+        import scipy.sparse as sp
+
+        from .classical_dist import (ClassicalDistOperators,
+                                     coarse_numbering, rap_dist)
+        from ..distributed.manager import DistributedManager
+        from ..ops import cpu as cpu_ops
+        A = self.A
+        cf = self.cf_map.numpy()
+        cf_ext, coarse_offs = coarse_numbering(mgr, cf, self.num_coarse)
+        A_host = CSRMatrix(A.row_offsets.cpu(), A.col_indices.cpu(),
+                           A.values.cpu(), n_cols=A.n_cols)
+        P_m = cpu_ops.interp_d1(A_host, torch.from_numpy(self._strong_out),
+                                torch.from_numpy(cf_ext),
+                                int(coarse_offs[-1]))
+        tf = float(self.scope.get("interp_truncation_factor"))
+        me = int(self.scope.get("interp_max_elements"))
+        if tf > 0.0 or me >= 0:
+            P_m = ops.truncate_rows(P_m, tf, me)
+        P_own = sp.csr_matrix(
+            (P_m.values.cpu().numpy().astype(np.float64),
+             P_m.col_indices.cpu().numpy().astype(np.int64),
+             P_m.row_offsets.cpu().numpy().astype(np.int64)),
+            shape=(A.n_rows, int(coarse_offs[-1])))
+        rap = rap_dist(A, mgr, P_own, coarse_offs)
+        Ac = DistributedManager.upload_global_csr(
+            rap.indptr, rap.indices, rap.data, self.num_coarse,
+            int(coarse_offs[mgr.rank]), int(coarse_offs[-1]),
+            device=A.device, block_dim=1, dtype=A.dtype)
+        self._dist_ops = ClassicalDistOperators(A, mgr, P_own, Ac,
+                                                coarse_offs)
+        return Ac
+
     def restrict_residual(self, r, bc):
+        if getattr(self, "_dist_ops", None) is not None:
+            self._dist_ops.restrict(r, bc)
+            return
         ops.spmv(self.R, r, bc)
 
     def prolongate_and_apply(self, xc, x):
+        if getattr(self, "_dist_ops", None) is not None:
+            self._dist_ops.prolongate_add(xc, x)
+            return
         tmp = torch.zeros_like(x)
         ops.spmv(self.P, xc, tmp)
         ops.axpy(x, tmp, 1.0)

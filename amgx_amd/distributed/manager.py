@@ -238,6 +238,40 @@ class DistributedManager:
             rq.wait()
         return None
 
+    def add_from_halo(self, x: torch.Tensor, block_override: int = None):
+        """Reverse halo exchange: every rank sends its halo tail slices back
+        to the owners, which sum them into the owned entries (reference
+        DistributedComms::add_from_halo, include/distributed/
+        distributed_comms.h:197-201 — used by classical restriction and
+        consolidation)."""
+        if not self.neighbors:
+            return
+        xf = x.reshape(-1)
+        b = self.block_dim if block_override is None else block_override
+        p2p, recv = [], []
+        for i, r in enumerate(self.neighbors):
+            lo, hi = self.halo_slices[i]
+            if hi > lo:
+                sbuf = xf[(self.n_local + lo) * b:(self.n_local + hi) * b] \
+                    .contiguous()
+                p2p.append(dist.P2POp(dist.isend, sbuf, r))
+            rbuf = torch.empty(int(self.b2l[i].numel()) * b, dtype=x.dtype,
+                               device=x.device)
+            recv.append(rbuf)
+            if rbuf.numel():
+                p2p.append(dist.P2POp(dist.irecv, rbuf, r))
+        if p2p:
+            for rq in dist.batch_isend_irecv(p2p):
+                rq.wait()
+        for i in range(len(self.neighbors)):
+            if not recv[i].numel():
+                continue
+            idx = self.b2l[i].to(torch.int64)
+            if b != 1:
+                idx = (idx[:, None] * b + torch.arange(
+                    b, dtype=torch.int64, device=x.device)[None, :]).reshape(-1)
+            xf.index_add_(0, idx, recv[i])
+
     # -------------------------------------------------------------- reductions
     def global_sum(self, v: float) -> float:
         t = torch.tensor([v], dtype=torch.float64,
@@ -283,3 +317,161 @@ class DistributedManager:
         idx = (self.row_iperm[:, None] * b
                + torch.arange(b, device=v.device)[None, :]).reshape(-1)
         return vf[idx]
+
+
+class HaloExchange:
+    """Generic gather/scatter exchange for an arbitrary set of needed global
+    ids over a 1-D row partition. This factors the reference's B2L-map
+    machinery (include/distributed/distributed_arranger.h create_B2L_from_maps
+    / create_neighbors_v2) into a reusable object: the classical-AMG
+    interpolation operator P has its own coarse-column halo, distinct from the
+    matrix's fine halo, and drives its prolongation gather / restriction
+    scatter-add through one of these.
+
+    Collective constructor. ``needed_global``: sorted unique global ids this
+    rank needs but does not own. ``part_offsets``: partition offsets of the
+    owned space (len world+1). ``owner_local_map``: optional owner-side
+    renumbering (old-local -> internal id) applied to the ids other ranks
+    request from me — e.g. the coarse matrix's interior-first permutation.
+    """
+
+    def __init__(self, needed_global, part_offsets, device="cpu",
+                 owner_local_map=None):
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        self.world = dist.get_world_size() if dist.is_initialized() else 1
+        needed = np.asarray(needed_global, dtype=np.int64)
+        offs = np.asarray(part_offsets, dtype=np.int64)
+        self.device = torch.device(device)
+        self.n_needed = int(needed.size)
+        owners = np.searchsorted(offs, needed, side="right") - 1
+        needed_by_owner = [needed[owners == r] for r in range(self.world)]
+        all_needs = [None] * self.world
+        dist.all_gather_object(all_needs, needed_by_owner)
+        self.neighbors, self.b2l, self.slices = [], [], []
+        for r in range(self.world):
+            if r == self.rank:
+                continue
+            they = np.asarray(all_needs[r][self.rank], dtype=np.int64)
+            mine = needed_by_owner[r]
+            if they.size == 0 and mine.size == 0:
+                continue
+            self.neighbors.append(r)
+            loc = they - offs[self.rank]
+            if owner_local_map is not None:
+                olm = np.asarray(owner_local_map, dtype=np.int64)
+                loc = olm[loc]
+            self.b2l.append(torch.from_numpy(loc).to(self.device))
+            if mine.size:
+                lo = int(np.searchsorted(needed, mine[0]))
+                hi = lo + int(mine.size)
+            else:
+                lo = hi = 0
+            self.slices.append((lo, hi))
+
+    def forward(self, src: torch.Tensor, dst_tail: torch.Tensor):
+        """Fill ``dst_tail`` (length n_needed, sorted-global order) with the
+        owners' values of the needed ids; ``src`` is each owner's owned vector
+        in its internal order."""
+        p2p, keep = [], []
+        for i, r in enumerate(self.neighbors):
+            buf = src[self.b2l[i]].contiguous()
+            keep.append(buf)
+            if buf.numel():
+                p2p.append(dist.P2POp(dist.isend, buf, r))
+            lo, hi = self.slices[i]
+            if hi > lo:
+                p2p.append(dist.P2POp(dist.irecv, dst_tail[lo:hi], r))
+        if p2p:
+            for rq in dist.batch_isend_irecv(p2p):
+                rq.wait()
+
+    def reverse_add(self, contrib_tail: torch.Tensor, dst: torch.Tensor):
+        """Send my per-id contributions (``contrib_tail``, length n_needed) to
+        the owners, which sum them into ``dst`` (owner internal order)."""
+        p2p, recv = [], []
+        for i, r in enumerate(self.neighbors):
+            lo, hi = self.slices[i]
+            if hi > lo:
+                p2p.append(dist.P2POp(
+                    dist.isend, contrib_tail[lo:hi].contiguous(), r))
+            rbuf = torch.empty(int(self.b2l[i].numel()), dtype=dst.dtype,
+                               device=dst.device)
+            recv.append(rbuf)
+            if rbuf.numel():
+                p2p.append(dist.P2POp(dist.irecv, rbuf, r))
+        if p2p:
+            for rq in dist.batch_isend_irecv(p2p):
+                rq.wait()
+        for i in range(len(self.neighbors)):
+            if recv[i].numel():
+                dst.index_add_(0, self.b2l[i], recv[i])
+
+
+def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
+    """Matrix-halo exchange (reference DistributedComms::exchange_matrix_halo,
+    comms_mpi_hostbuffer_stream.cu:801-1000, tensor-payload redesign): every
+    rank sends, per neighbor, the CSR rows listed in its B2L map and receives
+    the rows backing its own halo slots. Two phases of batched isend/irecv —
+    per-row counts (sizes known from the maps), then column/value payloads.
+
+    indptr/indices/data: the LOCAL sparse rows being shared (numpy, global
+    column ids). Returns (counts, cols, vals) per halo slot: list over halo
+    positions 0..n_halo of (cols, vals) arrays.
+    """
+    indptr = np.asarray(indptr, dtype=np.int64)
+    counts_all = np.diff(indptr)
+    # phase 1: per-row nnz counts
+    p2p = []
+    send_counts, recv_counts = [], []
+    for i, r in enumerate(mgr.neighbors):
+        rows = mgr.b2l[i].cpu().numpy().astype(np.int64)
+        sc = torch.from_numpy(counts_all[rows].astype(np.int64))
+        send_counts.append((rows, sc))
+        if sc.numel():
+            p2p.append(dist.P2POp(dist.isend, sc, r))
+        lo, hi = mgr.halo_slices[i]
+        rc = torch.empty(hi - lo, dtype=torch.int64)
+        recv_counts.append(rc)
+        if rc.numel():
+            p2p.append(dist.P2POp(dist.irecv, rc, r))
+    if p2p:
+        for rq in dist.batch_isend_irecv(p2p):
+            rq.wait()
+    # phase 2: concatenated cols + vals per neighbor
+    p2p = []
+    recv_payload = []
+    keep = []
+    for i, r in enumerate(mgr.neighbors):
+        rows, sc = send_counts[i]
+        nz = np.concatenate([np.arange(indptr[j], indptr[j + 1])
+                             for j in rows]) if rows.size else \
+            np.zeros(0, dtype=np.int64)
+        scol = torch.from_numpy(np.asarray(indices, dtype=np.int64)[nz])
+        sval = torch.from_numpy(np.asarray(data, dtype=np.float64)[nz])
+        keep += [scol, sval]
+        if scol.numel():
+            p2p.append(dist.P2POp(dist.isend, scol, r))
+            p2p.append(dist.P2POp(dist.isend, sval, r))
+        tot = int(recv_counts[i].sum())
+        rcol = torch.empty(tot, dtype=torch.int64)
+        rval = torch.empty(tot, dtype=torch.float64)
+        recv_payload.append((rcol, rval))
+        if tot:
+            p2p.append(dist.P2POp(dist.irecv, rcol, r))
+            p2p.append(dist.P2POp(dist.irecv, rval, r))
+    if p2p:
+        for rq in dist.batch_isend_irecv(p2p):
+            rq.wait()
+    # scatter into per-halo-slot rows
+    out = [None] * mgr.n_halo
+    for i in range(len(mgr.neighbors)):
+        lo, hi = mgr.halo_slices[i]
+        rc = recv_counts[i].numpy()
+        rcol, rval = recv_payload[i]
+        rcol, rval = rcol.numpy(), rval.numpy()
+        pos = 0
+        for k in range(hi - lo):
+            c = int(rc[k])
+            out[lo + k] = (rcol[pos:pos + c], rval[pos:pos + c])
+            pos += c
+    return out
