@@ -1,10 +1,27 @@
-"""Matrix coloring attachment (reference src/matrix_coloring/, 6,860 LoC).
+"""Matrix coloring subsystem (reference src/matrix_coloring/, 6,860 LoC,
+registered schemes at src/core.cu:489-506).
 
-A coloring partitions rows so no two adjacent rows (distance-1) share a
-color; multicolor smoothers (GS/DILU/ILU/Kaczmarz) sweep color by color with
-full parallelism inside a color. The GPU implements the parallel MIN_MAX
-hash-based local-maximum scheme (reference src/matrix_coloring/min_max.cu);
-the host reference is sequential greedy.
+A coloring partitions rows so no two adjacent rows (distance-1, or
+distance-2 when ``coloring_level=2``) share a color; multicolor smoothers
+(GS/DILU/ILU/Kaczmarz) sweep color by color with full parallelism inside a
+color. Scheme registry mirrors the reference factory names:
+
+  MIN_MAX               hash local-maximum independent sets (min_max.cu) —
+                        the default; GPU path is a hand-written gfx950 kernel
+  PARALLEL_GREEDY       Jones-Plassmann greedy, smallest feasible color
+                        (parallel_greedy.cu)
+  SERIAL_GREEDY_BFS     BFS-ordered sequential greedy (serial_greedy_bfs.cu)
+  GREEDY_RECOLOR        MIN_MAX then greedy recolor-down pass
+                        (min_max_2ring/greedy_recolor.cu)
+  MULTI_HASH            per-round hash local-max (multi_hash.cu)
+  ROUND_ROBIN           greedy with rotating first-fit start (round_robin.cu)
+  UNIFORM               index-pattern coloring for banded/structured rows
+                        (uniform.cu; validity not guaranteed on general
+                        graphs, exactly like the reference)
+  MIN_MAX_2RING         MIN_MAX on the distance-2 graph (min_max_2ring.cu)
+  GREEDY_MIN_MAX_2RING  greedy on the distance-2 graph
+  LOCALLY_DOWNWIND      downwind-ordered greedy for Kaczmarz sweeps
+                        (locally_downwind.cu)
 
 The attachment precomputes ``rows_sorted`` (row ids stably sorted by color,
 device-resident) and host-side ``bounds`` so every per-color kernel slice is
@@ -13,9 +30,262 @@ a view — no per-sweep index rebuilds.
 
 from __future__ import annotations
 
+from typing import Callable, Dict
+
+import numpy as np
 import torch
 
 from .. import ops
+
+COLORING_REGISTRY: Dict[str, Callable] = {}
+
+
+def register_coloring(name: str):
+    def deco(fn):
+        COLORING_REGISTRY[name] = fn
+        return fn
+    return deco
+
+
+def _host_adj(A, level: int = 1):
+    """Row adjacency (indptr, indices) on host, self-edges removed, halo
+    columns dropped; level=2 squares the graph (distance-2 neighbors)."""
+    import scipy.sparse as sp
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    n = A.n_rows
+    keep = ci < n
+    rows = np.repeat(np.arange(n), np.diff(ro))[keep]
+    cols = ci[keep]
+    g = sp.csr_matrix((np.ones(rows.size, dtype=np.int8), (rows, cols)),
+                      shape=(n, n))
+    g = g + g.T
+    if level >= 2:
+        g = g @ g
+    g = g.tocsr()
+    g.setdiag(0)
+    g.eliminate_zeros()
+    return g.indptr, g.indices
+
+
+def _greedy(indptr, indices, order):
+    n = indptr.size - 1
+    colors = np.full(n, -1, dtype=np.int32)
+    for i in order:
+        nb = indices[indptr[i]:indptr[i + 1]]
+        used = set(int(colors[j]) for j in nb if colors[j] >= 0)
+        c = 0
+        while c in used:
+            c += 1
+        colors[i] = c
+    return colors, (int(colors.max()) + 1 if n else 0)
+
+
+def _hash(i: np.ndarray, salt: int) -> np.ndarray:
+    """uint64 bijection of the index (tie-free weights)."""
+    s = np.uint64((salt * 0x9E3779B97F4A7C15) % (1 << 64))
+    x = (i.astype(np.uint64) + s) * np.uint64(0xBF58476D1CE4E5B9)
+    return (x ^ (x >> np.uint64(31))).astype(np.uint64)
+
+
+@register_coloring("MIN_MAX")
+def _min_max(A, scope, level):
+    if level <= 1:
+        frac = scope.get("max_uncolored_percentage") if scope is not None \
+            else 0.0
+        colors, num = ops.color_matrix(A, max_uncolored_frac=float(frac or 0))
+        return colors.cpu().numpy().astype(np.int32), num
+    return _min_max_rounds(*_host_adj(A, level))
+
+
+def _min_max_rounds(indptr, indices):
+    """Luby rounds: row is colored c at round c if its hashed weight is a
+    local max/min among uncolored neighbors (reference min_max.cu: two colors
+    per round, max and min)."""
+    n = indptr.size - 1
+    colors = np.full(n, -1, dtype=np.int32)
+    w = _hash(np.arange(n), 1)
+    c = 0
+    guard = 0
+    while (colors < 0).any():
+        un = colors < 0          # frozen snapshot: all decisions this round
+        rows = np.nonzero(un)[0]  # are made against the same uncolored set
+        newc = []
+        for i in rows:
+            nb = indices[indptr[i]:indptr[i + 1]]
+            nb = nb[un[nb]]
+            nbw = w[nb[nb != i]]
+            # _hash is a uint64 bijection, so weights are tie-free
+            if nbw.size == 0 or w[i] > nbw.max():
+                newc.append((i, c))
+            elif w[i] < nbw.min():
+                newc.append((i, c + 1))
+        for i, cc in newc:
+            colors[i] = cc
+        c += 2
+        guard += 1
+        if guard > n + 2:
+            raise RuntimeError("MIN_MAX coloring did not converge")
+    return colors, int(colors.max()) + 1
+
+
+@register_coloring("PARALLEL_GREEDY")
+def _parallel_greedy(A, scope, level):
+    # Jones-Plassmann with smallest-feasible-color (same algorithm the gfx950
+    # kernel runs); host model is sequential over JP rounds
+    indptr, indices = _host_adj(A, level)
+    n = indptr.size - 1
+    colors = np.full(n, -1, dtype=np.int32)
+    w = _hash(np.arange(n), 7)
+    guard = 0
+    while (colors < 0).any():
+        newly = []
+        for i in np.nonzero(colors < 0)[0]:
+            nb = indices[indptr[i]:indptr[i + 1]]
+            un_nb = nb[colors[nb] < 0]
+            un_nb = un_nb[un_nb != i]
+            if un_nb.size and not (w[i] > w[un_nb]).all():
+                continue
+            used = set(int(colors[j]) for j in nb if colors[j] >= 0)
+            c = 0
+            while c in used:
+                c += 1
+            newly.append((i, c))
+        for i, c in newly:
+            colors[i] = c
+        guard += 1
+        if guard > n + 2:
+            raise RuntimeError("PARALLEL_GREEDY coloring did not converge")
+    return colors, int(colors.max()) + 1 if n else 0
+
+
+@register_coloring("SERIAL_GREEDY_BFS")
+def _serial_greedy_bfs(A, scope, level):
+    import collections
+    indptr, indices = _host_adj(A, level)
+    n = indptr.size - 1
+    order = []
+    seen = np.zeros(n, dtype=bool)
+    for s in range(n):
+        if seen[s]:
+            continue
+        q = collections.deque([s])
+        seen[s] = True
+        while q:
+            i = q.popleft()
+            order.append(i)
+            for j in indices[indptr[i]:indptr[i + 1]]:
+                if not seen[j]:
+                    seen[j] = True
+                    q.append(j)
+    return _greedy(indptr, indices, np.asarray(order, dtype=np.int64))
+
+
+@register_coloring("GREEDY_RECOLOR")
+def _greedy_recolor(A, scope, level):
+    """MIN_MAX first, then one greedy pass in descending-color order that
+    moves every row down to its smallest feasible color."""
+    colors, num = COLORING_REGISTRY["MIN_MAX"](A, scope, level)
+    indptr, indices = _host_adj(A, level)
+    order = np.argsort(-colors, kind="stable")
+    for i in order:
+        nb = indices[indptr[i]:indptr[i + 1]]
+        used = set(int(colors[j]) for j in nb)
+        c = 0
+        while c in used:
+            c += 1
+        if c < colors[i]:
+            colors[i] = c
+    return colors, int(colors.max()) + 1 if colors.size else 0
+
+
+@register_coloring("MULTI_HASH")
+def _multi_hash(A, scope, level):
+    indptr, indices = _host_adj(A, level)
+    n = indptr.size - 1
+    max_hash = int(scope.get("max_num_hash")) if scope is not None else 7
+    colors = np.full(n, -1, dtype=np.int32)
+    for salt in range(max_hash):
+        w = _hash(np.arange(n), salt + 11)
+        un = colors < 0          # frozen snapshot per round
+        un_rows = np.nonzero(un)[0]
+        if un_rows.size == 0:
+            break
+        newc = []
+        for i in un_rows:
+            nb = indices[indptr[i]:indptr[i + 1]]
+            nb = nb[un[nb]]
+            nbw = w[nb[nb != i]]
+            if nbw.size == 0 or w[i] > nbw.max():
+                newc.append(i)
+        colors[np.asarray(newc, dtype=np.int64)] = salt
+    if (colors < 0).any():   # leftover rows: greedy cleanup
+        for i in np.nonzero(colors < 0)[0]:
+            nb = indices[indptr[i]:indptr[i + 1]]
+            used = set(int(colors[j]) for j in nb if colors[j] >= 0)
+            c = 0
+            while c in used:
+                c += 1
+            colors[i] = max(c, 0)
+    return colors, int(colors.max()) + 1 if n else 0
+
+
+@register_coloring("ROUND_ROBIN")
+def _round_robin(A, scope, level):
+    """Greedy with a rotating first-fit start color — spreads rows evenly
+    over colors (reference round_robin.cu intent)."""
+    indptr, indices = _host_adj(A, level)
+    n = indptr.size - 1
+    colors = np.full(n, -1, dtype=np.int32)
+    k = 1
+    for i in range(n):
+        nb = indices[indptr[i]:indptr[i + 1]]
+        used = set(int(colors[j]) for j in nb if colors[j] >= 0)
+        start = i % max(k, 1)
+        c = start
+        while c in used:
+            c += 1
+        colors[i] = c
+        k = max(k, c + 1)
+    return colors, int(colors.max()) + 1 if n else 0
+
+
+@register_coloring("UNIFORM")
+def _uniform(A, scope, level):
+    n = A.n_rows
+    deg = np.diff(A.row_offsets.cpu().numpy())
+    k = int(deg.max()) + 1 if n else 1
+    colors = (np.arange(n) % max(k, 1)).astype(np.int32)
+    return colors, max(k, 1) if n else 0
+
+
+@register_coloring("MIN_MAX_2RING")
+def _min_max_2ring(A, scope, level):
+    return _min_max_rounds(*_host_adj(A, 2))
+
+
+@register_coloring("GREEDY_MIN_MAX_2RING")
+def _greedy_min_max_2ring(A, scope, level):
+    indptr, indices = _host_adj(A, 2)
+    return _greedy(indptr, indices, np.arange(indptr.size - 1))
+
+
+@register_coloring("LOCALLY_DOWNWIND")
+def _locally_downwind(A, scope, level):
+    """Downwind-ordered greedy for Kaczmarz (reference locally_downwind.cu):
+    rows ordered by descending signed off-diagonal outflow so sweeps follow
+    the flow direction."""
+    indptr, indices = _host_adj(A, level)
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    v = A.values.cpu().numpy().reshape(A.nnz, -1)[:, 0]
+    n = A.n_rows
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    off = (rows != ci) & (ci < n)
+    outflow = np.zeros(n)
+    np.add.at(outflow, rows[off], -v[off])
+    order = np.argsort(-outflow, kind="stable")
+    return _greedy(indptr, indices, order)
 
 
 class MatrixColoring:
@@ -32,22 +302,27 @@ class MatrixColoring:
 
     @classmethod
     def create(cls, A, scope=None) -> "MatrixColoring":
-        frac = scope.get("max_uncolored_percentage") if scope is not None else 0.0
-        colors, num = ops.color_matrix(A, max_uncolored_frac=float(frac or 0.0))
+        scheme = (scope.get("matrix_coloring_scheme")
+                  if scope is not None else "MIN_MAX") or "MIN_MAX"
+        level = int(scope.get("coloring_level")) if scope is not None else 1
+        fn = COLORING_REGISTRY.get(scheme)
+        if fn is None:
+            raise KeyError(f"unknown coloring scheme {scheme!r}; known: "
+                           f"{sorted(COLORING_REGISTRY)}")
+        colors, num = fn(A, scope, max(level, 1))
+        if isinstance(colors, np.ndarray):
+            colors = torch.from_numpy(colors)
         return cls(colors.to(A.row_offsets.device), num)
 
     def rows_of(self, c: int) -> torch.Tensor:
         return self.rows_sorted[self.bounds[c]:self.bounds[c + 1]]
 
-    def validate(self, A) -> bool:
-        """Distance-1 validity check (reference src/tests/valid_coloring.cu)."""
-        ro = A.row_offsets.cpu().numpy()
-        ci = A.col_indices.cpu().numpy()
+    def validate(self, A, level: int = 1) -> bool:
+        """Distance-1/2 validity check (reference src/tests/valid_coloring.cu)."""
+        indptr, indices = _host_adj(A, level)
         col = self.colors.cpu().numpy()
-        n = A.n_rows
-        for i in range(n):
-            for k in range(ro[i], ro[i + 1]):
-                j = ci[k]
-                if j != i and j < n and col[j] == col[i]:
+        for i in range(indptr.size - 1):
+            for j in indices[indptr[i]:indptr[i + 1]]:
+                if col[j] == col[i]:
                     return False
         return True

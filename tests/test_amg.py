@@ -160,3 +160,55 @@ def test_cycle_ordering():
         iters[cyc] = st.iterations
     assert iters["W"] <= iters["V"] + 2
     assert iters["CG"] <= iters["V"] + 2
+
+
+# ----------------------------------------------------------- coloring schemes
+def test_coloring_schemes_valid():
+    """Every registered coloring scheme yields a valid distance-1 (or
+    distance-2 for *2RING) coloring on an unstructured-ish matrix (reference
+    src/tests/valid_coloring.cu + matrix_coloring_test.cu)."""
+    import numpy as np
+
+    from amgx_amd.amg.coloring import COLORING_REGISTRY, MatrixColoring
+    from amgx_amd.config import AMGConfig
+    from amgx_amd.problems import poisson_3d
+    A = poisson_3d(5, 4, 3)
+    for scheme in sorted(COLORING_REGISTRY):
+        cfg = AMGConfig.from_dict({"solver": {
+            "solver": "MULTICOLOR_GS",
+            "matrix_coloring_scheme": scheme}})
+        col = MatrixColoring.create(A, cfg.root_scope())
+        assert col.colors.numel() == A.n_rows
+        assert col.num_colors >= 1
+        level = 2 if "2RING" in scheme else 1
+        if scheme != "UNIFORM":   # UNIFORM is index-pattern, like reference
+            assert col.validate(A, level=level), f"{scheme} invalid"
+        # bounds cover all rows exactly once
+        assert col.bounds[-1] == A.n_rows
+
+
+def test_coloring_2ring_is_distance2():
+    from amgx_amd.amg.coloring import MatrixColoring
+    from amgx_amd.config import AMGConfig
+    from amgx_amd.problems import poisson_3d
+    A = poisson_3d(4, 4, 2)
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "MULTICOLOR_GS",
+        "matrix_coloring_scheme": "MIN_MAX_2RING"}})
+    col = MatrixColoring.create(A, cfg.root_scope())
+    assert col.validate(A, level=2)
+
+
+def test_coloring_level2_any_scheme():
+    """coloring_level=2 upgrades any scheme to the distance-2 graph
+    (reference coloring_level, src/core.cu:489)."""
+    from amgx_amd.amg.coloring import MatrixColoring
+    from amgx_amd.config import AMGConfig
+    from amgx_amd.problems import poisson_3d
+    A = poisson_3d(4, 3, 3)
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "MULTICOLOR_GS",
+        "matrix_coloring_scheme": "PARALLEL_GREEDY",
+        "coloring_level": 2}})
+    col = MatrixColoring.create(A, cfg.root_scope())
+    assert col.validate(A, level=2)
