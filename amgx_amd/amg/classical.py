@@ -1,0 +1,395 @@
+"""Classical-AMG component registries: strength metrics, C/F selectors and
+interpolators (reference src/classical/strength/, src/classical/selectors/,
+src/classical/interpolators/; factory registration src/core.cu:560-690).
+
+Host implementations are the numpy/scipy references; the hot defaults
+(AHAT + PMIS + D1) dispatch to the gfx950 kernels through the ops backend.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Dict
+
+import numpy as np
+import scipy.sparse as sp
+import torch
+
+from .. import ops
+
+STRENGTH_REGISTRY: Dict[str, Callable] = {}
+SELECTOR_REGISTRY: Dict[str, Callable] = {}
+INTERP_REGISTRY: Dict[str, Callable] = {}
+
+
+def _register(reg, name):
+    def deco(fn):
+        reg[name] = fn
+        return fn
+    return deco
+
+
+def _csr_parts(A):
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    v = A.values.cpu().numpy().astype(np.float64).reshape(A.nnz, -1)[:, 0] \
+        if A.values.numel() else np.zeros(0)
+    return ro, ci, v
+
+
+# ============================================================ strength metrics
+@_register(STRENGTH_REGISTRY, "AHAT")
+def strength_ahat(A, scope):
+    return ops._backend(A).strength_ahat(
+        A, float(scope.get("strength_threshold")),
+        float(scope.get("max_row_sum")))
+
+
+@_register(STRENGTH_REGISTRY, "ALL")
+def strength_all(A, scope):
+    """Every off-diagonal connection is strong (reference
+    src/classical/strength/all.cu)."""
+    ro, ci, _ = _csr_parts(A)
+    rows = np.repeat(np.arange(A.n_rows), np.diff(ro))
+    return torch.from_numpy((rows != ci) & (ci < A.n_rows))
+
+
+@_register(STRENGTH_REGISTRY, "AFFINITY")
+def strength_affinity(A, scope, n_vecs: int = 4, n_sweeps: int = 8):
+    """Affinity strength from smoothed random test vectors (reference
+    src/classical/strength/affinity.cu): relax A x = 0 from random starts;
+    entries whose endpoints move together are strong:
+    aff_ij = (sum_k x_ki x_kj)^2 / (sum_k x_ki^2 * sum_k x_kj^2)."""
+    ro, ci, v = _csr_parts(A)
+    n = A.n_rows
+    m = sp.csr_matrix((v, ci, ro), shape=(n, A.n_cols))[:, :n]
+    d = m.diagonal()
+    d = np.where(np.abs(d) > 0, d, 1.0)
+    rng = np.random.RandomState(20017)
+    X = rng.rand(n, n_vecs) * 2.0 - 1.0
+    for _ in range(n_sweeps):
+        X = X - 0.6 * ((m @ X) / d[:, None])
+    norms = (X * X).sum(1)
+    norms = np.where(norms > 0, norms, 1.0)
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    own = ci < n
+    dots = np.zeros(ci.size)
+    dots[own] = (X[rows[own]] * X[ci[own]]).sum(1)
+    aff = dots * dots / (norms[rows] * np.where(own, norms[np.minimum(ci, n - 1)], 1.0))
+    off = (rows != ci) & own
+    theta = float(scope.get("strength_threshold"))
+    rowmax = np.zeros(n)
+    np.maximum.at(rowmax, rows[off], aff[off])
+    strong = off & (aff >= theta * rowmax[rows]) & (rowmax[rows] > 0)
+    return torch.from_numpy(strong)
+
+
+# ================================================================== selectors
+def _strong_adj(A, S):
+    """Symmetrized strong adjacency S union S^T as scipy CSR of 0/1."""
+    ro, ci, _ = _csr_parts(A)
+    n = A.n_rows
+    strong = S.cpu().numpy() if torch.is_tensor(S) else np.asarray(S)
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    own = strong & (ci < n)
+    sr, sc = rows[own], ci[own]
+    adj = sp.csr_matrix((np.ones(2 * sr.size, dtype=np.int8),
+                         (np.concatenate([sr, sc]),
+                          np.concatenate([sc, sr]))), shape=(n, n))
+    adj.sum_duplicates()
+    adj.setdiag(0)
+    adj.eliminate_zeros()
+    return adj
+
+
+def _mis_select(adj: sp.csr_matrix, lam: np.ndarray, seed: int = 10007):
+    """Luby independent-set rounds with (lam + rand, id) total order — the
+    PMIS core over an arbitrary adjacency (reference pmis.cu rounds)."""
+    n = adj.shape[0]
+    rng = np.random.RandomState(seed)
+    w = lam + rng.rand(n)
+    state = np.zeros(n, dtype=np.int8)  # 0 undecided, 1 C, -1 F
+    deg = np.diff(adj.indptr)
+    state[deg == 0] = -1
+    indptr, indices = adj.indptr, adj.indices
+    while (state == 0).any():
+        und = state == 0
+        newC = []
+        for i in np.nonzero(und)[0]:
+            nb = indices[indptr[i]:indptr[i + 1]]
+            nb = nb[state[nb] == 0]
+            beat = False
+            for j in nb:
+                if j != i and (w[j] > w[i] or (w[j] == w[i] and j > i)):
+                    beat = True
+                    break
+            if not beat:
+                newC.append(i)
+        state[np.asarray(newC, dtype=np.int64)] = 1
+        for i in newC:
+            nb = indices[indptr[i]:indptr[i + 1]]
+            nb = nb[state[nb] == 0]
+            state[nb] = -1
+    cf = np.full(n, -1, dtype=np.int32)
+    c_rows = np.nonzero(state == 1)[0]
+    cf[c_rows] = np.arange(c_rows.size, dtype=np.int32)
+    return torch.from_numpy(cf), int(c_rows.size)
+
+
+@_register(SELECTOR_REGISTRY, "PMIS")
+def select_pmis(A, S, scope):
+    return ops._backend(A).pmis_select(A, S)
+
+
+@_register(SELECTOR_REGISTRY, "HMIS")
+def select_hmis(A, S, scope):
+    """HMIS = PMIS on the distance-2 strong graph (reference
+    src/classical/selectors/hmis.cu 'reuses PMIS on 2-ring')."""
+    adj = _strong_adj(A, S)
+    adj2 = (adj @ adj + adj).tocsr()
+    adj2.setdiag(0)
+    adj2.eliminate_zeros()
+    adj2.data[:] = 1
+    lam = np.asarray(adj.sum(0)).ravel().astype(np.float64)
+    return _mis_select(adj2, lam)
+
+
+@_register(SELECTOR_REGISTRY, "RS")
+def select_rs(A, S, scope):
+    """Classic sequential Ruge-Stueben first pass (reference
+    src/classical/selectors/rs.cu, host-only legacy): pick the max-measure
+    undecided point as C, mark its strong dependents F, increment measures of
+    their dependencies."""
+    import heapq
+    adj = _strong_adj(A, S)      # symmetric; measures from S^T counts
+    ro, ci, _ = _csr_parts(A)
+    n = A.n_rows
+    strong = S.cpu().numpy()
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    own = strong & (ci < n)
+    lam = np.zeros(n)
+    np.add.at(lam, ci[own], 1.0)       # points i influences
+    Smat = sp.csr_matrix((np.ones(own.sum(), dtype=np.int8),
+                          (rows[own], ci[own])), shape=(n, n))
+    St = Smat.tocsc()
+    st_rows = [Smat.indices[Smat.indptr[i]:Smat.indptr[i + 1]]
+               for i in range(n)]                               # S_i
+    infl = [St.indices[St.indptr[i]:St.indptr[i + 1]]
+            for i in range(n)]                                  # S^T_i
+    state = np.zeros(n, dtype=np.int8)
+    heap = [(-lam[i], i) for i in range(n)]
+    heapq.heapify(heap)
+    version = lam.copy()
+    while heap:
+        negm, i = heapq.heappop(heap)
+        if state[i] != 0 or -negm != version[i]:
+            continue
+        state[i] = 1
+        for j in infl[i]:
+            if state[j] == 0:
+                state[j] = -1
+                for k in st_rows[j]:
+                    if state[k] == 0:
+                        version[k] += 1
+                        heapq.heappush(heap, (-version[k], k))
+    cf = np.full(n, -1, dtype=np.int32)
+    c_rows = np.nonzero(state == 1)[0]
+    cf[c_rows] = np.arange(c_rows.size, dtype=np.int32)
+    return torch.from_numpy(cf), int(c_rows.size)
+
+
+@_register(SELECTOR_REGISTRY, "CR")
+def select_cr(A, S, scope, nu: int = 5, rounds: int = 4, tol: float = 0.7):
+    """Compatible-relaxation selector (reference src/classical/selectors/
+    cr.cu): rows where F-point relaxation of the homogeneous system converges
+    slowly are promoted to C, a max-independent subset at a time."""
+    ro, ci, v = _csr_parts(A)
+    n = A.n_rows
+    m = sp.csr_matrix((v, ci, ro), shape=(n, A.n_cols))[:, :n].tocsr()
+    d = m.diagonal()
+    d = np.where(np.abs(d) > 0, d, 1.0)
+    adj = _strong_adj(A, S)
+    state = np.full(n, -1, dtype=np.int8)   # all F initially
+    rng = np.random.RandomState(4242)
+    for _ in range(rounds):
+        e = rng.rand(n)
+        e[state == 1] = 0.0
+        for _ in range(nu):
+            e = e - 0.8 * ((m @ e) / d)
+            e[state == 1] = 0.0
+        slow = (np.abs(e) > tol * np.abs(e).max()) & (state == -1) \
+            if np.abs(e).max() > 0 else np.zeros(n, dtype=bool)
+        if not slow.any():
+            break
+        # independent subset of the slow candidates
+        cand = np.nonzero(slow)[0]
+        sub = adj[cand][:, cand].tocsr()
+        lam = np.asarray(adj.sum(0)).ravel()[cand].astype(np.float64)
+        cf_sub, _nc = _mis_select(sub, lam)
+        state[cand[cf_sub.numpy() >= 0]] = 1
+    if (state == 1).sum() == 0:
+        return SELECTOR_REGISTRY["PMIS"](A, S, scope)
+    cf = np.full(n, -1, dtype=np.int32)
+    c_rows = np.nonzero(state == 1)[0]
+    cf[c_rows] = np.arange(c_rows.size, dtype=np.int32)
+    return torch.from_numpy(cf), int(c_rows.size)
+
+
+def aggressive_select(A, S, scope, base: str):
+    """Two-pass aggressive coarsening (reference aggressive_pmis.cu /
+    aggressive_hmis.cu + aggressive_levels, src/core.cu:461): run the base
+    selector, then re-select among the C points over strong paths of length
+    <= 2, keeping only the surviving subset as C."""
+    cf1, nc1 = SELECTOR_REGISTRY[base](A, S, scope)
+    cf1np = cf1.numpy() if torch.is_tensor(cf1) else cf1
+    c_idx = np.nonzero(cf1np >= 0)[0]
+    if c_idx.size <= 1:
+        return cf1, nc1
+    adj = _strong_adj(A, S)
+    adj2 = (adj @ adj + adj).tocsr()
+    sub = adj2[c_idx][:, c_idx].tocsr()
+    sub.setdiag(0)
+    sub.eliminate_zeros()
+    lam = np.asarray(adj.sum(0)).ravel()[c_idx].astype(np.float64)
+    cf_sub, nc2 = _mis_select(sub, lam)
+    cf = np.full(A.n_rows, -1, dtype=np.int32)
+    keep = c_idx[cf_sub.numpy() >= 0]
+    cf[keep] = np.arange(keep.size, dtype=np.int32)
+    return torch.from_numpy(cf), int(keep.size)
+
+
+SELECTOR_REGISTRY["AGGRESSIVE_PMIS"] = \
+    lambda A, S, scope: aggressive_select(A, S, scope, "PMIS")
+SELECTOR_REGISTRY["AGGRESSIVE_HMIS"] = \
+    lambda A, S, scope: aggressive_select(A, S, scope, "HMIS")
+
+
+# ============================================================== interpolators
+@_register(INTERP_REGISTRY, "D1")
+def interp_d1(A, S, cf_map, num_coarse, scope):
+    return ops._backend(A).interp_d1(A, S, cf_map, num_coarse)
+
+
+@_register(INTERP_REGISTRY, "D2")
+def interp_d2(A, S, cf_map, num_coarse, scope):
+    """Distance-2 (standard/extended) interpolation (reference
+    src/classical/interpolators/distance2.cu): an F point i interpolates
+    from its strong C neighbors AND the strong C neighbors of its strong F
+    neighbors, with a_ij distributed over j's C points proportionally to
+    a_jk (classical standard-interpolation formula)."""
+    from ..matrix import CSRMatrix
+    ro, ci, v = _csr_parts(A)
+    strong = S.cpu().numpy()
+    cf = cf_map.cpu().numpy().astype(np.int64) if torch.is_tensor(cf_map) \
+        else np.asarray(cf_map, dtype=np.int64)
+    n = A.n_rows
+    Pc, Pv, indptr = [], [], [0]
+    for i in range(n):
+        if cf[i] >= 0:
+            Pc.append(cf[i])
+            Pv.append(1.0)
+            indptr.append(indptr[-1] + 1)
+            continue
+        s, e = ro[i], ro[i + 1]
+        diag = 0.0
+        acc = {}           # coarse id -> accumulated coupling
+        weak_sum = 0.0
+        for k in range(s, e):
+            j, a = ci[k], v[k]
+            if j == i:
+                diag = a
+                continue
+            if not strong[k]:
+                weak_sum += a
+                continue
+            if j < cf.size and cf[j] >= 0:
+                acc[cf[j]] = acc.get(cf[j], 0.0) + a
+            elif j < n:
+                # strong F neighbor: distribute a_ij over j's strong C points
+                js, je = ro[j], ro[j + 1]
+                cpts, cvals = [], []
+                for kk in range(js, je):
+                    jj = ci[kk]
+                    if strong[kk] and jj < cf.size and cf[jj] >= 0:
+                        cpts.append(cf[jj])
+                        cvals.append(v[kk])
+                tot = sum(cvals)
+                if cpts and tot != 0.0:
+                    for cc, av in zip(cpts, cvals):
+                        acc[cc] = acc.get(cc, 0.0) + a * av / tot
+                else:
+                    weak_sum += a     # dead-end F neighbor: lump
+            else:
+                weak_sum += a
+        denom = diag + weak_sum
+        if not acc or denom == 0.0:
+            indptr.append(indptr[-1])
+            continue
+        for cc in sorted(acc):
+            Pc.append(cc)
+            Pv.append(-acc[cc] / denom)
+        indptr.append(indptr[-1] + len(acc))
+    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
+                       np.asarray(indptr)), shape=(n, num_coarse))
+    return CSRMatrix.from_scipy(P, dtype=A.dtype)
+
+
+@_register(INTERP_REGISTRY, "MULTIPASS")
+def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
+    """Multipass interpolation for aggressive coarsening (reference
+    src/classical/interpolators/multipass.cu): pass 0 = C points (identity);
+    pass 1 = F points with strong C neighbors (direct D1 weights); pass k>1 =
+    F points whose strong neighbors were interpolated in earlier passes,
+    composing their rows."""
+    from ..matrix import CSRMatrix
+    ro, ci, v = _csr_parts(A)
+    strong = S.cpu().numpy()
+    cf = cf_map.cpu().numpy().astype(np.int64) if torch.is_tensor(cf_map) \
+        else np.asarray(cf_map, dtype=np.int64)
+    n = A.n_rows
+    rows_P: list = [None] * n           # dict coarse->weight per row
+    done = np.zeros(n, dtype=bool)
+    for i in range(n):
+        if cf[i] >= 0:
+            rows_P[i] = {int(cf[i]): 1.0}
+            done[i] = True
+    for _pass in range(max_passes):
+        progressed = False
+        newly = []
+        for i in range(n):
+            if done[i]:
+                continue
+            s, e = ro[i], ro[i + 1]
+            diag = 0.0
+            acc: Dict[int, float] = {}
+            lump = 0.0
+            ok = False
+            for k in range(s, e):
+                j, a = ci[k], v[k]
+                if j == i:
+                    diag = a
+                    continue
+                if strong[k] and j < n and done[j]:
+                    for cc, wjc in rows_P[j].items():
+                        acc[cc] = acc.get(cc, 0.0) + a * wjc
+                    ok = True
+                else:
+                    lump += a if not strong[k] else 0.0
+            if ok and (diag + lump) != 0.0:
+                newly.append((i, {cc: -aw / (diag + lump)
+                                  for cc, aw in acc.items()}))
+        for i, row in newly:
+            rows_P[i] = row
+            done[i] = True
+            progressed = True
+        if not progressed:
+            break
+    Pc, Pv, indptr = [], [], [0]
+    for i in range(n):
+        row = rows_P[i] or {}
+        for cc in sorted(row):
+            Pc.append(cc)
+            Pv.append(row[cc])
+        indptr.append(indptr[-1] + len(row))
+    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
+                       np.asarray(indptr)), shape=(n, num_coarse))
+    return CSRMatrix.from_scipy(P, dtype=A.dtype)

@@ -177,10 +177,20 @@ class ClassicalLevel(AMGLevel):
             self.num_coarse = nc
             self.A._cache["cf_map"] = self.cf_map
             return nc
-        S = ops._backend(self.A).strength_ahat(
-            self.A, float(self.scope.get("strength_threshold")),
-            float(self.scope.get("max_row_sum")))
-        cf, nc = ops._backend(self.A).pmis_select(self.A, S)
+        from .classical import SELECTOR_REGISTRY, STRENGTH_REGISTRY
+        strength = self.scope.get("strength") or "AHAT"
+        S = STRENGTH_REGISTRY[strength](self.A, self.scope)
+        sel = self.scope.get("selector") or "PMIS"
+        if sel in ("SIZE_2", "SIZE_4", "SIZE_8"):  # aggregation default leaks
+            sel = "PMIS"
+        if self.index < int(self.scope.get("aggressive_levels") or 0):
+            agg_sel = self.scope.get("aggressive_selector")
+            sel = ("AGGRESSIVE_" + sel) if agg_sel in (None, "DEFAULT") \
+                else agg_sel
+            self._aggressive = True
+        else:
+            self._aggressive = False
+        cf, nc = SELECTOR_REGISTRY[sel](self.A, S, self.scope)
         self.S = S
         self.cf_map = cf
         self.num_coarse = nc
@@ -191,8 +201,12 @@ class ClassicalLevel(AMGLevel):
         mgr = getattr(self.A, "manager", None)
         if mgr is not None:
             return self._create_coarse_matrix_distributed(mgr)
-        self.P = ops._backend(self.A).interp_d1(self.A, self.S, self.cf_map,
-                                                self.num_coarse)
+        from .classical import INTERP_REGISTRY
+        interp = self.scope.get("interpolator") or "D1"
+        if getattr(self, "_aggressive", False):
+            interp = self.scope.get("aggressive_interpolator") or "MULTIPASS"
+        self.P = INTERP_REGISTRY[interp](self.A, self.S, self.cf_map,
+                                         self.num_coarse, self.scope)
         tf = float(self.scope.get("interp_truncation_factor"))
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:
