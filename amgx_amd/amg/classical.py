@@ -393,3 +393,52 @@ def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
     P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
                        np.asarray(indptr)), shape=(n, num_coarse))
     return CSRMatrix.from_scipy(P, dtype=A.dtype)
+
+
+@_register(INTERP_REGISTRY, "EM")
+def interp_em(A, S, cf_map, num_coarse, scope):
+    """Energy-minimization interpolation (reference src/energymin/
+    interpolators/em.cu: per-patch dense solves): each F row i minimizes the
+    A-energy of the extended basis function over its strong C patch —
+    w0 = A_CC^{-1} A_Ci (local harmonic extension) — then enforces the
+    constant constraint sum(w)=1 with a Lagrange correction
+    w = w0 + A_CC^{-1}1 (1 - 1^T w0)/(1^T A_CC^{-1} 1)."""
+    from ..matrix import CSRMatrix
+    ro, ci, v = _csr_parts(A)
+    strong = S.cpu().numpy()
+    cf = cf_map.cpu().numpy().astype(np.int64) if torch.is_tensor(cf_map) \
+        else np.asarray(cf_map, dtype=np.int64)
+    n = A.n_rows
+    m = sp.csr_matrix((v, ci, ro), shape=(n, A.n_cols)).tocsr()
+    Pc, Pv, indptr = [], [], [0]
+    for i in range(n):
+        if cf[i] >= 0:
+            Pc.append(cf[i]); Pv.append(1.0)
+            indptr.append(indptr[-1] + 1)
+            continue
+        s, e = ro[i], ro[i + 1]
+        patch = [int(ci[k]) for k in range(s, e)
+                 if strong[k] and ci[k] < cf.size and cf[ci[k]] >= 0]
+        if not patch:
+            indptr.append(indptr[-1])
+            continue
+        Acc = m[patch][:, patch].toarray()
+        aci = np.asarray(m[patch][:, [i]].todense()).ravel()
+        try:
+            w0 = np.linalg.solve(Acc, -aci)
+            z = np.linalg.solve(Acc, np.ones(len(patch)))
+        except np.linalg.LinAlgError:
+            w0, *_ = np.linalg.lstsq(Acc, -aci, rcond=None)
+            z, *_ = np.linalg.lstsq(Acc, np.ones(len(patch)), rcond=None)
+        denom = float(np.ones(len(patch)) @ z)
+        if denom != 0.0:
+            w = w0 + z * (1.0 - float(np.ones(len(patch)) @ w0)) / denom
+        else:
+            w = w0
+        order = np.argsort(np.asarray([cf[p] for p in patch]))
+        for o in order:
+            Pc.append(cf[patch[o]]); Pv.append(float(w[o]))
+        indptr.append(indptr[-1] + len(patch))
+    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
+                       np.asarray(indptr)), shape=(n, num_coarse))
+    return CSRMatrix.from_scipy(P, dtype=A.dtype)

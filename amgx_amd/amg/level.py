@@ -64,23 +64,21 @@ class AggregationLevel(AMGLevel):
         self.num_aggregates = 0
 
     def create_coarse_vertices(self) -> int:
-        selector = self.scope.get("selector")
-        passes = {"SIZE_2": 1, "SIZE_4": 2, "SIZE_8": 3,
-                  "MULTI_PAIRWISE": 2, "DUMMY": 1}.get(selector, 1)
+        from .aggregation import AGG_SELECTOR_REGISTRY
+        selector = self.scope.get("selector") or "SIZE_2"
         mgr = getattr(self.A, "manager", None)
         if mgr is not None:
-            passes = 1   # distributed multi-pass selectors: later round
-        maxit = self.scope.get("max_matching_iterations")
-        agg, num = ops.size2_matching(self.A, max_iterations=maxit)
-        work = self.A
-        for _ in range(passes - 1):
-            if num <= self.scope.get("min_coarse_rows"):
-                break
-            work = ops.galerkin_aggregation(work, agg.to(work.row_offsets.device),
-                                            num)
-            agg2, num2 = ops.size2_matching(work, max_iterations=maxit)
-            agg = agg2.to(agg.device)[agg.long()]
-            num = num2
+            # distributed: rank-local pairwise matching (halo columns are
+            # excluded from matching; remote coupling enters via the coarse
+            # level's own halo). Multi-pass selectors collapse to one pass.
+            maxit = self.scope.get("max_matching_iterations")
+            agg, num = ops.size2_matching(self.A, max_iterations=maxit)
+        else:
+            fn = AGG_SELECTOR_REGISTRY.get(selector)
+            if fn is None:
+                raise KeyError(f"unknown aggregation selector {selector!r}; "
+                               f"known: {sorted(AGG_SELECTOR_REGISTRY)}")
+            agg, num = fn(self.A, self.scope)
         self.aggregates = agg.to(self.A.row_offsets.device)
         self.num_aggregates = num
         self._build_r_structure()
@@ -101,8 +99,20 @@ class AggregationLevel(AMGLevel):
     def create_coarse_matrix(self) -> CSRMatrix:
         mgr = getattr(self.A, "manager", None)
         if mgr is None:
-            return ops.galerkin_aggregation(self.A, self.aggregates,
-                                            self.num_aggregates)
+            Ac = ops.galerkin_aggregation(self.A, self.aggregates,
+                                          self.num_aggregates)
+            geom = self.A._cache.get("geometry")
+            if geom is not None:
+                # aggregate centroids keep GEO usable on coarse levels
+                # (reference geo_selector propagates coordinates)
+                g = np.asarray(geom, dtype=np.float64)
+                agg = self.aggregates.cpu().numpy().astype(np.int64)
+                sums = np.zeros((self.num_aggregates, g.shape[1]))
+                cnts = np.zeros(self.num_aggregates)
+                np.add.at(sums, agg, g)
+                np.add.at(cnts, agg, 1.0)
+                Ac._cache["geometry"] = sums / np.maximum(cnts, 1.0)[:, None]
+            return Ac
         return self._create_coarse_matrix_distributed(mgr)
 
     def _create_coarse_matrix_distributed(self, mgr) -> CSRMatrix:
@@ -157,6 +167,9 @@ class ClassicalLevel(AMGLevel):
     src/classical/classical_amg_level.cu): AHAT strength -> PMIS C/F split ->
     distance-1 (direct) interpolation -> R = P^T -> RAP Galerkin product."""
 
+    default_selector = "PMIS"
+    default_interp = "D1"
+
     def __init__(self, A, scope, index):
         super().__init__(A, scope, index)
         self.P = None
@@ -180,9 +193,10 @@ class ClassicalLevel(AMGLevel):
         from .classical import SELECTOR_REGISTRY, STRENGTH_REGISTRY
         strength = self.scope.get("strength") or "AHAT"
         S = STRENGTH_REGISTRY[strength](self.A, self.scope)
-        sel = self.scope.get("selector") or "PMIS"
-        if sel in ("SIZE_2", "SIZE_4", "SIZE_8"):  # aggregation default leaks
-            sel = "PMIS"
+        sel = self.scope.get("selector") if self.scope.has("selector") \
+            else self.default_selector
+        if sel in (None, "SIZE_2", "SIZE_4", "SIZE_8"):  # aggregation default
+            sel = self.default_selector
         if self.index < int(self.scope.get("aggressive_levels") or 0):
             agg_sel = self.scope.get("aggressive_selector")
             sel = ("AGGRESSIVE_" + sel) if agg_sel in (None, "DEFAULT") \
@@ -202,7 +216,8 @@ class ClassicalLevel(AMGLevel):
         if mgr is not None:
             return self._create_coarse_matrix_distributed(mgr)
         from .classical import INTERP_REGISTRY
-        interp = self.scope.get("interpolator") or "D1"
+        interp = self.scope.get("interpolator") \
+            if self.scope.has("interpolator") else self.default_interp
         if getattr(self, "_aggressive", False):
             interp = self.scope.get("aggressive_interpolator") or "MULTIPASS"
         self.P = INTERP_REGISTRY[interp](self.A, self.S, self.cf_map,
@@ -268,9 +283,20 @@ class ClassicalLevel(AMGLevel):
         ops.axpy(x, tmp, 1.0)
 
 
+class EnergyminLevel(ClassicalLevel):
+    """Energy-minimization AMG level (reference
+    src/energymin/energymin_amg_level.cu): compatible-relaxation C/F
+    selection + per-patch energy-minimizing interpolation (EM)."""
+
+    default_selector = "CR"
+    default_interp = "EM"
+
+
 def create_level(algorithm: str, A, scope, index) -> AMGLevel:
     if algorithm == "AGGREGATION":
         return AggregationLevel(A, scope, index)
-    if algorithm in ("CLASSICAL", "ENERGYMIN"):
+    if algorithm == "CLASSICAL":
         return ClassicalLevel(A, scope, index)
+    if algorithm == "ENERGYMIN":
+        return EnergyminLevel(A, scope, index)
     raise KeyError(f"unknown AMG algorithm {algorithm!r}")

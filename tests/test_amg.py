@@ -321,3 +321,50 @@ def test_d2_interp_reproduces_constant():
     Pv = P.to_scipy() @ ones_c
     assert np.allclose(Pv, 1.0, atol=1e-10), \
         f"max dev {np.abs(Pv - 1).max()}"
+
+
+# -------------------------------------------------- aggregation selectors
+def test_aggregation_selectors():
+    """Every aggregation selector yields a valid aggregate map and a
+    convergent FGMRES+AGG hierarchy (reference aggregates_*.cu tests)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.amg.aggregation import AGG_SELECTOR_REGISTRY
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    A = poisson_3d(8, 8, 8)
+    A._cache["geometry"] = __import__("numpy").stack(
+        __import__("numpy").meshgrid(range(8), range(8), range(8),
+                                     indexing="ij"), -1).reshape(-1, 3)
+    for sel in sorted(AGG_SELECTOR_REGISTRY):
+        cfg = AMGConfig.from_dict({"solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "AGGREGATION",
+                "smoother": "BLOCK_JACOBI", "presweeps": 1, "postsweeps": 1,
+                "max_iters": 1, "min_coarse_rows": 16, "cycle": "V",
+                "selector": sel, "aggregate_size": 4,
+            },
+            "solver": "FGMRES", "max_iters": 120, "gmres_n_restart": 20,
+            "monitor_residual": 1, "convergence": "RELATIVE_INI",
+            "tolerance": 1e-6,
+        }})
+        scope = cfg.root_scope().sub_solver("preconditioner", "AMG")[1]
+        agg, num = AGG_SELECTOR_REGISTRY[sel](A, scope)
+        assert agg.numel() == A.n_rows
+        assert 0 < num < A.n_rows, f"{sel}: num={num}"
+        assert int(agg.max()) == num - 1 and int(agg.min()) == 0
+        s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+        assert st.converged and rel < 1e-5, f"{sel}: {st} rel={rel}"
+
+
+def test_energymin_level():
+    """ENERGYMIN algorithm (CR selector + EM interpolation) solves Poisson
+    (reference src/tests/energymin_algorithm.cu)."""
+    st, rel = _solve_classical(_classical_cfg(algorithm="ENERGYMIN"), n=8)
+    assert st.converged and rel < 1e-7, f"{st}, rel={rel}"
