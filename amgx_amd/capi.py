@@ -550,10 +550,12 @@ def AMGX_solver_get_iteration_residual(s: _SolverHandle, it: int = -1,
 @_amgx_try
 def AMGX_read_system(m: _MatrixHandle, rhs: _VectorHandle,
                      sol: _VectorHandle, path: str):
+    from .io.binary import is_binary_file, read_system_binary
     from .io.matrix_market import read_system
     mem, vecprec, matprec = _parse_mode(m.mode)
     dev = mem if mem == "cpu" else m.res.device
-    A, b, x0 = read_system(path, device=dev, dtype=matprec)
+    reader = read_system_binary if is_binary_file(path) else read_system
+    A, b, x0 = reader(path, device=dev, dtype=matprec)
     m.A = A
     n = A.n_rows
     if rhs is not None:
@@ -571,9 +573,13 @@ def AMGX_read_system(m: _MatrixHandle, rhs: _VectorHandle,
 def AMGX_write_system(m: _MatrixHandle, rhs: _VectorHandle,
                       sol: _VectorHandle, path: str):
     from .io.matrix_market import write_system
-    write_system(path, m.A,
-                 rhs.v if rhs is not None else None,
-                 sol.v if sol is not None else None)
+    b = rhs.v if rhs is not None else None
+    x = sol.v if sol is not None else None
+    if path.endswith((".bin", ".amgxb")):     # binary writer (reference
+        from .io.binary import write_system_binary   # matrix_writer=binary)
+        write_system_binary(path, m.A, b, x)
+        return RC_OK
+    write_system(path, m.A, b, x)
     return RC_OK
 
 
@@ -632,3 +638,406 @@ def AMGX_unpin_memory(ptr):
 
 def AMGX_SOLVE_SUCCESS():
     return 0
+
+
+# -------------------------------------------------- remaining API surface
+_ERROR_STRINGS = {
+    RC_OK: "No error",
+    RC_BAD_PARAMETERS: "Incorrect parameters",
+    RC_UNKNOWN: "Unknown error",
+    RC_NOT_SUPPORTED_TARGET: "Unsupported target",
+    RC_NOT_SUPPORTED_BLOCKSIZE: "Unsupported block size",
+    RC_CUDA_FAILURE: "HIP/GPU failure",
+    RC_IO_ERROR: "I/O error",
+    RC_BAD_MODE: "Invalid mode",
+    RC_CORE: "Core library error",
+    RC_PLUGIN: "Plugin error",
+    RC_BAD_CONFIGURATION: "Invalid configuration",
+    RC_NOT_IMPLEMENTED: "Not implemented",
+    RC_LICENSE_NOT_FOUND: "License not found",
+    RC_INTERNAL: "Internal error",
+}
+
+
+def AMGX_get_error_string(rc: int):
+    """Reference AMGX_get_error_string (include/amgx_c.h)."""
+    return _ERROR_STRINGS.get(rc, "Unknown error code")
+
+
+@_amgx_try
+def AMGX_config_create_from_file_and_string(path: str, options: str):
+    rc, h = AMGX_config_create_from_file(path)
+    assert rc == RC_OK
+    rc = AMGX_config_add_parameters(h, options)
+    assert rc == RC_OK
+    return RC_OK, h
+
+
+@_amgx_try
+def AMGX_distribution_set_32bit_colindices(dist: _DistributionHandle,
+                                           use32: bool):
+    dist.colindices_32bit = bool(use32)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_get_nnz(m: _MatrixHandle):
+    return RC_OK, m.A.nnz
+
+
+@_amgx_try
+def AMGX_matrix_attach_geometry(m: _MatrixHandle, geox, geoy=None, geoz=None,
+                                n=None, dimension=None):
+    """Reference AMGX_matrix_attach_geometry: coordinates for the GEO
+    selector. Accepts either one (n,dim) array or per-axis arrays."""
+    if geoy is None and geoz is None:
+        coords = np.asarray(geox, dtype=np.float64)
+    else:
+        axes = [np.asarray(a, dtype=np.float64)
+                for a in (geox, geoy, geoz) if a is not None]
+        coords = np.stack(axes, axis=-1)
+    m.A._cache["geometry"] = coords
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_attach_coloring(m: _MatrixHandle, row_coloring, num_rows=None,
+                                num_colors=None):
+    """Reference AMGX_matrix_attach_coloring: user-provided row coloring used
+    by multicolor smoothers instead of a computed one."""
+    from .amg.coloring import MatrixColoring
+    colors = torch.as_tensor(np.asarray(row_coloring), dtype=torch.int32)
+    num = int(num_colors) if num_colors is not None \
+        else int(colors.max()) + 1
+    m.A._cache["coloring"] = MatrixColoring(
+        colors.to(m.A.row_offsets.device), num)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_set_boundary_separation(m: _MatrixHandle, flag: int):
+    m.boundary_separation = int(flag)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_upload_all_global_32(m, n_global, n, nnz, block_dimx,
+                                     block_dimy, row_ptrs,
+                                     col_indices_global, data,
+                                     diag_data=None, allocated_halo_depth=1,
+                                     num_import_rings=1,
+                                     partition_vector=None):
+    """32-bit global column variant (reference amgx_c.h
+    AMGX_matrix_upload_all_global_32); indices are widened internally."""
+    cols = np.asarray(col_indices_global, dtype=np.int64)
+    return AMGX_matrix_upload_all_global(
+        m, n_global, n, nnz, block_dimx, block_dimy, row_ptrs, cols, data,
+        diag_data, allocated_halo_depth, num_import_rings, partition_vector)
+
+
+@_amgx_try
+def AMGX_matrix_vector_multiply(m: _MatrixHandle, x: _VectorHandle,
+                                y: _VectorHandle):
+    """y = A x (reference AMGX_matrix_vector_multiply)."""
+    if y.v is None or y.v.numel() != x.v.numel():
+        y.v = torch.zeros_like(x.v)
+        y.n = x.n
+        y.block_dim = x.block_dim
+    ops.spmv(m.A, x.v, y.v)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_comm_from_maps_one_ring(m: _MatrixHandle, allocated_halo_depth,
+                                        num_neighbors, neighbors,
+                                        send_sizes, send_maps,
+                                        recv_sizes, recv_maps):
+    """Build the distributed halo structure from user-provided B2L maps
+    (reference AMGX_matrix_comm_from_maps_one_ring, src/amgx_c.cu ->
+    cacheMaps + updateMapsReorder). The local matrix must already index halo
+    columns as n_local + position, where positions follow the concatenated
+    recv maps."""
+    import torch.distributed as tdist
+    from .distributed.manager import DistributedManager
+    A = m.A
+    assert A is not None, "upload the local matrix first"
+    dev = A.row_offsets.device
+    mgr = DistributedManager(dev, A.block_dim)
+    mgr.n_local = A.n_rows
+    n_halo = int(sum(int(s) for s in recv_sizes))
+    mgr.n_halo = n_halo
+    if tdist.is_initialized():
+        counts = [None] * mgr.world
+        tdist.all_gather_object(counts, A.n_rows)
+        mgr.row_start = int(sum(counts[:mgr.rank]))
+        mgr.n_global = int(sum(counts))
+    else:
+        mgr.row_start, mgr.n_global = 0, A.n_rows
+    pos = 0
+    for k in range(int(num_neighbors)):
+        nb = int(neighbors[k])
+        mgr.neighbors.append(nb)
+        smap = np.asarray(send_maps[k], dtype=np.int64)
+        mgr.b2l.append(torch.from_numpy(smap.astype(np.int32)).to(dev))
+        sz = int(recv_sizes[k])
+        rmap = np.asarray(recv_maps[k], dtype=np.int64)
+        # our layout needs per-neighbor contiguous halo slices in map order
+        assert (rmap == np.arange(pos, pos + sz) + mgr.n_local).all() or \
+               (rmap == np.arange(pos, pos + sz)).all(), \
+            "recv maps must be consecutive halo positions"
+        mgr.halo_slices.append((pos, pos + sz))
+        pos += sz
+    mgr.halo_global = np.arange(n_halo)   # opaque (maps-based upload)
+    # rows are NOT renumbered interior-first here, so no row is provably
+    # halo-free: treat every row as boundary (computed after the exchange)
+    mgr.boundary_start = 0
+    mgr.row_perm = torch.arange(A.n_rows, dtype=torch.int64, device=dev)
+    mgr.row_iperm = mgr.row_perm.clone()
+    mgr._alloc_send_bufs(A.dtype)
+    A.manager = mgr
+    A.n_cols = A.n_rows + n_halo
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_matrix_comm_from_maps(m, allocated_halo_depth, num_import_rings,
+                               num_neighbors, neighbors, send_sizes,
+                               send_maps, recv_sizes, recv_maps):
+    assert int(num_import_rings) == 1, "only one-ring maps supported"
+    return AMGX_matrix_comm_from_maps_one_ring(
+        m, allocated_halo_depth, num_neighbors, neighbors, send_sizes,
+        send_maps, recv_sizes, recv_maps)
+
+
+@_amgx_try
+def AMGX_read_system_maps_one_ring(m: _MatrixHandle):
+    """Return the halo maps of a distributed matrix (reference
+    AMGX_read_system_maps_one_ring out-params)."""
+    mgr = getattr(m.A, "manager", None)
+    assert mgr is not None, "matrix is not distributed"
+    neighbors = list(mgr.neighbors)
+    send_maps = [b.cpu().numpy().copy() for b in mgr.b2l]
+    send_sizes = [int(b.numel()) for b in mgr.b2l]
+    recv_maps = [np.arange(lo, hi) + mgr.n_local
+                 for (lo, hi) in mgr.halo_slices]
+    recv_sizes = [int(hi - lo) for (lo, hi) in mgr.halo_slices]
+    return (RC_OK, len(neighbors), neighbors, send_sizes, send_maps,
+            recv_sizes, recv_maps)
+
+
+@_amgx_try
+def AMGX_free_system_maps_one_ring(*args):
+    return RC_OK   # Python GC owns the arrays
+
+
+@_amgx_try
+def AMGX_solver_calculate_residual_norm(s: _SolverHandle, m: _MatrixHandle,
+                                        rhs: _VectorHandle,
+                                        sol: _VectorHandle):
+    r = ops.residual(m.A, sol.v, rhs.v)
+    mgr = getattr(m.A, "manager", None)
+    if mgr is not None:
+        nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+            r.reshape(-1)[:mgr.owned_size])), "L2")
+    else:
+        nrm = float(torch.linalg.vector_norm(r))
+    return RC_OK, nrm
+
+
+@_amgx_try
+def AMGX_solver_register_print_callback(s: _SolverHandle, cb):
+    s.print_callback = cb
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_read_system_distributed(m: _MatrixHandle, rhs: _VectorHandle,
+                                 sol: _VectorHandle, path: str,
+                                 allocated_halo_depth=1, num_partitions=None,
+                                 partition_sizes=None, partition_vector=None):
+    """Rank-partitioned read (reference src/distributed/distributed_io.cu
+    DistributedRead): every rank reads the file and keeps its contiguous row
+    slab; halo structure is rebuilt from global column ids."""
+    import torch.distributed as tdist
+    from .distributed.manager import DistributedManager
+    from .io.matrix_market import read_system
+    from .io.binary import is_binary_file, read_system_binary
+    mem, vecprec, matprec = _parse_mode(m.mode)
+    dev = mem if mem == "cpu" else m.res.device
+    reader = read_system_binary if is_binary_file(path) else read_system
+    A, b, x0 = reader(path, device="cpu", dtype=matprec)
+    world = tdist.get_world_size() if tdist.is_initialized() else 1
+    rank = tdist.get_rank() if tdist.is_initialized() else 0
+    n = A.n_rows
+    if partition_vector is not None:
+        pv = np.asarray(partition_vector)
+        sizes = [int((pv == r).sum()) for r in range(world)]
+    elif partition_sizes is not None:
+        sizes = [int(s) for s in partition_sizes]
+    else:
+        base = n // world
+        sizes = [base + (1 if r < n % world else 0) for r in range(world)]
+    offs = np.zeros(world + 1, dtype=np.int64)
+    offs[1:] = np.cumsum(sizes)
+    lo, hi = int(offs[rank]), int(offs[rank + 1])
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    va = A.values.cpu().numpy()
+    s0, s1 = ro[lo], ro[hi]
+    local_ro = ro[lo:hi + 1] - s0
+    m.A = DistributedManager.upload_global_csr(
+        local_ro, ci[s0:s1], va[s0:s1], hi - lo, lo, n,
+        device=dev, block_dim=A.block_dim, dtype=matprec)
+    mgr = m.A.manager
+    if rhs is not None:
+        rhs.v = mgr.new_ext_vec(vecprec)
+        src = b if b is not None else torch.ones(n, dtype=vecprec)
+        rhs.v[:mgr.owned_size] = mgr.permute_in(
+            src[lo * A.block_dim:hi * A.block_dim].to(vecprec).to(dev)
+        )[:mgr.owned_size]
+        rhs.n = mgr.n_local
+        rhs.bound = m
+    if sol is not None:
+        sol.v = mgr.new_ext_vec(vecprec)
+        if x0 is not None:
+            sol.v[:mgr.owned_size] = mgr.permute_in(
+                x0[lo * A.block_dim:hi * A.block_dim].to(vecprec).to(dev)
+            )[:mgr.owned_size]
+        sol.n = mgr.n_local
+        sol.bound = m
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_read_system_global(m: _MatrixHandle, rhs: _VectorHandle,
+                            sol: _VectorHandle, path: str):
+    return AMGX_read_system_distributed(m, rhs, sol, path)
+
+
+@_amgx_try
+def AMGX_write_system_distributed(m: _MatrixHandle, rhs: _VectorHandle,
+                                  sol: _VectorHandle, path: str,
+                                  allocated_halo_depth=1, num_partitions=None,
+                                  partition_sizes=None,
+                                  partition_vector=None):
+    """Gather the distributed system to rank 0 and write one file (reference
+    AMGX_write_system_distributed consolidates partitions)."""
+    import torch.distributed as tdist
+    from .io.matrix_market import write_system
+    from .io.binary import write_system_binary
+    mgr = getattr(m.A, "manager", None)
+    if mgr is None:
+        return AMGX_write_system(m, rhs, sol, path)
+    A = m.A
+    perm = mgr.row_perm.cpu().numpy()          # internal new -> old local
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    va = A.values.cpu().numpy()
+    # back to user (old-local) row order with GLOBAL columns
+    iold = np.argsort(perm, kind="stable")      # old -> new position
+    halo_map = mgr.halo_global
+    rows_out, cols_out, vals_out = [], [], []
+    for old in range(mgr.n_local):
+        new = int(iold[old])
+        s, e = ro[new], ro[new + 1]
+        cols = ci[s:e].copy()
+        own = cols < mgr.n_local
+        gcols = np.empty_like(cols)
+        gcols[own] = perm[cols[own]] + mgr.row_start
+        gcols[~own] = halo_map[cols[~own] - mgr.n_local]
+        order = np.argsort(gcols)
+        rows_out.append(np.full(e - s, old + mgr.row_start, dtype=np.int64))
+        cols_out.append(gcols[order])
+        vals_out.append(va[s:e][order])
+    payload = (np.concatenate(rows_out) if rows_out else np.zeros(0),
+               np.concatenate(cols_out) if cols_out else np.zeros(0),
+               np.concatenate(vals_out) if vals_out else np.zeros(0),
+               mgr.permute_out(rhs.v).cpu().numpy() if rhs is not None
+               and rhs.v is not None else None,
+               mgr.permute_out(sol.v).cpu().numpy() if sol is not None
+               and sol.v is not None else None)
+    gathered = [None] * mgr.world
+    tdist.all_gather_object(gathered, payload)
+    if mgr.rank == 0:
+        import scipy.sparse as sp
+        rows = np.concatenate([g[0] for g in gathered])
+        cols = np.concatenate([g[1] for g in gathered])
+        vals = np.concatenate([g[2] for g in gathered], axis=0)
+        n = mgr.n_global
+        if A.block_dim == 1:
+            sm = sp.csr_matrix((vals, (rows, cols)), shape=(n, n))
+            Afull = CSRMatrix.from_scipy(sm, dtype=A.dtype)
+        else:
+            order = np.argsort(rows * n + cols, kind="stable")
+            ro_f = np.zeros(n + 1, dtype=np.int64)
+            np.add.at(ro_f[1:], rows, 1)
+            np.cumsum(ro_f, out=ro_f)
+            Afull = CSRMatrix(
+                torch.from_numpy(ro_f.astype(np.int32)),
+                torch.from_numpy(cols[order].astype(np.int32)),
+                torch.from_numpy(vals[order]).to(A.dtype),
+                n_cols=n, block_dim=A.block_dim)
+        bfull = (np.concatenate([g[3] for g in gathered])
+                 if gathered[0][3] is not None else None)
+        xfull = (np.concatenate([g[4] for g in gathered])
+                 if gathered[0][4] is not None else None)
+        bt = torch.from_numpy(bfull) if bfull is not None else None
+        xt = torch.from_numpy(xfull) if xfull is not None else None
+        if path.endswith((".bin", ".amgxb")):
+            write_system_binary(path, Afull, bt, xt)
+        else:
+            write_system(path, Afull, bt, xt)
+    if tdist.is_initialized():
+        tdist.barrier()
+    return RC_OK
+
+
+# ------------------------------------------------------------- eigensolvers
+class _EigenSolverHandle:
+    def __init__(self, res: Resources, mode: str, cfg: AMGConfig):
+        self.res = res
+        self.mode = mode
+        self.cfg = cfg
+        self.solver = None
+        self.status = None
+
+
+@_amgx_try
+def AMGX_eigensolver_create(res: _ResourcesHandle, mode: str,
+                            cfg: _ConfigHandle):
+    """Reference AMGX_eigensolver_create (include/amgx_eig_c.h:16)."""
+    _parse_mode(mode)
+    return RC_OK, _EigenSolverHandle(res.res, mode, cfg.cfg)
+
+
+@_amgx_try
+def AMGX_eigensolver_setup(s: _EigenSolverHandle, m: _MatrixHandle):
+    from .eigensolvers import create_eigensolver
+    s.solver = create_eigensolver(s.cfg.root_scope(), resources=s.res)
+    s.solver.setup(m.A)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_eigensolver_pagerank_setup(s: _EigenSolverHandle, m: _MatrixHandle):
+    from .eigensolvers import create_eigensolver
+    s.solver = create_eigensolver(s.cfg.root_scope(), resources=s.res)
+    s.solver.pagerank_setup(m.A)
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_eigensolver_solve(s: _EigenSolverHandle, x0: _VectorHandle = None):
+    st = s.solver.solve(x0.v if x0 is not None else None)
+    s.status = st
+    if x0 is not None and st.eigenvector is not None:
+        v = st.eigenvector
+        x0.v = v[:, 0] if v.dim() > 1 else v
+        x0.n = int(x0.v.numel())
+    return RC_OK
+
+
+@_amgx_try
+def AMGX_eigensolver_destroy(s: _EigenSolverHandle):
+    s.solver = None
+    return RC_OK

@@ -128,3 +128,131 @@ def test_capi_symmetry_checks():
     assert struct_sym and sym
     rc, dd = capi.AMGX_matrix_check_diag_dominant(A)
     assert dd
+
+
+# ------------------------------------------------- binary IO + API additions
+def test_binary_io_roundtrip(tmp_path):
+    """Binary system format round-trips matrix+rhs+sol (reference
+    ReadNVAMGBinary, src/readers.cu:1676)."""
+    import numpy as np
+    import torch
+
+    from amgx_amd.io.binary import (is_binary_file, read_system_binary,
+                                    write_system_binary)
+    from amgx_amd.problems import poisson_3d
+    A = poisson_3d(4, 4, 4)
+    b = torch.rand(A.n_rows, dtype=torch.float64)
+    x = torch.rand(A.n_rows, dtype=torch.float64)
+    p = str(tmp_path / "sys.bin")
+    write_system_binary(p, A, b, x)
+    assert is_binary_file(p)
+    A2, b2, x2 = read_system_binary(p)
+    assert torch.equal(A2.row_offsets, A.row_offsets)
+    assert torch.equal(A2.col_indices, A.col_indices)
+    assert torch.allclose(A2.values, A.values)
+    assert torch.allclose(b2, b) and torch.allclose(x2, x)
+
+
+def test_capi_read_write_binary(tmp_path):
+    import torch
+
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create("config_version=2, solver=CG, max_iters=40,"
+                                   " tolerance=1e-8, convergence=RELATIVE_INI,"
+                                   " monitor_residual=1")
+    assert rc == C.RC_OK
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    rc, rhs = C.AMGX_vector_create(res, "hDDI")
+    rc, sol = C.AMGX_vector_create(res, "hDDI")
+    rc = C.AMGX_generate_distributed_poisson_7pt(m, rhs, sol, 1, 1, 5, 5, 5)
+    assert rc == C.RC_OK
+    p = str(tmp_path / "sys.bin")
+    rc = C.AMGX_write_system(m, rhs, sol, p)
+    assert rc == C.RC_OK
+    rc, m2 = C.AMGX_matrix_create(res, "hDDI")
+    rc, rhs2 = C.AMGX_vector_create(res, "hDDI")
+    rc, sol2 = C.AMGX_vector_create(res, "hDDI")
+    rc = C.AMGX_read_system(m2, rhs2, sol2, p)
+    assert rc == C.RC_OK
+    assert m2.A.n_rows == 125
+    rc, nnz = C.AMGX_matrix_get_nnz(m2)
+    assert rc == C.RC_OK and nnz == m.A.nnz
+    # solve from the reloaded system
+    rc, s = C.AMGX_solver_create(res, "hDDI", cfg)
+    assert C.AMGX_solver_setup(s, m2) == C.RC_OK
+    assert C.AMGX_solver_solve(s, rhs2, sol2) == C.RC_OK
+    rc, nrm = C.AMGX_solver_calculate_residual_norm(s, m2, rhs2, sol2)
+    assert rc == C.RC_OK and nrm < 1e-6 * 125
+
+
+def test_capi_matrix_vector_multiply():
+    import numpy as np
+    import torch
+
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create("config_version=2, solver=CG")
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    rc, x = C.AMGX_vector_create(res, "hDDI")
+    rc, y = C.AMGX_vector_create(res, "hDDI")
+    n = 4
+    ro = [0, 1, 2, 3, 4]
+    ci = [0, 1, 2, 3]
+    va = [2.0, 3.0, 4.0, 5.0]
+    assert C.AMGX_matrix_upload_all(m, n, 4, 1, 1, ro, ci, va) == C.RC_OK
+    C.AMGX_vector_upload(x, n, 1, [1.0, 1.0, 1.0, 1.0])
+    assert C.AMGX_matrix_vector_multiply(m, x, y) == C.RC_OK
+    assert np.allclose(y.v.numpy(), [2, 3, 4, 5])
+
+
+def test_capi_attach_geometry_and_coloring():
+    import numpy as np
+
+    from amgx_amd import capi as C
+    from amgx_amd.problems import poisson_3d
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create("config_version=2, solver=CG")
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    A = poisson_3d(3, 3, 3)
+    m.A = A
+    g = np.random.rand(27, 3)
+    assert C.AMGX_matrix_attach_geometry(m, g) == C.RC_OK
+    assert "geometry" in A._cache
+    colors = np.arange(27) % 2
+    # invalid colorings are accepted like the reference (user responsibility)
+    assert C.AMGX_matrix_attach_coloring(m, colors) == C.RC_OK
+    assert A._cache["coloring"].num_colors == 2
+    assert C.AMGX_get_error_string(C.RC_OK) == "No error"
+    assert "configuration" in C.AMGX_get_error_string(
+        C.RC_BAD_CONFIGURATION).lower()
+
+
+def test_capi_eigensolver():
+    """AMGX_eigensolver_* flow (reference include/amgx_eig_c.h:16-26)."""
+    import torch
+
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create(
+        "config_version=2, eig_solver=POWER_ITERATION, eig_max_iters=200,"
+        " eig_tolerance=1e-8")
+    assert rc == C.RC_OK
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    n = 3
+    ro = [0, 1, 2, 3]
+    ci = [0, 1, 2]
+    va = [3.0, 2.0, 1.0]
+    assert C.AMGX_matrix_upload_all(m, n, 3, 1, 1, ro, ci, va) == C.RC_OK
+    rc, es = C.AMGX_eigensolver_create(res, "hDDI", cfg)
+    assert rc == C.RC_OK
+    assert C.AMGX_eigensolver_setup(es, m) == C.RC_OK
+    rc, v0 = C.AMGX_vector_create(res, "hDDI")
+    C.AMGX_vector_upload(v0, n, 1, [1.0, 1.0, 1.0])
+    assert C.AMGX_eigensolver_solve(es, v0) == C.RC_OK
+    assert abs(es.status.eigenvalues[-1] - 3.0) < 1e-6
+    assert C.AMGX_eigensolver_destroy(es) == C.RC_OK
