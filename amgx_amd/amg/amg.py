@@ -36,6 +36,9 @@ class AMGHierarchy:
         self.cycle_iters = scope.get("cycle_iters")
         self.print_grid_stats = bool(scope.get("print_grid_stats"))
         self.setup_time = 0.0
+        from ..utils.profiler import PhaseProfiler
+        self.profiler = PhaseProfiler(
+            enabled=bool(scope.get("obtain_timings")))
 
     # ------------------------------------------------------------------ setup
     def setup(self, A):
@@ -116,20 +119,29 @@ class AMGHierarchy:
     def _cycle(self, li: int, b, x, zero_guess: bool, ctype: str):
         """Reference FixedCycle::cycle (src/cycles/fixed_cycle.cu:59-230)."""
         level = self.levels[li]
+        prof = self.profiler
         if li == len(self.levels) - 1:
+            prof.tic("coarseSolve")
             if self.coarse_solver is not None:
                 self.coarse_solver.solve(b, x, zero_initial_guess=True)
             else:
                 if zero_guess:
                     x.zero_()
                 level.smoother.sweep(b, x, self.coarsest_sweeps)
+            prof.toc("coarseSolve")
             return
         if zero_guess:
             x.zero_()
         if self.presweeps > 0:
+            prof.tic("Smoother")
             level.smoother.sweep(b, x, self.presweeps)
+            prof.toc("Smoother")
+        prof.tic("computeResidual")
         ops.residual(level.A, x, b, level.r)
+        prof.toc("computeResidual")
+        prof.tic("restrictResidual")
         level.restrict_residual(level.r, level.bc)
+        prof.toc("restrictResidual")
         if ctype in ("CG", "CGF"):
             # K-cycle (reference src/cycles/cg_cycle.cu + cg_flex_cycle.cu):
             # accelerate the coarse correction with cycle_iters FCG steps
@@ -146,9 +158,13 @@ class AMGHierarchy:
             for rep in range(repeats):
                 sub = "V" if (ctype == "F" and rep > 0) else ctype
                 self._cycle(li + 1, level.bc, level.xc, rep == 0, sub)
+        prof.tic("prolongate")
         level.prolongate_and_apply(level.xc, x)
+        prof.toc("prolongate")
         if self.postsweeps > 0:
+            prof.tic("Smoother")
             level.smoother.sweep(b, x, self.postsweeps)
+            prof.toc("Smoother")
 
     def _kcycle_coarse(self, li: int, b, x):
         """Notay K-cycle coarse correction (reference src/cycles/cg_cycle.cu):

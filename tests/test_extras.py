@@ -76,3 +76,72 @@ def test_cf_jacobi_in_classical_amg():
     s.setup(A)
     st = s.solve(b, x, zero_initial_guess=True)
     assert st.converged and st.iterations <= 40
+
+
+# ----------------------------------------------------------- aux subsystems
+def test_phase_profiler_and_timings():
+    """obtain_timings=1 accumulates per-phase timers in the AMG cycle
+    (reference per-level Profile.tic/toc, src/cycles/fixed_cycle.cu)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "AMG", "algorithm": "AGGREGATION",
+        "smoother": "BLOCK_JACOBI", "presweeps": 1, "postsweeps": 1,
+        "max_iters": 4, "min_coarse_rows": 8, "cycle": "V",
+        "obtain_timings": 1, "monitor_residual": 1, "tolerance": 1e-10,
+    }})
+    A = poisson_3d(6, 6, 6)
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    s.solve(b, x, zero_initial_guess=True)
+    prof = s.hierarchy.profiler
+    assert prof.enabled
+    for phase in ("Smoother", "restrictResidual", "prolongate",
+                  "coarseSolve"):
+        assert phase in prof.acc and prof.acc[phase][0] > 0, phase
+    rep = prof.report()
+    assert "Smoother" in rep and "avg_ms" in rep
+
+
+def test_determinism_checker():
+    """Two identical AMG setups produce identical checkpoint hashes
+    (reference determinism_checker.cu + determinism_flag)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    from amgx_amd.utils import DeterminismChecker
+
+    def run():
+        cfg = AMGConfig.from_dict({"solver": {
+            "solver": "AMG", "algorithm": "AGGREGATION",
+            "smoother": "BLOCK_JACOBI", "max_iters": 2,
+            "min_coarse_rows": 8, "cycle": "V", "monitor_residual": 1,
+        }})
+        A = poisson_3d(5, 5, 5)
+        s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+        s.setup(A)
+        chk = DeterminismChecker()
+        for i, lvl in enumerate(s.hierarchy.levels):
+            chk.checkpoint(f"level{i}.A", lvl.A.row_offsets,
+                           lvl.A.col_indices, lvl.A.values)
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.solve(b, x, zero_initial_guess=True)
+        chk.checkpoint("x", x)
+        return chk
+
+    c1, c2 = run(), run()
+    assert c1.same_as(c2), c1.diff(c2)
+
+
+def test_memory_info():
+    from amgx_amd.utils import MemoryInfo
+    u = MemoryInfo.get_max_memory_usage()
+    assert u["host_peak_mib"] > 1.0
