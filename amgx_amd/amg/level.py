@@ -205,6 +205,10 @@ class ClassicalLevel(AMGLevel):
         else:
             self._aggressive = False
         cf, nc = SELECTOR_REGISTRY[sel](self.A, S, self.scope)
+        # host-computed metrics/selectors (ALL/AFFINITY/HMIS/RS/CR) return
+        # CPU tensors: normalize to the matrix device for the kernel path
+        dev = self.A.row_offsets.device
+        S, cf = S.to(dev), cf.to(dev)
         self.S = S
         self.cf_map = cf
         self.num_coarse = nc
@@ -222,6 +226,8 @@ class ClassicalLevel(AMGLevel):
             interp = self.scope.get("aggressive_interpolator") or "MULTIPASS"
         self.P = INTERP_REGISTRY[interp](self.A, self.S, self.cf_map,
                                          self.num_coarse, self.scope)
+        if self.P.row_offsets.device != self.A.row_offsets.device:
+            self.P = self.P.to(self.A.row_offsets.device)  # host interps
         tf = float(self.scope.get("interp_truncation_factor"))
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:
