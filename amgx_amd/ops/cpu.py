@@ -153,13 +153,16 @@ def jacobi_dinv(A, l1: bool = False) -> torch.Tensor:
 
 def jacobi_smooth(A, dinv, b, x_in, x_out, omega: float):
     r = residual(A, x_in, b)
-    if A.block_dim == 1:
-        upd = dinv.reshape(-1) * r.reshape(-1)
+    bd = A.block_dim
+    n_owned = A.n_rows * bd      # vectors may be halo-extended (distributed)
+    if bd == 1:
+        upd = dinv.reshape(-1) * r.reshape(-1)[:n_owned]
     else:
-        bd = A.block_dim
-        rb = r.reshape(-1, bd, 1).to(dinv.dtype)
+        rb = r.reshape(-1)[:n_owned].reshape(-1, bd, 1).to(dinv.dtype)
         upd = torch.bmm(dinv, rb).reshape(-1).to(x_in.dtype)
-    x_out.reshape(-1).copy_(x_in.reshape(-1) + omega * upd)
+    xo = x_out.reshape(-1)
+    xo[:n_owned] = x_in.reshape(-1)[:n_owned] + omega * upd
+    xo[n_owned:] = x_in.reshape(-1)[n_owned:]
     return x_out
 
 
