@@ -472,7 +472,9 @@ def dilu_setup(A, coloring):
             E = D[i].copy()
             for k in range(ro[i], ro[i + 1]):
                 j = ci[k]
-                if j != i and colors[j] < c and AT[k] >= 0:
+                # halo columns (j >= n_rows) carry no local Einv: DILU is
+                # rank-local, like the reference's per-partition smoother
+                if j != i and j < A.n_rows and colors[j] < c and AT[k] >= 0:
                     E -= vals[k] @ einv[j] @ vals[AT[k]]
             if abs(np.linalg.det(E)) < 1e-300:
                 E = np.eye(A.block_dim)

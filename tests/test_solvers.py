@@ -144,3 +144,74 @@ def test_dense_lu_direct():
     s.solve(b, x)
     r = ops.residual(A, x, b)
     assert ops.nrm2(r) < 1e-10
+
+
+def test_block4_dilu_smoother_and_fgmres():
+    """BASELINE config #4 shape: block-4 coupled system, multicolor DILU
+    (reference multicolor_dilu_solver.cu NxN block path)."""
+    A = block_laplacian(8, 8, block_dim=4)
+    s = make({"solver": "MULTICOLOR_DILU", "max_iters": 8,
+              "relaxation_factor": 1.0})
+    n = A.n_rows * A.block_dim
+    b = torch.ones(n, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    r0 = ops.nrm2(ops.residual(A, x, b))
+    s.solve(b, x)
+    r1 = ops.nrm2(ops.residual(A, x, b))
+    assert r1 < 0.5 * r0, f"{r1} !< 0.5*{r0}"
+    # outer FGMRES + DILU preconditioner on the block system
+    s2 = make({"solver": "FGMRES", "max_iters": 200, "gmres_n_restart": 30,
+               "monitor_residual": 1, "tolerance": 1e-8,
+               "convergence": "RELATIVE_INI",
+               "preconditioner": {"solver": "MULTICOLOR_DILU",
+                                  "max_iters": 2,
+                                  "relaxation_factor": 1.0}})
+    x2 = torch.zeros_like(b)
+    s2.setup(A)
+    st = s2.solve(b, x2, zero_initial_guess=True)
+    rel = float(ops.nrm2(ops.residual(A, x2, b)) / ops.nrm2(b))
+    assert st.converged and rel < 1e-7
+
+
+def test_block_sizes_2_to_5():
+    """Block sizes 2..5 through block-Jacobi+PCG (reference
+    smoother_blocksizes.cu covers 2-10)."""
+    for bd in (2, 3, 4, 5):
+        A = block_laplacian(6, 6, block_dim=bd)
+        s = make({"solver": "PCG", "preconditioner": "BLOCK_JACOBI",
+                  "max_iters": 600, "monitor_residual": 1,
+                  "tolerance": 1e-8, "convergence": "RELATIVE_INI"})
+        n = A.n_rows * bd
+        b = torch.ones(n, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+        assert st.converged and rel < 1e-6, f"bd={bd}: {st} rel={rel}"
+
+
+def test_mixed_precision_mode():
+    """dDFI-analog: float32 matrix x float64 vectors (reference mixed mode
+    dDFI, include/amgx_config.h:79-120)."""
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create(
+        "config_version=2, solver=PCG, preconditioner=BLOCK_JACOBI,"
+        " max_iters=400, tolerance=1e-6, convergence=RELATIVE_INI,"
+        " monitor_residual=1")
+    assert rc == C.RC_OK
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDFI")
+    rc, b = C.AMGX_vector_create(res, "hDFI")
+    rc, x = C.AMGX_vector_create(res, "hDFI")
+    assert C.AMGX_generate_distributed_poisson_7pt(
+        m, b, x, 1, 1, 8, 8, 8) == C.RC_OK
+    assert m.A.values.dtype == torch.float32      # matrix precision F
+    assert b.v.dtype == torch.float64             # vector precision D
+    rc, s = C.AMGX_solver_create(res, "hDFI", cfg)
+    assert C.AMGX_solver_setup(s, m) == C.RC_OK
+    assert C.AMGX_solver_solve(s, b, x) == C.RC_OK
+    assert s.status.converged
+    rc, nrm = C.AMGX_solver_calculate_residual_norm(s, m, b, x)
+    assert nrm < 1e-3
