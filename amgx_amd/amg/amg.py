@@ -81,6 +81,61 @@ class AMGHierarchy:
         if self.print_grid_stats and self.res.rank == 0:
             print(self.grid_stats())
 
+    def resetup(self, A, reuse_levels: int):
+        """Values-only re-setup: keep the coarsening structure (aggregates /
+        C-F splits) of the first ``reuse_levels`` levels (<0 = all) and
+        rebuild only Galerkin values, smoother factors and the coarse solver
+        (reference structure_reuse_levels)."""
+        t0 = time.perf_counter()
+        if reuse_levels < 0:
+            reuse_levels = len(self.levels)
+        self.levels[0].A = A
+        A._cache.pop("cf_map", None)
+        for i, lvl in enumerate(self.levels[:-1]):
+            if i >= reuse_levels:
+                # from here down, recoarsen from scratch
+                self._truncate_and_recoarsen(i)
+                break
+            Ac = lvl.rebuild_coarse_values()
+            if Ac is None:          # level cannot reuse -> full recoarsen
+                self._truncate_and_recoarsen(i)
+                break
+            self.levels[i + 1].A = Ac
+        for lvl in self.levels[:-1]:
+            if lvl.smoother is None:
+                lvl.smoother = self._make_smoother()
+            lvl.smoother.setup(lvl.A)
+        self._setup_coarse_solver()
+        if self.res.is_cuda:
+            torch.cuda.synchronize()
+        self.setup_time = time.perf_counter() - t0
+
+    def _truncate_and_recoarsen(self, i: int):
+        """Drop levels below i and continue the setup loop from level i."""
+        level = self.levels[i]
+        self.levels = self.levels[:i + 1]
+        level.next = None
+        dist = getattr(level.A, "manager", None)
+        while True:
+            A_l = level.A
+            n_global = dist.global_rows(A_l.n_rows) if dist is not None \
+                else A_l.n_rows
+            if (len(self.levels) >= self.max_levels
+                    or n_global <= max(self.min_coarse_rows, 2)):
+                break
+            nc = level.create_coarse_vertices()
+            nc_global = dist.global_rows(nc) if dist is not None else nc
+            if nc_global >= n_global * self.coarsen_threshold \
+                    or nc_global in (0, n_global):
+                break
+            Ac = level.create_coarse_matrix()
+            level.alloc_coarse_vectors(Ac)
+            nxt = create_level(self.algorithm, Ac, self.scope,
+                               len(self.levels))
+            level.next = nxt
+            self.levels.append(nxt)
+            level = nxt
+
     def _make_smoother(self):
         name, sub = self.scope.sub_solver("smoother", "BLOCK_JACOBI")
         # smoother inherits the amg scope's relaxation_factor unless it sets one

@@ -45,6 +45,11 @@ class AMGLevel:
     def prolongate_and_apply(self, xc, x):
         raise NotImplementedError
 
+    def rebuild_coarse_values(self):
+        """Recompute the coarse operator's VALUES on the cached structure
+        (structure_reuse_levels); None = this level cannot reuse."""
+        return None
+
     def alloc_coarse_vectors(self, Ac: CSRMatrix):
         mgr = getattr(Ac, "manager", None)
         nc = mgr.ext_size if mgr is not None else Ac.n_rows * Ac.block_dim
@@ -152,6 +157,13 @@ class AggregationLevel(AMGLevel):
         self._build_r_structure()
         return Ac
 
+    def rebuild_coarse_values(self):
+        if self.aggregates is None \
+                or getattr(self.A, "manager", None) is not None:
+            return None
+        return ops.galerkin_aggregation(self.A, self.aggregates,
+                                        self.num_aggregates)
+
     def restrict_residual(self, r, bc):
         out = ops.restrict_agg(r, self.aggregates, self.num_aggregates,
                                self.A.block_dim,
@@ -235,6 +247,20 @@ class ClassicalLevel(AMGLevel):
         self.R = ops.transpose(self.P)
         Ac = ops.galerkin_rap(self.R, self.A, self.P)
         return Ac
+
+    def rebuild_coarse_values(self):
+        """Classical reuse: keep the C/F split, recompute strength-dependent
+        interpolation values and RAP (reference: selector output is the
+        reused structure)."""
+        if self.cf_map is None \
+                or getattr(self.A, "manager", None) is not None:
+            return None
+        from .classical import STRENGTH_REGISTRY
+        strength = self.scope.get("strength") or "AHAT"
+        self.S = STRENGTH_REGISTRY[strength](self.A, self.scope) \
+            .to(self.A.row_offsets.device)
+        self.A._cache["cf_map"] = self.cf_map
+        return self.create_coarse_matrix()
 
     def _create_coarse_matrix_distributed(self, mgr) -> CSRMatrix:
         """Distributed classical coarsening (reference

@@ -34,8 +34,17 @@ class AMGSolver(Solver):
 
     def solver_setup(self):
         from ..amg.amg import AMGHierarchy
-        self.hierarchy = AMGHierarchy(self.scope, self.res)
-        self.hierarchy.setup(self.A)
+        reuse = int(self.scope.get("structure_reuse_levels") or 0)
+        old = getattr(self, "hierarchy", None)
+        if (reuse != 0 and old is not None and old.levels
+                and old.levels[0].A.n_rows == self.A.n_rows
+                and old.levels[0].A.nnz == self.A.nnz):
+            # values-only re-setup on the cached structure (reference
+            # structure_reuse_levels, src/core.cu:437 + amg.cu reuse path)
+            old.resetup(self.A, reuse)
+        else:
+            self.hierarchy = AMGHierarchy(self.scope, self.res)
+            self.hierarchy.setup(self.A)
         self._graph = None
         self._graph_failed = False
 

@@ -27,7 +27,10 @@ class PCGSolver(Solver):
         self.flexible = bool(scope.get("pcg_flexible"))
 
     def solver_setup(self):
-        self.precond = self.make_preconditioner()
+        # keep the preconditioner object across resetup so its hierarchy can
+        # honor structure_reuse_levels (reference AMGX_solver_resetup)
+        if getattr(self, "precond", None) is None:
+            self.precond = self.make_preconditioner()
         if self.precond is not None:
             self.precond.setup(self.A)
 
@@ -90,7 +93,10 @@ class BiCGStabSolver(Solver):
     """(Preconditioned) BiCGStab (reference src/solvers/pbicgstab_solver.cu)."""
 
     def solver_setup(self):
-        self.precond = self.make_preconditioner()
+        # keep the preconditioner object across resetup so its hierarchy can
+        # honor structure_reuse_levels (reference AMGX_solver_resetup)
+        if getattr(self, "precond", None) is None:
+            self.precond = self.make_preconditioner()
         if self.precond is not None:
             self.precond.setup(self.A)
 
@@ -161,7 +167,10 @@ class FGMRESSolver(Solver):
         self.restart = scope.get("gmres_n_restart")
 
     def solver_setup(self):
-        self.precond = self.make_preconditioner()
+        # keep the preconditioner object across resetup so its hierarchy can
+        # honor structure_reuse_levels (reference AMGX_solver_resetup)
+        if getattr(self, "precond", None) is None:
+            self.precond = self.make_preconditioner()
         if self.precond is not None:
             self.precond.setup(self.A)
 
@@ -261,7 +270,10 @@ class IDRSolver(Solver):
         self.s = scope.get("subspace_dim_s")
 
     def solver_setup(self):
-        self.precond = self.make_preconditioner()
+        # keep the preconditioner object across resetup so its hierarchy can
+        # honor structure_reuse_levels (reference AMGX_solver_resetup)
+        if getattr(self, "precond", None) is None:
+            self.precond = self.make_preconditioner()
         if self.precond is not None:
             self.precond.setup(self.A)
 

@@ -368,3 +368,75 @@ def test_energymin_level():
     (reference src/tests/energymin_algorithm.cu)."""
     st, rel = _solve_classical(_classical_cfg(algorithm="ENERGYMIN"), n=8)
     assert st.converged and rel < 1e-7, f"{st}, rel={rel}"
+
+
+# --------------------------------------------------------- structure reuse
+def test_structure_reuse_resetup():
+    """structure_reuse_levels: resetup after a value change keeps the
+    aggregates and rebuilds only Galerkin values (reference
+    amg_levels_reuse.cu + AMGX_solver_resetup)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.matrix import CSRMatrix
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {
+            "solver": "AMG", "algorithm": "AGGREGATION",
+            "smoother": "BLOCK_JACOBI", "presweeps": 1, "postsweeps": 1,
+            "max_iters": 1, "min_coarse_rows": 16, "cycle": "V",
+            "structure_reuse_levels": -1, "scope": "amg",
+        },
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+    }})
+    A = poisson_3d(8, 8, 8)
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    h1 = s.precond.hierarchy
+    aggs1 = [l.aggregates for l in h1.levels[:-1]]
+    st1 = s.solve(b, x, zero_initial_guess=True)
+    assert st1.converged
+    # scale the values (same structure), resetup, solve again
+    A2 = CSRMatrix(A.row_offsets, A.col_indices, A.values * 2.0,
+                   n_cols=A.n_cols)
+    s.resetup(A2)
+    h2 = s.precond.hierarchy
+    assert h2 is h1, "hierarchy object must be reused"
+    for l, a1 in zip(h2.levels[:-1], aggs1):
+        assert l.aggregates is a1, "aggregates must be reused"
+    x2 = torch.zeros_like(b)
+    st2 = s.solve(b, x2, zero_initial_guess=True)
+    assert st2.converged
+    rel = float(ops.nrm2(ops.residual(A2, x2, b)) / ops.nrm2(b))
+    assert rel < 1e-7
+    # solution of 2A x = b is x/2
+    assert torch.allclose(x2, x / 2.0, atol=1e-6)
+
+
+def test_classical_structure_reuse():
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.matrix import CSRMatrix
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    cfg = _classical_cfg(structure_reuse_levels=-1)
+    A = poisson_3d(8, 8, 8)
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    h1 = s.precond.hierarchy
+    cf1 = [l.cf_map for l in h1.levels[:-1]]
+    A2 = CSRMatrix(A.row_offsets, A.col_indices, A.values * 3.0,
+                   n_cols=A.n_cols)
+    s.resetup(A2)
+    assert s.precond.hierarchy is h1
+    for l, c in zip(h1.levels[:-1], cf1):
+        assert l.cf_map is c
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged
