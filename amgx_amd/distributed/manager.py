@@ -475,3 +475,36 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
             out[lo + k] = (rcol[pos:pos + c], rval[pos:pos + c])
             pos += c
     return out
+
+
+def halo_matrix(mgr: DistributedManager, A):
+    """Fetch the matrix rows backing this rank's halo columns (reference
+    DistributedManager::createOneRingHaloRows, distributed_manager.cu:
+    1542-1596 — the num_import_rings=2 structure): returns a CSR fragment
+    (row_offsets, col_global, values) with one row per halo slot, columns as
+    GLOBAL ids. The columns of these rows are the 2-ring; feeding them back
+    through another exchange extends the ring again."""
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    va = A.values.cpu().numpy().reshape(A.nnz, -1)
+    # local -> global column map for the shared rows
+    own_gid = mgr.row_perm.cpu().numpy() + mgr.row_start
+    gcol = np.empty(A.n_cols, dtype=np.int64)
+    gcol[:mgr.n_local] = own_gid
+    if mgr.n_halo:
+        gcol[mgr.n_local:] = mgr.halo_global
+    rows = exchange_csr_rows(mgr, ro, gcol[ci],
+                             va[:, 0] if va.shape[1] == 1 else va[:, 0])
+    out_ro = np.zeros(mgr.n_halo + 1, dtype=np.int64)
+    cols_l, vals_l = [], []
+    for k, rowdat in enumerate(rows):
+        if rowdat is None:
+            out_ro[k + 1] = out_ro[k]
+            continue
+        cols, vals = rowdat
+        out_ro[k + 1] = out_ro[k] + cols.size
+        cols_l.append(cols)
+        vals_l.append(vals)
+    cols = np.concatenate(cols_l) if cols_l else np.zeros(0, dtype=np.int64)
+    vals = np.concatenate(vals_l) if vals_l else np.zeros(0)
+    return out_ro, cols, vals

@@ -87,7 +87,8 @@ class CSRMatrix:
     @classmethod
     def from_scipy(cls, m, device="cpu", dtype=torch.float64) -> "CSRMatrix":
         m = m.tocsr()
-        m.sort_indices()
+        m.sum_duplicates()     # canonicalize: kron/add chains leave duplicate
+        m.sort_indices()       # (i,j) entries that break per-entry ops
         ro = torch.from_numpy(np.ascontiguousarray(m.indptr, dtype=np.int32))
         ci = torch.from_numpy(np.ascontiguousarray(m.indices, dtype=np.int32))
         v = torch.from_numpy(np.ascontiguousarray(m.data)).to(dtype)

@@ -23,8 +23,10 @@ def poisson_2d(nx: int, ny: int, stencil: int = 5, device="cpu",
         D = sp.diags([-ex[:-1], np.zeros(nx), -ex[:-1]], [-1, 0, 1])
         Dy = sp.diags([-ey[:-1], np.zeros(ny), -ey[:-1]], [-1, 0, 1])
         A = A + 0.5 * sp.kron(Dy, D)
-        A = A + sp.identity(A.shape[0]) * 0  # keep structure
-    return CSRMatrix.from_scipy(A.tocsr(), device=device, dtype=dtype)
+    A = A.tocsr()
+    A.sum_duplicates()
+    A.eliminate_zeros()     # kron chains leave explicit zeros off-stencil
+    return CSRMatrix.from_scipy(A, device=device, dtype=dtype)
 
 
 def poisson_3d(nx: int, ny: int, nz: int, device="cpu",
@@ -37,8 +39,10 @@ def poisson_3d(nx: int, ny: int, nz: int, device="cpu",
     Ix, Iy, Iz = sp.identity(nx), sp.identity(ny), sp.identity(nz)
     A = (sp.kron(sp.kron(Iz, Iy), lap1(nx))
          + sp.kron(sp.kron(Iz, lap1(ny)), Ix)
-         + sp.kron(sp.kron(lap1(nz), Iy), Ix))
-    return CSRMatrix.from_scipy(A.tocsr(), device=device, dtype=dtype)
+         + sp.kron(sp.kron(lap1(nz), Iy), Ix)).tocsr()
+    A.sum_duplicates()
+    A.eliminate_zeros()     # kron chains leave explicit zeros off-stencil
+    return CSRMatrix.from_scipy(A, device=device, dtype=dtype)
 
 
 def poisson_3d_local(nx: int, ny: int, nz: int, rank: int, world: int):

@@ -422,3 +422,34 @@ def _impl_test_dist_block4_dilu(rank, world, tmp):
         st2 = s2.solve(bs, xs, zero_initial_guess=True)
         assert st2.converged
         assert abs(st.iterations - st2.iterations) <= 8
+
+
+def test_dist_halo_matrix_two_ring():
+    _run_dist(test_dist_halo_matrix_two_ring)
+
+
+def _impl_test_dist_halo_matrix_two_ring(rank, world, tmp):
+    """halo_matrix returns exactly the owner's rows for my halo columns
+    (reference createOneRingHaloRows / num_import_rings=2)."""
+    import numpy as np
+
+    from amgx_amd.distributed.manager import halo_matrix
+    from amgx_amd.problems import poisson_3d
+    n = 4
+    A = _make_dist_A(rank, world, n)
+    mgr = A.manager
+    ro_h, cols_h, vals_h = halo_matrix(mgr, A)
+    # global reference matrix
+    Afull = poisson_3d(n, n, n * world).to_scipy().tocsr()
+    for k in range(mgr.n_halo):
+        g = int(mgr.halo_global[k])
+        got_cols = np.sort(cols_h[ro_h[k]:ro_h[k + 1]])
+        ref_cols = np.sort(Afull.indices[Afull.indptr[g]:Afull.indptr[g + 1]])
+        assert (got_cols == ref_cols).all(), f"halo row {g}"
+        got = dict(zip(cols_h[ro_h[k]:ro_h[k + 1]],
+                       vals_h[ro_h[k]:ro_h[k + 1]]))
+        for j, v in zip(ref_cols,
+                        Afull.data[Afull.indptr[g]:Afull.indptr[g + 1]][
+                            np.argsort(Afull.indices[
+                                Afull.indptr[g]:Afull.indptr[g + 1]])]):
+            assert abs(got[int(j)] - v) < 1e-14
