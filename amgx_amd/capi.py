@@ -79,8 +79,10 @@ def _parse_mode(mode: str):
     if not isinstance(mode, str) or len(mode) != 4:
         raise ValueError(f"bad mode {mode!r}")
     mem = {"d": "cuda", "h": "cpu"}[mode[0]]
-    vec = {"D": torch.float64, "F": torch.float32}[mode[1]]
-    mat = {"D": torch.float64, "F": torch.float32}[mode[2]]
+    prec = {"D": torch.float64, "F": torch.float32,
+            "Z": torch.complex128, "C": torch.complex64}  # Z/C: complex modes
+    vec = prec[mode[1]]
+    mat = prec[mode[2]]
     assert mode[3] == "I"
     return mem, vec, mat
 

@@ -86,19 +86,25 @@ def residual(A, x, b, r=None, row_begin=0, row_end=-1):
 
 # ---------------------------------------------------------------------- BLAS-1
 def dot(x, y):
-    return float(torch.dot(x.reshape(-1).double(), y.reshape(-1).double()))
+    xf, yf = x.reshape(-1), y.reshape(-1)
+    if xf.is_complex():
+        return complex(torch.vdot(xf, yf))   # conj(x) . y (complex modes)
+    return float(torch.dot(xf.double(), yf.double()))
 
 
 def nrm2(x):
-    return float(torch.linalg.vector_norm(x.reshape(-1).double()))
+    xf = x.reshape(-1)
+    if xf.is_complex():
+        return float(torch.linalg.vector_norm(xf))
+    return float(torch.linalg.vector_norm(xf.double()))
 
 
 def nrm1(x):
-    return float(x.reshape(-1).double().abs().sum())
+    return float(x.reshape(-1).abs().double().sum())
 
 
 def nrmmax(x):
-    return float(x.reshape(-1).double().abs().max()) if x.numel() else 0.0
+    return float(x.reshape(-1).abs().double().max()) if x.numel() else 0.0
 
 
 def axpy(y, x, alpha):
@@ -123,7 +129,9 @@ def jacobi_dinv(A, l1: bool = False) -> torch.Tensor:
     (l1 adds the off-diagonal block-row L1 norms to the block diagonal,
     reference src/solvers/jacobi_l1_solver.cu)."""
     if A.block_dim == 1:
-        d = extract_diagonal(A).double().clone()
+        work_dtype = torch.complex128 if A.values.is_complex() \
+            else torch.float64
+        d = extract_diagonal(A).to(work_dtype).clone()
         if l1:
             m = _csr(A)
             abssum = np.abs(m).sum(axis=1).A1 - np.abs(_np(extract_diagonal(A)))

@@ -184,16 +184,23 @@ class FGMRESSolver(Solver):
         m = self.restart
         self.V = []          # Krylov basis
         self.Z = []          # preconditioned vectors (flexible)
-        self.H = torch.zeros(m + 1, m, dtype=torch.float64)
+        # complex modes keep a complex Hessenberg; rotations have real c,
+        # complex s (LAPACK zrotg convention)
+        hd = torch.complex128 if b.is_complex() else torch.float64
+        self._hd = hd
+        self.H = torch.zeros(m + 1, m, dtype=hd)
         self.cs = torch.zeros(m, dtype=torch.float64)
-        self.sn = torch.zeros(m, dtype=torch.float64)
-        self.g = torch.zeros(m + 1, dtype=torch.float64)
+        self.sn = torch.zeros(m, dtype=hd)
+        self.g = torch.zeros(m + 1, dtype=hd)
         self._restart_init(b, x)
+
+    def _sc(self, v):
+        """Hessenberg scalar: complex in complex mode, float otherwise."""
+        return complex(v) if self._hd is torch.complex128 else float(v)
 
     def _restart_init(self, b, x):
         r = ops.residual(self.A, x, b)
-        beta = self.compute_norm(r) if self.norm == "L2" else math.sqrt(max(self.dot(r, r), 0.0))
-        beta = math.sqrt(max(self.dot(r, r), 0.0))
+        beta = math.sqrt(abs(self.dot(r, r)))
         self.beta = beta
         self.j = 0
         self.V = [r / beta if beta > 0 else r]
@@ -211,30 +218,42 @@ class FGMRESSolver(Solver):
         self.Z.append(z)
         w = self.new_vec(vj)
         ops.spmv(self.A, z, w)
-        # modified Gram-Schmidt
+        # modified Gram-Schmidt: h_ij = <v_i, w> (conjugated in the first arg)
         for i in range(j + 1):
-            hij = self.dot(w, self.V[i])
+            hij = self.dot(self.V[i], w)
             self.H[i, j] = hij
             ops.axpy(w, self.V[i], -hij)
-        hnext = math.sqrt(max(self.dot(w, w), 0.0))
+        hnext = math.sqrt(abs(self.dot(w, w)))
         self.H[j + 1, j] = hnext
         # apply stored Givens rotations to column j
         for i in range(j):
-            t = self.cs[i] * self.H[i, j] + self.sn[i] * self.H[i + 1, j]
-            self.H[i + 1, j] = -self.sn[i] * self.H[i, j] + self.cs[i] * self.H[i + 1, j]
-            self.H[i, j] = t
-        # new rotation
-        denom = math.hypot(float(self.H[j, j]), hnext)
+            ci, si = float(self.cs[i]), self._sc(self.sn[i])
+            si_c = si.conjugate() if isinstance(si, complex) else si
+            hi, hi1 = self._sc(self.H[i, j]), self._sc(self.H[i + 1, j])
+            self.H[i, j] = ci * hi + si * hi1
+            self.H[i + 1, j] = -si_c * hi + ci * hi1
+        # new rotation zeroing H[j+1, j] (hnext real >= 0)
+        a = self._sc(self.H[j, j])
+        denom = math.sqrt(abs(a) ** 2 + hnext * hnext)
         if denom == 0.0:
             self._update_x(x, j)
             return True
-        self.cs[j] = self.H[j, j] / denom
-        self.sn[j] = hnext / denom
-        self.H[j, j] = denom
+        if abs(a) == 0.0:
+            c, s, rr = 0.0, 1.0, hnext
+        else:
+            phase = a / abs(a)
+            c = abs(a) / denom
+            s = phase * (hnext / denom)
+            rr = phase * denom
+        self.cs[j] = c
+        self.sn[j] = s
+        self.H[j, j] = rr
         self.H[j + 1, j] = 0.0
-        self.g[j + 1] = -self.sn[j] * self.g[j]
-        self.g[j] = self.cs[j] * self.g[j]
-        self._rnorm = abs(float(self.g[j + 1]))
+        gj = self._sc(self.g[j])
+        sconj = s.conjugate() if isinstance(s, complex) else s
+        self.g[j + 1] = -sconj * gj
+        self.g[j] = c * gj
+        self._rnorm = abs(self._sc(self.g[j + 1]))
         lucky = hnext == 0.0
         converged = self.convergence.converged(self._rnorm)
         if converged or lucky or j + 1 == m:
@@ -248,12 +267,14 @@ class FGMRESSolver(Solver):
 
     def _update_x(self, x, j):
         # back-substitute y from the j+1 x j+1 triangular system
-        y = torch.zeros(j + 1, dtype=torch.float64)
+        y = torch.zeros(j + 1, dtype=self._hd)
         for i in range(j, -1, -1):
-            s = float(self.g[i]) - float(self.H[i, i + 1:j + 1] @ y[i + 1:j + 1])
-            y[i] = s / float(self.H[i, i]) if float(self.H[i, i]) != 0.0 else 0.0
+            s = self._sc(self.g[i]) - self._sc(
+                self.H[i, i + 1:j + 1] @ y[i + 1:j + 1])
+            hii = self._sc(self.H[i, i])
+            y[i] = s / hii if hii != 0.0 else 0.0
         for i in range(j + 1):
-            ops.axpy(x, self.Z[i], float(y[i]))
+            ops.axpy(x, self.Z[i], self._sc(y[i]))
 
     def last_residual_norm(self, b, x):
         return self._rnorm

@@ -215,3 +215,68 @@ def test_mixed_precision_mode():
     assert s.status.converged
     rc, nrm = C.AMGX_solver_calculate_residual_norm(s, m, b, x)
     assert nrm < 1e-3
+
+
+def test_complex_mode_cg_and_gmres():
+    """hZZI complex-double mode (reference complex modes,
+    include/amgx_config.h:79-120): Hermitian positive-definite system solved
+    by CG (conjugated dots) and GMRES on the host backend."""
+    import numpy as np
+    import scipy.sparse as sp
+
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.matrix import CSRMatrix
+    from amgx_amd.resources import Resources
+    rng = np.random.RandomState(11)
+    n = 60
+    B = sp.random(n, n, density=0.06, random_state=rng, format="csr")
+    C = sp.random(n, n, density=0.06, random_state=rng, format="csr")
+    M = (B + 1j * C)
+    H = (M + M.conj().T) * 0.5 + sp.identity(n) * 8.0   # Hermitian PD
+    A = CSRMatrix.from_scipy(H.tocsr(), dtype=torch.complex128)
+    xref = torch.from_numpy(rng.randn(n) + 1j * rng.randn(n))
+    b = torch.from_numpy(H @ xref.numpy())
+    for solver_name in ("CG", "GMRES"):
+        cfg = AMGConfig.from_dict({"solver": {
+            "solver": solver_name, "max_iters": 300,
+            "gmres_n_restart": 40, "monitor_residual": 1,
+            "convergence": "RELATIVE_INI", "tolerance": 1e-10,
+        }})
+        s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+        x = torch.zeros(n, dtype=torch.complex128)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        err = float(torch.linalg.vector_norm(x - xref)
+                    / torch.linalg.vector_norm(xref))
+        assert st.converged, f"{solver_name}: {st}"
+        assert err < 1e-7, f"{solver_name}: err={err}"
+
+
+def test_complex_mode_capi():
+    """hZZI through the AMGX_* API with block-Jacobi preconditioning."""
+    import numpy as np
+
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create(
+        "config_version=2, solver=PCG, preconditioner=BLOCK_JACOBI,"
+        " max_iters=300, tolerance=1e-10, convergence=RELATIVE_INI,"
+        " monitor_residual=1")
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hZZI")
+    n = 4
+    ro = [0, 2, 4, 6, 8]
+    ci = [0, 1, 0, 1, 2, 3, 2, 3]
+    va = np.asarray([4.0, 1 - 1j, 1 + 1j, 4.0, 5.0, 2j, -2j, 5.0],
+                    dtype=np.complex128)
+    assert C.AMGX_matrix_upload_all(m, n, 8, 1, 1, ro, ci, va) == C.RC_OK
+    assert m.A.values.dtype == torch.complex128
+    rc, bh = C.AMGX_vector_create(res, "hZZI")
+    rc, xh = C.AMGX_vector_create(res, "hZZI")
+    C.AMGX_vector_upload(bh, n, 1, np.ones(n, dtype=np.complex128))
+    C.AMGX_vector_set_zero(xh, n, 1)
+    rc, s = C.AMGX_solver_create(res, "hZZI", cfg)
+    assert C.AMGX_solver_setup(s, m) == C.RC_OK
+    assert C.AMGX_solver_solve(s, bh, xh) == C.RC_OK
+    rc, nrm = C.AMGX_solver_calculate_residual_norm(s, m, bh, xh)
+    assert rc == C.RC_OK and nrm < 1e-8
