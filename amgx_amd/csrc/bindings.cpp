@@ -263,6 +263,9 @@ std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
         int left = counter.cpu().item<int>();
         if (left == 0) break;
     }
+    TORCH_CHECK((colors.min().cpu().item<int>()) >= 0,
+                "MIN_MAX coloring did not converge in ", max_rounds,
+                " rounds");
     int64_t ncolors = (colors.max().cpu().item<int>()) + 1;
     return {colors, ncolors};
 }

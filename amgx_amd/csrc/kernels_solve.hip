@@ -451,7 +451,7 @@ __global__ void gs_rows_block(const int* __restrict__ ro,
     if (t >= count) return;
     int i = rows[t];
     int s = ro[i], e = ro[i + 1];
-    T res[8], upd[8];
+    T res[16], upd[16];   // b <= 16 (reference supports block sizes to 10)
     for (int c = 0; c < b; ++c) {
         T sum = T(0);
         for (int k = s; k < e; ++k) {
@@ -563,8 +563,12 @@ void dilu_setup_color(const int* ro, const int* ci, const T* va,
         hipLaunchKernelGGL((dilu_setup_scalar<T>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, didx, tidx,
                            colors, rows, count, color, einv);
-    } else {
+    } else if (b <= 8) {
         hipLaunchKernelGGL((dilu_setup_block<T, 8>), dim3(grid_1d(count)),
+                           dim3(AMGX_BLOCK), 0, s, ro, ci, va, didx, tidx,
+                           colors, rows, count, color, einv, b);
+    } else {
+        hipLaunchKernelGGL((dilu_setup_block<T, 16>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, didx, tidx,
                            colors, rows, count, color, einv, b);
     }
@@ -599,7 +603,7 @@ __global__ void dilu_fwd_block(const int* __restrict__ ro,
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
-    T acc[8];
+    T acc[16];            // b <= 16
     for (int c = 0; c < b; ++c) {
         T sum = T(0);
         for (int k = ro[i]; k < ro[i + 1]; ++k) {
@@ -646,7 +650,7 @@ __global__ void dilu_bwd_block(const int* __restrict__ ro,
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
-    T acc[8];
+    T acc[16];            // b <= 16
     for (int c = 0; c < b; ++c) {
         T sum = T(0);
         for (int k = ro[i]; k < ro[i + 1]; ++k) {

@@ -166,6 +166,14 @@ class DistributedManager:
         own_mask = cols_perm < mgr.n_local
         cols_perm[own_mask] = iperm[cols_perm[own_mask]]
         vals_perm = vals_np[gather_nz]
+        # re-sort columns within each row: the interior-first remap breaks
+        # per-row ordering, and the GPU diag_index kernel binary-searches
+        # sorted columns (csrc/kernels_solve.hip diag_index_kernel)
+        if total:
+            row_of = np.repeat(np.arange(mgr.n_local), counts)
+            order = np.lexsort((cols_perm, row_of))
+            cols_perm = cols_perm[order]
+            vals_perm = vals_perm[order]
         # B2L maps and halo refer to OLD local ids -> remap b2l through iperm
         mgr.b2l = [torch.from_numpy(
             iperm[b.cpu().numpy().astype(np.int64)].astype(np.int32)).to(device)

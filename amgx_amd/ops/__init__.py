@@ -56,7 +56,7 @@ def spmv(A, x, y=None, alpha=1.0, beta=0.0, row_begin=0, row_end=-1):
     assert row_begin == 0 and row_end < 0, "windowed distributed spmv: internal"
     if y is None:
         y = _new_dist_vec(A, x)
-    reqs = mgr.exchange_halo(x, async_start=True)
+    reqs = mgr.exchange_halo(x, async_start=True) or []
     B.spmv(A, x, y, alpha, beta, 0, mgr.boundary_start)
     for rq in reqs:
         rq.wait()
