@@ -233,6 +233,12 @@ class Solver:
                 if not math.isfinite(nrm):
                     st.status = SolveStatus.DIVERGED
                     break
+                # explicit divergence check (reference rel_div_tolerance)
+                rdt = self.scope.get("rel_div_tolerance")
+                if (rdt is not None and rdt > 0 and st.residuals
+                        and nrm > rdt * st.residuals[0]):
+                    st.status = SolveStatus.DIVERGED
+                    break
             if done:
                 if not monitoring:
                     st.status = SolveStatus.SUCCESS
