@@ -537,7 +537,7 @@ def dilu_solve(A, Einv, coloring, r, relaxation, x):
         else:
             z[rb] = w[rb] - np.matmul(ei[rows], s.reshape(-1, bd, 1)).reshape(-1)
         later[rb] = z[rb]
-    _np(x).reshape(-1)[:] += relaxation * z
+    _np(x).reshape(-1)[:z.size] += relaxation * z
     return x
 
 
@@ -561,8 +561,8 @@ def ilu0_setup(A, coloring):
     n = A.n_rows
     # color-order position of each row
     order = np.lexsort((np.arange(n), colors))
-    pos = np.empty(n, dtype=np.int64)
-    pos[order] = np.arange(n)
+    pos = np.full(A.n_cols, np.iinfo(np.int64).max, dtype=np.int64)
+    pos[order] = np.arange(n)   # halo columns keep +inf: never pivots
     # pattern lookup
     lut = {}
     for i in range(n):
@@ -594,15 +594,15 @@ def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
     colors = _np(coloring.colors).astype(np.int64)
     n = A.n_rows
     order = np.lexsort((np.arange(n), colors))
-    pos = np.empty(n, dtype=np.int64)
-    pos[order] = np.arange(n)
+    pos = np.full(A.n_cols, np.iinfo(np.int64).max, dtype=np.int64)
+    pos[order] = np.arange(n)   # halo columns: rank-local sweeps skip them
     rv = _np(r).reshape(-1).astype(np.float64)
     y = np.zeros(n)
     for i in order:
         s = rv[i]
         for k in range(ro[i], ro[i + 1]):
             j = ci[k]
-            if pos[j] < pos[i]:
+            if j < n and pos[j] < pos[i]:
                 s -= vals[k] * y[j]
         y[i] = s
     z = np.zeros(n)
@@ -611,12 +611,12 @@ def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
         d = 1.0
         for k in range(ro[i], ro[i + 1]):
             j = ci[k]
-            if pos[j] > pos[i]:
+            if j < n and pos[j] > pos[i]:
                 s -= vals[k] * z[j]
             elif j == i:
                 d = vals[k]
         z[i] = s / (d if d != 0.0 else 1.0)
-    _np(x).reshape(-1)[:] += relaxation * z
+    _np(x).reshape(-1)[:n] += relaxation * z
     return x
 
 

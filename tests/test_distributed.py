@@ -453,3 +453,37 @@ def _impl_test_dist_halo_matrix_two_ring(rank, world, tmp):
                             np.argsort(Afull.indices[
                                 Afull.indptr[g]:Afull.indptr[g + 1]])]):
             assert abs(got[int(j)] - v) < 1e-14
+
+
+def test_dist_ilu0_fgmres():
+    _run_dist(test_dist_ilu0_fgmres)
+
+
+def _impl_test_dist_ilu0_fgmres(rank, world, tmp):
+    """BASELINE config #5 shape: FGMRES + ILU(0) across ranks (rank-local
+    factorization, halo columns excluded from pivoting)."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {"solver": "MULTICOLOR_ILU", "max_iters": 1,
+                           "scope": "ilu"},
+        "solver": "FGMRES", "max_iters": 100, "gmres_n_restart": 25,
+        "monitor_residual": 1, "convergence": "RELATIVE_INI",
+        "tolerance": 1e-8,
+    }})
+    A = _make_dist_A(rank, world, 6)
+    mgr = A.manager
+    s = create_solver(cfg.root_scope(), resources=Resources(
+        "cpu", distributed=True))
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged, f"rank {rank}: {st}"
+    r = ops.residual(A, x, b)
+    nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+        r[:mgr.owned_size])), "L2")
+    bn = mgr.global_norm(float(torch.linalg.vector_norm(
+        b[:mgr.owned_size])), "L2")
+    assert nrm / bn < 1e-7
