@@ -142,13 +142,11 @@ class AggregationLevel(AMGLevel):
         agg_col = aggv.round().to(torch.int32)
         Ac_local = ops.galerkin_aggregation(A, self.aggregates, nc_local,
                                             agg_col, n_global_c)
-        # rebuild distributed structure from global column ids (host pass)
-        ro = Ac_local.row_offsets.cpu().numpy()
-        ci = Ac_local.col_indices.cpu().numpy()
-        va = Ac_local.values.cpu().numpy().reshape(Ac_local.nnz, -1) \
-            if A.block_dim > 1 else Ac_local.values.cpu().numpy()
+        # rebuild distributed structure from global column ids; tensors stay
+        # on device (upload_global_csr is torch-native)
         Ac = DistributedManager.upload_global_csr(
-            ro, ci, va, nc_local, coarse_start, n_global_c,
+            Ac_local.row_offsets, Ac_local.col_indices, Ac_local.values,
+            nc_local, coarse_start, n_global_c,
             device=A.device, block_dim=A.block_dim, dtype=A.dtype)
         # coarse rows were renumbered interior-first by the upload: compose
         # the fine->coarse map with that renumbering

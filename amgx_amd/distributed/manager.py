@@ -96,28 +96,38 @@ class DistributedManager:
         offs[mgr.world] = n_global
         mgr.part_offsets = offs
 
-        cols_global = np.asarray(cols_global, dtype=np.int64)
-        ro = np.asarray(ro, dtype=np.int64)
-        vals_np = np.asarray(vals)
-        own_lo, own_hi = mgr.row_start, mgr.row_start + mgr.n_local
-        is_halo = (cols_global < own_lo) | (cols_global >= own_hi)
-        halo_cols = np.unique(cols_global[is_halo])
-        mgr.n_halo = int(halo_cols.size)
-        mgr.halo_global = halo_cols
+        # device-native structure build (torch ops run on the GPU for the
+        # per-level coarse uploads inside distributed AMG setup; only the
+        # small per-neighbor metadata goes through host collectives)
+        n_loc = mgr.n_local
+        ro_t = torch.as_tensor(np.asarray(ro) if not torch.is_tensor(ro)
+                               else ro, dtype=torch.int64).to(device)
+        ci_t = torch.as_tensor(np.asarray(cols_global)
+                               if not torch.is_tensor(cols_global)
+                               else cols_global, dtype=torch.int64).to(device)
+        if torch.is_tensor(vals):
+            va_t = vals.to(dtype).to(device).reshape(ci_t.numel(), -1)
+        else:
+            va_t = torch.as_tensor(np.ascontiguousarray(vals)).to(dtype) \
+                .to(device).reshape(ci_t.numel(), -1)
+        own_lo, own_hi = mgr.row_start, mgr.row_start + n_loc
+        is_halo = (ci_t < own_lo) | (ci_t >= own_hi)
+        halo_cols = torch.unique(ci_t[is_halo])        # sorted global ids
+        mgr.n_halo = int(halo_cols.numel())
+        halo_np = halo_cols.cpu().numpy()
+        mgr.halo_global = halo_np
 
-        # owner of each halo col; contiguous per owner because halo_cols sorted
-        owners = np.searchsorted(offs, halo_cols, side="right") - 1
+        # owner of each halo col; contiguous per owner because sorted
+        owners = np.searchsorted(offs, halo_np, side="right") - 1
         # local renumbering: own g -> g-row_start ; halo -> n_local + pos
-        new_cols = np.where(
-            is_halo,
-            mgr.n_local + np.searchsorted(halo_cols, cols_global),
-            cols_global - own_lo).astype(np.int32)
+        new_cols = torch.where(
+            is_halo, n_loc + torch.searchsorted(halo_cols, ci_t),
+            ci_t - own_lo)
 
         # tell each owner which of its rows we need (global ids)
-        needed_by_owner = [halo_cols[owners == r] for r in range(mgr.world)]
+        needed_by_owner = [halo_np[owners == r] for r in range(mgr.world)]
         all_needs = [None] * mgr.world
         dist.all_gather_object(all_needs, needed_by_owner)
-        # B2L: rows of MINE that rank r needs
         for r in range(mgr.world):
             if r == mgr.rank:
                 continue
@@ -131,65 +141,65 @@ class DistributedManager:
                 .astype(np.int32)).to(device)
             mgr.b2l.append(b2l)
             if len(i_need):
-                lo = int(np.searchsorted(halo_cols, i_need[0]))
+                lo = int(np.searchsorted(halo_np, i_need[0]))
                 hi = lo + len(i_need)
             else:
                 lo = hi = 0
             mgr.halo_slices.append((lo, hi))
 
         # interior/boundary row split: renumber rows interior-first
-        deg = np.diff(ro)
-        row_has_halo = np.zeros(mgr.n_local, dtype=bool)
-        rows_rep = np.repeat(np.arange(mgr.n_local), deg)
-        np.logical_or.at(row_has_halo, rows_rep, is_halo)
-        interior = np.nonzero(~row_has_halo)[0]
-        boundary = np.nonzero(row_has_halo)[0]
-        perm = np.concatenate([interior, boundary])        # new -> old
-        iperm = np.empty_like(perm)
-        iperm[perm] = np.arange(mgr.n_local)               # old -> new
-        mgr.boundary_start = int(interior.size)
-        mgr.row_perm = torch.from_numpy(perm.astype(np.int64)).to(device)
-        mgr.row_iperm = torch.from_numpy(iperm.astype(np.int64)).to(device)
+        deg = ro_t[1:] - ro_t[:-1]
+        row_of = torch.repeat_interleave(
+            torch.arange(n_loc, dtype=torch.int64, device=device), deg)
+        row_has_halo = torch.zeros(n_loc, dtype=torch.bool, device=device)
+        row_has_halo[row_of[is_halo]] = True
+        interior = torch.nonzero(~row_has_halo).reshape(-1)
+        boundary = torch.nonzero(row_has_halo).reshape(-1)
+        perm = torch.cat([interior, boundary])             # new -> old
+        iperm = torch.empty_like(perm)
+        iperm[perm] = torch.arange(n_loc, dtype=torch.int64, device=device)
+        mgr.boundary_start = int(interior.numel())
+        mgr.row_perm = perm
+        mgr.row_iperm = iperm
 
         # permute rows of the CSR and remap owned column ids through iperm
-        new_ro = np.zeros(mgr.n_local + 1, dtype=np.int64)
-        new_ro[1:] = np.cumsum(deg[perm])
         counts = deg[perm]
-        total = int(counts.sum())
-        if mgr.n_local and total:
-            starts = ro[perm]
-            gather_nz = (np.repeat(starts, counts) + np.arange(total)
-                         - np.repeat(new_ro[:-1], counts))
+        new_ro = torch.zeros(n_loc + 1, dtype=torch.int64, device=device)
+        torch.cumsum(counts, 0, out=new_ro[1:])
+        total = int(new_ro[-1].item()) if n_loc else 0
+        if total:
+            gather_nz = (torch.repeat_interleave(ro_t[perm], counts)
+                         + torch.arange(total, dtype=torch.int64,
+                                        device=device)
+                         - torch.repeat_interleave(new_ro[:-1], counts))
         else:
-            gather_nz = np.zeros(0, dtype=np.int64)
+            gather_nz = torch.zeros(0, dtype=torch.int64, device=device)
         cols_perm = new_cols[gather_nz]
-        own_mask = cols_perm < mgr.n_local
+        own_mask = cols_perm < n_loc
         cols_perm[own_mask] = iperm[cols_perm[own_mask]]
-        vals_perm = vals_np[gather_nz]
+        vals_perm = va_t[gather_nz]
         # re-sort columns within each row: the interior-first remap breaks
         # per-row ordering, and the GPU diag_index kernel binary-searches
         # sorted columns (csrc/kernels_solve.hip diag_index_kernel)
         if total:
-            row_of = np.repeat(np.arange(mgr.n_local), counts)
-            order = np.lexsort((cols_perm, row_of))
+            ncols_ext = n_loc + mgr.n_halo
+            row_new = torch.repeat_interleave(
+                torch.arange(n_loc, dtype=torch.int64, device=device), counts)
+            order = torch.argsort(row_new * max(ncols_ext, 1) + cols_perm)
             cols_perm = cols_perm[order]
             vals_perm = vals_perm[order]
-        # B2L maps and halo refer to OLD local ids -> remap b2l through iperm
-        mgr.b2l = [torch.from_numpy(
-            iperm[b.cpu().numpy().astype(np.int64)].astype(np.int32)).to(device)
-            for b in mgr.b2l]
+        # B2L maps refer to OLD local ids -> remap through iperm
+        mgr.b2l = [iperm[b.to(device).to(torch.int64)].to(torch.int32)
+                   .contiguous() for b in mgr.b2l]
 
-        if block_dim == 1:
-            values = torch.from_numpy(np.ascontiguousarray(vals_perm)) \
-                .to(dtype).to(device)
-        else:
-            values = torch.from_numpy(np.ascontiguousarray(vals_perm)) \
-                .to(dtype).to(device).reshape(-1, block_dim, block_dim)
+        values = vals_perm.contiguous()
+        values = values.reshape(-1) if block_dim == 1 else \
+            values.reshape(-1, block_dim, block_dim)
         A = CSRMatrix(
-            torch.from_numpy(new_ro.astype(np.int32)).to(device),
-            torch.from_numpy(cols_perm.astype(np.int32)).to(device),
+            new_ro.to(torch.int32).contiguous(),
+            cols_perm.to(torch.int32).contiguous(),
             values,
-            n_cols=mgr.n_local + mgr.n_halo, block_dim=block_dim)
+            n_cols=n_loc + mgr.n_halo, block_dim=block_dim)
         A.manager = mgr
         mgr._alloc_send_bufs(dtype)
         return A
