@@ -119,7 +119,9 @@ def main():
     A, manager = build_local_matrix(args.size, rank, world, device)
     n_local = args.size ** 3
     g = torch.Generator().manual_seed(1234 + rank)
-    b = torch.rand(n_local, generator=g, dtype=torch.float64).to(device)
+    b_user = torch.rand(n_local, generator=g, dtype=torch.float64).to(device)
+    # distributed solves run on halo-extended vectors in internal ordering
+    b = manager.permute_in(b_user) if manager is not None else b_user
 
     def barrier_sync():
         if dist is not None:

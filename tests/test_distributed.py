@@ -487,3 +487,24 @@ def _impl_test_dist_ilu0_fgmres(rank, world, tmp):
     bn = mgr.global_norm(float(torch.linalg.vector_norm(
         b[:mgr.owned_size])), "L2")
     assert nrm / bn < 1e-7
+
+
+def test_bench_contract_two_ranks():
+    """bench.py must run under torch.distributed.run with world=2 (the
+    driver's scale-run invocation shape) and print one JSON line."""
+    import json
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29621", os.path.join(repo, "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "0", "--size", "10"],
+        capture_output=True, text=True, timeout=300, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout[-2000:]
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["config"]["converged"]
+    assert rec["config"]["final_rel_residual"] < 1e-6
