@@ -45,6 +45,28 @@ def poisson_3d(nx: int, ny: int, nz: int, device="cpu",
     return CSRMatrix.from_scipy(A, device=device, dtype=dtype)
 
 
+def poisson_3d_27pt(nx: int, ny: int, nz: int, device="cpu",
+                    dtype=torch.float64) -> CSRMatrix:
+    """27-point 3D stencil (reference cusp::gallery::poisson27pt used by the
+    unit tests): tensor-product of 1D [-1, 2, -1] graphs with all diagonal
+    couplings, SPD."""
+    def lap1(n):
+        e = np.ones(n)
+        return sp.diags([-e[:-1], 2 * e, -e[:-1]], [-1, 0, 1])
+
+    def mass1(n):
+        e = np.ones(n)
+        return sp.diags([e[:-1] * 0.5, e, e[:-1] * 0.5], [-1, 0, 1])
+
+    Lx, Ly, Lz = lap1(nx), lap1(ny), lap1(nz)
+    Mx, My, Mz = mass1(nx), mass1(ny), mass1(nz)
+    A = (sp.kron(sp.kron(Mz, My), Lx) + sp.kron(sp.kron(Mz, Ly), Mx)
+         + sp.kron(sp.kron(Lz, My), Mx)).tocsr()
+    A.sum_duplicates()
+    A.eliminate_zeros()
+    return CSRMatrix.from_scipy(A, device=device, dtype=dtype)
+
+
 def poisson_3d_local(nx: int, ny: int, nz: int, rank: int, world: int):
     """Rank-local slab rows (z-partition) of a global nx*ny*(nz*world) 7-pt
     Poisson, as (row_offsets, col_indices GLOBAL, values, row_start) numpy.

@@ -24,20 +24,39 @@ class Scaler:
         raise NotImplementedError
 
     def scale_matrix(self, A):
-        """A <- D A D in place (values only)."""
+        """A <- D A D in place (values only). Distributed: halo columns use
+        the OWNER's d (one halo exchange) so every rank scales the same
+        global operator."""
         d = self.d
         rows = torch.repeat_interleave(
             torch.arange(A.n_rows, device=A.device),
             (A.row_offsets[1:] - A.row_offsets[:-1]).to(torch.int64))
         dr = d[rows]
-        dcol_src = torch.ones(A.n_cols, dtype=d.dtype, device=d.device)
-        dcol_src[:d.numel()] = d
+        mgr = getattr(A, "manager", None)
+        if mgr is not None and mgr.n_halo:
+            ext = mgr.new_ext_vec(torch.float64)
+            ext[:mgr.owned_size] = d.reshape(-1).to(torch.float64)
+            mgr.exchange_halo(ext, block_override=1)
+            dcol_src = ext.to(d.dtype)
+        else:
+            dcol_src = torch.ones(A.n_cols, dtype=d.dtype, device=d.device)
+            dcol_src[:d.numel()] = d
         dc = dcol_src[A.col_indices.to(torch.int64)]
         if A.block_dim == 1:
             A.values.mul_(dr * dc)
         else:
             A.values.mul_((dr * dc).view(-1, 1, 1))
         A.clear_cache()
+
+    def unscale_matrix(self, A):
+        """Restore A (divide the scaling back out; reference
+        scaleMatrix(..., amgx::UNSCALE))."""
+        d = self.d
+        self.d = 1.0 / d
+        try:
+            self.scale_matrix(A)
+        finally:
+            self.d = d
 
     def scale_rhs(self, b):
         out = b.clone()
@@ -85,8 +104,7 @@ class BinormalizationScaler(Scaler):
         for _ in range(self.sweeps):
             s = B @ w
             s = np.where(s > 0, s, 1.0)
-            w = np.sqrt(w / np.sqrt(s / n * B.sum() / n)) if False else \
-                w / np.sqrt(np.sqrt(s))
+            w = w / np.sqrt(np.sqrt(s))
             w *= n / w.sum()
         d = np.sqrt(w)
         self.d = torch.from_numpy(d).to(A.dtype).to(A.device)

@@ -127,17 +127,19 @@ class Solver:
         self.scaler = None
         scaling = self.scope.get("scaling")
         if scaling and scaling != "NONE":
+            # reference lifecycle (src/solvers/solver.cu:443-476): scale the
+            # matrix around preconditioner setup, restore it afterwards;
+            # solve() re-scales around the iteration (':667-676, :853-880)
             from ..scalers import create_scaler
-            prev = getattr(A, "_scaled_by", None)
-            if prev is not None:      # re-setup: undo the previous scaling
-                prev.d = 1.0 / prev.d
-                prev.scale_matrix(A)
-                A._scaled_by = None
             self.scaler = create_scaler(scaling)
             self.scaler.setup(A)
             self.scaler.scale_matrix(A)
-            A._scaled_by = self.scaler
-        self.solver_setup()
+            try:
+                self.solver_setup()
+            finally:
+                self.scaler.unscale_matrix(A)
+        else:
+            self.solver_setup()
         self.res.synchronize() if self.res.is_cuda else None
         self.status.setup_time = time.perf_counter() - t0
 
@@ -184,11 +186,17 @@ class Solver:
         initial residual + norm, solve_init, iterate until
         converged/max_iters."""
         if getattr(self, "scaler", None) is not None:
+            # reference Solver::solve (:667-676): scale A and b, map x into
+            # the scaled space; at exit (:853-861) restore A and map x back
+            self.scaler.scale_matrix(self.A)
             bs = self.scaler.scale_rhs(b)
             if not zero_initial_guess:
                 self.scaler.scale_guess(x)
-            st = self._solve_inner(bs, x, zero_initial_guess)
-            self.scaler.unscale_solution(x)
+            try:
+                st = self._solve_inner(bs, x, zero_initial_guess)
+            finally:
+                self.scaler.unscale_solution(x)
+                self.scaler.unscale_matrix(self.A)
             return st
         return self._solve_inner(b, x, zero_initial_guess)
 

@@ -508,3 +508,34 @@ def test_bench_contract_two_ranks():
     rec = json.loads(lines[0])
     assert rec["n_gpus"] == 2 and rec["config"]["converged"]
     assert rec["config"]["final_rel_residual"] < 1e-6
+
+
+def test_dist_scaler():
+    _run_dist(test_dist_scaler)
+
+
+def _impl_test_dist_scaler(rank, world, tmp):
+    """Scalers in the distributed solve lifecycle (reference scaler hook,
+    src/solvers/solver.cu:667-676)."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "PCG", "max_iters": 200, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+        "scaling": "DIAGONAL_SYMMETRIC",
+        "preconditioner": {"solver": "BLOCK_JACOBI", "max_iters": 1},
+    }})
+    A = _make_dist_A(rank, world, 5)
+    mgr = A.manager
+    s = create_solver(cfg.root_scope(), resources=Resources(
+        "cpu", distributed=True))
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged, f"rank {rank}: {st}"
+    r = ops.residual(A, x, b)
+    nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+        r[:mgr.owned_size])), "L2")
+    assert nrm < 1e-5
