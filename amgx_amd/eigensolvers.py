@@ -87,10 +87,9 @@ class PageRankOperator(Operator):
     def __init__(self, A, damping: float = 0.85):
         super().__init__(A)
         self.d = damping
-        # column-stochastic normalization: P = A D_out^{-1}
-        deg = ops.spmv(ops.transpose(A) if False else A,
-                       torch.ones(self.n, dtype=A.dtype, device=A.device))
+        # column-stochastic normalization: P = A D_out^{-1};
         # out-degree = row sums of A
+        deg = ops.spmv(A, torch.ones(self.n, dtype=A.dtype, device=A.device))
         self.out_inv = torch.where(deg != 0, 1.0 / deg, torch.zeros_like(deg))
         self.AT = ops.transpose(A)
 

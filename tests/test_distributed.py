@@ -539,3 +539,40 @@ def _impl_test_dist_scaler(rank, world, tmp):
     nrm = mgr.global_norm(float(torch.linalg.vector_norm(
         r[:mgr.owned_size])), "L2")
     assert nrm < 1e-5
+
+
+def test_dist_classical_three_ranks():
+    _run_dist(test_dist_classical_three_ranks, world=3)
+
+
+def _impl_test_dist_classical_three_ranks(rank, world, tmp):
+    """Classical distributed AMG with world=3 (uneven neighbor counts: the
+    middle rank has two neighbors, ends one) — exercises multi-neighbor B2L
+    bookkeeping beyond the 2-rank case."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {
+            "solver": "AMG", "algorithm": "CLASSICAL",
+            "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+            "max_iters": 1, "min_coarse_rows": 12, "cycle": "V",
+        },
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+    }})
+    A = _make_dist_A(rank, world, 6)
+    mgr = A.manager
+    s = create_solver(cfg.root_scope(), resources=Resources(
+        "cpu", distributed=True))
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged, f"rank {rank}: {st}"
+    r = ops.residual(A, x, b)
+    nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+        r[:mgr.owned_size])), "L2")
+    bn = mgr.global_norm(float(torch.linalg.vector_norm(
+        b[:mgr.owned_size])), "L2")
+    assert nrm / bn < 1e-7
