@@ -28,7 +28,12 @@ def register_agg_selector(name: str):
 
 def _compose_passes(A, scope, passes: int, matcher=None):
     """Repeated pairwise matching: pass k matches the level-k aggregates
-    (reference size4 = 'two SIZE_2 passes')."""
+    (reference size4 = 'two SIZE_2 passes').
+    ``aggregation_passes`` (reference param) overrides the selector's pass
+    count when set."""
+    override = int(scope.get("aggregation_passes") or 0)
+    if override > 0:
+        passes = override
     maxit = scope.get("max_matching_iterations")
     match = matcher or (lambda M: ops.size2_matching(M, max_iterations=maxit))
     agg, num = match(A)            # composed fine->current map

@@ -152,6 +152,10 @@ class AMGHierarchy:
             self.coarse_solver.setup(coarsest.A)
             return
         name, sub = self.scope.sub_solver("coarse_solver", "DENSE_LU_SOLVER")
+        if (sub is not None and not sub.has("max_iters")
+                and self.scope.has("max_coarse_iters")):
+            sub = sub.child(dict(sub.node,
+                                 max_iters=self.scope.get("max_coarse_iters")))
         if name == "DENSE_LU_SOLVER":
             maxr = self.scope.get("dense_lu_max_rows")
             if maxr and coarsest.A.n_rows > maxr:
@@ -187,9 +191,14 @@ class AMGHierarchy:
             return
         if zero_guess:
             x.zero_()
-        if self.presweeps > 0:
+        pre = self.presweeps
+        if li == 0:
+            fs = int(self.scope.get("finest_sweeps") or -1)
+            if fs >= 0:       # reference finest_sweeps override
+                pre = fs
+        if pre > 0:
             prof.tic("Smoother")
-            level.smoother.sweep(b, x, self.presweeps)
+            level.smoother.sweep(b, x, pre)
             prof.toc("Smoother")
         prof.tic("computeResidual")
         ops.residual(level.A, x, b, level.r)

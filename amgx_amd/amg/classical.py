@@ -54,11 +54,16 @@ def strength_all(A, scope):
 
 
 @_register(STRENGTH_REGISTRY, "AFFINITY")
-def strength_affinity(A, scope, n_vecs: int = 4, n_sweeps: int = 8):
+def strength_affinity(A, scope, n_vecs: int = None, n_sweeps: int = None):
     """Affinity strength from smoothed random test vectors (reference
-    src/classical/strength/affinity.cu): relax A x = 0 from random starts;
-    entries whose endpoints move together are strong:
+    src/classical/strength/affinity.cu, affinity_vectors/affinity_iterations
+    params): relax A x = 0 from random starts; entries whose endpoints move
+    together are strong:
     aff_ij = (sum_k x_ki x_kj)^2 / (sum_k x_ki^2 * sum_k x_kj^2)."""
+    if n_vecs is None:
+        n_vecs = int(scope.get("affinity_vectors") or 4)
+    if n_sweeps is None:
+        n_sweeps = int(scope.get("affinity_iterations") or 8)
     ro, ci, v = _csr_parts(A)
     n = A.n_rows
     m = sp.csr_matrix((v, ci, ro), shape=(n, A.n_cols))[:, :n]

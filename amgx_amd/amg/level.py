@@ -316,10 +316,16 @@ class ClassicalLevel(AMGLevel):
 class EnergyminLevel(ClassicalLevel):
     """Energy-minimization AMG level (reference
     src/energymin/energymin_amg_level.cu): compatible-relaxation C/F
-    selection + per-patch energy-minimizing interpolation (EM)."""
+    selection + per-patch energy-minimizing interpolation (EM);
+    energymin_selector / energymin_interpolator params override."""
 
-    default_selector = "CR"
-    default_interp = "EM"
+    @property
+    def default_selector(self):
+        return self.scope.get("energymin_selector") or "CR"
+
+    @property
+    def default_interp(self):
+        return self.scope.get("energymin_interpolator") or "EM"
 
 
 def create_level(algorithm: str, A, scope, index) -> AMGLevel:

@@ -486,6 +486,13 @@ def AMGX_vector_get_size(v: _VectorHandle):
 @_amgx_try
 def AMGX_solver_create(res: _ResourcesHandle, mode: str, cfg: _ConfigHandle):
     _parse_mode(mode)
+    tree = cfg.cfg.tree
+    solver_node = tree.get("solver")
+    wants_print = tree.get("print_config") or (
+        isinstance(solver_node, dict) and solver_node.get("print_config"))
+    if wants_print:   # reference print_config param
+        import json as _json
+        print(_json.dumps(tree, indent=2, default=str))
     return RC_OK, _SolverHandle(res.res, mode, cfg.cfg)
 
 

@@ -150,6 +150,83 @@ def _register_core_parameters() -> None:
     P("eig_eigenvector", int, 1, "number of eigenvectors")
     P("eig_wanted_count", int, 1, "number of wanted eigenpairs")
 
+    # --- remaining reference registry (src/core.cu:307-545) ------------------
+    # Wired parameters:
+    P("affinity_vectors", int, 4, "AFFINITY strength: number of test vectors")
+    P("affinity_iterations", int, 8, "AFFINITY strength: smoothing sweeps")
+    P("aggregation_passes", int, 0,
+      "pairwise-matching passes (0 = selector default)")
+    P("energymin_selector", str, "CR", "selector for ENERGYMIN levels")
+    P("energymin_interpolator", str, "EM", "interpolator for ENERGYMIN levels")
+    P("matrix_writer", str, "matrixmarket", "write_system format",
+      ("matrixmarket", "binary"))
+    P("finest_sweeps", int, -1, "sweeps on the finest level (-1 = presweeps)")
+    P("max_coarse_iters", int, 100, "coarse-solver iteration cap")
+    P("rhs_from_a", int, 0, "synthesize rhs = A*ones when the file has none")
+    P("print_config", int, 0, "print the parsed config tree at create")
+    P("kpz_order", int, 3, "KPZ polynomial order")
+    P("kpz_mu", float, 2.0, "KPZ polynomial mu parameter")
+    # Accepted for config compatibility; behavior governed by the MI355X-first
+    # design (COMPONENTS.md): torch caching allocator is the memory pool, the
+    # gathered dense coarse solve replaces consolidation/gluing, RCCL halo
+    # exchange replaces the comm knobs.
+    for name, typ, dflt, doc in (
+        ("aggregation_edge_weight_component", int, 0, "block weight component"),
+        ("weight_formula", int, 0, "pairwise matching weight formula"),
+        ("notay_weights", int, 0, "Notay quality weights in matching"),
+        ("filter_weights", int, 0, "filter small matching weights"),
+        ("filter_weights_alpha", float, 0.25, "weight filter threshold"),
+        ("serial_matching", int, 0, "serial matching fallback"),
+        ("handshaking_phases", int, 1, "matching handshake phases"),
+        ("modified_handshake", int, 0, "modified handshaking"),
+        ("late_rejection", int, 0, "late rejection in matching"),
+        ("merge_singletons", int, 1, "merge unmatched rows into aggregates"),
+        ("weakness_bound", float, 0.0, "weak-edge bound in matching"),
+        ("ghost_offdiag_limit", int, 0, "ghost off-diagonal limit"),
+        ("coarse_smoother", str, "NOSOLVER", "smoother on coarse levels"),
+        ("fine_smoother", str, "NOSOLVER", "smoother on fine levels"),
+        ("smoother_amg_list", str, "", "per-level smoother list"),
+        ("fine_levels", int, 0, "levels treated as fine"),
+        ("min_fine_rows", int, 1, "minimum fine-level rows"),
+        ("coarseAgenerator", str, "LOW_DEG", "aggregation Galerkin generator"),
+        ("coarseAgenerator_coarse", str, "LOW_DEG", "coarse-level generator"),
+        ("boundary_coloring", str, "SYNC_COLORS", "halo coloring mode"),
+        ("halo_coloring", int, 1, "color halo rows"),
+        ("initial_color", int, 0, "first color index"),
+        ("coloring_custom_arg", str, "", "scheme-specific coloring arg"),
+        ("coloring_try_remove_last_colors", int, 0, "recolor-down passes"),
+        ("print_coloring_info", int, 0, "print coloring statistics"),
+        ("print_aggregation_info", int, 0, "print aggregation statistics"),
+        ("convergence_analysis", int, 0, "convergence analysis output"),
+        ("verbosity_level", int, 3, "output verbosity"),
+        ("device_mem_pool_size", int, 0, "device pool size (torch allocator)"),
+        ("device_mem_pool_size_limit", int, 0, "device pool limit"),
+        ("device_mem_pool_max_alloc_size", int, 0, "device pool max alloc"),
+        ("device_alloc_scaling_factor", int, 10, "pool over-alloc scaling"),
+        ("device_alloc_scaling_threshold", int, 16384, "pool scaling cutoff"),
+        ("device_consolidation_pool_size", int, 0, "consolidation pool size"),
+        ("num_streams", int, 0, "worker streams (HIP streams used directly)"),
+        ("high_priority_stream", int, 0, "high-priority compute stream"),
+        ("serialize_threads", int, 0, "serialize launches (debug)"),
+        ("use_cuda_ipc_consolidation", int, 0, "IPC consolidation (n/a)"),
+        ("amg_consolidation_flag", int, 0, "consolidation mode (n/a)"),
+        ("fine_level_consolidation", int, 0, "fine consolidation (n/a)"),
+        ("matrix_halo_exchange", int, 0, "matrix halo exchange depth"),
+        ("separation_interior", int, 0, "interior separation flag"),
+        ("separation_exterior", int, 0, "exterior separation flag"),
+        ("use_bsrxmv", int, 0, "masked block SpMV routing"),
+        ("block_convert", int, 0, "scalar<->block conversion"),
+        ("complex_conversion", int, 0, "complex conversion mode"),
+        ("geometric_dim", int, 2, "geometry dimension for GEO"),
+        ("jacobi_iters", int, 5, "inner Jacobi iterations (polynomial)"),
+        ("scaling_smoother_steps", int, 2, "scaler smoother steps"),
+        ("reuse_scale", int, 0, "reuse scaler across solves"),
+        ("use_sum_stopping_criteria", int, 0, "sum-based stopping"),
+        ("max_uncolored_percentage", float, 0.15, "allowed uncolored rows"),
+    ):
+        if name not in PARAM_REGISTRY:
+            P(name, typ, dflt, doc)
+
 
 _register_core_parameters()
 
