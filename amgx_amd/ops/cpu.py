@@ -414,8 +414,11 @@ def galerkin_rap(R, A, P):
 
 def truncate_rows(P, trunc_factor: float = 0.0, max_elements: int = -1):
     """Drop |p| < factor*rowmax (and cap per-row count), rescale to preserve the
-    row sum (reference src/truncate.cu truncateAndScale_kernel)."""
+    row sum (reference src/truncate.cu truncateAndScale_kernel). Factors
+    outside (0, 1) disable dropping (reference default 1.1 = off)."""
     from ..matrix import CSRMatrix
+    if not (0.0 < trunc_factor < 1.0):
+        trunc_factor = 0.0
     if trunc_factor <= 0.0 and max_elements < 0:
         return P
     m = _csr(P).copy()

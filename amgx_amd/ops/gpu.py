@@ -251,7 +251,7 @@ def galerkin_rap(R, A, P):
 
 
 def truncate_rows(P, trunc_factor: float = 0.0, max_elements: int = -1):
-    if trunc_factor <= 0.0 and max_elements < 0:
+    if not (0.0 < trunc_factor < 1.0) and max_elements < 0:
         return P
     from . import cpu
     return cpu.truncate_rows(P.to("cpu"), trunc_factor, max_elements) \
