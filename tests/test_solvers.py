@@ -280,3 +280,31 @@ def test_complex_mode_capi():
     assert C.AMGX_solver_solve(s, bh, xh) == C.RC_OK
     rc, nrm = C.AMGX_solver_calculate_residual_norm(s, m, bh, xh)
     assert rc == C.RC_OK and nrm < 1e-8
+
+
+def test_ilu_k_levels():
+    """ILU(1) (level-1 fill, reference csr_sparsity_ilu1 + ilu_sparsity_level)
+    preconditions better than ILU(0)."""
+    from amgx_amd.solvers.ilu import extended_sparsity
+    A = poisson_2d(14, 14)
+    E = extended_sparsity(A, 1)
+    assert E.nnz > A.nnz            # fill added
+    assert E.n_rows == A.n_rows
+
+    def iters(level):
+        s = make({"solver": "FGMRES", "max_iters": 200,
+                  "gmres_n_restart": 40, "monitor_residual": 1,
+                  "tolerance": 1e-10, "convergence": "RELATIVE_INI",
+                  "preconditioner": {"solver": "MULTICOLOR_ILU",
+                                     "max_iters": 1,
+                                     "ilu_sparsity_level": level}})
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+        assert st.converged and rel < 1e-8, f"ILU({level}): {st} rel={rel}"
+        return st.iterations
+
+    it0, it1 = iters(0), iters(1)
+    assert it1 <= it0, f"ILU(1) {it1} !<= ILU(0) {it0}"
