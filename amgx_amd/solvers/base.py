@@ -114,6 +114,7 @@ class Solver:
         self.store_res_history = bool(scope.get("store_res_history"))
         self.print_solve_stats = bool(scope.get("print_solve_stats"))
         self.norm = scope.get("norm")
+        self.min_iters = int(scope.get("min_iters") or 0)
         self.convergence = Convergence(scope.get("convergence"),
                                        scope.get("tolerance"),
                                        scope.get("alt_rel_tolerance"))
@@ -165,6 +166,14 @@ class Solver:
             nrm = ops.nrm2(r)
         elif self.norm == "L1":
             nrm = ops.nrm1(r)
+        elif self.norm == "L1_SCALED":
+            # reference L1_SCALED: L1 norm divided by the global length
+            n = r.numel()
+            mgr = getattr(self.A, "manager", None) if self.A is not None \
+                else None
+            if mgr is not None:
+                n = mgr.n_global * mgr.block_dim
+            nrm = ops.nrm1(r) / max(n, 1)
         elif self.norm == "LMAX":
             nrm = ops.nrmmax(r)
         else:
@@ -235,7 +244,8 @@ class Solver:
                     rate = (st.residuals[-1] / st.residuals[-2]
                             if st.residuals[-2] else 0.0)
                     print(f"           {it:4d} {nrm:14.6e}  {rate:6.4f}")
-                if self.convergence.converged(nrm):
+                if self.convergence.converged(nrm) \
+                        and it + 1 >= self.min_iters:
                     st.status = SolveStatus.SUCCESS
                     break
                 if not math.isfinite(nrm):

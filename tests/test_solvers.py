@@ -308,3 +308,26 @@ def test_ilu_k_levels():
 
     it0, it1 = iters(0), iters(1)
     assert it1 <= it0, f"ILU(1) {it1} !<= ILU(0) {it0}"
+
+
+def test_min_iters_and_l1_scaled_norm():
+    """min_iters forces extra iterations; L1_SCALED divides by length
+    (reference norm machinery, src/solvers/solver.cu:209-260)."""
+    A = poisson_2d(8, 8)
+    s = make({"solver": "CG", "max_iters": 60, "min_iters": 12,
+              "monitor_residual": 1, "tolerance": 1e-3,
+              "convergence": "RELATIVE_INI"})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged and st.iterations >= 12
+    s2 = make({"solver": "CG", "max_iters": 200, "monitor_residual": 1,
+               "tolerance": 1e-10, "convergence": "ABSOLUTE",
+               "norm": "L1_SCALED"})
+    x2 = torch.zeros_like(b)
+    s2.setup(A)
+    st2 = s2.solve(b, x2, zero_initial_guess=True)
+    assert st2.converged
+    r = ops.residual(A, x2, b)
+    assert float(ops.nrm1(r)) / A.n_rows < 1e-9
