@@ -145,3 +145,146 @@ def test_memory_info():
     from amgx_amd.utils import MemoryInfo
     u = MemoryInfo.get_max_memory_usage()
     assert u["host_peak_mib"] > 1.0
+
+
+# ---------------------------------------------- reference edge-case analogues
+def test_zero_diagonal_handling():
+    """Rows with zero (or missing) diagonal must not produce NaN/Inf in
+    Jacobi/GS/DILU (reference zero_in_diagonal_handling.cu +
+    zero_values_handling.cu)."""
+    import numpy as np
+    import scipy.sparse as sp
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.matrix import CSRMatrix
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    m = poisson_2d(6, 6).to_scipy().tolil()
+    m[7, 7] = 0.0          # explicit zero diagonal
+    A = CSRMatrix.from_scipy(m.tocsr())
+    for smoother in ("BLOCK_JACOBI", "MULTICOLOR_GS", "MULTICOLOR_DILU"):
+        cfg = AMGConfig.from_dict({"solver": {
+            "solver": smoother, "max_iters": 5, "monitor_residual": 1}})
+        s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        s.solve(b, x)
+        assert torch.isfinite(x).all(), smoother
+
+
+def test_fgmres_zero_initial_residual():
+    """b = 0 with zero guess: immediate success, x stays 0 (reference
+    fgmres_zero_initial_residual.cu)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    A = poisson_2d(6, 6)
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "FGMRES", "max_iters": 50, "monitor_residual": 1,
+        "convergence": "ABSOLUTE", "tolerance": 1e-12}})
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.zeros(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged and st.iterations == 0
+    assert torch.equal(x, torch.zeros_like(x))
+
+
+def test_nested_solvers():
+    """Solver-composition depth 3: FGMRES -> PCG preconditioner -> AMG
+    preconditioner (reference nested_solvers.cu)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "FGMRES", "max_iters": 60, "gmres_n_restart": 20,
+        "monitor_residual": 1, "convergence": "RELATIVE_INI",
+        "tolerance": 1e-8,
+        "preconditioner": {
+            "solver": "PCG", "max_iters": 4, "monitor_residual": 1,
+            "scope": "inner",
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "AGGREGATION",
+                "smoother": "BLOCK_JACOBI", "max_iters": 1,
+                "min_coarse_rows": 16, "cycle": "V", "scope": "amg"}},
+    }})
+    A = poisson_3d(7, 7, 7)
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+    assert st.converged and rel < 1e-7
+
+
+def test_preconditioner_reuse_across_solves():
+    """One setup, many solves with different rhs (reference
+    preconditioner_usage.cu)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    A = poisson_2d(10, 10)
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-10,
+        "preconditioner": {"solver": "AMG", "algorithm": "AGGREGATION",
+                           "smoother": "BLOCK_JACOBI", "max_iters": 1,
+                           "min_coarse_rows": 8, "scope": "amg"}}})
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    s.setup(A)
+    g = torch.Generator().manual_seed(5)
+    for trial in range(3):
+        b = torch.rand(A.n_rows, generator=g, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+        assert st.converged and rel < 1e-9, f"trial {trial}"
+
+
+def test_capi_object_destruction_order():
+    """Destroying handles in any order must not corrupt others (reference
+    object_destruction.cu)."""
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create("config_version=2, solver=CG, max_iters=20,"
+                                   " monitor_residual=1, tolerance=1e-8,"
+                                   " convergence=RELATIVE_INI")
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    rc, b = C.AMGX_vector_create(res, "hDDI")
+    rc, x = C.AMGX_vector_create(res, "hDDI")
+    assert C.AMGX_generate_distributed_poisson_7pt(
+        m, b, x, 1, 1, 6, 6, 6) == C.RC_OK
+    rc, s = C.AMGX_solver_create(res, "hDDI", cfg)
+    assert C.AMGX_solver_setup(s, m) == C.RC_OK
+    # destroy config before solve (reference allows it: solver holds a copy)
+    assert C.AMGX_config_destroy(cfg) == C.RC_OK
+    assert C.AMGX_solver_solve(s, b, x) == C.RC_OK
+    assert C.AMGX_matrix_destroy(m) == C.RC_OK
+    assert C.AMGX_vector_destroy(b) == C.RC_OK
+    assert C.AMGX_solver_destroy(s) == C.RC_OK
+    assert C.AMGX_resources_destroy(res) == C.RC_OK
+    assert C.AMGX_finalize() == C.RC_OK
+
+
+def test_memory_use_accounting():
+    """MemoryInfo peaks grow after a large allocation (reference
+    memory_use.cu analogue)."""
+    import numpy as np
+
+    from amgx_amd.utils import MemoryInfo
+    before = MemoryInfo.get_max_memory_usage()["host_peak_mib"]
+    blob = np.ones((64, 1024, 1024))      # ~512 MiB
+    after = MemoryInfo.get_max_memory_usage()["host_peak_mib"]
+    assert after >= before
+    assert blob.sum() > 0
