@@ -700,55 +700,56 @@ def interp_d1(A, S, cf_map, num_coarse):
     src/classical/interpolators/distance1.cu):
       w_ij = -alpha_i * a_ij / a_ii,  alpha = sum(neg a) / sum(neg a over C),
     likewise beta for positive entries; positives lumped into the diagonal
-    when no positive C connection exists."""
+    when no positive C connection exists. Fully vectorized (the per-row
+    formula has no cross-row coupling), so host-mode and the distributed
+    classical setup scale to multi-million-row partitions."""
     from ..matrix import CSRMatrix
     ro = _np(A.row_offsets).astype(np.int64)
     ci = _np(A.col_indices).astype(np.int64)
     v = _np(A.values).astype(np.float64)
-    strong = _np(S)
+    strong = np.asarray(_np(S), dtype=bool)
     cf = _np(cf_map).astype(np.int64)
     n = A.n_rows
-    Pr, Pc, Pv = [], [], []
-    indptr = [0]
-    for i in range(n):
-        if cf[i] >= 0:
-            Pc.append(cf[i]); Pv.append(1.0)
-            indptr.append(indptr[-1] + 1)
-            continue
-        s, e = ro[i], ro[i + 1]
-        diag = 0.0
-        neg_all = pos_all = neg_c = pos_c = 0.0
-        entries = []
-        for k in range(s, e):
-            j, a = ci[k], v[k]
-            if j == i:
-                diag = a
-                continue
-            if a < 0:
-                neg_all += a
-            else:
-                pos_all += a
-            if strong[k] and cf[j] >= 0:
-                if a < 0:
-                    neg_c += a
-                else:
-                    pos_c += a
-                entries.append((cf[j], a))
-        if not entries or diag == 0.0:
-            indptr.append(indptr[-1])
-            continue
-        if pos_c == 0.0:
-            diag += pos_all
-            pos_all = 0.0
-        alpha = neg_all / neg_c if neg_c != 0.0 else 0.0
-        beta = pos_all / pos_c if pos_c != 0.0 else 0.0
-        acc = {}
-        for jc, a in entries:
-            wgt = -(alpha if a < 0 else beta) * a / diag
-            acc[jc] = acc.get(jc, 0.0) + wgt
-        for jc in sorted(acc):
-            Pc.append(jc); Pv.append(acc[jc])
-        indptr.append(indptr[-1] + len(acc))
-    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
-                       np.asarray(indptr)), shape=(n, num_coarse))
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    offd = rows != ci
+    neg = v < 0
+    # per-row sums over off-diagonals / strong-C off-diagonals
+    cf_of_col = np.where(ci < cf.size, cf[np.minimum(ci, cf.size - 1)], -1)
+    strongC = strong & offd & (cf_of_col >= 0)
+    neg_all = np.bincount(rows[offd & neg], weights=v[offd & neg],
+                          minlength=n)
+    pos_all = np.bincount(rows[offd & ~neg], weights=v[offd & ~neg],
+                          minlength=n)
+    neg_c = np.bincount(rows[strongC & neg], weights=v[strongC & neg],
+                        minlength=n)
+    pos_c = np.bincount(rows[strongC & ~neg], weights=v[strongC & ~neg],
+                        minlength=n)
+    diag = np.zeros(n)
+    dmask = rows == ci
+    diag[rows[dmask]] = v[dmask]
+    # positives lumped into the diagonal when no positive C connection
+    lump = pos_c == 0.0
+    diag2 = diag + np.where(lump, pos_all, 0.0)
+    alpha = np.divide(neg_all, neg_c, out=np.zeros(n), where=neg_c != 0.0)
+    beta = np.divide(np.where(lump, 0.0, pos_all), pos_c,
+                     out=np.zeros(n), where=pos_c != 0.0)
+    # F-row entries: one per strong C neighbor (cols unique within a row)
+    has_c = np.bincount(rows[strongC], minlength=n) > 0
+    frow_ok = (cf[:n] < 0) & (diag2 != 0.0) & has_c
+    emit = strongC & frow_ok[rows]
+    er = rows[emit]
+    ec = cf_of_col[emit]
+    coef = np.where(neg[emit], alpha[er], beta[er])
+    ev = -coef * v[emit] / diag2[er]
+    # C rows: identity entries
+    c_rows = np.nonzero(cf[:n] >= 0)[0]
+    Pr = np.concatenate([er, c_rows])
+    Pc = np.concatenate([ec, cf[c_rows]])
+    Pv = np.concatenate([ev, np.ones(c_rows.size)])
+    order = np.lexsort((Pc, Pr))
+    counts = np.bincount(Pr, minlength=n)
+    indptr = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(counts, out=indptr[1:])
+    P = sp.csr_matrix((Pv[order], Pc[order], indptr),
+                      shape=(n, num_coarse))
     return CSRMatrix.from_scipy(P, dtype=A.dtype)
