@@ -651,48 +651,50 @@ def strength_ahat(A, theta: float = 0.25, max_row_sum: float = 1.1):
 def pmis_select(A, S):
     """PMIS C/F splitting (reference src/classical/selectors/pmis.cu:
     random-weight independent-set rounds). Returns (cf_map int32: >=0 coarse
-    index for C, -1 for F, and num_coarse)."""
+    index for C, -1 for F, and num_coarse). Vectorized Luby rounds with a
+    (weight, id) total order — each round's decisions are made against a
+    frozen snapshot, like the gfx950 kernel."""
     ro = _np(A.row_offsets).astype(np.int64)
     ci = _np(A.col_indices).astype(np.int64)
     n = A.n_rows
-    strong = _np(S)
-    # symmetrized strong adjacency (S union S^T) as CSR of indices
+    strong = np.asarray(_np(S), dtype=bool)
     rows = np.repeat(np.arange(n), np.diff(ro))
-    sr, sc = rows[strong], ci[strong]
-    adj = sp.csr_matrix((np.ones(2 * sr.size), (np.concatenate([sr, sc]),
-                                                np.concatenate([sc, sr]))),
-                        shape=(n, n))
-    adj.sum_duplicates()
+    own = strong & (ci < n)
+    # symmetrized strong edge list (S union S^T), both directions present
+    er = np.concatenate([rows[own], ci[own]])
+    ec = np.concatenate([ci[own], rows[own]])
+    keep = er != ec
+    er, ec = er[keep], ec[keep]
     # lambda_i = number of points strongly influenced by i (= S^T row count)
     lam = np.zeros(n)
-    np.add.at(lam, sc, 1.0)
+    np.add.at(lam, ci[own], 1.0)
     rng = np.random.RandomState(10007)
     w = lam + rng.rand(n)
     state = np.zeros(n, dtype=np.int8)  # 0 undecided, 1 C, -1 F
-    # isolated points (no strong edges at all): F (smoother-only rows)
-    deg = np.diff(adj.indptr)
-    state[deg == 0] = -1
+    has_edge = np.zeros(n, dtype=bool)
+    has_edge[er] = True
+    state[~has_edge] = -1    # isolated points: F (smoother-only rows)
+    guard = 0
     while (state == 0).any():
+        guard += 1
+        if guard > 8 * (int(np.log2(n + 2)) + 8):
+            raise RuntimeError("PMIS failed to converge")
         und = state == 0
-        # local max among undecided neighbors
-        newC = []
-        for i in np.nonzero(und)[0]:
-            nb = adj.indices[adj.indptr[i]:adj.indptr[i + 1]]
-            nb = nb[state[nb] == 0]
-            if all(w[i] > w[j] for j in nb if j != i):
-                newC.append(i)
-        if not newC:
-            newC = [np.nonzero(und)[0][0]]
-        state[np.asarray(newC, dtype=np.int64)] = 1
-        # neighbors of new C become F
-        for i in newC:
-            nb = adj.indices[adj.indptr[i]:adj.indptr[i + 1]]
-            nb = nb[state[nb] == 0]
-            state[nb] = -1
+        act = und[er] & und[ec]
+        beaten = np.zeros(n, dtype=bool)
+        if act.any():
+            ar, ac = er[act], ec[act]
+            loses = (w[ac] > w[ar]) | ((w[ac] == w[ar]) & (ac > ar))
+            np.logical_or.at(beaten, ar, loses)
+        new_c = und & ~beaten
+        state[new_c] = 1
+        # undecided strong neighbors of new C become F
+        f_edge = (state[er] == 0) & (state[ec] == 1)
+        state[np.unique(er[f_edge])] = -1
     cf = np.full(n, -1, dtype=np.int32)
-    cpts = np.nonzero(state == 1)[0]
-    cf[cpts] = np.arange(cpts.size, dtype=np.int32)
-    return torch.from_numpy(cf), int(cpts.size)
+    c_rows = np.nonzero(state == 1)[0]
+    cf[c_rows] = np.arange(c_rows.size, dtype=np.int32)
+    return torch.from_numpy(cf), int(c_rows.size)
 
 
 def interp_d1(A, S, cf_map, num_coarse):
