@@ -30,11 +30,12 @@ def compute_diag_index(A) -> torch.Tensor:
     ci = _np(A.col_indices).astype(np.int64)
     n = A.n_rows
     out = np.full(n, -1, dtype=np.int32)
-    for i in range(n):
-        s, e = ro[i], ro[i + 1]
-        hits = np.nonzero(ci[s:e] == i)[0]
-        if hits.size:
-            out[i] = s + hits[0]
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    hit = rows == ci
+    # first hit wins (canonical CSRs have a unique diagonal entry):
+    # reversed assignment keeps the FIRST index on duplicates
+    idx = np.nonzero(hit)[0][::-1]
+    out[rows[idx]] = idx.astype(np.int32)
     return torch.from_numpy(out)
 
 
@@ -499,12 +500,17 @@ def transpose_block_index(A) -> np.ndarray:
     ci = _np(A.col_indices).astype(np.int64)
     n = A.n_rows
     rows = np.repeat(np.arange(n), np.diff(ro))
-    out = np.full(ci.size, -1, dtype=np.int64)
-    pos = {}
-    for k in range(ci.size):
-        pos[(rows[k], ci[k])] = k
-    for k in range(ci.size):
-        out[k] = pos.get((ci[k], rows[k]), -1)
+    # find index of (j, i) for each nz (i, j) by key search on sorted keys
+    ncols = max(int(A.n_cols), n)
+    keys = rows * ncols + ci
+    order = np.argsort(keys, kind="stable")
+    skeys = keys[order]
+    valid = ci < n                      # halo columns have no transpose row
+    want = np.where(valid, ci * ncols + rows, -1)
+    pos = np.searchsorted(skeys, want)
+    pos = np.minimum(pos, skeys.size - 1)
+    found = valid & (skeys[pos] == want)
+    out = np.where(found, order[pos], -1).astype(np.int64)
     return out
 
 
