@@ -208,4 +208,9 @@ class ChebyshevPolySolver(ChebyshevSolver):
 @register_solver("KPZ_POLYNOMIAL")
 class PolynomialSolver(ChebyshevPolySolver):
     """Polynomial smoother family; maps to the Chebyshev polynomial engine
-    (reference src/solvers/polynomial_solver.cu)."""
+    (reference src/solvers/polynomial_solver.cu, kpz_order param)."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        if scope.has("kpz_order"):
+            self.order = int(scope.get("kpz_order"))
