@@ -33,95 +33,109 @@ from ..distributed.manager import (DistributedManager, HaloExchange,
                                    exchange_csr_rows)
 
 
-def _hash01(gid: np.ndarray) -> np.ndarray:
+def _lsr(x: torch.Tensor, k: int) -> torch.Tensor:
+    """Logical right shift on int64 (torch >> is arithmetic)."""
+    return (x >> k) & ((1 << (64 - k)) - 1)
+
+
+def _hash01(gid: torch.Tensor) -> torch.Tensor:
     """Deterministic partition-invariant pseudo-random weight in [0,1) from a
-    global row id (reference: hashed random weights in pmis.cu)."""
-    x = gid.astype(np.uint64, copy=True)
-    x = (x ^ (x >> np.uint64(33))) * np.uint64(0xFF51AFD7ED558CCD)
-    x = (x ^ (x >> np.uint64(33))) * np.uint64(0xC4CEB9FE1A85EC53)
-    x = x ^ (x >> np.uint64(33))
-    return (x >> np.uint64(11)).astype(np.float64) / float(1 << 53)
+    global row id (reference: hashed random weights in pmis.cu). int64
+    wrapping arithmetic reproduces the uint64 mix bit-exactly."""
+    x = gid.to(torch.int64)
+    M1 = -47286287463422404   # 0xFF51AFD7ED558CCD as int64
+    M2 = -4265267296055464877  # 0xC4CEB9FE1A85EC53 as int64
+    x = (x ^ _lsr(x, 33)) * M1
+    x = (x ^ _lsr(x, 33)) * M2
+    x = x ^ _lsr(x, 33)
+    return _lsr(x, 11).to(torch.float64) / float(1 << 53)
 
 
 def strength_dist(A, mgr: DistributedManager, theta: float,
                   max_row_sum: float):
-    """Return (strong_out, strong_union) boolean masks over A's entries.
+    """Return (strong_out, strong_union) boolean masks over A's entries
+    (torch, on A's device — the whole computation stays GPU-resident apart
+    from the halo exchange).
     strong_out: |a_ij| >= theta * rowmax_i (the serial AHAT criterion on
     owned rows). strong_union additionally includes incoming strength
     |a_ij| >= theta * rowmax_j using the exchanged halo rowmax."""
-    ro = A.row_offsets.cpu().numpy().astype(np.int64)
-    ci = A.col_indices.cpu().numpy().astype(np.int64)
-    v = A.values.cpu().numpy().astype(np.float64).reshape(-1)
+    dev = A.row_offsets.device
+    ro = A.row_offsets.to(torch.int64)
+    ci = A.col_indices.to(torch.int64)
+    v = A.values.reshape(A.nnz, -1)[:, 0].to(torch.float64)
     n = A.n_rows
-    rows = np.repeat(np.arange(n), np.diff(ro))
+    rows = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int64, device=dev), ro[1:] - ro[:-1])
     off = rows != ci
-    absv = np.abs(v)
-    rowmax = np.zeros(n)
-    np.maximum.at(rowmax, rows[off], absv[off])
+    absv = v.abs()
+    rowmax = torch.zeros(n, dtype=torch.float64, device=dev)
+    rowmax.scatter_reduce_(0, rows[off], absv[off], reduce="amax")
     # halo rowmax by one exchange
     rm_ext = mgr.new_ext_vec(torch.float64)
-    rm_ext[:n] = torch.from_numpy(rowmax)
+    rm_ext[:n] = rowmax
     mgr.exchange_halo(rm_ext, block_override=1)
-    rm_all = rm_ext.cpu().numpy()
     strong_out = off & (absv >= theta * rowmax[rows]) & (rowmax[rows] > 0)
-    strong_in = off & (absv >= theta * rm_all[ci]) & (rm_all[ci] > 0)
+    strong_in = off & (absv >= theta * rm_ext[ci]) & (rm_ext[ci] > 0)
     if max_row_sum < 1.0:
-        rs = np.zeros(n)
-        np.add.at(rs, rows, v)
-        d = np.zeros(n)
+        rs = torch.zeros(n, dtype=torch.float64, device=dev)
+        rs.index_add_(0, rows, v)
+        d = torch.zeros(n, dtype=torch.float64, device=dev)
         diag_mask = rows == ci
-        d[rows[diag_mask]] = np.abs(v[diag_mask])
-        weak_rows = np.abs(rs) > max_row_sum * np.where(d > 0, d, 1.0)
+        d[rows[diag_mask]] = absv[diag_mask]
+        weak_rows = rs.abs() > max_row_sum * torch.where(
+            d > 0, d, torch.ones_like(d))
         strong_out &= ~weak_rows[rows]
         strong_in &= ~weak_rows[rows]
     return strong_out, strong_out | strong_in
 
 
-def pmis_dist(A, mgr: DistributedManager, strong_union: np.ndarray):
+def pmis_dist(A, mgr: DistributedManager, strong_union: torch.Tensor):
     """Distributed PMIS: Luby-style independent-set rounds over the strong
-    graph with per-round halo state exchange. Returns (cf_local int32 with
-    the local coarse index for C rows / -1 for F, nc_local)."""
-    ro = A.row_offsets.cpu().numpy().astype(np.int64)
-    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    graph with per-round halo state exchange, device-resident torch ops.
+    Returns (cf_local int32 tensor on A's device with the local coarse index
+    for C rows / -1 for F, nc_local)."""
+    dev = A.row_offsets.device
+    ro = A.row_offsets.to(torch.int64)
+    ci = A.col_indices.to(torch.int64)
     n = A.n_rows
-    rows = np.repeat(np.arange(n), np.diff(ro))
-    er, ec = rows[strong_union], ci[strong_union]   # edges incident to my rows
+    rows = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int64, device=dev), ro[1:] - ro[:-1])
+    er, ec = rows[strong_union], ci[strong_union]  # edges incident to my rows
 
     # lam_i = number of strong incoming edges (S^T row count); contributions
     # to halo columns are summed into their owners
     lam_ext = mgr.new_ext_vec(torch.float64)
-    lam_host = np.zeros(mgr.n_local + mgr.n_halo)
-    np.add.at(lam_host, ec, 1.0)
-    lam_ext.copy_(torch.from_numpy(lam_host).to(lam_ext.device))
+    lam_ext.index_add_(0, ec, torch.ones_like(ec, dtype=torch.float64))
     mgr.add_from_halo(lam_ext, block_override=1)
 
     # partition-invariant tie-broken weights
-    gid_own = (mgr.row_perm.cpu().numpy() + mgr.row_start)
+    gid_own = mgr.row_perm.to(torch.int64) + mgr.row_start
     w_ext = mgr.new_ext_vec(torch.float64)
-    w_ext[:n] = lam_ext[:n] + torch.from_numpy(_hash01(gid_own)).to(w_ext.dtype)
+    w_ext[:n] = lam_ext[:n] + _hash01(gid_own)
     mgr.exchange_halo(w_ext, block_override=1)
-    w = w_ext.cpu().numpy()
-    gid_ext = (np.concatenate([gid_own, mgr.halo_global]) if mgr.n_halo
-               else gid_own)
+    w = w_ext
+    if mgr.n_halo:
+        gid_ext = torch.cat([gid_own, torch.from_numpy(
+            mgr.halo_global.astype(np.int64)).to(dev)])
+    else:
+        gid_ext = gid_own
 
-    state_ext = mgr.new_ext_vec(torch.float64)   # 0 undec, 1 C, -1 F
-    st = np.zeros(mgr.n_local + mgr.n_halo)
+    st = mgr.new_ext_vec(torch.float64)          # 0 undec, 1 C, -1 F
     # isolated rows (no strong edges): F
-    has_edge = np.zeros(n, dtype=bool)
+    has_edge = torch.zeros(n, dtype=torch.bool, device=dev)
     has_edge[er] = True
-    st[:n][~has_edge] = -1.0
+    st[:n] = torch.where(has_edge, torch.zeros(n, device=dev,
+                                               dtype=torch.float64), -1.0)
 
     def _sync_state():
-        state_ext[:mgr.n_local + mgr.n_halo] = torch.from_numpy(st) \
-            .to(state_ext.dtype)
-        mgr.exchange_halo(state_ext, block_override=1)
-        st[mgr.n_local:] = state_ext[mgr.n_local:].cpu().numpy()
+        mgr.exchange_halo(st, block_override=1)
 
     _sync_state()
     guard = 0
+    ones_b = torch.ones(1, dtype=torch.float64, device=dev)
     while True:
-        undec_local = int((st[:n] == 0).sum())
-        total = mgr.global_sum(float(undec_local))
+        und_row = st[:n] == 0
+        total = mgr.global_sum(float(und_row.sum().item()))
         if total == 0:
             break
         guard += 1
@@ -129,43 +143,46 @@ def pmis_dist(A, mgr: DistributedManager, strong_union: np.ndarray):
             raise RuntimeError("distributed PMIS failed to converge")
         # a row is new-C if undecided and a strict (w, gid) local max among
         # its undecided strong neighbors (total order -> guaranteed progress)
-        und_row = st[:n] == 0
         e_act = und_row[er] & (st[ec] == 0) & (er != ec)
-        beaten = np.zeros(n, dtype=bool)
-        if e_act.any():
+        beaten = torch.zeros(n, dtype=torch.float64, device=dev)
+        if bool(e_act.any()):
             ea_r, ea_c = er[e_act], ec[e_act]
-            wr, wc = w[ea_r], w[ea_c]
-            gr, gc = gid_ext[ea_r], gid_ext[ea_c]
-            loses = (wc > wr) | ((wc == wr) & (gc > gr))
-            np.logical_or.at(beaten, ea_r, loses)
-        new_c = und_row & ~beaten
-        st[:n][new_c] = 1.0
+            loses = (w[ea_c] > w[ea_r]) | ((w[ea_c] == w[ea_r])
+                                           & (gid_ext[ea_c] > gid_ext[ea_r]))
+            beaten.index_add_(0, ea_r, loses.to(torch.float64))
+        new_c = und_row & (beaten == 0)
+        st[:n] = torch.where(new_c, ones_b, st[:n])
         _sync_state()
         # undecided rows with a C strong neighbor -> F
-        e_f = (st[:n][er] == 0) & (st[ec] == 1.0)
-        st[:n][np.unique(er[e_f])] = -1.0
+        e_f = (st[er] == 0) & (st[ec] == 1.0)
+        st[er[e_f]] = -1.0
         _sync_state()
 
-    cf = np.full(n, -1, dtype=np.int32)
-    c_rows = np.nonzero(st[:n] == 1.0)[0]
-    cf[c_rows] = np.arange(c_rows.size, dtype=np.int32)
-    return cf, int(c_rows.size)
+    c_mask = st[:n] == 1.0
+    cf = torch.full((n,), -1, dtype=torch.int32, device=dev)
+    csum = torch.cumsum(c_mask.to(torch.int64), 0)
+    cf[c_mask] = (csum[c_mask] - 1).to(torch.int32)
+    return cf, int(c_mask.sum().item())
 
 
-def coarse_numbering(mgr: DistributedManager, cf: np.ndarray, nc_local: int):
+def coarse_numbering(mgr: DistributedManager, cf: torch.Tensor,
+                     nc_local: int):
     """Global coarse ids: owned C rows numbered coarse_start + local index;
     halo columns' coarse ids fetched by one halo exchange. Returns
-    (cf_ext int64 len n_cols with global coarse id or -1, coarse_offsets)."""
+    (cf_ext int32 device tensor, len n_cols, global coarse id or -1;
+    coarse_offsets numpy)."""
     counts = [None] * mgr.world
     tdist.all_gather_object(counts, nc_local)
     coarse_offs = np.zeros(mgr.world + 1, dtype=np.int64)
     coarse_offs[1:] = np.cumsum([int(c) for c in counts])
+    assert int(coarse_offs[-1]) < 2 ** 31, "global coarse ids exceed int32"
     cs = int(coarse_offs[mgr.rank])
     gc_ext = mgr.new_ext_vec(torch.float64)
-    own = np.where(cf >= 0, cf.astype(np.float64) + cs, -1.0)
-    gc_ext[:mgr.n_local] = torch.from_numpy(own).to(gc_ext.dtype)
+    cff = cf.to(torch.float64)
+    gc_ext[:mgr.n_local] = torch.where(cff >= 0, cff + cs,
+                                       torch.full_like(cff, -1.0))
     mgr.exchange_halo(gc_ext, block_override=1)
-    cf_ext = gc_ext.cpu().numpy().round().astype(np.int64)
+    cf_ext = gc_ext.round().to(torch.int32)
     return cf_ext, coarse_offs
 
 

@@ -196,7 +196,7 @@ class ClassicalLevel(AMGLevel):
             self._strong_out, strong_union = strength_dist(
                 self.A, mgr, theta, mrs)
             cf, nc = pmis_dist(self.A, mgr, strong_union)
-            self.cf_map = torch.from_numpy(cf)
+            self.cf_map = cf          # device tensor
             self.num_coarse = nc
             self.A._cache["cf_map"] = self.cf_map
             return nc
@@ -271,15 +271,16 @@ class ClassicalLevel(AMGLevel):
         from .classical_dist import (ClassicalDistOperators,
                                      coarse_numbering, rap_dist)
         from ..distributed.manager import DistributedManager
-        from ..ops import cpu as cpu_ops
         A = self.A
-        cf = self.cf_map.numpy()
-        cf_ext, coarse_offs = coarse_numbering(mgr, cf, self.num_coarse)
-        A_host = CSRMatrix(A.row_offsets.cpu(), A.col_indices.cpu(),
-                           A.values.cpu(), n_cols=A.n_cols)
-        P_m = cpu_ops.interp_d1(A_host, torch.from_numpy(self._strong_out),
-                                torch.from_numpy(cf_ext),
-                                int(coarse_offs[-1]))
+        cf_ext, coarse_offs = coarse_numbering(mgr, self.cf_map,
+                                               self.num_coarse)
+        # D1 interpolation onto GLOBAL coarse columns, on A's device (the
+        # gfx950 interp kernels take the ext-length cf array directly)
+        S_dev = self._strong_out
+        if A.row_offsets.is_cuda:
+            S_dev = S_dev.to(torch.uint8)
+        P_m = ops._backend(A).interp_d1(A, S_dev, cf_ext,
+                                        int(coarse_offs[-1]))
         tf = float(self.scope.get("interp_truncation_factor"))
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:

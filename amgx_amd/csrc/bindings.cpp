@@ -486,7 +486,8 @@ std::tuple<Tensor, Tensor, Tensor> interp_d1(Tensor ro, Tensor ci, Tensor va,
             ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
             strong.data_ptr<unsigned char>(), cf.data_ptr<int>(),
             didx.data_ptr<int>(), p_ro.data_ptr<int>(), n,
-            p_ci.data_ptr<int>(), p_va.data_ptr<scalar_t>(), cur_stream());
+            (int)cf.numel(), p_ci.data_ptr<int>(),
+            p_va.data_ptr<scalar_t>(), cur_stream());
     });
     return {p_ro, p_ci, p_va};
 }
@@ -496,8 +497,8 @@ Tensor interp_d1_count(Tensor ro, Tensor ci, Tensor strong, Tensor cf) {
     auto counts = torch::empty({n}, ro.options());
     amgx_hip::interp_d1_count(ro.data_ptr<int>(), ci.data_ptr<int>(),
                               strong.data_ptr<unsigned char>(),
-                              cf.data_ptr<int>(), n, counts.data_ptr<int>(),
-                              cur_stream());
+                              cf.data_ptr<int>(), n, (int)cf.numel(),
+                              counts.data_ptr<int>(), cur_stream());
     return counts;
 }
 

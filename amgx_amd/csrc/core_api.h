@@ -164,11 +164,13 @@ void pmis_one_round(const int* ro, const int* ci, const int* tidx,
                     const signed char* state, signed char* state_mid,
                     signed char* state_out, int* n_undecided, hipStream_t s);
 void interp_d1_count(const int* ro, const int* ci, const unsigned char* strong,
-                     const int* cf, int n, int* counts, hipStream_t s);
+                     const int* cf, int n, int ncols, int* counts,
+                     hipStream_t s);
 template <typename T>
 void interp_d1(const int* ro, const int* ci, const T* va,
                const unsigned char* strong, const int* cf, const int* didx,
-               const int* p_ro, int n, int* p_ci, T* p_va, hipStream_t s);
+               const int* p_ro, int n, int ncols, int* p_ci, T* p_va,
+               hipStream_t s);
 
 // ---- ILU(0) (kernels_classical.hip) ------------------------------------------
 template <typename T>

@@ -175,17 +175,19 @@ void pmis_one_round(const int* ro, const int* ci, const int* tidx,
 
 // ============================================================ D1 interpolation
 // count pass: C rows -> 1 entry; F rows -> #strong C neighbors
+// cf spans ncols entries (= n for single-process, n_local+n_halo with
+// GLOBAL coarse ids for the distributed path)
 __global__ void d1_count(const int* __restrict__ ro,
                          const int* __restrict__ ci,
                          const unsigned char* __restrict__ strong,
-                         const int* __restrict__ cf, int n,
+                         const int* __restrict__ cf, int n, int ncols,
                          int* __restrict__ counts) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     if (cf[i] >= 0) { counts[i] = 1; return; }
     int c = 0;
     for (int k = ro[i]; k < ro[i + 1]; ++k)
-        if (strong[k] && ci[k] < n && cf[ci[k]] >= 0) ++c;
+        if (strong[k] && ci[k] < ncols && cf[ci[k]] >= 0) ++c;
     counts[i] = c;
 }
 
@@ -198,7 +200,7 @@ __global__ void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
                         const T* __restrict__ va,
                         const unsigned char* __restrict__ strong,
                         const int* __restrict__ cf, const int* __restrict__ didx,
-                        const int* __restrict__ p_ro, int n,
+                        const int* __restrict__ p_ro, int n, int ncols,
                         int* __restrict__ p_ci, T* __restrict__ p_va) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -216,14 +218,14 @@ __global__ void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
         if (ci[k] == i) continue;
         double a = (double)va[k];
         if (a < 0) neg_all += a; else pos_all += a;
-        if (strong[k] && ci[k] < n && cf[ci[k]] >= 0) {
+        if (strong[k] && ci[k] < ncols && cf[ci[k]] >= 0) {
             if (a < 0) neg_c += a; else pos_c += a;
         }
     }
     if (diag == 0.0) {
         // degenerate row: emit zero weights for its counted slots
         for (int k = s; k < e; ++k) {
-            if (ci[k] == i || !strong[k] || ci[k] >= n || cf[ci[k]] < 0)
+            if (ci[k] == i || !strong[k] || ci[k] >= ncols || cf[ci[k]] < 0)
                 continue;
             p_ci[out] = cf[ci[k]];
             p_va[out] = T(0);
@@ -235,7 +237,8 @@ __global__ void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
     double alpha = neg_c != 0.0 ? neg_all / neg_c : 0.0;
     double beta = pos_c != 0.0 ? pos_all / pos_c : 0.0;
     for (int k = s; k < e; ++k) {
-        if (ci[k] == i || !strong[k] || ci[k] >= n || cf[ci[k]] < 0) continue;
+        if (ci[k] == i || !strong[k] || ci[k] >= ncols || cf[ci[k]] < 0)
+            continue;
         double a = (double)va[k];
         p_ci[out] = cf[ci[k]];
         p_va[out] = (T)(-(a < 0 ? alpha : beta) * a / diag);
@@ -246,15 +249,18 @@ __global__ void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
 template <typename T>
 void interp_d1(const int* ro, const int* ci, const T* va,
                const unsigned char* strong, const int* cf, const int* didx,
-               const int* p_ro, int n, int* p_ci, T* p_va, hipStream_t s) {
+               const int* p_ro, int n, int ncols, int* p_ci, T* p_va,
+               hipStream_t s) {
     hipLaunchKernelGGL((d1_fill<T>), dim3(grid_1d(n)), dim3(AMGX_BLOCK), 0, s,
-                       ro, ci, va, strong, cf, didx, p_ro, n, p_ci, p_va);
+                       ro, ci, va, strong, cf, didx, p_ro, n, ncols, p_ci,
+                       p_va);
 }
 
 void interp_d1_count(const int* ro, const int* ci, const unsigned char* strong,
-                     const int* cf, int n, int* counts, hipStream_t s) {
+                     const int* cf, int n, int ncols, int* counts,
+                     hipStream_t s) {
     hipLaunchKernelGGL(d1_count, dim3(grid_1d(n)), dim3(AMGX_BLOCK), 0, s, ro,
-                       ci, strong, cf, n, counts);
+                       ci, strong, cf, n, ncols, counts);
 }
 
 // ============================================================ ILU(0)
@@ -384,7 +390,7 @@ void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
                                    unsigned char*, hipStream_t);               \
     template void interp_d1<T>(const int*, const int*, const T*,               \
                                const unsigned char*, const int*, const int*,   \
-                               const int*, int, int*, T*, hipStream_t);        \
+                               const int*, int, int, int*, T*, hipStream_t);   \
     template void ilu0_factor_color_launch<T>(const int*, const int*,          \
                                               const int*, const int*,          \
                                               const int*, int, T*, int,        \
