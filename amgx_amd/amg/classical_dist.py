@@ -186,42 +186,54 @@ def coarse_numbering(mgr: DistributedManager, cf: torch.Tensor,
     return cf_ext, coarse_offs
 
 
-def rap_dist(A, mgr: DistributedManager, P_own: sp.csr_matrix,
+def rap_dist(A, mgr: DistributedManager, P_own,
              coarse_offs: np.ndarray):
-    """Distributed Galerkin RAP. P_own: owned fine rows x GLOBAL coarse cols.
-    Exchanges halo rows of P, forms P_own^T (A_loc P_ext) and ships external
-    coarse-row contributions to their owners. Returns the owned coarse rows
-    as a scipy CSR with global coarse columns."""
+    """Distributed Galerkin RAP on the backend primitives (device-resident
+    when A is on GPU). P_own: CSRMatrix, owned fine rows x GLOBAL coarse
+    cols. Exchanges halo rows of P, forms P_own^T (A_loc P_ext) via the
+    SpGEMM/transpose kernels, and ships external coarse-row contributions to
+    their owners. Returns a CSRMatrix of the owned coarse rows with global
+    coarse columns."""
     n, n_ext = mgr.n_local, mgr.n_local + mgr.n_halo
     ngc = int(coarse_offs[-1])
-    # halo rows of P (reference exchange_halo_rows_P)
-    halo_rows = exchange_csr_rows(mgr, P_own.indptr, P_own.indices,
-                                  P_own.data)
-    hr, hc, hv = [], [], []
-    for pos, rowdat in enumerate(halo_rows):
-        if rowdat is None:
-            continue
-        cols, vals = rowdat
-        hr.append(np.full(cols.size, n + pos, dtype=np.int64))
-        hc.append(cols)
-        hv.append(vals)
-    if hr:
-        halo_part = sp.csr_matrix(
-            (np.concatenate(hv), (np.concatenate(hr), np.concatenate(hc))),
-            shape=(n_ext, ngc))
-    else:
-        halo_part = sp.csr_matrix((n_ext, ngc))
-    P_ext = sp.vstack([P_own, sp.csr_matrix((mgr.n_halo, ngc))]).tocsr() \
-        + halo_part
-    A_loc = sp.csr_matrix(
-        (A.values.cpu().numpy().astype(np.float64).reshape(-1),
-         A.col_indices.cpu().numpy().astype(np.int64),
-         A.row_offsets.cpu().numpy().astype(np.int64)), shape=(n, n_ext))
-    T = A_loc @ P_ext                      # owned fine x global coarse
-    C = (P_own.T @ T).tocsr()              # global coarse x global coarse
-    C.sum_duplicates()
+    from .. import ops as O
+    from ..matrix import CSRMatrix
+    dev = A.row_offsets.device
+    # halo rows of P (reference exchange_halo_rows_P): payload transfer is
+    # host-staged (boundary-sized), products run on A's device
+    ro_np = P_own.row_offsets.cpu().numpy().astype(np.int64)
+    ci_np = P_own.col_indices.cpu().numpy().astype(np.int64)
+    va_np = P_own.values.cpu().numpy().astype(np.float64)
+    halo_rows = exchange_csr_rows(mgr, ro_np, ci_np, va_np)
+    h_counts = np.asarray([0 if r is None else r[0].size
+                           for r in halo_rows], dtype=np.int64)
+    h_cols = (np.concatenate([r[0] for r in halo_rows if r is not None])
+              if h_counts.sum() else np.zeros(0, dtype=np.int64))
+    h_vals = (np.concatenate([r[1] for r in halo_rows if r is not None])
+              if h_counts.sum() else np.zeros(0))
+    nnz_p = int(ro_np[-1])
+    ext_ro = np.concatenate([ro_np, nnz_p + np.cumsum(h_counts)])
+    P_ext = CSRMatrix(
+        torch.from_numpy(ext_ro.astype(np.int32)).to(dev),
+        torch.cat([P_own.col_indices.to(torch.int32),
+                   torch.from_numpy(h_cols.astype(np.int32)).to(dev)]),
+        torch.cat([P_own.values.reshape(-1),
+                   torch.from_numpy(h_vals).to(A.dtype).to(dev)]),
+        n_cols=ngc)
+    T = O.spgemm(A, P_ext)                 # owned fine x global coarse
+    Pt = O.transpose(P_own)                # global coarse x owned fine
+    C = O.spgemm(Pt, T)                    # global coarse x global coarse
     cs, ce = int(coarse_offs[mgr.rank]), int(coarse_offs[mgr.rank + 1])
-    mine = C[cs:ce]
+    ro_C = C.row_offsets.to(torch.int64)
+
+    def _slice(lo, hi):
+        s, e = int(ro_C[lo].item()), int(ro_C[hi].item())
+        rows = torch.repeat_interleave(
+            torch.arange(hi - lo, dtype=torch.int64, device=dev),
+            ro_C[lo + 1:hi + 1] - ro_C[lo:hi])
+        return (rows, C.col_indices[s:e].to(torch.int64),
+                C.values.reshape(-1)[s:e])
+
     # external rows -> owners (reference exchange_RAP_ext + sparse_add)
     frags = []
     for r in range(mgr.world):
@@ -229,25 +241,39 @@ def rap_dist(A, mgr: DistributedManager, P_own: sp.csr_matrix,
             frags.append(None)
             continue
         lo, hi = int(coarse_offs[r]), int(coarse_offs[r + 1])
-        block = C[lo:hi]
-        frags.append((block.indptr, block.indices, block.data)
-                     if block.nnz else None)
+        rr, cc, vv = _slice(lo, hi)
+        frags.append((rr.cpu().numpy(), cc.cpu().numpy(),
+                      vv.cpu().numpy()) if rr.numel() else None)
     gathered = [None] * mgr.world
     tdist.all_gather_object(gathered, frags)
-    acc = mine
-    add = sp.csr_matrix(acc.shape)
+    rows_l, cols_l, vals_l = _slice(cs, ce)
+    parts_r, parts_c, parts_v = [rows_l], [cols_l], [vals_l]
     for r in range(mgr.world):
         if r == mgr.rank or gathered[r] is None:
             continue
         frag = gathered[r][mgr.rank]
         if frag is None:
             continue
-        indptr, indices, data = frag
-        add = add + sp.csr_matrix((data, indices, indptr), shape=acc.shape)
-    out = (acc + add).tocsr()
-    out.sum_duplicates()
-    out.eliminate_zeros()
-    return out
+        rr, cc, vv = frag
+        parts_r.append(torch.from_numpy(rr).to(dev))
+        parts_c.append(torch.from_numpy(cc).to(dev))
+        parts_v.append(torch.from_numpy(vv).to(A.dtype).to(dev))
+    rows = torch.cat(parts_r)
+    cols = torch.cat(parts_c)
+    vals = torch.cat(parts_v)
+    nc_local = ce - cs
+    # dedupe/sum by (row, col) key — device sort, no scipy
+    key = rows * ngc + cols
+    uk, inv = torch.unique(key, return_inverse=True)
+    vsum = torch.zeros(uk.numel(), dtype=vals.dtype, device=dev)
+    vsum.index_add_(0, inv, vals)
+    rows_out = (uk // ngc)
+    cols_out = (uk % ngc).to(torch.int32)
+    ro_out = torch.zeros(nc_local + 1, dtype=torch.int64, device=dev)
+    cnt = torch.bincount(rows_out, minlength=nc_local)
+    torch.cumsum(cnt, 0, out=ro_out[1:])
+    return CSRMatrix(ro_out.to(torch.int32), cols_out.contiguous(),
+                     vsum.contiguous(), n_cols=ngc)
 
 
 class ClassicalDistOperators:
@@ -256,30 +282,29 @@ class ClassicalDistOperators:
     HaloExchange that completes P^T r (scatter-add to owners) and P xc
     (gather of halo coarse values)."""
 
-    def __init__(self, A, mgr, P_own: sp.csr_matrix, Ac, coarse_offs):
+    def __init__(self, A, mgr, P_own, Ac, coarse_offs):
         from ..matrix import CSRMatrix
         mgr_c = Ac.manager
+        dev = A.row_offsets.device
         cs, ce = int(coarse_offs[mgr.rank]), int(coarse_offs[mgr.rank + 1])
         nc_local = ce - cs
-        cols = P_own.indices.astype(np.int64)
+        cols = P_own.col_indices.to(torch.int64)
         own_mask = (cols >= cs) & (cols < ce)
-        halo_gids = np.unique(cols[~own_mask])
-        iperm_c = mgr_c.row_iperm.cpu().numpy().astype(np.int64)
-        self.halomap = HaloExchange(halo_gids, coarse_offs,
-                                    device=A.device,
-                                    owner_local_map=iperm_c)
-        new_cols = np.empty_like(cols)
+        halo_gids_t = torch.unique(cols[~own_mask])      # sorted
+        self.halomap = HaloExchange(
+            halo_gids_t.cpu().numpy(), coarse_offs, device=dev,
+            owner_local_map=mgr_c.row_iperm.cpu().numpy().astype(np.int64))
+        iperm_c = mgr_c.row_iperm.to(torch.int64)
+        new_cols = torch.empty_like(cols)
         new_cols[own_mask] = iperm_c[cols[own_mask] - cs]
-        new_cols[~own_mask] = nc_local + np.searchsorted(halo_gids,
-                                                         cols[~own_mask])
+        new_cols[~own_mask] = nc_local + torch.searchsorted(
+            halo_gids_t, cols[~own_mask])
         self.nc_local = nc_local
-        self.n_halo_p = int(halo_gids.size)
-        dev = A.device
+        self.n_halo_p = int(halo_gids_t.numel())
         self.P = CSRMatrix(
-            torch.from_numpy(P_own.indptr.astype(np.int32)).to(dev),
-            torch.from_numpy(new_cols.astype(np.int32)).to(dev),
-            torch.from_numpy(P_own.data.astype(np.float64)).to(A.dtype)
-            .to(dev),
+            P_own.row_offsets.to(torch.int32).contiguous(),
+            new_cols.to(torch.int32).contiguous(),
+            P_own.values.reshape(-1).to(A.dtype).contiguous(),
             n_cols=nc_local + self.n_halo_p)
         from .. import ops
         self.R = ops.transpose(self.P)

@@ -285,17 +285,12 @@ class ClassicalLevel(AMGLevel):
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:
             P_m = ops.truncate_rows(P_m, tf, me)
-        P_own = sp.csr_matrix(
-            (P_m.values.cpu().numpy().astype(np.float64),
-             P_m.col_indices.cpu().numpy().astype(np.int64),
-             P_m.row_offsets.cpu().numpy().astype(np.int64)),
-            shape=(A.n_rows, int(coarse_offs[-1])))
-        rap = rap_dist(A, mgr, P_own, coarse_offs)
+        rap = rap_dist(A, mgr, P_m, coarse_offs)
         Ac = DistributedManager.upload_global_csr(
-            rap.indptr, rap.indices, rap.data, self.num_coarse,
+            rap.row_offsets, rap.col_indices, rap.values, self.num_coarse,
             int(coarse_offs[mgr.rank]), int(coarse_offs[-1]),
-            device=A.device, block_dim=1, dtype=A.dtype)
-        self._dist_ops = ClassicalDistOperators(A, mgr, P_own, Ac,
+            device=A.row_offsets.device, block_dim=1, dtype=A.dtype)
+        self._dist_ops = ClassicalDistOperators(A, mgr, P_m, Ac,
                                                 coarse_offs)
         return Ac
 
