@@ -204,19 +204,54 @@ def gs_smooth_color(A, dinv, b, x, color_rows, omega: float):
 
 # ---------------------------------------------------------------------- coloring
 def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0):
-    """Sequential greedy distance-1 coloring (host reference;
-    GPU uses the parallel min-max kernels — colors differ, validity tested)."""
+    """Vectorized Jones-Plassmann greedy rounds (host analogue of the gfx950
+    MIN_MAX/greedy kernel, src/matrix_coloring/min_max.cu): each round the
+    hash-maxima of the uncolored subgraph take the SMALLEST color unused by
+    their already-colored neighbors — greedy color counts, parallel rounds."""
     ro = _np(A.row_offsets).astype(np.int64)
     ci = _np(A.col_indices).astype(np.int64)
     n = A.n_rows
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    keep = (rows != ci) & (ci < n)
+    er, ec = rows[keep], ci[keep]
+    # tie-free weights: bijective uint64 mix of the row id
+    x = (np.arange(n, dtype=np.uint64)
+         + np.uint64((seed * 0x9E3779B97F4A7C15) % (1 << 64)))
+    x = x * np.uint64(0xBF58476D1CE4E5B9)
+    w = x ^ (x >> np.uint64(31))
     colors = np.full(n, -1, dtype=np.int32)
-    for i in range(n):
-        nb = ci[ro[i]:ro[i + 1]]
-        used = set(colors[j] for j in nb if j != i and j < n and colors[j] >= 0)
-        c = 0
-        while c in used:
-            c += 1
-        colors[i] = c
+    isolated = np.ones(n, dtype=bool)
+    isolated[er] = False
+    colors[isolated] = 0
+    guard = 0
+    while (colors < 0).any():
+        guard += 1
+        if guard > n + 2:
+            raise RuntimeError("coloring did not converge")
+        un = colors < 0
+        act = un[er] & un[ec]
+        beaten = np.zeros(n, dtype=bool)
+        np.logical_or.at(beaten, er[act], w[ec[act]] > w[er[act]])
+        cand = un & ~beaten
+        # smallest color not used by already-colored neighbors (bitmask <64;
+        # fallback past the mask like the GPU kernel)
+        nb_colored = colors[ec] >= 0
+        used = np.zeros(n, dtype=np.uint64)
+        sel = nb_colored & cand[er]
+        np.bitwise_or.at(used, er[sel],
+                         np.uint64(1) << colors[ec[sel]].astype(np.uint64))
+        free = (~used).astype(np.uint64)
+        # index of lowest set bit of `free`
+        low = (free & (~free + np.uint64(1)))
+        c_new = np.zeros(n, dtype=np.int32)
+        nz = low > 0
+        c_new[nz] = (np.log2(low[nz].astype(np.float64))).astype(np.int32)
+        big = cand & ~nz          # all 64 low colors taken (rare)
+        if big.any():
+            mx = np.zeros(n, dtype=np.int32)
+            np.maximum.at(mx, er[sel], colors[ec[sel]])
+            c_new[big] = mx[big] + 1
+        colors[cand] = c_new[cand]
     num = int(colors.max()) + 1 if n else 0
     return torch.from_numpy(colors), num
 
