@@ -290,51 +290,57 @@ def _block_norm_csr(A) -> sp.csr_matrix:
 
 def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
                    seed: int = 0):
-    w = _strength_weights(A)
+    """Handshaking pairwise matching (host analogue of the gfx950 SIZE_2
+    kernels): each round every unaggregated node proposes to its strongest
+    unaggregated neighbor (ties broken by a seeded random key), mutual
+    proposals merge; leftovers join their strongest aggregate or stay
+    singletons. Vectorized over the edge list."""
+    w = _strength_weights(A).tocoo()
     n = A.n_rows
+    er = w.row.astype(np.int64)
+    ec = w.col.astype(np.int64)
+    ew = w.data.astype(np.float64)
+    keep = er != ec
+    er, ec, ew = er[keep], ec[keep], ew[keep]
     agg = np.full(n, -1, dtype=np.int64)
     rng = np.random.RandomState(seed)
     tie = rng.rand(n)
     next_id = 0
     for _ in range(max_iterations):
-        un = np.nonzero(agg < 0)[0]
-        if un.size == 0:
+        if not (agg < 0).any():
             break
-        # each unaggregated node proposes to its strongest unaggregated neighbor
+        un = agg < 0
+        act = un[er] & un[ec]
+        if not act.any():
+            break
+        ar, ac, aw = er[act], ec[act], ew[act]
+        # per-row argmax of (weight, tie[j]): ascending lexsort then
+        # sequential overwrite — the last write per row is the max key
+        order = np.lexsort((tie[ac], aw, ar))
         prop = np.full(n, -1, dtype=np.int64)
-        for i in un:
-            s, e = w.indptr[i], w.indptr[i + 1]
-            best, bw = -1, 0.0
-            for k in range(s, e):
-                j = w.indices[k]
-                if agg[j] >= 0 or j == i:
-                    continue
-                wk = w.data[k]
-                if (wk > bw) or (wk == bw and best >= 0 and tie[j] > tie[best]):
-                    best, bw = j, wk
-            prop[i] = best
-        merged = False
-        for i in un:
-            j = prop[i]
-            if j >= 0 and prop[j] == i and i < j:
-                agg[i] = agg[j] = next_id
-                next_id += 1
-                merged = True
-        if not merged:
+        prop[ar[order]] = ac[order]
+        cand = np.nonzero(prop >= 0)[0]
+        mutual = cand[(prop[prop[cand]] == cand) & (cand < prop[cand])]
+        if mutual.size == 0:
             break
-    # leftovers: join strongest aggregated neighbor, else singleton
-    for i in np.nonzero(agg < 0)[0]:
-        s, e = w.indptr[i], w.indptr[i + 1]
-        best, bw = -1, 0.0
-        for k in range(s, e):
-            j = w.indices[k]
-            if agg[j] >= 0 and w.data[k] >= bw:
-                best, bw = j, w.data[k]
-        if best >= 0:
-            agg[i] = agg[best]
-        else:
-            agg[i] = next_id
-            next_id += 1
+        ids = next_id + np.arange(mutual.size)
+        agg[mutual] = ids
+        agg[prop[mutual]] = ids
+        next_id += int(mutual.size)
+    # leftovers: join the strongest aggregated neighbor, else singleton
+    left = agg < 0
+    if left.any():
+        sel = left[er] & (agg[ec] >= 0)
+        if sel.any():
+            sr, sc, sw = er[sel], ec[sel], ew[sel]
+            order = np.lexsort((np.arange(sr.size), sw, sr))
+            join = np.full(n, -1, dtype=np.int64)
+            join[sr[order]] = sc[order]
+            can_join = left & (join >= 0)
+            agg[can_join] = agg[join[can_join]]
+        singles = np.nonzero(agg < 0)[0]
+        agg[singles] = next_id + np.arange(singles.size)
+        next_id += int(singles.size)
     return torch.from_numpy(agg.astype(np.int32)), int(next_id)
 
 
