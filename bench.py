@@ -54,9 +54,69 @@ FGMRES_AGG = {
 }
 
 
-def build_local_matrix(size, rank, world, device):
+# Secondary measurable configs (BASELINE.json configs #3/#4/#5 shapes);
+# the driver contract always uses the default fgmres_agg.
+PCG_CLASSICAL = {
+    "config_version": 2,
+    "solver": {
+        "preconditioner": {
+            "algorithm": "CLASSICAL", "solver": "AMG",
+            "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+            "max_iters": 1, "min_coarse_rows": 32, "scope": "amg",
+            "cycle": "V"},
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-6, "norm": "L2"},
+}
+FGMRES_ILU0 = {
+    "config_version": 2,
+    "solver": {
+        "preconditioner": {"solver": "MULTICOLOR_ILU", "max_iters": 1,
+                           "scope": "ilu"},
+        "solver": "FGMRES", "max_iters": 200, "gmres_n_restart": 30,
+        "monitor_residual": 1, "convergence": "RELATIVE_INI",
+        "tolerance": 1e-6},
+}
+FGMRES_BLOCK4_DILU = {
+    "config_version": 2,
+    "solver": {
+        "preconditioner": {"solver": "MULTICOLOR_DILU", "max_iters": 2,
+                           "relaxation_factor": 1.0, "scope": "dilu"},
+        "solver": "FGMRES", "max_iters": 200, "gmres_n_restart": 30,
+        "monitor_residual": 1, "convergence": "RELATIVE_INI",
+        "tolerance": 1e-6},
+}
+CONFIGS = {
+    "fgmres_agg": (FGMRES_AGG, "poisson"),
+    "classical_pcg": (PCG_CLASSICAL, "poisson"),
+    "ilu0": (FGMRES_ILU0, "poisson"),
+    "block4_dilu": (FGMRES_BLOCK4_DILU, "block4"),
+}
+
+
+def build_local_matrix(size, rank, world, device, problem="poisson"):
     """Rank-local z-slab of the global size x size x (size*world) 7-pt
-    Poisson. For world=1 this is exactly the 256^3 single-GPU config."""
+    Poisson (or a block-4 coupled system for block4_dilu). For world=1 this
+    is exactly the 256^3 single-GPU config."""
+    if problem == "block4":
+        import numpy as np
+
+        from amgx_amd.problems import block_laplacian
+        side = max(int(round(size ** 1.5 / world ** 0.5)), 8)
+        Afull = block_laplacian(side, side * world, block_dim=4, seed=9)
+        if world == 1:
+            return Afull.to(device), None
+        from amgx_amd.distributed.manager import DistributedManager
+        n = Afull.n_rows
+        per = n // world
+        lo = rank * per
+        hi = n if rank == world - 1 else lo + per
+        ro = Afull.row_offsets.numpy().astype(np.int64)
+        s0, s1 = ro[lo], ro[hi]
+        A = DistributedManager.upload_global_csr(
+            ro[lo:hi + 1] - s0, Afull.col_indices.numpy()[s0:s1],
+            Afull.values.numpy().reshape(Afull.nnz, 16)[s0:s1],
+            hi - lo, lo, n, device=device, block_dim=4)
+        return A, A.manager
     if world == 1:
         from amgx_amd.problems import poisson_3d
         return poisson_3d(size, size, size, device=device), None
@@ -96,6 +156,8 @@ def main():
     ap.add_argument("--size", type=int, default=256,
                     help="per-rank subdomain edge (size^3 rows per GPU)")
     ap.add_argument("--device", default=None)
+    ap.add_argument("--config", default="fgmres_agg", choices=sorted(CONFIGS),
+                    help="solver composition (driver contract: fgmres_agg)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -115,9 +177,10 @@ def main():
         import torch.distributed as tdist
         dist = tdist
 
-    cfg = AMGConfig.from_dict(FGMRES_AGG)
-    A, manager = build_local_matrix(args.size, rank, world, device)
-    n_local = args.size ** 3
+    cfg_dict, problem = CONFIGS[args.config]
+    cfg = AMGConfig.from_dict(cfg_dict)
+    A, manager = build_local_matrix(args.size, rank, world, device, problem)
+    n_local = A.n_rows * A.block_dim
     g = torch.Generator().manual_seed(1234 + rank)
     b_user = torch.rand(n_local, generator=g, dtype=torch.float64).to(device)
     # distributed solves run on halo-extended vectors in internal ordering
@@ -150,7 +213,9 @@ def main():
     dof = args.size ** 3 * n_gpus
     if rank == 0:
         line = {
-            "metric": "amg_setup_solve_seconds_poisson256_fgmres_agg",
+            "metric": ("amg_setup_solve_seconds_poisson256_fgmres_agg"
+                       if args.config == "fgmres_agg" else
+                       f"amg_setup_solve_seconds_{args.config}"),
             "value": sec_per_step,
             "unit": "s",
             "n_gpus": n_gpus,
@@ -163,8 +228,17 @@ def main():
             "dtype": "fp64",
             "data": "synthetic",
             "config": {
-                "model": "3D 7-pt Poisson, FGMRES + aggregation-AMG V-cycle "
-                         "(FGMRES_AGGREGATION.json), tol 1e-6 RELATIVE_INI",
+                "model": {
+                    "fgmres_agg": "3D 7-pt Poisson, FGMRES + aggregation-AMG"
+                                  " V-cycle (FGMRES_AGGREGATION.json), tol"
+                                  " 1e-6 RELATIVE_INI",
+                    "classical_pcg": "3D 7-pt Poisson, PCG + classical"
+                                     " Ruge-Stueben AMG V-cycle, tol 1e-6",
+                    "ilu0": "3D 7-pt Poisson, FGMRES + multicolor ILU(0),"
+                            " tol 1e-6",
+                    "block4_dilu": "block-4 coupled system, FGMRES +"
+                                   " multicolor DILU, tol 1e-6",
+                }[args.config],
                 "global_batch": dof,
                 "seq_len": args.size,
                 "parallelism": f"dd{n_gpus}",
