@@ -25,7 +25,6 @@ capability on top of torch.distributed (RCCL over xGMI / gloo on host):
 from __future__ import annotations
 
 import numpy as np
-import scipy.sparse as sp
 import torch
 import torch.distributed as tdist
 

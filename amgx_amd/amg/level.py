@@ -265,9 +265,6 @@ class ClassicalLevel(AMGLevel):
         computeAOperator_1x1_distributed, src/classical/
         classical_amg_level.cu:657-850): D1 interpolation onto global coarse
         columns, halo-row exchange of P, RAP with external-row shipping."""
-        import numpy as np
-        import scipy.sparse as sp
-
         from .classical_dist import (ClassicalDistOperators,
                                      coarse_numbering, rap_dist)
         from ..distributed.manager import DistributedManager

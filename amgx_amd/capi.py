@@ -17,7 +17,7 @@ matrix.
 from __future__ import annotations
 
 import threading
-from typing import Any, Dict, Optional
+from typing import Optional
 
 import numpy as np
 import torch

@@ -221,7 +221,6 @@ class DistributedManager:
         xf = x.reshape(-1)
         b = self.block_dim if block_override is None else block_override
         ops = []
-        from .. import ops as O
         for i, r in enumerate(self.neighbors):
             if block_override is None:
                 buf = self._send_bufs[i]

@@ -22,7 +22,7 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-from amgx_amd import AMGConfig, CSRMatrix, create_solver, ops  # noqa: E402
+from amgx_amd import AMGConfig, create_solver  # noqa: E402
 from amgx_amd.resources import Resources  # noqa: E402
 
 FGMRES_AGG = {
