@@ -73,28 +73,29 @@ def poisson_3d_local(nx: int, ny: int, nz: int, rank: int, world: int):
     NZg = nz * world
     n_local = nx * ny * nz
     row_start = rank * n_local
-    # vectorized stencil assembly
-    z = np.arange(nz).repeat(ny * nx)
-    y = np.tile(np.arange(ny).repeat(nx), nz)
-    x = np.tile(np.arange(nx), nz * ny)
-    gz = z + rank * nz
-    gid = (gz * ny + y) * nx + x
-    neigh = [
-        (gz > 0, -nx * ny), (y > 0, -nx), (x > 0, -1),
-        (np.ones(n_local, dtype=bool), 0),
-        (x < nx - 1, 1), (y < ny - 1, nx), (gz < NZg - 1, nx * ny),
-    ]
-    mask = np.stack([m for m, _ in neigh])            # (7, n)
-    offs = np.asarray([o for _, o in neigh])
-    deg = mask.sum(axis=0)
+    # vectorized stencil assembly in row-major (n, 7) layout so the boolean
+    # select emits rows in order with ascending columns — no transposes
+    lid = np.arange(n_local, dtype=np.int64)
+    x = lid % nx
+    y = (lid // nx) % ny
+    gz = lid // (nx * ny) + rank * nz
+    gid = lid + row_start
+    mask = np.empty((n_local, 7), dtype=bool)
+    mask[:, 0] = gz > 0
+    mask[:, 1] = y > 0
+    mask[:, 2] = x > 0
+    mask[:, 3] = True
+    mask[:, 4] = x < nx - 1
+    mask[:, 5] = y < ny - 1
+    mask[:, 6] = gz < NZg - 1
+    offs = np.asarray([-nx * ny, -nx, -1, 0, 1, nx, nx * ny], dtype=np.int64)
+    deg = mask.sum(axis=1)
     ro = np.zeros(n_local + 1, dtype=np.int64)
     np.cumsum(deg, out=ro[1:])
-    colmat = gid[None, :] + offs[:, None]             # (7, n)
-    valmat = np.where(offs[:, None] == 0, 6.0, -1.0) * np.ones((7, n_local))
-    sel = mask.T.reshape(-1)                          # row-major per row
-    cols = colmat.T.reshape(-1)[sel]
-    vals = valmat.T.reshape(-1)[sel]
-    return ro, cols.astype(np.int64), vals.astype(np.float64), row_start
+    cols = (gid[:, None] + offs[None, :])[mask]
+    valrow = np.asarray([-1.0, -1.0, -1.0, 6.0, -1.0, -1.0, -1.0])
+    vals = np.broadcast_to(valrow, (n_local, 7))[mask]
+    return ro, cols, vals, row_start
 
 
 def block_laplacian(nx: int, ny: int, block_dim: int = 4, device="cpu",
