@@ -144,3 +144,35 @@ def block_laplacian(nx: int, ny: int, block_dim: int = 4, device="cpu",
     return CSRMatrix.from_bsr(A.indptr.astype(np.int32),
                               A.indices.astype(np.int32), blocks,
                               n_cols=n, device=device, dtype=dtype)
+
+
+def poisson_3d_local_device(nx: int, ny: int, nz: int, rank: int, world: int,
+                            device="cpu", dtype=torch.float64):
+    """poisson_3d_local with torch assembly on the target device (the
+    multi-GPU bench path: each rank builds its slab on its own GPU).
+    Returns (ro, cols_global, vals, row_start) tensors."""
+    dev = torch.device(device)
+    NZg = nz * world
+    n = nx * ny * nz
+    row_start = rank * n
+    lid = torch.arange(n, dtype=torch.int64, device=dev)
+    x = lid % nx
+    y = (lid // nx) % ny
+    gz = lid // (nx * ny) + rank * nz
+    mask = torch.empty((n, 7), dtype=torch.bool, device=dev)
+    mask[:, 0] = gz > 0
+    mask[:, 1] = y > 0
+    mask[:, 2] = x > 0
+    mask[:, 3] = True
+    mask[:, 4] = x < nx - 1
+    mask[:, 5] = y < ny - 1
+    mask[:, 6] = gz < NZg - 1
+    offs = torch.tensor([-nx * ny, -nx, -1, 0, 1, nx, nx * ny],
+                        dtype=torch.int64, device=dev)
+    ro = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(mask.sum(1), 0, out=ro[1:])
+    cols = ((lid + row_start)[:, None] + offs[None, :])[mask]
+    valrow = torch.tensor([-1., -1., -1., 6., -1., -1., -1.],
+                          dtype=dtype, device=dev)
+    vals = valrow.expand(n, 7)[mask].contiguous()
+    return ro, cols, vals, row_start

@@ -121,9 +121,10 @@ def build_local_matrix(size, rank, world, device, problem="poisson"):
         from amgx_amd.problems import poisson_3d
         return poisson_3d(size, size, size, device=device), None
     from amgx_amd.distributed.manager import DistributedManager
-    from amgx_amd.problems import poisson_3d_local
-    ro, cols, vals, row_start = poisson_3d_local(size, size, size, rank, world)
-    n_local = len(ro) - 1
+    from amgx_amd.problems import poisson_3d_local_device
+    ro, cols, vals, row_start = poisson_3d_local_device(
+        size, size, size, rank, world, device=device)
+    n_local = int(ro.numel() - 1)
     A = DistributedManager.upload_global_csr(
         ro, cols, vals, n_local, row_start, size * size * size * world,
         device=device)

@@ -474,3 +474,19 @@ def test_ilu1_gpu():
     }}
     st, rel = _solve_gpu(cfg, A, tol=1e-9)
     assert st.converged and rel < 1e-8
+
+
+def test_poisson_3d_device_assembly():
+    """On-device Poisson assembly matches the host construction."""
+    from amgx_amd.problems import poisson_3d, poisson_3d_local_device
+    Ah = poisson_3d(6, 5, 4)
+    Ad = poisson_3d(6, 5, 4, device="cuda:0")
+    assert Ad.row_offsets.device.type == "cuda"
+    assert torch.equal(Ad.row_offsets.cpu(), Ah.row_offsets)
+    assert torch.equal(Ad.col_indices.cpu(), Ah.col_indices)
+    assert torch.allclose(Ad.values.cpu(), Ah.values)
+    ro, cols, vals, rs = poisson_3d_local_device(4, 4, 4, 0, 1,
+                                                 device="cuda:0")
+    A2 = poisson_3d(4, 4, 4)
+    assert torch.equal(ro.cpu().to(torch.int32), A2.row_offsets)
+    assert torch.equal(cols.cpu().to(torch.int32), A2.col_indices)
