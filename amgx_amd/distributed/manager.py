@@ -218,41 +218,55 @@ class DistributedManager:
         pinned-host staging)."""
         if not self.neighbors:
             return None
-        xf = x.reshape(-1)
-        b = self.block_dim if block_override is None else block_override
-        ops = []
-        for i, r in enumerate(self.neighbors):
-            if block_override is None:
-                buf = self._send_bufs[i]
-                if buf.dtype != x.dtype:
-                    buf = self._send_bufs[i] = buf.to(x.dtype)
-            else:
-                buf = torch.empty(int(self.b2l[i].numel()) * b, dtype=x.dtype,
-                                  device=x.device)
-            if self.device.type == "cuda":
-                from .. import _core
-                _core.gather(xf, self.b2l[i], b, buf)
-            else:
-                idx = self.b2l[i].to(torch.int64)
-                if b == 1:
-                    buf.copy_(xf[idx])
+
+        def _pack_and_post():
+            xf = x.reshape(-1)
+            b = self.block_dim if block_override is None else block_override
+            ops = []
+            for i, r in enumerate(self.neighbors):
+                if block_override is None:
+                    buf = self._send_bufs[i]
+                    if buf.dtype != x.dtype:
+                        buf = self._send_bufs[i] = buf.to(x.dtype)
                 else:
-                    ii = (idx[:, None] * b
-                          + torch.arange(b, dtype=torch.int64)[None, :])
-                    buf.copy_(xf[ii.reshape(-1)])
-            if buf.numel() > 0:
-                ops.append(dist.P2POp(dist.isend, buf, r))
-            lo, hi = self.halo_slices[i]
-            if hi > lo:
-                view = xf[(self.n_local + lo) * b:(self.n_local + hi) * b]
-                ops.append(dist.P2POp(dist.irecv, view, r))
-        if not ops:
-            return [] if async_start else None
-        reqs = dist.batch_isend_irecv(ops)
+                    buf = torch.empty(int(self.b2l[i].numel()) * b,
+                                      dtype=x.dtype, device=x.device)
+                if self.device.type == "cuda":
+                    from .. import _core
+                    _core.gather(xf, self.b2l[i], b, buf)
+                else:
+                    idx = self.b2l[i].to(torch.int64)
+                    if b == 1:
+                        buf.copy_(xf[idx])
+                    else:
+                        ii = (idx[:, None] * b
+                              + torch.arange(b, dtype=torch.int64)[None, :])
+                        buf.copy_(xf[ii.reshape(-1)])
+                if buf.numel() > 0:
+                    ops.append(dist.P2POp(dist.isend, buf, r))
+                lo, hi = self.halo_slices[i]
+                if hi > lo:
+                    view = xf[(self.n_local + lo) * b:
+                              (self.n_local + hi) * b]
+                    ops.append(dist.P2POp(dist.irecv, view, r))
+            return dist.batch_isend_irecv(ops) if ops else []
+
+        if self.comm_stream is not None:
+            # pack + post on the dedicated comm stream so the interior SpMV
+            # queued on the compute stream overlaps the exchange (reference
+            # m_bdy_stream latency hiding). rq.wait() later inserts the
+            # compute-stream dependency.
+            self.comm_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self.comm_stream):
+                reqs = _pack_and_post()
+        else:
+            reqs = _pack_and_post()
         if async_start:
             return reqs
         for rq in reqs:
             rq.wait()
+        if self.comm_stream is not None:
+            torch.cuda.current_stream().wait_stream(self.comm_stream)
         return None
 
     def add_from_halo(self, x: torch.Tensor, block_override: int = None):
