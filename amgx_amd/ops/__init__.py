@@ -78,7 +78,7 @@ def residual(A, x, b, r=None):
         return B.residual(A, x, b, r)
     if r is None:
         r = _new_dist_vec(A, b)
-    reqs = mgr.exchange_halo(x, async_start=True)
+    reqs = mgr.exchange_halo(x, async_start=True) or []
     B.residual(A, x, b, r, 0, mgr.boundary_start)
     for rq in reqs:
         rq.wait()
