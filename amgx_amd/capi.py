@@ -282,26 +282,91 @@ def AMGX_matrix_destroy(m: _MatrixHandle):
     return RC_OK
 
 
+def _fold_external_diag(n, b, ro, ci, va, diag):
+    """Merge the DIAG-property external diagonal into the CSR structure
+    (reference block-DIA-CSR, include/matrix.h:24-26). The gfx950 kernels
+    take one folded CSR — the external layout is an input format here."""
+    deg = np.diff(ro)
+    rows = np.repeat(np.arange(n, dtype=np.int64), deg)
+    all_rows = np.concatenate([rows, np.arange(n, dtype=np.int64)])
+    all_cols = np.concatenate([ci.astype(np.int64),
+                               np.arange(n, dtype=np.int64)])
+    vab = va.reshape(len(ci), -1)
+    db = diag.reshape(n, -1)
+    all_vals = np.concatenate([vab, db], axis=0)
+    order = np.lexsort((all_cols, all_rows))
+    ro2 = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(np.bincount(all_rows, minlength=n), out=ro2[1:])
+    cols2 = all_cols[order]
+    vals2 = all_vals[order]
+    # sum duplicates (a CSR that already stores an explicit diagonal)
+    ncols = int(max(cols2.max() + 1 if cols2.size else 1, n))
+    key = all_rows[order] * ncols + cols2
+    uniq, first = np.unique(key, return_index=True)
+    if uniq.size != key.size:
+        sums = np.add.reduceat(vals2, first, axis=0)
+        cols2 = cols2[first]
+        vals2 = sums
+        counts = np.bincount((uniq // ncols).astype(np.int64), minlength=n)
+        ro2 = np.zeros(n + 1, dtype=np.int64)
+        np.cumsum(counts, out=ro2[1:])
+    return ro2, cols2, vals2
+
+
 @_amgx_try
 def AMGX_matrix_upload_all(m: _MatrixHandle, n, nnz, block_dimx, block_dimy,
                            row_ptrs, col_indices, data, diag_data=None):
     mem, _, matprec = _parse_mode(m.mode)
     assert block_dimx == block_dimy, "rectangular blocks unsupported"
     device = mem if mem == "cpu" else m.res.device
-    ro = torch.as_tensor(np.asarray(row_ptrs), dtype=torch.int32)
-    ci = torch.as_tensor(np.asarray(col_indices), dtype=torch.int32)
     b = int(block_dimx)
-    va = torch.as_tensor(np.asarray(data)).to(matprec)
-    if b > 1:
-        va = va.reshape(nnz, b, b)
-    diag = None
+    ro_np = np.asarray(row_ptrs, dtype=np.int64)
+    ci_np = np.asarray(col_indices, dtype=np.int64)
+    va_np = np.asarray(data)
     if diag_data is not None:
-        diag = torch.as_tensor(np.asarray(diag_data)).to(matprec)
-        diag = diag.reshape(n, b, b) if b > 1 else diag.reshape(n)
-        diag = diag.to(device)
+        dg = np.asarray(diag_data)
+        ro_np, ci_np, va_np = _fold_external_diag(int(n), b, ro_np, ci_np,
+                                                  va_np, dg)
+        m._upload_structure = (np.asarray(row_ptrs, dtype=np.int64),
+                               np.asarray(col_indices, dtype=np.int64))
+    ro = torch.as_tensor(ro_np, dtype=torch.int32)
+    ci = torch.as_tensor(ci_np, dtype=torch.int32)
+    va = torch.as_tensor(np.ascontiguousarray(va_np)).to(matprec)
+    va = va.reshape(-1, b, b) if b > 1 else va.reshape(-1)
     m.A = CSRMatrix(ro.to(device), ci.to(device), va.to(device),
-                    n_cols=n, block_dim=b, diag=diag)
+                    n_cols=n, block_dim=b)
     return RC_OK
+
+
+def _fold_external_diag(n, b, ro, ci, va, diag):
+    """Merge the DIAG-property external diagonal into the CSR structure
+    (reference block-DIA-CSR, include/matrix.h:24-26). The gfx950 kernels
+    take one folded CSR — the external layout is an input format here."""
+    deg = np.diff(ro)
+    rows = np.repeat(np.arange(n, dtype=np.int64), deg)
+    all_rows = np.concatenate([rows, np.arange(n, dtype=np.int64)])
+    all_cols = np.concatenate([ci.astype(np.int64),
+                               np.arange(n, dtype=np.int64)])
+    vab = va.reshape(len(ci), -1)
+    db = diag.reshape(n, -1)
+    all_vals = np.concatenate([vab, db], axis=0)
+    order = np.lexsort((all_cols, all_rows))
+    ro2 = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(np.bincount(all_rows, minlength=n), out=ro2[1:])
+    cols2 = all_cols[order]
+    vals2 = all_vals[order]
+    # sum duplicates (a CSR that already stores an explicit diagonal)
+    ncols = int(max(cols2.max() + 1 if cols2.size else 1, n))
+    key = all_rows[order] * ncols + cols2
+    uniq, first = np.unique(key, return_index=True)
+    if uniq.size != key.size:
+        sums = np.add.reduceat(vals2, first, axis=0)
+        cols2 = cols2[first]
+        vals2 = sums
+        counts = np.bincount((uniq // ncols).astype(np.int64), minlength=n)
+        ro2 = np.zeros(n + 1, dtype=np.int64)
+        np.cumsum(counts, out=ro2[1:])
+    return ro2, cols2, vals2
 
 
 @_amgx_try
@@ -355,13 +420,21 @@ def AMGX_matrix_upload_distributed(m, n_global, n, nnz, block_dimx,
 def AMGX_matrix_replace_coefficients(m: _MatrixHandle, n, nnz, data,
                                      diag_data=None):
     _, _, matprec = _parse_mode(m.mode)
-    va = torch.as_tensor(np.asarray(data)).to(matprec)
-    if m.A.block_dim > 1:
-        va = va.reshape(-1, m.A.block_dim, m.A.block_dim)
-    diag = None
+    b = m.A.block_dim
     if diag_data is not None:
-        diag = torch.as_tensor(np.asarray(diag_data)).to(matprec)
-    m.A.replace_coefficients(va, diag)
+        # re-fold through the ORIGINAL upload structure so positions match
+        ro0, ci0 = m._upload_structure
+        _, _, va_np = _fold_external_diag(int(n), b, ro0, ci0,
+                                          np.asarray(data),
+                                          np.asarray(diag_data))
+        va = torch.as_tensor(np.ascontiguousarray(va_np)).to(matprec)
+        va = va.reshape(-1, b, b) if b > 1 else va.reshape(-1)
+        m.A.replace_coefficients(va)
+        return RC_OK
+    va = torch.as_tensor(np.asarray(data)).to(matprec)
+    if b > 1:
+        va = va.reshape(-1, b, b)
+    m.A.replace_coefficients(va)
     return RC_OK
 
 
@@ -727,6 +800,37 @@ def AMGX_matrix_attach_coloring(m: _MatrixHandle, row_coloring, num_rows=None,
 def AMGX_matrix_set_boundary_separation(m: _MatrixHandle, flag: int):
     m.boundary_separation = int(flag)
     return RC_OK
+
+
+def _fold_external_diag(n, b, ro, ci, va, diag):
+    """Merge the DIAG-property external diagonal into the CSR structure
+    (reference block-DIA-CSR, include/matrix.h:24-26). The gfx950 kernels
+    take one folded CSR — the external layout is an input format here."""
+    deg = np.diff(ro)
+    rows = np.repeat(np.arange(n, dtype=np.int64), deg)
+    all_rows = np.concatenate([rows, np.arange(n, dtype=np.int64)])
+    all_cols = np.concatenate([ci.astype(np.int64),
+                               np.arange(n, dtype=np.int64)])
+    vab = va.reshape(len(ci), -1)
+    db = diag.reshape(n, -1)
+    all_vals = np.concatenate([vab, db], axis=0)
+    order = np.lexsort((all_cols, all_rows))
+    ro2 = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(np.bincount(all_rows, minlength=n), out=ro2[1:])
+    cols2 = all_cols[order]
+    vals2 = all_vals[order]
+    # sum duplicates (a CSR that already stores an explicit diagonal)
+    ncols = int(max(cols2.max() + 1 if cols2.size else 1, n))
+    key = all_rows[order] * ncols + cols2
+    uniq, first = np.unique(key, return_index=True)
+    if uniq.size != key.size:
+        sums = np.add.reduceat(vals2, first, axis=0)
+        cols2 = cols2[first]
+        vals2 = sums
+        counts = np.bincount((uniq // ncols).astype(np.int64), minlength=n)
+        ro2 = np.zeros(n + 1, dtype=np.int64)
+        np.cumsum(counts, out=ro2[1:])
+    return ro2, cols2, vals2
 
 
 @_amgx_try

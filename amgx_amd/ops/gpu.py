@@ -58,6 +58,8 @@ def extract_diagonal(A):
 
 # ---------------------------------------------------------------------- SpMV
 def spmv(A, x, y=None, alpha=1.0, beta=0.0, row_begin=0, row_end=-1):
+    assert A.diag is None, \
+        "external DIAG matrices must be folded at upload (capi does this)"
     if row_end < 0:
         row_end = A.n_rows
     if y is None:
