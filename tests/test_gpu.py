@@ -490,3 +490,29 @@ def test_poisson_3d_device_assembly():
     A2 = poisson_3d(4, 4, 4)
     assert torch.equal(ro.cpu().to(torch.int32), A2.row_offsets)
     assert torch.equal(cols.cpu().to(torch.int32), A2.col_indices)
+
+
+def test_determinism_gpu():
+    """Two identical GPU setups+solves produce bitwise-identical results
+    (deterministic reductions/sorts; reference determinism_flag)."""
+    from amgx_amd.utils import DeterminismChecker
+
+    def run():
+        from tests.test_amg import FGMRES_AGG
+        A = to_gpu(poisson_3d(12, 12, 12))
+        cfg = AMGConfig.from_dict(FGMRES_AGG)
+        s = create_solver(cfg.root_scope(), resources=Resources("cuda:0"))
+        b = torch.ones(A.n_rows, dtype=torch.float64, device="cuda:0")
+        x = torch.zeros_like(b)
+        s.setup(A)
+        s.solve(b, x, zero_initial_guess=True)
+        chk = DeterminismChecker()
+        h = s.precond.hierarchy
+        for i, lvl in enumerate(h.levels):
+            chk.checkpoint(f"level{i}", lvl.A.row_offsets, lvl.A.col_indices,
+                           lvl.A.values)
+        chk.checkpoint("x", x)
+        return chk
+
+    c1, c2 = run(), run()
+    assert c1.same_as(c2), c1.diff(c2)[:3]
