@@ -160,8 +160,10 @@ class ChebyshevSolver(_SmootherBase):
             v.mul_(self.dinv.reshape(-1))
         else:
             bd = self.A.block_dim
-            v.copy_(torch.bmm(self.dinv.to(v.dtype),
-                              v.reshape(-1, bd, 1)).reshape(v.shape))
+            n_owned = self.A.n_rows * bd     # halo tail untouched (dist)
+            vo = v.reshape(-1)[:n_owned]
+            vo.copy_(torch.bmm(self.dinv.to(v.dtype),
+                               vo.reshape(-1, bd, 1)).reshape(-1))
 
     def _init_cheb(self):
         self.theta = 0.5 * (self.lmax + self.lmin)
