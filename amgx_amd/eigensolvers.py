@@ -124,6 +124,12 @@ class EigenSolver:
         self.status = EigenStatus()
 
     def setup(self, A):
+        if getattr(A, "manager", None) is not None:
+            # eigensolver reductions are single-process here; failing loudly
+            # beats silently-local dot products (multi-GPU: next round)
+            raise NotImplementedError(
+                "distributed eigensolvers are not wired yet — run the "
+                "eigensolver on a single process/GPU")
         self.A = A
         if self.which == "smallest":
             from .solvers import create_solver
