@@ -20,7 +20,7 @@ namespace amgx_hip {
 // strong iff |a_ij| >= theta * max_{k!=i}|a_ik|; optional all-weak rows when
 // |row sum| > max_row_sum * |a_ii|.
 template <typename T>
-__global__ void strength_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void strength_kernel(const int* __restrict__ ro,
                                 const int* __restrict__ ci,
                                 const T* __restrict__ va,
                                 const int* __restrict__ didx, int n,
@@ -71,7 +71,7 @@ __device__ __forceinline__ unsigned int pmis_hash(unsigned int a) {
     return a;
 }
 
-__global__ void pmis_lambda_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void pmis_lambda_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci,
                                    const int* __restrict__ tidx,
                                    const unsigned char* __restrict__ strong,
@@ -88,7 +88,7 @@ __global__ void pmis_lambda_kernel(const int* __restrict__ ro,
 }
 
 // state: 0 undecided, 1 C, -1 F
-__global__ void pmis_round1(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void pmis_round1(const int* __restrict__ ro,
                             const int* __restrict__ ci,
                             const int* __restrict__ tidx,
                             const unsigned char* __restrict__ strong, int n,
@@ -115,7 +115,7 @@ __global__ void pmis_round1(const int* __restrict__ ro,
     if (!ismax) atomicAdd(n_undecided, 1);
 }
 
-__global__ void pmis_round2(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void pmis_round2(const int* __restrict__ ro,
                             const int* __restrict__ ci,
                             const int* __restrict__ tidx,
                             const unsigned char* __restrict__ strong, int n,
@@ -136,7 +136,7 @@ __global__ void pmis_round2(const int* __restrict__ ro,
 }
 
 // isolated rows (no strong edges either way) -> F
-__global__ void pmis_isolated(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void pmis_isolated(const int* __restrict__ ro,
                               const int* __restrict__ ci,
                               const int* __restrict__ tidx,
                               const unsigned char* __restrict__ strong, int n,
@@ -177,7 +177,7 @@ void pmis_one_round(const int* ro, const int* ci, const int* tidx,
 // count pass: C rows -> 1 entry; F rows -> #strong C neighbors
 // cf spans ncols entries (= n for single-process, n_local+n_halo with
 // GLOBAL coarse ids for the distributed path)
-__global__ void d1_count(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void d1_count(const int* __restrict__ ro,
                          const int* __restrict__ ci,
                          const unsigned char* __restrict__ strong,
                          const int* __restrict__ cf, int n, int ncols,
@@ -196,7 +196,7 @@ __global__ void d1_count(const int* __restrict__ ro,
 //   alpha = sum_neg(all) / sum_neg(strong C), beta likewise for positives;
 //   positives lumped into the diagonal when no positive C connection.
 template <typename T>
-__global__ void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
+__global__ __launch_bounds__(AMGX_BLOCK) void d1_fill(const int* __restrict__ ro, const int* __restrict__ ci,
                         const T* __restrict__ va,
                         const unsigned char* __restrict__ strong,
                         const int* __restrict__ cf, const int* __restrict__ didx,
@@ -271,7 +271,7 @@ void interp_d1_count(const int* ro, const int* ci, const unsigned char* strong,
 //   l_ik = a_ik / u_kk ; for j in row(i) with pos[j] > pos[k] and (k,j) in
 //   pattern: a_ij -= l_ik * u_kj.
 template <typename T>
-__global__ void ilu0_factor_color(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_factor_color(const int* __restrict__ ro,
                                   const int* __restrict__ ci,
                                   const int* __restrict__ pos,
                                   const int* __restrict__ didx,
@@ -320,7 +320,7 @@ __global__ void ilu0_factor_color(const int* __restrict__ ro,
 
 // forward: y_i = r_i - sum_{pos[j]<pos[i]} l_ij y_j (unit L); per color.
 template <typename T>
-__global__ void ilu0_fwd(const int* __restrict__ ro, const int* __restrict__ ci,
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_fwd(const int* __restrict__ ro, const int* __restrict__ ci,
                          const int* __restrict__ pos, const T* __restrict__ lu,
                          const int* __restrict__ rows, int count,
                          const T* __restrict__ r, T* __restrict__ y, int n) {
@@ -338,7 +338,7 @@ __global__ void ilu0_fwd(const int* __restrict__ ro, const int* __restrict__ ci,
 
 // backward: z_i = (y_i - sum_{pos[j]>pos[i]} u_ij z_j) / u_ii; per color desc.
 template <typename T>
-__global__ void ilu0_bwd(const int* __restrict__ ro, const int* __restrict__ ci,
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_bwd(const int* __restrict__ ro, const int* __restrict__ ci,
                          const int* __restrict__ pos, const T* __restrict__ lu,
                          const int* __restrict__ didx,
                          const int* __restrict__ rows, int count,

@@ -41,7 +41,7 @@ __device__ __forceinline__ unsigned int hash_u32(unsigned int a,
 // already-colored neighbors — greedy-quality color counts (~max degree) with
 // Jones-Plassmann parallel rounds. Hash fixed across rounds, tie-break on
 // row id => deterministic. Colors >= 64 fall back past the bitmask (rare).
-__global__ void color_round_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void color_round_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci, int n,
                                    int* __restrict__ colors, int iter,
                                    int seed, int* __restrict__ n_uncolored) {
@@ -99,7 +99,7 @@ __device__ __forceinline__ double edge_weight(const T* va, const int* tidx,
 }
 
 template <typename T>
-__global__ void agg_propose_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void agg_propose_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci,
                                    const T* __restrict__ va,
                                    const int* __restrict__ tidx,
@@ -122,7 +122,7 @@ __global__ void agg_propose_kernel(const int* __restrict__ ro,
     prop[i] = best;
 }
 
-__global__ void agg_match_kernel(const int* __restrict__ prop, int n,
+__global__ __launch_bounds__(AMGX_BLOCK) void agg_match_kernel(const int* __restrict__ prop, int n,
                                  int* __restrict__ agg,
                                  int* __restrict__ changed) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -136,7 +136,7 @@ __global__ void agg_match_kernel(const int* __restrict__ prop, int n,
 }
 
 template <typename T>
-__global__ void agg_singleton_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void agg_singleton_kernel(const int* __restrict__ ro,
                                      const int* __restrict__ ci,
                                      const T* __restrict__ va,
                                      const int* __restrict__ tidx,
@@ -191,7 +191,7 @@ static void* dev_alloc(size_t bytes, hipStream_t s) {
 }
 static void dev_free(void* p, hipStream_t s) { HIP_CHECK(hipFreeAsync(p, s)); }
 
-__global__ void fill_row_ids(const int* __restrict__ ro, int n,
+__global__ __launch_bounds__(AMGX_BLOCK) void fill_row_ids(const int* __restrict__ ro, int n,
                              int* __restrict__ rows) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
@@ -199,7 +199,7 @@ __global__ void fill_row_ids(const int* __restrict__ ro, int n,
 }
 
 template <typename T>
-__global__ void make_agg_keys(const int* __restrict__ rows,
+__global__ __launch_bounds__(AMGX_BLOCK) void make_agg_keys(const int* __restrict__ rows,
                               const int* __restrict__ ci,
                               const T* __restrict__ va,
                               const int* __restrict__ agg_row,
@@ -215,7 +215,7 @@ __global__ void make_agg_keys(const int* __restrict__ rows,
 }
 
 // decompose sorted unique keys -> CSR of the coarse matrix
-__global__ void keys_to_csr(const unsigned long long* __restrict__ keys,
+__global__ __launch_bounds__(AMGX_BLOCK) void keys_to_csr(const unsigned long long* __restrict__ keys,
                             long long nnz_c, long long ncmod, long long nrows,
                             int* __restrict__ ro_c, int* __restrict__ ci_c) {
     long long stride = (long long)gridDim.x * blockDim.x;
@@ -233,7 +233,7 @@ __global__ void keys_to_csr(const unsigned long long* __restrict__ keys,
     }
 }
 
-__global__ void make_agg_keys_perm(const int* __restrict__ rows,
+__global__ __launch_bounds__(AMGX_BLOCK) void make_agg_keys_perm(const int* __restrict__ rows,
                                    const int* __restrict__ ci,
                                    const int* __restrict__ agg_row,
                                    const int* __restrict__ agg_col,
@@ -248,7 +248,7 @@ __global__ void make_agg_keys_perm(const int* __restrict__ rows,
     }
 }
 
-__global__ void find_run_starts(const unsigned long long* __restrict__ keys,
+__global__ __launch_bounds__(AMGX_BLOCK) void find_run_starts(const unsigned long long* __restrict__ keys,
                                 long long nnz, int* __restrict__ starts,
                                 unsigned int* __restrict__ nruns) {
     long long stride = (long long)gridDim.x * blockDim.x;
@@ -261,7 +261,7 @@ __global__ void find_run_starts(const unsigned long long* __restrict__ keys,
     }
 }
 
-__global__ void gather_keys(const unsigned long long* __restrict__ keys,
+__global__ __launch_bounds__(AMGX_BLOCK) void gather_keys(const unsigned long long* __restrict__ keys,
                             const int* __restrict__ starts, long long n,
                             unsigned long long* __restrict__ out) {
     long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -270,7 +270,7 @@ __global__ void gather_keys(const unsigned long long* __restrict__ keys,
 
 // block variant: payload is the nz INDEX, blocks summed in a second pass
 template <typename T>
-__global__ void sum_blocks_by_run(const int* __restrict__ run_starts,
+__global__ __launch_bounds__(AMGX_BLOCK) void sum_blocks_by_run(const int* __restrict__ run_starts,
                                   const int* __restrict__ perm,
                                   const T* __restrict__ va_in, long long nnz_c,
                                   long long nnz, int bb, T* __restrict__ va_out) {
@@ -392,7 +392,7 @@ long long galerkin_agg(const int* ro, const int* ci, const T* va, int n,
 }
 
 // ============================================================ ESC SpGEMM
-__global__ void expand_degree(const int* __restrict__ ciA,
+__global__ __launch_bounds__(AMGX_BLOCK) void expand_degree(const int* __restrict__ ciA,
                               const int* __restrict__ roB, long long nnzA,
                               unsigned long long* __restrict__ deg) {
     long long stride = (long long)gridDim.x * blockDim.x;
@@ -404,7 +404,7 @@ __global__ void expand_degree(const int* __restrict__ ciA,
 }
 
 template <typename T>
-__global__ void expand_fill(const int* __restrict__ rowsA,
+__global__ __launch_bounds__(AMGX_BLOCK) void expand_fill(const int* __restrict__ rowsA,
                             const int* __restrict__ ciA,
                             const T* __restrict__ vaA,
                             const int* __restrict__ roB,
@@ -496,7 +496,7 @@ long long spgemm_esc(const int* roA, const int* ciA, const T* vaA, int m,
 
 // ============================================================ transpose
 template <typename T>
-__global__ void gather_vals_rows(const int* __restrict__ perm,
+__global__ __launch_bounds__(AMGX_BLOCK) void gather_vals_rows(const int* __restrict__ perm,
                                  const T* __restrict__ va,
                                  const int* __restrict__ rows, long long nnz,
                                  T* __restrict__ va_t, int* __restrict__ ci_t) {
@@ -509,7 +509,7 @@ __global__ void gather_vals_rows(const int* __restrict__ perm,
     }
 }
 
-__global__ void cols_to_ro(const int* __restrict__ cols_sorted, long long nnz,
+__global__ __launch_bounds__(AMGX_BLOCK) void cols_to_ro(const int* __restrict__ cols_sorted, long long nnz,
                            int n, int* __restrict__ ro_t) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -523,7 +523,7 @@ __global__ void cols_to_ro(const int* __restrict__ cols_sorted, long long nnz,
     }
 }
 
-__global__ void iota_kernel(int* p, long long n) {
+__global__ __launch_bounds__(AMGX_BLOCK) void iota_kernel(int* p, long long n) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < n;
          t += stride)

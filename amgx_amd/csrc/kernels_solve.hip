@@ -18,7 +18,7 @@ namespace amgx_hip {
 // ============================================================ SpMV
 // thread-per-row: right shape for stencil-like rows (<= ~16 nnz).
 template <typename T, int UNROLL>
-__global__ void csrmv_tpr(const int* __restrict__ ro, const int* __restrict__ ci,
+__global__ __launch_bounds__(AMGX_BLOCK) void csrmv_tpr(const int* __restrict__ ro, const int* __restrict__ ci,
                           const T* __restrict__ va, const T* __restrict__ x,
                           T* __restrict__ y, const T* __restrict__ bvec,
                           T alpha, T beta, T gamma, int r0, int r1) {
@@ -40,7 +40,7 @@ __global__ void csrmv_tpr(const int* __restrict__ ro, const int* __restrict__ ci
 
 // L lanes cooperate on one row: for high-degree rows (unstructured matrices).
 template <typename T, int L>
-__global__ void csrmv_vec(const int* __restrict__ ro, const int* __restrict__ ci,
+__global__ __launch_bounds__(AMGX_BLOCK) void csrmv_vec(const int* __restrict__ ro, const int* __restrict__ ci,
                           const T* __restrict__ va, const T* __restrict__ x,
                           T* __restrict__ y, const T* __restrict__ bvec,
                           T alpha, T beta, T gamma, int r0, int r1) {
@@ -83,7 +83,7 @@ void csrmv(const int* ro, const int* ci, const T* va, const T* x, T* y,
 
 // block-CSR: one thread per output row component (row i, comp r).
 template <typename T>
-__global__ void bsrmv_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void bsrmv_kernel(const int* __restrict__ ro,
                              const int* __restrict__ ci,
                              const T* __restrict__ va, int b,
                              const T* __restrict__ x, T* __restrict__ y,
@@ -125,7 +125,7 @@ void bsrmv(const int* ro, const int* ci, const T* va, int b, const T* x, T* y,
 #define NPART 1024
 
 template <typename T, int OP>
-__global__ void reduce_stage1(const T* __restrict__ x, const T* __restrict__ y,
+__global__ __launch_bounds__(AMGX_BLOCK) void reduce_stage1(const T* __restrict__ x, const T* __restrict__ y,
                               long long n, T* __restrict__ part) {
     long long stride = (long long)gridDim.x * blockDim.x;
     T acc = T(0);
@@ -143,7 +143,7 @@ __global__ void reduce_stage1(const T* __restrict__ x, const T* __restrict__ y,
 }
 
 template <typename T, int OP>
-__global__ void reduce_stage2(const T* __restrict__ part, int nparts, T* out) {
+__global__ __launch_bounds__(AMGX_BLOCK) void reduce_stage2(const T* __restrict__ part, int nparts, T* out) {
     T acc = T(0);
     for (int i = threadIdx.x; i < nparts; i += blockDim.x) {
         if (OP == 2) acc = part[i] > acc ? part[i] : acc;
@@ -179,7 +179,7 @@ void reduce(const T* x, const T* y, long long n, int op, T* ws, T* out,
 }
 
 template <typename T>
-__global__ void axpy_kernel(T* y, const T* x, T a, long long n) {
+__global__ __launch_bounds__(AMGX_BLOCK) void axpy_kernel(T* y, const T* x, T a, long long n) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += stride)
@@ -187,7 +187,7 @@ __global__ void axpy_kernel(T* y, const T* x, T a, long long n) {
 }
 
 template <typename T>
-__global__ void axpby_kernel(T* y, const T* x, T a, T b, long long n) {
+__global__ __launch_bounds__(AMGX_BLOCK) void axpby_kernel(T* y, const T* x, T a, T b, long long n) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += stride)
@@ -195,7 +195,7 @@ __global__ void axpby_kernel(T* y, const T* x, T a, T b, long long n) {
 }
 
 template <typename T>
-__global__ void scal_kernel(T* x, T a, long long n) {
+__global__ __launch_bounds__(AMGX_BLOCK) void scal_kernel(T* x, T a, long long n) {
     long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += stride)
@@ -219,7 +219,7 @@ void scal(T* x, T a, long long n, hipStream_t s) {
 }
 
 // ============================================================ structure
-__global__ void diag_index_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void diag_index_kernel(const int* __restrict__ ro,
                                   const int* __restrict__ ci, int n,
                                   int* __restrict__ out) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -242,7 +242,7 @@ void diag_index(const int* ro, const int* ci, int n, int* out, hipStream_t s) {
 }
 
 template <typename T>
-__global__ void extract_diag_kernel(const T* __restrict__ va,
+__global__ __launch_bounds__(AMGX_BLOCK) void extract_diag_kernel(const T* __restrict__ va,
                                     const int* __restrict__ didx, int n, int b,
                                     T* __restrict__ out) {
     long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -263,7 +263,7 @@ void extract_diag(const int* ro, const int* ci, const T* va, const int* didx,
 }
 
 // transpose-entry lookup: for nz k in row i with col j, the index of (j, i).
-__global__ void trans_index_kernel(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void trans_index_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci, int n,
                                    int* __restrict__ out) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -292,7 +292,7 @@ void trans_index(const int* ro, const int* ci, int n, int nnz, int* out,
 
 // ============================================================ Jacobi
 template <typename T>
-__global__ void jacobi_dinv_scalar(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void jacobi_dinv_scalar(const int* __restrict__ ro,
                                    const int* __restrict__ ci,
                                    const T* __restrict__ va,
                                    const int* __restrict__ didx, int n,
@@ -312,7 +312,7 @@ __global__ void jacobi_dinv_scalar(const int* __restrict__ ro,
 }
 
 template <typename T, int BMAX>
-__global__ void jacobi_dinv_block(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void jacobi_dinv_block(const int* __restrict__ ro,
                                   const int* __restrict__ ci,
                                   const T* __restrict__ va,
                                   const int* __restrict__ didx, int n, int b,
@@ -358,7 +358,7 @@ void jacobi_dinv(const int* ro, const int* ci, const T* va, const int* didx,
 
 // fused damped Jacobi sweep: xo = xi + omega*dinv*(b - A xi), single pass.
 template <typename T>
-__global__ void jacobi_smooth_scalar(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void jacobi_smooth_scalar(const int* __restrict__ ro,
                                      const int* __restrict__ ci,
                                      const T* __restrict__ va,
                                      const T* __restrict__ dinv,
@@ -374,7 +374,7 @@ __global__ void jacobi_smooth_scalar(const int* __restrict__ ro,
 }
 
 template <typename T>
-__global__ void jacobi_smooth_block(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void jacobi_smooth_block(const int* __restrict__ ro,
                                     const int* __restrict__ ci,
                                     const T* __restrict__ va,
                                     const T* __restrict__ dinv,
@@ -419,7 +419,7 @@ void jacobi_smooth(const int* ro, const int* ci, const T* va, const T* dinv,
 
 // ============================================================ multicolor GS
 template <typename T>
-__global__ void gs_rows_scalar(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_scalar(const int* __restrict__ ro,
                                const int* __restrict__ ci,
                                const T* __restrict__ va,
                                const T* __restrict__ dinv,
@@ -440,7 +440,7 @@ __global__ void gs_rows_scalar(const int* __restrict__ ro,
 // under a valid distance-1 coloring; the only intra-row hazard is the
 // diagonal block, which this thread owns entirely).
 template <typename T>
-__global__ void gs_rows_block(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_block(const int* __restrict__ ro,
                               const int* __restrict__ ci,
                               const T* __restrict__ va,
                               const T* __restrict__ dinv,
@@ -488,7 +488,7 @@ void gs_smooth_rows(const int* ro, const int* ci, const T* va, const T* dinv,
 
 // ============================================================ DILU
 template <typename T>
-__global__ void dilu_setup_scalar(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_setup_scalar(const int* __restrict__ ro,
                                   const int* __restrict__ ci,
                                   const T* __restrict__ va,
                                   const int* __restrict__ didx,
@@ -512,7 +512,7 @@ __global__ void dilu_setup_scalar(const int* __restrict__ ro,
 }
 
 template <typename T, int BMAX>
-__global__ void dilu_setup_block(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_setup_block(const int* __restrict__ ro,
                                  const int* __restrict__ ci,
                                  const T* __restrict__ va,
                                  const int* __restrict__ didx,
@@ -578,7 +578,7 @@ void dilu_setup_color(const int* ro, const int* ci, const T* va,
 // full-row product only picks up earlier colors (valid coloring => no
 // same-color off-diagonals; diagonal contributes w_i = 0).
 template <typename T>
-__global__ void dilu_fwd_scalar(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar(const int* __restrict__ ro,
                                 const int* __restrict__ ci,
                                 const T* __restrict__ va,
                                 const T* __restrict__ einv,
@@ -593,7 +593,7 @@ __global__ void dilu_fwd_scalar(const int* __restrict__ ro,
 }
 
 template <typename T>
-__global__ void dilu_fwd_block(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_block(const int* __restrict__ ro,
                                const int* __restrict__ ci,
                                const T* __restrict__ va,
                                const T* __restrict__ einv,
@@ -625,7 +625,7 @@ __global__ void dilu_fwd_block(const int* __restrict__ ro,
 // backward: z_i = w_i - Einv_i sum_{color(j)>c} A_ij z_j; z pre-zeroed and
 // filled color-descending, so a full-row product sees only later colors.
 template <typename T>
-__global__ void dilu_bwd_scalar(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_scalar(const int* __restrict__ ro,
                                 const int* __restrict__ ci,
                                 const T* __restrict__ va,
                                 const T* __restrict__ einv,
@@ -640,7 +640,7 @@ __global__ void dilu_bwd_scalar(const int* __restrict__ ro,
 }
 
 template <typename T>
-__global__ void dilu_bwd_block(const int* __restrict__ ro,
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_block(const int* __restrict__ ro,
                                const int* __restrict__ ci,
                                const T* __restrict__ va,
                                const T* __restrict__ einv,
@@ -701,7 +701,7 @@ void dilu_bwd_color(const int* ro, const int* ci, const T* va, const T* einv,
 
 // ============================================================ transfers
 template <typename T>
-__global__ void restrict_kernel(const T* __restrict__ r,
+__global__ __launch_bounds__(AMGX_BLOCK) void restrict_kernel(const T* __restrict__ r,
                                 const int* __restrict__ agg, long long n,
                                 int b, T* __restrict__ rc) {
     long long stride = (long long)gridDim.x * blockDim.x;
@@ -725,7 +725,7 @@ void restrict_agg(const T* r, const int* agg, int n, int b, T* rc,
 // fine ids sorted by aggregate) — reference fillRowOffsetsAndColIndices R
 // storage, src/aggregation/aggregation_amg_level.cu:323.
 template <typename T>
-__global__ void restrict_csr_kernel(const int* __restrict__ off,
+__global__ __launch_bounds__(AMGX_BLOCK) void restrict_csr_kernel(const int* __restrict__ off,
                                     const int* __restrict__ fids,
                                     const T* __restrict__ r, long long nc,
                                     int b, T* __restrict__ rc) {
@@ -751,7 +751,7 @@ void restrict_csr(const int* off, const int* fids, const T* r, int nc, int b,
 }
 
 template <typename T>
-__global__ void prolongate_kernel(T* __restrict__ x, const T* __restrict__ xc,
+__global__ __launch_bounds__(AMGX_BLOCK) void prolongate_kernel(T* __restrict__ x, const T* __restrict__ xc,
                                   const int* __restrict__ agg, long long n,
                                   int b) {
     long long stride = (long long)gridDim.x * blockDim.x;
@@ -774,7 +774,7 @@ void prolongate_agg(T* x, const T* xc, const int* agg, int n, int b,
 // ============================================================ dense GEMV
 // coarse solve x = Ainv b; n <= a few hundred -> wave-per-row.
 template <typename T>
-__global__ void dense_gemv_kernel(const T* __restrict__ Ainv,
+__global__ __launch_bounds__(AMGX_BLOCK) void dense_gemv_kernel(const T* __restrict__ Ainv,
                                   const T* __restrict__ b, T* __restrict__ x,
                                   int n) {
     int row = blockIdx.x * (blockDim.x / WAVE_SIZE) + threadIdx.x / WAVE_SIZE;
@@ -798,7 +798,7 @@ void dense_gemv(const T* Ainv, const T* b, T* x, int n, hipStream_t s) {
 
 // ============================================================ gather/scatter
 template <typename T>
-__global__ void gather_kernel(const T* __restrict__ src,
+__global__ __launch_bounds__(AMGX_BLOCK) void gather_kernel(const T* __restrict__ src,
                               const int* __restrict__ idx, long long count,
                               int b, T* __restrict__ dst) {
     long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -809,7 +809,7 @@ __global__ void gather_kernel(const T* __restrict__ src,
 }
 
 template <typename T>
-__global__ void scatter_kernel(const T* __restrict__ src,
+__global__ __launch_bounds__(AMGX_BLOCK) void scatter_kernel(const T* __restrict__ src,
                                const int* __restrict__ idx, long long count,
                                int b, T* __restrict__ dst, bool add) {
     long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
