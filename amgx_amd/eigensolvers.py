@@ -86,6 +86,11 @@ class PageRankOperator(Operator):
 
     def __init__(self, A, damping: float = 0.85):
         super().__init__(A)
+        if getattr(A, "manager", None) is not None:
+            # a rank-local transpose is not the global transpose; the
+            # add_from_halo-composed distributed A^T apply is future work
+            raise NotImplementedError(
+                "distributed PageRank needs the transpose-apply exchange")
         self.d = damping
         # column-stochastic normalization: P = A D_out^{-1};
         # out-degree = row sums of A
@@ -94,10 +99,21 @@ class PageRankOperator(Operator):
         self.AT = ops.transpose(A)
 
     def apply(self, x, y):
+        mgr = getattr(self.A, "manager", None)
         ops.spmv(self.AT, x * self.out_inv, y)
         ops.scal(y, self.d)
-        dangling = float(ops.dot(x, (self.out_inv == 0).to(x.dtype)))
-        y += (self.d * dangling + (1.0 - self.d) * float(x.sum())) / self.n
+        if mgr is None:
+            dangling = float(ops.dot(x, (self.out_inv == 0).to(x.dtype)))
+            total = float(x.sum())
+            n_glob = self.n
+        else:
+            xo = mgr.owned(x)
+            mask = (self.out_inv == 0).to(x.dtype).reshape(-1)[
+                :mgr.owned_size]
+            dangling = mgr.global_sum(float(ops.dot(xo, mask)))
+            total = mgr.global_sum(float(xo.sum()))
+            n_glob = mgr.n_global
+        y += (self.d * dangling + (1.0 - self.d) * total) / n_glob
         return y
 
 
@@ -123,13 +139,51 @@ class EigenSolver:
         self.op: Optional[Operator] = None
         self.status = EigenStatus()
 
+    # ---------------------------------------------------- dist-aware algebra
+    def _mgr(self):
+        return getattr(self.A, "manager", None) if hasattr(self, "A") \
+            else None
+
+    def dot(self, x, y) -> float:
+        """Globally-reduced dot over owned entries (reference eigensolvers
+        run on the distributed vector ops)."""
+        mgr = self._mgr()
+        if mgr is None:
+            return ops.dot(x, y)
+        return mgr.global_sum(ops.dot(mgr.owned(x), mgr.owned(y)))
+
+    def nrm2(self, x) -> float:
+        return math.sqrt(max(self.dot(x, x), 0.0))
+
+    def _qr(self, B: torch.Tensor) -> torch.Tensor:
+        """Thin QR of a tall-skinny block; distributed via Cholesky-QR
+        (p x p Gram allreduce) — the TSQR idiom over RCCL."""
+        mgr = self._mgr()
+        if mgr is None:
+            Q, _ = torch.linalg.qr(B)
+            return Q
+        import torch.distributed as tdist
+        Bo = B[:mgr.owned_size]
+        G = (Bo.T @ Bo).double()
+        tdist.all_reduce(G)
+        L = torch.linalg.cholesky(G + 1e-14 * torch.eye(
+            G.shape[0], dtype=G.dtype, device=G.device))
+        Q = B.clone()
+        Q[:] = torch.linalg.solve_triangular(
+            L, B.double().T, upper=False).T.to(B.dtype)
+        return Q
+
+    def _block_inner(self, X: torch.Tensor, Y: torch.Tensor) -> np.ndarray:
+        """X^T Y over owned rows, globally reduced (p x p, host numpy)."""
+        mgr = self._mgr()
+        if mgr is None:
+            return (X.double().T @ Y.double()).cpu().numpy()
+        import torch.distributed as tdist
+        T = (X[:mgr.owned_size].double().T @ Y[:mgr.owned_size].double())
+        tdist.all_reduce(T)
+        return T.cpu().numpy()
+
     def setup(self, A):
-        if getattr(A, "manager", None) is not None:
-            # eigensolver reductions are single-process here; failing loudly
-            # beats silently-local dot products (multi-GPU: next round)
-            raise NotImplementedError(
-                "distributed eigensolvers are not wired yet — run the "
-                "eigensolver on a single process/GPU")
         self.A = A
         if self.which == "smallest":
             from .solvers import create_solver
@@ -157,8 +211,13 @@ class EigenSolver:
         pass
 
     def _rand_vec(self, seed=42):
-        g = torch.Generator().manual_seed(seed)
-        v = torch.rand(self.op.n, generator=g, dtype=torch.float64)
+        mgr = self._mgr()
+        n = mgr.ext_size if mgr is not None else self.op.n
+        g = torch.Generator().manual_seed(
+            seed + 7919 * (mgr.rank if mgr is not None else 0))
+        v = torch.rand(n, generator=g, dtype=torch.float64)
+        if mgr is not None:
+            v[mgr.owned_size:] = 0.0
         return v.to(self.A.dtype).to(self.A.device)
 
     def _to_true_eig(self, lam: float) -> float:
@@ -184,16 +243,16 @@ class PowerIteration(EigenSolver):
         Av = torch.zeros_like(v)
         lam = 0.0
         for it in range(self.max_iters):
-            nv = ops.nrm2(v)
+            nv = self.nrm2(v)
             if nv == 0:
                 break
             ops.scal(v, 1.0 / nv)
             self.op.apply(v, Av)
-            lam = ops.dot(v, Av)
+            lam = self.dot(v, Av)
             # residual ||Av - lam v||
             r = Av.clone()
             ops.axpy(r, v, -lam)
-            rn = ops.nrm2(r) / max(abs(lam), 1e-300)
+            rn = self.nrm2(r) / max(abs(lam), 1e-300)
             st.residuals.append(rn)
             st.iterations = it + 1
             v, Av = Av.clone(), Av
@@ -201,7 +260,7 @@ class PowerIteration(EigenSolver):
                 st.converged = True
                 break
         st.eigenvalues = [self._to_true_eig(lam)]
-        nv = ops.nrm2(v)
+        nv = self.nrm2(v)
         st.eigenvector = v / nv if nv else v
         return st
 
@@ -215,21 +274,21 @@ class Lanczos(EigenSolver):
         st = self.status = EigenStatus()
         m = min(self.max_iters, self.op.n)
         v = x0.clone() if x0 is not None else self._rand_vec()
-        ops.scal(v, 1.0 / ops.nrm2(v))
+        ops.scal(v, 1.0 / self.nrm2(v))
         V = [v]
         alphas, betas = [], []
         w = torch.zeros_like(v)
         for j in range(m):
             self.op.apply(V[j], w)
-            a = ops.dot(w, V[j])
+            a = self.dot(w, V[j])
             alphas.append(a)
             ops.axpy(w, V[j], -a)
             if j > 0:
                 ops.axpy(w, V[j - 1], -betas[-1])
             # full reorthogonalization
             for q in V:
-                ops.axpy(w, q, -ops.dot(w, q))
-            b = ops.nrm2(w)
+                ops.axpy(w, q, -self.dot(w, q))
+            b = self.nrm2(w)
             st.iterations = j + 1
             if j >= 1:
                 T = np.diag(alphas) + np.diag(betas, 1) + np.diag(betas, -1)
@@ -276,16 +335,16 @@ class Arnoldi(EigenSolver):
         st = self.status = EigenStatus()
         m = min(self.max_iters, self.op.n)
         v = x0.clone() if x0 is not None else self._rand_vec()
-        ops.scal(v, 1.0 / ops.nrm2(v))
+        ops.scal(v, 1.0 / self.nrm2(v))
         V = [v]
         H = np.zeros((m + 1, m))
         w = torch.zeros_like(v)
         for j in range(m):
             self.op.apply(V[j], w)
             for i in range(j + 1):
-                H[i, j] = ops.dot(w, V[i])
+                H[i, j] = self.dot(w, V[i])
                 ops.axpy(w, V[i], -H[i, j])
-            H[j + 1, j] = ops.nrm2(w)
+            H[j + 1, j] = self.nrm2(w)
             st.iterations = j + 1
             if j >= 1:
                 evals, evecs = np.linalg.eig(H[:j + 1, :j + 1])
@@ -322,21 +381,24 @@ class SubspaceIteration(EigenSolver):
         st = self.status = EigenStatus()
         k = max(self.wanted, 1)
         p = min(k + 2, self.op.n)
-        n = self.op.n
-        g = torch.Generator().manual_seed(17)
+        mgr = self._mgr()
+        n = mgr.ext_size if mgr is not None else self.op.n
+        g = torch.Generator().manual_seed(
+            17 + 7919 * (mgr.rank if mgr is not None else 0))
         X = torch.rand(n, p, generator=g, dtype=torch.float64) \
             .to(self.A.dtype).to(self.A.device)
+        if mgr is not None:
+            X[mgr.owned_size:] = 0.0
         Y = torch.zeros_like(X)
         lam_old = None
         for it in range(self.max_iters):
-            X, _ = torch.linalg.qr(X.double())
-            X = X.to(self.A.dtype)
+            X = self._qr(X.double()).to(self.A.dtype)
             for c in range(p):
                 xc = X[:, c].contiguous()
                 yc = torch.zeros_like(xc)
                 self.op.apply(xc, yc)
                 Y[:, c] = yc
-            T = (X.double().T @ Y.double()).cpu().numpy()
+            T = self._block_inner(X, Y)
             evals, evecs = np.linalg.eig(T)
             order = np.argsort(-np.abs(evals))
             lam = np.real(evals[order[:k]])
@@ -370,15 +432,15 @@ class LOBPCG(EigenSolver):
     def solve(self, x0=None):
         st = self.status = EigenStatus()
         x = x0.clone() if x0 is not None else self._rand_vec()
-        ops.scal(x, 1.0 / ops.nrm2(x))
+        ops.scal(x, 1.0 / self.nrm2(x))
         Ax = torch.zeros_like(x)
         self.op.apply(x, Ax)
-        lam = ops.dot(x, Ax)
+        lam = self.dot(x, Ax)
         p = None
         for it in range(self.max_iters):
             r = Ax.clone()
             ops.axpy(r, x, -lam)
-            rn = ops.nrm2(r) / max(abs(lam), 1e-300)
+            rn = self.nrm2(r) / max(abs(lam), 1e-300)
             st.residuals.append(rn)
             st.iterations = it + 1
             if rn < self.tol:
@@ -388,7 +450,7 @@ class LOBPCG(EigenSolver):
             self.prec.solve(r, w, zero_initial_guess=True)
             basis = [x, w] + ([p] if p is not None else [])
             B = torch.stack([v.double() for v in basis], dim=1)
-            Q, _ = torch.linalg.qr(B)
+            Q = self._qr(B)
             m = Q.shape[1]
             AQ = torch.zeros_like(Q)
             for c in range(m):
@@ -396,15 +458,16 @@ class LOBPCG(EigenSolver):
                 tmp = torch.zeros_like(qc)
                 self.op.apply(qc, tmp)
                 AQ[:, c] = tmp.double()
-            T = (Q.T @ AQ).cpu().numpy()
+            T = self._block_inner(Q, AQ)
+            T = 0.5 * (T + T.T)
             evals, evecs = np.linalg.eigh(T)
             y = evecs[:, 0]
             x_new = (Q @ torch.from_numpy(y).to(Q.device)).to(self.A.dtype)
             p = (x_new - x * float(y[0])).contiguous()
             x = x_new.contiguous()
-            ops.scal(x, 1.0 / ops.nrm2(x))
+            ops.scal(x, 1.0 / self.nrm2(x))
             self.op.apply(x, Ax)
-            lam = ops.dot(x, Ax)
+            lam = self.dot(x, Ax)
         st.eigenvalues = [float(lam)]
         st.eigenvector = x
         return st

@@ -632,3 +632,33 @@ def _impl_test_dist_unstructured_halo_stress(rank, world, tmp):
     x_user = mgr.permute_out(x)
     assert torch.allclose(x_user, xs[lo:hi], atol=1e-6), \
         float((x_user - xs[lo:hi]).abs().max())
+
+
+def test_dist_eigensolvers():
+    _run_dist(test_dist_eigensolvers)
+
+
+def _impl_test_dist_eigensolvers(rank, world, tmp):
+    """Distributed eigensolvers: globally-reduced dots and Cholesky-QR give
+    the same extreme eigenvalues as a serial dense reference."""
+    import numpy as np
+
+    from amgx_amd.config import ConfigScope
+    from amgx_amd.eigensolvers import create_eigensolver
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    n = 5
+    A = _make_dist_A(rank, world, n)
+    dense = poisson_3d(n, n, n * world).to_scipy().toarray()
+    lam_ref = float(np.linalg.eigvalsh(dense).max())
+    for name, iters in (("POWER_ITERATION", 4000), ("LANCZOS", 300),
+                        ("SUBSPACE_ITERATION", 500)):
+        es = create_eigensolver(
+            ConfigScope(None, {"eig_solver": name, "eig_max_iters": iters,
+                               "eig_tolerance": 1e-8}),
+            resources=Resources("cpu", distributed=True))
+        es.setup(A)
+        st = es.solve()
+        assert st.converged, f"{name}: {st.iterations} iters"
+        assert abs(st.eigenvalues[-1] - lam_ref) < 1e-5 * lam_ref, \
+            f"{name}: {st.eigenvalues[-1]} vs {lam_ref}"
