@@ -83,13 +83,12 @@ def write_system(path: str, A: CSRMatrix, b=None, x=None):
         if flags:
             f.write("%%NVAMG " + " ".join(flags) + "\n")
         f.write(f"{m.shape[0]} {m.shape[1]} {m.nnz}\n")
-        for r, c, v in zip(m.row, m.col, m.data):
-            f.write(f"{r + 1} {c + 1} {v:.17g}\n")
+        np.savetxt(f, np.column_stack([m.row + 1, m.col + 1, m.data]),
+                   fmt="%d %d %.17g")
         import torch as _torch
         for vec in (b, x):
             if vec is None:
                 continue
             arr = vec.detach().cpu().numpy().reshape(-1) \
                 if isinstance(vec, _torch.Tensor) else np.asarray(vec).reshape(-1)
-            for val in arr[:m.shape[0]]:
-                f.write(f"{val:.17g}\n")
+            np.savetxt(f, arr[:m.shape[0]], fmt="%.17g")
