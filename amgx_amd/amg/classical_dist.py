@@ -330,3 +330,133 @@ class ClassicalDistOperators:
         tmp = torch.zeros(self.P.n_rows, dtype=x.dtype, device=x.device)
         ops.spmv(self.P, xc_ext, tmp)
         x.reshape(-1)[:self.P.n_rows] += tmp
+
+
+def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
+                   cf_ext: torch.Tensor, coarse_offs, theta: float):
+    """Distributed distance-2 (standard) interpolation (reference
+    src/classical/interpolators/distance2.cu + the 2-ring halo,
+    num_import_rings=2): F rows distribute couplings to strong F neighbors
+    over those neighbors' strong C points. Halo F rows are fetched with
+    halo_matrix (one matrix-halo exchange); the C/F status and global coarse
+    ids of the resulting 2-ring columns come from one more HaloExchange of
+    the owners' cf array. Host assembly (like the serial D2 reference);
+    returns a CSRMatrix P with GLOBAL coarse columns on A's device."""
+    from ..distributed.manager import halo_matrix
+    from ..matrix import CSRMatrix
+    n = mgr.n_local
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    va = A.values.cpu().numpy().astype(np.float64).reshape(-1)
+    strong = strong_out.cpu().numpy()
+    cfx = cf_ext.cpu().numpy().astype(np.int64)      # len n_cols
+    # 1-ring halo rows with global columns
+    ro_h, cols_h, vals_h = halo_matrix(mgr, A)
+    # cf for every 2-ring global id
+    lo, hi = mgr.row_start, mgr.row_start + n
+    ids2 = np.unique(cols_h) if cols_h.size else np.zeros(0, dtype=np.int64)
+    needed2 = ids2[(ids2 < lo) | (ids2 >= hi)]
+    hx = HaloExchange(needed2, mgr.part_offsets, device=A.row_offsets.device,
+                      owner_local_map=mgr.row_iperm.cpu().numpy()
+                      .astype(np.int64))
+    cf_gid_own = torch.full((n,), -1.0, dtype=torch.float64,
+                            device=A.row_offsets.device)
+    cf_gid_own[:] = cf_ext[:n].to(torch.float64)
+    tail = torch.full((max(int(needed2.size), 1),), -1.0,
+                      dtype=torch.float64, device=A.row_offsets.device)
+    if needed2.size:
+        hx.forward(cf_gid_own, tail[:needed2.size])
+    cf2 = tail.cpu().numpy().round().astype(np.int64)
+
+    iperm = mgr.row_iperm.cpu().numpy().astype(np.int64)
+    halo_pos = {int(g): p for p, g in enumerate(mgr.halo_global)}
+    # strength of halo rows: |a| >= theta * rowmax over off-diagonals
+    rowmax_h = np.zeros(mgr.n_halo)
+    for p in range(mgr.n_halo):
+        s0, s1 = ro_h[p], ro_h[p + 1]
+        g = int(mgr.halo_global[p])
+        cc, vv = cols_h[s0:s1], vals_h[s0:s1]
+        off = cc != g
+        if off.any():
+            rowmax_h[p] = np.abs(vv[off]).max()
+
+    def cf_any(g):
+        """Global coarse id of global fine id g (or -1)."""
+        if lo <= g < hi:
+            return int(cfx[iperm[g - lo]])
+        p = halo_pos.get(int(g))
+        if p is not None:
+            return int(cfx[n + p])
+        k = np.searchsorted(needed2, g)
+        if k < needed2.size and needed2[k] == g:
+            return int(cf2[k])
+        return -1
+
+    Pc, Pv, indptr = [], [], [0]
+    gid_own = mgr.row_perm.cpu().numpy() + lo     # internal -> global
+    for i in range(n):
+        if cfx[i] >= 0:
+            Pc.append(int(cfx[i]))
+            Pv.append(1.0)
+            indptr.append(indptr[-1] + 1)
+            continue
+        s, e = ro[i], ro[i + 1]
+        diag = 0.0
+        acc = {}
+        weak_sum = 0.0
+        for k in range(s, e):
+            j, a = int(ci[k]), va[k]
+            if j == i:
+                diag = a
+                continue
+            if not strong[k]:
+                weak_sum += a
+                continue
+            cfj = int(cfx[j]) if j < cfx.size else -1
+            if cfj >= 0:
+                acc[cfj] = acc.get(cfj, 0.0) + a
+                continue
+            # strong F neighbor: distribute over its strong C points
+            cpts, cvals = [], []
+            if j < n:                      # owned F neighbor
+                for kk in range(ro[j], ro[j + 1]):
+                    jj = int(ci[kk])
+                    cfk = int(cfx[jj]) if jj < cfx.size else -1
+                    if strong[kk] and cfk >= 0:
+                        cpts.append(cfk)
+                        cvals.append(va[kk])
+            else:                          # halo F neighbor: fetched row
+                p = j - n
+                gj = int(mgr.halo_global[p])
+                s0, s1 = ro_h[p], ro_h[p + 1]
+                for kk in range(s0, s1):
+                    gk = int(cols_h[kk])
+                    if gk == gj:
+                        continue
+                    if abs(vals_h[kk]) < theta * rowmax_h[p] \
+                            or rowmax_h[p] == 0.0:
+                        continue
+                    cfk = cf_any(gk)
+                    if cfk >= 0:
+                        cpts.append(cfk)
+                        cvals.append(vals_h[kk])
+            tot = sum(cvals)
+            if cpts and tot != 0.0:
+                for cc_, av in zip(cpts, cvals):
+                    acc[cc_] = acc.get(cc_, 0.0) + a * av / tot
+            else:
+                weak_sum += a
+        denom = diag + weak_sum
+        if not acc or denom == 0.0:
+            indptr.append(indptr[-1])
+            continue
+        for cc_ in sorted(acc):
+            Pc.append(cc_)
+            Pv.append(-acc[cc_] / denom)
+        indptr.append(indptr[-1] + len(acc))
+    dev = A.row_offsets.device
+    return CSRMatrix(
+        torch.as_tensor(np.asarray(indptr, dtype=np.int32)).to(dev),
+        torch.as_tensor(np.asarray(Pc, dtype=np.int32)).to(dev),
+        torch.as_tensor(np.asarray(Pv, dtype=np.float64)).to(A.dtype).to(dev),
+        n_cols=int(coarse_offs[-1]))

@@ -271,13 +271,26 @@ class ClassicalLevel(AMGLevel):
         A = self.A
         cf_ext, coarse_offs = coarse_numbering(mgr, self.cf_map,
                                                self.num_coarse)
-        # D1 interpolation onto GLOBAL coarse columns, on A's device (the
-        # gfx950 interp kernels take the ext-length cf array directly)
-        S_dev = self._strong_out
-        if A.row_offsets.is_cuda:
-            S_dev = S_dev.to(torch.uint8)
-        P_m = ops._backend(A).interp_d1(A, S_dev, cf_ext,
-                                        int(coarse_offs[-1]))
+        interp = self.scope.get("interpolator") \
+            if self.scope.has("interpolator") else "D1"
+        if interp == "D2":
+            # distance-2 through the 2-ring (halo-row fetch)
+            from .classical_dist import interp_d2_dist
+            P_m = interp_d2_dist(A, mgr, self._strong_out, cf_ext,
+                                 coarse_offs,
+                                 float(self.scope.get("strength_threshold")))
+        elif interp in (None, "D1"):
+            # D1 onto GLOBAL coarse columns, on A's device (the gfx950
+            # interp kernels take the ext-length cf array directly)
+            S_dev = self._strong_out
+            if A.row_offsets.is_cuda:
+                S_dev = S_dev.to(torch.uint8)
+            P_m = ops._backend(A).interp_d1(A, S_dev, cf_ext,
+                                            int(coarse_offs[-1]))
+        else:
+            raise NotImplementedError(
+                f"distributed interpolator {interp!r}: D1 and D2 are wired"
+                " (MULTIPASS: next round)")
         tf = float(self.scope.get("interp_truncation_factor"))
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:

@@ -662,3 +662,46 @@ def _impl_test_dist_eigensolvers(rank, world, tmp):
         assert st.converged, f"{name}: {st.iterations} iters"
         assert abs(st.eigenvalues[-1] - lam_ref) < 1e-5 * lam_ref, \
             f"{name}: {st.eigenvalues[-1]} vs {lam_ref}"
+
+
+def test_dist_classical_d2():
+    _run_dist(test_dist_classical_d2)
+
+
+def _impl_test_dist_classical_d2(rank, world, tmp):
+    """Distributed D2 (standard) interpolation through the 2-ring halo-row
+    fetch: PCG+classical converges at least as well as D1."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+
+    def solve(interp):
+        cfg = AMGConfig.from_dict({"solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "CLASSICAL",
+                "interpolator": interp,
+                "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+                "max_iters": 1, "min_coarse_rows": 12, "cycle": "V",
+            },
+            "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+            "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+        }})
+        A = _make_dist_A(rank, world, 8)
+        mgr = A.manager
+        s = create_solver(cfg.root_scope(), resources=Resources(
+            "cpu", distributed=True))
+        b = mgr.new_ext_vec(torch.float64)
+        b[:mgr.owned_size] = 1.0
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        r = ops.residual(A, x, b)
+        nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+            r[:mgr.owned_size])), "L2")
+        bn = mgr.global_norm(float(torch.linalg.vector_norm(
+            b[:mgr.owned_size])), "L2")
+        assert st.converged and nrm / bn < 1e-7, f"{interp}: {st}"
+        return st.iterations
+
+    it_d2 = solve("D2")
+    it_d1 = solve("D1")
+    assert it_d2 <= it_d1 + 5, f"D2 {it_d2} vs D1 {it_d1}"
