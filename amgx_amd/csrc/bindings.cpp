@@ -492,6 +492,31 @@ std::tuple<Tensor, Tensor, Tensor> interp_d1(Tensor ro, Tensor ci, Tensor va,
     return {p_ro, p_ci, p_va};
 }
 
+std::tuple<Tensor, Tensor, Tensor> truncate_rows(Tensor ro, Tensor ci,
+                                                 Tensor va, double factor) {
+    int n = (int)(ro.numel() - 1);
+    auto counts = torch::empty({n}, ro.options());
+    DISPATCH_FT(va, "truncate_count", [&] {
+        amgx_hip::truncate_rows_gpu<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            n, factor, nullptr, nullptr, nullptr, counts.data_ptr<int>(),
+            cur_stream());
+    });
+    auto ro_out = torch::zeros({n + 1}, ro.options());
+    ro_out.slice(0, 1, n + 1).copy_(
+        torch::cumsum(counts.to(torch::kLong), 0).to(torch::kInt));
+    int64_t nnz_out = n > 0 ? ro_out[n].item<int64_t>() : 0;
+    auto ci_out = torch::empty({nnz_out}, ro.options());
+    auto va_out = torch::empty({nnz_out}, va.options());
+    DISPATCH_FT(va, "truncate_fill", [&] {
+        amgx_hip::truncate_fill_gpu<scalar_t>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
+            n, factor, ro_out.data_ptr<int>(), ci_out.data_ptr<int>(),
+            va_out.data_ptr<scalar_t>(), cur_stream());
+    });
+    return {ro_out, ci_out, va_out};
+}
+
 Tensor interp_d1_count(Tensor ro, Tensor ci, Tensor strong, Tensor cf) {
     int n = (int)(ro.numel() - 1);
     auto counts = torch::empty({n}, ro.options());
@@ -586,6 +611,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pmis_select", &pmis_select);
     m.def("interp_d1", &interp_d1);
     m.def("interp_d1_count", &interp_d1_count);
+    m.def("truncate_rows", &truncate_rows);
     m.def("ilu0_setup", &ilu0_setup);
     m.def("ilu0_apply", &ilu0_apply);
 }

@@ -406,3 +406,81 @@ INSTANTIATE_CLASSICAL(double)
 INSTANTIATE_CLASSICAL(float)
 
 }  // namespace amgx_hip
+
+namespace amgx_hip {
+
+// ============================================================ truncate
+// Drop |p_ij| < factor * rowmax_i, rescale survivors to preserve the row sum
+// (reference src/truncate.cu truncate_kernel:388 + truncateAndScale:506).
+template <typename T>
+__global__ __launch_bounds__(AMGX_BLOCK) void truncate_count(
+    const int* __restrict__ ro, const T* __restrict__ va, int n,
+    double factor, int* __restrict__ counts) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    int s = ro[i], e = ro[i + 1];
+    double mx = 0.0;
+    for (int k = s; k < e; ++k) mx = fmax(mx, fabs((double)va[k]));
+    int c = 0;
+    for (int k = s; k < e; ++k)
+        if (fabs((double)va[k]) >= factor * mx) ++c;
+    counts[i] = c;
+}
+
+template <typename T>
+__global__ __launch_bounds__(AMGX_BLOCK) void truncate_fill(
+    const int* __restrict__ ro, const int* __restrict__ ci,
+    const T* __restrict__ va, int n, double factor,
+    const int* __restrict__ ro_out, int* __restrict__ ci_out,
+    T* __restrict__ va_out) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    int s = ro[i], e = ro[i + 1];
+    double mx = 0.0, sum_old = 0.0, sum_new = 0.0;
+    for (int k = s; k < e; ++k) {
+        double a = (double)va[k];
+        sum_old += a;
+        mx = fmax(mx, fabs(a));
+    }
+    for (int k = s; k < e; ++k)
+        if (fabs((double)va[k]) >= factor * mx) sum_new += (double)va[k];
+    double scale = (sum_new != 0.0 && sum_old != 0.0) ? sum_old / sum_new
+                                                      : 1.0;
+    int out = ro_out[i];
+    for (int k = s; k < e; ++k) {
+        if (fabs((double)va[k]) < factor * mx) continue;
+        ci_out[out] = ci[k];
+        va_out[out] = (T)((double)va[k] * scale);
+        ++out;
+    }
+}
+
+template <typename T>
+void truncate_rows_gpu(const int* ro, const int* ci, const T* va, int n,
+                       double factor, const int* ro_out, int* ci_out,
+                       T* va_out, int* counts, hipStream_t s) {
+    hipLaunchKernelGGL((truncate_count<T>), dim3(grid_1d(n)),
+                       dim3(AMGX_BLOCK), 0, s, ro, va, n, factor, counts);
+}
+
+template <typename T>
+void truncate_fill_gpu(const int* ro, const int* ci, const T* va, int n,
+                       double factor, const int* ro_out, int* ci_out,
+                       T* va_out, hipStream_t s) {
+    hipLaunchKernelGGL((truncate_fill<T>), dim3(grid_1d(n)),
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, va, n, factor, ro_out,
+                       ci_out, va_out);
+}
+
+#define INSTANTIATE_TRUNC(T)                                                   \
+    template void truncate_rows_gpu<T>(const int*, const int*, const T*, int, \
+                                       double, const int*, int*, T*, int*,    \
+                                       hipStream_t);                          \
+    template void truncate_fill_gpu<T>(const int*, const int*, const T*, int, \
+                                       double, const int*, int*, T*,          \
+                                       hipStream_t);
+
+INSTANTIATE_TRUNC(double)
+INSTANTIATE_TRUNC(float)
+
+}  // namespace amgx_hip

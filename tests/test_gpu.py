@@ -516,3 +516,16 @@ def test_determinism_gpu():
 
     c1, c2 = run(), run()
     assert c1.same_as(c2), c1.diff(c2)[:3]
+
+
+def test_truncate_rows_gpu():
+    """GPU factor truncation matches the host reference (drop + row-sum
+    rescale; reference src/truncate.cu truncateAndScale_kernel)."""
+    from amgx_amd.ops import cpu as cpu_ops
+    from amgx_amd.ops import gpu as gpu_ops
+    A = rand_csr(300, 0.04, seed=9)
+    ref = cpu_ops.truncate_rows(A, trunc_factor=0.4)
+    got = gpu_ops.truncate_rows(to_gpu(A), trunc_factor=0.4)
+    assert torch.equal(got.row_offsets.cpu(), ref.row_offsets)
+    assert torch.equal(got.col_indices.cpu(), ref.col_indices)
+    assert torch.allclose(got.values.cpu(), ref.values, atol=1e-13)
