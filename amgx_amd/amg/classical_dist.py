@@ -460,3 +460,97 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
         torch.as_tensor(np.asarray(Pc, dtype=np.int32)).to(dev),
         torch.as_tensor(np.asarray(Pv, dtype=np.float64)).to(A.dtype).to(dev),
         n_cols=int(coarse_offs[-1]))
+
+
+def interp_multipass_dist(A, mgr: DistributedManager,
+                          strong_out: torch.Tensor, cf_ext: torch.Tensor,
+                          coarse_offs, max_passes: int = 10):
+    """Distributed multipass interpolation (reference
+    src/classical/interpolators/multipass.cu): pass 0 = C rows; each later
+    pass interpolates F rows through already-interpolated strong neighbors.
+    Neighbor P rows cross ranks by one CSR-row exchange per pass, done flags
+    by one vector exchange per pass."""
+    from ..matrix import CSRMatrix
+    n = mgr.n_local
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    va = A.values.cpu().numpy().astype(np.float64).reshape(-1)
+    strong = strong_out.cpu().numpy()
+    cfx = cf_ext.cpu().numpy().astype(np.int64)
+    rows_P = [None] * n
+    done = np.zeros(n, dtype=bool)
+    for i in range(n):
+        if cfx[i] >= 0:
+            rows_P[i] = {int(cfx[i]): 1.0}
+            done[i] = True
+
+    done_ext = mgr.new_ext_vec(torch.float64)
+    for _pass in range(max_passes):
+        # share done flags and current P rows of the boundary
+        done_ext[:n] = torch.from_numpy(done.astype(np.float64))
+        mgr.exchange_halo(done_ext, block_override=1)
+        halo_done = done_ext[n:].cpu().numpy() > 0.5
+        indptr = [0]
+        cols_l, vals_l = [], []
+        for i in range(n):
+            row = rows_P[i] or {}
+            for cc in sorted(row):
+                cols_l.append(cc)
+                vals_l.append(row[cc])
+            indptr.append(len(cols_l))
+        halo_rows = exchange_csr_rows(
+            mgr, np.asarray(indptr, dtype=np.int64),
+            np.asarray(cols_l, dtype=np.int64), np.asarray(vals_l))
+        progressed = False
+        newly = []
+        for i in range(n):
+            if done[i]:
+                continue
+            s, e = ro[i], ro[i + 1]
+            diag = 0.0
+            acc = {}
+            lump = 0.0
+            ok = False
+            for k in range(s, e):
+                j, a = int(ci[k]), va[k]
+                if j == i:
+                    diag = a
+                    continue
+                src = None
+                if strong[k]:
+                    if j < n and done[j]:
+                        src = rows_P[j]
+                    elif j >= n and halo_done[j - n] \
+                            and halo_rows[j - n] is not None:
+                        cc_, vv_ = halo_rows[j - n]
+                        src = dict(zip(cc_.tolist(), vv_.tolist()))
+                if src is not None:
+                    for cc2, wjc in src.items():
+                        acc[int(cc2)] = acc.get(int(cc2), 0.0) + a * wjc
+                    ok = True
+                elif not strong[k]:
+                    lump += a
+            if ok and (diag + lump) != 0.0:
+                newly.append((i, {cc2: -aw / (diag + lump)
+                                  for cc2, aw in acc.items()}))
+        for i, row in newly:
+            rows_P[i] = row
+            done[i] = True
+            progressed = True
+        total_left = mgr.global_sum(float((~done).sum()))
+        any_prog = mgr.global_sum(1.0 if progressed else 0.0)
+        if total_left == 0 or any_prog == 0:
+            break
+    Pc, Pv, indptr = [], [], [0]
+    for i in range(n):
+        row = rows_P[i] or {}
+        for cc in sorted(row):
+            Pc.append(cc)
+            Pv.append(row[cc])
+        indptr.append(indptr[-1] + len(row))
+    dev = A.row_offsets.device
+    return CSRMatrix(
+        torch.as_tensor(np.asarray(indptr, dtype=np.int32)).to(dev),
+        torch.as_tensor(np.asarray(Pc, dtype=np.int32)).to(dev),
+        torch.as_tensor(np.asarray(Pv, dtype=np.float64)).to(A.dtype).to(dev),
+        n_cols=int(coarse_offs[-1]))

@@ -287,10 +287,14 @@ class ClassicalLevel(AMGLevel):
                 S_dev = S_dev.to(torch.uint8)
             P_m = ops._backend(A).interp_d1(A, S_dev, cf_ext,
                                             int(coarse_offs[-1]))
+        elif interp == "MULTIPASS":
+            from .classical_dist import interp_multipass_dist
+            P_m = interp_multipass_dist(A, mgr, self._strong_out, cf_ext,
+                                        coarse_offs)
         else:
             raise NotImplementedError(
-                f"distributed interpolator {interp!r}: D1 and D2 are wired"
-                " (MULTIPASS: next round)")
+                f"distributed interpolator {interp!r}: D1/D2/MULTIPASS are"
+                " wired (EM: next round)")
         tf = float(self.scope.get("interp_truncation_factor"))
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:

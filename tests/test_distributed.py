@@ -705,3 +705,38 @@ def _impl_test_dist_classical_d2(rank, world, tmp):
     it_d2 = solve("D2")
     it_d1 = solve("D1")
     assert it_d2 <= it_d1 + 5, f"D2 {it_d2} vs D1 {it_d1}"
+
+
+def test_dist_classical_multipass():
+    _run_dist(test_dist_classical_multipass)
+
+
+def _impl_test_dist_classical_multipass(rank, world, tmp):
+    """Distributed MULTIPASS interpolation (per-pass P-row + done-flag
+    exchange) converges on Poisson."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {
+            "solver": "AMG", "algorithm": "CLASSICAL",
+            "interpolator": "MULTIPASS",
+            "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+            "max_iters": 1, "min_coarse_rows": 12, "cycle": "V",
+        },
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+    }})
+    A = _make_dist_A(rank, world, 7)
+    mgr = A.manager
+    s = create_solver(cfg.root_scope(), resources=Resources(
+        "cpu", distributed=True))
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged, f"rank {rank}: {st}"
+    r = ops.residual(A, x, b)
+    nrm = mgr.global_norm(float(torch.linalg.vector_norm(
+        r[:mgr.owned_size])), "L2")
+    assert nrm < 1e-5
