@@ -191,6 +191,10 @@ class ClassicalLevel(AMGLevel):
         mgr = getattr(self.A, "manager", None)
         if mgr is not None:
             from .classical_dist import pmis_dist, strength_dist
+            if self.index < int(self.scope.get("aggressive_levels") or 0):
+                import warnings
+                warnings.warn("distributed aggressive coarsening is not "
+                              "wired; using plain PMIS on this level")
             theta = float(self.scope.get("strength_threshold"))
             mrs = float(self.scope.get("max_row_sum"))
             self._strong_out, strong_union = strength_dist(

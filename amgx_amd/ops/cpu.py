@@ -566,11 +566,14 @@ def dilu_solve(A, Einv, coloring, r, relaxation, x):
     def rowsb(rows):
         return (rows[:, None] * bd + np.arange(bd)[None, :]).reshape(-1)
     # forward: valid coloring => same-color off-diagonals absent, and the
-    # diagonal contributes 0 while w[rows] == 0
+    # diagonal contributes 0 while w[rows] == 0. Row-sliced products: each
+    # color sweep reads only its rows' entries (O(nnz) per direction total,
+    # not O(colors * nnz)).
+    mc = m.tocsr()
     for c in range(nc):
         rows = np.nonzero(colors == c)[0]
         rb = rowsb(rows)
-        tmp = rv[rb] - (m @ w)[rb]
+        tmp = rv[rb] - (mc[rb, :] @ w)
         if bd == 1:
             w[rb] = ei[rows] * tmp
         else:
@@ -581,7 +584,7 @@ def dilu_solve(A, Einv, coloring, r, relaxation, x):
     for c in range(nc - 1, -1, -1):
         rows = np.nonzero(colors == c)[0]
         rb = rowsb(rows)
-        s = (m @ later)[rb]
+        s = mc[rb, :] @ later
         if bd == 1:
             z[rb] = w[rb] - ei[rows] * s
         else:
