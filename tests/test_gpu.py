@@ -529,3 +529,70 @@ def test_truncate_rows_gpu():
     assert torch.equal(got.row_offsets.cpu(), ref.row_offsets)
     assert torch.equal(got.col_indices.cpu(), ref.col_indices)
     assert torch.allclose(got.values.cpu(), ref.values, atol=1e-13)
+
+
+def test_wf_cycles_gpu():
+    """W and F cycles on device converge (reference src/cycles/)."""
+    for cyc in ("W", "F", "CG"):
+        cfg = {"solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "AGGREGATION",
+                "smoother": "BLOCK_JACOBI", "presweeps": 1, "postsweeps": 1,
+                "max_iters": 1, "min_coarse_rows": 16, "cycle": cyc,
+            },
+            "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+            "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+        }}
+        A = to_gpu(poisson_3d(10, 10, 10))
+        st, rel = _solve_gpu(cfg, A, tol=1e-8)
+        assert st.converged and rel < 1e-7, f"{cyc}: {st}"
+
+
+def test_structure_reuse_gpu():
+    """Resetup on device keeps hierarchy structure, updates values."""
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {
+            "solver": "AMG", "algorithm": "AGGREGATION",
+            "smoother": "BLOCK_JACOBI", "max_iters": 1,
+            "min_coarse_rows": 16, "cycle": "V",
+            "structure_reuse_levels": -1, "scope": "amg",
+        },
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+    }})
+    A = to_gpu(poisson_3d(8, 8, 8))
+    s = create_solver(cfg.root_scope(), resources=Resources("cuda:0"))
+    b = torch.ones(A.n_rows, dtype=torch.float64, device="cuda:0")
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st1 = s.solve(b, x, zero_initial_guess=True)
+    h1 = s.precond.hierarchy
+    A2 = CSRMatrix(A.row_offsets, A.col_indices, A.values * 2.0,
+                   n_cols=A.n_cols)
+    s.resetup(A2)
+    assert s.precond.hierarchy is h1
+    x2 = torch.zeros_like(b)
+    st2 = s.solve(b, x2, zero_initial_guess=True)
+    assert st1.converged and st2.converged
+    assert torch.allclose(x2, x / 2.0, atol=1e-6)
+
+
+def test_scaler_gpu():
+    """Scaler lifecycle on device: matrix restored after solve."""
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "PCG", "max_iters": 200, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8,
+        "scaling": "BINORMALIZATION",
+        "preconditioner": {"solver": "BLOCK_JACOBI", "max_iters": 1},
+    }})
+    A = to_gpu(poisson_3d(8, 8, 8))
+    vals_before = A.values.clone()
+    s = create_solver(cfg.root_scope(), resources=Resources("cuda:0"))
+    b = torch.ones(A.n_rows, dtype=torch.float64, device="cuda:0")
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged
+    assert torch.allclose(A.values, vals_before, atol=1e-12)
+    r = ops.residual(A, x, b)
+    assert float(torch.linalg.vector_norm(r)) < 1e-5
