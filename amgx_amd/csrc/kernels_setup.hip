@@ -186,6 +186,7 @@ void agg_merge_singletons(const int* ro, const int* ci, const T* va,
 // ============================================================ sort helpers
 static void* dev_alloc(size_t bytes, hipStream_t s) {
     void* p = nullptr;
+    if (bytes == 0) bytes = 16;   // degenerate empty-matrix paths
     HIP_CHECK(hipMallocAsync(&p, bytes, s));
     return p;
 }
