@@ -157,6 +157,8 @@ class AMGHierarchy:
             sub = sub.child(dict(sub.node,
                                  max_iters=self.scope.get("max_coarse_iters")))
         if name == "DENSE_LU_SOLVER":
+            # a stalled coarsening must never densify a huge matrix (an
+            # n^2 inverse would OOM the box): beyond the cap, smooth instead
             maxr = self.scope.get("dense_lu_max_rows")
             if maxr and coarsest.A.n_rows > maxr:
                 name = None

@@ -95,7 +95,8 @@ def _register_core_parameters() -> None:
     P("intensive_smoothing", int, 0, "accepted for parity")
     P("cycle_iters", int, 2, "inner iterations of CG/CGF cycles")
     P("dense_lu_num_rows", int, 128, "DENSE_LU activates at/below this many rows")
-    P("dense_lu_max_rows", int, 0, "DENSE_LU refuses above this many rows (0=inf)")
+    P("dense_lu_max_rows", int, 8192,
+      "DENSE_LU refuses above this many rows (falls back to coarsest sweeps)")
     P("exact_coarse_solve", int, 0, "gather global coarse problem for DENSE_LU")
     # --- aggregation ------------------------------------------------------------------
     P("selector", str, "SIZE_2", "aggregation selector / classical CF selector")
