@@ -146,10 +146,20 @@ class AMGHierarchy:
 
     def _setup_coarse_solver(self):
         coarsest = self.levels[-1]
-        if getattr(coarsest.A, "manager", None) is not None:
-            from ..solvers.dense_lu import GatheredDenseLU
-            self.coarse_solver = GatheredDenseLU(self.res)
-            self.coarse_solver.setup(coarsest.A)
+        mgr = getattr(coarsest.A, "manager", None)
+        if mgr is not None:
+            maxr = self.scope.get("dense_lu_max_rows")
+            n_glob = mgr.global_rows(coarsest.A.n_rows)
+            if not maxr or n_glob <= maxr:
+                from ..solvers.dense_lu import GatheredDenseLU
+                self.coarse_solver = GatheredDenseLU(self.res)
+                self.coarse_solver.setup(coarsest.A)
+                return
+            # stalled distributed coarsening: never densify a huge gathered
+            # matrix — smooth the coarsest level instead
+            self.coarse_solver = None
+            coarsest.smoother = self._make_smoother()
+            coarsest.smoother.setup(coarsest.A)
             return
         name, sub = self.scope.sub_solver("coarse_solver", "DENSE_LU_SOLVER")
         if (sub is not None and not sub.has("max_iters")
