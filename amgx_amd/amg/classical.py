@@ -309,14 +309,17 @@ def interp_d2(A, S, cf_map, num_coarse, scope):
             if j < cf.size and cf[j] >= 0:
                 acc[cf[j]] = acc.get(cf[j], 0.0) + a
             elif j < n:
-                # strong F neighbor: distribute a_ij over j's strong C points
+                # strong F neighbor: distribute a_ij over j's strong C
+                # points by MAGNITUDE (sums to a_ij exactly; a signed sum
+                # can cancel to ~0 and blow the weights up on coarse
+                # re-coarsened operators)
                 js, je = ro[j], ro[j + 1]
                 cpts, cvals = [], []
                 for kk in range(js, je):
                     jj = ci[kk]
                     if strong[kk] and jj < cf.size and cf[jj] >= 0:
                         cpts.append(cf[jj])
-                        cvals.append(v[kk])
+                        cvals.append(abs(v[kk]))
                 tot = sum(cvals)
                 if cpts and tot != 0.0:
                     for cc, av in zip(cpts, cvals):

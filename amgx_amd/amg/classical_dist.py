@@ -417,6 +417,8 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
                 acc[cfj] = acc.get(cfj, 0.0) + a
                 continue
             # strong F neighbor: distribute over its strong C points
+            # magnitude-proportional distribution (sums to a_ij exactly;
+            # signed sums can cancel and blow the weights up)
             cpts, cvals = [], []
             if j < n:                      # owned F neighbor
                 for kk in range(ro[j], ro[j + 1]):
@@ -424,7 +426,7 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
                     cfk = int(cfx[jj]) if jj < cfx.size else -1
                     if strong[kk] and cfk >= 0:
                         cpts.append(cfk)
-                        cvals.append(va[kk])
+                        cvals.append(abs(va[kk]))
             else:                          # halo F neighbor: fetched row
                 p = j - n
                 gj = int(mgr.halo_global[p])
@@ -439,7 +441,7 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
                     cfk = cf_any(gk)
                     if cfk >= 0:
                         cpts.append(cfk)
-                        cvals.append(vals_h[kk])
+                        cvals.append(abs(vals_h[kk]))
             tot = sum(cvals)
             if cpts and tot != 0.0:
                 for cc_, av in zip(cpts, cvals):

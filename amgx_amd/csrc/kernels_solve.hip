@@ -500,14 +500,18 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_setup_scalar(const int* __res
     if (t >= count) return;
     int i = rows[t];
     int dk = didx[i];
-    T e = dk >= 0 ? va[dk] : T(0);
+    T d = dk >= 0 ? va[dk] : T(0);
+    T e = d;
     for (int k = ro[i]; k < ro[i + 1]; ++k) {
         int j = ci[k];
         if (j == i) continue;
         int tk = tidx[k];
         if (tk >= 0 && colors[j] < color) e -= va[k] * einv[j] * va[tk];
     }
-    if (e == T(0)) e = T(1);
+    // tiny-pivot safeguard: a near-zero modified pivot cascades huge Einv
+    // through later colors — fall back to the plain diagonal
+    T dref = (d != T(0)) ? d : T(1);
+    if (fabs((double)e) < 1e-10 * fabs((double)dref)) e = dref;
     einv[i] = T(1) / e;
 }
 
@@ -548,6 +552,20 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_setup_block(const int* __rest
                     E[r * b + c] -= s;
                 }
         }
+    }
+    // stabilized fallback: wildly grown E means an earlier near-singular
+    // pivot — revert to the plain diagonal block
+    double emax = 0.0, dmax = 0.0;
+    for (int q = 0; q < b * b; ++q) {
+        emax = fmax(emax, fabs((double)E[q]));
+        double dv = dk >= 0 ? fabs((double)va[(long long)dk * b * b + q])
+                            : 0.0;
+        dmax = fmax(dmax, dv);
+    }
+    if (!(emax <= 1e10 * (dmax + 1.0))) {
+        for (int q = 0; q < b * b; ++q)
+            E[q] = dk >= 0 ? va[(long long)dk * b * b + q]
+                           : (q % (b + 1) == 0 ? T(1) : T(0));
     }
     small_mat_inv(E, Inv, b);
     for (int q = 0; q < b * b; ++q) einv[(long long)i * b * b + q] = Inv[q];

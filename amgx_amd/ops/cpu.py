@@ -507,7 +507,12 @@ def dilu_setup(A, coloring):
         for c in range(coloring.num_colors):
             rows = np.nonzero(colors == c)[0]
             e = d[rows] - (B[rows, :] @ (einv * done))
-            e = np.where(e != 0.0, e, 1.0)
+            # tiny-pivot safeguard: a near-zero modified pivot would cascade
+            # huge Einv through later colors — fall back to the plain
+            # diagonal for that row (stabilized DILU)
+            dref = np.where(d[rows] != 0.0, d[rows], 1.0)
+            bad = np.abs(e) < 1e-10 * np.abs(dref)
+            e = np.where(bad, dref, e)
             einv[rows] = 1.0 / e
             done[rows] = 1.0
         return torch.from_numpy(einv).to(A.dtype)
@@ -529,8 +534,12 @@ def dilu_setup(A, coloring):
                 # rank-local, like the reference's per-partition smoother
                 if j != i and j < A.n_rows and colors[j] < c and AT[k] >= 0:
                     E -= vals[k] @ einv[j] @ vals[AT[k]]
-            if abs(np.linalg.det(E)) < 1e-300:
-                E = np.eye(A.block_dim)
+            Dref = D[i] if np.abs(np.diag(D[i])).min() > 0 \
+                else np.eye(A.block_dim)
+            if not np.isfinite(E).all() \
+                    or abs(np.linalg.det(E)) < 1e-10 * max(
+                        abs(np.linalg.det(Dref)), 1e-300):
+                E = Dref          # stabilized fallback to the block diagonal
             einv[i] = np.linalg.inv(E)
     return torch.from_numpy(einv).to(A.dtype)
 

@@ -31,10 +31,9 @@ def dilu_solve(A, Einv, coloring, r, relaxation, x):
 
 @register_solver("MULTICOLOR_DILU")
 class MulticolorDILUSolver(_SmootherBase):
-    def __init__(self, scope, resources):
-        super().__init__(scope, resources)
-        if not scope.has("relaxation_factor"):
-            self.relaxation_factor = 1.0
+    # relaxation_factor comes from the registry default (0.9), matching the
+    # reference (src/core.cu:398): undamped DILU can diverge on the dense
+    # rows of D2/aggressive coarse operators
 
     def solver_setup(self):
         A = self.A

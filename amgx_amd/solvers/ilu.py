@@ -66,8 +66,6 @@ def extended_sparsity(A, level: int):
 class MulticolorILUSolver(_SmootherBase):
     def __init__(self, scope, resources):
         super().__init__(scope, resources)
-        if not scope.has("relaxation_factor"):
-            self.relaxation_factor = 1.0
         self.sparsity_level = int(scope.get("ilu_sparsity_level") or 0)
 
     def solver_setup(self):
