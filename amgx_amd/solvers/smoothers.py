@@ -174,6 +174,13 @@ class ChebyshevSolver(_SmootherBase):
         self._rho = 1.0 / self._sigma if self._sigma else 1.0
         self._d = None
 
+    def sweep(self, b, x, n: int = 1):
+        # smoother entry (AMG per-level sweeps bypass solve()): the
+        # Chebyshev recurrence state must be re-seeded per sweep sequence
+        self.solve_init(b, x, False)
+        for _ in range(n):
+            self.solve_iteration(b, x)
+
     def solve_iteration(self, b, x):
         r = ops.residual(self.A, x, b)
         self._apply_dinv(r)

@@ -470,3 +470,10 @@ def test_poisson_27pt_classical():
     rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
     assert st.converged and rel < 1e-7
     assert st.iterations <= 30
+
+
+def test_chebyshev_as_amg_smoother():
+    """Chebyshev used as the per-level smoother (the sweep() entry must
+    re-seed the recurrence; regression for a missing solve_init)."""
+    st, rel = _solve_classical(_classical_cfg(smoother="CHEBYSHEV"), n=8)
+    assert st.converged and rel < 1e-7, f"{st}, rel={rel}"
