@@ -280,64 +280,64 @@ def interp_d2(A, S, cf_map, num_coarse, scope):
     src/classical/interpolators/distance2.cu): an F point i interpolates
     from its strong C neighbors AND the strong C neighbors of its strong F
     neighbors, with a_ij distributed over j's C points proportionally to
-    a_jk (classical standard-interpolation formula)."""
+    |a_jk| (magnitude-proportional: the distribution sums to a_ij exactly
+    and cannot cancel). Fully vectorized: the through-F pass is one sparse
+    product F @ W."""
     from ..matrix import CSRMatrix
     ro, ci, v = _csr_parts(A)
-    strong = S.cpu().numpy()
+    strong = np.asarray(S.cpu().numpy() if torch.is_tensor(S) else S,
+                        dtype=bool)
     cf = cf_map.cpu().numpy().astype(np.int64) if torch.is_tensor(cf_map) \
         else np.asarray(cf_map, dtype=np.int64)
     n = A.n_rows
-    Pc, Pv, indptr = [], [], [0]
-    for i in range(n):
-        if cf[i] >= 0:
-            Pc.append(cf[i])
-            Pv.append(1.0)
-            indptr.append(indptr[-1] + 1)
-            continue
-        s, e = ro[i], ro[i + 1]
-        diag = 0.0
-        acc = {}           # coarse id -> accumulated coupling
-        weak_sum = 0.0
-        for k in range(s, e):
-            j, a = ci[k], v[k]
-            if j == i:
-                diag = a
-                continue
-            if not strong[k]:
-                weak_sum += a
-                continue
-            if j < cf.size and cf[j] >= 0:
-                acc[cf[j]] = acc.get(cf[j], 0.0) + a
-            elif j < n:
-                # strong F neighbor: distribute a_ij over j's strong C
-                # points by MAGNITUDE (sums to a_ij exactly; a signed sum
-                # can cancel to ~0 and blow the weights up on coarse
-                # re-coarsened operators)
-                js, je = ro[j], ro[j + 1]
-                cpts, cvals = [], []
-                for kk in range(js, je):
-                    jj = ci[kk]
-                    if strong[kk] and jj < cf.size and cf[jj] >= 0:
-                        cpts.append(cf[jj])
-                        cvals.append(abs(v[kk]))
-                tot = sum(cvals)
-                if cpts and tot != 0.0:
-                    for cc, av in zip(cpts, cvals):
-                        acc[cc] = acc.get(cc, 0.0) + a * av / tot
-                else:
-                    weak_sum += a     # dead-end F neighbor: lump
-            else:
-                weak_sum += a
-        denom = diag + weak_sum
-        if not acc or denom == 0.0:
-            indptr.append(indptr[-1])
-            continue
-        for cc in sorted(acc):
-            Pc.append(cc)
-            Pv.append(-acc[cc] / denom)
-        indptr.append(indptr[-1] + len(acc))
-    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
-                       np.asarray(indptr)), shape=(n, num_coarse))
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    local = ci < n
+    cf_col = np.where(local, cf[np.minimum(ci, cf.size - 1)], -1)
+    offd = rows != ci
+    strongC = strong & offd & (cf_col >= 0)
+    strongF = strong & offd & local & (cf_col < 0)
+    # distribution weights of each F row j over its C points
+    absC = np.abs(v) * strongC
+    sC_sum = np.bincount(rows[strongC], weights=np.abs(v[strongC]),
+                         minlength=n)
+    alive = sC_sum > 0.0                       # F rows that can distribute
+    # W[j, cf[k]] = |a_jk| / sC_sum[j] over strongC entries of row j
+    wr = rows[strongC]
+    W = sp.csr_matrix(
+        (np.abs(v[strongC]) / sC_sum[wr], (wr, cf_col[strongC])),
+        shape=(n, num_coarse))
+    # F[i, j] = a_ij over strong-F edges into LIVE rows j; dead-end edges
+    # are lumped into the denominator instead
+    live_edge = strongF & alive[np.minimum(ci, n - 1)] & local
+    F = sp.csr_matrix((v[live_edge], (rows[live_edge], ci[live_edge])),
+                      shape=(n, n))
+    D = sp.csr_matrix((v[strongC], (rows[strongC], cf_col[strongC])),
+                      shape=(n, num_coarse))
+    acc = (D + F @ W).tocsr()
+    acc.sum_duplicates()
+    # denominators: diag + all weak couplings + dead-end strong-F couplings
+    diag = np.zeros(n)
+    dmask = rows == ci
+    diag[rows[dmask]] = v[dmask]
+    weak = offd & ~strong
+    dead = strongF & ~alive[np.minimum(ci, n - 1)]
+    lump = (np.bincount(rows[weak], weights=v[weak], minlength=n)
+            + np.bincount(rows[dead], weights=v[dead], minlength=n))
+    denom = diag + lump
+    row_nnz = np.diff(acc.indptr)
+    f_ok = (cf[:n] < 0) & (denom != 0.0) & (row_nnz > 0)
+    # scale F rows by -1/denom; C rows become identity; everything else empty
+    scale = np.where(f_ok, np.divide(-1.0, denom, out=np.ones(n),
+                                     where=denom != 0.0), 0.0)
+    acc = sp.diags(scale) @ acc
+    acc = acc.tocsr()
+    acc.eliminate_zeros()
+    c_rows = np.nonzero(cf[:n] >= 0)[0]
+    ident = sp.csr_matrix((np.ones(c_rows.size),
+                           (c_rows, cf[c_rows])), shape=(n, num_coarse))
+    P = (acc + ident).tocsr()
+    P.sum_duplicates()
+    P.sort_indices()
     return CSRMatrix.from_scipy(P, dtype=A.dtype)
 
 
