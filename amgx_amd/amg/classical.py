@@ -345,61 +345,53 @@ def interp_d2(A, S, cf_map, num_coarse, scope):
 def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
     """Multipass interpolation for aggressive coarsening (reference
     src/classical/interpolators/multipass.cu): pass 0 = C points (identity);
-    pass 1 = F points with strong C neighbors (direct D1 weights); pass k>1 =
-    F points whose strong neighbors were interpolated in earlier passes,
-    composing their rows."""
+    pass k = F points whose strong neighbors were interpolated in earlier
+    passes, composing their rows. Vectorized: each pass is one SpGEMM of the
+    strong-edges-into-done-rows matrix with the current P."""
     from ..matrix import CSRMatrix
     ro, ci, v = _csr_parts(A)
-    strong = S.cpu().numpy()
+    strong = np.asarray(S.cpu().numpy() if torch.is_tensor(S) else S,
+                        dtype=bool)
     cf = cf_map.cpu().numpy().astype(np.int64) if torch.is_tensor(cf_map) \
         else np.asarray(cf_map, dtype=np.int64)
     n = A.n_rows
-    rows_P: list = [None] * n           # dict coarse->weight per row
-    done = np.zeros(n, dtype=bool)
-    for i in range(n):
-        if cf[i] >= 0:
-            rows_P[i] = {int(cf[i]): 1.0}
-            done[i] = True
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    local = ci < n
+    offd = rows != ci
+    diag = np.zeros(n)
+    dmask = rows == ci
+    diag[rows[dmask]] = v[dmask]
+    weak_lump = np.bincount(rows[offd & ~strong], weights=v[offd & ~strong],
+                            minlength=n)
+    denom = diag + weak_lump
+    sedge = strong & offd & local        # strong edges usable for composing
+    done = cf[:n] >= 0
+    c_rows = np.nonzero(done)[0]
+    P = sp.csr_matrix((np.ones(c_rows.size), (c_rows, cf[c_rows])),
+                      shape=(n, num_coarse))
     for _pass in range(max_passes):
-        progressed = False
-        newly = []
-        for i in range(n):
-            if done[i]:
-                continue
-            s, e = ro[i], ro[i + 1]
-            diag = 0.0
-            acc: Dict[int, float] = {}
-            lump = 0.0
-            ok = False
-            for k in range(s, e):
-                j, a = ci[k], v[k]
-                if j == i:
-                    diag = a
-                    continue
-                if strong[k] and j < n and done[j]:
-                    for cc, wjc in rows_P[j].items():
-                        acc[cc] = acc.get(cc, 0.0) + a * wjc
-                    ok = True
-                else:
-                    lump += a if not strong[k] else 0.0
-            if ok and (diag + lump) != 0.0:
-                newly.append((i, {cc: -aw / (diag + lump)
-                                  for cc, aw in acc.items()}))
-        for i, row in newly:
-            rows_P[i] = row
-            done[i] = True
-            progressed = True
-        if not progressed:
+        undone = ~done
+        if not undone.any():
             break
-    Pc, Pv, indptr = [], [], [0]
-    for i in range(n):
-        row = rows_P[i] or {}
-        for cc in sorted(row):
-            Pc.append(cc)
-            Pv.append(row[cc])
-        indptr.append(indptr[-1] + len(row))
-    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
-                       np.asarray(indptr)), shape=(n, num_coarse))
+        # edges from undone rows into done rows
+        e = sedge & undone[rows] & done[np.minimum(ci, n - 1)] & local
+        if not e.any():
+            break
+        ok_rows = np.zeros(n, dtype=bool)
+        ok_rows[rows[e]] = True
+        ok_rows &= denom != 0.0
+        if not ok_rows.any():
+            break
+        F = sp.csr_matrix((v[e], (rows[e], ci[e])), shape=(n, n))
+        contrib = (F @ P).tocsr()
+        scale = np.where(ok_rows, np.divide(-1.0, denom, out=np.ones(n),
+                                            where=denom != 0.0), 0.0)
+        newP = sp.diags(scale) @ contrib
+        P = (P + newP).tocsr()
+        done = done | ok_rows
+    P.sum_duplicates()
+    P.sort_indices()
+    P.eliminate_zeros()
     return CSRMatrix.from_scipy(P, dtype=A.dtype)
 
 
