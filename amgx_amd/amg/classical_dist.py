@@ -337,14 +337,19 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
     """Distributed distance-2 (standard) interpolation (reference
     src/classical/interpolators/distance2.cu + the 2-ring halo,
     num_import_rings=2): F rows distribute couplings to strong F neighbors
-    over those neighbors' strong C points. Halo F rows are fetched with
-    halo_matrix (one matrix-halo exchange); the C/F status and global coarse
-    ids of the resulting 2-ring columns come from one more HaloExchange of
-    the owners' cf array. Host assembly (like the serial D2 reference);
-    returns a CSRMatrix P with GLOBAL coarse columns on A's device."""
+    over those neighbors' strong C points, magnitude-proportionally. Halo F
+    rows come from one matrix-halo exchange; the C/F status of the resulting
+    2-ring columns from one HaloExchange of the owners' cf array. The
+    through-F pass is one SpGEMM over the owned+halo extended operator; only
+    the boundary-sized halo rows are assembled in a loop. Returns a
+    CSRMatrix P with GLOBAL coarse columns on A's device."""
+    import scipy.sparse as sp
+
     from ..distributed.manager import halo_matrix
     from ..matrix import CSRMatrix
     n = mgr.n_local
+    n_ext = n + mgr.n_halo
+    ngc = int(coarse_offs[-1])
     ro = A.row_offsets.cpu().numpy().astype(np.int64)
     ci = A.col_indices.cpu().numpy().astype(np.int64)
     va = A.values.cpu().numpy().astype(np.float64).reshape(-1)
@@ -359,26 +364,14 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
     hx = HaloExchange(needed2, mgr.part_offsets, device=A.row_offsets.device,
                       owner_local_map=mgr.row_iperm.cpu().numpy()
                       .astype(np.int64))
-    cf_gid_own = torch.full((n,), -1.0, dtype=torch.float64,
-                            device=A.row_offsets.device)
-    cf_gid_own[:] = cf_ext[:n].to(torch.float64)
+    cf_gid_own = cf_ext[:n].to(torch.float64)
     tail = torch.full((max(int(needed2.size), 1),), -1.0,
                       dtype=torch.float64, device=A.row_offsets.device)
     if needed2.size:
         hx.forward(cf_gid_own, tail[:needed2.size])
     cf2 = tail.cpu().numpy().round().astype(np.int64)
-
     iperm = mgr.row_iperm.cpu().numpy().astype(np.int64)
     halo_pos = {int(g): p for p, g in enumerate(mgr.halo_global)}
-    # strength of halo rows: |a| >= theta * rowmax over off-diagonals
-    rowmax_h = np.zeros(mgr.n_halo)
-    for p in range(mgr.n_halo):
-        s0, s1 = ro_h[p], ro_h[p + 1]
-        g = int(mgr.halo_global[p])
-        cc, vv = cols_h[s0:s1], vals_h[s0:s1]
-        off = cc != g
-        if off.any():
-            rowmax_h[p] = np.abs(vv[off]).max()
 
     def cf_any(g):
         """Global coarse id of global fine id g (or -1)."""
@@ -392,76 +385,86 @@ def interp_d2_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
             return int(cf2[k])
         return -1
 
-    Pc, Pv, indptr = [], [], [0]
-    gid_own = mgr.row_perm.cpu().numpy() + lo     # internal -> global
-    for i in range(n):
-        if cfx[i] >= 0:
-            Pc.append(int(cfx[i]))
-            Pv.append(1.0)
-            indptr.append(indptr[-1] + 1)
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    offd = rows != ci
+    cf_col = np.where(ci < cfx.size, cfx[np.minimum(ci, cfx.size - 1)], -1)
+    strongC = strong & offd & (cf_col >= 0)
+    strongF = strong & offd & (cf_col < 0)
+    # W rows for OWNED F rows (vectorized)
+    sC_sum = np.bincount(rows[strongC], weights=np.abs(va[strongC]),
+                         minlength=n) if strongC.any() else np.zeros(n)
+    alive = np.zeros(n_ext, dtype=bool)
+    alive[:n] = sC_sum > 0.0
+    wr = rows[strongC]
+    W_parts_r = [wr]
+    W_parts_c = [cf_col[strongC]]
+    W_parts_v = [np.abs(va[strongC]) / np.where(sC_sum[wr] > 0,
+                                                sC_sum[wr], 1.0)]
+    # W rows for HALO F slots (boundary-sized loop over fetched rows)
+    for p in range(mgr.n_halo):
+        if cfx[n + p] >= 0:      # halo C point: direct, not a W row
             continue
-        s, e = ro[i], ro[i + 1]
-        diag = 0.0
-        acc = {}
-        weak_sum = 0.0
-        for k in range(s, e):
-            j, a = int(ci[k]), va[k]
-            if j == i:
-                diag = a
-                continue
-            if not strong[k]:
-                weak_sum += a
-                continue
-            cfj = int(cfx[j]) if j < cfx.size else -1
-            if cfj >= 0:
-                acc[cfj] = acc.get(cfj, 0.0) + a
-                continue
-            # strong F neighbor: distribute over its strong C points
-            # magnitude-proportional distribution (sums to a_ij exactly;
-            # signed sums can cancel and blow the weights up)
-            cpts, cvals = [], []
-            if j < n:                      # owned F neighbor
-                for kk in range(ro[j], ro[j + 1]):
-                    jj = int(ci[kk])
-                    cfk = int(cfx[jj]) if jj < cfx.size else -1
-                    if strong[kk] and cfk >= 0:
-                        cpts.append(cfk)
-                        cvals.append(abs(va[kk]))
-            else:                          # halo F neighbor: fetched row
-                p = j - n
-                gj = int(mgr.halo_global[p])
-                s0, s1 = ro_h[p], ro_h[p + 1]
-                for kk in range(s0, s1):
-                    gk = int(cols_h[kk])
-                    if gk == gj:
-                        continue
-                    if abs(vals_h[kk]) < theta * rowmax_h[p] \
-                            or rowmax_h[p] == 0.0:
-                        continue
-                    cfk = cf_any(gk)
-                    if cfk >= 0:
-                        cpts.append(cfk)
-                        cvals.append(abs(vals_h[kk]))
-            tot = sum(cvals)
-            if cpts and tot != 0.0:
-                for cc_, av in zip(cpts, cvals):
-                    acc[cc_] = acc.get(cc_, 0.0) + a * av / tot
-            else:
-                weak_sum += a
-        denom = diag + weak_sum
-        if not acc or denom == 0.0:
-            indptr.append(indptr[-1])
+        s0, s1 = int(ro_h[p]), int(ro_h[p + 1])
+        gj = int(mgr.halo_global[p])
+        cc, vv = cols_h[s0:s1], vals_h[s0:s1]
+        off = cc != gj
+        if not off.any():
             continue
-        for cc_ in sorted(acc):
-            Pc.append(cc_)
-            Pv.append(-acc[cc_] / denom)
-        indptr.append(indptr[-1] + len(acc))
+        rmax = np.abs(vv[off]).max()
+        if rmax == 0.0:
+            continue
+        cpts, cvals = [], []
+        for gk, av in zip(cc, vv):
+            if gk == gj or abs(av) < theta * rmax:
+                continue
+            cfk = cf_any(int(gk))
+            if cfk >= 0:
+                cpts.append(cfk)
+                cvals.append(abs(av))
+        tot = sum(cvals)
+        if not cpts or tot == 0.0:
+            continue
+        alive[n + p] = True
+        W_parts_r.append(np.full(len(cpts), n + p, dtype=np.int64))
+        W_parts_c.append(np.asarray(cpts, dtype=np.int64))
+        W_parts_v.append(np.asarray(cvals) / tot)
+    W = sp.csr_matrix(
+        (np.concatenate(W_parts_v), (np.concatenate(W_parts_r),
+                                     np.concatenate(W_parts_c))),
+        shape=(n_ext, ngc))
+    live_edge = strongF & alive[np.minimum(ci, n_ext - 1)] & (ci < n_ext)
+    F = sp.csr_matrix((va[live_edge], (rows[live_edge], ci[live_edge])),
+                      shape=(n, n_ext))
+    D = sp.csr_matrix((va[strongC], (rows[strongC], cf_col[strongC])),
+                      shape=(n, ngc))
+    acc = (D + F @ W).tocsr()
+    acc.sum_duplicates()
+    diag = np.zeros(n)
+    dmask = rows == ci
+    diag[rows[dmask]] = va[dmask]
+    weak = offd & ~strong
+    dead = strongF & ~alive[np.minimum(ci, n_ext - 1)]
+    lump = (np.bincount(rows[weak], weights=va[weak], minlength=n)
+            + np.bincount(rows[dead], weights=va[dead], minlength=n))
+    denom = diag + lump
+    row_nnz = np.diff(acc.indptr)
+    f_ok = (cfx[:n] < 0) & (denom != 0.0) & (row_nnz > 0)
+    scale = np.where(f_ok, np.divide(-1.0, denom, out=np.ones(n),
+                                     where=denom != 0.0), 0.0)
+    acc = (sp.diags(scale) @ acc).tocsr()
+    acc.eliminate_zeros()
+    c_rows = np.nonzero(cfx[:n] >= 0)[0]
+    ident = sp.csr_matrix((np.ones(c_rows.size), (c_rows, cfx[c_rows])),
+                          shape=(n, ngc))
+    P = (acc + ident).tocsr()
+    P.sum_duplicates()
+    P.sort_indices()
     dev = A.row_offsets.device
     return CSRMatrix(
-        torch.as_tensor(np.asarray(indptr, dtype=np.int32)).to(dev),
-        torch.as_tensor(np.asarray(Pc, dtype=np.int32)).to(dev),
-        torch.as_tensor(np.asarray(Pv, dtype=np.float64)).to(A.dtype).to(dev),
-        n_cols=int(coarse_offs[-1]))
+        torch.from_numpy(P.indptr.astype(np.int32)).to(dev),
+        torch.from_numpy(P.indices.astype(np.int32)).to(dev),
+        torch.from_numpy(P.data).to(A.dtype).to(dev),
+        n_cols=ngc)
 
 
 def interp_multipass_dist(A, mgr: DistributedManager,
