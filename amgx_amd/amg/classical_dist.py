@@ -474,88 +474,84 @@ def interp_multipass_dist(A, mgr: DistributedManager,
     src/classical/interpolators/multipass.cu): pass 0 = C rows; each later
     pass interpolates F rows through already-interpolated strong neighbors.
     Neighbor P rows cross ranks by one CSR-row exchange per pass, done flags
-    by one vector exchange per pass."""
+    by one vector exchange per pass; the composition itself is one SpGEMM
+    over the owned+halo extended P."""
+    import scipy.sparse as sp
+
     from ..matrix import CSRMatrix
     n = mgr.n_local
+    n_ext = n + mgr.n_halo
+    ngc = int(coarse_offs[-1])
     ro = A.row_offsets.cpu().numpy().astype(np.int64)
     ci = A.col_indices.cpu().numpy().astype(np.int64)
     va = A.values.cpu().numpy().astype(np.float64).reshape(-1)
     strong = strong_out.cpu().numpy()
     cfx = cf_ext.cpu().numpy().astype(np.int64)
-    rows_P = [None] * n
-    done = np.zeros(n, dtype=bool)
-    for i in range(n):
-        if cfx[i] >= 0:
-            rows_P[i] = {int(cfx[i]): 1.0}
-            done[i] = True
-
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    offd = rows != ci
+    diag = np.zeros(n)
+    dmask = rows == ci
+    diag[rows[dmask]] = va[dmask]
+    weak_lump = np.bincount(rows[offd & ~strong],
+                            weights=va[offd & ~strong], minlength=n)
+    denom = diag + weak_lump
+    sedge = strong & offd & (ci < n_ext)
+    done = np.zeros(n_ext, dtype=bool)
+    done[:n] = cfx[:n] >= 0
+    c_rows = np.nonzero(done[:n])[0]
+    P = sp.csr_matrix((np.ones(c_rows.size), (c_rows, cfx[c_rows])),
+                      shape=(n, ngc))
     done_ext = mgr.new_ext_vec(torch.float64)
     for _pass in range(max_passes):
-        # share done flags and current P rows of the boundary
-        done_ext[:n] = torch.from_numpy(done.astype(np.float64))
+        done_ext[:n] = torch.from_numpy(done[:n].astype(np.float64)) \
+            .to(done_ext.dtype)
         mgr.exchange_halo(done_ext, block_override=1)
-        halo_done = done_ext[n:].cpu().numpy() > 0.5
-        indptr = [0]
-        cols_l, vals_l = [], []
-        for i in range(n):
-            row = rows_P[i] or {}
-            for cc in sorted(row):
-                cols_l.append(cc)
-                vals_l.append(row[cc])
-            indptr.append(len(cols_l))
-        halo_rows = exchange_csr_rows(
-            mgr, np.asarray(indptr, dtype=np.int64),
-            np.asarray(cols_l, dtype=np.int64), np.asarray(vals_l))
-        progressed = False
-        newly = []
-        for i in range(n):
-            if done[i]:
-                continue
-            s, e = ro[i], ro[i + 1]
-            diag = 0.0
-            acc = {}
-            lump = 0.0
-            ok = False
-            for k in range(s, e):
-                j, a = int(ci[k]), va[k]
-                if j == i:
-                    diag = a
-                    continue
-                src = None
-                if strong[k]:
-                    if j < n and done[j]:
-                        src = rows_P[j]
-                    elif j >= n and halo_done[j - n] \
-                            and halo_rows[j - n] is not None:
-                        cc_, vv_ = halo_rows[j - n]
-                        src = dict(zip(cc_.tolist(), vv_.tolist()))
-                if src is not None:
-                    for cc2, wjc in src.items():
-                        acc[int(cc2)] = acc.get(int(cc2), 0.0) + a * wjc
-                    ok = True
-                elif not strong[k]:
-                    lump += a
-            if ok and (diag + lump) != 0.0:
-                newly.append((i, {cc2: -aw / (diag + lump)
-                                  for cc2, aw in acc.items()}))
-        for i, row in newly:
-            rows_P[i] = row
-            done[i] = True
-            progressed = True
-        total_left = mgr.global_sum(float((~done).sum()))
-        any_prog = mgr.global_sum(1.0 if progressed else 0.0)
-        if total_left == 0 or any_prog == 0:
+        done[n:] = done_ext[n:].cpu().numpy() > 0.5
+        # exchange current P rows for the halo slots
+        halo_rows = exchange_csr_rows(mgr, P.indptr.astype(np.int64),
+                                      P.indices.astype(np.int64), P.data)
+        h_counts = np.asarray([0 if r is None else r[0].size
+                               for r in halo_rows], dtype=np.int64)
+        tot = int(h_counts.sum())
+        if tot:
+            h_cols = np.concatenate([r[0] for r in halo_rows
+                                     if r is not None])
+            h_vals = np.concatenate([r[1] for r in halo_rows
+                                     if r is not None])
+        else:
+            h_cols = np.zeros(0, dtype=np.int64)
+            h_vals = np.zeros(0)
+        ext_indptr = np.concatenate([P.indptr.astype(np.int64),
+                                     int(P.indptr[-1]) + np.cumsum(h_counts)])
+        P_ext = sp.csr_matrix(
+            (np.concatenate([P.data, h_vals]),
+             np.concatenate([P.indices.astype(np.int64), h_cols]),
+             ext_indptr), shape=(n_ext, ngc))
+        undone = ~done[:n]
+        total_left = mgr.global_sum(float(undone.sum()))
+        if total_left == 0:
             break
-    Pc, Pv, indptr = [], [], [0]
-    for i in range(n):
-        row = rows_P[i] or {}
-        for cc in sorted(row):
-            Pc.append(cc)
-            Pv.append(row[cc])
-        indptr.append(indptr[-1] + len(row))
+        e = sedge & undone[rows] & done[np.minimum(ci, n_ext - 1)]
+        ok_rows = np.zeros(n, dtype=bool)
+        ok_rows[rows[e]] = True
+        ok_rows &= denom != 0.0
+        any_prog = mgr.global_sum(float(ok_rows.sum()))
+        if any_prog == 0:
+            break
+        if ok_rows.any():
+            F = sp.csr_matrix((va[e], (rows[e], ci[e])), shape=(n, n_ext))
+            contrib = (F @ P_ext).tocsr()
+            scale = np.where(ok_rows, np.divide(-1.0, denom,
+                                                out=np.ones(n),
+                                                where=denom != 0.0), 0.0)
+            P = (P + sp.diags(scale) @ contrib).tocsr()
+            done[:n] |= ok_rows
+    P.sum_duplicates()
+    P.sort_indices()
+    P.eliminate_zeros()
     dev = A.row_offsets.device
     return CSRMatrix(
-        torch.as_tensor(np.asarray(indptr, dtype=np.int32)).to(dev),
-        torch.as_tensor(np.asarray(Pc, dtype=np.int32)).to(dev),
-        torch.as_tensor(np.asarray(Pv, dtype=np.float64)).to(A.dtype).to(dev),
-        n_cols=int(coarse_offs[-1]))
+        torch.from_numpy(P.indptr.astype(np.int32)).to(dev),
+        torch.from_numpy(P.indices.astype(np.int32)).to(dev),
+        torch.from_numpy(P.data).to(A.dtype).to(dev),
+        n_cols=ngc)
