@@ -72,22 +72,39 @@ class AggregationLevel(AMGLevel):
         from .aggregation import AGG_SELECTOR_REGISTRY
         selector = self.scope.get("selector") or "SIZE_2"
         mgr = getattr(self.A, "manager", None)
+        fn = AGG_SELECTOR_REGISTRY.get(selector)
+        if fn is None:
+            raise KeyError(f"unknown aggregation selector {selector!r}; "
+                           f"known: {sorted(AGG_SELECTOR_REGISTRY)}")
         if mgr is not None:
-            # distributed: rank-local pairwise matching (halo columns are
-            # excluded from matching; remote coupling enters via the coarse
-            # level's own halo). Multi-pass selectors collapse to one pass.
-            maxit = self.scope.get("max_matching_iterations")
-            agg, num = ops.size2_matching(self.A, max_iterations=maxit)
+            # distributed: selection is rank-local (reference
+            # setAggregates operates on the owned partition; remote coupling
+            # enters via the coarse level's own halo) — run the SAME
+            # selector, passes included, on the halo-dropped local view
+            agg, num = fn(self._local_square_view(), self.scope)
         else:
-            fn = AGG_SELECTOR_REGISTRY.get(selector)
-            if fn is None:
-                raise KeyError(f"unknown aggregation selector {selector!r}; "
-                               f"known: {sorted(AGG_SELECTOR_REGISTRY)}")
             agg, num = fn(self.A, self.scope)
         self.aggregates = agg.to(self.A.row_offsets.device)
         self.num_aggregates = num
         self._build_r_structure()
         return num
+
+    def _local_square_view(self) -> CSRMatrix:
+        """Owned rows with halo columns dropped (device-vectorized)."""
+        A = self.A
+        n = A.n_rows
+        ro = A.row_offsets.to(torch.int64)
+        keep = A.col_indices.to(torch.int64) < n
+        rows = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=keep.device),
+            ro[1:] - ro[:-1])
+        counts = torch.bincount(rows[keep], minlength=n)
+        ro2 = torch.zeros(n + 1, dtype=torch.int64, device=keep.device)
+        torch.cumsum(counts, 0, out=ro2[1:])
+        return CSRMatrix(ro2.to(torch.int32),
+                         A.col_indices[keep].contiguous(),
+                         A.values[keep].contiguous(),
+                         n_cols=n, block_dim=A.block_dim)
 
     def _build_r_structure(self):
         """Aggregate-CSR (offsets, fine ids sorted by aggregate) for

@@ -740,3 +740,43 @@ def _impl_test_dist_classical_multipass(rank, world, tmp):
     nrm = mgr.global_norm(float(torch.linalg.vector_norm(
         r[:mgr.owned_size])), "L2")
     assert nrm < 1e-5
+
+
+def test_dist_agg_multipass_selectors():
+    _run_dist(test_dist_agg_multipass_selectors)
+
+
+def _impl_test_dist_agg_multipass_selectors(rank, world, tmp):
+    """SIZE_4/SIZE_8/MULTI_PAIRWISE run their full pass composition on the
+    rank-local view in distributed mode (no silent SIZE_2 collapse)."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    for sel in ("SIZE_4", "SIZE_8", "MULTI_PAIRWISE"):
+        cfg = AMGConfig.from_dict({"solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "AGGREGATION",
+                "selector": sel, "aggregate_size": 4,
+                "smoother": "MULTICOLOR_DILU", "presweeps": 0,
+                "postsweeps": 3, "relaxation_factor": 0.75,
+                "max_iters": 1, "min_coarse_rows": 16, "cycle": "V",
+                "scope": "amg",
+            },
+            "solver": "FGMRES", "max_iters": 120, "gmres_n_restart": 25,
+            "monitor_residual": 1, "convergence": "RELATIVE_INI",
+            "tolerance": 1e-7,
+        }})
+        A = _make_dist_A(rank, world, 7)
+        mgr = A.manager
+        s = create_solver(cfg.root_scope(), resources=Resources(
+            "cpu", distributed=True))
+        b = mgr.new_ext_vec(torch.float64)
+        b[:mgr.owned_size] = 1.0
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        assert st.converged, f"{sel} rank {rank}: {st}"
+        # SIZE_4/8 must actually coarsen faster than SIZE_2 would
+        h = s.precond.hierarchy
+        if sel in ("SIZE_4", "SIZE_8"):
+            ratio = h.levels[0].A.n_rows / max(h.levels[1].A.n_rows, 1)
+            assert ratio > 2.5, f"{sel}: ratio {ratio}"
