@@ -555,3 +555,121 @@ def interp_multipass_dist(A, mgr: DistributedManager,
         torch.from_numpy(P.indices.astype(np.int32)).to(dev),
         torch.from_numpy(P.data).to(A.dtype).to(dev),
         n_cols=ngc)
+
+
+def aggressive_thin_dist(A, mgr: DistributedManager,
+                         strong_union: torch.Tensor, cf: torch.Tensor):
+    """Distributed aggressive second pass (reference aggressive_pmis.cu):
+    thin the PMIS C set by an independent set over the C-C graph connected
+    by strong paths of length <= 2. Paths through halo intermediates use the
+    fetched halo rows (2-ring); cross-rank MIS state syncs through a
+    HaloExchange over exactly the remote C ids involved. Returns the thinned
+    cf (local coarse renumbered) and count."""
+    n = mgr.n_local
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    strong = strong_union.cpu().numpy()
+    cfx = cf.cpu().numpy()
+    lo = mgr.row_start
+    gid_own = mgr.row_perm.cpu().numpy() + lo          # internal -> global
+    rows = np.repeat(np.arange(n), np.diff(ro))
+    er, ec = rows[strong], ci[strong]
+    is_c = cfx >= 0
+    # C status of owned + halo columns
+    c_ext = np.zeros(n + mgr.n_halo, dtype=bool)
+    c_ext[:n] = is_c
+    st_ext = mgr.new_ext_vec(torch.float64)
+    st_ext[:n] = torch.from_numpy(is_c.astype(np.float64))
+    mgr.exchange_halo(st_ext, block_override=1)
+    c_ext[n:] = st_ext[n:].cpu().numpy() > 0.5
+    gid_ext = np.concatenate([gid_own, mgr.halo_global]) if mgr.n_halo \
+        else gid_own
+    # 1-hop C-C edges (i C, j C): (gid_i, gid_j)
+    e1 = is_c[er] & c_ext[ec]
+    pairs = [(gid_own[er[e1]], gid_ext[ec[e1]])]
+    # 2-hop pairs via one SpGEMM: M[k, c] = 1 when intermediate k (owned or
+    # halo) strongly couples to C node c; pairs = nonzeros of M^T M
+    import scipy.sparse as sp
+    c_gid_parts = [gid_own[is_c]]
+    if mgr.n_halo:
+        c_gid_parts.append(gid_ext[n:][c_ext[n:]])
+    c_gid_all = np.unique(np.concatenate(c_gid_parts)) if c_gid_parts else \
+        np.zeros(0, dtype=np.int64)
+    nC = int(c_gid_all.size)
+    mr, mc = [], []
+    # (a) strong edge i -> k with i owned C: incidence at (k, gid_i)
+    a_mask = is_c[er]
+    mr.append(ec[a_mask])
+    mc.append(np.searchsorted(c_gid_all, gid_own[er[a_mask]]))
+    # (b) strong entry k -> j with k owned, j C: incidence at (k, gid_j)
+    b_mask = c_ext[ec]
+    mr.append(er[b_mask])
+    mc.append(np.searchsorted(c_gid_all, gid_ext[ec[b_mask]]))
+    M = sp.csr_matrix((np.ones(sum(x.size for x in mr), dtype=np.int8),
+                       (np.concatenate(mr), np.concatenate(mc))),
+                      shape=(n + mgr.n_halo, max(nC, 1)))
+    CC = (M.T @ M).tocoo()
+    keep2 = CC.row != CC.col
+    pairs.append((c_gid_all[CC.row[keep2]], c_gid_all[CC.col[keep2]]))
+    pa = np.concatenate([p[0] for p in pairs]) if pairs else \
+        np.zeros(0, dtype=np.int64)
+    pb = np.concatenate([p[1] for p in pairs]) if pairs else \
+        np.zeros(0, dtype=np.int64)
+    own_a = (pa >= lo) & (pa < lo + n)
+    pa, pb = pa[own_a], pb[own_a]           # keep edges whose LEFT is owned
+    # MIS over C nodes with weight hash01(gid): remote state via exchange
+    remote = np.unique(pb[(pb < lo) | (pb >= lo + n)])
+    hxr = HaloExchange(remote, mgr.part_offsets,
+                       device=A.row_offsets.device,
+                       owner_local_map=mgr.row_iperm.cpu().numpy()
+                       .astype(np.int64))
+    iperm = mgr.row_iperm.cpu().numpy().astype(np.int64)
+    la = iperm[pa - lo]                      # owned internal index
+    w_own = _hash01(torch.from_numpy(gid_own)).numpy()
+    w_b = _hash01(torch.from_numpy(pb)).numpy()
+    state = np.where(is_c, 0.0, -1.0)        # 0 undecided-C, 1 keep, -1 out
+    st_t = torch.zeros(n, dtype=torch.float64)
+    tailbuf = torch.zeros(max(int(remote.size), 1), dtype=torch.float64)
+    guard = 0
+    while True:
+        undec = state == 0.0
+        total = mgr.global_sum(float(undec.sum()))
+        if total == 0:
+            break
+        guard += 1
+        if guard > 10 * max(1, int(np.log2(mgr.n_global + 2)) + 8):
+            raise RuntimeError("aggressive MIS failed to converge")
+        st_t[:] = torch.from_numpy(state)
+        if remote.size:
+            hxr.forward(st_t, tailbuf[:remote.size])
+        rstate = tailbuf.numpy()
+        bpos = np.searchsorted(remote, pb)
+        b_remote = (pb < lo) | (pb >= lo + n)
+        sb = np.where(b_remote, rstate[np.minimum(bpos, max(remote.size - 1,
+                                                            0))],
+                      state[iperm[np.clip(pb - lo, 0, n - 1)]])
+        act = undec[la] & (sb == 0.0)
+        beaten = np.zeros(n, dtype=bool)
+        if act.any():
+            wa = w_own[la[act]]
+            wbv = w_b[act]
+            ga, gb = pa[act], pb[act]
+            loses = (wbv > wa) | ((wbv == wa) & (gb > ga))
+            np.logical_or.at(beaten, la[act], loses)
+        new_keep = undec & ~beaten
+        state[new_keep] = 1.0
+        # sync then drop undecided neighbors of kept nodes
+        st_t[:] = torch.from_numpy(state)
+        if remote.size:
+            hxr.forward(st_t, tailbuf[:remote.size])
+        rstate = tailbuf.numpy()
+        sb = np.where(b_remote, rstate[np.minimum(bpos, max(remote.size - 1,
+                                                            0))],
+                      state[iperm[np.clip(pb - lo, 0, n - 1)]])
+        drop_edges = (state[la] == 0.0) & (sb == 1.0)
+        state[np.unique(la[drop_edges])] = -1.0
+    keep = state == 1.0
+    cf2 = np.full(n, -1, dtype=np.int32)
+    krows = np.nonzero(keep)[0]
+    cf2[krows] = np.arange(krows.size, dtype=np.int32)
+    return torch.from_numpy(cf2).to(cf.device), int(krows.size)

@@ -207,16 +207,17 @@ class ClassicalLevel(AMGLevel):
     def create_coarse_vertices(self) -> int:
         mgr = getattr(self.A, "manager", None)
         if mgr is not None:
-            from .classical_dist import pmis_dist, strength_dist
-            if self.index < int(self.scope.get("aggressive_levels") or 0):
-                import warnings
-                warnings.warn("distributed aggressive coarsening is not "
-                              "wired; using plain PMIS on this level")
+            from .classical_dist import (aggressive_thin_dist, pmis_dist,
+                                         strength_dist)
             theta = float(self.scope.get("strength_threshold"))
             mrs = float(self.scope.get("max_row_sum"))
             self._strong_out, strong_union = strength_dist(
                 self.A, mgr, theta, mrs)
             cf, nc = pmis_dist(self.A, mgr, strong_union)
+            self._aggressive = self.index < int(
+                self.scope.get("aggressive_levels") or 0)
+            if self._aggressive:
+                cf, nc = aggressive_thin_dist(self.A, mgr, strong_union, cf)
             self.cf_map = cf          # device tensor
             self.num_coarse = nc
             self.A._cache["cf_map"] = self.cf_map
@@ -294,6 +295,8 @@ class ClassicalLevel(AMGLevel):
                                                self.num_coarse)
         interp = self.scope.get("interpolator") \
             if self.scope.has("interpolator") else "D1"
+        if getattr(self, "_aggressive", False):
+            interp = self.scope.get("aggressive_interpolator") or "MULTIPASS"
         if interp == "D2":
             # distance-2 through the 2-ring (halo-row fetch)
             from .classical_dist import interp_d2_dist

@@ -780,3 +780,46 @@ def _impl_test_dist_agg_multipass_selectors(rank, world, tmp):
         if sel in ("SIZE_4", "SIZE_8"):
             ratio = h.levels[0].A.n_rows / max(h.levels[1].A.n_rows, 1)
             assert ratio > 2.5, f"{sel}: ratio {ratio}"
+
+
+def test_dist_aggressive_classical():
+    _run_dist(test_dist_aggressive_classical)
+
+
+def _impl_test_dist_aggressive_classical(rank, world, tmp):
+    """Distributed aggressive coarsening: the C-C strong-2 MIS thins the
+    PMIS split consistently across ranks (no adjacent kept C across the
+    boundary), and MULTIPASS interpolation carries the aggressive level."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+
+    def solve(aggr):
+        cfg = AMGConfig.from_dict({"solver": {
+            "preconditioner": {
+                "solver": "AMG", "algorithm": "CLASSICAL",
+                "aggressive_levels": aggr,
+                "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+                "max_iters": 1, "min_coarse_rows": 12, "cycle": "V",
+                "scope": "amg",
+            },
+            "solver": "PCG", "max_iters": 120, "monitor_residual": 1,
+            "convergence": "RELATIVE_INI", "tolerance": 1e-7,
+        }})
+        A = _make_dist_A(rank, world, 8)
+        mgr = A.manager
+        s = create_solver(cfg.root_scope(), resources=Resources(
+            "cpu", distributed=True))
+        b = mgr.new_ext_vec(torch.float64)
+        b[:mgr.owned_size] = 1.0
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        assert st.converged, f"aggr={aggr} rank {rank}: {st}"
+        h = s.precond.hierarchy
+        lvl1 = mgr.global_sum(float(h.levels[1].A.n_rows)) \
+            if len(h.levels) > 1 else mgr.n_global
+        return lvl1
+
+    plain = solve(0)
+    aggr = solve(1)
+    assert aggr < plain, f"aggressive {aggr} !< plain {plain}"
