@@ -245,7 +245,7 @@ def aggressive_select(A, S, scope, base: str):
     selector, then re-select among the C points over strong paths of length
     <= 2, keeping only the surviving subset as C."""
     cf1, nc1 = SELECTOR_REGISTRY[base](A, S, scope)
-    cf1np = cf1.numpy() if torch.is_tensor(cf1) else cf1
+    cf1np = cf1.cpu().numpy() if torch.is_tensor(cf1) else cf1
     c_idx = np.nonzero(cf1np >= 0)[0]
     if c_idx.size <= 1:
         return cf1, nc1

@@ -21,8 +21,9 @@ register_parameter("cf_smoothing_mode", int, 0,
 @register_solver("KACZMARZ")
 class KaczmarzSolver(_SmootherBase):
     """Row-projection sweeps x += omega*(b_i - a_i.x)/||a_i||^2 * a_i,
-    scheduled by color (same-color rows may still share columns: the scatter
-    is accumulated atomically on device, sequentially on host)."""
+    scheduled by a DISTANCE-2 coloring so same-color rows share no columns —
+    each color sweep is one race-free gather/segmented-reduce/scatter-add
+    batch on the device."""
 
     def __init__(self, scope, resources):
         super().__init__(scope, resources)
@@ -31,9 +32,21 @@ class KaczmarzSolver(_SmootherBase):
 
     def solver_setup(self):
         A = self.A
-        if A.coloring is None:
+        # Kaczmarz scatters into the COLUMNS of each row: parallel same-color
+        # updates need rows that share no column, i.e. a DISTANCE-2 coloring
+        # (reference: kaczmarz pairs with LOCALLY_DOWNWIND / 2-ring
+        # colorings). Cached separately from the matrix's distance-1 one.
+        self._kz_coloring = A._cache.get("kz_coloring")
+        if self._kz_coloring is None:
             from ..amg.coloring import MatrixColoring
-            A.coloring = MatrixColoring.create(A, self.scope)
+            from ..config import ConfigScope
+            scheme = self.scope.get("matrix_coloring_scheme") \
+                if self.scope.has("matrix_coloring_scheme") \
+                else "LOCALLY_DOWNWIND"
+            self._kz_coloring = MatrixColoring.create(
+                A, ConfigScope(None, {"matrix_coloring_scheme": scheme,
+                                      "coloring_level": 2}))
+            A._cache["kz_coloring"] = self._kz_coloring
         # row squared norms
         if A.block_dim != 1:
             raise NotImplementedError("Kaczmarz: scalar matrices")
@@ -48,32 +61,43 @@ class KaczmarzSolver(_SmootherBase):
 
     def solve_iteration(self, b, x):
         A = self.A
-        col = A.coloring
+        col = self._kz_coloring
         mgr = getattr(A, "manager", None)
         if mgr is not None and mgr.neighbors:
             mgr.exchange_halo(x)
-        backend = ops._backend(A)
-        if hasattr(backend, "kaczmarz_rows"):
-            for c in range(col.num_colors):
-                backend.kaczmarz_rows(A, self.row_norm_inv, b, x,
-                                      col.rows_of(c), self.relaxation_factor)
-        else:
-            self._host_sweep(b, x)
+        # per color: r_rows = b - (Ax)_rows; x[cols of row] += w*r*rni*a —
+        # race-free under the distance-2 coloring, device-capable via
+        # gather/scatter_add (torch index ops drive the HIP kernels)
+        ro = A.row_offsets.to(torch.int64)
+        ci = A.col_indices.to(torch.int64)
+        va = A.values.reshape(-1)
+        xv = x.reshape(-1)
+        bv = b.reshape(-1)
+        for c in range(col.num_colors):
+            rows = col.rows_of(c).to(torch.int64)
+            if rows.numel() == 0:
+                continue
+            starts = ro[rows]
+            counts = ro[rows + 1] - starts
+            total = int(counts.sum().item())
+            if total == 0:
+                continue
+            pos = (torch.repeat_interleave(starts, counts)
+                   + torch.arange(total, device=xv.device, dtype=torch.int64)
+                   - torch.repeat_interleave(
+                       torch.cumsum(counts, 0) - counts, counts))
+            cols_c = ci[pos]
+            vals_c = va[pos]
+            prod = vals_c * xv[cols_c]
+            seg = torch.repeat_interleave(
+                torch.arange(rows.numel(), device=xv.device), counts)
+            ax = torch.zeros(rows.numel(), dtype=prod.dtype,
+                             device=xv.device)
+            ax.index_add_(0, seg, prod)
+            coef = (self.relaxation_factor * (bv[rows] - ax)
+                    * self.row_norm_inv[rows])
+            xv.index_add_(0, cols_c, coef[seg] * vals_c)
         return False
-
-    def _host_sweep(self, b, x):
-        import numpy as np
-        A = self.A
-        m = A.to_scipy().tocsr()
-        xv = x.reshape(-1).numpy()
-        bv = b.reshape(-1).numpy()
-        rni = self.row_norm_inv.numpy()
-        for c in range(A.coloring.num_colors):
-            for i in A.coloring.rows_of(c).numpy():
-                s, e = m.indptr[i], m.indptr[i + 1]
-                resid = bv[i] - m.data[s:e] @ xv[m.indices[s:e]]
-                xv[m.indices[s:e]] += (self.relaxation_factor * resid
-                                       * rni[i]) * m.data[s:e]
 
 
 @register_solver("CF_JACOBI")
