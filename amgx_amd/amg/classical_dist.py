@@ -627,9 +627,11 @@ def aggressive_thin_dist(A, mgr: DistributedManager,
     la = iperm[pa - lo]                      # owned internal index
     w_own = _hash01(torch.from_numpy(gid_own)).numpy()
     w_b = _hash01(torch.from_numpy(pb)).numpy()
+    dev = A.row_offsets.device
     state = np.where(is_c, 0.0, -1.0)        # 0 undecided-C, 1 keep, -1 out
-    st_t = torch.zeros(n, dtype=torch.float64)
-    tailbuf = torch.zeros(max(int(remote.size), 1), dtype=torch.float64)
+    st_t = torch.zeros(n, dtype=torch.float64, device=dev)
+    tailbuf = torch.zeros(max(int(remote.size), 1), dtype=torch.float64,
+                          device=dev)
     guard = 0
     while True:
         undec = state == 0.0
@@ -639,10 +641,10 @@ def aggressive_thin_dist(A, mgr: DistributedManager,
         guard += 1
         if guard > 10 * max(1, int(np.log2(mgr.n_global + 2)) + 8):
             raise RuntimeError("aggressive MIS failed to converge")
-        st_t[:] = torch.from_numpy(state)
+        st_t[:] = torch.from_numpy(state).to(dev)
         if remote.size:
             hxr.forward(st_t, tailbuf[:remote.size])
-        rstate = tailbuf.numpy()
+        rstate = tailbuf.cpu().numpy()
         bpos = np.searchsorted(remote, pb)
         b_remote = (pb < lo) | (pb >= lo + n)
         sb = np.where(b_remote, rstate[np.minimum(bpos, max(remote.size - 1,
@@ -659,10 +661,10 @@ def aggressive_thin_dist(A, mgr: DistributedManager,
         new_keep = undec & ~beaten
         state[new_keep] = 1.0
         # sync then drop undecided neighbors of kept nodes
-        st_t[:] = torch.from_numpy(state)
+        st_t[:] = torch.from_numpy(state).to(dev)
         if remote.size:
             hxr.forward(st_t, tailbuf[:remote.size])
-        rstate = tailbuf.numpy()
+        rstate = tailbuf.cpu().numpy()
         sb = np.where(b_remote, rstate[np.minimum(bpos, max(remote.size - 1,
                                                             0))],
                       state[iperm[np.clip(pb - lo, 0, n - 1)]])
