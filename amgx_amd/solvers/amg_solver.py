@@ -108,8 +108,11 @@ class AMGSolver(Solver):
         self._first = zero_initial_guess
 
     def solve(self, b, x, zero_initial_guess=False):
-        # preconditioner fast path: one (or a few) cycles, no monitoring
-        if not self.monitor_residual:
+        # preconditioner fast path: one (or a few) cycles, no monitoring.
+        # With a scaler configured, the base path must wrap the cycles so
+        # the system is scaled/unscaled correctly.
+        if not self.monitor_residual and getattr(self, "scaler", None) \
+                is None:
             self._first = zero_initial_guess
             for _ in range(self.max_iters):
                 self.solve_iteration(b, x)

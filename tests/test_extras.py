@@ -449,3 +449,29 @@ def test_singular_neumann_consistent():
     st = s.solve(bn, x, zero_initial_guess=True)
     assert st.converged
     assert float(ops.nrm2(ops.residual(An, x, bn))) < 1e-6
+
+
+def test_scaler_with_standalone_amg():
+    """Scaling applies on the AMG-as-solver fast path too (the
+    no-monitoring branch must not bypass the scaler)."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    A = poisson_2d(12, 12)
+    vals0 = A.values.clone()
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "AMG", "algorithm": "AGGREGATION",
+        "smoother": "BLOCK_JACOBI", "max_iters": 60,
+        "min_coarse_rows": 8, "cycle": "V",
+        "scaling": "DIAGONAL_SYMMETRIC",
+    }})
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    s.solve(b, x, zero_initial_guess=True)
+    assert torch.allclose(A.values, vals0, atol=1e-14)   # matrix restored
+    r = float(ops.nrm2(ops.residual(A, x, b)))
+    assert r < 1e-6, r
