@@ -165,6 +165,9 @@ class FGMRESSolver(Solver):
     def __init__(self, scope, resources):
         super().__init__(scope, resources)
         self.restart = scope.get("gmres_n_restart")
+        # truncated orthogonalization window (reference gmres_krylov_dim,
+        # src/core.cu:391; 0 = full restart-length orthogonalization)
+        self.krylov_dim = int(scope.get("gmres_krylov_dim") or 0)
 
     def solver_setup(self):
         # keep the preconditioner object across resetup so its hierarchy can
@@ -218,8 +221,10 @@ class FGMRESSolver(Solver):
         self.Z.append(z)
         w = self.new_vec(vj)
         ops.spmv(self.A, z, w)
-        # modified Gram-Schmidt: h_ij = <v_i, w> (conjugated in the first arg)
-        for i in range(j + 1):
+        # modified Gram-Schmidt: h_ij = <v_i, w> (conjugated in the first
+        # arg); truncated to the last krylov_dim basis vectors when set
+        i0 = max(0, j + 1 - self.krylov_dim) if self.krylov_dim > 0 else 0
+        for i in range(i0, j + 1):
             hij = self.dot(self.V[i], w)
             self.H[i, j] = hij
             ops.axpy(w, self.V[i], -hij)

@@ -331,3 +331,22 @@ def test_min_iters_and_l1_scaled_norm():
     assert st2.converged
     r = ops.residual(A, x2, b)
     assert float(ops.nrm1(r)) / A.n_rows < 1e-9
+
+
+def test_truncated_fgmres():
+    """gmres_krylov_dim truncates the orthogonalization window (reference
+    src/core.cu:391) — still solves SPD Poisson with AMG preconditioning."""
+    A = poisson_3d(8, 8, 8)
+    s = make({"solver": "FGMRES", "max_iters": 150, "gmres_n_restart": 50,
+              "gmres_krylov_dim": 4, "monitor_residual": 1,
+              "tolerance": 1e-8, "convergence": "RELATIVE_INI",
+              "preconditioner": {"solver": "AMG",
+                                 "algorithm": "AGGREGATION",
+                                 "smoother": "BLOCK_JACOBI", "max_iters": 1,
+                                 "min_coarse_rows": 16, "scope": "amg"}})
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+    assert st.converged and rel < 1e-7
