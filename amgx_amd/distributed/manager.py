@@ -451,17 +451,22 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
     """
     indptr = np.asarray(indptr, dtype=np.int64)
     counts_all = np.diff(indptr)
+    # payload tensors must live where the comm backend wants them (NCCL =
+    # device buffers over xGMI; gloo = host)
+    comm_dev = mgr.device if (dist.is_initialized()
+                              and dist.get_backend() == "nccl") else \
+        torch.device("cpu")
     # phase 1: per-row nnz counts
     p2p = []
     send_counts, recv_counts = [], []
     for i, r in enumerate(mgr.neighbors):
         rows = mgr.b2l[i].cpu().numpy().astype(np.int64)
-        sc = torch.from_numpy(counts_all[rows].astype(np.int64))
+        sc = torch.from_numpy(counts_all[rows].astype(np.int64)).to(comm_dev)
         send_counts.append((rows, sc))
         if sc.numel():
             p2p.append(dist.P2POp(dist.isend, sc, r))
         lo, hi = mgr.halo_slices[i]
-        rc = torch.empty(hi - lo, dtype=torch.int64)
+        rc = torch.empty(hi - lo, dtype=torch.int64, device=comm_dev)
         recv_counts.append(rc)
         if rc.numel():
             p2p.append(dist.P2POp(dist.irecv, rc, r))
@@ -477,15 +482,17 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
         nz = np.concatenate([np.arange(indptr[j], indptr[j + 1])
                              for j in rows]) if rows.size else \
             np.zeros(0, dtype=np.int64)
-        scol = torch.from_numpy(np.asarray(indices, dtype=np.int64)[nz])
-        sval = torch.from_numpy(np.asarray(data, dtype=np.float64)[nz])
+        scol = torch.from_numpy(np.asarray(indices, dtype=np.int64)[nz]) \
+            .to(comm_dev)
+        sval = torch.from_numpy(np.asarray(data, dtype=np.float64)[nz]) \
+            .to(comm_dev)
         keep += [scol, sval]
         if scol.numel():
             p2p.append(dist.P2POp(dist.isend, scol, r))
             p2p.append(dist.P2POp(dist.isend, sval, r))
         tot = int(recv_counts[i].sum())
-        rcol = torch.empty(tot, dtype=torch.int64)
-        rval = torch.empty(tot, dtype=torch.float64)
+        rcol = torch.empty(tot, dtype=torch.int64, device=comm_dev)
+        rval = torch.empty(tot, dtype=torch.float64, device=comm_dev)
         recv_payload.append((rcol, rval))
         if tot:
             p2p.append(dist.P2POp(dist.irecv, rcol, r))
@@ -497,9 +504,9 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
     out = [None] * mgr.n_halo
     for i in range(len(mgr.neighbors)):
         lo, hi = mgr.halo_slices[i]
-        rc = recv_counts[i].numpy()
+        rc = recv_counts[i].cpu().numpy()
         rcol, rval = recv_payload[i]
-        rcol, rval = rcol.numpy(), rval.numpy()
+        rcol, rval = rcol.cpu().numpy(), rval.cpu().numpy()
         pos = 0
         for k in range(hi - lo):
             c = int(rc[k])
