@@ -283,7 +283,7 @@ class AMGHierarchy:
     def grid_stats(self) -> str:
         """Reference AMG::getGridStatisticsString (src/amg.cu:1230-1330)."""
         lines = ["AMG Grid:", "   Number of Levels: %d" % len(self.levels),
-                 "      LVL        ROWS         NNZ    SPRSTY"]
+                 "      LVL        ROWS         NNZ    SPRSTY   MEM(MB)"]
         total_rows = total_nnz = 0
         fine_rows = self.levels[0].A.n_rows
         fine_nnz = self.levels[0].A.nnz
@@ -292,7 +292,11 @@ class AMGHierarchy:
             total_rows += n
             total_nnz += nnz
             sp = nnz / (n * n) if n else 0.0
-            lines.append(f"      {i:3d} {n:11d} {nnz:11d}  {sp:8.3g}")
+            bd = l.A.block_dim
+            mem = (nnz * (bd * bd * l.A.values.element_size() + 4)
+                   + (n + 1) * 4) / 1e6
+            lines.append(f"      {i:3d} {n:11d} {nnz:11d}  {sp:8.3g}"
+                         f"  {mem:8.1f}")
         lines.append(f"      Grid Complexity: {total_rows / max(fine_rows, 1):.5g}")
         lines.append(f"      Operator Complexity: {total_nnz / max(fine_nnz, 1):.5g}")
         return "\n".join(lines)

@@ -219,8 +219,15 @@ def AMGX_config_add_parameters(cfg: _ConfigHandle, options: str):
 
 @_amgx_try
 def AMGX_config_get_default_number_of_rings(cfg: _ConfigHandle):
-    # reference: 1 ring unless aggregation-style configs request 2
-    return RC_OK, 1
+    """Reference semantics (src/amgx_c.cu): aggregation-style compositions
+    ask for 2 import rings, everything else 1."""
+    def _has_agg(node):
+        if not isinstance(node, dict):
+            return False
+        if node.get("algorithm") == "AGGREGATION":
+            return True
+        return any(_has_agg(v) for v in node.values())
+    return RC_OK, 2 if _has_agg(cfg.cfg.tree) else 1
 
 
 @_amgx_try
