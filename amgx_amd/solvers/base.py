@@ -162,6 +162,28 @@ class Solver:
 
     def compute_norm(self, r: torch.Tensor) -> float:
         r = self._owned(r)
+        bd = self.A.block_dim if self.A is not None else 1
+        if bd > 1 and not self.scope.get("use_scalar_norm"):
+            # blocked norms (reference src/norm.cu per-block-component
+            # norms): converge on the worst component
+            import torch.distributed as tdist
+            comp = r.reshape(-1, bd).abs().double()
+            mgr = getattr(self.A, "manager", None)
+            if self.norm == "L1":
+                per = comp.sum(0)
+                if mgr is not None:
+                    tdist.all_reduce(per)
+            elif self.norm == "LMAX":
+                per = comp.amax(0) if comp.numel() else \
+                    torch.zeros(bd, dtype=torch.float64, device=r.device)
+                if mgr is not None:
+                    tdist.all_reduce(per, op=tdist.ReduceOp.MAX)
+            else:                      # L2 default
+                per = (comp * comp).sum(0)
+                if mgr is not None:
+                    tdist.all_reduce(per)
+                per = per.sqrt()
+            return float(per.max())
         if self.norm == "L2":
             nrm = ops.nrm2(r)
         elif self.norm == "L1":
