@@ -86,34 +86,47 @@ class PageRankOperator(Operator):
 
     def __init__(self, A, damping: float = 0.85):
         super().__init__(A)
-        if getattr(A, "manager", None) is not None:
-            # a rank-local transpose is not the global transpose; the
-            # add_from_halo-composed distributed A^T apply is future work
-            raise NotImplementedError(
-                "distributed PageRank needs the transpose-apply exchange")
+        if A.block_dim != 1:
+            raise ValueError("PageRank operates on scalar matrices")
         self.d = damping
-        # column-stochastic normalization: P = A D_out^{-1};
-        # out-degree = row sums of A
-        deg = ops.spmv(A, torch.ones(self.n, dtype=A.dtype, device=A.device))
+        self.mgr = getattr(A, "manager", None)
+        # column-stochastic normalization: P[j,i] = A[i,j] / outdeg(i).
+        # out-degree = row sums of A; full rows are rank-local even in the
+        # distributed partition, so no communication is needed here
+        cnt = (A.row_offsets[1:] - A.row_offsets[:-1]).to(torch.int64)
+        rid = torch.repeat_interleave(
+            torch.arange(A.n_rows, device=A.device), cnt)
+        deg = torch.zeros(A.n_rows, dtype=A.dtype, device=A.device)
+        deg.index_add_(0, rid, A.values.reshape(-1))
         self.out_inv = torch.where(deg != 0, 1.0 / deg, torch.zeros_like(deg))
+        # local transpose: for a row partition the local A^T maps owned x
+        # into the extended (owned + halo) column space; accumulating the
+        # halo tail onto its owners (add_from_halo) completes the global
+        # A^T apply (reference pagerank_operator.h runs on the transposed
+        # distributed matrix)
         self.AT = ops.transpose(A)
 
     def apply(self, x, y):
-        mgr = getattr(self.A, "manager", None)
-        ops.spmv(self.AT, x * self.out_inv, y)
-        ops.scal(y, self.d)
+        mgr = self.mgr
         if mgr is None:
+            ops.spmv(self.AT, x * self.out_inv, y)
+            ops.scal(y, self.d)
             dangling = float(ops.dot(x, (self.out_inv == 0).to(x.dtype)))
             total = float(x.sum())
-            n_glob = self.n
-        else:
-            xo = mgr.owned(x)
-            mask = (self.out_inv == 0).to(x.dtype).reshape(-1)[
-                :mgr.owned_size]
-            dangling = mgr.global_sum(float(ops.dot(xo, mask)))
-            total = mgr.global_sum(float(xo.sum()))
-            n_glob = mgr.n_global
-        y += (self.d * dangling + (1.0 - self.d) * total) / n_glob
+            y += (self.d * dangling + (1.0 - self.d) * total) / self.n
+            return y
+        xo = x.reshape(-1)[:mgr.owned_size]
+        z = (xo * self.out_inv).contiguous()
+        yext = ops.spmv(self.AT, z)            # ext-sized partial columns
+        mgr.add_from_halo(yext, block_override=1)
+        yf = y.reshape(-1)
+        yf[:mgr.owned_size] = self.d * yext[:mgr.owned_size]
+        yf[mgr.owned_size:] = 0.0
+        mask = (self.out_inv == 0).to(x.dtype)
+        dangling = mgr.global_sum(float(ops.dot(xo, mask)))
+        total = mgr.global_sum(float(xo.sum()))
+        yf[:mgr.owned_size] += (self.d * dangling +
+                                (1.0 - self.d) * total) / mgr.n_global
         return y
 
 

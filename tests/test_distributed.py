@@ -823,3 +823,61 @@ def _impl_test_dist_aggressive_classical(rank, world, tmp):
     plain = solve(0)
     aggr = solve(1)
     assert aggr < plain, f"aggressive {aggr} !< plain {plain}"
+
+
+def test_dist_pagerank():
+    _run_dist(test_dist_pagerank)
+
+
+def _impl_test_dist_pagerank(rank, world, tmp):
+    """Distributed PageRank: local-transpose apply + add_from_halo equals the
+    dense serial Google-matrix stationary vector (dangling nodes included)."""
+    import scipy.sparse as sp
+    import torch.distributed as dist
+
+    from amgx_amd.config import ConfigScope
+    from amgx_amd.distributed.manager import DistributedManager
+    from amgx_amd.eigensolvers import create_eigensolver
+    from amgx_amd.resources import Resources
+    n, d = 60, 0.85
+    rng = np.random.RandomState(3)            # same digraph on all ranks
+    M = sp.random(n, n, density=0.08, random_state=rng, format="csr")
+    M.setdiag(0)
+    M.eliminate_zeros()
+    M.data[:] = 1.0
+    M.indptr[-6:] = M.indptr[-6]              # last 5 rows dangling
+    M = sp.csr_matrix((M.data[:M.indptr[-1]],
+                       M.indices[:M.indptr[-1]], M.indptr), shape=(n, n))
+    per = n // world
+    lo = rank * per
+    hi = n if rank == world - 1 else lo + per
+    s0, s1 = M.indptr[lo], M.indptr[hi]
+    A = DistributedManager.upload_global_csr(
+        M.indptr[lo:hi + 1] - s0, M.indices[s0:s1], M.data[s0:s1],
+        hi - lo, lo, n, device="cpu")
+    mgr = A.manager
+    es = create_eigensolver(
+        ConfigScope(None, {"eig_solver": "POWER_ITERATION",
+                           "eig_max_iters": 2000, "eig_tolerance": 1e-10}),
+        resources=Resources("cpu", distributed=True))
+    es.pagerank_setup(A, damping=d)
+    st = es.solve()
+    assert st.converged
+    assert abs(st.eigenvalues[0] - 1.0) < 1e-6   # Google matrix: lambda=1
+    x_user = mgr.permute_out(st.eigenvector)
+    tot = torch.tensor([float(x_user.sum())])
+    dist.all_reduce(tot)
+    pr = (x_user / tot.item()).numpy()
+    # dense reference
+    out = np.asarray(M.sum(axis=1)).ravel()
+    P = np.zeros((n, n))
+    nz = out > 0
+    P[:, nz] = (M.toarray()[nz] / out[nz, None]).T
+    G = d * P + d * np.where(out == 0, 1.0 / n, 0.0)[None, :] \
+        + (1 - d) / n
+    evals, evecs = np.linalg.eig(G)
+    k = int(np.argmax(np.real(evals)))
+    ref = np.real(evecs[:, k])
+    ref = ref / ref.sum()
+    assert np.allclose(pr, ref[lo:hi], atol=1e-6), \
+        float(np.abs(pr - ref[lo:hi]).max())
