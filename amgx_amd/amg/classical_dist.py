@@ -675,3 +675,90 @@ def aggressive_thin_dist(A, mgr: DistributedManager,
     krows = np.nonzero(keep)[0]
     cf2[krows] = np.arange(krows.size, dtype=np.int32)
     return torch.from_numpy(cf2).to(cf.device), int(krows.size)
+
+
+def interp_em_dist(A, mgr: DistributedManager, strong_out: torch.Tensor,
+                   cf_ext: torch.Tensor, coarse_offs):
+    """Distributed energy-minimization interpolation (reference
+    src/energymin/interpolators/em.cu run under the distributed level):
+    each owned F row minimizes the A-energy over its strong C patch with
+    the constant constraint, exactly as the serial EM — but a patch may
+    reach across the partition, so the A-rows of halo patch members come
+    from one matrix-halo exchange (global columns). Returns P with GLOBAL
+    coarse columns on A's device."""
+    import scipy.sparse as sp
+
+    from ..distributed.manager import halo_matrix
+    from ..matrix import CSRMatrix
+    n = mgr.n_local
+    ngc = int(coarse_offs[-1])
+    ro = A.row_offsets.cpu().numpy().astype(np.int64)
+    ci = A.col_indices.cpu().numpy().astype(np.int64)
+    va = A.values.cpu().numpy().astype(np.float64).reshape(-1)
+    strong = strong_out.cpu().numpy()
+    cfx = cf_ext.cpu().numpy().astype(np.int64)      # len n_cols, global C
+    ro_h, cols_h, vals_h = halo_matrix(mgr, A)
+    vals_h = np.asarray(vals_h, dtype=np.float64).reshape(-1)
+    # global fine id of every extended (owned + halo) column slot
+    gid = np.empty(A.n_cols, dtype=np.int64)
+    gid[:n] = mgr.row_perm.cpu().numpy().astype(np.int64) + mgr.row_start
+    if mgr.n_halo:
+        gid[n:] = np.asarray(mgr.halo_global, dtype=np.int64)
+
+    def row_map(p):
+        """A-row of ext slot p as {global col: value}."""
+        if p < n:
+            s, e = int(ro[p]), int(ro[p + 1])
+            return dict(zip(gid[ci[s:e]].tolist(), va[s:e].tolist()))
+        s, e = int(ro_h[p - n]), int(ro_h[p - n + 1])
+        return dict(zip(np.asarray(cols_h[s:e]).tolist(),
+                        vals_h[s:e].tolist()))
+
+    Pc, Pv, indptr = [], [], [0]
+    for i in range(n):
+        if cfx[i] >= 0:
+            Pc.append(int(cfx[i])); Pv.append(1.0)
+            indptr.append(indptr[-1] + 1)
+            continue
+        s, e = int(ro[i]), int(ro[i + 1])
+        patch = [int(ci[k]) for k in range(s, e)
+                 if strong[k] and cfx[ci[k]] >= 0]
+        if not patch:
+            indptr.append(indptr[-1])
+            continue
+        m = len(patch)
+        gp = gid[np.asarray(patch)]
+        gi = int(gid[i])
+        Acc = np.zeros((m, m))
+        aci = np.zeros(m)
+        for a, p in enumerate(patch):
+            rm = row_map(p)
+            for b, gq in enumerate(gp):
+                Acc[a, b] = rm.get(int(gq), 0.0)
+            aci[a] = rm.get(gi, 0.0)
+        try:
+            w0 = np.linalg.solve(Acc, -aci)
+            z = np.linalg.solve(Acc, np.ones(m))
+        except np.linalg.LinAlgError:
+            w0, *_ = np.linalg.lstsq(Acc, -aci, rcond=None)
+            z, *_ = np.linalg.lstsq(Acc, np.ones(m), rcond=None)
+        denom = float(np.ones(m) @ z)
+        if denom != 0.0:
+            w = w0 + z * (1.0 - float(np.ones(m) @ w0)) / denom
+        else:
+            w = w0
+        cp = cfx[np.asarray(patch)]
+        order = np.argsort(cp)
+        for o in order:
+            Pc.append(int(cp[o])); Pv.append(float(w[o]))
+        indptr.append(indptr[-1] + m)
+    P = sp.csr_matrix((np.asarray(Pv), np.asarray(Pc, dtype=np.int64),
+                       np.asarray(indptr)), shape=(n, ngc))
+    P.sum_duplicates()
+    P.sort_indices()
+    dev = A.row_offsets.device
+    return CSRMatrix(
+        torch.from_numpy(P.indptr.astype(np.int32)).to(dev),
+        torch.from_numpy(P.indices.astype(np.int32)).to(dev),
+        torch.from_numpy(P.data).to(A.dtype).to(dev),
+        n_cols=ngc)

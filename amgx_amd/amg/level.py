@@ -294,7 +294,7 @@ class ClassicalLevel(AMGLevel):
         cf_ext, coarse_offs = coarse_numbering(mgr, self.cf_map,
                                                self.num_coarse)
         interp = self.scope.get("interpolator") \
-            if self.scope.has("interpolator") else "D1"
+            if self.scope.has("interpolator") else self.default_interp
         if getattr(self, "_aggressive", False):
             interp = self.scope.get("aggressive_interpolator") or "MULTIPASS"
         if interp == "D2":
@@ -315,10 +315,14 @@ class ClassicalLevel(AMGLevel):
             from .classical_dist import interp_multipass_dist
             P_m = interp_multipass_dist(A, mgr, self._strong_out, cf_ext,
                                         coarse_offs)
+        elif interp == "EM":
+            from .classical_dist import interp_em_dist
+            P_m = interp_em_dist(A, mgr, self._strong_out, cf_ext,
+                                 coarse_offs)
         else:
             raise NotImplementedError(
-                f"distributed interpolator {interp!r}: D1/D2/MULTIPASS are"
-                " wired (EM: next round)")
+                f"distributed interpolator {interp!r}: D1/D2/MULTIPASS/EM"
+                " are wired")
         tf = float(self.scope.get("interp_truncation_factor"))
         me = int(self.scope.get("interp_max_elements"))
         if tf > 0.0 or me >= 0:
