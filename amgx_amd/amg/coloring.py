@@ -321,8 +321,5 @@ class MatrixColoring:
         """Distance-1/2 validity check (reference src/tests/valid_coloring.cu)."""
         indptr, indices = _host_adj(A, level)
         col = self.colors.cpu().numpy()
-        for i in range(indptr.size - 1):
-            for j in indices[indptr[i]:indptr[i + 1]]:
-                if col[j] == col[i]:
-                    return False
-        return True
+        rows = np.repeat(np.arange(indptr.size - 1), np.diff(indptr))
+        return not bool((col[rows] == col[indices]).any())

@@ -110,6 +110,41 @@ class GSSolver(MulticolorGSSolver):
     sweep has no MI355X-native expression, color-parallel is the idiom)."""
 
 
+@register_solver("FIXCOLOR_GS")
+class FixcolorGSSolver(MulticolorGSSolver):
+    """Fixed 8-coloring GS for structured cubic grids (reference
+    src/solvers/fixcolor_gauss_seidel_solver.cu: color = the 2x2x2 lattice
+    parity of the row's (x,y,z) grid coordinates, num_colors=8, no coloring
+    computation). Falls back to the computed multicolor schedule when the
+    matrix is not a row-major cube (the fixed lattice would race)."""
+
+    def solver_setup(self):
+        A = self.A
+        if A.coloring is None:
+            A.coloring = self._lattice_coloring(A) or \
+                __import__("amgx_amd.amg.coloring",
+                           fromlist=["MatrixColoring"]) \
+                .MatrixColoring.create(A, self.scope)
+        self.dinv = ops.jacobi_dinv(A, l1=bool(self.scope.get("GS_L1_variant")))
+
+    @staticmethod
+    def _lattice_coloring(A):
+        """color(g) = parity bits of (x, y, z) for a row-major n1d^3 grid;
+        validated distance-1 against A's actual structure before use."""
+        from ..amg.coloring import MatrixColoring
+        n = A.n_rows
+        n1d = round(n ** (1.0 / 3.0))
+        if n1d ** 3 != n:
+            return None
+        g = torch.arange(n, device=A.row_offsets.device)
+        x = g % n1d
+        y = (g // n1d) % n1d
+        z = g // (n1d * n1d)
+        colors = ((z & 1) | ((y & 1) << 1) | ((x & 1) << 2)).to(torch.int32)
+        col = MatrixColoring(colors, 8)
+        return col if col.validate(A) else None
+
+
 @register_solver("CHEBYSHEV")
 class ChebyshevSolver(_SmootherBase):
     """Chebyshev iteration preconditioned by the (L1-)diagonal; lambda_max

@@ -350,3 +350,29 @@ def test_truncated_fgmres():
     st = s.solve(b, x, zero_initial_guess=True)
     rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
     assert st.converged and rel < 1e-7
+
+
+def test_fixcolor_gs():
+    """FIXCOLOR_GS: the fixed 2x2x2 lattice coloring is valid on a cube and
+    smooths like multicolor GS; non-cubes fall back to a computed coloring."""
+    from amgx_amd.problems import poisson_3d, poisson_2d
+    from amgx_amd.config import ConfigScope
+    for prob in (poisson_3d(8, 8, 8), poisson_2d(9, 7)):
+        s = create_solver(ConfigScope(None, {"solver": "FIXCOLOR_GS",
+                                             "max_iters": 30}),
+                          resources=Resources("cpu"))
+        b = torch.ones(prob.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(prob)
+        r0 = ops.nrm2(ops.residual(prob, x, b))
+        s.solve(b, x)
+        r1 = ops.nrm2(ops.residual(prob, x, b))
+        assert r1 < 0.2 * r0, (prob.n_rows, r1 / r0)
+    # the cube case really used the lattice: exactly 8 colors
+    A = poisson_3d(8, 8, 8)
+    s = create_solver(ConfigScope(None, {"solver": "FIXCOLOR_GS",
+                                         "max_iters": 1}),
+                      resources=Resources("cpu"))
+    s.setup(A)
+    assert A.coloring.num_colors == 8
+    assert A.coloring.validate(A)
