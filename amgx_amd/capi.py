@@ -388,7 +388,24 @@ def AMGX_matrix_upload_all_global(m: _MatrixHandle, n_global, n, nnz,
     from .distributed.manager import DistributedManager
     mem, _, matprec = _parse_mode(m.mode)
     assert block_dimx == block_dimy
-    assert tdist.is_initialized(), "upload_all_global needs torch.distributed"
+    if not tdist.is_initialized():
+        # single-process global upload (reference: 1-rank communicator) —
+        # global columns ARE local columns, no halo structure needed
+        assert int(n) == int(n_global), \
+            "upload_all_global without torch.distributed needs n == n_global"
+        from .matrix import CSRMatrix
+        device = mem if mem == "cpu" else m.res.device
+        ro_t = torch.as_tensor(np.asarray(row_ptrs)).to(torch.int32)
+        ci_t = torch.as_tensor(
+            np.asarray(col_indices_global)).to(torch.int32)
+        vals = torch.as_tensor(np.asarray(data)).to(matprec)
+        if block_dimx > 1:
+            m.A = CSRMatrix.from_bsr(
+                ro_t, ci_t, vals.reshape(nnz, block_dimx, block_dimx),
+                n_cols=int(n)).to(device)
+        else:
+            m.A = CSRMatrix(ro_t, ci_t, vals, n_cols=int(n)).to(device)
+        return RC_OK
     rank = tdist.get_rank()
     if partition_vector is not None:
         pv = np.asarray(partition_vector)
