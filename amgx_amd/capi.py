@@ -742,8 +742,10 @@ def AMGX_unpin_memory(ptr):
     return RC_OK
 
 
-def AMGX_SOLVE_SUCCESS():
-    return 0
+# solve-status enum (reference AMGX_SOLVE_STATUS, include/amgx_c.h)
+AMGX_SOLVE_SUCCESS = 0
+AMGX_SOLVE_FAILED = 1
+AMGX_SOLVE_DIVERGED = 2
 
 
 # -------------------------------------------------- remaining API surface

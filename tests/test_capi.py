@@ -301,3 +301,39 @@ def test_upload_with_external_diag():
         m, n, offd.nnz, offd.data * 2.0, diag_data=diag * 2.0) == C.RC_OK
     got2 = m.A.to_scipy().tocsr()
     assert np.allclose(got2.toarray(), 2.0 * A_full.toarray())
+
+
+def test_upload_all_global_single_process():
+    """upload_all_global without torch.distributed = the 1-rank communicator
+    case (reference runs on MPI_COMM_SELF): serial matrix, full solve."""
+    import numpy as np
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create(
+        "config_version=2, solver=PCG, preconditioner=BLOCK_JACOBI,"
+        " max_iters=500, tolerance=1e-10, convergence=RELATIVE_INI,"
+        " monitor_residual=1")
+    rc, res = C.AMGX_resources_create(cfg, comm=None, device_num=0)
+    rc, A = C.AMGX_matrix_create(res, "hDDI")
+    n = 50
+    ro, cols, vals = [0], [], []
+    for i in range(n):
+        for j, v in ((i - 1, -1.0), (i, 2.0), (i + 1, -1.0)):
+            if 0 <= j < n:
+                cols.append(j)
+                vals.append(v)
+        ro.append(len(cols))
+    rc = C.AMGX_matrix_upload_all_global(
+        A, n, n, len(cols), 1, 1, np.asarray(ro),
+        np.asarray(cols, dtype=np.int64), np.asarray(vals))
+    assert rc == C.RC_OK
+    rc, b = C.AMGX_vector_create(res, "hDDI")
+    rc, x = C.AMGX_vector_create(res, "hDDI")
+    C.AMGX_vector_upload(b, n, 1, np.ones(n))
+    C.AMGX_vector_set_zero(x, n, 1)
+    rc, s = C.AMGX_solver_create(res, "hDDI", cfg)
+    assert C.AMGX_solver_setup(s, A) == C.RC_OK
+    assert C.AMGX_solver_solve(s, b, x) == C.RC_OK
+    rc, st = C.AMGX_solver_get_status(s)
+    assert st == C.AMGX_SOLVE_SUCCESS
+    C.AMGX_finalize()
