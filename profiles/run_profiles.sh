@@ -23,10 +23,12 @@ cd /tmp && timeout 900 rocprofv3 --kernel-trace --stats \
     > "$GRAFT_REPO_ROOT"/gpurun_out/prof/rocprof_stats.log 2>&1
 
 # 2b. kernel microbenchmarks (per-kernel ms + achieved GB/s)
+cd "$GRAFT_REPO_ROOT" || cd /root/repo
 timeout 600 python bench_kernels.py --size 192 --iters 20 \
     > gpurun_out/prof/kernel_bench.jsonl 2> gpurun_out/prof/kernel_bench.txt
 
 # 3. PMC counters in their OWN run (pool rule: never combined with traces)
+cd /tmp
 timeout 900 rocprofv3 --pmc FETCH_SIZE WRITE_SIZE SQ_INSTS_MFMA \
     -d "$GRAFT_REPO_ROOT"/gpurun_out/prof/pmc -- \
     python "$GRAFT_REPO_ROOT"/bench.py --gpus 1 --steps 1 --warmup 0 --size 128 \
