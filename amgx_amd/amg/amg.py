@@ -136,12 +136,23 @@ class AMGHierarchy:
             self.levels.append(nxt)
             level = nxt
 
+    # smoother tuning knobs a string-declared smoother inherits from the
+    # declaring AMG scope (reference: a sub-solver without its own scope
+    # resolves parameters in the scope it was declared in)
+    _SMOOTHER_INHERIT = (
+        "relaxation_factor", "symmetric_GS", "GS_L1_variant",
+        "jacobi_l1_variant", "matrix_coloring_scheme", "coloring_level",
+        "max_uncolored_percentage", "num_colors", "max_num_hash",
+        "ilu_sparsity_level", "chebyshev_polynomial_order",
+        "chebyshev_lambda_estimate_mode", "cheby_max_lambda",
+        "cheby_min_lambda", "kpz_order", "cf_smoothing_mode")
+
     def _make_smoother(self):
         name, sub = self.scope.sub_solver("smoother", "BLOCK_JACOBI")
-        # smoother inherits the amg scope's relaxation_factor unless it sets one
-        if not sub.has("relaxation_factor") and self.scope.has("relaxation_factor"):
-            sub = sub.child(dict(sub.node,
-                                 relaxation_factor=self.scope.get("relaxation_factor")))
+        inh = {k: self.scope.get(k) for k in self._SMOOTHER_INHERIT
+               if not sub.has(k) and self.scope.has(k)}
+        if inh:
+            sub = sub.child(dict(sub.node, **inh))
         return create_solver(name, sub, self.res)
 
     def _setup_coarse_solver(self):

@@ -29,6 +29,11 @@ class KaczmarzSolver(_SmootherBase):
         super().__init__(scope, resources)
         if not scope.has("relaxation_factor"):
             self.relaxation_factor = 1.0
+        # symmetric color schedule (ascending then descending), same knob
+        # as symmetric_GS. Note: the sweep is l2-symmetric, not A-adjoint,
+        # so a Kaczmarz-smoothed AMG preconditioner still needs a flexible
+        # Krylov method (FGMRES/PCGF), not plain PCG
+        self.symmetric = bool(scope.get("symmetric_GS"))
 
     def solver_setup(self):
         A = self.A
@@ -73,7 +78,10 @@ class KaczmarzSolver(_SmootherBase):
         va = A.values.reshape(-1)
         xv = x.reshape(-1)
         bv = b.reshape(-1)
-        for c in range(col.num_colors):
+        order = list(range(col.num_colors))
+        if self.symmetric:
+            order = order + order[::-1]
+        for c in order:
             rows = col.rows_of(c).to(torch.int64)
             if rows.numel() == 0:
                 continue
