@@ -176,3 +176,26 @@ def poisson_3d_local_device(nx: int, ny: int, nz: int, rank: int, world: int,
                           dtype=dtype, device=dev)
     vals = valrow.expand(n, 7)[mask].contiguous()
     return ro, cols, vals, row_start
+
+
+def random_laplacian(n: int, avg_degree: int = 12, seed: int = 11,
+                     device="cpu", dtype=torch.float64):
+    """Unstructured SPD graph Laplacian (+I): the offline stand-in for the
+    SuiteSparse/Florida unstructured matrices the reference benchmarks
+    against (BASELINE.json config #5) — irregular row lengths and a
+    scattered sparsity pattern stress ILU/halo paths the way a real
+    unstructured mesh does."""
+    import scipy.sparse as sp
+
+    from .matrix import CSRMatrix
+    rng = np.random.RandomState(seed)
+    density = avg_degree / float(n)
+    G = sp.random(n, n, density=density, random_state=rng, format="csr")
+    G = G + G.T
+    G.data[:] = np.abs(G.data)
+    L = sp.diags(np.asarray(G.sum(1)).ravel()) - G + sp.identity(n)
+    L = L.tocsr()
+    L.sum_duplicates()
+    L.sort_indices()
+    L.eliminate_zeros()
+    return CSRMatrix.from_scipy(L, dtype=dtype).to(device)

@@ -88,7 +88,7 @@ FGMRES_BLOCK4_DILU = {
 CONFIGS = {
     "fgmres_agg": (FGMRES_AGG, "poisson"),
     "classical_pcg": (PCG_CLASSICAL, "poisson"),
-    "ilu0": (FGMRES_ILU0, "poisson"),
+    "ilu0": (FGMRES_ILU0, "unstructured"),
     "block4_dilu": (FGMRES_BLOCK4_DILU, "block4"),
 }
 
@@ -116,6 +116,34 @@ def build_local_matrix(size, rank, world, device, problem="poisson"):
             ro[lo:hi + 1] - s0, Afull.col_indices.numpy()[s0:s1],
             Afull.values.numpy().reshape(Afull.nnz, 16)[s0:s1],
             hi - lo, lo, n, device=device, block_dim=4)
+        return A, A.manager
+    if problem == "unstructured":
+        # scrambled Poisson: random node relabeling destroys the banded
+        # locality (offline SuiteSparse stand-in, BASELINE config #5) while
+        # keeping the spectrum — irregular halo + gather patterns
+        import numpy as np
+        import scipy.sparse as sp
+
+        from amgx_amd.matrix import CSRMatrix
+        from amgx_amd.problems import poisson_3d
+        n = size ** 3 * world
+        Ah = poisson_3d(size, size, size * world).to_scipy()
+        rng = np.random.RandomState(7)
+        perm = rng.permutation(n)
+        Pm = sp.csr_matrix((np.ones(n), (perm, np.arange(n))), shape=(n, n))
+        As = (Pm @ Ah @ Pm.T).tocsr()
+        As.sum_duplicates()
+        As.sort_indices()
+        if world == 1:
+            return CSRMatrix.from_scipy(As).to(device), None
+        from amgx_amd.distributed.manager import DistributedManager
+        per = n // world
+        lo = rank * per
+        hi = n if rank == world - 1 else lo + per
+        s0, s1 = As.indptr[lo], As.indptr[hi]
+        A = DistributedManager.upload_global_csr(
+            As.indptr[lo:hi + 1] - s0, As.indices[s0:s1], As.data[s0:s1],
+            hi - lo, lo, n, device=device)
         return A, A.manager
     if world == 1:
         from amgx_amd.problems import poisson_3d
@@ -235,8 +263,9 @@ def main():
                                   " 1e-6 RELATIVE_INI",
                     "classical_pcg": "3D 7-pt Poisson, PCG + classical"
                                      " Ruge-Stueben AMG V-cycle, tol 1e-6",
-                    "ilu0": "3D 7-pt Poisson, FGMRES + multicolor ILU(0),"
-                            " tol 1e-6",
+                    "ilu0": "scrambled-Poisson unstructured Laplacian"
+                            " (SuiteSparse stand-in), FGMRES +"
+                            " multicolor ILU(0), tol 1e-6",
                     "block4_dilu": "block-4 coupled system, FGMRES +"
                                    " multicolor DILU, tol 1e-6",
                 }[args.config],
