@@ -419,3 +419,34 @@ def test_blocked_norms():
     s.setup(A)
     st = s.solve(b, x, zero_initial_guess=True)
     assert st.converged
+
+
+def test_unstructured_random_laplacian():
+    """FGMRES+ILU(0) and PCG+AMG on an unstructured random graph Laplacian
+    (the offline SuiteSparse stand-in, irregular degrees)."""
+    from amgx_amd import AMGConfig
+    from amgx_amd.problems import random_laplacian
+    A = random_laplacian(400, avg_degree=10, seed=3)
+    for cfg in (
+        {"solver": {"preconditioner": {"solver": "MULTICOLOR_ILU",
+                                       "max_iters": 1, "scope": "i"},
+                    "solver": "FGMRES", "max_iters": 200,
+                    "gmres_n_restart": 30, "monitor_residual": 1,
+                    "convergence": "RELATIVE_INI", "tolerance": 1e-8}},
+        {"solver": {"preconditioner": {"solver": "AMG",
+                                       "algorithm": "AGGREGATION",
+                                       "smoother": "MULTICOLOR_GS",
+                                       "symmetric_GS": 1, "max_iters": 1,
+                                       "min_coarse_rows": 10, "cycle": "V"},
+                    "solver": "PCG", "max_iters": 200,
+                    "monitor_residual": 1,
+                    "convergence": "RELATIVE_INI", "tolerance": 1e-8}},
+    ):
+        s = create_solver(AMGConfig.from_dict(cfg).root_scope(),
+                          resources=Resources("cpu"))
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
+        assert st.converged and rel < 1e-7, st
