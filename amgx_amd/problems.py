@@ -199,3 +199,46 @@ def random_laplacian(n: int, avg_degree: int = 12, seed: int = 11,
     L.sort_indices()
     L.eliminate_zeros()
     return CSRMatrix.from_scipy(L, dtype=dtype).to(device)
+
+
+def anisotropic_2d(nx: int, ny: int, eps: float = 0.01, device="cpu",
+                   dtype=torch.float64):
+    """2D anisotropic diffusion -eps*u_xx - u_yy (5-pt): the classic
+    strength-of-connection stress (reference CUSP generator role)."""
+    import scipy.sparse as sp
+    ex = np.ones(nx)
+    ey = np.ones(ny)
+    Tx = sp.diags([-ex[:-1] * eps, 2 * eps * ex, -ex[:-1] * eps],
+                  [-1, 0, 1], shape=(nx, nx))
+    Ty = sp.diags([-ey[:-1], 2 * ey, -ey[:-1]], [-1, 0, 1], shape=(ny, ny))
+    L = (sp.kron(sp.identity(ny), Tx) + sp.kron(Ty, sp.identity(nx))).tocsr()
+    L.sum_duplicates()
+    L.sort_indices()
+    L.eliminate_zeros()
+    from .matrix import CSRMatrix
+    return CSRMatrix.from_scipy(L, dtype=dtype).to(device)
+
+
+def convection_diffusion_2d(nx: int, ny: int, beta: float = 20.0,
+                            device="cpu", dtype=torch.float64):
+    """2D convection-diffusion with upwind convection (NONSYMMETRIC): the
+    BiCGStab/GMRES/IDR stress problem."""
+    import scipy.sparse as sp
+    h = 1.0 / (nx + 1)
+    n = nx * ny
+    rows, cols, vals = [], [], []
+    for j in range(ny):
+        for i in range(nx):
+            r = j * nx + i
+            diag = 4.0 + beta * h
+            rows.append(r); cols.append(r); vals.append(diag)
+            for di, dj, v in ((-1, 0, -1.0 - beta * h), (1, 0, -1.0),
+                              (0, -1, -1.0), (0, 1, -1.0)):
+                ii, jj = i + di, j + dj
+                if 0 <= ii < nx and 0 <= jj < ny:
+                    rows.append(r); cols.append(jj * nx + ii); vals.append(v)
+    L = sp.csr_matrix((vals, (rows, cols)), shape=(n, n))
+    L.sum_duplicates()
+    L.sort_indices()
+    from .matrix import CSRMatrix
+    return CSRMatrix.from_scipy(L, dtype=dtype).to(device)

@@ -450,3 +450,41 @@ def test_unstructured_random_laplacian():
         st = s.solve(b, x, zero_initial_guess=True)
         rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
         assert st.converged and rel < 1e-7, st
+
+
+def test_anisotropic_and_nonsymmetric():
+    """Anisotropic diffusion under classical AMG+PCG and nonsymmetric
+    convection-diffusion under BiCGStab/GMRES/IDR+ILU(0) (reference CUSP
+    generator roles; nonsymmetric Krylov coverage)."""
+    from amgx_amd import AMGConfig
+    from amgx_amd.problems import anisotropic_2d, convection_diffusion_2d
+    A = anisotropic_2d(30, 30, eps=0.01)
+    cfg = {"solver": {"preconditioner": {
+        "solver": "AMG", "algorithm": "CLASSICAL",
+        "smoother": "MULTICOLOR_GS", "symmetric_GS": 1, "max_iters": 1,
+        "min_coarse_rows": 10, "cycle": "V"},
+        "solver": "PCG", "max_iters": 200, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8}}
+    s = create_solver(AMGConfig.from_dict(cfg).root_scope(),
+                      resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged and st.iterations < 40, st
+
+    An = convection_diffusion_2d(30, 30, beta=20.0)
+    for outer in ("BICGSTAB", "GMRES", "IDR"):
+        cfg = {"solver": {"preconditioner": {"solver": "MULTICOLOR_ILU",
+                                             "max_iters": 1, "scope": "i"},
+               "solver": outer, "max_iters": 300, "gmres_n_restart": 40,
+               "monitor_residual": 1, "convergence": "RELATIVE_INI",
+               "tolerance": 1e-8}}
+        s = create_solver(AMGConfig.from_dict(cfg).root_scope(),
+                          resources=Resources("cpu"))
+        b = torch.ones(An.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(An)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = ops.nrm2(ops.residual(An, x, b)) / ops.nrm2(b)
+        assert st.converged and rel < 1e-6, (outer, st)
