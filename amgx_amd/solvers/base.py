@@ -287,6 +287,24 @@ class Solver:
         if self.res.is_cuda:
             self.res.synchronize()
         st.solve_time = time.perf_counter() - t0
+        if (self.print_solve_stats and self.res.rank == 0 and monitoring
+                and st.residuals):
+            # reference Solver::print_final (src/solvers/solver.cu:895-934)
+            ini, last = st.residuals[0], st.residuals[-1]
+            its = max(st.iterations, 1)
+            rate = (last / ini) ** (1.0 / its) if ini > 0 else ini
+            print(f"           ----------------------------")
+            print(f"         Total Iterations: {st.iterations}")
+            print(f"         Avg Convergence Rate: {rate:15.4f}")
+            print(f"         Final Residual: {last:15.6e}")
+            print(f"         Total Reduction in Residual: "
+                  f"{(last / ini if ini > 0 else ini):15.6e}")
+            if self.scope.get("obtain_timings"):
+                print(f"         Total Time: {st.setup_time + st.solve_time:10.4f} s")
+                print(f"             setup: {st.setup_time:10.4f} s")
+                print(f"             solve: {st.solve_time:10.4f} s")
+                print(f"             solve(per iteration): "
+                      f"{st.solve_time / its:10.6f} s")
         return st
 
     # default residual-norm recomputation; Krylov solvers override with their
