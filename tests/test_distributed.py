@@ -925,3 +925,42 @@ def _impl_test_dist_classical_em(rank, world, tmp):
     it_em = solve("EM")
     it_d1 = solve("D1")
     assert it_em <= it_d1 + 5, f"EM {it_em} vs D1 {it_d1}"
+
+
+def test_dist_resetup():
+    _run_dist(test_dist_resetup)
+
+
+def _impl_test_dist_resetup(rank, world, tmp):
+    """Distributed resetup: replace coefficients (values*2), solver resetup,
+    solution halves (structure identical)."""
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.matrix import CSRMatrix
+    from amgx_amd.distributed.manager import DistributedManager
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {
+            "solver": "AMG", "algorithm": "AGGREGATION",
+            "smoother": "BLOCK_JACOBI", "max_iters": 1,
+            "min_coarse_rows": 12, "cycle": "V"},
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-10,
+    }})
+    from amgx_amd.resources import Resources
+    n = 6
+    A = _make_dist_A(rank, world, n)
+    mgr = A.manager
+    s = create_solver(cfg.root_scope(), resources=Resources(
+        "cpu", distributed=True))
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x1 = torch.zeros_like(b)
+    s.setup(A)
+    st1 = s.solve(b, x1, zero_initial_guess=True)
+    A.values.mul_(2.0)
+    s.resetup(A)
+    x2 = torch.zeros_like(b)
+    st2 = s.solve(b, x2, zero_initial_guess=True)
+    assert st1.converged and st2.converged
+    assert torch.allclose(x2[:mgr.owned_size], x1[:mgr.owned_size] / 2.0,
+                          atol=1e-7), \
+        float((x2 - x1 / 2)[:mgr.owned_size].abs().max())
