@@ -488,3 +488,53 @@ def test_anisotropic_and_nonsymmetric():
         st = s.solve(b, x, zero_initial_guess=True)
         rel = ops.nrm2(ops.residual(An, x, b)) / ops.nrm2(b)
         assert st.converged and rel < 1e-6, (outer, st)
+
+
+def test_degenerate_matrices():
+    """Reference zero_{in,off}_diagonal / zero_values handling tests:
+    diagonal-only operators, zero diagonal entries and explicit zeros must
+    not produce NaN/inf anywhere in setup or solve."""
+    import numpy as np
+    import scipy.sparse as sp
+    from amgx_amd import AMGConfig
+    from amgx_amd.matrix import CSRMatrix
+
+    # 1. diagonal-only matrix: no strong connections, AMG degenerates cleanly
+    D = CSRMatrix.from_scipy(sp.diags(np.arange(1.0, 41.0)).tocsr())
+    cfg = {"solver": {"preconditioner": {
+        "solver": "AMG", "algorithm": "AGGREGATION",
+        "smoother": "BLOCK_JACOBI", "max_iters": 1, "min_coarse_rows": 4,
+        "cycle": "V"},
+        "solver": "PCG", "max_iters": 60, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-10}}
+    s = create_solver(AMGConfig.from_dict(cfg).root_scope(),
+                      resources=Resources("cpu"))
+    b = torch.ones(40, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(D)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged and torch.isfinite(x).all()
+    assert torch.allclose(x, 1.0 / torch.arange(1.0, 41.0,
+                                                dtype=torch.float64),
+                          atol=1e-8)
+
+    # 2. one zero diagonal entry: Jacobi dinv must stay finite
+    m = sp.diags([np.full(19, -1.0), np.full(20, 4.0), np.full(19, -1.0)],
+                 [-1, 0, 1]).tocsr().astype(float)
+    m = m.tolil()
+    m[7, 7] = 0.0
+    A = CSRMatrix.from_scipy(m.tocsr())
+    dinv = ops.jacobi_dinv(A)
+    assert torch.isfinite(dinv).all()
+    dinv = ops.jacobi_dinv(A, l1=True)
+    assert torch.isfinite(dinv).all()
+
+    # 3. explicit zeros in a user-provided pattern are PRESERVED (the
+    # structure is the replace_coefficients contract) and ops handle them
+    m2 = sp.csr_matrix((np.array([4.0, 0.0, -1.0, 4.0]),
+                        np.array([0, 1, 0, 1]), np.array([0, 2, 4])),
+                       shape=(2, 2))
+    A2 = CSRMatrix.from_scipy(m2)
+    assert A2.nnz == 4
+    y = ops.spmv(A2, torch.tensor([1.0, 2.0], dtype=torch.float64))
+    assert torch.allclose(y, torch.tensor([4.0, 7.0], dtype=torch.float64))
