@@ -202,6 +202,14 @@ class AMGHierarchy:
         """Reference FixedCycle::cycle (src/cycles/fixed_cycle.cu:59-230)."""
         level = self.levels[li]
         prof = self.profiler
+        if level.r is not None and level.r.dtype != b.dtype:
+            # mixed precision (dDFI): matrix stores fp32, but ALL cycle
+            # vectors carry the vector precision — re-key the scratch
+            # vectors lazily off the rhs dtype
+            level.r = level.r.to(b.dtype)
+            if level.bc is not None:
+                level.bc = level.bc.to(b.dtype)
+                level.xc = level.xc.to(b.dtype)
         if li == len(self.levels) - 1:
             prof.tic("coarseSolve")
             if self.coarse_solver is not None:

@@ -23,6 +23,31 @@ hipStream_t cur_stream() {
 #define DISPATCH_FT(TENSOR, NAME, ...)                                  \
     AT_DISPATCH_FLOATING_TYPES(TENSOR.scalar_type(), NAME, __VA_ARGS__)
 
+// two-type dispatch for matrix-value x vector mixed ops (reference value
+// modes dDDI / dFFI / dDFI): VA drives scalar_a, X drives scalar_v
+#define DISPATCH_FT2(VA, X, NAME, ...)                                       \
+    do {                                                                     \
+        auto _ta = (VA).scalar_type();                                       \
+        auto _tv = (X).scalar_type();                                        \
+        if (_ta == torch::kFloat64 && _tv == torch::kFloat64) {              \
+            using scalar_a = double;                                         \
+            using scalar_v = double;                                         \
+            __VA_ARGS__();                                                   \
+        } else if (_ta == torch::kFloat32 && _tv == torch::kFloat32) {       \
+            using scalar_a = float;                                          \
+            using scalar_v = float;                                          \
+            __VA_ARGS__();                                                   \
+        } else if (_ta == torch::kFloat32 && _tv == torch::kFloat64) {       \
+            using scalar_a = float;                                          \
+            using scalar_v = double;                                         \
+            __VA_ARGS__();                                                   \
+        } else {                                                             \
+            TORCH_CHECK(false, NAME,                                         \
+                        ": unsupported (matrix, vector) dtype pair ", _ta,   \
+                        " x ", _tv);                                         \
+        }                                                                    \
+    } while (0)
+
 inline void check_dev(const Tensor& t) {
     TORCH_CHECK(t.is_cuda(), "amgx_amd._core requires device tensors");
 }
@@ -33,25 +58,22 @@ void csrmv(Tensor ro, Tensor ci, Tensor va, int64_t block_dim, Tensor x,
            double gamma, int64_t r0, int64_t r1) {
     check_dev(va);
     double avg = ci.numel() / std::max<double>(1.0, ro.numel() - 1);
-    DISPATCH_FT(va, "csrmv", [&] {
-        const scalar_t* bp =
-            bvec.has_value() ? bvec->data_ptr<scalar_t>() : nullptr;
+    DISPATCH_FT2(va, x, "csrmv", [&] {
+        const scalar_v* bp =
+            bvec.has_value() ? bvec->data_ptr<scalar_v>() : nullptr;
         if (block_dim == 1) {
-            amgx_hip::csrmv<scalar_t>(ro.data_ptr<int>(), ci.data_ptr<int>(),
-                                      va.data_ptr<scalar_t>(),
-                                      x.data_ptr<scalar_t>(),
-                                      y.data_ptr<scalar_t>(), bp,
-                                      (scalar_t)alpha, (scalar_t)beta,
-                                      (scalar_t)gamma, (int)r0, (int)r1, avg,
-                                      cur_stream());
+            amgx_hip::csrmv<scalar_a, scalar_v>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_a>(), x.data_ptr<scalar_v>(),
+                y.data_ptr<scalar_v>(), bp, (scalar_v)alpha, (scalar_v)beta,
+                (scalar_v)gamma, (int)r0, (int)r1, avg, cur_stream());
         } else {
-            amgx_hip::bsrmv<scalar_t>(ro.data_ptr<int>(), ci.data_ptr<int>(),
-                                      va.data_ptr<scalar_t>(), (int)block_dim,
-                                      x.data_ptr<scalar_t>(),
-                                      y.data_ptr<scalar_t>(), bp,
-                                      (scalar_t)alpha, (scalar_t)beta,
-                                      (scalar_t)gamma, (int)r0, (int)r1,
-                                      cur_stream());
+            amgx_hip::bsrmv<scalar_a, scalar_v>(
+                ro.data_ptr<int>(), ci.data_ptr<int>(),
+                va.data_ptr<scalar_a>(), (int)block_dim,
+                x.data_ptr<scalar_v>(), y.data_ptr<scalar_v>(), bp,
+                (scalar_v)alpha, (scalar_v)beta, (scalar_v)gamma, (int)r0,
+                (int)r1, cur_stream());
         }
     });
 }
@@ -143,11 +165,11 @@ void jacobi_smooth(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
                    Tensor bvec, Tensor xi, Tensor xo, double omega) {
     int n = (int)(ro.numel() - 1);
     double avg = ci.numel() / std::max<double>(1.0, n);
-    DISPATCH_FT(va, "jacobi_smooth", [&] {
-        amgx_hip::jacobi_smooth<scalar_t>(
-            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
-            dinv.data_ptr<scalar_t>(), bvec.data_ptr<scalar_t>(),
-            xi.data_ptr<scalar_t>(), xo.data_ptr<scalar_t>(), (scalar_t)omega,
+    DISPATCH_FT2(va, xi, "jacobi_smooth", [&] {
+        amgx_hip::jacobi_smooth<scalar_a, scalar_v>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_a>(),
+            dinv.data_ptr<scalar_a>(), bvec.data_ptr<scalar_v>(),
+            xi.data_ptr<scalar_v>(), xo.data_ptr<scalar_v>(), (scalar_v)omega,
             n, (int)b, avg, cur_stream());
     });
 }
@@ -155,12 +177,12 @@ void jacobi_smooth(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
 void gs_smooth_rows(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
                     Tensor bvec, Tensor x, Tensor rows, double omega) {
     int n = (int)(ro.numel() - 1);
-    DISPATCH_FT(va, "gs_smooth_rows", [&] {
-        amgx_hip::gs_smooth_rows<scalar_t>(
-            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
-            dinv.data_ptr<scalar_t>(), bvec.data_ptr<scalar_t>(),
-            x.data_ptr<scalar_t>(), rows.data_ptr<int>(), (int)rows.numel(),
-            (scalar_t)omega, n, (int)b, cur_stream());
+    DISPATCH_FT2(va, x, "gs_smooth_rows", [&] {
+        amgx_hip::gs_smooth_rows<scalar_a, scalar_v>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_a>(),
+            dinv.data_ptr<scalar_a>(), bvec.data_ptr<scalar_v>(),
+            x.data_ptr<scalar_v>(), rows.data_ptr<int>(), (int)rows.numel(),
+            (scalar_v)omega, n, (int)b, cur_stream());
     });
 }
 
@@ -170,16 +192,16 @@ void gs_sweep(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
               std::vector<int64_t> bounds, double omega, bool symmetric) {
     int n = (int)(ro.numel() - 1);
     int nc = (int)bounds.size() - 1;
-    DISPATCH_FT(va, "gs_sweep", [&] {
+    DISPATCH_FT2(va, x, "gs_sweep", [&] {
         auto run = [&](int c) {
             int64_t s = bounds[c], e = bounds[c + 1];
             if (e <= s) return;
-            amgx_hip::gs_smooth_rows<scalar_t>(
+            amgx_hip::gs_smooth_rows<scalar_a, scalar_v>(
                 ro.data_ptr<int>(), ci.data_ptr<int>(),
-                va.data_ptr<scalar_t>(), dinv.data_ptr<scalar_t>(),
-                bvec.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                va.data_ptr<scalar_a>(), dinv.data_ptr<scalar_a>(),
+                bvec.data_ptr<scalar_v>(), x.data_ptr<scalar_v>(),
                 rows_sorted.data_ptr<int>() + s, (int)(e - s),
-                (scalar_t)omega, n, (int)b, cur_stream());
+                (scalar_v)omega, n, (int)b, cur_stream());
         };
         for (int c = 0; c < nc; ++c) run(c);
         if (symmetric)
@@ -218,30 +240,30 @@ void dilu_apply(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor einv,
     int nc = (int)bounds.size() - 1;
     w.zero_();
     z.zero_();
-    DISPATCH_FT(va, "dilu_apply", [&] {
+    DISPATCH_FT2(va, x, "dilu_apply", [&] {
         hipStream_t st = cur_stream();
         for (int c = 0; c < nc; ++c) {
             int64_t s = bounds[c], e = bounds[c + 1];
             if (e <= s) continue;
-            amgx_hip::dilu_fwd_color<scalar_t>(
+            amgx_hip::dilu_fwd_color<scalar_a, scalar_v>(
                 ro.data_ptr<int>(), ci.data_ptr<int>(),
-                va.data_ptr<scalar_t>(), einv.data_ptr<scalar_t>(),
+                va.data_ptr<scalar_a>(), einv.data_ptr<scalar_a>(),
                 colors.data_ptr<int>(), rows_sorted.data_ptr<int>() + s,
-                (int)(e - s), c, r.data_ptr<scalar_t>(),
-                w.data_ptr<scalar_t>(), (int)b, st);
+                (int)(e - s), c, r.data_ptr<scalar_v>(),
+                w.data_ptr<scalar_v>(), (int)b, st);
         }
         for (int c = nc - 1; c >= 0; --c) {
             int64_t s = bounds[c], e = bounds[c + 1];
             if (e <= s) continue;
-            amgx_hip::dilu_bwd_color<scalar_t>(
+            amgx_hip::dilu_bwd_color<scalar_a, scalar_v>(
                 ro.data_ptr<int>(), ci.data_ptr<int>(),
-                va.data_ptr<scalar_t>(), einv.data_ptr<scalar_t>(),
+                va.data_ptr<scalar_a>(), einv.data_ptr<scalar_a>(),
                 colors.data_ptr<int>(), rows_sorted.data_ptr<int>() + s,
-                (int)(e - s), c, w.data_ptr<scalar_t>(),
-                z.data_ptr<scalar_t>(), (int)b, st);
+                (int)(e - s), c, w.data_ptr<scalar_v>(),
+                z.data_ptr<scalar_v>(), (int)b, st);
         }
-        amgx_hip::axpy<scalar_t>(x.data_ptr<scalar_t>(),
-                                 z.data_ptr<scalar_t>(), (scalar_t)relax,
+        amgx_hip::axpy<scalar_v>(x.data_ptr<scalar_v>(),
+                                 z.data_ptr<scalar_v>(), (scalar_v)relax,
                                  x.numel(), st);
     });
 }
@@ -335,11 +357,10 @@ void prolongate_agg(Tensor x, Tensor xc, Tensor agg, int64_t b) {
 
 // ---------------------------------------------------------------- dense
 void dense_gemv(Tensor Ainv, Tensor b, Tensor x) {
-    DISPATCH_FT(b, "dense_gemv", [&] {
-        amgx_hip::dense_gemv<scalar_t>(Ainv.data_ptr<scalar_t>(),
-                                       b.data_ptr<scalar_t>(),
-                                       x.data_ptr<scalar_t>(),
-                                       (int)b.numel(), cur_stream());
+    DISPATCH_FT2(Ainv, b, "dense_gemv", [&] {
+        amgx_hip::dense_gemv<scalar_a, scalar_v>(
+            Ainv.data_ptr<scalar_a>(), b.data_ptr<scalar_v>(),
+            x.data_ptr<scalar_v>(), (int)b.numel(), cur_stream());
     });
 }
 
@@ -553,28 +574,28 @@ void ilu0_apply(Tensor ro, Tensor ci, Tensor lu, Tensor didx, Tensor pos,
     int nc = (int)bounds.size() - 1;
     y.zero_();
     z.zero_();
-    DISPATCH_FT(lu, "ilu0_apply", [&] {
+    DISPATCH_FT2(lu, x, "ilu0_apply", [&] {
         hipStream_t st = cur_stream();
         for (int c = 0; c < nc; ++c) {
             int64_t s = bounds[c], e = bounds[c + 1];
             if (e <= s) continue;
-            amgx_hip::ilu0_fwd_launch<scalar_t>(
+            amgx_hip::ilu0_fwd_launch<scalar_a, scalar_v>(
                 ro.data_ptr<int>(), ci.data_ptr<int>(), pos.data_ptr<int>(),
-                lu.data_ptr<scalar_t>(), rows_sorted.data_ptr<int>() + s,
-                (int)(e - s), r.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                lu.data_ptr<scalar_a>(), rows_sorted.data_ptr<int>() + s,
+                (int)(e - s), r.data_ptr<scalar_v>(), y.data_ptr<scalar_v>(),
                 n, st);
         }
         for (int c = nc - 1; c >= 0; --c) {
             int64_t s = bounds[c], e = bounds[c + 1];
             if (e <= s) continue;
-            amgx_hip::ilu0_bwd_launch<scalar_t>(
+            amgx_hip::ilu0_bwd_launch<scalar_a, scalar_v>(
                 ro.data_ptr<int>(), ci.data_ptr<int>(), pos.data_ptr<int>(),
-                lu.data_ptr<scalar_t>(), didx.data_ptr<int>(),
+                lu.data_ptr<scalar_a>(), didx.data_ptr<int>(),
                 rows_sorted.data_ptr<int>() + s, (int)(e - s),
-                y.data_ptr<scalar_t>(), z.data_ptr<scalar_t>(), n, st);
+                y.data_ptr<scalar_v>(), z.data_ptr<scalar_v>(), n, st);
         }
-        amgx_hip::axpy<scalar_t>(x.data_ptr<scalar_t>(),
-                                 z.data_ptr<scalar_t>(), (scalar_t)relax,
+        amgx_hip::axpy<scalar_v>(x.data_ptr<scalar_v>(),
+                                 z.data_ptr<scalar_v>(), (scalar_v)relax,
                                  x.numel(), st);
     });
 }

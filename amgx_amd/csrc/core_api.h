@@ -10,16 +10,18 @@
 namespace amgx_hip {
 
 // ---- SpMV family (csrmv.hip) ------------------------------------------------
+// Mixed precision: TA = matrix-value type, TV = vector type (reference dDFI);
+// instantiated for (d,d), (f,f), (f,d).
 // y[i] = alpha * (A x)[i] + beta * y[i] + gamma * b[i]  over rows [r0, r1)
-template <typename T>
-void csrmv(const int* ro, const int* ci, const T* va, const T* x, T* y,
-           const T* bvec, T alpha, T beta, T gamma, int r0, int r1,
+template <typename TA, typename TV>
+void csrmv(const int* ro, const int* ci, const TA* va, const TV* x, TV* y,
+           const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            double avg_deg, hipStream_t s);
 
 // block-CSR variant, block_dim b in [2,8]; values (nnz, b, b) row-major
-template <typename T>
-void bsrmv(const int* ro, const int* ci, const T* va, int b, const T* x, T* y,
-           const T* bvec, T alpha, T beta, T gamma, int r0, int r1,
+template <typename TA, typename TV>
+void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
+           TV* y, const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            hipStream_t s);
 
 // ---- BLAS-1 (blas.hip) ------------------------------------------------------
@@ -50,15 +52,15 @@ template <typename T>
 void jacobi_dinv(const int* ro, const int* ci, const T* va, const int* didx,
                  int n, int b, bool l1, T* dinv, hipStream_t s);
 // xo = xi + omega * dinv * (b - A xi)   (block-aware; b=1 scalar fast path)
-template <typename T>
-void jacobi_smooth(const int* ro, const int* ci, const T* va, const T* dinv,
-                   const T* bvec, const T* xi, T* xo, T omega, int n, int b,
-                   double avg_deg, hipStream_t s);
+template <typename TA, typename TV>
+void jacobi_smooth(const int* ro, const int* ci, const TA* va, const TA* dinv,
+                   const TV* bvec, const TV* xi, TV* xo, TV omega, int n,
+                   int b, double avg_deg, hipStream_t s);
 // in-place GS update of rows[count]: x[r] += omega*dinv[r]*(b - A x)[r]
-template <typename T>
-void gs_smooth_rows(const int* ro, const int* ci, const T* va, const T* dinv,
-                    const T* bvec, T* x, const int* rows, int count, T omega,
-                    int n, int b, hipStream_t s);
+template <typename TA, typename TV>
+void gs_smooth_rows(const int* ro, const int* ci, const TA* va,
+                    const TA* dinv, const TV* bvec, TV* x, const int* rows,
+                    int count, TV omega, int n, int b, hipStream_t s);
 
 // ---- DILU (dilu.hip) --------------------------------------------------------
 template <typename T>
@@ -66,14 +68,16 @@ void dilu_setup_color(const int* ro, const int* ci, const T* va,
                       const int* didx, const int* tidx, const int* colors,
                       const int* rows, int count, int color, T* einv, int b,
                       hipStream_t s);
-template <typename T>
-void dilu_fwd_color(const int* ro, const int* ci, const T* va, const T* einv,
-                    const int* colors, const int* rows, int count, int color,
-                    const T* r, T* w, int b, hipStream_t s);
-template <typename T>
-void dilu_bwd_color(const int* ro, const int* ci, const T* va, const T* einv,
-                    const int* colors, const int* rows, int count, int color,
-                    const T* w, T* z, int b, hipStream_t s);
+template <typename TA, typename TV>
+void dilu_fwd_color(const int* ro, const int* ci, const TA* va,
+                    const TA* einv, const int* colors, const int* rows,
+                    int count, int color, const TV* r, TV* w, int b,
+                    hipStream_t s);
+template <typename TA, typename TV>
+void dilu_bwd_color(const int* ro, const int* ci, const TA* va,
+                    const TA* einv, const int* colors, const int* rows,
+                    int count, int color, const TV* w, TV* z, int b,
+                    hipStream_t s);
 
 // ---- coloring (setup.hip) ---------------------------------------------------
 // one min-max hash round; returns (via counter) number newly colored.
@@ -104,8 +108,8 @@ void prolongate_agg(T* x, const T* xc, const int* agg, int n, int b,
                     hipStream_t s);
 
 // ---- dense coarse solve (misc.hip) -------------------------------------------
-template <typename T>
-void dense_gemv(const T* Ainv, const T* b, T* x, int n, hipStream_t s);
+template <typename TA, typename TV>
+void dense_gemv(const TA* Ainv, const TV* b, TV* x, int n, hipStream_t s);
 
 // ---- gather/scatter for halo pack (misc.hip) ---------------------------------
 template <typename T>
@@ -177,14 +181,14 @@ template <typename T>
 void ilu0_factor_color_launch(const int* ro, const int* ci, const int* pos,
                               const int* didx, const int* rows, int count,
                               T* lu, int n, hipStream_t s);
-template <typename T>
-void ilu0_fwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
-                     const int* rows, int count, const T* r, T* y, int n,
-                     hipStream_t s);
-template <typename T>
-void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
-                     const int* didx, const int* rows, int count, const T* y,
-                     T* z, int n, hipStream_t s);
+template <typename TA, typename TV>
+void ilu0_fwd_launch(const int* ro, const int* ci, const int* pos,
+                     const TA* lu, const int* rows, int count, const TV* r,
+                     TV* y, int n, hipStream_t s);
+template <typename TA, typename TV>
+void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos,
+                     const TA* lu, const int* didx, const int* rows, int count,
+                     const TV* y, TV* z, int n, hipStream_t s);
 
 }  // namespace amgx_hip
 

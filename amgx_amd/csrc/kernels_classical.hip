@@ -319,42 +319,44 @@ __global__ __launch_bounds__(AMGX_BLOCK) void ilu0_factor_color(const int* __res
 }
 
 // forward: y_i = r_i - sum_{pos[j]<pos[i]} l_ij y_j (unit L); per color.
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void ilu0_fwd(const int* __restrict__ ro, const int* __restrict__ ci,
-                         const int* __restrict__ pos, const T* __restrict__ lu,
+                         const int* __restrict__ pos,
+                         const TA* __restrict__ lu,
                          const int* __restrict__ rows, int count,
-                         const T* __restrict__ r, T* __restrict__ y, int n) {
+                         const TV* __restrict__ r, TV* __restrict__ y, int n) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
     int pi = pos[i];
-    T sum = r[i];
+    TV sum = r[i];
     for (int k = ro[i]; k < ro[i + 1]; ++k) {
         int j = ci[k];
-        if (j < n && pos[j] < pi) sum -= lu[k] * y[j];
+        if (j < n && pos[j] < pi) sum -= (TV)lu[k] * y[j];
     }
     y[i] = sum;
 }
 
 // backward: z_i = (y_i - sum_{pos[j]>pos[i]} u_ij z_j) / u_ii; per color desc.
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void ilu0_bwd(const int* __restrict__ ro, const int* __restrict__ ci,
-                         const int* __restrict__ pos, const T* __restrict__ lu,
+                         const int* __restrict__ pos,
+                         const TA* __restrict__ lu,
                          const int* __restrict__ didx,
                          const int* __restrict__ rows, int count,
-                         const T* __restrict__ y, T* __restrict__ z, int n) {
+                         const TV* __restrict__ y, TV* __restrict__ z, int n) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
     int pi = pos[i];
-    T sum = y[i];
+    TV sum = y[i];
     for (int k = ro[i]; k < ro[i + 1]; ++k) {
         int j = ci[k];
-        if (j < n && pos[j] > pi) sum -= lu[k] * z[j];
+        if (j < n && pos[j] > pi) sum -= (TV)lu[k] * z[j];
     }
     int dk = didx[i];
-    T d = dk >= 0 ? lu[dk] : T(1);
-    if (d == T(0)) d = T(1);
+    TV d = dk >= 0 ? (TV)lu[dk] : TV(1);
+    if (d == TV(0)) d = TV(1);
     z[i] = sum / d;
 }
 
@@ -367,21 +369,23 @@ void ilu0_factor_color_launch(const int* ro, const int* ci, const int* pos,
                        dim3(AMGX_BLOCK), 0, s, ro, ci, pos, didx, rows, count,
                        lu, n);
 }
-template <typename T>
-void ilu0_fwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
-                     const int* rows, int count, const T* r, T* y, int n,
-                     hipStream_t s) {
+template <typename TA, typename TV>
+void ilu0_fwd_launch(const int* ro, const int* ci, const int* pos,
+                     const TA* lu, const int* rows, int count, const TV* r,
+                     TV* y, int n, hipStream_t s) {
     if (count <= 0) return;
-    hipLaunchKernelGGL((ilu0_fwd<T>), dim3(grid_1d(count)), dim3(AMGX_BLOCK),
-                       0, s, ro, ci, pos, lu, rows, count, r, y, n);
+    hipLaunchKernelGGL((ilu0_fwd<TA, TV>), dim3(grid_1d(count)),
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, pos, lu, rows, count,
+                       r, y, n);
 }
-template <typename T>
-void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
-                     const int* didx, const int* rows, int count, const T* y,
-                     T* z, int n, hipStream_t s) {
+template <typename TA, typename TV>
+void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos,
+                     const TA* lu, const int* didx, const int* rows, int count,
+                     const TV* y, TV* z, int n, hipStream_t s) {
     if (count <= 0) return;
-    hipLaunchKernelGGL((ilu0_bwd<T>), dim3(grid_1d(count)), dim3(AMGX_BLOCK),
-                       0, s, ro, ci, pos, lu, didx, rows, count, y, z, n);
+    hipLaunchKernelGGL((ilu0_bwd<TA, TV>), dim3(grid_1d(count)),
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, pos, lu, didx, rows,
+                       count, y, z, n);
 }
 
 #define INSTANTIATE_CLASSICAL(T)                                               \
@@ -395,15 +399,25 @@ void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos, const T* lu,
                                               const int*, const int*,          \
                                               const int*, int, T*, int,        \
                                               hipStream_t);                    \
-    template void ilu0_fwd_launch<T>(const int*, const int*, const int*,       \
-                                     const T*, const int*, int, const T*, T*,  \
-                                     int, hipStream_t);                        \
-    template void ilu0_bwd_launch<T>(const int*, const int*, const int*,       \
-                                     const T*, const int*, const int*, int,    \
-                                     const T*, T*, int, hipStream_t);
+    template void ilu0_fwd_launch<T, T>(const int*, const int*, const int*,    \
+                                        const T*, const int*, int, const T*,   \
+                                        T*, int, hipStream_t);                 \
+    template void ilu0_bwd_launch<T, T>(const int*, const int*, const int*,    \
+                                        const T*, const int*, const int*,      \
+                                        int, const T*, T*, int, hipStream_t);
 
 INSTANTIATE_CLASSICAL(double)
 INSTANTIATE_CLASSICAL(float)
+// mixed dDFI ILU apply: float factor, double vectors
+template void ilu0_fwd_launch<float, double>(const int*, const int*,
+                                             const int*, const float*,
+                                             const int*, int, const double*,
+                                             double*, int, hipStream_t);
+template void ilu0_bwd_launch<float, double>(const int*, const int*,
+                                             const int*, const float*,
+                                             const int*, const int*, int,
+                                             const double*, double*, int,
+                                             hipStream_t);
 
 }  // namespace amgx_hip
 

@@ -16,105 +16,107 @@
 namespace amgx_hip {
 
 // ============================================================ SpMV
+// Mixed precision (reference dDFI modes): TA = matrix-value type, TV =
+// vector type; products accumulate in TV (double when mixed).
 // thread-per-row: right shape for stencil-like rows (<= ~16 nnz).
-template <typename T, int UNROLL>
+template <typename TA, typename TV, int UNROLL>
 __global__ __launch_bounds__(AMGX_BLOCK) void csrmv_tpr(const int* __restrict__ ro, const int* __restrict__ ci,
-                          const T* __restrict__ va, const T* __restrict__ x,
-                          T* __restrict__ y, const T* __restrict__ bvec,
-                          T alpha, T beta, T gamma, int r0, int r1) {
+                          const TA* __restrict__ va, const TV* __restrict__ x,
+                          TV* __restrict__ y, const TV* __restrict__ bvec,
+                          TV alpha, TV beta, TV gamma, int r0, int r1) {
     int i = r0 + blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= r1) return;
     int s = ro[i], e = ro[i + 1];
-    T sum = T(0);
+    TV sum = TV(0);
     int k = s;
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
-        if (k < e) { sum += va[k] * x[ci[k]]; ++k; }
+        if (k < e) { sum += (TV)va[k] * x[ci[k]]; ++k; }
     }
-    for (; k < e; ++k) sum += va[k] * x[ci[k]];
-    T out = alpha * sum;
-    if (beta != T(0)) out += beta * y[i];
+    for (; k < e; ++k) sum += (TV)va[k] * x[ci[k]];
+    TV out = alpha * sum;
+    if (beta != TV(0)) out += beta * y[i];
     if (bvec) out += gamma * bvec[i];
     y[i] = out;
 }
 
 // L lanes cooperate on one row: for high-degree rows (unstructured matrices).
-template <typename T, int L>
+template <typename TA, typename TV, int L>
 __global__ __launch_bounds__(AMGX_BLOCK) void csrmv_vec(const int* __restrict__ ro, const int* __restrict__ ci,
-                          const T* __restrict__ va, const T* __restrict__ x,
-                          T* __restrict__ y, const T* __restrict__ bvec,
-                          T alpha, T beta, T gamma, int r0, int r1) {
+                          const TA* __restrict__ va, const TV* __restrict__ x,
+                          TV* __restrict__ y, const TV* __restrict__ bvec,
+                          TV alpha, TV beta, TV gamma, int r0, int r1) {
     const int lane = threadIdx.x & (L - 1);
     int i = r0 + (blockIdx.x * blockDim.x + threadIdx.x) / L;
     if (i >= r1) return;
     int s = ro[i], e = ro[i + 1];
-    T sum = T(0);
-    for (int k = s + lane; k < e; k += L) sum += va[k] * x[ci[k]];
+    TV sum = TV(0);
+    for (int k = s + lane; k < e; k += L) sum += (TV)va[k] * x[ci[k]];
 #pragma unroll
     for (int off = L / 2; off > 0; off >>= 1) sum += __shfl_down(sum, off, L);
     if (lane == 0) {
-        T out = alpha * sum;
-        if (beta != T(0)) out += beta * y[i];
+        TV out = alpha * sum;
+        if (beta != TV(0)) out += beta * y[i];
         if (bvec) out += gamma * bvec[i];
         y[i] = out;
     }
 }
 
-template <typename T>
-void csrmv(const int* ro, const int* ci, const T* va, const T* x, T* y,
-           const T* bvec, T alpha, T beta, T gamma, int r0, int r1,
+template <typename TA, typename TV>
+void csrmv(const int* ro, const int* ci, const TA* va, const TV* x, TV* y,
+           const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            double avg_deg, hipStream_t s) {
     long long rows = (long long)r1 - r0;
     if (rows <= 0) return;
     if (avg_deg <= 16.0) {
-        hipLaunchKernelGGL((csrmv_tpr<T, 8>), dim3(grid_1d(rows)),
+        hipLaunchKernelGGL((csrmv_tpr<TA, TV, 8>), dim3(grid_1d(rows)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, x, y, bvec,
                            alpha, beta, gamma, r0, r1);
     } else if (avg_deg <= 64.0) {
-        hipLaunchKernelGGL((csrmv_vec<T, 8>), dim3(grid_1d(rows * 8)),
+        hipLaunchKernelGGL((csrmv_vec<TA, TV, 8>), dim3(grid_1d(rows * 8)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, x, y, bvec,
                            alpha, beta, gamma, r0, r1);
     } else {
-        hipLaunchKernelGGL((csrmv_vec<T, 32>), dim3(grid_1d(rows * 32)),
+        hipLaunchKernelGGL((csrmv_vec<TA, TV, 32>), dim3(grid_1d(rows * 32)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, x, y, bvec,
                            alpha, beta, gamma, r0, r1);
     }
 }
 
 // block-CSR: one thread per output row component (row i, comp r).
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void bsrmv_kernel(const int* __restrict__ ro,
                              const int* __restrict__ ci,
-                             const T* __restrict__ va, int b,
-                             const T* __restrict__ x, T* __restrict__ y,
-                             const T* __restrict__ bvec, T alpha, T beta,
-                             T gamma, int r0, int r1) {
+                             const TA* __restrict__ va, int b,
+                             const TV* __restrict__ x, TV* __restrict__ y,
+                             const TV* __restrict__ bvec, TV alpha, TV beta,
+                             TV gamma, int r0, int r1) {
     long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
     long long nrows = (long long)(r1 - r0) * b;
     if (t >= nrows) return;
     int i = r0 + (int)(t / b);
     int comp = (int)(t % b);
     int st = ro[i], e = ro[i + 1];
-    T sum = T(0);
+    TV sum = TV(0);
     for (int k = st; k < e; ++k) {
-        const T* blk = va + (long long)k * b * b + (long long)comp * b;
-        const T* xs = x + (long long)ci[k] * b;
-        for (int c = 0; c < b; ++c) sum += blk[c] * xs[c];
+        const TA* blk = va + (long long)k * b * b + (long long)comp * b;
+        const TV* xs = x + (long long)ci[k] * b;
+        for (int c = 0; c < b; ++c) sum += (TV)blk[c] * xs[c];
     }
     long long oi = (long long)i * b + comp;
-    T out = alpha * sum;
-    if (beta != T(0)) out += beta * y[oi];
+    TV out = alpha * sum;
+    if (beta != TV(0)) out += beta * y[oi];
     if (bvec) out += gamma * bvec[oi];
     y[oi] = out;
 }
 
-template <typename T>
-void bsrmv(const int* ro, const int* ci, const T* va, int b, const T* x, T* y,
-           const T* bvec, T alpha, T beta, T gamma, int r0, int r1,
+template <typename TA, typename TV>
+void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
+           TV* y, const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            hipStream_t s) {
     long long rows = ((long long)r1 - r0) * b;
     if (rows <= 0) return;
-    hipLaunchKernelGGL((bsrmv_kernel<T>), dim3(grid_1d(rows)),
+    hipLaunchKernelGGL((bsrmv_kernel<TA, TV>), dim3(grid_1d(rows)),
                        dim3(AMGX_BLOCK), 0, s, ro, ci, va, b, x, y, bvec,
                        alpha, beta, gamma, r0, r1);
 }
@@ -357,30 +359,31 @@ void jacobi_dinv(const int* ro, const int* ci, const T* va, const int* didx,
 }
 
 // fused damped Jacobi sweep: xo = xi + omega*dinv*(b - A xi), single pass.
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void jacobi_smooth_scalar(const int* __restrict__ ro,
                                      const int* __restrict__ ci,
-                                     const T* __restrict__ va,
-                                     const T* __restrict__ dinv,
-                                     const T* __restrict__ bvec,
-                                     const T* __restrict__ xi,
-                                     T* __restrict__ xo, T omega, int n) {
+                                     const TA* __restrict__ va,
+                                     const TA* __restrict__ dinv,
+                                     const TV* __restrict__ bvec,
+                                     const TV* __restrict__ xi,
+                                     TV* __restrict__ xo, TV omega, int n) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     int s = ro[i], e = ro[i + 1];
-    T sum = T(0);
-    for (int k = s; k < e; ++k) sum += va[k] * xi[ci[k]];
-    xo[i] = xi[i] + omega * dinv[i] * (bvec[i] - sum);
+    TV sum = TV(0);
+    for (int k = s; k < e; ++k) sum += (TV)va[k] * xi[ci[k]];
+    xo[i] = xi[i] + omega * (TV)dinv[i] * (bvec[i] - sum);
 }
 
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void jacobi_smooth_block(const int* __restrict__ ro,
                                     const int* __restrict__ ci,
-                                    const T* __restrict__ va,
-                                    const T* __restrict__ dinv,
-                                    const T* __restrict__ bvec,
-                                    const T* __restrict__ xi,
-                                    T* __restrict__ xo, T omega, int n, int b) {
+                                    const TA* __restrict__ va,
+                                    const TA* __restrict__ dinv,
+                                    const TV* __restrict__ bvec,
+                                    const TV* __restrict__ xi,
+                                    TV* __restrict__ xo, TV omega, int n,
+                                    int b) {
     long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= (long long)n * b) return;
     int i = (int)(t / b);
@@ -388,99 +391,99 @@ __global__ __launch_bounds__(AMGX_BLOCK) void jacobi_smooth_block(const int* __r
     int s = ro[i], e = ro[i + 1];
     // residual components of the whole block-row are needed for dinv apply;
     // recompute per output comp: r_c = b_c - sum_k blk[c,:] x  (c = 0..b-1)
-    T upd = T(0);
+    TV upd = TV(0);
     for (int c = 0; c < b; ++c) {
-        T sum = T(0);
+        TV sum = TV(0);
         for (int k = s; k < e; ++k) {
-            const T* blk = va + ((long long)k * b + c) * b;
-            const T* xs = xi + (long long)ci[k] * b;
-            for (int q = 0; q < b; ++q) sum += blk[q] * xs[q];
+            const TA* blk = va + ((long long)k * b + c) * b;
+            const TV* xs = xi + (long long)ci[k] * b;
+            for (int q = 0; q < b; ++q) sum += (TV)blk[q] * xs[q];
         }
-        T rc = bvec[(long long)i * b + c] - sum;
-        upd += dinv[((long long)i * b + rcomp) * b + c] * rc;
+        TV rc = bvec[(long long)i * b + c] - sum;
+        upd += (TV)dinv[((long long)i * b + rcomp) * b + c] * rc;
     }
     xo[t] = xi[t] + omega * upd;
 }
 
-template <typename T>
-void jacobi_smooth(const int* ro, const int* ci, const T* va, const T* dinv,
-                   const T* bvec, const T* xi, T* xo, T omega, int n, int b,
-                   double avg_deg, hipStream_t s) {
+template <typename TA, typename TV>
+void jacobi_smooth(const int* ro, const int* ci, const TA* va, const TA* dinv,
+                   const TV* bvec, const TV* xi, TV* xo, TV omega, int n,
+                   int b, double avg_deg, hipStream_t s) {
     if (b == 1) {
-        hipLaunchKernelGGL((jacobi_smooth_scalar<T>), dim3(grid_1d(n)),
+        hipLaunchKernelGGL((jacobi_smooth_scalar<TA, TV>), dim3(grid_1d(n)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, dinv, bvec, xi,
                            xo, omega, n);
     } else {
-        hipLaunchKernelGGL((jacobi_smooth_block<T>),
+        hipLaunchKernelGGL((jacobi_smooth_block<TA, TV>),
                            dim3(grid_1d((long long)n * b)), dim3(AMGX_BLOCK),
                            0, s, ro, ci, va, dinv, bvec, xi, xo, omega, n, b);
     }
 }
 
 // ============================================================ multicolor GS
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_scalar(const int* __restrict__ ro,
                                const int* __restrict__ ci,
-                               const T* __restrict__ va,
-                               const T* __restrict__ dinv,
-                               const T* __restrict__ bvec, T* __restrict__ x,
+                               const TA* __restrict__ va,
+                               const TA* __restrict__ dinv,
+                               const TV* __restrict__ bvec, TV* __restrict__ x,
                                const int* __restrict__ rows, int count,
-                               T omega) {
+                               TV omega) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
     int s = ro[i], e = ro[i + 1];
-    T sum = T(0);
-    for (int k = s; k < e; ++k) sum += va[k] * x[ci[k]];
-    x[i] += omega * dinv[i] * (bvec[i] - sum);
+    TV sum = TV(0);
+    for (int k = s; k < e; ++k) sum += (TV)va[k] * x[ci[k]];
+    x[i] += omega * (TV)dinv[i] * (bvec[i] - sum);
 }
 
 // one THREAD per block-row: all b components are computed before any write,
 // so the in-place update is race-free (same-color rows have no mutual edges
 // under a valid distance-1 coloring; the only intra-row hazard is the
 // diagonal block, which this thread owns entirely).
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_block(const int* __restrict__ ro,
                               const int* __restrict__ ci,
-                              const T* __restrict__ va,
-                              const T* __restrict__ dinv,
-                              const T* __restrict__ bvec, T* __restrict__ x,
-                              const int* __restrict__ rows, int count, T omega,
-                              int b) {
+                              const TA* __restrict__ va,
+                              const TA* __restrict__ dinv,
+                              const TV* __restrict__ bvec, TV* __restrict__ x,
+                              const int* __restrict__ rows, int count,
+                              TV omega, int b) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
     int s = ro[i], e = ro[i + 1];
-    T res[16], upd[16];   // b <= 16 (reference supports block sizes to 10)
+    TV res[16], upd[16];  // b <= 16 (reference supports block sizes to 10)
     for (int c = 0; c < b; ++c) {
-        T sum = T(0);
+        TV sum = TV(0);
         for (int k = s; k < e; ++k) {
-            const T* blk = va + ((long long)k * b + c) * b;
-            const T* xs = x + (long long)ci[k] * b;
-            for (int q = 0; q < b; ++q) sum += blk[q] * xs[q];
+            const TA* blk = va + ((long long)k * b + c) * b;
+            const TV* xs = x + (long long)ci[k] * b;
+            for (int q = 0; q < b; ++q) sum += (TV)blk[q] * xs[q];
         }
         res[c] = bvec[(long long)i * b + c] - sum;
     }
-    const T* D = dinv + (long long)i * b * b;
+    const TA* D = dinv + (long long)i * b * b;
     for (int c = 0; c < b; ++c) {
-        T s2 = T(0);
-        for (int q = 0; q < b; ++q) s2 += D[c * b + q] * res[q];
+        TV s2 = TV(0);
+        for (int q = 0; q < b; ++q) s2 += (TV)D[c * b + q] * res[q];
         upd[c] = s2;
     }
     for (int c = 0; c < b; ++c) x[(long long)i * b + c] += omega * upd[c];
 }
 
-template <typename T>
-void gs_smooth_rows(const int* ro, const int* ci, const T* va, const T* dinv,
-                    const T* bvec, T* x, const int* rows, int count, T omega,
-                    int n, int b, hipStream_t s) {
+template <typename TA, typename TV>
+void gs_smooth_rows(const int* ro, const int* ci, const TA* va,
+                    const TA* dinv, const TV* bvec, TV* x, const int* rows,
+                    int count, TV omega, int n, int b, hipStream_t s) {
     if (count <= 0) return;
     if (b == 1) {
-        hipLaunchKernelGGL((gs_rows_scalar<T>), dim3(grid_1d(count)),
+        hipLaunchKernelGGL((gs_rows_scalar<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, dinv, bvec, x,
                            rows, count, omega);
     } else {
-        hipLaunchKernelGGL((gs_rows_block<T>), dim3(grid_1d(count)),
+        hipLaunchKernelGGL((gs_rows_block<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, dinv, bvec, x,
                            rows, count, omega, b);
     }
@@ -595,124 +598,126 @@ void dilu_setup_color(const int* ro, const int* ci, const T* va,
 // forward: w_i = Einv_i (r_i - sum_{color(j)<c} A_ij w_j); w pre-zeroed so the
 // full-row product only picks up earlier colors (valid coloring => no
 // same-color off-diagonals; diagonal contributes w_i = 0).
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar(const int* __restrict__ ro,
                                 const int* __restrict__ ci,
-                                const T* __restrict__ va,
-                                const T* __restrict__ einv,
+                                const TA* __restrict__ va,
+                                const TA* __restrict__ einv,
                                 const int* __restrict__ rows, int count,
-                                const T* __restrict__ r, T* __restrict__ w) {
+                                const TV* __restrict__ r, TV* __restrict__ w) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
-    T sum = T(0);
-    for (int k = ro[i]; k < ro[i + 1]; ++k) sum += va[k] * w[ci[k]];
-    w[i] = einv[i] * (r[i] - sum);
+    TV sum = TV(0);
+    for (int k = ro[i]; k < ro[i + 1]; ++k) sum += (TV)va[k] * w[ci[k]];
+    w[i] = (TV)einv[i] * (r[i] - sum);
 }
 
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_block(const int* __restrict__ ro,
                                const int* __restrict__ ci,
-                               const T* __restrict__ va,
-                               const T* __restrict__ einv,
+                               const TA* __restrict__ va,
+                               const TA* __restrict__ einv,
                                const int* __restrict__ rows, int count,
-                               const T* __restrict__ r, T* __restrict__ w,
+                               const TV* __restrict__ r, TV* __restrict__ w,
                                int b) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
-    T acc[16];            // b <= 16
+    TV acc[16];           // b <= 16
     for (int c = 0; c < b; ++c) {
-        T sum = T(0);
+        TV sum = TV(0);
         for (int k = ro[i]; k < ro[i + 1]; ++k) {
             if (ci[k] == i) continue;
-            const T* blk = va + ((long long)k * b + c) * b;
-            const T* ws = w + (long long)ci[k] * b;
-            for (int q = 0; q < b; ++q) sum += blk[q] * ws[q];
+            const TA* blk = va + ((long long)k * b + c) * b;
+            const TV* ws = w + (long long)ci[k] * b;
+            for (int q = 0; q < b; ++q) sum += (TV)blk[q] * ws[q];
         }
         acc[c] = r[(long long)i * b + c] - sum;
     }
-    const T* E = einv + (long long)i * b * b;
+    const TA* E = einv + (long long)i * b * b;
     for (int c = 0; c < b; ++c) {
-        T s = T(0);
-        for (int q = 0; q < b; ++q) s += E[c * b + q] * acc[q];
+        TV s = TV(0);
+        for (int q = 0; q < b; ++q) s += (TV)E[c * b + q] * acc[q];
         w[(long long)i * b + c] = s;
     }
 }
 
 // backward: z_i = w_i - Einv_i sum_{color(j)>c} A_ij z_j; z pre-zeroed and
 // filled color-descending, so a full-row product sees only later colors.
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_scalar(const int* __restrict__ ro,
                                 const int* __restrict__ ci,
-                                const T* __restrict__ va,
-                                const T* __restrict__ einv,
+                                const TA* __restrict__ va,
+                                const TA* __restrict__ einv,
                                 const int* __restrict__ rows, int count,
-                                const T* __restrict__ w, T* __restrict__ z) {
+                                const TV* __restrict__ w, TV* __restrict__ z) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
-    T sum = T(0);
-    for (int k = ro[i]; k < ro[i + 1]; ++k) sum += va[k] * z[ci[k]];
-    z[i] = w[i] - einv[i] * sum;
+    TV sum = TV(0);
+    for (int k = ro[i]; k < ro[i + 1]; ++k) sum += (TV)va[k] * z[ci[k]];
+    z[i] = w[i] - (TV)einv[i] * sum;
 }
 
-template <typename T>
+template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_block(const int* __restrict__ ro,
                                const int* __restrict__ ci,
-                               const T* __restrict__ va,
-                               const T* __restrict__ einv,
+                               const TA* __restrict__ va,
+                               const TA* __restrict__ einv,
                                const int* __restrict__ rows, int count,
-                               const T* __restrict__ w, T* __restrict__ z,
+                               const TV* __restrict__ w, TV* __restrict__ z,
                                int b) {
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
-    T acc[16];            // b <= 16
+    TV acc[16];           // b <= 16
     for (int c = 0; c < b; ++c) {
-        T sum = T(0);
+        TV sum = TV(0);
         for (int k = ro[i]; k < ro[i + 1]; ++k) {
             if (ci[k] == i) continue;
-            const T* blk = va + ((long long)k * b + c) * b;
-            const T* zs = z + (long long)ci[k] * b;
-            for (int q = 0; q < b; ++q) sum += blk[q] * zs[q];
+            const TA* blk = va + ((long long)k * b + c) * b;
+            const TV* zs = z + (long long)ci[k] * b;
+            for (int q = 0; q < b; ++q) sum += (TV)blk[q] * zs[q];
         }
         acc[c] = sum;
     }
-    const T* E = einv + (long long)i * b * b;
+    const TA* E = einv + (long long)i * b * b;
     for (int c = 0; c < b; ++c) {
-        T s = T(0);
-        for (int q = 0; q < b; ++q) s += E[c * b + q] * acc[q];
+        TV s = TV(0);
+        for (int q = 0; q < b; ++q) s += (TV)E[c * b + q] * acc[q];
         z[(long long)i * b + c] = w[(long long)i * b + c] - s;
     }
 }
 
-template <typename T>
-void dilu_fwd_color(const int* ro, const int* ci, const T* va, const T* einv,
-                    const int* colors, const int* rows, int count, int color,
-                    const T* r, T* w, int b, hipStream_t s) {
+template <typename TA, typename TV>
+void dilu_fwd_color(const int* ro, const int* ci, const TA* va,
+                    const TA* einv, const int* colors, const int* rows,
+                    int count, int color, const TV* r, TV* w, int b,
+                    hipStream_t s) {
     if (count <= 0) return;
     if (b == 1)
-        hipLaunchKernelGGL((dilu_fwd_scalar<T>), dim3(grid_1d(count)),
+        hipLaunchKernelGGL((dilu_fwd_scalar<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
                            count, r, w);
     else
-        hipLaunchKernelGGL((dilu_fwd_block<T>), dim3(grid_1d(count)),
+        hipLaunchKernelGGL((dilu_fwd_block<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
                            count, r, w, b);
 }
 
-template <typename T>
-void dilu_bwd_color(const int* ro, const int* ci, const T* va, const T* einv,
-                    const int* colors, const int* rows, int count, int color,
-                    const T* w, T* z, int b, hipStream_t s) {
+template <typename TA, typename TV>
+void dilu_bwd_color(const int* ro, const int* ci, const TA* va,
+                    const TA* einv, const int* colors, const int* rows,
+                    int count, int color, const TV* w, TV* z, int b,
+                    hipStream_t s) {
     if (count <= 0) return;
     if (b == 1)
-        hipLaunchKernelGGL((dilu_bwd_scalar<T>), dim3(grid_1d(count)),
+        hipLaunchKernelGGL((dilu_bwd_scalar<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
                            count, w, z);
     else
-        hipLaunchKernelGGL((dilu_bwd_block<T>), dim3(grid_1d(count)),
+        hipLaunchKernelGGL((dilu_bwd_block<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
                            count, w, z, b);
 }
@@ -791,27 +796,27 @@ void prolongate_agg(T* x, const T* xc, const int* agg, int n, int b,
 
 // ============================================================ dense GEMV
 // coarse solve x = Ainv b; n <= a few hundred -> wave-per-row.
-template <typename T>
-__global__ __launch_bounds__(AMGX_BLOCK) void dense_gemv_kernel(const T* __restrict__ Ainv,
-                                  const T* __restrict__ b, T* __restrict__ x,
-                                  int n) {
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void dense_gemv_kernel(const TA* __restrict__ Ainv,
+                                  const TV* __restrict__ b,
+                                  TV* __restrict__ x, int n) {
     int row = blockIdx.x * (blockDim.x / WAVE_SIZE) + threadIdx.x / WAVE_SIZE;
     int lane = threadIdx.x & (WAVE_SIZE - 1);
     if (row >= n) return;
-    T sum = T(0);
-    const T* arow = Ainv + (long long)row * n;
-    for (int j = lane; j < n; j += WAVE_SIZE) sum += arow[j] * b[j];
+    TV sum = TV(0);
+    const TA* arow = Ainv + (long long)row * n;
+    for (int j = lane; j < n; j += WAVE_SIZE) sum += (TV)arow[j] * b[j];
     sum = wave_reduce_sum(sum);
     if (lane == 0) x[row] = sum;
 }
 
-template <typename T>
-void dense_gemv(const T* Ainv, const T* b, T* x, int n, hipStream_t s) {
+template <typename TA, typename TV>
+void dense_gemv(const TA* Ainv, const TV* b, TV* x, int n, hipStream_t s) {
     if (n <= 0) return;
     int waves_per_block = AMGX_BLOCK / WAVE_SIZE;
     int g = (n + waves_per_block - 1) / waves_per_block;
-    hipLaunchKernelGGL((dense_gemv_kernel<T>), dim3(g), dim3(AMGX_BLOCK), 0, s,
-                       Ainv, b, x, n);
+    hipLaunchKernelGGL((dense_gemv_kernel<TA, TV>), dim3(g), dim3(AMGX_BLOCK),
+                       0, s, Ainv, b, x, n);
 }
 
 // ============================================================ gather/scatter
@@ -868,11 +873,8 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
 }
 
 // ============================================================ instantiation
+// vector-typed ops (single type)
 #define INSTANTIATE(T)                                                          \
-    template void csrmv<T>(const int*, const int*, const T*, const T*, T*,      \
-                           const T*, T, T, T, int, int, double, hipStream_t);   \
-    template void bsrmv<T>(const int*, const int*, const T*, int, const T*,     \
-                           T*, const T*, T, T, T, int, int, hipStream_t);       \
     template void reduce<T>(const T*, const T*, long long, int, T*, T*,         \
                             hipStream_t);                                       \
     template void axpy<T>(T*, const T*, T, long long, hipStream_t);             \
@@ -882,35 +884,52 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
                                   const int*, int, int, T*, hipStream_t);       \
     template void jacobi_dinv<T>(const int*, const int*, const T*, const int*,  \
                                  int, int, bool, T*, hipStream_t);              \
-    template void jacobi_smooth<T>(const int*, const int*, const T*, const T*,  \
-                                   const T*, const T*, T*, T, int, int,         \
-                                   double, hipStream_t);                        \
-    template void gs_smooth_rows<T>(const int*, const int*, const T*,           \
-                                    const T*, const T*, T*, const int*, int,    \
-                                    T, int, int, hipStream_t);                  \
     template void dilu_setup_color<T>(const int*, const int*, const T*,         \
                                       const int*, const int*, const int*,       \
                                       const int*, int, int, T*, int,            \
                                       hipStream_t);                             \
-    template void dilu_fwd_color<T>(const int*, const int*, const T*,           \
-                                    const T*, const int*, const int*, int,      \
-                                    int, const T*, T*, int, hipStream_t);       \
-    template void dilu_bwd_color<T>(const int*, const int*, const T*,           \
-                                    const T*, const int*, const int*, int,      \
-                                    int, const T*, T*, int, hipStream_t);       \
     template void restrict_agg<T>(const T*, const int*, int, int, T*,           \
                                   hipStream_t);                                 \
     template void restrict_csr<T>(const int*, const int*, const T*, int, int,   \
-                                  T*, hipStream_t);                                 \
+                                  T*, hipStream_t);                             \
     template void prolongate_agg<T>(T*, const T*, const int*, int, int,         \
                                     hipStream_t);                               \
-    template void dense_gemv<T>(const T*, const T*, T*, int, hipStream_t);      \
     template void gather<T>(const T*, const int*, int, int, T*, hipStream_t);   \
     template void scatter<T>(const T*, const int*, int, int, T*, hipStream_t);  \
     template void scatter_add<T>(const T*, const int*, int, int, T*,            \
                                  hipStream_t);
 
+// matrix-value x vector mixed ops: (TA, TV) in {(d,d), (f,f), (f,d)} — the
+// reference's dDDI / dFFI / dDFI value modes
+#define INSTANTIATE_MIXED(TA, TV)                                               \
+    template void csrmv<TA, TV>(const int*, const int*, const TA*, const TV*,   \
+                                TV*, const TV*, TV, TV, TV, int, int, double,   \
+                                hipStream_t);                                   \
+    template void bsrmv<TA, TV>(const int*, const int*, const TA*, int,         \
+                                const TV*, TV*, const TV*, TV, TV, TV, int,     \
+                                int, hipStream_t);                              \
+    template void jacobi_smooth<TA, TV>(const int*, const int*, const TA*,      \
+                                        const TA*, const TV*, const TV*, TV*,   \
+                                        TV, int, int, double, hipStream_t);     \
+    template void gs_smooth_rows<TA, TV>(const int*, const int*, const TA*,     \
+                                         const TA*, const TV*, TV*,             \
+                                         const int*, int, TV, int, int,         \
+                                         hipStream_t);                          \
+    template void dilu_fwd_color<TA, TV>(const int*, const int*, const TA*,     \
+                                         const TA*, const int*, const int*,     \
+                                         int, int, const TV*, TV*, int,         \
+                                         hipStream_t);                          \
+    template void dilu_bwd_color<TA, TV>(const int*, const int*, const TA*,     \
+                                         const TA*, const int*, const int*,     \
+                                         int, int, const TV*, TV*, int,         \
+                                         hipStream_t);                          \
+    template void dense_gemv<TA, TV>(const TA*, const TV*, TV*, int,            \
+                                     hipStream_t);
+
 INSTANTIATE(double)
 INSTANTIATE(float)
+INSTANTIATE_MIXED(double, double)
+INSTANTIATE_MIXED(float, float)
+INSTANTIATE_MIXED(float, double)
 
 }  // namespace amgx_hip

@@ -35,11 +35,12 @@ def _tidx(A):
     return t
 
 
-def _scratch(A, name, numel):
-    key = ("scratch", name, numel)
+def _scratch(A, name, numel, dtype=None):
+    dtype = dtype if dtype is not None else A.dtype
+    key = ("scratch", name, numel, dtype)
     t = A._cache.get(key)
     if t is None:
-        t = torch.empty(numel, dtype=A.dtype, device=A.device)
+        t = torch.empty(numel, dtype=dtype, device=A.device)
         A._cache[key] = t
     return t
 
@@ -163,8 +164,8 @@ def dilu_setup(A, coloring):
 
 def dilu_solve(A, Einv, coloring, r, relaxation, x):
     n = A.n_cols * A.block_dim   # ext size: halo tails stay zero in the sweeps
-    w = _scratch(A, "dilu_w", n)
-    z = _scratch(A, "dilu_z", n)
+    w = _scratch(A, "dilu_w", n, r.dtype)   # vector precision (dDFI mixed)
+    z = _scratch(A, "dilu_z", n, r.dtype)
     _core.dilu_apply(A.row_offsets, A.col_indices, A.values, A.block_dim,
                      Einv, coloring.colors, coloring.rows_sorted,
                      coloring.bounds, r.reshape(-1), w, z, x.reshape(-1),
@@ -321,8 +322,8 @@ def ilu0_setup(A, coloring):
 
 def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
     n = A.n_cols * A.block_dim
-    y = _scratch(A, "ilu_y", n)
-    z = _scratch(A, "ilu_z", n)
+    y = _scratch(A, "ilu_y", n, r.dtype)    # vector precision (dDFI mixed)
+    z = _scratch(A, "ilu_z", n, r.dtype)
     pos = _color_pos(A, coloring)
     _core.ilu0_apply(A.row_offsets, A.col_indices, factors, _didx(A), pos,
                      coloring.rows_sorted, coloring.bounds, r.reshape(-1),

@@ -103,6 +103,12 @@ def build_local_matrix(size, rank, world, device, problem="poisson"):
         from amgx_amd.problems import block_laplacian
         side = max(int(round(size ** 1.5 / world ** 0.5)), 8)
         Afull = block_laplacian(side, side * world, block_dim=4, seed=9)
+        # BASELINE config #4 is fp32/fp64 MIXED: fp32 matrix storage
+        # (half the SpMV bytes), fp64 vectors end to end
+        from amgx_amd.matrix import CSRMatrix
+        Afull = CSRMatrix(Afull.row_offsets, Afull.col_indices,
+                          Afull.values.to(torch.float32),
+                          n_cols=Afull.n_cols, block_dim=4)
         if world == 1:
             return Afull.to(device), None
         from amgx_amd.distributed.manager import DistributedManager
@@ -115,7 +121,8 @@ def build_local_matrix(size, rank, world, device, problem="poisson"):
         A = DistributedManager.upload_global_csr(
             ro[lo:hi + 1] - s0, Afull.col_indices.numpy()[s0:s1],
             Afull.values.numpy().reshape(Afull.nnz, 16)[s0:s1],
-            hi - lo, lo, n, device=device, block_dim=4)
+            hi - lo, lo, n, device=device, block_dim=4,
+            dtype=torch.float32)
         return A, A.manager
     if problem == "unstructured":
         # scrambled Poisson: random node relabeling destroys the banded
@@ -254,7 +261,8 @@ def main():
             "higher_is_better": False,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp64",
+            "dtype": "fp32/fp64 mixed" if args.config == "block4_dilu"
+                     else "fp64",
             "data": "synthetic",
             "config": {
                 "model": {

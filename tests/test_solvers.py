@@ -538,3 +538,30 @@ def test_degenerate_matrices():
     assert A2.nnz == 4
     y = ops.spmv(A2, torch.tensor([1.0, 2.0], dtype=torch.float64))
     assert torch.allclose(y, torch.tensor([4.0, 7.0], dtype=torch.float64))
+
+
+def test_mixed_precision_amg_cycle():
+    """hDFI through the full AMG stack: fp32 matrix/hierarchy, fp64 vectors
+    (reference mixed mode dDFI; cycle scratch re-keys to the rhs dtype)."""
+    from amgx_amd import AMGConfig
+    from amgx_amd.matrix import CSRMatrix
+    from amgx_amd.problems import poisson_3d
+    A64 = poisson_3d(10, 10, 10)
+    A = CSRMatrix(A64.row_offsets, A64.col_indices,
+                  A64.values.to(torch.float32), n_cols=A64.n_cols)
+    cfg = {"solver": {"preconditioner": {
+        "solver": "AMG", "algorithm": "AGGREGATION",
+        "smoother": "MULTICOLOR_DILU", "max_iters": 1,
+        "min_coarse_rows": 10, "cycle": "V"},
+        "solver": "FGMRES", "max_iters": 150, "gmres_n_restart": 30,
+        "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-6}}
+    s = create_solver(AMGConfig.from_dict(cfg).root_scope(),
+                      resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert x.dtype == torch.float64
+    rel = ops.nrm2(ops.residual(A64, x, b)) / ops.nrm2(b)
+    assert st.converged and rel < 1e-5, (st, rel)
