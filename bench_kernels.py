@@ -61,6 +61,12 @@ def main():
 
     rec("csrmv", bench(lambda: ops.spmv(A, x, y), args.iters, sync),
         spmv_bytes)
+    from amgx_amd.matrix import CSRMatrix
+    A32 = CSRMatrix(A.row_offsets, A.col_indices,
+                    A.values.to(torch.float32), n_cols=A.n_cols)
+    rec("csrmv_mixed_f32A", bench(lambda: ops.spmv(A32, x, y),
+                                  args.iters, sync),
+        nnz * 8 + (n + 1) * 4 + n * 16)   # fp32 vals: half the value bytes
     r = torch.zeros_like(x)
     rec("residual_fused", bench(lambda: ops.residual(A, x, b, r),
                                 args.iters, sync), spmv_bytes + n * 8)
