@@ -73,3 +73,29 @@ def test_flat_string():
 def test_registry_dump():
     d = json.loads(write_parameters_description())
     assert "tolerance" in d and d["strength_threshold"]["default"] == 0.25
+
+
+def test_all_shipped_configs_solve():
+    """Every JSON config in configs/ drives a working solve on a small
+    Poisson system (reference ships its configs/ as the contract)."""
+    import glob
+    import os
+
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+    here = os.path.join(os.path.dirname(__file__), "..", "configs")
+    files = sorted(glob.glob(os.path.join(here, "*.json")))
+    assert len(files) >= 12
+    for f in files:
+        cfg = AMGConfig.from_file(f)
+        s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+        A = poisson_3d(8, 8, 8)
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        st = s.solve(b, x, zero_initial_guess=True)
+        rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
+        assert st.converged and rel < 1e-4, (os.path.basename(f), st, rel)
