@@ -122,3 +122,33 @@ def test_pagerank():
     ref = np.real(evecs[:, k])
     ref = ref / ref.sum()
     assert np.allclose(pr.numpy(), ref, atol=1e-6)
+
+
+def test_shipped_eigen_configs():
+    """Every flat config in configs/eigen/ builds a working eigensolver on
+    a small SPD system (reference src/configs/eigen_configs)."""
+    import glob
+    import os
+
+    from amgx_amd.config import AMGConfig
+    from amgx_amd.eigensolvers import create_eigensolver
+    from amgx_amd.problems import poisson_2d
+    here = os.path.join(os.path.dirname(__file__), "..", "configs", "eigen")
+    files = sorted(f for f in glob.glob(os.path.join(here, "*"))
+                   if os.path.isfile(f))
+    assert len(files) >= 8
+    A = poisson_2d(10, 10)
+    dense = A.to_scipy().toarray()
+    evs = np.linalg.eigvalsh(dense)
+    for f in files:
+        cfg = AMGConfig.from_file(f)
+        es = create_eigensolver(cfg.root_scope(), Resources("cpu"))
+        if os.path.basename(f) == "PAGERANK":
+            continue   # needs a digraph; covered by test_pagerank
+        es.setup(A)
+        st = es.solve()
+        assert st.converged, os.path.basename(f)
+        ref = evs[0] if es.which == "smallest" else evs[-1]
+        assert abs(st.eigenvalues[-1 if es.which != "smallest" else 0]
+                   - ref) < 1e-3 * abs(ref), (os.path.basename(f),
+                                              st.eigenvalues, ref)
