@@ -1068,3 +1068,31 @@ def _impl_test_dist_block4_amg(rank, world, tmp):
     bn = mgr.global_norm(float(torch.linalg.vector_norm(
         b[:mgr.owned_size])), "L2")
     assert nrm / bn < 1e-7
+
+
+def test_dist_energymin():
+    _run_dist(test_dist_energymin)
+
+
+def _impl_test_dist_energymin(rank, world, tmp):
+    """Distributed ENERGYMIN level (PMIS selection + distributed EM
+    interpolation) under PCG."""
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.resources import Resources
+    A = _make_dist_A(rank, world, 6)
+    mgr = A.manager
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {"solver": "AMG", "algorithm": "ENERGYMIN",
+                           "smoother": "MULTICOLOR_GS", "symmetric_GS": 1,
+                           "max_iters": 1, "min_coarse_rows": 10,
+                           "cycle": "V"},
+        "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8}})
+    s = create_solver(cfg.root_scope(), resources=Resources(
+        "cpu", distributed=True))
+    b = mgr.new_ext_vec(torch.float64)
+    b[:mgr.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged and st.iterations <= 20, f"rank {rank}: {st}"
