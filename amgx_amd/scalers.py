@@ -77,7 +77,10 @@ class DiagonalSymmetricScaler(Scaler):
     name = "DIAGONAL_SYMMETRIC"
 
     def setup(self, A):
-        assert A.block_dim == 1, "scalar scaling only"
+        if A.block_dim != 1:
+            raise NotImplementedError(
+                "scalers operate on scalar matrices (reference parity: "
+                "src/scalers/* are 1x1-only)")
         d = A.diagonal().abs()
         d = torch.where(d > 0, d, torch.ones_like(d))
         self.d = (1.0 / torch.sqrt(d)).to(A.dtype)
@@ -95,7 +98,9 @@ class BinormalizationScaler(Scaler):
         self.sweeps = sweeps
 
     def setup(self, A):
-        assert A.block_dim == 1
+        if A.block_dim != 1:
+            raise NotImplementedError(
+                "scalers operate on scalar matrices (reference parity)")
         import scipy.sparse as sp
         m = A.to_scipy().tocsr()
         B = m.multiply(m)       # a_ij^2
