@@ -246,7 +246,7 @@ def main():
 
     sec_per_step = t_total / args.steps
     last = stats[-1]
-    dof = args.size ** 3 * n_gpus
+    dof = n_local * n_gpus   # actual per-rank unknowns (exact for block/unstructured too)
     if rank == 0:
         line = {
             "metric": ("amg_setup_solve_seconds_poisson256_fgmres_agg"
@@ -280,7 +280,7 @@ def main():
                 "global_batch": dof,
                 "seq_len": args.size,
                 "parallelism": f"dd{n_gpus}",
-                "rows_per_gpu": args.size ** 3,
+                "rows_per_gpu": n_local,
                 "iterations": last["iterations"],
                 "converged": last["converged"],
                 "setup_s": last["setup_s"],
