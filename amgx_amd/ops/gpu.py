@@ -3,6 +3,11 @@
 Import FAILS LOUDLY if the extension is missing — on a GPU machine there is no
 silent eager fallback (the HIP kernels ARE the product). Structural byproducts
 (diag index, transpose index, scratch vectors) are cached on the matrix.
+
+Per-kernel reference citations live in the HIP sources (csrc/kernels_*.hip:
+reference src/multiply.cu, src/blas.cu, src/norm.cu, src/solvers/*.cu,
+src/csr_multiply*.cu, src/matrix_coloring/*.cu, src/aggregation/*,
+src/classical/*); this module only routes tensors into them.
 """
 
 from __future__ import annotations
