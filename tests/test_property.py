@@ -93,3 +93,31 @@ def test_pcg_amg_always_converges_on_spd(n, seed):
     st_ = s.solve(b, x, zero_initial_guess=True)
     rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
     assert st_.converged and rel < 1e-6
+
+
+@settings(max_examples=15, deadline=None)
+@given(n=st.integers(4, 80), density=st.floats(0.02, 0.3),
+       seed=st.integers(0, 10_000), block=st.sampled_from([1, 2, 4]))
+def test_binary_io_roundtrip(n, density, seed, block):
+    import os
+    import tempfile
+
+    from amgx_amd.io.binary import read_system_binary, write_system_binary
+    if block == 1:
+        A = CSRMatrix.from_scipy(_rand_csr(n, density, seed))
+    else:
+        m = _rand_csr(n, density, seed)
+        bv = np.random.RandomState(seed).rand(m.nnz, block, block)
+        A = CSRMatrix.from_bsr(torch.from_numpy(m.indptr.astype(np.int32)),
+                               torch.from_numpy(m.indices.astype(np.int32)),
+                               torch.from_numpy(bv))
+    b = torch.rand(A.n_rows * block, dtype=torch.float64)
+    with tempfile.TemporaryDirectory() as d:
+        p = os.path.join(d, "s.bin")
+        write_system_binary(p, A, b, None)
+        A2, b2, _ = read_system_binary(p)
+    assert torch.equal(A2.row_offsets, A.row_offsets)
+    assert torch.equal(A2.col_indices, A.col_indices)
+    assert torch.allclose(A2.values.reshape(-1), A.values.reshape(-1))
+    assert A2.block_dim == block
+    assert torch.allclose(b2, b)
