@@ -121,3 +121,42 @@ def test_binary_io_roundtrip(n, density, seed, block):
     assert torch.allclose(A2.values.reshape(-1), A.values.reshape(-1))
     assert A2.block_dim == block
     assert torch.allclose(b2, b)
+
+
+@settings(max_examples=15, deadline=None)
+@given(n=st.integers(8, 80), density=st.floats(0.03, 0.25),
+       seed=st.integers(0, 10_000))
+def test_galerkin_matches_scipy_triple_product(n, density, seed):
+    m = _rand_csr(n, density, seed, spd=True)
+    A = CSRMatrix.from_scipy(m)
+    agg, nc = ops.size2_matching(A)
+    Ac = ops.galerkin_aggregation(A, agg, nc)
+    P = sp.csr_matrix((np.ones(n), (np.arange(n), agg.numpy())),
+                      shape=(n, nc))
+    ref = (P.T @ m @ P).toarray()
+    assert np.allclose(Ac.to_scipy().toarray(), ref, rtol=1e-12, atol=1e-12)
+
+
+@settings(max_examples=10, deadline=None)
+@given(n=st.integers(10, 60), seed=st.integers(0, 10_000),
+       scaling=st.sampled_from(["BINORMALIZATION", "DIAGONAL_SYMMETRIC",
+                                "NBINORMALIZATION"]))
+def test_scaler_lifecycle_restores_matrix(n, seed, scaling):
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.resources import Resources
+    A = CSRMatrix.from_scipy(_rand_csr(n, 0.15, seed, spd=True))
+    before = A.values.clone()
+    cfg = {"solver": {"solver": "PCG", "preconditioner": "BLOCK_JACOBI",
+                      "max_iters": 300, "monitor_residual": 1,
+                      "tolerance": 1e-8, "convergence": "RELATIVE_INI",
+                      "scaling": scaling}}
+    s = create_solver(AMGConfig.from_dict(cfg).root_scope(),
+                      resources=Resources("cpu"))
+    b = torch.ones(n, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st_ = s.solve(b, x, zero_initial_guess=True)
+    assert torch.allclose(A.values, before, rtol=1e-12, atol=1e-14)
+    assert st_.converged
+    rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
+    assert rel < 1e-5
