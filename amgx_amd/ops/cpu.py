@@ -615,7 +615,11 @@ def ilu0_setup(A, coloring):
     """ILU(0) factorization in color order (scalar). Returns factored values
     aligned with A's CSR structure. Host reference for the per-color GPU
     kernels (reference src/solvers/multicolor_ilu_solver.cu)."""
-    assert A.block_dim == 1, "block ILU: use DILU or scalar path"
+    if A.block_dim != 1:
+        raise NotImplementedError(
+            "MULTICOLOR_ILU is scalar-only here (block systems: use "
+            "MULTICOLOR_DILU, the reference's recommended block smoother; "
+            "block ILU(0) port pending)")
     ro = _np(A.row_offsets).astype(np.int64)
     ci = _np(A.col_indices).astype(np.int64)
     vals = _np(A.values).astype(np.float64).copy()
