@@ -319,6 +319,10 @@ def _color_pos(A, coloring):
 
 
 def ilu0_setup(A, coloring):
+    if A.block_dim != 1:
+        raise NotImplementedError(
+            "device block ILU(0) kernels pending — host modes support it; "
+            "on device use MULTICOLOR_DILU for block systems")
     pos = _color_pos(A, coloring)
     lu = _core.ilu0_setup(A.row_offsets, A.col_indices, A.values, _didx(A),
                           pos, coloring.rows_sorted, coloring.bounds)
@@ -326,6 +330,10 @@ def ilu0_setup(A, coloring):
 
 
 def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
+    if A.block_dim != 1:
+        raise NotImplementedError(
+            "device block ILU(0) kernels pending — host modes support it; "
+            "on device use MULTICOLOR_DILU for block systems")
     n = A.n_cols * A.block_dim
     y = _scratch(A, "ilu_y", n, r.dtype)    # vector precision (dDFI mixed)
     z = _scratch(A, "ilu_z", n, r.dtype)

@@ -70,6 +70,10 @@ class MulticolorILUSolver(_SmootherBase):
 
     def solver_setup(self):
         A = self.A
+        if self.sparsity_level > 0 and A.block_dim != 1:
+            raise NotImplementedError(
+                "ILU(k>0) fill patterns are scalar-only (block ILU(0) is "
+                "supported on host modes)")
         if self.sparsity_level > 0:
             # ILU(k): factor on the extended pattern; coloring computed on
             # that pattern so same-color rows stay decoupled (reference

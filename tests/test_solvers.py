@@ -565,3 +565,32 @@ def test_mixed_precision_amg_cycle():
     assert x.dtype == torch.float64
     rel = ops.nrm2(ops.residual(A64, x, b)) / ops.nrm2(b)
     assert st.converged and rel < 1e-5, (st, rel)
+
+
+def test_block_ilu0():
+    """Block ILU(0) on the host path (reference multicolor_ilu_solver.cu
+    bxb setup_LU): smooths a block-4 system and preconditions FGMRES."""
+    for bd in (2, 4):
+        A = block_laplacian(8, 8, block_dim=bd)
+        n = A.n_rows * bd
+        s = make({"solver": "MULTICOLOR_ILU", "max_iters": 6})
+        b = torch.ones(n, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        r0 = ops.nrm2(ops.residual(A, x, b))
+        s.solve(b, x)
+        r1 = ops.nrm2(ops.residual(A, x, b))
+        assert r1 < 0.6 * r0, (bd, r1 / r0)
+    A = block_laplacian(10, 10, block_dim=4)
+    s = make({"solver": "FGMRES", "max_iters": 200, "gmres_n_restart": 30,
+              "monitor_residual": 1, "tolerance": 1e-8,
+              "convergence": "RELATIVE_INI",
+              "preconditioner": {"solver": "MULTICOLOR_ILU",
+                                 "max_iters": 1}})
+    n = A.n_rows * 4
+    b = torch.ones(n, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+    assert st.converged and rel < 1e-7, (st, rel)
