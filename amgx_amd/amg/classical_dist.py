@@ -233,30 +233,52 @@ def rap_dist(A, mgr: DistributedManager, P_own,
         return (rows, C.col_indices[s:e].to(torch.int64),
                 C.values.reshape(-1)[s:e])
 
-    # external rows -> owners (reference exchange_RAP_ext + sparse_add)
-    frags = []
+    # external rows -> owners, point-to-point (reference exchange_RAP_ext +
+    # sparse_add, distributed_arranger exchange_RAP_ext): each rank ships its
+    # fragment for owner r straight to r as a packed (row*ngc+col, value)
+    # pair — no O(world^2) all-gather of everyone's fragments.
+    comm_dev = dev if (tdist.is_initialized()
+                       and tdist.get_backend() == "nccl") else \
+        torch.device("cpu")
+    send_keys, send_vals = {}, {}
+    counts = torch.zeros(mgr.world, dtype=torch.int64)
     for r in range(mgr.world):
         if r == mgr.rank:
-            frags.append(None)
             continue
         lo, hi = int(coarse_offs[r]), int(coarse_offs[r + 1])
         rr, cc, vv = _slice(lo, hi)
-        frags.append((rr.cpu().numpy(), cc.cpu().numpy(),
-                      vv.cpu().numpy()) if rr.numel() else None)
-    gathered = [None] * mgr.world
-    tdist.all_gather_object(gathered, frags)
+        if rr.numel():
+            # row ids shipped relative to the owner's slice start
+            send_keys[r] = (rr * ngc + cc).to(comm_dev)
+            send_vals[r] = vv.to(torch.float64).to(comm_dev)
+            counts[r] = rr.numel()
+    all_counts = [torch.zeros(mgr.world, dtype=torch.int64)
+                  for _ in range(mgr.world)]
+    tdist.all_gather(all_counts, counts)
+    p2p, recv = [], {}
+    for r in range(mgr.world):
+        if r == mgr.rank:
+            continue
+        if r in send_keys:
+            p2p.append(tdist.P2POp(tdist.isend, send_keys[r], r))
+            p2p.append(tdist.P2POp(tdist.isend, send_vals[r], r))
+        cnt = int(all_counts[r][mgr.rank])
+        if cnt:
+            rk = torch.empty(cnt, dtype=torch.int64, device=comm_dev)
+            rv = torch.empty(cnt, dtype=torch.float64, device=comm_dev)
+            recv[r] = (rk, rv)
+            p2p.append(tdist.P2POp(tdist.irecv, rk, r))
+            p2p.append(tdist.P2POp(tdist.irecv, rv, r))
+    if p2p:
+        for rq in tdist.batch_isend_irecv(p2p):
+            rq.wait()
     rows_l, cols_l, vals_l = _slice(cs, ce)
     parts_r, parts_c, parts_v = [rows_l], [cols_l], [vals_l]
-    for r in range(mgr.world):
-        if r == mgr.rank or gathered[r] is None:
-            continue
-        frag = gathered[r][mgr.rank]
-        if frag is None:
-            continue
-        rr, cc, vv = frag
-        parts_r.append(torch.from_numpy(rr).to(dev))
-        parts_c.append(torch.from_numpy(cc).to(dev))
-        parts_v.append(torch.from_numpy(vv).to(A.dtype).to(dev))
+    for r, (rk, rv) in sorted(recv.items()):
+        rk = rk.to(dev)
+        parts_r.append(rk // ngc)
+        parts_c.append(rk % ngc)
+        parts_v.append(rv.to(A.dtype).to(dev))
     rows = torch.cat(parts_r)
     cols = torch.cat(parts_c)
     vals = torch.cat(parts_v)

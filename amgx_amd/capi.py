@@ -345,37 +345,6 @@ def AMGX_matrix_upload_all(m: _MatrixHandle, n, nnz, block_dimx, block_dimy,
     return RC_OK
 
 
-def _fold_external_diag(n, b, ro, ci, va, diag):
-    """Merge the DIAG-property external diagonal into the CSR structure
-    (reference block-DIA-CSR, include/matrix.h:24-26). The gfx950 kernels
-    take one folded CSR — the external layout is an input format here."""
-    deg = np.diff(ro)
-    rows = np.repeat(np.arange(n, dtype=np.int64), deg)
-    all_rows = np.concatenate([rows, np.arange(n, dtype=np.int64)])
-    all_cols = np.concatenate([ci.astype(np.int64),
-                               np.arange(n, dtype=np.int64)])
-    vab = va.reshape(len(ci), -1)
-    db = diag.reshape(n, -1)
-    all_vals = np.concatenate([vab, db], axis=0)
-    order = np.lexsort((all_cols, all_rows))
-    ro2 = np.zeros(n + 1, dtype=np.int64)
-    np.cumsum(np.bincount(all_rows, minlength=n), out=ro2[1:])
-    cols2 = all_cols[order]
-    vals2 = all_vals[order]
-    # sum duplicates (a CSR that already stores an explicit diagonal)
-    ncols = int(max(cols2.max() + 1 if cols2.size else 1, n))
-    key = all_rows[order] * ncols + cols2
-    uniq, first = np.unique(key, return_index=True)
-    if uniq.size != key.size:
-        sums = np.add.reduceat(vals2, first, axis=0)
-        cols2 = cols2[first]
-        vals2 = sums
-        counts = np.bincount((uniq // ncols).astype(np.int64), minlength=n)
-        ro2 = np.zeros(n + 1, dtype=np.int64)
-        np.cumsum(counts, out=ro2[1:])
-    return ro2, cols2, vals2
-
-
 @_amgx_try
 def AMGX_matrix_upload_all_global(m: _MatrixHandle, n_global, n, nnz,
                                   block_dimx, block_dimy, row_ptrs,
@@ -656,11 +625,11 @@ def AMGX_solver_get_iteration_residual(s: _SolverHandle, it: int = -1,
 @_amgx_try
 def AMGX_read_system(m: _MatrixHandle, rhs: _VectorHandle,
                      sol: _VectorHandle, path: str):
-    from .io.binary import is_binary_file, read_system_binary
+    from .io.binary import is_binary_file, read_system_any
     from .io.matrix_market import read_system
     mem, vecprec, matprec = _parse_mode(m.mode)
     dev = mem if mem == "cpu" else m.res.device
-    reader = read_system_binary if is_binary_file(path) else read_system
+    reader = read_system_any if is_binary_file(path) else read_system
     A, b, x0 = reader(path, device=dev, dtype=matprec)
     m.A = A
     n = A.n_rows
@@ -828,37 +797,6 @@ def AMGX_matrix_set_boundary_separation(m: _MatrixHandle, flag: int):
     return RC_OK
 
 
-def _fold_external_diag(n, b, ro, ci, va, diag):
-    """Merge the DIAG-property external diagonal into the CSR structure
-    (reference block-DIA-CSR, include/matrix.h:24-26). The gfx950 kernels
-    take one folded CSR — the external layout is an input format here."""
-    deg = np.diff(ro)
-    rows = np.repeat(np.arange(n, dtype=np.int64), deg)
-    all_rows = np.concatenate([rows, np.arange(n, dtype=np.int64)])
-    all_cols = np.concatenate([ci.astype(np.int64),
-                               np.arange(n, dtype=np.int64)])
-    vab = va.reshape(len(ci), -1)
-    db = diag.reshape(n, -1)
-    all_vals = np.concatenate([vab, db], axis=0)
-    order = np.lexsort((all_cols, all_rows))
-    ro2 = np.zeros(n + 1, dtype=np.int64)
-    np.cumsum(np.bincount(all_rows, minlength=n), out=ro2[1:])
-    cols2 = all_cols[order]
-    vals2 = all_vals[order]
-    # sum duplicates (a CSR that already stores an explicit diagonal)
-    ncols = int(max(cols2.max() + 1 if cols2.size else 1, n))
-    key = all_rows[order] * ncols + cols2
-    uniq, first = np.unique(key, return_index=True)
-    if uniq.size != key.size:
-        sums = np.add.reduceat(vals2, first, axis=0)
-        cols2 = cols2[first]
-        vals2 = sums
-        counts = np.bincount((uniq // ncols).astype(np.int64), minlength=n)
-        ro2 = np.zeros(n + 1, dtype=np.int64)
-        np.cumsum(counts, out=ro2[1:])
-    return ro2, cols2, vals2
-
-
 @_amgx_try
 def AMGX_matrix_upload_all_global_32(m, n_global, n, nnz, block_dimx,
                                      block_dimy, row_ptrs,
@@ -1000,10 +938,10 @@ def AMGX_read_system_distributed(m: _MatrixHandle, rhs: _VectorHandle,
     import torch.distributed as tdist
     from .distributed.manager import DistributedManager
     from .io.matrix_market import read_system
-    from .io.binary import is_binary_file, read_system_binary
+    from .io.binary import is_binary_file, read_system_any
     mem, vecprec, matprec = _parse_mode(m.mode)
     dev = mem if mem == "cpu" else m.res.device
-    reader = read_system_binary if is_binary_file(path) else read_system
+    reader = read_system_any if is_binary_file(path) else read_system
     A, b, x0 = reader(path, device="cpu", dtype=matprec)
     world = tdist.get_world_size() if tdist.is_initialized() else 1
     rank = tdist.get_rank() if tdist.is_initialized() else 0

@@ -294,7 +294,7 @@ std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
 
 // ---------------------------------------------------------------- aggregation
 Tensor size2_match(Tensor ro, Tensor ci, Tensor va, Tensor tidx, Tensor diag,
-                   int64_t n, int64_t max_iters) {
+                   int64_t n, int64_t max_iters, int64_t seed) {
     auto agg = torch::full({n}, -1, ro.options().dtype(torch::kInt32));
     auto prop = torch::empty({n}, ro.options().dtype(torch::kInt32));
     auto changed = torch::zeros({1}, ro.options().dtype(torch::kInt32));
@@ -305,7 +305,7 @@ Tensor size2_match(Tensor ro, Tensor ci, Tensor va, Tensor tidx, Tensor diag,
                 ro.data_ptr<int>(), ci.data_ptr<int>(),
                 va.data_ptr<scalar_t>(), tidx.data_ptr<int>(),
                 diag.data_ptr<scalar_t>(), (int)n, agg.data_ptr<int>(),
-                prop.data_ptr<int>(), st);
+                prop.data_ptr<int>(), (int)seed, st);
             changed.zero_();
             amgx_hip::agg_match(prop.data_ptr<int>(), (int)n,
                                 agg.data_ptr<int>(), changed.data_ptr<int>(),

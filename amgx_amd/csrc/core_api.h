@@ -87,7 +87,7 @@ void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
 // ---- aggregation (setup.hip) ------------------------------------------------
 template <typename T>
 void agg_propose(const int* ro, const int* ci, const T* va, const int* tidx,
-                 const T* diag, int n, const int* agg, int* prop,
+                 const T* diag, int n, const int* agg, int* prop, int seed,
                  hipStream_t s);
 void agg_match(const int* prop, int n, int* agg, int* changed, hipStream_t s);
 template <typename T>

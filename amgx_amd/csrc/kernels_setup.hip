@@ -105,19 +105,29 @@ __global__ __launch_bounds__(AMGX_BLOCK) void agg_propose_kernel(const int* __re
                                    const int* __restrict__ tidx,
                                    const T* __restrict__ diag, int n,
                                    const int* __restrict__ agg,
-                                   int* __restrict__ prop) {
+                                   int* __restrict__ prop, int seed) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     if (agg[i] >= 0) { prop[i] = -1; return; }
     int best = -1;
     double bw = 0.0;
+    unsigned int bt = 0;
     for (int k = ro[i]; k < ro[i + 1]; ++k) {
         int j = ci[k];
         if (j == i || j >= n || agg[j] >= 0) continue;
         double w = edge_weight(va, tidx, diag, k, i, j);
-        // deterministic tie-break on larger j (matches host reference's
-        // stable preference order closely enough for structure)
-        if (w > bw || (w == bw && best >= 0 && j > best)) { bw = w; best = j; }
+        // Weight ties broken by a seeded per-node hash key (mirrors the host
+        // path's random tie[] and the reference's hashed weights,
+        // src/aggregation/selectors/size2_selector.cu) — a pure row-id
+        // tie-break makes every node on a uniform-weight matrix (Poisson!)
+        // propose to its highest-numbered neighbor and handshakes stall.
+        // Final fallback on row id keeps the result fully deterministic.
+        unsigned int tj = hash_u32((unsigned)j, (unsigned)seed);
+        if (w > bw ||
+            (w == bw && best >= 0 &&
+             (tj > bt || (tj == bt && j > best)))) {
+            bw = w; best = j; bt = tj;
+        }
     }
     prop[i] = best;
 }
@@ -162,11 +172,11 @@ __global__ __launch_bounds__(AMGX_BLOCK) void agg_singleton_kernel(const int* __
 
 template <typename T>
 void agg_propose(const int* ro, const int* ci, const T* va, const int* tidx,
-                 const T* diag, int n, const int* agg, int* prop,
+                 const T* diag, int n, const int* agg, int* prop, int seed,
                  hipStream_t s) {
     hipLaunchKernelGGL((agg_propose_kernel<T>), dim3(grid_1d(n)),
                        dim3(AMGX_BLOCK), 0, s, ro, ci, va, tidx, diag, n, agg,
-                       prop);
+                       prop, seed);
 }
 
 void agg_match(const int* prop, int n, int* agg, int* changed, hipStream_t s) {
@@ -568,7 +578,7 @@ void transpose_csr(const int* ro, const int* ci, const T* va, int m, int n,
 #define INSTANTIATE_SETUP(T)                                                   \
     template void agg_propose<T>(const int*, const int*, const T*,             \
                                  const int*, const T*, int, const int*, int*,  \
-                                 hipStream_t);                                 \
+                                 int, hipStream_t);                            \
     template void agg_merge_singletons<T>(const int*, const int*, const T*,    \
                                           const int*, const T*, int,           \
                                           const int*, int*, hipStream_t);      \

@@ -446,10 +446,16 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
     per-row counts (sizes known from the maps), then column/value payloads.
 
     indptr/indices/data: the LOCAL sparse rows being shared (numpy, global
-    column ids). Returns (counts, cols, vals) per halo slot: list over halo
-    positions 0..n_halo of (cols, vals) arrays.
+    column ids). `data` may be 1-D (scalar entries) or 2-D of shape
+    (nnz, width) — block matrices ship `width = block_dim^2` values per
+    entry. Returns (counts, cols, vals) per halo slot: list over halo
+    positions 0..n_halo of (cols, vals) arrays; vals keeps data's shape
+    convention ((count,) for 1-D input, (count, width) for 2-D).
     """
     indptr = np.asarray(indptr, dtype=np.int64)
+    data = np.asarray(data, dtype=np.float64)
+    width = 1 if data.ndim == 1 else int(data.shape[1])
+    data2 = data.reshape(-1, width)
     counts_all = np.diff(indptr)
     # payload tensors must live where the comm backend wants them (NCCL =
     # device buffers over xGMI; gloo = host)
@@ -484,15 +490,14 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
             np.zeros(0, dtype=np.int64)
         scol = torch.from_numpy(np.asarray(indices, dtype=np.int64)[nz]) \
             .to(comm_dev)
-        sval = torch.from_numpy(np.asarray(data, dtype=np.float64)[nz]) \
-            .to(comm_dev)
+        sval = torch.from_numpy(data2[nz].reshape(-1).copy()).to(comm_dev)
         keep += [scol, sval]
         if scol.numel():
             p2p.append(dist.P2POp(dist.isend, scol, r))
             p2p.append(dist.P2POp(dist.isend, sval, r))
         tot = int(recv_counts[i].sum())
         rcol = torch.empty(tot, dtype=torch.int64, device=comm_dev)
-        rval = torch.empty(tot, dtype=torch.float64, device=comm_dev)
+        rval = torch.empty(tot * width, dtype=torch.float64, device=comm_dev)
         recv_payload.append((rcol, rval))
         if tot:
             p2p.append(dist.P2POp(dist.irecv, rcol, r))
@@ -506,11 +511,14 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
         lo, hi = mgr.halo_slices[i]
         rc = recv_counts[i].cpu().numpy()
         rcol, rval = recv_payload[i]
-        rcol, rval = rcol.cpu().numpy(), rval.cpu().numpy()
+        rcol = rcol.cpu().numpy()
+        rval = rval.cpu().numpy().reshape(-1, width)
         pos = 0
         for k in range(hi - lo):
             c = int(rc[k])
-            out[lo + k] = (rcol[pos:pos + c], rval[pos:pos + c])
+            v = rval[pos:pos + c]
+            out[lo + k] = (rcol[pos:pos + c],
+                           v.reshape(-1) if data.ndim == 1 else v)
             pos += c
     return out
 
@@ -532,7 +540,7 @@ def halo_matrix(mgr: DistributedManager, A):
     if mgr.n_halo:
         gcol[mgr.n_local:] = mgr.halo_global
     rows = exchange_csr_rows(mgr, ro, gcol[ci],
-                             va[:, 0] if va.shape[1] == 1 else va[:, 0])
+                             va[:, 0] if va.shape[1] == 1 else va)
     out_ro = np.zeros(mgr.n_halo + 1, dtype=np.int64)
     cols_l, vals_l = [], []
     for k, rowdat in enumerate(rows):

@@ -188,10 +188,12 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
         va = torch.linalg.vector_norm(
             A.values.reshape(A.nnz, -1).to(torch.float64), dim=1).to(A.dtype)
         roots = _core.size2_match(A.row_offsets, A.col_indices, va, _tidx(A),
-                                  diag.to(A.dtype), A.n_rows, max_iterations)
+                                  diag.to(A.dtype), A.n_rows, max_iterations,
+                                  int(seed) + 0x9E3779B1)
     else:
         roots = _core.size2_match(A.row_offsets, A.col_indices, A.values,
-                                  _tidx(A), diag, A.n_rows, max_iterations)
+                                  _tidx(A), diag, A.n_rows, max_iterations,
+                                  int(seed) + 0x9E3779B1)
     uniq, agg = torch.unique(roots, sorted=True, return_inverse=True)
     return agg.to(torch.int32), int(uniq.numel())
 

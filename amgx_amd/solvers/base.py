@@ -173,6 +173,14 @@ class Solver:
                 per = comp.sum(0)
                 if mgr is not None:
                     tdist.all_reduce(per)
+            elif self.norm == "L1_SCALED":
+                # per-component L1 divided by the per-component global length
+                per = comp.sum(0)
+                if mgr is not None:
+                    tdist.all_reduce(per)
+                    per = per / max(mgr.n_global, 1)
+                else:
+                    per = per / max(comp.shape[0], 1)
             elif self.norm == "LMAX":
                 per = comp.amax(0) if comp.numel() else \
                     torch.zeros(bd, dtype=torch.float64, device=r.device)
