@@ -7,8 +7,8 @@ set -x
 cd /tmp && export TMPDIR=/tmp && cd "$GRAFT_REPO_ROOT" || cd /root/repo
 mkdir -p gpurun_out/prof
 
-# 0. correctness first: GPU suite + smoke
-timeout 900 python -m pytest tests -m gpu -x -q > gpurun_out/prof/pytest_gpu.log 2>&1
+# 0. correctness first: GPU suite + smoke (no -x: show ALL failures)
+timeout 900 python -m pytest tests -m gpu -q > gpurun_out/prof/pytest_gpu.log 2>&1
 timeout 300 python -c "import __graft_entry__ as g; g.smoke()" \
     > gpurun_out/prof/smoke.log 2>&1
 
