@@ -736,6 +736,13 @@ _ERROR_STRINGS = {
 }
 
 
+def _c_abi_world_size():
+    """Helper for the C ABI shim (csrc_capi/amgx_c_shim.cpp): world size of
+    the live torch.distributed group, 1 when single-process."""
+    import torch.distributed as tdist
+    return tdist.get_world_size() if tdist.is_initialized() else 1
+
+
 def AMGX_get_error_string(rc: int):
     """Reference AMGX_get_error_string (include/amgx_c.h)."""
     return _ERROR_STRINGS.get(rc, "Unknown error code")
