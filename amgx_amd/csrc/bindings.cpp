@@ -101,6 +101,24 @@ void axpy(Tensor y, Tensor x, double a) {
     });
 }
 
+void axpy_dalpha(Tensor y, Tensor x, Tensor alpha, double scale) {
+    DISPATCH_FT(x, "axpy_dalpha", [&] {
+        amgx_hip::axpy_dalpha<scalar_t>(y.data_ptr<scalar_t>(),
+                                        x.data_ptr<scalar_t>(),
+                                        alpha.data_ptr<scalar_t>(),
+                                        (scalar_t)scale, x.numel(),
+                                        cur_stream());
+    });
+}
+
+void scal_drsqrt(Tensor x, Tensor s2) {
+    DISPATCH_FT(x, "scal_drsqrt", [&] {
+        amgx_hip::scal_drsqrt<scalar_t>(x.data_ptr<scalar_t>(),
+                                        s2.data_ptr<scalar_t>(), x.numel(),
+                                        cur_stream());
+    });
+}
+
 void axpby(Tensor y, Tensor x, double a, double b) {
     DISPATCH_FT(x, "axpby", [&] {
         amgx_hip::axpby<scalar_t>(y.data_ptr<scalar_t>(),
@@ -606,6 +624,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("csrmv", &csrmv);
     m.def("reduce_op", &reduce_op);
     m.def("axpy", &axpy);
+    m.def("axpy_dalpha", &axpy_dalpha);
+    m.def("scal_drsqrt", &scal_drsqrt);
     m.def("axpby", &axpby);
     m.def("scal", &scal);
     m.def("diag_index", &diag_index);

@@ -35,6 +35,11 @@ void reduce(const T* x, const T* y, long long n, int op, T* ws, T* out,
 template <typename T>
 void axpy(T* y, const T* x, T a, long long n, hipStream_t s);
 template <typename T>
+void axpy_dalpha(T* y, const T* x, const T* alpha, T scale, long long n,
+                 hipStream_t s);
+template <typename T>
+void scal_drsqrt(T* x, const T* s2, long long n, hipStream_t s);
+template <typename T>
 void axpby(T* y, const T* x, T a, T b, long long n, hipStream_t s);
 template <typename T>
 void scal(T* x, T a, long long n, hipStream_t s);

@@ -111,6 +111,20 @@ def dot_async(x, y):
     return _core.reduce_op(x.reshape(-1), y.reshape(-1), 0)
 
 
+def axpy_dalpha(y, x, alpha, scale=1.0):
+    """y += scale * alpha[0] * x with device-resident alpha (no host
+    sync; Krylov MGS projections)."""
+    _core.axpy_dalpha(y.reshape(-1), x.reshape(-1), alpha.reshape(-1),
+                      float(scale))
+    return y
+
+
+def scal_drsqrt(x, s2):
+    """x *= rsqrt(s2[0]) with device-resident s2 (no-op when s2<=0)."""
+    _core.scal_drsqrt(x.reshape(-1), s2.reshape(-1))
+    return x
+
+
 def axpy(y, x, alpha):
     _core.axpy(y.reshape(-1), x.reshape(-1), float(alpha))
     return y

@@ -212,6 +212,45 @@ class FGMRESSolver(Solver):
         self.g[0] = beta
         self._rnorm = beta
 
+    # -- device fast path helpers (<=1 host sync per iteration) -----------
+    def _dot_dev(self, x, y):
+        """Device-scalar (possibly all-reduced) dot; no host sync."""
+        from ..ops import gpu as G
+        mgr = getattr(self.A, "manager", None)
+        if mgr is not None:
+            x = x.reshape(-1)[:mgr.owned_size]
+            y = y.reshape(-1)[:mgr.owned_size]
+        h = G.dot_async(x, y)
+        if mgr is not None:
+            import torch.distributed as tdist
+            tdist.all_reduce(h)
+        return h
+
+    def _iterate_device(self, j, w):
+        """MGS projections with device-resident coefficients: all h_ij stay
+        on the GPU (all-reduced on stream when distributed), w is normalized
+        on device, and the iteration's ONE host sync reads the whole
+        Hessenberg column afterwards (reference pipelined-dot intent,
+        src/solvers/fgmres_solver.cu)."""
+        from ..ops import gpu as G
+        i0 = max(0, j + 1 - self.krylov_dim) if self.krylov_dim > 0 else 0
+        if not hasattr(self, "_hbuf") or self._hbuf.numel() < self.restart + 2 \
+                or self._hbuf.dtype != w.dtype:
+            self._hbuf = torch.zeros(self.restart + 2, dtype=w.dtype,
+                                     device=w.device)
+        hbuf = self._hbuf
+        for i in range(i0, j + 1):
+            hi = self._dot_dev(self.V[i], w)
+            hbuf[i] = hi
+            G.axpy_dalpha(w, self.V[i], hi, -1.0)
+        ww = self._dot_dev(w, w)
+        hbuf[j + 1] = ww
+        G.scal_drsqrt(w, ww)              # w /= ||w|| (no-op on breakdown)
+        hcol = hbuf[i0:j + 2].double().cpu()   # the ONE sync
+        for i in range(i0, j + 1):
+            self.H[i, j] = float(hcol[i - i0])
+        return math.sqrt(max(float(hcol[j + 1 - i0]), 0.0)), True
+
     def solve_iteration(self, b, x):
         j = self.j
         m = self.restart
@@ -221,15 +260,21 @@ class FGMRESSolver(Solver):
         self.Z.append(z)
         w = self.new_vec(vj)
         ops.spmv(self.A, z, w)
-        # modified Gram-Schmidt: h_ij = <v_i, w> (conjugated in the first
-        # arg); truncated to the last krylov_dim basis vectors when set
-        i0 = max(0, j + 1 - self.krylov_dim) if self.krylov_dim > 0 else 0
-        for i in range(i0, j + 1):
-            hij = self.dot(self.V[i], w)
-            self.H[i, j] = hij
-            ops.axpy(w, self.V[i], -hij)
-        hnext = math.sqrt(abs(self.dot(w, w)))
-        self.H[j + 1, j] = hnext
+        fast = (torch.is_tensor(w) and w.is_cuda and not w.is_complex())
+        if fast:
+            hnext, w_normalized = self._iterate_device(j, w)
+            self.H[j + 1, j] = hnext
+        else:
+            # modified Gram-Schmidt: h_ij = <v_i, w> (conjugated in the
+            # first arg); truncated to the last krylov_dim vectors when set
+            w_normalized = False
+            i0 = max(0, j + 1 - self.krylov_dim) if self.krylov_dim > 0 else 0
+            for i in range(i0, j + 1):
+                hij = self.dot(self.V[i], w)
+                self.H[i, j] = hij
+                ops.axpy(w, self.V[i], -hij)
+            hnext = math.sqrt(abs(self.dot(w, w)))
+            self.H[j + 1, j] = hnext
         # apply stored Givens rotations to column j
         for i in range(j):
             ci, si = float(self.cs[i]), self._sc(self.sn[i])
@@ -266,7 +311,7 @@ class FGMRESSolver(Solver):
             if not (converged or lucky):
                 self._restart_init(b, x)   # restart
             return converged or lucky
-        self.V.append(w / hnext)
+        self.V.append(w if w_normalized else w / hnext)
         self.j += 1
         return False
 
