@@ -101,6 +101,14 @@ void axpy(Tensor y, Tensor x, double a) {
     });
 }
 
+Tensor mfma4_probe(Tensor a_frag, Tensor b_frag) {
+    auto c = torch::empty_like(a_frag);
+    amgx_hip::mfma4_probe(a_frag.data_ptr<double>(),
+                          b_frag.data_ptr<double>(), c.data_ptr<double>(),
+                          cur_stream());
+    return c;
+}
+
 void axpy_dalpha(Tensor y, Tensor x, Tensor alpha, double scale) {
     DISPATCH_FT(x, "axpy_dalpha", [&] {
         amgx_hip::axpy_dalpha<scalar_t>(y.data_ptr<scalar_t>(),
@@ -625,6 +633,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("reduce_op", &reduce_op);
     m.def("axpy", &axpy);
     m.def("axpy_dalpha", &axpy_dalpha);
+    m.def("mfma4_probe", &mfma4_probe);
     m.def("scal_drsqrt", &scal_drsqrt);
     m.def("axpby", &axpby);
     m.def("scal", &scal);

@@ -24,6 +24,28 @@ void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
            TV* y, const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            hipStream_t s);
 
+// ---- MFMA wave-structured block-4 kernels (kernels_mfma.hip) ----------------
+// v_mfma_f64_4x4x4_4b_f64 path: wave = 4 rows x 16 lanes, coalesced block
+// loads; used automatically by the b==4 dispatch of bsrmv/dilu_* above.
+void mfma4_probe(const double* a, const double* b, double* c, hipStream_t s);
+template <typename TA, typename TV>
+void bsrmv_b4(const int* ro, const int* ci, const TA* va, const TV* x, TV* y,
+              const TV* bvec, double alpha, double beta, double gamma,
+              int row_begin, int row_end, hipStream_t s);
+template <typename TA, typename TV>
+void dilu_fwd_b4(const int* ro, const int* ci, const TA* va, const TA* einv,
+                 const int* rows, int count, const TV* r, TV* w,
+                 hipStream_t s);
+template <typename TA, typename TV>
+void dilu_bwd_b4(const int* ro, const int* ci, const TA* va, const TA* einv,
+                 const int* rows, int count, const TV* w, TV* z,
+                 hipStream_t s);
+template <typename TA>
+void dilu_setup_b4(const int* ro, const int* ci, const TA* va,
+                   const int* didx, const int* tidx, const int* colors,
+                   const int* rows, int count, int color, TA* einv,
+                   hipStream_t s);
+
 // ---- BLAS-1 (blas.hip) ------------------------------------------------------
 // op: 0 = dot(x,y), 1 = sum|x| (L1), 2 = max|x| (Lmax). Deterministic
 // two-stage reduction; out is a device scalar, ws a device scratch of

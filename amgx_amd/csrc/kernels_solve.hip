@@ -116,6 +116,11 @@ void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
            hipStream_t s) {
     long long rows = ((long long)r1 - r0) * b;
     if (rows <= 0) return;
+    if (b == 4) {   // MFMA wave kernel (kernels_mfma.hip)
+        bsrmv_b4<TA, TV>(ro, ci, va, x, y, bvec, (double)alpha, (double)beta,
+                         (double)gamma, r0, r1, s);
+        return;
+    }
     hipLaunchKernelGGL((bsrmv_kernel<TA, TV>), dim3(grid_1d(rows)),
                        dim3(AMGX_BLOCK), 0, s, ro, ci, va, b, x, y, bvec,
                        alpha, beta, gamma, r0, r1);
@@ -630,6 +635,9 @@ void dilu_setup_color(const int* ro, const int* ci, const T* va,
         hipLaunchKernelGGL((dilu_setup_scalar<T>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, didx, tidx,
                            colors, rows, count, color, einv);
+    } else if (b == 4) {   // MFMA triple-product path (kernels_mfma.hip)
+        dilu_setup_b4<T>(ro, ci, va, didx, tidx, colors, rows, count, color,
+                         einv, s);
     } else if (b <= 8) {
         hipLaunchKernelGGL((dilu_setup_block<T, 8>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, didx, tidx,
@@ -746,6 +754,8 @@ void dilu_fwd_color(const int* ro, const int* ci, const TA* va,
         hipLaunchKernelGGL((dilu_fwd_scalar<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
                            count, r, w);
+    else if (b == 4)       // MFMA wave kernel (kernels_mfma.hip)
+        dilu_fwd_b4<TA, TV>(ro, ci, va, einv, rows, count, r, w, s);
     else
         hipLaunchKernelGGL((dilu_fwd_block<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
@@ -762,6 +772,8 @@ void dilu_bwd_color(const int* ro, const int* ci, const TA* va,
         hipLaunchKernelGGL((dilu_bwd_scalar<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,
                            count, w, z);
+    else if (b == 4)       // MFMA wave kernel (kernels_mfma.hip)
+        dilu_bwd_b4<TA, TV>(ro, ci, va, einv, rows, count, w, z, s);
     else
         hipLaunchKernelGGL((dilu_bwd_block<TA, TV>), dim3(grid_1d(count)),
                            dim3(AMGX_BLOCK), 0, s, ro, ci, va, einv, rows,

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(csrc, "kernels_solve.hip"),
         os.path.join(csrc, "kernels_setup.hip"),
         os.path.join(csrc, "kernels_classical.hip"),
+        os.path.join(csrc, "kernels_mfma.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
