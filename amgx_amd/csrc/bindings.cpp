@@ -101,6 +101,51 @@ void axpy(Tensor y, Tensor x, double a) {
     });
 }
 
+// LDS-hash SpGEMM / Galerkin (kernels_spgemm.hip).  Returns
+// (roC, ciC, vaC, big_rows): big_rows lists rows written unsorted (the
+// Python caller sorts those).  Empty ciC + roC[-1]==-1 signals "fall back
+// to the ESC path" (a row exceeded the big hash capacity).
+std::vector<Tensor> spgemm_hash(Tensor roA, Tensor ciA, Tensor vaA,
+                                Tensor roB, Tensor ciB, Tensor vaB,
+                                c10::optional<Tensor> aggcol, int64_t mode,
+                                int64_t cap_nnz) {
+    int m = (int)roA.numel() - 1;
+    auto roC = torch::empty({m + 1}, roA.options());
+    auto ciC = torch::empty({cap_nnz}, roA.options());
+    auto vaC = torch::empty({cap_nnz}, vaA.options());
+    auto big = torch::empty({0}, roA.options());
+    hipStream_t st = cur_stream();
+    long long nnz = -1;
+    DISPATCH_FT(vaA, "spgemm_hash", [&] {
+        int* big_rows = nullptr;
+        int n_big = 0;
+        nnz = amgx_hip::spgemm_hash<scalar_t>(
+            roA.data_ptr<int>(), ciA.data_ptr<int>(),
+            vaA.data_ptr<scalar_t>(), m, roB.data_ptr<int>(),
+            ciB.data_ptr<int>(), vaB.data_ptr<scalar_t>(),
+            aggcol ? aggcol->data_ptr<int>() : nullptr, (int)mode,
+            roC.data_ptr<int>(), ciC.data_ptr<int>(),
+            vaC.data_ptr<scalar_t>(), (long long)cap_nnz, &big_rows,
+            &n_big, st);
+        if (n_big > 0) {
+            big = torch::empty({n_big}, roA.options());
+            TORCH_CHECK(hipMemcpyAsync(big.data_ptr<int>(), big_rows,
+                                       n_big * sizeof(int),
+                                       hipMemcpyDeviceToDevice,
+                                       st) == hipSuccess,
+                        "spgemm_hash: big_rows copy failed");
+        }
+        if (big_rows) amgx_hip::free_device_buf(big_rows, st);
+    });
+    TORCH_CHECK(nnz != -2, "spgemm_hash: cap_nnz too small (internal)");
+    if (nnz < 0) {        // ESC fallback signal
+        roC.fill_(-1);
+        return {roC, torch::empty({0}, roA.options()),
+                torch::empty({0}, vaA.options()), big};
+    }
+    return {roC, ciC.narrow(0, 0, nnz), vaC.narrow(0, 0, nnz), big};
+}
+
 Tensor mfma4_probe(Tensor a_frag, Tensor b_frag) {
     auto c = torch::empty_like(a_frag);
     amgx_hip::mfma4_probe(a_frag.data_ptr<double>(),
@@ -634,6 +679,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("axpy", &axpy);
     m.def("axpy_dalpha", &axpy_dalpha);
     m.def("mfma4_probe", &mfma4_probe);
+    m.def("spgemm_hash", &spgemm_hash);
     m.def("scal_drsqrt", &scal_drsqrt);
     m.def("axpby", &axpby);
     m.def("scal", &scal);

@@ -24,6 +24,21 @@ void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
            TV* y, const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            hipStream_t s);
 
+// ---- LDS-hash SpGEMM (kernels_spgemm.hip) -----------------------------------
+// mode 0: C = A*B.  mode 1: aggregation Galerkin — A = aggregate-membership
+// CSR (coarse row -> fine rows), B = fine matrix, key = aggcol[col].
+// Returns nnz(C), or -1 when a row exceeds the big-table capacity (caller
+// falls back to the ESC sort path), or -2 when cap_nnz is too small.
+// big_rows_out (device, caller frees via free_device_buf) lists rows whose
+// entries were written UNSORTED (caller sorts those columns).
+template <typename T>
+long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
+                      const int* roB, const int* ciB, const T* vaB,
+                      const int* aggcol, int mode, int* roC_out,
+                      int* ciC_cap_buf, T* vaC_cap_buf, long long cap_nnz,
+                      int** big_rows_out, int* n_big_out, hipStream_t s);
+void free_device_buf(void* p, hipStream_t s);
+
 // ---- MFMA wave-structured block-4 kernels (kernels_mfma.hip) ----------------
 // v_mfma_f64_4x4x4_4b_f64 path: wave = 4 rows x 16 lanes, coalesced block
 // loads; used automatically by the b==4 dispatch of bsrmv/dilu_* above.

@@ -215,12 +215,34 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
 def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
                          ncols_mod=None):
     """agg_col/ncols_mod: distributed variant — per-column coarse ids (GLOBAL)
-    and the global coarse column count; defaults to the single-process case."""
+    and the global coarse column count; defaults to the single-process case.
+
+    Scalar matrices run the LDS-hash LOW_DEG-style generator
+    (kernels_spgemm.hip mode 1: wave-per-coarse-row hash accumulation —
+    reference src/aggregation/coarseAgenerators/low_deg_…); block matrices
+    and hash-capacity overflows use the one-sort rocPRIM generator."""
     from ..matrix import CSRMatrix
     if agg_col is None:
         agg_col = aggregates
     if ncols_mod is None:
         ncols_mod = num_aggregates
+    if A.block_dim == 1:
+        # aggregate-membership CSR: coarse row -> fine member rows
+        agg64 = aggregates.to(torch.int64)
+        counts = torch.bincount(agg64, minlength=num_aggregates)
+        m_ro = torch.zeros(num_aggregates + 1, dtype=torch.int64,
+                           device=A.device)
+        m_ro[1:] = torch.cumsum(counts, 0)
+        fids = torch.argsort(agg64, stable=True).to(torch.int32)
+        ro, ci, va, big = _core.spgemm_hash(
+            m_ro.to(torch.int32), fids, A.values,   # vaA unused in mode 1
+            A.row_offsets, A.col_indices, A.values,
+            agg_col, 1, max(A.nnz, 1))
+        if int(ro[0].item()) != -1:
+            if big.numel():
+                _sort_unsorted_rows(ro, ci, va, big)
+            return CSRMatrix(ro, ci.contiguous(), va.contiguous(),
+                             n_cols=int(ncols_mod), block_dim=1)
     ro_c, ci_c, va_c = _core.galerkin_agg(A.row_offsets, A.col_indices,
                                           A.values, aggregates, agg_col,
                                           num_aggregates, int(ncols_mod),
@@ -253,12 +275,35 @@ def _expansion_bound(A, B):
     return int(degB[A.col_indices.to(torch.int64)].sum().item())
 
 
+def _sort_unsorted_rows(ro, ci, va, big_rows):
+    """Column-sort the (rare) rows the big-capacity hash kernel wrote
+    unsorted (>512 nnz per row)."""
+    ro64 = ro.to(torch.int64)
+    for r in big_rows.tolist():
+        s, e = int(ro64[r]), int(ro64[r + 1])
+        order = torch.argsort(ci[s:e])
+        ci[s:e] = ci[s:e][order]
+        va[s:e] = va[s:e][order]
+
+
 def spgemm(A, B):
+    """C = A*B — LDS-hash kernels (kernels_spgemm.hip, role of reference
+    src/csr_multiply_detail.cu warp-hash family); ESC sort fallback for
+    pathological >8k-nnz rows."""
     from ..matrix import CSRMatrix
     cap = max(_expansion_bound(A, B), 1)
-    ro, ci, va = _core.spgemm(A.row_offsets, A.col_indices, A.values,
-                              B.row_offsets, B.col_indices, B.values,
-                              B.n_cols, cap)
+    ro, ci, va, big = _core.spgemm_hash(A.row_offsets, A.col_indices,
+                                        A.values, B.row_offsets,
+                                        B.col_indices, B.values, None, 0,
+                                        cap)
+    if int(ro[0].item()) == -1:   # a row exceeded the big hash capacity
+        ro, ci, va = _core.spgemm(A.row_offsets, A.col_indices, A.values,
+                                  B.row_offsets, B.col_indices, B.values,
+                                  B.n_cols, cap)
+        return CSRMatrix(ro, ci.contiguous(), va.contiguous(),
+                         n_cols=B.n_cols)
+    if big.numel():
+        _sort_unsorted_rows(ro, ci, va, big)
     return CSRMatrix(ro, ci.contiguous(), va.contiguous(), n_cols=B.n_cols)
 
 

@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(csrc, "kernels_setup.hip"),
         os.path.join(csrc, "kernels_classical.hip"),
         os.path.join(csrc, "kernels_mfma.hip"),
+        os.path.join(csrc, "kernels_spgemm.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
