@@ -274,8 +274,107 @@ def interp_d1(A, S, cf_map, num_coarse, scope):
     return ops._backend(A).interp_d1(A, S, cf_map, num_coarse)
 
 
+def _interp_d2_device(A, S, cf_map, num_coarse, scope):
+    """Device-resident distance-2 interpolation: the same magnitude-
+    proportional formulation as the host path below (acc = D + F @ W, then
+    row scaling), expressed in torch device ops + the LDS-hash spgemm — no
+    .cpu() round trip mid-setup (reference all-device D2,
+    src/classical/interpolators/distance2.cu)."""
+    from ..matrix import CSRMatrix
+    dev = A.values.device
+    n = A.n_rows
+    nc = int(num_coarse)
+    ro64 = A.row_offsets.to(torch.int64)
+    ci = A.col_indices.to(torch.int64)
+    v = A.values.reshape(A.nnz, -1)[:, 0].to(torch.float64)
+    strong = S.to(torch.bool).reshape(-1)
+    cf = cf_map.to(torch.int64)
+    deg = ro64[1:] - ro64[:-1]
+    rows = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int64, device=dev), deg)
+    local = ci < n
+    cf_col = torch.where(local, cf[torch.clamp(ci, max=cf.numel() - 1)],
+                         torch.tensor(-1, dtype=torch.int64, device=dev))
+    offd = rows != ci
+    strongC = strong & offd & (cf_col >= 0)
+    strongF = strong & offd & local & (cf_col < 0)
+    absv = v.abs()
+    sC_sum = torch.zeros(n, dtype=torch.float64, device=dev)
+    sC_sum.index_add_(0, rows[strongC], absv[strongC])
+    alive = sC_sum > 0.0
+    ci_n = torch.clamp(ci, max=n - 1)
+
+    def _csr(mask, cols, vals, ncols):
+        cnt = torch.bincount(rows[mask], minlength=n)
+        ro_ = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+        ro_[1:] = torch.cumsum(cnt, 0).to(torch.int32)
+        return CSRMatrix(ro_, cols[mask].to(torch.int32).contiguous(),
+                         vals[mask].contiguous(), n_cols=ncols)
+
+    wr = rows[strongC]
+    W = _csr(strongC, cf_col, absv / sC_sum.clamp(min=1e-300)[
+        torch.clamp(rows, max=n - 1)], nc)
+    live_edge = strongF & alive[ci_n] & local
+    F = _csr(live_edge, ci, v, n)
+    FW = ops._backend(A).spgemm(F, W)
+    # acc = D + FW  (merge by sorted (row, col) key)
+    fw_deg = FW.row_offsets.to(torch.int64)
+    fw_rows = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int64, device=dev),
+        fw_deg[1:] - fw_deg[:-1])
+    all_r = torch.cat([rows[strongC], fw_rows])
+    all_c = torch.cat([cf_col[strongC], FW.col_indices.to(torch.int64)])
+    all_v = torch.cat([v[strongC], FW.values.reshape(-1).to(torch.float64)])
+    key = all_r * nc + all_c
+    uk, inv = torch.unique(key, return_inverse=True)
+    vsum = torch.zeros(uk.numel(), dtype=torch.float64, device=dev)
+    vsum.index_add_(0, inv, all_v)
+    acc_r = uk // nc
+    acc_c = uk % nc
+    # denominators: diag + weak couplings + dead-end strong-F couplings
+    diag = torch.zeros(n, dtype=torch.float64, device=dev)
+    dmask = rows == ci
+    diag[rows[dmask]] = v[dmask]
+    weak = offd & ~strong
+    dead = strongF & ~alive[ci_n]
+    lump = torch.zeros(n, dtype=torch.float64, device=dev)
+    lump.index_add_(0, rows[weak], v[weak])
+    lump.index_add_(0, rows[dead], v[dead])
+    denom = diag + lump
+    # row nnz of acc
+    acc_cnt = torch.bincount(acc_r, minlength=n)
+    f_ok = (cf[:n] < 0) & (denom != 0.0) & (acc_cnt > 0)
+    scale = torch.where(
+        f_ok, -1.0 / torch.where(denom != 0.0, denom,
+                                 torch.ones_like(denom)),
+        torch.zeros(n, dtype=torch.float64, device=dev))
+    sv = vsum * scale[acc_r]
+    keep = sv != 0.0
+    acc_r, acc_c, sv = acc_r[keep], acc_c[keep], sv[keep]
+    # identity rows for C points, merged by key (already disjoint from acc)
+    c_rows = torch.nonzero(cf[:n] >= 0, as_tuple=True)[0]
+    all_r = torch.cat([acc_r, c_rows])
+    all_c = torch.cat([acc_c, cf[c_rows]])
+    all_v = torch.cat([sv, torch.ones(c_rows.numel(), dtype=torch.float64,
+                                      device=dev)])
+    key = all_r * nc + all_c
+    order = torch.argsort(key)
+    all_r, all_c, all_v = all_r[order], all_c[order], all_v[order]
+    cnt = torch.bincount(all_r, minlength=n)
+    p_ro = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+    p_ro[1:] = torch.cumsum(cnt, 0).to(torch.int32)
+    return CSRMatrix(p_ro, all_c.to(torch.int32).contiguous(),
+                     all_v.to(A.dtype).contiguous(), n_cols=nc)
+
+
 @_register(INTERP_REGISTRY, "D2")
 def interp_d2(A, S, cf_map, num_coarse, scope):
+    if A.values.is_cuda and A.block_dim == 1:
+        return _interp_d2_device(A, S, cf_map, num_coarse, scope)
+    return _interp_d2_host(A, S, cf_map, num_coarse, scope)
+
+
+def _interp_d2_host(A, S, cf_map, num_coarse, scope):
     """Distance-2 (standard/extended) interpolation (reference
     src/classical/interpolators/distance2.cu): an F point i interpolates
     from its strong C neighbors AND the strong C neighbors of its strong F
@@ -341,6 +440,94 @@ def interp_d2(A, S, cf_map, num_coarse, scope):
     return CSRMatrix.from_scipy(P, dtype=A.dtype)
 
 
+def _interp_multipass_device(A, S, cf_map, num_coarse, scope,
+                             max_passes: int = 10):
+    """Device-resident multipass interpolation: same pass structure as the
+    host path below (each pass = one spgemm of the strong-edges-into-done-
+    rows matrix with the current P), in torch device ops + the LDS-hash
+    spgemm — no .cpu() mid-setup (reference all-device MULTIPASS,
+    src/classical/interpolators/multipass.cu)."""
+    from ..matrix import CSRMatrix
+    dev = A.values.device
+    n = A.n_rows
+    nc = int(num_coarse)
+    ro64 = A.row_offsets.to(torch.int64)
+    ci = A.col_indices.to(torch.int64)
+    v = A.values.reshape(A.nnz, -1)[:, 0].to(torch.float64)
+    strong = S.to(torch.bool).reshape(-1)
+    cf = cf_map.to(torch.int64)
+    deg = ro64[1:] - ro64[:-1]
+    rows = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int64, device=dev), deg)
+    local = ci < n
+    offd = rows != ci
+    diag = torch.zeros(n, dtype=torch.float64, device=dev)
+    dmask = rows == ci
+    diag[rows[dmask]] = v[dmask]
+    wmask = offd & ~strong
+    weak_lump = torch.zeros(n, dtype=torch.float64, device=dev)
+    weak_lump.index_add_(0, rows[wmask], v[wmask])
+    denom = diag + weak_lump
+    sedge = strong & offd & local
+    ci_n = torch.clamp(ci, max=n - 1)
+    done = cf[:n] >= 0
+    c_rows = torch.nonzero(done, as_tuple=True)[0]
+
+    def _coo_to_csr(r_, c_, v_, ncols):
+        key = r_ * ncols + c_
+        uk, inv = torch.unique(key, return_inverse=True)
+        vs = torch.zeros(uk.numel(), dtype=torch.float64, device=dev)
+        vs.index_add_(0, inv, v_)
+        keep = vs != 0.0
+        uk, vs = uk[keep], vs[keep]
+        rr = uk // ncols
+        cnt = torch.bincount(rr, minlength=n)
+        ro_ = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+        ro_[1:] = torch.cumsum(cnt, 0).to(torch.int32)
+        return CSRMatrix(ro_, (uk % ncols).to(torch.int32).contiguous(),
+                         vs.contiguous(), n_cols=ncols)
+
+    P_r = c_rows
+    P_c = cf[c_rows]
+    P_v = torch.ones(c_rows.numel(), dtype=torch.float64, device=dev)
+    for _pass in range(max_passes):
+        undone = ~done
+        if not bool(undone.any()):
+            break
+        e = sedge & undone[rows] & done[ci_n] & local
+        if not bool(e.any()):
+            break
+        ok_rows = torch.zeros(n, dtype=torch.bool, device=dev)
+        ok_rows[rows[e]] = True
+        ok_rows &= denom != 0.0
+        if not bool(ok_rows.any()):
+            break
+        # F (undone -> done edges) and current P as device CSR
+        ecnt = torch.bincount(rows[e], minlength=n)
+        f_ro = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+        f_ro[1:] = torch.cumsum(ecnt, 0).to(torch.int32)
+        F = CSRMatrix(f_ro, ci[e].to(torch.int32).contiguous(),
+                      v[e].contiguous(), n_cols=n)
+        P = _coo_to_csr(P_r, P_c, P_v, nc)
+        contrib = ops._backend(A).spgemm(F, P)
+        scale = torch.where(
+            ok_rows, -1.0 / torch.where(denom != 0.0, denom,
+                                        torch.ones_like(denom)),
+            torch.zeros(n, dtype=torch.float64, device=dev))
+        cdeg = contrib.row_offsets.to(torch.int64)
+        c_rows2 = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=dev),
+            cdeg[1:] - cdeg[:-1])
+        newv = contrib.values.reshape(-1).to(torch.float64) * scale[c_rows2]
+        P_r = torch.cat([P_r, c_rows2])
+        P_c = torch.cat([P_c, contrib.col_indices.to(torch.int64)])
+        P_v = torch.cat([P_v, newv])
+        done = done | ok_rows
+    P = _coo_to_csr(P_r, P_c, P_v, nc)
+    return CSRMatrix(P.row_offsets, P.col_indices,
+                     P.values.to(A.dtype), n_cols=nc)
+
+
 @_register(INTERP_REGISTRY, "MULTIPASS")
 def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
     """Multipass interpolation for aggressive coarsening (reference
@@ -348,6 +535,9 @@ def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
     pass k = F points whose strong neighbors were interpolated in earlier
     passes, composing their rows. Vectorized: each pass is one SpGEMM of the
     strong-edges-into-done-rows matrix with the current P."""
+    if A.values.is_cuda and A.block_dim == 1:
+        return _interp_multipass_device(A, S, cf_map, num_coarse, scope,
+                                        max_passes)
     from ..matrix import CSRMatrix
     ro, ci, v = _csr_parts(A)
     strong = np.asarray(S.cpu().numpy() if torch.is_tensor(S) else S,

@@ -93,8 +93,7 @@ def _min_max(A, scope, level):
     if level <= 1:
         frac = scope.get("max_uncolored_percentage") if scope is not None \
             else 0.0
-        colors, num = ops.color_matrix(A, max_uncolored_frac=float(frac or 0))
-        return colors.cpu().numpy().astype(np.int32), num
+        return ops.color_matrix(A, max_uncolored_frac=float(frac or 0))
     return _min_max_rounds(*_host_adj(A, level))
 
 
@@ -133,6 +132,8 @@ def _min_max_rounds(indptr, indices):
 def _parallel_greedy(A, scope, level):
     # Jones-Plassmann with smallest-feasible-color (same algorithm the gfx950
     # kernel runs); host model is sequential over JP rounds
+    if level <= 1 and A.row_offsets.is_cuda:
+        return ops.color_matrix(A, seed=7)
     indptr, indices = _host_adj(A, level)
     n = indptr.size - 1
     colors = np.full(n, -1, dtype=np.int32)
@@ -186,6 +187,8 @@ def _greedy_recolor(A, scope, level):
     """MIN_MAX first, then one greedy pass in descending-color order that
     moves every row down to its smallest feasible color."""
     colors, num = COLORING_REGISTRY["MIN_MAX"](A, scope, level)
+    if torch.is_tensor(colors):
+        colors = colors.cpu().numpy().astype(np.int32)
     indptr, indices = _host_adj(A, level)
     order = np.argsort(-colors, kind="stable")
     for i in order:
@@ -201,6 +204,9 @@ def _greedy_recolor(A, scope, level):
 
 @register_coloring("MULTI_HASH")
 def _multi_hash(A, scope, level):
+    if level <= 1 and A.row_offsets.is_cuda:
+        max_hash = int(scope.get("max_num_hash")) if scope is not None else 7
+        return ops.color_matrix(A, seed=11, multihash_rounds=max_hash)
     indptr, indices = _host_adj(A, level)
     n = indptr.size - 1
     max_hash = int(scope.get("max_num_hash")) if scope is not None else 7

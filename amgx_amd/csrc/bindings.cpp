@@ -340,8 +340,12 @@ void dilu_apply(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor einv,
 }
 
 // ---------------------------------------------------------------- coloring
+// mode 0: greedy smallest-unused color (MIN_MAX / PARALLEL_GREEDY class);
+// mode 1: MULTI_HASH — `mode1_rounds` rounds assigning color = round id
+// with per-round re-hash, then greedy cleanup rounds for leftovers.
 std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
-                                         int64_t max_rounds, int64_t seed) {
+                                         int64_t max_rounds, int64_t seed,
+                                         int64_t mode1_rounds) {
     auto colors = torch::full({n}, -1,
                               ro.options().dtype(torch::kInt32));
     auto counter = torch::zeros({1}, ro.options().dtype(torch::kInt32));
@@ -349,9 +353,10 @@ std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
     int rounds = 0;
     for (; rounds < max_rounds; ++rounds) {
         counter.zero_();
+        int mode = rounds < mode1_rounds ? 1 : 0;
         amgx_hip::color_minmax_round(ro.data_ptr<int>(), ci.data_ptr<int>(),
                                      (int)n, colors.data_ptr<int>(), rounds,
-                                     (int)(seed + rounds * 7919),
+                                     (int)(seed + rounds * 7919), mode,
                                      counter.data_ptr<int>(), st);
         int left = counter.cpu().item<int>();
         if (left == 0) break;

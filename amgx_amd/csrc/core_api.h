@@ -124,7 +124,8 @@ void dilu_bwd_color(const int* ro, const int* ci, const TA* va,
 // ---- coloring (setup.hip) ---------------------------------------------------
 // one min-max hash round; returns (via counter) number newly colored.
 void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
-                        int iter, int seed, int* n_uncolored, hipStream_t s);
+                        int iter, int seed, int mode, int* n_uncolored,
+                        hipStream_t s);
 
 // ---- aggregation (setup.hip) ------------------------------------------------
 template <typename T>

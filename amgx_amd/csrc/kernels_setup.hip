@@ -41,10 +41,14 @@ __device__ __forceinline__ unsigned int hash_u32(unsigned int a,
 // already-colored neighbors — greedy-quality color counts (~max degree) with
 // Jones-Plassmann parallel rounds. Hash fixed across rounds, tie-break on
 // row id => deterministic. Colors >= 64 fall back past the bitmask (rare).
+// mode 0: local max takes the smallest color unused by colored neighbors
+// (greedy / PARALLEL_GREEDY quality); mode 1: local max takes color = iter
+// (MULTI_HASH semantics, reference src/matrix_coloring/multi_hash.cu).
 __global__ __launch_bounds__(AMGX_BLOCK) void color_round_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci, int n,
                                    int* __restrict__ colors, int iter,
-                                   int seed, int* __restrict__ n_uncolored) {
+                                   int seed, int mode,
+                                   int* __restrict__ n_uncolored) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     if (colors[i] >= 0) return;
@@ -69,18 +73,24 @@ __global__ __launch_bounds__(AMGX_BLOCK) void color_round_kernel(const int* __re
         if (h > mine) { is_max = false; break; }
     }
     if (is_max) {
-        int c = (int)(__builtin_ffsll((long long)~used) - 1);
-        if (c < 0 || c >= 64) c = big_used_max + 1;   // beyond-bitmask fallback
-        colors[i] = c;
+        if (mode == 1) {
+            colors[i] = iter;
+        } else {
+            int c = (int)(__builtin_ffsll((long long)~used) - 1);
+            if (c < 0 || c >= 64) c = big_used_max + 1;  // beyond-bitmask
+            colors[i] = c;
+        }
     } else {
         atomicAdd(n_uncolored, 1);
     }
 }
 
 void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
-                        int iter, int seed, int* n_uncolored, hipStream_t s) {
+                        int iter, int seed, int mode, int* n_uncolored,
+                        hipStream_t s) {
     hipLaunchKernelGGL(color_round_kernel, dim3(grid_1d(n)), dim3(AMGX_BLOCK),
-                       0, s, ro, ci, n, colors, iter, seed, n_uncolored);
+                       0, s, ro, ci, n, colors, iter, seed, mode,
+                       n_uncolored);
 }
 
 // ============================================================ aggregation

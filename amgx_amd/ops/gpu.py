@@ -168,9 +168,12 @@ def gs_sweep(A, dinv, b, x, coloring, omega: float, symmetric: bool = False):
 
 
 # ---------------------------------------------------------------------- coloring
-def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0):
+def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0,
+                 multihash_rounds: int = 0):
+    """Device Jones-Plassmann greedy coloring rounds; multihash_rounds > 0
+    prepends MULTI_HASH semantics (color = round id, per-round re-hash)."""
     colors, nc = _core.color_minmax(A.row_offsets, A.col_indices, A.n_rows,
-                                    64, int(seed))
+                                    64, int(seed), int(multihash_rounds))
     return colors, int(nc)
 
 
