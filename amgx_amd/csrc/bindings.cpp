@@ -78,6 +78,21 @@ void csrmv(Tensor ro, Tensor ci, Tensor va, int64_t block_dim, Tensor x,
     });
 }
 
+// baseline thread-per-row-component block SpMV (A/B target for the MFMA
+// b=4 kernel in bench_kernels.py)
+void bsrmv_generic(Tensor ro, Tensor ci, Tensor va, int64_t block_dim,
+                   Tensor x, Tensor y, double alpha, double beta) {
+    check_dev(va);
+    int r1 = (int)ro.numel() - 1;
+    DISPATCH_FT2(va, x, "bsrmv_generic", [&] {
+        amgx_hip::bsrmv_generic<scalar_a, scalar_v>(
+            ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_a>(),
+            (int)block_dim, x.data_ptr<scalar_v>(), y.data_ptr<scalar_v>(),
+            nullptr, (scalar_v)alpha, (scalar_v)beta, (scalar_v)0, 0, r1,
+            cur_stream());
+    });
+}
+
 // ---------------------------------------------------------------- BLAS
 Tensor reduce_op(Tensor x, c10::optional<Tensor> y, int64_t op) {
     check_dev(x);
@@ -685,6 +700,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("axpy_dalpha", &axpy_dalpha);
     m.def("mfma4_probe", &mfma4_probe);
     m.def("spgemm_hash", &spgemm_hash);
+    m.def("bsrmv_generic", &bsrmv_generic);
     m.def("scal_drsqrt", &scal_drsqrt);
     m.def("axpby", &axpby);
     m.def("scal", &scal);

@@ -24,6 +24,12 @@ void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
            TV* y, const TV* bvec, TV alpha, TV beta, TV gamma, int r0, int r1,
            hipStream_t s);
 
+// baseline thread-per-row-component block SpMV (bench A/B target)
+template <typename TA, typename TV>
+void bsrmv_generic(const int* ro, const int* ci, const TA* va, int b,
+                   const TV* x, TV* y, const TV* bvec, TV alpha, TV beta,
+                   TV gamma, int r0, int r1, hipStream_t s);
+
 // ---- LDS-hash SpGEMM (kernels_spgemm.hip) -----------------------------------
 // mode 0: C = A*B.  mode 1: aggregation Galerkin — A = aggregate-membership
 // CSR (coarse row -> fine rows), B = fine matrix, key = aggcol[col].

@@ -126,6 +126,19 @@ void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
                        alpha, beta, gamma, r0, r1);
 }
 
+// baseline scalar thread-per-row-component kernel, any b (A/B comparison
+// target for the MFMA b=4 path in bench_kernels.py)
+template <typename TA, typename TV>
+void bsrmv_generic(const int* ro, const int* ci, const TA* va, int b,
+                   const TV* x, TV* y, const TV* bvec, TV alpha, TV beta,
+                   TV gamma, int r0, int r1, hipStream_t s) {
+    long long rows = ((long long)r1 - r0) * b;
+    if (rows <= 0) return;
+    hipLaunchKernelGGL((bsrmv_kernel<TA, TV>), dim3(grid_1d(rows)),
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, va, b, x, y, bvec,
+                       alpha, beta, gamma, r0, r1);
+}
+
 // ============================================================ BLAS-1 reduce
 // Deterministic two-stage reduction: fixed grid of NPART partials, then one
 // block folds them. op: 0 dot, 1 L1, 2 Lmax.
@@ -969,6 +982,9 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
     template void bsrmv<TA, TV>(const int*, const int*, const TA*, int,         \
                                 const TV*, TV*, const TV*, TV, TV, TV, int,     \
                                 int, hipStream_t);                              \
+    template void bsrmv_generic<TA, TV>(const int*, const int*, const TA*,      \
+                                        int, const TV*, TV*, const TV*, TV,     \
+                                        TV, TV, int, int, hipStream_t);         \
     template void jacobi_smooth<TA, TV>(const int*, const int*, const TA*,      \
                                         const TA*, const TV*, const TV*, TV*,   \
                                         TV, int, int, double, hipStream_t);     \

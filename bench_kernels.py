@@ -87,6 +87,32 @@ def main():
         lambda: dilu_solve(A, einv, col, r, 0.75, y), args.iters, sync),
         2 * rows_bytes + n * 40)
 
+    # block-4 kernels: MFMA wave path vs the scalar baseline
+    from amgx_amd.problems import block_laplacian
+    side = max(args.size // 3, 24)
+    A4 = block_laplacian(side, side, block_dim=4, seed=3).to(dev)
+    n4 = A4.n_rows
+    x4 = torch.rand(n4 * 4, dtype=torch.float64, device=dev)
+    y4 = torch.zeros_like(x4)
+    b4_bytes = A4.nnz * (16 * 8 + 4) + (n4 + 1) * 4 + n4 * 4 * 16
+    rec("bsrmv_b4_mfma", bench(lambda: ops.spmv(A4, x4, y4),
+                               args.iters, sync), b4_bytes)
+    if dev != "cpu":
+        from amgx_amd import _core
+        rec("bsrmv_b4_scalar_base", bench(
+            lambda: _core.bsrmv_generic(A4.row_offsets, A4.col_indices,
+                                        A4.values.reshape(-1), 4,
+                                        x4, y4, 1.0, 0.0),
+            args.iters, sync), b4_bytes)
+    col4 = MatrixColoring.create(A4, ConfigScope(None, {}))
+    e4 = dilu_setup(A4, col4)
+    r4 = torch.rand_like(x4)
+    rec("dilu_b4_apply_mfma", bench(
+        lambda: dilu_solve(A4, e4, col4, r4, 1.0, y4), args.iters, sync),
+        2 * A4.nnz * 16 * 8 + n4 * 4 * 40)
+    rec("dilu_b4_setup_mfma", bench(
+        lambda: dilu_setup(A4, col4), max(args.iters // 4, 2), sync))
+
     # setup-path
     rec("coloring_minmax", bench(
         lambda: ops.color_matrix(A), max(args.iters // 4, 2), sync))
