@@ -238,6 +238,27 @@ template <typename TA, typename TV>
 void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos,
                      const TA* lu, const int* didx, const int* rows, int count,
                      const TV* y, TV* z, int n, hipStream_t s);
+// block ILU(0): L_ik = A_ik U_kk^{-1}, A_ij -= L_ik U_kj; dinv = inverted
+// pivot diagonal blocks, refreshed per color via ilu0_invert_diag_block
+template <typename T>
+void ilu0_factor_color_block_launch(const int* ro, const int* ci,
+                                    const int* pos, const int* didx,
+                                    const int* rows, int count, T* lu,
+                                    const T* dinv, int n, int b,
+                                    hipStream_t s);
+template <typename T>
+void ilu0_invert_diag_block_launch(const int* didx, const int* rows,
+                                   int count, const T* lu, T* dinv, int b,
+                                   hipStream_t s);
+template <typename TA, typename TV>
+void ilu0_fwd_block_launch(const int* ro, const int* ci, const int* pos,
+                           const TA* lu, const int* rows, int count,
+                           const TV* r, TV* y, int n, int b, hipStream_t s);
+template <typename TA, typename TV>
+void ilu0_bwd_block_launch(const int* ro, const int* ci, const int* pos,
+                           const TA* lu, const TA* dinv, const int* rows,
+                           int count, const TV* y, TV* z, int n, int b,
+                           hipStream_t s);
 
 }  // namespace amgx_hip
 

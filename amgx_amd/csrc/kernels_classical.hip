@@ -360,6 +360,222 @@ __global__ __launch_bounds__(AMGX_BLOCK) void ilu0_bwd(const int* __restrict__ r
     z[i] = sum / d;
 }
 
+// ------------------------------------------------------------ block ILU(0)
+// Same color-ordered elimination with b x b blocks (reference
+// src/solvers/multicolor_ilu_solver.cu block path):
+//   L_ik = A_ik * U_kk^{-1};  A_ij -= L_ik * U_kj
+// dinv[] holds the inverted pivot diagonal blocks of already-factored
+// colors (filled by ilu0_invert_diag_block after each color).
+template <typename T, int BMAX>
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_factor_color_block(
+    const int* __restrict__ ro, const int* __restrict__ ci,
+    const int* __restrict__ pos, const int* __restrict__ didx,
+    const int* __restrict__ rows, int count, T* __restrict__ lu,
+    const T* __restrict__ dinv, int n, int b) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    int s = ro[i], e = ro[i + 1];
+    int pi = pos[i];
+    int bb = b * b;
+    T L[BMAX * BMAX];
+    for (int step = 0;; ++step) {
+        int kidx = -1, kpos = 0x7fffffff;
+        for (int k = s; k < e; ++k) {
+            int j = ci[k];
+            if (j >= n) continue;
+            int pj = pos[j];
+            if (pj < pi && pj >= step && pj < kpos) { kpos = pj; kidx = k; }
+        }
+        if (kidx < 0) break;
+        int kcol = ci[kidx];
+        // L = lu[kidx] * dinv[kcol]
+        const T* Aik = lu + (long long)kidx * bb;
+        const T* Ukk_inv = dinv + (long long)kcol * bb;
+        for (int r = 0; r < b; ++r)
+            for (int c = 0; c < b; ++c) {
+                T acc = T(0);
+                for (int q = 0; q < b; ++q)
+                    acc += Aik[r * b + q] * Ukk_inv[q * b + c];
+                L[r * b + c] = acc;
+            }
+        for (int q = 0; q < bb; ++q) lu[(long long)kidx * bb + q] = L[q];
+        // row i -= L * U(kcol, j) over pattern entries with pos[j] > kpos
+        for (int k2 = ro[kcol]; k2 < ro[kcol + 1]; ++k2) {
+            int j = ci[k2];
+            if (j >= n || pos[j] <= kpos) continue;
+            int lo = s, hi = e;
+            while (lo < hi) {
+                int mid = (lo + hi) >> 1;
+                int c = ci[mid];
+                if (c == j) {
+                    const T* Ukj = lu + (long long)k2 * bb;
+                    T* Aij = lu + (long long)mid * bb;
+                    for (int r = 0; r < b; ++r)
+                        for (int cc = 0; cc < b; ++cc) {
+                            T acc = T(0);
+                            for (int q = 0; q < b; ++q)
+                                acc += L[r * b + q] * Ukj[q * b + cc];
+                            Aij[r * b + cc] -= acc;
+                        }
+                    break;
+                }
+                if (c < j) lo = mid + 1; else hi = mid;
+            }
+        }
+        step = kpos;
+    }
+}
+
+// invert the (freshly factored) pivot diagonal blocks of one color
+template <typename T, int BMAX>
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_invert_diag_block(
+    const int* __restrict__ didx, const int* __restrict__ rows, int count,
+    const T* __restrict__ lu, T* __restrict__ dinv, int b) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    int dk = didx[i];
+    int bb = b * b;
+    T D[BMAX * BMAX], Inv[BMAX * BMAX];
+    double dmax = 0.0;
+    for (int q = 0; q < bb; ++q) {
+        D[q] = dk >= 0 ? lu[(long long)dk * bb + q] : T(q % (b + 1) == 0);
+        dmax = fmax(dmax, fabs((double)D[q]));
+    }
+    if (dmax == 0.0)                      // fully zero pivot: identity
+        for (int q = 0; q < bb; ++q) D[q] = T(q % (b + 1) == 0);
+    small_mat_inv(D, Inv, b);
+    bool bad = false;
+    for (int q = 0; q < bb; ++q)
+        if (!isfinite((double)Inv[q]) ||
+            fabs((double)Inv[q]) > 1e12 * (1.0 + 1.0 / (dmax + 1e-300)))
+            bad = true;
+    if (bad) {                            // near-singular: diagonal fallback
+        for (int q = 0; q < bb; ++q) Inv[q] = T(0);
+        for (int r = 0; r < b; ++r) {
+            double d = dk >= 0 ? (double)lu[(long long)dk * bb + r * b + r]
+                               : 1.0;
+            Inv[r * b + r] = (T)(d != 0.0 ? 1.0 / d : 1.0);
+        }
+    }
+    for (int q = 0; q < bb; ++q) dinv[(long long)i * bb + q] = Inv[q];
+}
+
+// fwd: y_i = r_i - sum_{pos[j]<pos[i]} L_ij y_j   (unit diagonal blocks)
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_fwd_block(
+    const int* __restrict__ ro, const int* __restrict__ ci,
+    const int* __restrict__ pos, const TA* __restrict__ lu,
+    const int* __restrict__ rows, int count, const TV* __restrict__ r,
+    TV* __restrict__ y, int n, int b) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    int pi = pos[i];
+    int bb = b * b;
+    TV acc[16];
+    for (int c = 0; c < b; ++c) acc[c] = r[(long long)i * b + c];
+    for (int k = ro[i]; k < ro[i + 1]; ++k) {
+        int j = ci[k];
+        if (j >= n || pos[j] >= pi) continue;
+        const TA* blk = lu + (long long)k * bb;
+        const TV* ys = y + (long long)j * b;
+        for (int c = 0; c < b; ++c) {
+            TV s = TV(0);
+            for (int q = 0; q < b; ++q) s += (TV)blk[c * b + q] * ys[q];
+            acc[c] -= s;
+        }
+    }
+    for (int c = 0; c < b; ++c) y[(long long)i * b + c] = acc[c];
+}
+
+// bwd: z_i = dinv_i * (y_i - sum_{pos[j]>pos[i]} U_ij z_j)
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_bwd_block(
+    const int* __restrict__ ro, const int* __restrict__ ci,
+    const int* __restrict__ pos, const TA* __restrict__ lu,
+    const TA* __restrict__ dinv, const int* __restrict__ rows, int count,
+    const TV* __restrict__ y, TV* __restrict__ z, int n, int b) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    int pi = pos[i];
+    int bb = b * b;
+    TV acc[16];
+    for (int c = 0; c < b; ++c) acc[c] = y[(long long)i * b + c];
+    for (int k = ro[i]; k < ro[i + 1]; ++k) {
+        int j = ci[k];
+        if (j >= n || pos[j] <= pi) continue;
+        const TA* blk = lu + (long long)k * bb;
+        const TV* zs = z + (long long)j * b;
+        for (int c = 0; c < b; ++c) {
+            TV s = TV(0);
+            for (int q = 0; q < b; ++q) s += (TV)blk[c * b + q] * zs[q];
+            acc[c] -= s;
+        }
+    }
+    const TA* Dv = dinv + (long long)i * bb;
+    for (int c = 0; c < b; ++c) {
+        TV s = TV(0);
+        for (int q = 0; q < b; ++q) s += (TV)Dv[c * b + q] * acc[q];
+        z[(long long)i * b + c] = s;
+    }
+}
+
+template <typename T>
+void ilu0_factor_color_block_launch(const int* ro, const int* ci,
+                                    const int* pos, const int* didx,
+                                    const int* rows, int count, T* lu,
+                                    const T* dinv, int n, int b,
+                                    hipStream_t s) {
+    if (count <= 0) return;
+    if (b <= 8)
+        hipLaunchKernelGGL((ilu0_factor_color_block<T, 8>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s, ro,
+                           ci, pos, didx, rows, count, lu, dinv, n, b);
+    else
+        hipLaunchKernelGGL((ilu0_factor_color_block<T, 16>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s, ro,
+                           ci, pos, didx, rows, count, lu, dinv, n, b);
+}
+
+template <typename T>
+void ilu0_invert_diag_block_launch(const int* didx, const int* rows,
+                                   int count, const T* lu, T* dinv, int b,
+                                   hipStream_t s) {
+    if (count <= 0) return;
+    if (b <= 8)
+        hipLaunchKernelGGL((ilu0_invert_diag_block<T, 8>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
+                           didx, rows, count, lu, dinv, b);
+    else
+        hipLaunchKernelGGL((ilu0_invert_diag_block<T, 16>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
+                           didx, rows, count, lu, dinv, b);
+}
+
+template <typename TA, typename TV>
+void ilu0_fwd_block_launch(const int* ro, const int* ci, const int* pos,
+                           const TA* lu, const int* rows, int count,
+                           const TV* r, TV* y, int n, int b, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((ilu0_fwd_block<TA, TV>), dim3(grid_1d(count)),
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, pos, lu, rows, count,
+                       r, y, n, b);
+}
+
+template <typename TA, typename TV>
+void ilu0_bwd_block_launch(const int* ro, const int* ci, const int* pos,
+                           const TA* lu, const TA* dinv, const int* rows,
+                           int count, const TV* y, TV* z, int n, int b,
+                           hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((ilu0_bwd_block<TA, TV>), dim3(grid_1d(count)),
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, pos, lu, dinv, rows,
+                       count, y, z, n, b);
+}
+
 template <typename T>
 void ilu0_factor_color_launch(const int* ro, const int* ci, const int* pos,
                               const int* didx, const int* rows, int count,
@@ -404,7 +620,18 @@ void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos,
                                         T*, int, hipStream_t);                 \
     template void ilu0_bwd_launch<T, T>(const int*, const int*, const int*,    \
                                         const T*, const int*, const int*,      \
-                                        int, const T*, T*, int, hipStream_t);
+                                        int, const T*, T*, int, hipStream_t);  \
+    template void ilu0_factor_color_block_launch<T>(                           \
+        const int*, const int*, const int*, const int*, const int*, int, T*,  \
+        const T*, int, int, hipStream_t);                                      \
+    template void ilu0_invert_diag_block_launch<T>(                            \
+        const int*, const int*, int, const T*, T*, int, hipStream_t);          \
+    template void ilu0_fwd_block_launch<T, T>(                                 \
+        const int*, const int*, const int*, const T*, const int*, int,         \
+        const T*, T*, int, int, hipStream_t);                                  \
+    template void ilu0_bwd_block_launch<T, T>(                                 \
+        const int*, const int*, const int*, const T*, const T*, const int*,    \
+        int, const T*, T*, int, int, hipStream_t);
 
 INSTANTIATE_CLASSICAL(double)
 INSTANTIATE_CLASSICAL(float)
@@ -418,6 +645,12 @@ template void ilu0_bwd_launch<float, double>(const int*, const int*,
                                              const int*, const int*, int,
                                              const double*, double*, int,
                                              hipStream_t);
+template void ilu0_fwd_block_launch<float, double>(
+    const int*, const int*, const int*, const float*, const int*, int,
+    const double*, double*, int, int, hipStream_t);
+template void ilu0_bwd_block_launch<float, double>(
+    const int*, const int*, const int*, const float*, const float*,
+    const int*, int, const double*, double*, int, int, hipStream_t);
 
 }  // namespace amgx_hip
 

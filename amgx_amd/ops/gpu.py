@@ -383,25 +383,29 @@ def _color_pos(A, coloring):
 
 
 def ilu0_setup(A, coloring):
-    if A.block_dim != 1:
-        raise NotImplementedError(
-            "device block ILU(0) kernels pending — host modes support it; "
-            "on device use MULTICOLOR_DILU for block systems")
     pos = _color_pos(A, coloring)
+    if A.block_dim != 1:
+        lu, dinv = _core.ilu0_setup_block(
+            A.row_offsets, A.col_indices, A.values.reshape(-1), A.block_dim,
+            _didx(A), pos, coloring.rows_sorted, coloring.bounds)
+        return (lu, dinv)
     lu = _core.ilu0_setup(A.row_offsets, A.col_indices, A.values, _didx(A),
                           pos, coloring.rows_sorted, coloring.bounds)
     return lu
 
 
 def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
-    if A.block_dim != 1:
-        raise NotImplementedError(
-            "device block ILU(0) kernels pending — host modes support it; "
-            "on device use MULTICOLOR_DILU for block systems")
     n = A.n_cols * A.block_dim
     y = _scratch(A, "ilu_y", n, r.dtype)    # vector precision (dDFI mixed)
     z = _scratch(A, "ilu_z", n, r.dtype)
     pos = _color_pos(A, coloring)
+    if A.block_dim != 1:
+        lu, dinv = factors
+        _core.ilu0_apply_block(A.row_offsets, A.col_indices, lu, dinv,
+                               A.block_dim, pos, coloring.rows_sorted,
+                               coloring.bounds, r.reshape(-1), y, z,
+                               x.reshape(-1), float(relaxation))
+        return x
     _core.ilu0_apply(A.row_offsets, A.col_indices, factors, _didx(A), pos,
                      coloring.rows_sorted, coloring.bounds, r.reshape(-1),
                      y, z, x.reshape(-1), float(relaxation))

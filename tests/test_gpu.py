@@ -355,6 +355,39 @@ def test_ilu0_gpu_matches_cpu():
     assert torch.allclose(x2.cpu(), x1, rtol=1e-11, atol=1e-12)
 
 
+def test_ilu0_block_gpu_matches_cpu():
+    """Device block ILU(0) (kernels_classical.hip block path) against the
+    host block reference (close ops/gpu.py guard)."""
+    from amgx_amd.ops import cpu as cpu_ops
+    from amgx_amd.amg.coloring import MatrixColoring
+    A = block_laplacian(10, 10, block_dim=3, seed=5)
+    Ag = to_gpu(A)
+    colg = MatrixColoring.create(Ag)
+    col_cpu = MatrixColoring(colg.colors.cpu(), colg.num_colors)
+    f_ref = cpu_ops.ilu0_setup(A, col_cpu)
+    f_gpu = ops._backend(Ag).ilu0_setup(Ag, colg)
+    r = torch.rand(A.n_rows * 3, dtype=torch.float64)
+    x1 = torch.zeros_like(r)
+    x2 = x1.clone().cuda()
+    cpu_ops.ilu0_solve(A, f_ref, col_cpu, r, x1, 1.0)
+    ops._backend(Ag).ilu0_solve(Ag, f_gpu, colg, r.cuda(), x2, 1.0)
+    assert torch.allclose(x2.cpu(), x1, rtol=1e-9, atol=1e-10), \
+        f"max dev {(x2.cpu() - x1).abs().max()}"
+    # b=4 as well (no MFMA in ILU yet, but the dispatch must be correct)
+    A4 = block_laplacian(8, 8, block_dim=4, seed=6)
+    A4g = to_gpu(A4)
+    c4g = MatrixColoring.create(A4g)
+    c4 = MatrixColoring(c4g.colors.cpu(), c4g.num_colors)
+    fr = cpu_ops.ilu0_setup(A4, c4)
+    fg = ops._backend(A4g).ilu0_setup(A4g, c4g)
+    r4 = torch.rand(A4.n_rows * 4, dtype=torch.float64)
+    y1 = torch.zeros_like(r4)
+    y2 = y1.clone().cuda()
+    cpu_ops.ilu0_solve(A4, fr, c4, r4, y1, 0.9)
+    ops._backend(A4g).ilu0_solve(A4g, fg, c4g, r4.cuda(), y2, 0.9)
+    assert torch.allclose(y2.cpu(), y1, rtol=1e-9, atol=1e-10)
+
+
 def test_fgmres_ilu0_gpu():
     cfg = {"solver": {"solver": "FGMRES", "preconditioner": "MULTICOLOR_ILU",
                       "gmres_n_restart": 20, "max_iters": 200,
