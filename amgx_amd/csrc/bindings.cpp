@@ -295,6 +295,29 @@ void gs_sweep(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
     });
 }
 
+// color-sorted scalar GS sweep (reorder-by-color layout)
+void gs_sweep_sorted(Tensor ro_s, Tensor ci_s, Tensor va_s, Tensor dinv_s,
+                     Tensor bvec, Tensor x, Tensor rows_sorted,
+                     std::vector<int64_t> bounds, double omega,
+                     bool symmetric) {
+    int nc = (int)bounds.size() - 1;
+    DISPATCH_FT2(va_s, x, "gs_sweep_sorted", [&] {
+        auto run = [&](int c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) return;
+            amgx_hip::gs_rows_sorted<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>() + s, ci_s.data_ptr<int>(),
+                va_s.data_ptr<scalar_a>(), dinv_s.data_ptr<scalar_a>() + s,
+                bvec.data_ptr<scalar_v>(), x.data_ptr<scalar_v>(),
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                (scalar_v)omega, cur_stream());
+        };
+        for (int c = 0; c < nc; ++c) run(c);
+        if (symmetric)
+            for (int c = nc - 1; c >= 0; --c) run(c);
+    });
+}
+
 // ---------------------------------------------------------------- DILU
 Tensor dilu_setup(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor didx,
                   Tensor tidx, Tensor colors, Tensor rows_sorted,
@@ -354,6 +377,44 @@ void dilu_apply(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor einv,
     });
 }
 
+// color-sorted DILU apply: matrix arrays in rows_sorted order (one
+// contiguous slab per color — reference reorder-by-color layout)
+void dilu_apply_sorted(Tensor ro_s, Tensor ci_s, Tensor va_s, int64_t b,
+                       Tensor einv_s, Tensor rows_sorted,
+                       std::vector<int64_t> bounds, Tensor r, Tensor w,
+                       Tensor z, Tensor x, double relax) {
+    int nc = (int)bounds.size() - 1;
+    int bb = (int)(b * b);
+    w.zero_();
+    z.zero_();
+    DISPATCH_FT2(va_s, x, "dilu_apply_sorted", [&] {
+        hipStream_t st = cur_stream();
+        for (int c = 0; c < nc; ++c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::dilu_fwd_sorted<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>() + s, ci_s.data_ptr<int>(),
+                va_s.data_ptr<scalar_a>(),
+                einv_s.data_ptr<scalar_a>() + s * bb,
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                r.data_ptr<scalar_v>(), w.data_ptr<scalar_v>(), (int)b, st);
+        }
+        for (int c = nc - 1; c >= 0; --c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::dilu_bwd_sorted<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>() + s, ci_s.data_ptr<int>(),
+                va_s.data_ptr<scalar_a>(),
+                einv_s.data_ptr<scalar_a>() + s * bb,
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                w.data_ptr<scalar_v>(), z.data_ptr<scalar_v>(), (int)b, st);
+        }
+        amgx_hip::axpy<scalar_v>(x.data_ptr<scalar_v>(),
+                                 z.data_ptr<scalar_v>(), (scalar_v)relax,
+                                 x.numel(), st);
+    });
+}
+
 // ---------------------------------------------------------------- coloring
 // mode 0: greedy smallest-unused color (MIN_MAX / PARALLEL_GREEDY class);
 // mode 1: MULTI_HASH — `mode1_rounds` rounds assigning color = round id
@@ -363,6 +424,7 @@ std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
                                          int64_t mode1_rounds) {
     auto colors = torch::full({n}, -1,
                               ro.options().dtype(torch::kInt32));
+    auto colors_next = torch::empty_like(colors);
     auto counter = torch::zeros({1}, ro.options().dtype(torch::kInt32));
     hipStream_t st = cur_stream();
     int rounds = 0;
@@ -370,9 +432,11 @@ std::tuple<Tensor, int64_t> color_minmax(Tensor ro, Tensor ci, int64_t n,
         counter.zero_();
         int mode = rounds < mode1_rounds ? 1 : 0;
         amgx_hip::color_minmax_round(ro.data_ptr<int>(), ci.data_ptr<int>(),
-                                     (int)n, colors.data_ptr<int>(), rounds,
+                                     (int)n, colors.data_ptr<int>(),
+                                     colors_next.data_ptr<int>(), rounds,
                                      (int)(seed + rounds * 7919), mode,
                                      counter.data_ptr<int>(), st);
+        std::swap(colors, colors_next);
         int left = counter.cpu().item<int>();
         if (left == 0) break;
     }
@@ -774,8 +838,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("jacobi_smooth", &jacobi_smooth);
     m.def("gs_smooth_rows", &gs_smooth_rows);
     m.def("gs_sweep", &gs_sweep);
+    m.def("gs_sweep_sorted", &gs_sweep_sorted);
     m.def("dilu_setup", &dilu_setup);
     m.def("dilu_apply", &dilu_apply);
+    m.def("dilu_apply_sorted", &dilu_apply_sorted);
     m.def("color_minmax", &color_minmax);
     m.def("size2_match", &size2_match);
     m.def("restrict_agg", &restrict_agg);

@@ -45,6 +45,18 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
                       int** big_rows_out, int* n_big_out, hipStream_t s);
 void free_device_buf(void* p, hipStream_t s);
 
+// ---- color-sorted DILU sweeps (reorder-by-color layout) ---------------------
+// ro_s/rows/einv_s pre-offset to the color base; matrix arrays are the
+// rows_sorted-gathered copy so each color reads one contiguous slab.
+template <typename TA, typename TV>
+void dilu_fwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                     const TA* einv_s, const int* rows, int count,
+                     const TV* r, TV* w, int b, hipStream_t s);
+template <typename TA, typename TV>
+void dilu_bwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                     const TA* einv_s, const int* rows, int count,
+                     const TV* wv, TV* z, int b, hipStream_t s);
+
 // ---- MFMA wave-structured block-4 kernels (kernels_mfma.hip) ----------------
 // v_mfma_f64_4x4x4_4b_f64 path: wave = 4 rows x 16 lanes, coalesced block
 // loads; used automatically by the b==4 dispatch of bsrmv/dilu_* above.
@@ -129,9 +141,9 @@ void dilu_bwd_color(const int* ro, const int* ci, const TA* va,
 
 // ---- coloring (setup.hip) ---------------------------------------------------
 // one min-max hash round; returns (via counter) number newly colored.
-void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
-                        int iter, int seed, int mode, int* n_uncolored,
-                        hipStream_t s);
+void color_minmax_round(const int* ro, const int* ci, int n,
+                        const int* colors_prev, int* colors_next, int iter,
+                        int seed, int mode, int* n_uncolored, hipStream_t s);
 
 // ---- aggregation (setup.hip) ------------------------------------------------
 template <typename T>

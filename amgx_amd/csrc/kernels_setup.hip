@@ -44,14 +44,19 @@ __device__ __forceinline__ unsigned int hash_u32(unsigned int a,
 // mode 0: local max takes the smallest color unused by colored neighbors
 // (greedy / PARALLEL_GREEDY quality); mode 1: local max takes color = iter
 // (MULTI_HASH semantics, reference src/matrix_coloring/multi_hash.cu).
+// Reads a FROZEN snapshot (colors_prev) and writes colors_next: same-round
+// neighbor decisions must be invisible, or the result depends on wave
+// scheduling — valid but nondeterministic (host model's frozen-snapshot
+// semantics; determinism_flag discipline, reference src/core.cu:313).
 __global__ __launch_bounds__(AMGX_BLOCK) void color_round_kernel(const int* __restrict__ ro,
                                    const int* __restrict__ ci, int n,
-                                   int* __restrict__ colors, int iter,
+                                   const int* __restrict__ colors_prev,
+                                   int* __restrict__ colors_next, int iter,
                                    int seed, int mode,
                                    int* __restrict__ n_uncolored) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    if (colors[i] >= 0) return;
+    if (colors_prev[i] >= 0) { colors_next[i] = colors_prev[i]; return; }
     unsigned long long mine =
         ((unsigned long long)hash_u32((unsigned)i, (unsigned)seed) << 32) |
         (unsigned)i;
@@ -61,7 +66,7 @@ __global__ __launch_bounds__(AMGX_BLOCK) void color_round_kernel(const int* __re
     for (int k = ro[i]; k < ro[i + 1]; ++k) {
         int j = ci[k];
         if (j == i || j >= n) continue;
-        int cj = colors[j];
+        int cj = colors_prev[j];
         if (cj >= 0) {
             if (cj < 64) used |= (1ull << cj);
             else big_used_max = max(big_used_max, cj);
@@ -72,25 +77,27 @@ __global__ __launch_bounds__(AMGX_BLOCK) void color_round_kernel(const int* __re
             (unsigned)j;
         if (h > mine) { is_max = false; break; }
     }
+    int out = -1;
     if (is_max) {
         if (mode == 1) {
-            colors[i] = iter;
+            out = iter;
         } else {
             int c = (int)(__builtin_ffsll((long long)~used) - 1);
             if (c < 0 || c >= 64) c = big_used_max + 1;  // beyond-bitmask
-            colors[i] = c;
+            out = c;
         }
     } else {
         atomicAdd(n_uncolored, 1);
     }
+    colors_next[i] = out;
 }
 
-void color_minmax_round(const int* ro, const int* ci, int n, int* colors,
-                        int iter, int seed, int mode, int* n_uncolored,
-                        hipStream_t s) {
+void color_minmax_round(const int* ro, const int* ci, int n,
+                        const int* colors_prev, int* colors_next, int iter,
+                        int seed, int mode, int* n_uncolored, hipStream_t s) {
     hipLaunchKernelGGL(color_round_kernel, dim3(grid_1d(n)), dim3(AMGX_BLOCK),
-                       0, s, ro, ci, n, colors, iter, seed, mode,
-                       n_uncolored);
+                       0, s, ro, ci, n, colors_prev, colors_next, iter, seed,
+                       mode, n_uncolored);
 }
 
 // ============================================================ aggregation

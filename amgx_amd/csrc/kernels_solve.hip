@@ -485,6 +485,34 @@ void jacobi_smooth(const int* ro, const int* ci, const TA* va, const TA* dinv,
 }
 
 // ============================================================ multicolor GS
+// color-sorted GS row sweep (reorder-by-color layout; ro_s/rows/dinv_s
+// pre-offset to the color base, matrix arrays gathered in rows_sorted
+// order so the color reads one contiguous slab)
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_scalar_sorted(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ dinv_s,
+    const TV* __restrict__ bvec, TV* __restrict__ x,
+    const int* __restrict__ rows, int count, TV omega) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    TV sum = TV(0);
+    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+        sum += (TV)va_s[k] * x[ci_s[k]];
+    x[i] += omega * (TV)dinv_s[t] * (bvec[i] - sum);
+}
+
+template <typename TA, typename TV>
+void gs_rows_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                    const TA* dinv_s, const TV* bvec, TV* x, const int* rows,
+                    int count, TV omega, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((gs_rows_scalar_sorted<TA, TV>),
+                       dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s, ro_s,
+                       ci_s, va_s, dinv_s, bvec, x, rows, count, omega);
+}
+
 template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_scalar(const int* __restrict__ ro,
                                const int* __restrict__ ci,
@@ -757,6 +785,128 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_block(const int* __restri
     }
 }
 
+// ---- color-sorted sweep variants (reference reorder-by-color layout,
+// include/matrix.h:766, src/core.cu:489-506, redesigned as a sorted COPY):
+// the matrix rows are stored in rows_sorted order, so each color's sweep
+// reads a CONTIGUOUS slab of values/columns exactly once — the original
+// per-color launches re-fetched ~num_colors scattered cache lines per
+// sweep.  ro_s/rows/einv_s pointers arrive pre-offset to the color base.
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar_sorted(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, int count, const TV* __restrict__ r,
+    TV* __restrict__ w) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    TV sum = TV(0);
+    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+        sum += (TV)va_s[k] * w[ci_s[k]];
+    w[i] = (TV)einv_s[t] * (r[i] - sum);
+}
+
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_scalar_sorted(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, int count, const TV* __restrict__ wv,
+    TV* __restrict__ z) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    TV sum = TV(0);
+    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+        sum += (TV)va_s[k] * z[ci_s[k]];
+    z[i] = wv[i] - (TV)einv_s[t] * sum;
+}
+
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_block_sorted(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, int count, const TV* __restrict__ r,
+    TV* __restrict__ w, int b) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    TV acc[16];
+    for (int c = 0; c < b; ++c) {
+        TV sum = TV(0);
+        for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+            if (ci_s[k] == i) continue;
+            const TA* blk = va_s + ((long long)k * b + c) * b;
+            const TV* ws = w + (long long)ci_s[k] * b;
+            for (int q = 0; q < b; ++q) sum += (TV)blk[q] * ws[q];
+        }
+        acc[c] = r[(long long)i * b + c] - sum;
+    }
+    const TA* E = einv_s + (long long)t * b * b;
+    for (int c = 0; c < b; ++c) {
+        TV s = TV(0);
+        for (int q = 0; q < b; ++q) s += (TV)E[c * b + q] * acc[q];
+        w[(long long)i * b + c] = s;
+    }
+}
+
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_block_sorted(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, int count, const TV* __restrict__ wv,
+    TV* __restrict__ z, int b) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    TV acc[16];
+    for (int c = 0; c < b; ++c) {
+        TV sum = TV(0);
+        for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+            if (ci_s[k] == i) continue;
+            const TA* blk = va_s + ((long long)k * b + c) * b;
+            const TV* zs = z + (long long)ci_s[k] * b;
+            for (int q = 0; q < b; ++q) sum += (TV)blk[q] * zs[q];
+        }
+        acc[c] = sum;
+    }
+    const TA* E = einv_s + (long long)t * b * b;
+    for (int c = 0; c < b; ++c) {
+        TV s = TV(0);
+        for (int q = 0; q < b; ++q) s += (TV)E[c * b + q] * acc[q];
+        z[(long long)i * b + c] = wv[(long long)i * b + c] - s;
+    }
+}
+
+template <typename TA, typename TV>
+void dilu_fwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                     const TA* einv_s, const int* rows, int count,
+                     const TV* r, TV* w, int b, hipStream_t s) {
+    if (count <= 0) return;
+    if (b == 1)
+        hipLaunchKernelGGL((dilu_fwd_scalar_sorted<TA, TV>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
+                           ro_s, ci_s, va_s, einv_s, rows, count, r, w);
+    else
+        hipLaunchKernelGGL((dilu_fwd_block_sorted<TA, TV>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
+                           ro_s, ci_s, va_s, einv_s, rows, count, r, w, b);
+}
+
+template <typename TA, typename TV>
+void dilu_bwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                     const TA* einv_s, const int* rows, int count,
+                     const TV* wv, TV* z, int b, hipStream_t s) {
+    if (count <= 0) return;
+    if (b == 1)
+        hipLaunchKernelGGL((dilu_bwd_scalar_sorted<TA, TV>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
+                           ro_s, ci_s, va_s, einv_s, rows, count, wv, z);
+    else
+        hipLaunchKernelGGL((dilu_bwd_block_sorted<TA, TV>),
+                           dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
+                           ro_s, ci_s, va_s, einv_s, rows, count, wv, z, b);
+}
+
 template <typename TA, typename TV>
 void dilu_fwd_color(const int* ro, const int* ci, const TA* va,
                     const TA* einv, const int* colors, const int* rows,
@@ -988,6 +1138,15 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
     template void jacobi_smooth<TA, TV>(const int*, const int*, const TA*,      \
                                         const TA*, const TV*, const TV*, TV*,   \
                                         TV, int, int, double, hipStream_t);     \
+    template void dilu_fwd_sorted<TA, TV>(const int*, const int*, const TA*,    \
+                                          const TA*, const int*, int,           \
+                                          const TV*, TV*, int, hipStream_t);    \
+    template void dilu_bwd_sorted<TA, TV>(const int*, const int*, const TA*,    \
+                                          const TA*, const int*, int,           \
+                                          const TV*, TV*, int, hipStream_t);    \
+    template void gs_rows_sorted<TA, TV>(const int*, const int*, const TA*,     \
+                                         const TA*, const TV*, TV*,             \
+                                         const int*, int, TV, hipStream_t);     \
     template void gs_smooth_rows<TA, TV>(const int*, const int*, const TA*,     \
                                          const TA*, const TV*, TV*,             \
                                          const int*, int, TV, int, int,         \

@@ -161,6 +161,24 @@ def gs_smooth_color(A, dinv, b, x, color_rows, omega: float):
 
 
 def gs_sweep(A, dinv, b, x, coloring, omega: float, symmetric: bool = False):
+    if A.block_dim == 1:
+        # color-sorted slab layout (reference reorder-by-color)
+        ro_s, ci_s, pos, perm = _color_sorted_struct(A, coloring)
+        vkey = ("gs_va_s", id(coloring), A.values._version)
+        va_s = A._cache.get(vkey)
+        if va_s is None:
+            va_s = A.values[pos].contiguous()
+            A._cache[vkey] = va_s
+        dkey = ("gs_dinv_s", id(coloring), dinv._version, dinv.data_ptr())
+        dinv_s = A._cache.get(dkey)
+        if dinv_s is None:
+            dinv_s = dinv.reshape(-1)[perm].contiguous()
+            A._cache[dkey] = dinv_s
+        _core.gs_sweep_sorted(ro_s, ci_s, va_s, dinv_s, b.reshape(-1),
+                              x.reshape(-1), coloring.rows_sorted,
+                              coloring.bounds, float(omega),
+                              bool(symmetric))
+        return x
     _core.gs_sweep(A.row_offsets, A.col_indices, A.values, A.block_dim, dinv,
                    b.reshape(-1), x.reshape(-1), coloring.rows_sorted,
                    coloring.bounds, float(omega), bool(symmetric))
@@ -178,16 +196,68 @@ def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0,
 
 
 # ---------------------------------------------------------------------- DILU
+def _color_sorted_struct(A, coloring):
+    """Structure of the rows_sorted-gathered matrix copy (reference
+    reorder-by-color, include/matrix.h:766): slot-ordered row offsets,
+    gathered columns, and the nz gather map (for value re-gathers after
+    replace_coefficients). Cached per (matrix, coloring)."""
+    key = ("csorted", id(coloring))
+    st = A._cache.get(key)
+    if st is None:
+        perm = coloring.rows_sorted.to(torch.int64)
+        ro64 = A.row_offsets.to(torch.int64)
+        deg = ro64[1:] - ro64[:-1]
+        counts = deg[perm]
+        total = int(A.nnz)
+        ro_s = torch.zeros(A.n_rows + 1, dtype=torch.int32, device=A.device)
+        csum = torch.cumsum(counts, 0)
+        ro_s[1:] = csum.to(torch.int32)
+        pos = (torch.repeat_interleave(ro64[perm], counts)
+               + torch.arange(total, device=A.device, dtype=torch.int64)
+               - torch.repeat_interleave(csum - counts, counts))
+        ci_s = A.col_indices[pos].contiguous()
+        st = (ro_s, ci_s, pos, perm)
+        A._cache[key] = st
+    return st
+
+
+class DiluState:
+    """Device DILU factor state: Einv in original row order (einv) plus the
+    color-sorted matrix/Einv copies the sweeps read contiguously."""
+    __slots__ = ("einv", "va_s", "einv_s")
+
+    def __init__(self, einv, va_s, einv_s):
+        self.einv = einv
+        self.va_s = va_s
+        self.einv_s = einv_s
+
+    def cpu(self):          # test convenience: compare against host Einv
+        return self.einv.cpu()
+
+
 def dilu_setup(A, coloring):
-    return _core.dilu_setup(A.row_offsets, A.col_indices, A.values,
+    einv = _core.dilu_setup(A.row_offsets, A.col_indices, A.values,
                             A.block_dim, _didx(A), _tidx(A), coloring.colors,
                             coloring.rows_sorted, coloring.bounds)
+    ro_s, ci_s, pos, perm = _color_sorted_struct(A, coloring)
+    bb = A.block_dim * A.block_dim
+    va_s = A.values.reshape(A.nnz, -1)[pos].reshape(-1).contiguous()
+    einv_s = einv.reshape(A.n_rows, bb)[perm].reshape(-1).contiguous() \
+        if A.block_dim > 1 else einv[perm].contiguous()
+    return DiluState(einv, va_s, einv_s)
 
 
 def dilu_solve(A, Einv, coloring, r, relaxation, x):
     n = A.n_cols * A.block_dim   # ext size: halo tails stay zero in the sweeps
     w = _scratch(A, "dilu_w", n, r.dtype)   # vector precision (dDFI mixed)
     z = _scratch(A, "dilu_z", n, r.dtype)
+    if isinstance(Einv, DiluState):
+        ro_s, ci_s, pos, perm = _color_sorted_struct(A, coloring)
+        _core.dilu_apply_sorted(ro_s, ci_s, Einv.va_s, A.block_dim,
+                                Einv.einv_s, coloring.rows_sorted,
+                                coloring.bounds, r.reshape(-1), w, z,
+                                x.reshape(-1), float(relaxation))
+        return x
     _core.dilu_apply(A.row_offsets, A.col_indices, A.values, A.block_dim,
                      Einv, coloring.colors, coloring.rows_sorted,
                      coloring.bounds, r.reshape(-1), w, z, x.reshape(-1),
