@@ -45,6 +45,12 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
                       int** big_rows_out, int* n_big_out, hipStream_t s);
 void free_device_buf(void* p, hipStream_t s);
 
+// ---- color-sorted GS sweep (reorder-by-color layout) ------------------------
+template <typename TA, typename TV>
+void gs_rows_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                    const TA* dinv_s, const TV* bvec, TV* x, const int* rows,
+                    int count, TV omega, hipStream_t s);
+
 // ---- color-sorted DILU sweeps (reorder-by-color layout) ---------------------
 // ro_s/rows/einv_s pre-offset to the color base; matrix arrays are the
 // rows_sorted-gathered copy so each color reads one contiguous slab.
