@@ -9,6 +9,9 @@ cd /tmp && export TMPDIR=/tmp && cd "$GRAFT_REPO_ROOT" || cd /root/repo
 OUT="$GRAFT_REPO_ROOT/gpurun_out/prof"
 mkdir -p "$OUT"
 
+# -1. MFMA fragment-layout discovery (one-time decode for kernels_mfma.hip)
+timeout 300 python profiles/mfma_discover.py > "$OUT"/mfma_layout.txt 2>&1
+
 # 0. correctness first: GPU suite + smoke (no -x: show ALL failures)
 timeout 900 python -m pytest tests -m gpu -q > "$OUT"/pytest_gpu.log 2>&1
 timeout 300 python -c "import __graft_entry__ as g; g.smoke()" \
@@ -23,7 +26,7 @@ timeout 600 python bench.py --gpus 1 --steps 2 --warmup 1 --config block4_dilu \
     > "$OUT"/bench_block4.json 2> "$OUT"/bench_block4.err
 
 # 2. kernel-time table (trace+stats only; NO --pmc here) — keep stats only
-cd /tmp && timeout 900 rocprofv3 --kernel-trace --stats -d "$OUT"/trace -- \
+cd /tmp && timeout 900 rocprofv3 --kernel-trace --stats --output-format csv -d "$OUT"/trace -- \
     python "$GRAFT_REPO_ROOT"/bench.py --gpus 1 --steps 1 --warmup 1 --size 192 \
     > "$OUT"/rocprof_stats.log 2>&1
 # prune: keep only *stats* csvs (the raw kernel trace is huge)
