@@ -40,8 +40,8 @@ timeout 600 python bench_kernels.py --size 192 --iters 20 \
 
 # 3. PMC counters in their OWN run (pool rule: never combined with traces)
 cd /tmp
-timeout 600 rocprofv3 --pmc FETCH_SIZE WRITE_SIZE SQ_INSTS_MFMA \
-    -d "$OUT"/pmc -- \
+timeout 600 rocprofv3 --pmc FETCH_SIZE SQ_INSTS_MFMA \
+    --output-format csv -d "$OUT"/pmc -- \
     python "$GRAFT_REPO_ROOT"/bench.py --gpus 1 --steps 1 --warmup 0 --size 128 \
     > "$OUT"/rocprof_pmc.log 2>&1
 # summarize per-kernel counter totals, then drop the raw csv
@@ -67,7 +67,7 @@ rm -rf "$OUT"/pmc
 
 # 3b. PMC on the block4 DILU config: SQ_INSTS_MFMA must be > 0 now
 timeout 600 rocprofv3 --pmc SQ_INSTS_MFMA SQ_INSTS_VALU \
-    -d "$OUT"/pmc4 -- \
+    --output-format csv -d "$OUT"/pmc4 -- \
     python "$GRAFT_REPO_ROOT"/bench.py --gpus 1 --steps 1 --warmup 0 \
     --config block4_dilu --size 128 > "$OUT"/rocprof_pmc4.log 2>&1
 python - "$OUT"/pmc4 > "$OUT"/pmc4_summary.txt 2>&1 <<'EOF'
