@@ -438,6 +438,45 @@ def test_classical_pcg_full_gpu():
     assert st.converged and rel < 1e-7
 
 
+def test_d2_setup_stays_on_device():
+    """Device-resident D2 interpolation (VERDICT r01 item 7 done-criterion):
+    audit that building P from a device matrix performs no device->host
+    tensor transfer (monkeypatched Tensor.cpu/.numpy counters)."""
+    from amgx_amd.amg.classical import (SELECTOR_REGISTRY, STRENGTH_REGISTRY,
+                                        INTERP_REGISTRY)
+    from amgx_amd.config import ConfigScope
+    A = to_gpu(poisson_3d(10, 10, 10))
+    scope = ConfigScope(None, {"strength_threshold": 0.25})
+    S = STRENGTH_REGISTRY["AHAT"](A, scope)
+    cf, nc = SELECTOR_REGISTRY["PMIS"](A, S, scope)
+    transfers = []
+    orig_cpu = torch.Tensor.cpu
+
+    def audit_cpu(self, *a, **k):
+        if self.is_cuda:
+            transfers.append(self.shape)
+        return orig_cpu(self, *a, **k)
+
+    torch.Tensor.cpu = audit_cpu
+    try:
+        P = INTERP_REGISTRY["D2"](A, S, cf, nc, scope)
+    finally:
+        torch.Tensor.cpu = orig_cpu
+    assert P.values.is_cuda
+    assert not transfers, f"D2 setup pulled tensors to host: {transfers[:5]}"
+    # MULTIPASS: item-count syncs are fine, no bulk transfers
+    cf2, nc2 = SELECTOR_REGISTRY["AGGRESSIVE_PMIS"](A, S, scope)
+    transfers.clear()
+    torch.Tensor.cpu = audit_cpu
+    try:
+        P2 = INTERP_REGISTRY["MULTIPASS"](A, S, cf2, nc2, scope)
+    finally:
+        torch.Tensor.cpu = orig_cpu
+    assert P2.values.is_cuda
+    big = [s for s in transfers if len(s) and int(torch.tensor(s).prod()) > 4]
+    assert not big, f"MULTIPASS setup pulled tensors to host: {big[:5]}"
+
+
 def test_classical_d2_and_aggressive_gpu():
     """Host-pass components (D2 interp, aggressive PMIS) drive a GPU solve:
     tensors must land back on device and converge."""
