@@ -22,6 +22,7 @@
 #include <string>
 
 #include "../include/amgx_c.h"
+#include "../include/amgx_eig_c.h"
 
 namespace {
 
@@ -1092,6 +1093,46 @@ AMGX_RC AMGX_generate_distributed_poisson_7pt(
                                      obj(sol), allocated_halo_depth,
                                      num_import_rings, nx, ny, nz, px, py,
                                      pz);
+}
+
+/* ---------------------------------------------------------- eigensolvers */
+AMGX_RC AMGX_eigensolver_create(AMGX_eigensolver_handle *ret,
+                                AMGX_resources_handle rsc, AMGX_Mode mode,
+                                const AMGX_config_handle cfg) {
+    const char *ms = mode_str(mode);
+    if (!ms) return AMGX_RC_BAD_MODE;
+    return create_genericv("AMGX_eigensolver_create", (void **)ret,
+                           "eigensolver_create", "(OsO)", obj(rsc), ms,
+                           obj(cfg));
+}
+
+AMGX_RC AMGX_eigensolver_setup(AMGX_eigensolver_handle s,
+                               AMGX_matrix_handle mtx) {
+    return simple_callv("AMGX_eigensolver_setup", "eigensolver_setup",
+                        "(OO)", obj(s), obj(mtx));
+}
+
+AMGX_RC AMGX_eigensolver_pagerank_setup(AMGX_eigensolver_handle s,
+                                        AMGX_vector_handle a) {
+    return simple_callv("AMGX_eigensolver_pagerank_setup",
+                        "eigensolver_pagerank_setup", "(OO)", obj(s),
+                        obj(a));
+}
+
+AMGX_RC AMGX_eigensolver_solve(AMGX_eigensolver_handle s,
+                               AMGX_vector_handle x) {
+    return simple_callv("AMGX_eigensolver_solve", "eigensolver_solve",
+                        "(OO)", obj(s), x ? obj(x) : Py_None);
+}
+
+AMGX_RC AMGX_eigensolver_destroy(AMGX_eigensolver_handle s) {
+    if (!s) return AMGX_RC_BAD_PARAMETERS;
+    Gil gil;
+    AMGX_RC rc = unpack_rc(call_capi("AMGX_eigensolver_destroy",
+                                     Py_BuildValue("(O)", obj(s))),
+                           nullptr, 0, "eigensolver_destroy");
+    Py_DECREF(obj(s));
+    return rc;
 }
 
 AMGX_RC AMGX_write_parameters_description(char *filename,
