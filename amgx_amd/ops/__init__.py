@@ -152,10 +152,16 @@ def gs_sweep(A, dinv, b, x, coloring, omega: float, symmetric: bool = False):
 
 
 # ---------------------------------------------------------------------- coloring
-def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0):
+def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0,
+                 multihash_rounds: int = 0):
     """Distance-1 greedy-min-max coloring. Returns (colors int32 (n,), num_colors).
-    (reference src/matrix_coloring/min_max.cu)"""
-    return _backend(A).color_matrix(A, max_uncolored_frac, seed)
+    multihash_rounds > 0 prepends MULTI_HASH-semantics rounds (device path).
+    (reference src/matrix_coloring/min_max.cu, multi_hash.cu)"""
+    B = _backend(A)
+    if multihash_rounds:
+        return B.color_matrix(A, max_uncolored_frac, seed,
+                              multihash_rounds=multihash_rounds)
+    return B.color_matrix(A, max_uncolored_frac, seed)
 
 
 # ---------------------------------------------------------------------- aggregation
