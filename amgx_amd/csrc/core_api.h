@@ -79,6 +79,14 @@ template <typename TA, typename TV>
 void dilu_bwd_b4(const int* ro, const int* ci, const TA* va, const TA* einv,
                  const int* rows, int count, const TV* w, TV* z,
                  hipStream_t s);
+template <typename TA, typename TV>
+void dilu_fwd_b4_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                        const TA* einv_s, const int* rows, int count,
+                        const TV* r, TV* w, hipStream_t s);
+template <typename TA, typename TV>
+void dilu_bwd_b4_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                        const TA* einv_s, const int* rows, int count,
+                        const TV* w, TV* z, hipStream_t s);
 template <typename TA>
 void dilu_setup_b4(const int* ro, const int* ci, const TA* va,
                    const int* didx, const int* tidx, const int* colors,

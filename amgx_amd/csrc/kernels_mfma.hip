@@ -4,22 +4,23 @@
 // src/multiply.cu:952-1143 — redesigned for 64-lane waves + CDNA4 MFMA,
 // not translated).
 //
-// Geometry (b = 4): one wave = 4 row-groups x 16 lanes.  Within a group,
-// lane t (= threadIdx.x % 16) owns element (r = t/4, q = t%4) of the
-// current 4x4 block, so `va[k*16 + t]` is a fully coalesced 16-lane load of
-// one block.  Row sums reduce over q with two quad shuffles (every lane of
-// the quad ends up holding the full component sum).  The dense 4x4 algebra
-// (Einv apply, and the setup's Aij*Einvj*Aji triple products) runs on
-// `v_mfma_f64_4x4x4_4b_f64`: 4 independent 4x4x4 products per wave
-// instruction = exactly one row-group per MFMA block.
+// v_mfma_f64_4x4x4f64 fragment layout, decoded ON HARDWARE by the 64x64
+// one-hot probe (profiles/mfma_discover.py, gpurun_out/mfma_layout_full.txt)
+// — the 4 blocks are interleaved by quads, NOT contiguous 16-lane groups:
+//   A[b][m][k] at lane 16k + 4b + m
+//   B[b][k][n] at lane 16k + 4b + n
+//   C[b][m][n] at lane 16m + 4b + n
+// Lane roles used below: b = (lane>>2)&3 (block = row-group / nnz-slot),
+// o = lane>>4 (the k index of A/B, the m index of C), q = lane&3 (the m
+// index of A, the n index of B/C).
 //
-// Fragment layout (validated on hardware by mfma4_probe / test_gpu.py):
-// per 16-lane block with t = lane % 16:
-//   A[m][k]: m = t % 4, k = t / 4      (so A row-major elem = A[(t%4)*4+t/4])
-//   B[k][n]: n = t % 4, k = t / 4      (so B row-major elem = B[t])
-//   C[m][n]: m = t % 4, n = t / 4
-// Note A and C share the same (t%4 indexes m, t/4 indexes the other axis)
-// mapping, so a product C can chain directly into the next MFMA's A operand.
+// Geometry (b = 4): one wave covers 4 block-rows (one per MFMA block).
+// Each row's 16 lanes {16o + 4b + q} load its 4x4 values va[k*16 + o*4+q]
+// (element (o,q), a fully coalesced 16-lane 128B segment per nnz); row
+// sums reduce over q with two quad shuffles (every lane of the quad ends
+// holding the component-o total = exactly the B-fragment layout), and the
+// dense 4x4 algebra (Einv apply; the setup's Aij*Einvj*Aji) runs on the
+// matrix pipe — one 4-block MFMA per step.
 
 #include "common.h"
 #include "core_api.h"
@@ -42,8 +43,7 @@ __device__ __forceinline__ double quad_sum(double v) {
 }  // namespace
 
 // ---------------------------------------------------------------- probe
-// Writes the MFMA result for caller-supplied fragments so the Python-side
-// GPU test can validate the layout assumptions above against a CPU GEMM.
+// Raw MFMA for the layout-validation GPU test / discovery script.
 __global__ void mfma4_probe_kernel(const double* __restrict__ a_frag,
                                    const double* __restrict__ b_frag,
                                    double* __restrict__ c_frag) {
@@ -59,78 +59,90 @@ void mfma4_probe(const double* a, const double* b, double* c, hipStream_t s) {
 // ---------------------------------------------------------------- DILU b=4
 // Forward sweep, one color: w_i = Einv_i (r_i - sum_{color(j)<c} A_ij w_j).
 // w is pre-zeroed outside so the full-row sum only picks up earlier colors.
-template <typename TA, typename TV>
+// SORTED=1 reads the color-sorted matrix copy (ro_s/einv_s indexed by slot,
+// pointers pre-offset to the color base).
+template <typename TA, typename TV, int SORTED>
 __global__ __launch_bounds__(256) void dilu_fwd_b4_kernel(
     const int* __restrict__ ro, const int* __restrict__ ci,
     const TA* __restrict__ va, const TA* __restrict__ einv,
     const int* __restrict__ rows, int count, const TV* __restrict__ r,
     TV* __restrict__ w) {
-    int t = threadIdx.x & 15;            // lane in group
-    int slot = blockIdx.x * 16 + (threadIdx.x >> 4);
+    int lane = threadIdx.x & 63;
+    int b = (lane >> 2) & 3;             // MFMA block = row within wave
+    int o = lane >> 4;                   // k index of A/B frags
+    int q = lane & 3;                    // m of A frag / n of B,C frags
+    int slot = blockIdx.x * 16 + (threadIdx.x >> 6) * 4 + b;
     bool valid = slot < count;
     int i = valid ? rows[slot] : 0;
-    int rr = t >> 2;                     // block row  (va elem = rr*4+q)
-    int q = t & 3;                       // block col
     double acc = 0.0;
+    long long ebase;
     if (valid) {
-        int k0 = ro[i], k1 = ro[i + 1];
+        int k0, k1;
+        if (SORTED) { k0 = ro[slot]; k1 = ro[slot + 1];
+                      ebase = (long long)slot * 16; }
+        else { k0 = ro[i]; k1 = ro[i + 1]; ebase = (long long)i * 16; }
         for (int k = k0; k < k1; ++k) {
             int j = ci[k];
             if (j == i) continue;
-            double a = (double)va[(long long)k * 16 + t];   // blk[rr][q]
+            // element (row o, col q) of the 4x4 block — coalesced 16 lanes
+            double a = (double)va[(long long)k * 16 + o * 4 + q];
             double x = (double)w[(long long)j * 4 + q];
             acc += a * x;
         }
     }
-    // component sum for row rr lands in every lane of the quad
+    // component-o row sum lands in every lane of the quad = B frag layout
     acc = quad_sum(acc);
-    // B fragment: B[k][n] = rhs[k] broadcast across n; k = t/4 = rr
-    double bfrag = valid ? ((double)r[(long long)i * 4 + rr] - acc) : 0.0;
-    // A fragment: Einv[m][k] with m = t%4, k = t/4 -> row-major [q*4 + rr]
-    double afrag = valid ? (double)einv[(long long)i * 16 + q * 4 + rr] : 0.0;
-    double c = mfma4x4_4b(afrag, bfrag, 0.0);   // C[m][n], m=t%4, n=t/4
-    if (valid && rr == 0) w[(long long)i * 4 + q] = (TV)c;   // column n=0
+    double bfrag = valid ? ((double)r[(long long)i * 4 + o] - acc) : 0.0;
+    // A frag: Einv[m=q][k=o] -> row-major elem q*4 + o
+    double afrag = valid ? (double)einv[ebase + q * 4 + o] : 0.0;
+    double c = mfma4x4_4b(afrag, bfrag, 0.0);   // C[m=o][n=q] at this lane
+    if (valid && q == 0) w[(long long)i * 4 + o] = (TV)c;   // column n=0
 }
 
 // Backward sweep, one color: z_i = w_i - Einv_i sum_{color(j)>c} A_ij z_j.
-template <typename TA, typename TV>
+template <typename TA, typename TV, int SORTED>
 __global__ __launch_bounds__(256) void dilu_bwd_b4_kernel(
     const int* __restrict__ ro, const int* __restrict__ ci,
     const TA* __restrict__ va, const TA* __restrict__ einv,
-    const int* __restrict__ rows, int count, const TV* __restrict__ w,
+    const int* __restrict__ rows, int count, const TV* __restrict__ wv,
     TV* __restrict__ z) {
-    int t = threadIdx.x & 15;
-    int slot = blockIdx.x * 16 + (threadIdx.x >> 4);
+    int lane = threadIdx.x & 63;
+    int b = (lane >> 2) & 3;
+    int o = lane >> 4;
+    int q = lane & 3;
+    int slot = blockIdx.x * 16 + (threadIdx.x >> 6) * 4 + b;
     bool valid = slot < count;
     int i = valid ? rows[slot] : 0;
-    int rr = t >> 2;
-    int q = t & 3;
     double acc = 0.0;
+    long long ebase;
     if (valid) {
-        int k0 = ro[i], k1 = ro[i + 1];
+        int k0, k1;
+        if (SORTED) { k0 = ro[slot]; k1 = ro[slot + 1];
+                      ebase = (long long)slot * 16; }
+        else { k0 = ro[i]; k1 = ro[i + 1]; ebase = (long long)i * 16; }
         for (int k = k0; k < k1; ++k) {
             int j = ci[k];
             if (j == i) continue;
-            double a = (double)va[(long long)k * 16 + t];
+            double a = (double)va[(long long)k * 16 + o * 4 + q];
             double x = (double)z[(long long)j * 4 + q];
             acc += a * x;
         }
     }
     acc = quad_sum(acc);
     double bfrag = valid ? acc : 0.0;
-    double afrag = valid ? (double)einv[(long long)i * 16 + q * 4 + rr] : 0.0;
+    double afrag = valid ? (double)einv[ebase + q * 4 + o] : 0.0;
     double c = mfma4x4_4b(afrag, bfrag, 0.0);
-    if (valid && rr == 0)
-        z[(long long)i * 4 + q] =
-            w[(long long)i * 4 + q] - (TV)c;
+    if (valid && q == 0)
+        z[(long long)i * 4 + o] = wv[(long long)i * 4 + o] - (TV)c;
 }
 
 // DILU setup, one color: E_i = D_i - sum_{color(j)<c} A_ij Einv_j A_ji,
 // then einv_i = E_i^{-1}.  One wave per row; the wave walks the row 4 nnz
-// at a time (one nnz per 16-lane MFMA block) and chains two MFMAs per
-// chunk: T = Aij x Einvj, contrib = T x Aji.  Cross-group reduction folds
-// the 4 blocks' contributions, then lane groups cooperate on the 4x4
-// inversion through LDS.
+// at a time (one nnz per MFMA block b) and chains two MFMAs per chunk:
+// T = Aij x Einvj, contrib = T x Aji (the chain needs one transpose
+// shuffle: C layout has (m outer, n inner), the next A operand needs
+// (k outer, m inner)).  Cross-block xor-folds (masks 4, 8) sum the 4
+// chunks, then lane 0 inverts E through LDS.
 template <typename TA>
 __global__ __launch_bounds__(256) void dilu_setup_b4_kernel(
     const int* __restrict__ ro, const int* __restrict__ ci,
@@ -141,16 +153,14 @@ __global__ __launch_bounds__(256) void dilu_setup_b4_kernel(
     __shared__ double Es[4][16];         // one 4x4 E per wave
     __shared__ double Is[4][16];
     int lane = threadIdx.x & 63;
-    int wave = threadIdx.x >> 6;         // 4 waves per WG
+    int wave = threadIdx.x >> 6;         // 4 waves per WG, one row each
     int slot = blockIdx.x * 4 + wave;
     bool valid = slot < count;
     int i = valid ? rows[slot] : 0;
-    int g = lane >> 4;                   // nnz sub-slot 0..3
-    int t = lane & 15;
-    int q = t & 3;                       // within-block col index helpers
-    int rr = t >> 2;
+    int g = (lane >> 2) & 3;             // MFMA block = nnz sub-slot 0..3
+    int o = lane >> 4;                   // outer field
+    int q = lane & 3;                    // inner field
     int dk = valid ? didx[i] : -1;
-    // E starts as the diagonal block (C layout: m=t%4, n=t/4)
     double e_acc = 0.0;
     if (valid) {
         int k0 = ro[i], k1 = ro[i + 1];
@@ -160,27 +170,30 @@ __global__ __launch_bounds__(256) void dilu_setup_b4_kernel(
             int j = act ? ci[k] : i;
             int tk = act ? tidx[k] : -1;
             bool use = act && j != i && tk >= 0 && colors[j] < color;
-            // T = Aij x Einvj : A frag = Aij[m][k] = blk[(t%4)*4 + t/4],
-            // B frag = Einvj[k][n] = blk[t]
-            double a1 = use ? (double)va[(long long)k * 16 + q * 4 + rr]
+            // T = Aij x Einvj: A frag elem (m=q, k=o) -> Aij[q][o];
+            //                  B frag elem (k=o, n=q) -> Einvj[o][q]
+            double a1 = use ? (double)va[(long long)k * 16 + q * 4 + o]
                             : 0.0;
-            double b1 = use ? (double)einv[(long long)j * 16 + t] : 0.0;
-            double Tc = mfma4x4_4b(a1, b1, 0.0);   // C layout == A layout
-            // contrib = T x Aji : B frag = Aji[k][n] = blk[t]
-            double b2 = use ? (double)va[(long long)tk * 16 + t] : 0.0;
-            e_acc = mfma4x4_4b(Tc, b2, e_acc);
+            double b1 = use ? (double)einv[(long long)j * 16 + o * 4 + q]
+                            : 0.0;
+            double Tc = mfma4x4_4b(a1, b1, 0.0);  // T[m=o][n=q] here
+            // contrib = T x Aji: next A operand needs T[m=q][k=o], which
+            // lives at lane 16q + 4g + o -> transpose shuffle
+            double a2 = __shfl(Tc, 16 * q + 4 * g + o, 64);
+            double b2 = use ? (double)va[(long long)tk * 16 + o * 4 + q]
+                            : 0.0;
+            e_acc = mfma4x4_4b(a2, b2, e_acc);    // C[m=o][n=q]
         }
     }
-    // fold the 4 sub-slots: every lane t accumulates across groups
-    e_acc += __shfl_xor(e_acc, 16, 64);
-    e_acc += __shfl_xor(e_acc, 32, 64);
-    // E = D - sum  (convert C layout (m=t%4,n=t/4) to row-major m*4+n)
+    // fold the 4 nnz sub-slots: blocks live in the 4b lane field
+    e_acc += __shfl_xor(e_acc, 4, 64);
+    e_acc += __shfl_xor(e_acc, 8, 64);
+    // E = D - sum; element here is (m=o, n=q) -> row-major o*4 + q
     if (g == 0) {
         double d = (valid && dk >= 0)
-                       ? (double)va[(long long)dk * 16 + q * 4 + rr]
-                       : (valid && q == rr ? 1.0 : 0.0);
-        double e = d - e_acc;
-        Es[wave][q * 4 + rr] = e;        // row-major in LDS
+                       ? (double)va[(long long)dk * 16 + o * 4 + q]
+                       : (valid && o == q ? 1.0 : 0.0);
+        Es[wave][o * 4 + q] = d - e_acc;
     }
     __syncthreads();
     // stabilized fallback + inversion: lane 0 of each wave (setup-time)
@@ -200,8 +213,8 @@ __global__ __launch_bounds__(256) void dilu_setup_b4_kernel(
         small_mat_inv(Es[wave], Is[wave], 4);
     }
     __syncthreads();
-    if (valid && g == 0)
-        einv[(long long)i * 16 + t] = (TA)Is[wave][t];
+    if (valid && lane < 16)
+        einv[(long long)i * 16 + lane] = (TA)Is[wave][lane];
 }
 
 template <typename TA, typename TV>
@@ -209,7 +222,7 @@ void dilu_fwd_b4(const int* ro, const int* ci, const TA* va, const TA* einv,
                  const int* rows, int count, const TV* r, TV* w,
                  hipStream_t s) {
     if (count <= 0) return;
-    hipLaunchKernelGGL((dilu_fwd_b4_kernel<TA, TV>),
+    hipLaunchKernelGGL((dilu_fwd_b4_kernel<TA, TV, 0>),
                        dim3((count + 15) / 16), dim3(256), 0, s, ro, ci, va,
                        einv, rows, count, r, w);
 }
@@ -219,9 +232,30 @@ void dilu_bwd_b4(const int* ro, const int* ci, const TA* va, const TA* einv,
                  const int* rows, int count, const TV* w, TV* z,
                  hipStream_t s) {
     if (count <= 0) return;
-    hipLaunchKernelGGL((dilu_bwd_b4_kernel<TA, TV>),
+    hipLaunchKernelGGL((dilu_bwd_b4_kernel<TA, TV, 0>),
                        dim3((count + 15) / 16), dim3(256), 0, s, ro, ci, va,
                        einv, rows, count, w, z);
+}
+
+// color-sorted variants (reorder-by-color slabs; see kernels_solve.hip)
+template <typename TA, typename TV>
+void dilu_fwd_b4_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                        const TA* einv_s, const int* rows, int count,
+                        const TV* r, TV* w, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((dilu_fwd_b4_kernel<TA, TV, 1>),
+                       dim3((count + 15) / 16), dim3(256), 0, s, ro_s, ci_s,
+                       va_s, einv_s, rows, count, r, w);
+}
+
+template <typename TA, typename TV>
+void dilu_bwd_b4_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
+                        const TA* einv_s, const int* rows, int count,
+                        const TV* w, TV* z, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((dilu_bwd_b4_kernel<TA, TV, 1>),
+                       dim3((count + 15) / 16), dim3(256), 0, s, ro_s, ci_s,
+                       va_s, einv_s, rows, count, w, z);
 }
 
 template <typename TA>
@@ -236,29 +270,31 @@ void dilu_setup_b4(const int* ro, const int* ci, const TA* va,
 }
 
 // ---------------------------------------------------------------- bsrmv b=4
-// y = alpha * A x + beta * y (block-4 rows; same wave geometry and
-// coalesced block loads as the DILU sweeps).
+// y = alpha * A x + beta * y (block-4 rows; wave covers 4 rows with the
+// same coalesced element-(o,q) block loads; no MFMA needed — the row sums
+// are quad reductions and the result is written directly).
 template <typename TA, typename TV>
 __global__ __launch_bounds__(256) void bsrmv_b4_kernel(
     const int* __restrict__ ro, const int* __restrict__ ci,
     const TA* __restrict__ va, const TV* __restrict__ x,
     TV* __restrict__ y, const TV* __restrict__ bvec, double alpha,
     double beta, double gamma, int row_begin, int row_end) {
-    int t = threadIdx.x & 15;
-    int i = row_begin + blockIdx.x * 16 + (threadIdx.x >> 4);
+    int lane = threadIdx.x & 63;
+    int b = (lane >> 2) & 3;
+    int o = lane >> 4;
+    int q = lane & 3;
+    int i = row_begin + blockIdx.x * 16 + (threadIdx.x >> 6) * 4 + b;
     if (i >= row_end) return;
-    int rr = t >> 2;
-    int q = t & 3;
     double acc = 0.0;
     int k0 = ro[i], k1 = ro[i + 1];
     for (int k = k0; k < k1; ++k) {
-        double a = (double)va[(long long)k * 16 + t];
+        double a = (double)va[(long long)k * 16 + o * 4 + q];
         double xv = (double)x[(long long)ci[k] * 4 + q];
         acc += a * xv;
     }
     acc = quad_sum(acc);
     if (q == 0) {
-        long long idx = (long long)i * 4 + rr;
+        long long idx = (long long)i * 4 + o;
         double out = alpha * acc;
         if (beta != 0.0) out += beta * (double)y[idx];
         if (gamma != 0.0 && bvec) out += gamma * (double)bvec[idx];
@@ -285,6 +321,14 @@ void bsrmv_b4(const int* ro, const int* ci, const TA* va, const TV* x,
     template void dilu_bwd_b4<TA, TV>(const int*, const int*, const TA*,      \
                                       const TA*, const int*, int, const TV*,  \
                                       TV*, hipStream_t);                      \
+    template void dilu_fwd_b4_sorted<TA, TV>(const int*, const int*,          \
+                                             const TA*, const TA*,           \
+                                             const int*, int, const TV*,      \
+                                             TV*, hipStream_t);               \
+    template void dilu_bwd_b4_sorted<TA, TV>(const int*, const int*,          \
+                                             const TA*, const TA*,           \
+                                             const int*, int, const TV*,      \
+                                             TV*, hipStream_t);               \
     template void bsrmv_b4<TA, TV>(const int*, const int*, const TA*,         \
                                    const TV*, TV*, const TV*, double, double, \
                                    double, int, int, hipStream_t);

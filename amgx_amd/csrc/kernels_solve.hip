@@ -886,6 +886,9 @@ void dilu_fwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
         hipLaunchKernelGGL((dilu_fwd_scalar_sorted<TA, TV>),
                            dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
                            ro_s, ci_s, va_s, einv_s, rows, count, r, w);
+    else if (b == 4)       // MFMA wave kernel (kernels_mfma.hip)
+        dilu_fwd_b4_sorted<TA, TV>(ro_s, ci_s, va_s, einv_s, rows, count, r,
+                                   w, s);
     else
         hipLaunchKernelGGL((dilu_fwd_block_sorted<TA, TV>),
                            dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
@@ -901,6 +904,9 @@ void dilu_bwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
         hipLaunchKernelGGL((dilu_bwd_scalar_sorted<TA, TV>),
                            dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,
                            ro_s, ci_s, va_s, einv_s, rows, count, wv, z);
+    else if (b == 4)       // MFMA wave kernel (kernels_mfma.hip)
+        dilu_bwd_b4_sorted<TA, TV>(ro_s, ci_s, va_s, einv_s, rows, count,
+                                   wv, z, s);
     else
         hipLaunchKernelGGL((dilu_bwd_block_sorted<TA, TV>),
                            dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s,

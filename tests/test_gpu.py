@@ -35,10 +35,11 @@ def test_native_extension_loaded():
 
 
 def test_mfma4_fragment_layout():
-    """Validates the v_mfma_f64_4x4x4_4b_f64 fragment layout assumed by
-    kernels_mfma.hip: per 16-lane block t, A[m][k] at (m=t%4, k=t/4),
-    B[k][n] at (n=t%4, k=t/4), C[m][n] at (m=t%4, n=t/4).  On failure the
-    printed C fragment identifies the true permutation."""
+    """Validates the v_mfma_f64_4x4x4f64 fragment layout decoded on
+    hardware (profiles/mfma_discover.py 64x64 one-hot probe) and assumed by
+    kernels_mfma.hip: blocks interleaved by quads —
+    A[b][m][k] at lane 16k+4b+m, B[b][k][n] at 16k+4b+n,
+    C[b][m][n] at 16m+4b+n."""
     from amgx_amd import _core
     g = torch.Generator().manual_seed(7)
     Am = torch.rand(4, 4, 4, generator=g, dtype=torch.float64)  # per block
@@ -46,18 +47,19 @@ def test_mfma4_fragment_layout():
     a_frag = torch.empty(64, dtype=torch.float64)
     b_frag = torch.empty(64, dtype=torch.float64)
     for lane in range(64):
-        blk, t = lane // 16, lane % 16
-        a_frag[lane] = Am[blk][t % 4][t // 4]
-        b_frag[lane] = Bm[blk][t // 4][t % 4]
+        blk = (lane // 4) % 4
+        outer = lane // 16
+        inner = lane % 4
+        a_frag[lane] = Am[blk][inner][outer]   # A[b][m=inner][k=outer]
+        b_frag[lane] = Bm[blk][outer][inner]   # B[b][k=outer][n=inner]
     c_frag = _core.mfma4_probe(a_frag.cuda(), b_frag.cuda()).cpu()
     C = torch.einsum("bmk,bkn->bmn", Am, Bm)
     got = torch.empty_like(C)
     for lane in range(64):
-        blk, t = lane // 16, lane % 16
-        got[blk][t % 4][t // 4] = c_frag[lane]
+        blk = (lane // 4) % 4
+        got[blk][lane // 16][lane % 4] = c_frag[lane]   # C[b][m=o][n=inner]
     assert torch.allclose(got, C, atol=1e-12), \
-        f"MFMA layout mismatch:\nexpected {C[0]}\ngot {got[0]}\n" \
-        f"raw frag block0 {c_frag[:16].reshape(4, 4)}"
+        f"MFMA layout mismatch:\nexpected {C[0]}\ngot {got[0]}"
 
 
 def test_spmv_scalar():
