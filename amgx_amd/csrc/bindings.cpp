@@ -123,7 +123,7 @@ void axpy(Tensor y, Tensor x, double a) {
 std::vector<Tensor> spgemm_hash(Tensor roA, Tensor ciA, Tensor vaA,
                                 Tensor roB, Tensor ciB, Tensor vaB,
                                 c10::optional<Tensor> aggcol, int64_t mode,
-                                int64_t cap_nnz) {
+                                int64_t cap_nnz, int64_t cap0) {
     int m = (int)roA.numel() - 1;
     auto roC = torch::empty({m + 1}, roA.options());
     auto ciC = torch::empty({cap_nnz}, roA.options());
@@ -139,7 +139,7 @@ std::vector<Tensor> spgemm_hash(Tensor roA, Tensor ciA, Tensor vaA,
             vaA.data_ptr<scalar_t>(), m, roB.data_ptr<int>(),
             ciB.data_ptr<int>(), vaB.data_ptr<scalar_t>(),
             aggcol ? aggcol->data_ptr<int>() : nullptr, (int)mode,
-            roC.data_ptr<int>(), ciC.data_ptr<int>(),
+            (int)cap0, roC.data_ptr<int>(), ciC.data_ptr<int>(),
             vaC.data_ptr<scalar_t>(), (long long)cap_nnz, &big_rows,
             &n_big, st);
         if (n_big > 0) {

@@ -40,7 +40,7 @@ void bsrmv_generic(const int* ro, const int* ci, const TA* va, int b,
 template <typename T>
 long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
                       const int* roB, const int* ciB, const T* vaB,
-                      const int* aggcol, int mode, int* roC_out,
+                      const int* aggcol, int mode, int cap0, int* roC_out,
                       int* ciC_cap_buf, T* vaC_cap_buf, long long cap_nnz,
                       int** big_rows_out, int* n_big_out, hipStream_t s);
 void free_device_buf(void* p, hipStream_t s);

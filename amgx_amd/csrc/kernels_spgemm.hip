@@ -29,6 +29,8 @@ namespace amgx_hip {
 
 namespace {
 
+constexpr int TINY_CAP = 64;       // per-wave table, 8 waves per WG —
+                                   // right-sized for AMG rows (<=64 nnz)
 constexpr int SMALL_CAP = 512;     // per-wave table, 4 waves per WG
 constexpr int BIG_CAP = 8192;      // 1 wave per WG
 
@@ -148,9 +150,8 @@ __global__ __launch_bounds__(64 * WAVES) void spgemm_count_kernel(
     cnt = (int)wave_reduce_sum(cnt);
     ovf = __any(ovf);
     if (lane == 0) {
-        counts[i] = ovf ? 0 : cnt;
+        counts[i] = ovf ? -1 : cnt;     // -1 marks the row for a bigger tier
         if (ovf) atomicExch(overflow, 1);
-        if (ovf && row_list == nullptr) counts[i] = -1;   // mark for retry
     }
 }
 
@@ -233,10 +234,12 @@ __global__ __launch_bounds__(64 * WAVES) void spgemm_fill_kernel(
 // ============================================================ driver
 // Returns nnz(C) >= 0 on success, -1 when a row exceeded BIG_CAP (caller
 // falls back to the ESC sort path).  counts/roC share the (m+1) buffer.
+// cap0 chooses the first tier of the 64 -> 512 -> 8192 capacity ladder
+// (64 for AMG-shaped rows; rows past a tier re-run in the next).
 template <typename T>
 long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
                       const int* roB, const int* ciB, const T* vaB,
-                      const int* aggcol, int mode, int* roC_out,
+                      const int* aggcol, int mode, int cap0, int* roC_out,
                       int* ciC_cap_buf, T* vaC_cap_buf, long long cap_nnz,
                       int** big_rows_out, int* n_big_out, hipStream_t s) {
     int* counts = nullptr;
@@ -245,31 +248,58 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
     HIP_CHECK(hipMallocAsync(&ovf, sizeof(int), s));
     HIP_CHECK(hipMemsetAsync(ovf, 0, sizeof(int), s));
     HIP_CHECK(hipMemsetAsync(counts, 0, (m + 1) * sizeof(int), s));
-    constexpr int WAVES = 4;
-    int wg = (m + WAVES - 1) / WAVES;
-    hipLaunchKernelGGL((spgemm_count_kernel<SMALL_CAP, WAVES>), dim3(wg),
-                       dim3(64 * WAVES), 0, s, roA, ciA, roB, ciB, aggcol,
-                       mode, m, nullptr, m, counts, ovf);
+    bool tiny_first = cap0 <= TINY_CAP;
+    if (tiny_first) {
+        int wg = (m + 7) / 8;
+        hipLaunchKernelGGL((spgemm_count_kernel<TINY_CAP, 8>), dim3(wg),
+                           dim3(64 * 8), 0, s, roA, ciA, roB, ciB, aggcol,
+                           mode, m, nullptr, m, counts, ovf);
+    } else {
+        int wg = (m + 3) / 4;
+        hipLaunchKernelGGL((spgemm_count_kernel<SMALL_CAP, 4>), dim3(wg),
+                           dim3(64 * 4), 0, s, roA, ciA, roB, ciB, aggcol,
+                           mode, m, nullptr, m, counts, ovf);
+    }
     int h_ovf = 0;
     HIP_CHECK(hipMemcpyAsync(&h_ovf, ovf, sizeof(int), hipMemcpyDeviceToHost,
                              s));
     HIP_CHECK(hipStreamSynchronize(s));
-    // big rows: counts[i] == -1
-    int* big_rows = nullptr;
-    int n_big = 0;
-    if (h_ovf) {
-        // compact the list of overflowing rows on host (rare, setup-time)
+    // rows with counts[i] == -1 climb the ladder (rare, setup-time: the
+    // overflow-list compaction runs on host)
+    auto collect_marked = [&](int** rows_out) -> int {
         std::vector<int> h_counts(m);
         HIP_CHECK(hipMemcpyAsync(h_counts.data(), counts, m * sizeof(int),
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
-        std::vector<int> big;
+        std::vector<int> marked;
         for (int i = 0; i < m; ++i)
-            if (h_counts[i] == -1) big.push_back(i);
-        n_big = (int)big.size();
-        HIP_CHECK(hipMallocAsync(&big_rows, n_big * sizeof(int), s));
-        HIP_CHECK(hipMemcpyAsync(big_rows, big.data(), n_big * sizeof(int),
-                                 hipMemcpyHostToDevice, s));
+            if (h_counts[i] == -1) marked.push_back(i);
+        int n = (int)marked.size();
+        if (n) {
+            HIP_CHECK(hipMallocAsync(rows_out, n * sizeof(int), s));
+            HIP_CHECK(hipMemcpyAsync(*rows_out, marked.data(),
+                                     n * sizeof(int), hipMemcpyHostToDevice,
+                                     s));
+        }
+        return n;
+    };
+    int* mid_rows = nullptr;   // rows needing the 512 tier (only if tiny)
+    int n_mid = 0;
+    int* big_rows = nullptr;   // rows needing the 8192 tier
+    int n_big = 0;
+    if (h_ovf && tiny_first) {
+        n_mid = collect_marked(&mid_rows);
+        HIP_CHECK(hipMemsetAsync(ovf, 0, sizeof(int), s));
+        hipLaunchKernelGGL((spgemm_count_kernel<SMALL_CAP, 4>),
+                           dim3((n_mid + 3) / 4), dim3(64 * 4), 0, s, roA,
+                           ciA, roB, ciB, aggcol, mode, m, mid_rows, n_mid,
+                           counts, ovf);
+        HIP_CHECK(hipMemcpyAsync(&h_ovf, ovf, sizeof(int),
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+    }
+    if (h_ovf) {
+        n_big = collect_marked(&big_rows);
         HIP_CHECK(hipMemsetAsync(ovf, 0, sizeof(int), s));
         hipLaunchKernelGGL((spgemm_count_kernel<BIG_CAP, 1>), dim3(n_big),
                            dim3(64), 0, s, roA, ciA, roB, ciB, aggcol, mode,
@@ -278,6 +308,7 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
         if (h_ovf) {     // >BIG_CAP nnz in a row: give up -> ESC fallback
+            if (mid_rows) HIP_CHECK(hipFreeAsync(mid_rows, s));
             HIP_CHECK(hipFreeAsync(big_rows, s));
             HIP_CHECK(hipFreeAsync(counts, s));
             HIP_CHECK(hipFreeAsync(ovf, s));
@@ -303,15 +334,26 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
         if (big_rows) HIP_CHECK(hipFreeAsync(big_rows, s));
         return -2;   // caller must re-allocate and retry fill
     }
-    constexpr int FWAVES = 4;
-    int fwg = (m + FWAVES - 1) / FWAVES;
-    hipLaunchKernelGGL((spgemm_fill_kernel<T, SMALL_CAP, FWAVES, 1>),
-                       dim3(fwg), dim3(64 * FWAVES), 0, s, roA, ciA, vaA,
-                       roB, ciB, vaB, aggcol, mode, m, nullptr, m, roC_out,
-                       ciC_cap_buf, vaC_cap_buf);
-    // fill pass skipped the big rows?  No: the small-cap fill kernel would
-    // overflow its table on them.  Re-run those rows with the big kernel
-    // (their small-pass output slots are rewritten completely).
+    // fill tiers mirror the count tiers: each kernel's per-row guard skips
+    // rows bigger than its table, and the next tier rewrites them fully
+    if (tiny_first) {
+        hipLaunchKernelGGL((spgemm_fill_kernel<T, TINY_CAP, 8, 1>),
+                           dim3((m + 7) / 8), dim3(64 * 8), 0, s, roA, ciA,
+                           vaA, roB, ciB, vaB, aggcol, mode, m, nullptr, m,
+                           roC_out, ciC_cap_buf, vaC_cap_buf);
+        if (n_mid)
+            hipLaunchKernelGGL((spgemm_fill_kernel<T, SMALL_CAP, 4, 1>),
+                               dim3((n_mid + 3) / 4), dim3(64 * 4), 0, s,
+                               roA, ciA, vaA, roB, ciB, vaB, aggcol, mode,
+                               m, mid_rows, n_mid, roC_out, ciC_cap_buf,
+                               vaC_cap_buf);
+    } else {
+        hipLaunchKernelGGL((spgemm_fill_kernel<T, SMALL_CAP, 4, 1>),
+                           dim3((m + 3) / 4), dim3(64 * 4), 0, s, roA, ciA,
+                           vaA, roB, ciB, vaB, aggcol, mode, m, nullptr, m,
+                           roC_out, ciC_cap_buf, vaC_cap_buf);
+    }
+    if (mid_rows) HIP_CHECK(hipFreeAsync(mid_rows, s));
     if (n_big) {
         hipLaunchKernelGGL((spgemm_fill_kernel<T, BIG_CAP, 1, 0>),
                            dim3(n_big), dim3(64), 0, s, roA, ciA, vaA, roB,
@@ -320,6 +362,7 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
         *big_rows_out = big_rows;   // caller sorts these rows + frees
         *n_big_out = n_big;
     } else {
+        if (big_rows) HIP_CHECK(hipFreeAsync(big_rows, s));
         *big_rows_out = nullptr;
         *n_big_out = 0;
     }
@@ -329,7 +372,7 @@ long long spgemm_hash(const int* roA, const int* ciA, const T* vaA, int m,
 #define INSTANTIATE_SPGEMM_HASH(T)                                            \
     template long long spgemm_hash<T>(const int*, const int*, const T*, int, \
                                       const int*, const int*, const T*,      \
-                                      const int*, int, int*, int*, T*,       \
+                                      const int*, int, int, int*, int*, T*,  \
                                       long long, int**, int*, hipStream_t);
 
 INSTANTIATE_SPGEMM_HASH(double)

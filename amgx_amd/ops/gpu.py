@@ -310,7 +310,8 @@ def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
         ro, ci, va, big = _core.spgemm_hash(
             m_ro.to(torch.int32), fids, A.values,   # vaA unused in mode 1
             A.row_offsets, A.col_indices, A.values,
-            agg_col, 1, max(A.nnz, 1))
+            agg_col, 1, max(A.nnz, 1),
+            64 if A.nnz <= 48 * max(num_aggregates, 1) else 512)
         if int(ro[0].item()) != -1:
             if big.numel():
                 _sort_unsorted_rows(ro, ci, va, big)
@@ -365,10 +366,12 @@ def spgemm(A, B):
     pathological >8k-nnz rows."""
     from ..matrix import CSRMatrix
     cap = max(_expansion_bound(A, B), 1)
+    # first hash tier sized to the average output row (AMG rows are tiny)
+    cap0 = 64 if cap <= 48 * max(A.n_rows, 1) else 512
     ro, ci, va, big = _core.spgemm_hash(A.row_offsets, A.col_indices,
                                         A.values, B.row_offsets,
                                         B.col_indices, B.values, None, 0,
-                                        cap)
+                                        cap, cap0)
     if int(ro[0].item()) == -1:   # a row exceeded the big hash capacity
         ro, ci, va = _core.spgemm(A.row_offsets, A.col_indices, A.values,
                                   B.row_offsets, B.col_indices, B.values,
