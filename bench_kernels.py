@@ -89,7 +89,7 @@ def main():
 
     # block-4 kernels: MFMA wave path vs the scalar baseline
     from amgx_amd.problems import block_laplacian
-    side = max(args.size // 3, 24)
+    side = max(args.size, 64)          # side^2 block rows
     A4 = block_laplacian(side, side, block_dim=4, seed=3).to(dev)
     n4 = A4.n_rows
     x4 = torch.rand(n4 * 4, dtype=torch.float64, device=dev)
