@@ -305,6 +305,9 @@ class MatrixColoring:
         for c in counts.cpu().tolist():
             b.append(b[-1] + int(c))
         self.bounds = b
+        # device copy for single-launch fused sweeps (small AMG levels)
+        self.bounds_dev = torch.tensor(b, dtype=torch.int32,
+                                       device=colors.device)
 
     @classmethod
     def create(cls, A, scope=None) -> "MatrixColoring":

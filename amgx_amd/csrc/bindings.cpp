@@ -298,9 +298,22 @@ void gs_sweep(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor dinv,
 // color-sorted scalar GS sweep (reorder-by-color layout)
 void gs_sweep_sorted(Tensor ro_s, Tensor ci_s, Tensor va_s, Tensor dinv_s,
                      Tensor bvec, Tensor x, Tensor rows_sorted,
-                     std::vector<int64_t> bounds, double omega,
+                     std::vector<int64_t> bounds,
+                     c10::optional<Tensor> bounds_dev, double omega,
                      bool symmetric) {
     int nc = (int)bounds.size() - 1;
+    int64_t n = (int64_t)ro_s.numel() - 1;
+    if (n <= 16384 && bounds_dev.has_value()) {
+        DISPATCH_FT2(va_s, x, "gs_sweep_small", [&] {
+            amgx_hip::gs_sweep_small<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>(), ci_s.data_ptr<int>(),
+                va_s.data_ptr<scalar_a>(), dinv_s.data_ptr<scalar_a>(),
+                bvec.data_ptr<scalar_v>(), x.data_ptr<scalar_v>(),
+                rows_sorted.data_ptr<int>(), bounds_dev->data_ptr<int>(),
+                nc, (scalar_v)omega, symmetric, cur_stream());
+        });
+        return;
+    }
     DISPATCH_FT2(va_s, x, "gs_sweep_sorted", [&] {
         auto run = [&](int c) {
             int64_t s = bounds[c], e = bounds[c + 1];
@@ -381,10 +394,26 @@ void dilu_apply(Tensor ro, Tensor ci, Tensor va, int64_t b, Tensor einv,
 // contiguous slab per color — reference reorder-by-color layout)
 void dilu_apply_sorted(Tensor ro_s, Tensor ci_s, Tensor va_s, int64_t b,
                        Tensor einv_s, Tensor rows_sorted,
-                       std::vector<int64_t> bounds, Tensor r, Tensor w,
+                       std::vector<int64_t> bounds,
+                       c10::optional<Tensor> bounds_dev, Tensor r, Tensor w,
                        Tensor z, Tensor x, double relax) {
     int nc = (int)bounds.size() - 1;
     int bb = (int)(b * b);
+    int64_t n = (int64_t)ro_s.numel() - 1;
+    // small levels: ONE fused single-WG launch for the whole apply
+    // (zeroing + all colors both sweeps + relaxed axpy)
+    if (b == 1 && n <= 16384 && bounds_dev.has_value()) {
+        DISPATCH_FT2(va_s, x, "dilu_apply_small", [&] {
+            amgx_hip::dilu_apply_small<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>(), ci_s.data_ptr<int>(),
+                va_s.data_ptr<scalar_a>(), einv_s.data_ptr<scalar_a>(),
+                rows_sorted.data_ptr<int>(), bounds_dev->data_ptr<int>(),
+                nc, r.data_ptr<scalar_v>(), w.data_ptr<scalar_v>(),
+                z.data_ptr<scalar_v>(), x.data_ptr<scalar_v>(),
+                (scalar_v)relax, (long long)w.numel(), cur_stream());
+        });
+        return;
+    }
     w.zero_();
     z.zero_();
     DISPATCH_FT2(va_s, x, "dilu_apply_sorted", [&] {

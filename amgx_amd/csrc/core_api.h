@@ -47,6 +47,11 @@ void free_device_buf(void* p, hipStream_t s);
 
 // ---- color-sorted GS sweep (reorder-by-color layout) ------------------------
 template <typename TA, typename TV>
+void gs_sweep_small(const int* ro_s, const int* ci_s, const TA* va_s,
+                    const TA* dinv_s, const TV* bvec, TV* x, const int* rows,
+                    const int* bounds, int ncolors, TV omega, bool symmetric,
+                    hipStream_t s);
+template <typename TA, typename TV>
 void gs_rows_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
                     const TA* dinv_s, const TV* bvec, TV* x, const int* rows,
                     int count, TV omega, hipStream_t s);
@@ -54,6 +59,13 @@ void gs_rows_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
 // ---- color-sorted DILU sweeps (reorder-by-color layout) ---------------------
 // ro_s/rows/einv_s pre-offset to the color base; matrix arrays are the
 // rows_sorted-gathered copy so each color reads one contiguous slab.
+// fused whole-apply for small levels: one single-WG launch does zeroing,
+// all forward colors, all backward colors and the relaxed axpy
+template <typename TA, typename TV>
+void dilu_apply_small(const int* ro_s, const int* ci_s, const TA* va_s,
+                      const TA* einv_s, const int* rows, const int* bounds,
+                      int ncolors, const TV* r, TV* w, TV* z, TV* x,
+                      TV relax, long long vec_n, hipStream_t s);
 template <typename TA, typename TV>
 void dilu_fwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
                      const TA* einv_s, const int* rows, int count,

@@ -503,6 +503,44 @@ __global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_scalar_sorted(
     x[i] += omega * (TV)dinv_s[t] * (bvec[i] - sum);
 }
 
+// fused whole-sweep GS for small levels: one single-WG launch loops all
+// colors (and the descending pass when symmetric) with __syncthreads as
+// the color barrier — same launch-floor fix as dilu_apply_small
+template <typename TA, typename TV>
+__global__ __launch_bounds__(1024) void gs_sweep_small_kernel(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ dinv_s,
+    const TV* __restrict__ bvec, TV* __restrict__ x,
+    const int* __restrict__ rows, const int* __restrict__ bounds,
+    int ncolors, TV omega, int symmetric) {
+    int tid = threadIdx.x;
+    int nt = blockDim.x;
+    for (int pass = 0; pass < 1 + symmetric; ++pass) {
+        for (int cc = 0; cc < ncolors; ++cc) {
+            int c = pass == 0 ? cc : ncolors - 1 - cc;
+            int s0 = bounds[c], s1 = bounds[c + 1];
+            for (int t = s0 + tid; t < s1; t += nt) {
+                int i = rows[t];
+                TV sum = TV(0);
+                for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+                    sum += (TV)va_s[k] * x[ci_s[k]];
+                x[i] += omega * (TV)dinv_s[t] * (bvec[i] - sum);
+            }
+            __syncthreads();
+        }
+    }
+}
+
+template <typename TA, typename TV>
+void gs_sweep_small(const int* ro_s, const int* ci_s, const TA* va_s,
+                    const TA* dinv_s, const TV* bvec, TV* x, const int* rows,
+                    const int* bounds, int ncolors, TV omega, bool symmetric,
+                    hipStream_t s) {
+    hipLaunchKernelGGL((gs_sweep_small_kernel<TA, TV>), dim3(1), dim3(1024),
+                       0, s, ro_s, ci_s, va_s, dinv_s, bvec, x, rows, bounds,
+                       ncolors, omega, symmetric ? 1 : 0);
+}
+
 template <typename TA, typename TV>
 void gs_rows_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
                     const TA* dinv_s, const TV* bvec, TV* x, const int* rows,
@@ -877,6 +915,59 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_block_sorted(
     }
 }
 
+// Whole-apply fused kernel for SMALL levels (the deep AMG tail): ONE
+// single-workgroup launch runs zeroing, every forward color, every
+// backward color and the relaxed axpy, with __syncthreads as the color
+// barrier.  The per-color launch floor (~5-20us x ncolors x 2 sweeps x
+// sweeps-per-smooth x ~15 coarse levels) dominated the V-cycle before
+// this (rocprof r02: 95k launches, 2.0 of 3.0 s GPU time at 192^3).
+template <typename TA, typename TV>
+__global__ __launch_bounds__(1024) void dilu_apply_small_kernel(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, const int* __restrict__ bounds,
+    int ncolors, const TV* __restrict__ r, TV* __restrict__ w,
+    TV* __restrict__ z, TV* __restrict__ x, TV relax, long long vec_n) {
+    int tid = threadIdx.x;
+    int nt = blockDim.x;
+    for (long long i = tid; i < vec_n; i += nt) { w[i] = TV(0); z[i] = TV(0); }
+    __syncthreads();
+    for (int c = 0; c < ncolors; ++c) {
+        int s0 = bounds[c], s1 = bounds[c + 1];
+        for (int t = s0 + tid; t < s1; t += nt) {
+            int i = rows[t];
+            TV sum = TV(0);
+            for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+                sum += (TV)va_s[k] * w[ci_s[k]];
+            w[i] = (TV)einv_s[t] * (r[i] - sum);
+        }
+        __syncthreads();
+    }
+    for (int c = ncolors - 1; c >= 0; --c) {
+        int s0 = bounds[c], s1 = bounds[c + 1];
+        for (int t = s0 + tid; t < s1; t += nt) {
+            int i = rows[t];
+            TV sum = TV(0);
+            for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+                sum += (TV)va_s[k] * z[ci_s[k]];
+            z[i] = w[i] - (TV)einv_s[t] * sum;
+        }
+        __syncthreads();
+    }
+    long long xn = (long long)bounds[ncolors];
+    for (long long i = tid; i < xn; i += nt) x[i] += relax * z[i];
+}
+
+template <typename TA, typename TV>
+void dilu_apply_small(const int* ro_s, const int* ci_s, const TA* va_s,
+                      const TA* einv_s, const int* rows, const int* bounds,
+                      int ncolors, const TV* r, TV* w, TV* z, TV* x,
+                      TV relax, long long vec_n, hipStream_t s) {
+    hipLaunchKernelGGL((dilu_apply_small_kernel<TA, TV>), dim3(1),
+                       dim3(1024), 0, s, ro_s, ci_s, va_s, einv_s, rows,
+                       bounds, ncolors, r, w, z, x, relax, vec_n);
+}
+
 template <typename TA, typename TV>
 void dilu_fwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
                      const TA* einv_s, const int* rows, int count,
@@ -1144,12 +1235,21 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
     template void jacobi_smooth<TA, TV>(const int*, const int*, const TA*,      \
                                         const TA*, const TV*, const TV*, TV*,   \
                                         TV, int, int, double, hipStream_t);     \
+    template void dilu_apply_small<TA, TV>(const int*, const int*,              \
+                                           const TA*, const TA*, const int*,    \
+                                           const int*, int, const TV*, TV*,     \
+                                           TV*, TV*, TV, long long,             \
+                                           hipStream_t);                        \
     template void dilu_fwd_sorted<TA, TV>(const int*, const int*, const TA*,    \
                                           const TA*, const int*, int,           \
                                           const TV*, TV*, int, hipStream_t);    \
     template void dilu_bwd_sorted<TA, TV>(const int*, const int*, const TA*,    \
                                           const TA*, const int*, int,           \
                                           const TV*, TV*, int, hipStream_t);    \
+    template void gs_sweep_small<TA, TV>(const int*, const int*, const TA*,     \
+                                         const TA*, const TV*, TV*,             \
+                                         const int*, const int*, int, TV,       \
+                                         bool, hipStream_t);                    \
     template void gs_rows_sorted<TA, TV>(const int*, const int*, const TA*,     \
                                          const TA*, const TV*, TV*,             \
                                          const int*, int, TV, hipStream_t);     \

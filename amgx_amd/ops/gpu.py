@@ -176,8 +176,9 @@ def gs_sweep(A, dinv, b, x, coloring, omega: float, symmetric: bool = False):
             A._cache[dkey] = dinv_s
         _core.gs_sweep_sorted(ro_s, ci_s, va_s, dinv_s, b.reshape(-1),
                               x.reshape(-1), coloring.rows_sorted,
-                              coloring.bounds, float(omega),
-                              bool(symmetric))
+                              coloring.bounds,
+                              getattr(coloring, "bounds_dev", None),
+                              float(omega), bool(symmetric))
         return x
     _core.gs_sweep(A.row_offsets, A.col_indices, A.values, A.block_dim, dinv,
                    b.reshape(-1), x.reshape(-1), coloring.rows_sorted,
@@ -255,7 +256,9 @@ def dilu_solve(A, Einv, coloring, r, relaxation, x):
         ro_s, ci_s, pos, perm = _color_sorted_struct(A, coloring)
         _core.dilu_apply_sorted(ro_s, ci_s, Einv.va_s, A.block_dim,
                                 Einv.einv_s, coloring.rows_sorted,
-                                coloring.bounds, r.reshape(-1), w, z,
+                                coloring.bounds,
+                                getattr(coloring, "bounds_dev", None),
+                                r.reshape(-1), w, z,
                                 x.reshape(-1), float(relaxation))
         return x
     _core.dilu_apply(A.row_offsets, A.col_indices, A.values, A.block_dim,
