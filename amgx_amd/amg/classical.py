@@ -287,8 +287,9 @@ def _interp_d2_device(A, S, cf_map, num_coarse, scope):
     ro64 = A.row_offsets.to(torch.int64)
     ci = A.col_indices.to(torch.int64)
     v = A.values.reshape(A.nnz, -1)[:, 0].to(torch.float64)
-    strong = S.to(torch.bool).reshape(-1)
-    cf = cf_map.to(torch.int64)
+    # S / cf_map may come from host-side selectors: land them on dev
+    strong = torch.as_tensor(S).to(dev).to(torch.bool).reshape(-1)
+    cf = torch.as_tensor(cf_map).to(dev).to(torch.int64)
     deg = ro64[1:] - ro64[:-1]
     rows = torch.repeat_interleave(
         torch.arange(n, dtype=torch.int64, device=dev), deg)
@@ -454,8 +455,9 @@ def _interp_multipass_device(A, S, cf_map, num_coarse, scope,
     ro64 = A.row_offsets.to(torch.int64)
     ci = A.col_indices.to(torch.int64)
     v = A.values.reshape(A.nnz, -1)[:, 0].to(torch.float64)
-    strong = S.to(torch.bool).reshape(-1)
-    cf = cf_map.to(torch.int64)
+    # S / cf_map may come from host-side selectors: land them on dev
+    strong = torch.as_tensor(S).to(dev).to(torch.bool).reshape(-1)
+    cf = torch.as_tensor(cf_map).to(dev).to(torch.int64)
     deg = ro64[1:] - ro64[:-1]
     rows = torch.repeat_interleave(
         torch.arange(n, dtype=torch.int64, device=dev), deg)

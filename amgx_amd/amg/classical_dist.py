@@ -241,7 +241,7 @@ def rap_dist(A, mgr: DistributedManager, P_own,
                        and tdist.get_backend() == "nccl") else \
         torch.device("cpu")
     send_keys, send_vals = {}, {}
-    counts = torch.zeros(mgr.world, dtype=torch.int64)
+    counts = torch.zeros(mgr.world, dtype=torch.int64, device=comm_dev)
     for r in range(mgr.world):
         if r == mgr.rank:
             continue
@@ -252,9 +252,10 @@ def rap_dist(A, mgr: DistributedManager, P_own,
             send_keys[r] = (rr * ngc + cc).to(comm_dev)
             send_vals[r] = vv.to(torch.float64).to(comm_dev)
             counts[r] = rr.numel()
-    all_counts = [torch.zeros(mgr.world, dtype=torch.int64)
+    all_counts = [torch.zeros(mgr.world, dtype=torch.int64, device=comm_dev)
                   for _ in range(mgr.world)]
     tdist.all_gather(all_counts, counts)
+    all_counts = [t.cpu() for t in all_counts]
     p2p, recv = [], {}
     for r in range(mgr.world):
         if r == mgr.rank:
