@@ -523,13 +523,25 @@ def exchange_csr_rows(mgr: DistributedManager, indptr, indices, data):
     return out
 
 
-def halo_matrix(mgr: DistributedManager, A):
+def halo_matrix(mgr: DistributedManager, A, cache: bool = True):
     """Fetch the matrix rows backing this rank's halo columns (reference
     DistributedManager::createOneRingHaloRows, distributed_manager.cu:
     1542-1596 — the num_import_rings=2 structure): returns a CSR fragment
     (row_offsets, col_global, values) with one row per halo slot, columns as
     GLOBAL ids. The columns of these rows are the 2-ring; feeding them back
-    through another exchange extends the ring again."""
+    through another exchange extends the ring again.
+
+    The fragment is the manager's first-class ring-2 structure: it is
+    cached per (manager, matrix) so repeated consumers (D2 interpolation,
+    EM patches, resetup) reuse one exchange instead of re-fetching."""
+    if cache:
+        key = ("ring2", id(A), int(A.values._version))
+        hit = getattr(mgr, "_ring2_cache", None)
+        if hit is not None and hit[0] == key:
+            return hit[1]
+        out = halo_matrix(mgr, A, cache=False)
+        mgr._ring2_cache = (key, out)
+        return out
     ro = A.row_offsets.cpu().numpy().astype(np.int64)
     ci = A.col_indices.cpu().numpy().astype(np.int64)
     va = A.values.cpu().numpy().reshape(A.nnz, -1)
