@@ -67,6 +67,21 @@ void dilu_apply_small(const int* ro_s, const int* ci_s, const TA* va_s,
                       int ncolors, const TV* r, TV* w, TV* z, TV* x,
                       TV relax, long long vec_n, hipStream_t s);
 template <typename TA, typename TV>
+void dilu_fwd_sorted_fused(const int* ro_s, const int* ci_s, const TA* va_s,
+                           const TA* einv_s, const int* rows, int count,
+                           const TV* bvec, const TV* x, TV* w,
+                           hipStream_t s);
+template <typename TA, typename TV>
+void dilu_smooth_small(const int* ro_s, const int* ci_s, const TA* va_s,
+                       const TA* einv_s, const int* rows, const int* bounds,
+                       int ncolors, const TV* bvec, TV* w, TV* z, TV* x,
+                       TV relax, long long vec_n, hipStream_t s);
+template <typename TA, typename TV>
+void dilu_fwd_b4_sorted_fused(const int* ro_s, const int* ci_s,
+                              const TA* va_s, const TA* einv_s,
+                              const int* rows, int count, const TV* bvec,
+                              const TV* x, TV* w, hipStream_t s);
+template <typename TA, typename TV>
 void dilu_fwd_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
                      const TA* einv_s, const int* rows, int count,
                      const TV* r, TV* w, int b, hipStream_t s);

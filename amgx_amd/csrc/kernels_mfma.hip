@@ -61,12 +61,15 @@ void mfma4_probe(const double* a, const double* b, double* c, hipStream_t s) {
 // w is pre-zeroed outside so the full-row sum only picks up earlier colors.
 // SORTED=1 reads the color-sorted matrix copy (ro_s/einv_s indexed by slot,
 // pointers pre-offset to the color base).
-template <typename TA, typename TV, int SORTED>
+// FUSED=1 folds the residual into the sweep: the gather becomes
+// x[j]+w[j] against the FULL row (incl. diagonal; w_i is still zero), so
+// w_i = Einv_i (b_i - A_i.x - sum_{earlier} A_ij w_j) in one matrix read.
+template <typename TA, typename TV, int SORTED, int FUSED = 0>
 __global__ __launch_bounds__(256) void dilu_fwd_b4_kernel(
     const int* __restrict__ ro, const int* __restrict__ ci,
     const TA* __restrict__ va, const TA* __restrict__ einv,
     const int* __restrict__ rows, int count, const TV* __restrict__ r,
-    TV* __restrict__ w) {
+    TV* __restrict__ w, const TV* __restrict__ xv = nullptr) {
     int lane = threadIdx.x & 63;
     int b = (lane >> 2) & 3;             // MFMA block = row within wave
     int o = lane >> 4;                   // k index of A/B frags
@@ -83,10 +86,11 @@ __global__ __launch_bounds__(256) void dilu_fwd_b4_kernel(
         else { k0 = ro[i]; k1 = ro[i + 1]; ebase = (long long)i * 16; }
         for (int k = k0; k < k1; ++k) {
             int j = ci[k];
-            if (j == i) continue;
+            if (!FUSED && j == i) continue;
             // element (row o, col q) of the 4x4 block — coalesced 16 lanes
             double a = (double)va[(long long)k * 16 + o * 4 + q];
             double x = (double)w[(long long)j * 4 + q];
+            if (FUSED) x += (double)xv[(long long)j * 4 + q];
             acc += a * x;
         }
     }
@@ -249,6 +253,17 @@ void dilu_fwd_b4_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
 }
 
 template <typename TA, typename TV>
+void dilu_fwd_b4_sorted_fused(const int* ro_s, const int* ci_s,
+                              const TA* va_s, const TA* einv_s,
+                              const int* rows, int count, const TV* bvec,
+                              const TV* x, TV* w, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((dilu_fwd_b4_kernel<TA, TV, 1, 1>),
+                       dim3((count + 15) / 16), dim3(256), 0, s, ro_s, ci_s,
+                       va_s, einv_s, rows, count, bvec, w, x);
+}
+
+template <typename TA, typename TV>
 void dilu_bwd_b4_sorted(const int* ro_s, const int* ci_s, const TA* va_s,
                         const TA* einv_s, const int* rows, int count,
                         const TV* w, TV* z, hipStream_t s) {
@@ -325,6 +340,11 @@ void bsrmv_b4(const int* ro, const int* ci, const TA* va, const TV* x,
                                              const TA*, const TA*,           \
                                              const int*, int, const TV*,      \
                                              TV*, hipStream_t);               \
+    template void dilu_fwd_b4_sorted_fused<TA, TV>(const int*, const int*,    \
+                                                   const TA*, const TA*,      \
+                                                   const int*, int,           \
+                                                   const TV*, const TV*,      \
+                                                   TV*, hipStream_t);         \
     template void dilu_bwd_b4_sorted<TA, TV>(const int*, const int*,          \
                                              const TA*, const TA*,           \
                                              const int*, int, const TV*,      \

@@ -844,6 +844,89 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar_sorted(
     w[i] = (TV)einv_s[t] * (r[i] - sum);
 }
 
+// fused-residual forward sweep: w_i = Einv_i (b_i - sum_k a_k (x_k + w_k)).
+// Since w is pre-zeroed and only earlier colors are filled, the single
+// gather x[j]+w[j] folds the residual b-Ax INTO the substitution — one
+// matrix read instead of two (residual kernel + sweep), mathematically
+// identical to r = b - A x; M w = r.
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar_sorted_fused(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, int count, const TV* __restrict__ bvec,
+    const TV* __restrict__ x, TV* __restrict__ w) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    TV sum = TV(0);
+    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+        int j = ci_s[k];
+        sum += (TV)va_s[k] * (x[j] + w[j]);
+    }
+    w[i] = (TV)einv_s[t] * (bvec[i] - sum);
+}
+
+template <typename TA, typename TV>
+void dilu_fwd_sorted_fused(const int* ro_s, const int* ci_s, const TA* va_s,
+                           const TA* einv_s, const int* rows, int count,
+                           const TV* bvec, const TV* x, TV* w,
+                           hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((dilu_fwd_scalar_sorted_fused<TA, TV>),
+                       dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s, ro_s,
+                       ci_s, va_s, einv_s, rows, count, bvec, x, w);
+}
+
+// whole-apply fused-residual kernel for small levels (cf. dilu_apply_small)
+template <typename TA, typename TV>
+__global__ __launch_bounds__(1024) void dilu_smooth_small_kernel(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ va_s, const TA* __restrict__ einv_s,
+    const int* __restrict__ rows, const int* __restrict__ bounds,
+    int ncolors, const TV* __restrict__ bvec, TV* __restrict__ w,
+    TV* __restrict__ z, TV* __restrict__ x, TV relax, long long vec_n) {
+    int tid = threadIdx.x;
+    int nt = blockDim.x;
+    for (long long i = tid; i < vec_n; i += nt) { w[i] = TV(0); z[i] = TV(0); }
+    __syncthreads();
+    for (int c = 0; c < ncolors; ++c) {
+        int s0 = bounds[c], s1 = bounds[c + 1];
+        for (int t = s0 + tid; t < s1; t += nt) {
+            int i = rows[t];
+            TV sum = TV(0);
+            for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+                int j = ci_s[k];
+                sum += (TV)va_s[k] * (x[j] + w[j]);
+            }
+            w[i] = (TV)einv_s[t] * (bvec[i] - sum);
+        }
+        __syncthreads();
+    }
+    for (int c = ncolors - 1; c >= 0; --c) {
+        int s0 = bounds[c], s1 = bounds[c + 1];
+        for (int t = s0 + tid; t < s1; t += nt) {
+            int i = rows[t];
+            TV sum = TV(0);
+            for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
+                sum += (TV)va_s[k] * z[ci_s[k]];
+            z[i] = w[i] - (TV)einv_s[t] * sum;
+        }
+        __syncthreads();
+    }
+    long long xn = (long long)bounds[ncolors];
+    for (long long i = tid; i < xn; i += nt) x[i] += relax * z[i];
+}
+
+template <typename TA, typename TV>
+void dilu_smooth_small(const int* ro_s, const int* ci_s, const TA* va_s,
+                       const TA* einv_s, const int* rows, const int* bounds,
+                       int ncolors, const TV* bvec, TV* w, TV* z, TV* x,
+                       TV relax, long long vec_n, hipStream_t s) {
+    hipLaunchKernelGGL((dilu_smooth_small_kernel<TA, TV>), dim3(1),
+                       dim3(1024), 0, s, ro_s, ci_s, va_s, einv_s, rows,
+                       bounds, ncolors, bvec, w, z, x, relax, vec_n);
+}
+
 template <typename TA, typename TV>
 __global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_scalar_sorted(
     const int* __restrict__ ro_s, const int* __restrict__ ci_s,
@@ -1235,6 +1318,15 @@ void scatter_add(const T* src, const int* idx, int count, int b, T* dst,
     template void jacobi_smooth<TA, TV>(const int*, const int*, const TA*,      \
                                         const TA*, const TV*, const TV*, TV*,   \
                                         TV, int, int, double, hipStream_t);     \
+    template void dilu_fwd_sorted_fused<TA, TV>(const int*, const int*,         \
+                                                const TA*, const TA*,           \
+                                                const int*, int, const TV*,     \
+                                                const TV*, TV*, hipStream_t);   \
+    template void dilu_smooth_small<TA, TV>(const int*, const int*,             \
+                                            const TA*, const TA*, const int*,   \
+                                            const int*, int, const TV*, TV*,    \
+                                            TV*, TV*, TV, long long,            \
+                                            hipStream_t);                       \
     template void dilu_apply_small<TA, TV>(const int*, const int*,              \
                                            const TA*, const TA*, const int*,    \
                                            const int*, int, const TV*, TV*,     \

@@ -248,6 +248,28 @@ def dilu_setup(A, coloring):
     return DiluState(einv, va_s, einv_s)
 
 
+def dilu_smooth(A, Einv, coloring, b, x, relaxation):
+    """x += relax * M^{-1}(b - A x) with the residual fused into the
+    forward sweep (one matrix read fewer than residual + dilu_solve);
+    b=1 scalar and b=4 MFMA paths. Returns False when this block size has
+    no fused kernel (caller falls back to residual + dilu_solve)."""
+    if not isinstance(Einv, DiluState) or A.block_dim not in (1, 4):
+        return False
+    if getattr(A, "manager", None) is not None:
+        return False   # distributed path must exchange halos of x first
+    n = A.n_cols * A.block_dim
+    w = _scratch(A, "dilu_w", n, x.dtype)
+    z = _scratch(A, "dilu_z", n, x.dtype)
+    ro_s, ci_s, pos, perm = _color_sorted_struct(A, coloring)
+    _core.dilu_smooth_sorted(ro_s, ci_s, Einv.va_s, A.block_dim,
+                             Einv.einv_s, coloring.rows_sorted,
+                             coloring.bounds,
+                             getattr(coloring, "bounds_dev", None),
+                             b.reshape(-1), x.reshape(-1), w, z,
+                             float(relaxation))
+    return True
+
+
 def dilu_solve(A, Einv, coloring, r, relaxation, x):
     n = A.n_cols * A.block_dim   # ext size: halo tails stay zero in the sweeps
     w = _scratch(A, "dilu_w", n, r.dtype)   # vector precision (dDFI mixed)

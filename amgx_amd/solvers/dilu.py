@@ -43,6 +43,11 @@ class MulticolorDILUSolver(_SmootherBase):
         self.Einv = dilu_setup(A, A.coloring)
 
     def solve_iteration(self, b, x):
+        B = ops._backend(self.A)
+        fused = getattr(B, "dilu_smooth", None)
+        if fused is not None and fused(self.A, self.Einv, self.A.coloring,
+                                       b, x, self.relaxation_factor):
+            return False
         r = ops.residual(self.A, x, b)
         dilu_solve(self.A, self.Einv, self.A.coloring, r,
                    self.relaxation_factor, x)
