@@ -497,9 +497,13 @@ __global__ __launch_bounds__(AMGX_BLOCK) void gs_rows_scalar_sorted(
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
+    int k = ro_s[t], e = ro_s[t + 1];
     TV sum = TV(0);
-    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
-        sum += (TV)va_s[k] * x[ci_s[k]];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        if (k < e) { sum += (TV)va_s[k] * x[ci_s[k]]; ++k; }
+    }
+    for (; k < e; ++k) sum += (TV)va_s[k] * x[ci_s[k]];
     x[i] += omega * (TV)dinv_s[t] * (bvec[i] - sum);
 }
 
@@ -838,9 +842,13 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar_sorted(
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
+    int k = ro_s[t], e = ro_s[t + 1];
     TV sum = TV(0);
-    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
-        sum += (TV)va_s[k] * w[ci_s[k]];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        if (k < e) { sum += (TV)va_s[k] * w[ci_s[k]]; ++k; }
+    }
+    for (; k < e; ++k) sum += (TV)va_s[k] * w[ci_s[k]];
     w[i] = (TV)einv_s[t] * (r[i] - sum);
 }
 
@@ -858,8 +866,14 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_fwd_scalar_sorted_fused(
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
+    int k = ro_s[t], e = ro_s[t + 1];
     TV sum = TV(0);
-    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        if (k < e) { int j = ci_s[k];
+                     sum += (TV)va_s[k] * (x[j] + w[j]); ++k; }
+    }
+    for (; k < e; ++k) {
         int j = ci_s[k];
         sum += (TV)va_s[k] * (x[j] + w[j]);
     }
@@ -936,9 +950,13 @@ __global__ __launch_bounds__(AMGX_BLOCK) void dilu_bwd_scalar_sorted(
     int t = blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= count) return;
     int i = rows[t];
+    int k = ro_s[t], e = ro_s[t + 1];
     TV sum = TV(0);
-    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k)
-        sum += (TV)va_s[k] * z[ci_s[k]];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        if (k < e) { sum += (TV)va_s[k] * z[ci_s[k]]; ++k; }
+    }
+    for (; k < e; ++k) sum += (TV)va_s[k] * z[ci_s[k]];
     z[i] = wv[i] - (TV)einv_s[t] * sum;
 }
 
