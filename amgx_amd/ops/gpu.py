@@ -191,8 +191,10 @@ def color_matrix(A, max_uncolored_frac: float = 0.0, seed: int = 0,
                  multihash_rounds: int = 0):
     """Device Jones-Plassmann greedy coloring rounds; multihash_rounds > 0
     prepends MULTI_HASH semantics (color = round id, per-round re-hash)."""
+    # dense coarse-level graphs (classical RAP) need ~max-clique rounds:
+    # each greedy round colors only the local maxima of the uncolored set
     colors, nc = _core.color_minmax(A.row_offsets, A.col_indices, A.n_rows,
-                                    64, int(seed), int(multihash_rounds))
+                                    4096, int(seed), int(multihash_rounds))
     return colors, int(nc)
 
 
