@@ -72,7 +72,7 @@ FGMRES_ILU0 = {
     "solver": {
         "preconditioner": {"solver": "MULTICOLOR_ILU", "max_iters": 1,
                            "scope": "ilu"},
-        "solver": "FGMRES", "max_iters": 200, "gmres_n_restart": 30,
+        "solver": "FGMRES", "max_iters": 500, "gmres_n_restart": 30,
         "monitor_residual": 1, "convergence": "RELATIVE_INI",
         "tolerance": 1e-6},
 }
@@ -189,8 +189,11 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--size", type=int, default=256,
-                    help="per-rank subdomain edge (size^3 rows per GPU)")
+    ap.add_argument("--size", type=int, default=None,
+                    help="per-rank subdomain edge (size^3 rows per GPU); "
+                         "defaults: 256 (fgmres_agg/classical_pcg/block4), "
+                         "128 (ilu0 — the unstructured ILU(0) iteration "
+                         "count grows with size)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--config", default="fgmres_agg", choices=sorted(CONFIGS),
                     help="solver composition (driver contract: fgmres_agg)")
@@ -213,6 +216,8 @@ def main():
         import torch.distributed as tdist
         dist = tdist
 
+    if args.size is None:
+        args.size = 128 if args.config == "ilu0" else 256
     cfg_dict, problem = CONFIGS[args.config]
     cfg = AMGConfig.from_dict(cfg_dict)
     A, manager = build_local_matrix(args.size, rank, world, device, problem)
