@@ -56,12 +56,17 @@ FGMRES_AGG = {
 
 # Secondary measurable configs (BASELINE.json configs #3/#4/#5 shapes);
 # the driver contract always uses the default fgmres_agg.
+# symmetric_GS: ascending+descending color sweeps make the smoother (and
+# with equal pre/post sweeps, the whole V-cycle) symmetric — PCG needs an
+# SPD preconditioner (plain one-direction multicolor GS is not adjoint to
+# itself, which stalls PCG at scale)
 PCG_CLASSICAL = {
     "config_version": 2,
     "solver": {
         "preconditioner": {
             "algorithm": "CLASSICAL", "solver": "AMG",
-            "smoother": "MULTICOLOR_GS", "presweeps": 1, "postsweeps": 1,
+            "smoother": "MULTICOLOR_GS", "symmetric_GS": 1,
+            "presweeps": 1, "postsweeps": 1,
             "max_iters": 1, "min_coarse_rows": 32, "scope": "amg",
             "cycle": "V"},
         "solver": "PCG", "max_iters": 100, "monitor_residual": 1,
