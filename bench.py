@@ -65,6 +65,7 @@ PCG_CLASSICAL = {
     "solver": {
         "preconditioner": {
             "algorithm": "CLASSICAL", "solver": "AMG",
+            "interpolator": "D2",      # PMIS needs distance-2 interpolation
             "smoother": "MULTICOLOR_GS", "symmetric_GS": 1,
             "presweeps": 1, "postsweeps": 1,
             "max_iters": 1, "min_coarse_rows": 32, "scope": "amg",
