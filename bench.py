@@ -66,6 +66,7 @@ PCG_CLASSICAL = {
         "preconditioner": {
             "algorithm": "CLASSICAL", "solver": "AMG",
             "interpolator": "D2",      # PMIS needs distance-2 interpolation
+            "interp_truncation_factor": 0.25,   # keep the D2 operator sparse
             "smoother": "MULTICOLOR_GS", "symmetric_GS": 1,
             "presweeps": 1, "postsweeps": 1,
             "max_iters": 1, "min_coarse_rows": 32, "scope": "amg",
