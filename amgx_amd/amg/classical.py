@@ -274,6 +274,51 @@ def interp_d1(A, S, cf_map, num_coarse, scope):
     return ops._backend(A).interp_d1(A, S, cf_map, num_coarse)
 
 
+def _csr_add_device(n, nc, roA, ciA, vaA, roB, ciB, vaB, dev, dtype):
+    """C = A + B for two device CSRs with the same shape: rowwise concat
+    (pure index arithmetic, no sort) then ONE hash dedupe pass
+    (spgemm_hash mode 1 with identity membership/column map) — replaces a
+    full torch.unique sort over the combined nnz."""
+    from ..matrix import CSRMatrix
+    from ..ops import gpu as G
+    cntA = (roA[1:] - roA[:-1]).to(torch.int64)
+    cntB = (roB[1:] - roB[:-1]).to(torch.int64)
+    cnt = cntA + cntB
+    roK = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    roK[1:] = torch.cumsum(cnt, 0)
+    nnzK = int(roK[-1].item())
+    ciK = torch.empty(nnzK, dtype=torch.int32, device=dev)
+    vaK = torch.empty(nnzK, dtype=dtype, device=dev)
+    # scatter positions for A's entries then B's (per row: A block first)
+    def _pos(ro_src, cnt_src, base_off):
+        csum = torch.cumsum(cnt_src, 0) - cnt_src
+        rows_l = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=dev), cnt_src)
+        local = (torch.arange(int(cnt_src.sum().item()), device=dev,
+                              dtype=torch.int64)
+                 - torch.repeat_interleave(csum, cnt_src))
+        return roK[rows_l] + base_off[rows_l] + local
+    zero = torch.zeros(n, dtype=torch.int64, device=dev)
+    posA = _pos(roA.to(torch.int64), cntA, zero)
+    posB = _pos(roB.to(torch.int64), cntB, cntA)
+    ciK[posA] = ciA.to(torch.int32)
+    vaK[posA] = vaA.to(dtype)
+    ciK[posB] = ciB.to(torch.int32)
+    vaK[posB] = vaB.to(dtype)
+    from .. import _core
+    ident = torch.arange(nc, dtype=torch.int32, device=dev)
+    m_ro = torch.arange(n + 1, dtype=torch.int32, device=dev)
+    m_fid = torch.arange(n, dtype=torch.int32, device=dev)
+    ro, ci2, va2, big = _core.spgemm_hash(
+        m_ro, m_fid, vaK, roK.to(torch.int32), ciK, vaK, ident, 1,
+        max(nnzK, 1), 64)
+    if int(ro[0].item()) == -1:   # pathological row: fall back to unique
+        return None
+    if big.numel():
+        G._sort_unsorted_rows(ro, ci2, va2, big)
+    return CSRMatrix(ro, ci2.contiguous(), va2.contiguous(), n_cols=nc)
+
+
 def _interp_d2_device(A, S, cf_map, num_coarse, scope):
     """Device-resident distance-2 interpolation: the same magnitude-
     proportional formulation as the host path below (acc = D + F @ W, then
@@ -318,20 +363,37 @@ def _interp_d2_device(A, S, cf_map, num_coarse, scope):
     live_edge = strongF & alive[ci_n] & local
     F = _csr(live_edge, ci, v, n)
     FW = ops._backend(A).spgemm(F, W)
-    # acc = D + FW  (merge by sorted (row, col) key)
-    fw_deg = FW.row_offsets.to(torch.int64)
-    fw_rows = torch.repeat_interleave(
-        torch.arange(n, dtype=torch.int64, device=dev),
-        fw_deg[1:] - fw_deg[:-1])
-    all_r = torch.cat([rows[strongC], fw_rows])
-    all_c = torch.cat([cf_col[strongC], FW.col_indices.to(torch.int64)])
-    all_v = torch.cat([v[strongC], FW.values.reshape(-1).to(torch.float64)])
-    key = all_r * nc + all_c
-    uk, inv = torch.unique(key, return_inverse=True)
-    vsum = torch.zeros(uk.numel(), dtype=torch.float64, device=dev)
-    vsum.index_add_(0, inv, all_v)
-    acc_r = uk // nc
-    acc_c = uk % nc
+    # acc = D + FW: rowwise concat + one hash dedupe (no global sort);
+    # unique-sort fallback on CPU (parity tests) or hash overflow
+    ACC = None
+    if dev.type == "cuda":
+        D = _csr(strongC, cf_col, v, nc)
+        ACC = _csr_add_device(n, nc, D.row_offsets, D.col_indices,
+                              D.values.reshape(-1), FW.row_offsets,
+                              FW.col_indices, FW.values.reshape(-1), dev,
+                              torch.float64)
+    if ACC is not None:
+        acc_deg = ACC.row_offsets.to(torch.int64)
+        acc_r = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=dev),
+            acc_deg[1:] - acc_deg[:-1])
+        acc_c = ACC.col_indices.to(torch.int64)
+        vsum = ACC.values.reshape(-1).to(torch.float64)
+    else:
+        fw_deg = FW.row_offsets.to(torch.int64)
+        fw_rows = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=dev),
+            fw_deg[1:] - fw_deg[:-1])
+        all_r = torch.cat([rows[strongC], fw_rows])
+        all_c = torch.cat([cf_col[strongC], FW.col_indices.to(torch.int64)])
+        all_v = torch.cat([v[strongC],
+                           FW.values.reshape(-1).to(torch.float64)])
+        key = all_r * nc + all_c
+        uk, inv = torch.unique(key, return_inverse=True)
+        vsum = torch.zeros(uk.numel(), dtype=torch.float64, device=dev)
+        vsum.index_add_(0, inv, all_v)
+        acc_r = uk // nc
+        acc_c = uk % nc
     # denominators: diag + weak couplings + dead-end strong-F couplings
     diag = torch.zeros(n, dtype=torch.float64, device=dev)
     dmask = rows == ci
