@@ -48,6 +48,7 @@ class AMGHierarchy:
         t0 = time.perf_counter()
         self.levels = []
         dist = getattr(A, "manager", None)
+        A = self._permute_fine(A)
         level = create_level(self.algorithm, A, self.scope, 0)
         self.levels.append(level)
         while True:
@@ -89,7 +90,18 @@ class AMGHierarchy:
         t0 = time.perf_counter()
         if reuse_levels < 0:
             reuse_levels = len(self.levels)
-        self.levels[0].A = A
+        if getattr(self, "_fine_perm", None) is not None:
+            # values-only refresh of the permuted fine copy (same structure)
+            A0 = self.levels[0].A
+            if A0.nnz == A.nnz:
+                A0.values = A.values.reshape(-1)[self._fine_pos].contiguous()
+                A0._cache.clear()
+                A0._diag_idx = None
+                A0.coloring = self._fine_coloring
+            else:                     # structure changed: re-permute fully
+                self.levels[0].A = self._permute_fine(A)
+        else:
+            self.levels[0].A = A
         A._cache.pop("cf_map", None)
         for i, lvl in enumerate(self.levels[:-1]):
             if i >= reuse_levels:
@@ -193,9 +205,69 @@ class AMGHierarchy:
             coarsest.smoother.setup(coarsest.A)
 
     # ------------------------------------------------------------------ cycles
+    def _permute_fine(self, A):
+        """Color-renumber the FINE level (rows + columns) so its smoother
+        sweeps access vectors color-contiguously (full reorder-by-color,
+        reference include/matrix.h:766): the hierarchy then lives entirely
+        in the permuted numbering and cycle() maps b/x at the boundary (two
+        gathers per cycle vs ~num_colors x random-line vector traffic per
+        sweep). Single-process scalar device matrices only."""
+        self._fine_perm = None
+        if (not A.values.is_cuda or A.block_dim != 1
+                or getattr(A, "manager", None) is not None
+                or A.n_rows < 4096):
+            return A
+        from ..matrix import CSRMatrix
+        from .coloring import MatrixColoring
+        col = MatrixColoring.create(A, self.scope)
+        dev = A.values.device
+        perm = col.rows_sorted.to(torch.int64)
+        iperm = torch.empty_like(perm)
+        iperm[perm] = torch.arange(A.n_rows, dtype=torch.int64, device=dev)
+        ro64 = A.row_offsets.to(torch.int64)
+        deg = ro64[1:] - ro64[:-1]
+        counts = deg[perm]
+        csum = torch.cumsum(counts, 0)
+        ro_p = torch.zeros(A.n_rows + 1, dtype=torch.int32, device=dev)
+        ro_p[1:] = csum.to(torch.int32)
+        pos = (torch.repeat_interleave(ro64[perm], counts)
+               + torch.arange(int(A.nnz), device=dev, dtype=torch.int64)
+               - torch.repeat_interleave(csum - counts, counts))
+        rows_p = torch.repeat_interleave(
+            torch.arange(A.n_rows, dtype=torch.int64, device=dev), counts)
+        ci_p = iperm[A.col_indices.to(torch.int64)[pos]]
+        # per-row column sort (one radix argsort of the combined key)
+        key = rows_p * A.n_rows + ci_p
+        order = torch.argsort(key)
+        pos = pos[order]
+        Ap = CSRMatrix(ro_p, ci_p[order].to(torch.int32).contiguous(),
+                       A.values[pos].contiguous(), n_cols=A.n_cols)
+        Ap.coloring = MatrixColoring(col.colors[perm].contiguous(),
+                                     col.num_colors)
+        self._fine_perm = perm
+        self._fine_iperm = iperm
+        self._fine_pos = pos          # values gather map for resetup
+        self._fine_coloring = Ap.coloring
+        self._pb = torch.empty(A.n_rows, dtype=A.values.dtype, device=dev)
+        self._px = torch.empty_like(self._pb)
+        return Ap
+
     def cycle(self, b: torch.Tensor, x: torch.Tensor,
               zero_initial_guess: bool = True):
         """One AMG cycle on the finest level."""
+        if self._fine_perm is not None:
+            bp = self._pb
+            xp = self._px
+            if bp.dtype != b.dtype:
+                bp = self._pb = torch.empty_like(b.reshape(-1))
+                xp = self._px = torch.empty_like(bp)
+            torch.index_select(b.reshape(-1), 0, self._fine_perm, out=bp)
+            if not zero_initial_guess:
+                torch.index_select(x.reshape(-1), 0, self._fine_perm,
+                                   out=xp)
+            self._cycle(0, bp, xp, zero_initial_guess, self.cycle_type)
+            x.reshape(-1)[self._fine_perm] = xp
+            return
         self._cycle(0, b, x, zero_initial_guess, self.cycle_type)
 
     def _cycle(self, li: int, b, x, zero_guess: bool, ctype: str):

@@ -123,6 +123,7 @@ class AggregationLevel(AMGLevel):
         if mgr is None:
             Ac = ops.galerkin_aggregation(self.A, self.aggregates,
                                           self.num_aggregates)
+            Ac = self._color_renumber(Ac)
             geom = self.A._cache.get("geometry")
             if geom is not None:
                 # aggregate centroids keep GEO usable on coarse levels
@@ -172,12 +173,48 @@ class AggregationLevel(AMGLevel):
         self._build_r_structure()
         return Ac
 
+    def _color_renumber(self, Ac):
+        """Renumber the coarse level into color order (full reorder-by-color,
+        reference include/matrix.h:766 + src/core.cu:489-506): color Ac,
+        compose the color-sort permutation into the fine->coarse aggregate
+        map, and re-run the Galerkin in the new numbering — the rebuilt Ac
+        has color-contiguous rows AND columns, so every smoother sweep on
+        this level reads vectors contiguously (rows_sorted becomes the
+        identity). Single-process device path only."""
+        import torch
+
+        from .coloring import MatrixColoring
+        if (not Ac.values.is_cuda or Ac.block_dim != 1
+                or getattr(self.A, "manager", None) is not None
+                or Ac.n_rows < 512):
+            return Ac
+        col = MatrixColoring.create(Ac, self.scope)
+        perm = col.rows_sorted.to(torch.int64)          # slot -> old id
+        iperm = torch.empty_like(perm)
+        iperm[perm] = torch.arange(Ac.n_rows, dtype=torch.int64,
+                                   device=perm.device)
+        self.aggregates = iperm[self.aggregates.to(torch.int64)] \
+            .to(torch.int32)
+        Ac2 = ops.galerkin_aggregation(self.A, self.aggregates,
+                                       self.num_aggregates)
+        # colors in the new numbering are ascending by construction
+        self._coarse_coloring = MatrixColoring(
+            col.colors[perm].contiguous(), col.num_colors)
+        Ac2.coloring = self._coarse_coloring
+        return Ac2
+
     def rebuild_coarse_values(self):
         if self.aggregates is None \
                 or getattr(self.A, "manager", None) is not None:
             return None
-        return ops.galerkin_aggregation(self.A, self.aggregates,
-                                        self.num_aggregates)
+        Ac = ops.galerkin_aggregation(self.A, self.aggregates,
+                                      self.num_aggregates)
+        # aggregates already carry the color renumbering; re-attach the
+        # structure-valid coloring so sweeps keep the identity ordering
+        coloring = getattr(self, "_coarse_coloring", None)
+        if coloring is not None and Ac.values.is_cuda:
+            Ac.coloring = coloring
+        return Ac
 
     def restrict_residual(self, r, bc):
         out = ops.restrict_agg(r, self.aggregates, self.num_aggregates,
