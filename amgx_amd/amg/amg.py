@@ -213,6 +213,9 @@ class AMGHierarchy:
         gathers per cycle vs ~num_colors x random-line vector traffic per
         sweep). Single-process scalar device matrices only."""
         self._fine_perm = None
+        import os
+        if os.environ.get("AMGX_AMD_NO_FINE_PERM"):
+            return A
         if (not A.values.is_cuda or A.block_dim != 1
                 or getattr(A, "manager", None) is not None
                 or A.n_rows < 4096):

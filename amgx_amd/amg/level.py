@@ -181,9 +181,13 @@ class AggregationLevel(AMGLevel):
         has color-contiguous rows AND columns, so every smoother sweep on
         this level reads vectors contiguously (rows_sorted becomes the
         identity). Single-process device path only."""
+        import os
+
         import torch
 
         from .coloring import MatrixColoring
+        if os.environ.get("AMGX_AMD_NO_COARSE_RENUMBER"):
+            return Ac
         if (not Ac.values.is_cuda or Ac.block_dim != 1
                 or getattr(self.A, "manager", None) is not None
                 or Ac.n_rows < 512):
