@@ -199,6 +199,7 @@ class AggregationLevel(AMGLevel):
                                    device=perm.device)
         self.aggregates = iperm[self.aggregates.to(torch.int64)] \
             .to(torch.int32)
+        self._build_r_structure()     # restriction map follows the new ids
         Ac2 = ops.galerkin_aggregation(self.A, self.aggregates,
                                        self.num_aggregates)
         # colors in the new numbering are ascending by construction
