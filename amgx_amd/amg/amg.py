@@ -214,7 +214,12 @@ class AMGHierarchy:
         sweep). Single-process scalar device matrices only."""
         self._fine_perm = None
         import os
-        if os.environ.get("AMGX_AMD_NO_FINE_PERM"):
+        # Measured on MI355X (gpurun call 17): color-permuting the FINE
+        # level costs ~8% on the 256^3 bench — the natural band ordering's
+        # +-1-neighbor cache-line sharing in the x/w gathers beats
+        # color-contiguous vector access. Opt-in for matrices without
+        # banded structure.
+        if not os.environ.get("AMGX_AMD_FINE_PERM"):
             return A
         if (not A.values.is_cuda or A.block_dim != 1
                 or getattr(A, "manager", None) is not None
