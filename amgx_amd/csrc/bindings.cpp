@@ -761,14 +761,15 @@ std::tuple<Tensor, Tensor, Tensor> interp_d1(Tensor ro, Tensor ci, Tensor va,
 }
 
 std::tuple<Tensor, Tensor, Tensor> truncate_rows(Tensor ro, Tensor ci,
-                                                 Tensor va, double factor) {
+                                                 Tensor va, double factor,
+                                                 int64_t max_elem) {
     int n = (int)(ro.numel() - 1);
     auto counts = torch::empty({n}, ro.options());
     DISPATCH_FT(va, "truncate_count", [&] {
         amgx_hip::truncate_rows_gpu<scalar_t>(
             ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
-            n, factor, nullptr, nullptr, nullptr, counts.data_ptr<int>(),
-            cur_stream());
+            n, factor, (int)max_elem, nullptr, nullptr, nullptr,
+            counts.data_ptr<int>(), cur_stream());
     });
     auto ro_out = torch::zeros({n + 1}, ro.options());
     ro_out.slice(0, 1, n + 1).copy_(
@@ -779,8 +780,9 @@ std::tuple<Tensor, Tensor, Tensor> truncate_rows(Tensor ro, Tensor ci,
     DISPATCH_FT(va, "truncate_fill", [&] {
         amgx_hip::truncate_fill_gpu<scalar_t>(
             ro.data_ptr<int>(), ci.data_ptr<int>(), va.data_ptr<scalar_t>(),
-            n, factor, ro_out.data_ptr<int>(), ci_out.data_ptr<int>(),
-            va_out.data_ptr<scalar_t>(), cur_stream());
+            n, factor, (int)max_elem, ro_out.data_ptr<int>(),
+            ci_out.data_ptr<int>(), va_out.data_ptr<scalar_t>(),
+            cur_stream());
     });
     return {ro_out, ci_out, va_out};
 }

@@ -319,10 +319,10 @@ void ilu0_bwd_block_launch(const int* ro, const int* ci, const int* pos,
 namespace amgx_hip {
 template <typename T>
 void truncate_rows_gpu(const int* ro, const int* ci, const T* va, int n,
-                       double factor, const int* ro_out, int* ci_out,
-                       T* va_out, int* counts, hipStream_t s);
+                       double factor, int max_elem, const int* ro_out,
+                       int* ci_out, T* va_out, int* counts, hipStream_t s);
 template <typename T>
 void truncate_fill_gpu(const int* ro, const int* ci, const T* va, int n,
-                       double factor, const int* ro_out, int* ci_out,
-                       T* va_out, hipStream_t s);
+                       double factor, int max_elem, const int* ro_out,
+                       int* ci_out, T* va_out, hipStream_t s);
 }  // namespace amgx_hip

@@ -659,10 +659,52 @@ namespace amgx_hip {
 // ============================================================ truncate
 // Drop |p_ij| < factor * rowmax_i, rescale survivors to preserve the row sum
 // (reference src/truncate.cu truncate_kernel:388 + truncateAndScale:506).
+// kth-largest key threshold among the factor-filtered entries of a row,
+// ordered by (|value| desc, position asc) — a strict total order, so ties
+// resolve like the host's stable argsort.  O(m * deg) selection without
+// marks: each round finds the max strictly below the previous pick.
+template <typename T>
+__device__ __forceinline__ void topk_threshold(const T* va, int s, int e,
+                                               double factor, double mx,
+                                               int m, double* th_v,
+                                               int* th_p) {
+    double cv = 1.0 / 0.0;   // +inf sentinel: everything is below it
+    int cp = -1;
+    for (int r = 0; r < m; ++r) {
+        double bv = -1.0;
+        int bp = -1;
+        for (int k = s; k < e; ++k) {
+            double a = fabs((double)va[k]);
+            if (a < factor * mx) continue;
+            // strictly below (cv, cp) in (desc, asc-pos) order
+            bool below = a < cv || (a == cv && k > cp);
+            if (!below) continue;
+            bool better = a > bv || (a == bv && k < bp);
+            if (better) { bv = a; bp = k; }
+        }
+        if (bp < 0) break;      // fewer than m filtered entries
+        cv = bv;
+        cp = bp;
+    }
+    *th_v = cv;
+    *th_p = cp;
+}
+
+template <typename T>
+__device__ __forceinline__ bool trunc_keep(const T* va, int k, double factor,
+                                           double mx, int m, double th_v,
+                                           int th_p) {
+    double a = fabs((double)va[k]);
+    if (a < factor * mx) return false;
+    if (m < 0) return true;
+    // keep entries at or above the m-th key in the total order
+    return a > th_v || (a == th_v && k <= th_p);
+}
+
 template <typename T>
 __global__ __launch_bounds__(AMGX_BLOCK) void truncate_count(
     const int* __restrict__ ro, const T* __restrict__ va, int n,
-    double factor, int* __restrict__ counts) {
+    double factor, int max_elem, int* __restrict__ counts) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     int s = ro[i], e = ro[i + 1];
@@ -671,13 +713,14 @@ __global__ __launch_bounds__(AMGX_BLOCK) void truncate_count(
     int c = 0;
     for (int k = s; k < e; ++k)
         if (fabs((double)va[k]) >= factor * mx) ++c;
+    if (max_elem >= 0 && c > max_elem) c = max_elem;
     counts[i] = c;
 }
 
 template <typename T>
 __global__ __launch_bounds__(AMGX_BLOCK) void truncate_fill(
     const int* __restrict__ ro, const int* __restrict__ ci,
-    const T* __restrict__ va, int n, double factor,
+    const T* __restrict__ va, int n, double factor, int max_elem,
     const int* __restrict__ ro_out, int* __restrict__ ci_out,
     T* __restrict__ va_out) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -689,13 +732,24 @@ __global__ __launch_bounds__(AMGX_BLOCK) void truncate_fill(
         sum_old += a;
         mx = fmax(mx, fabs(a));
     }
+    double th_v = -1.0;
+    int th_p = -1;
+    int m = max_elem;
+    if (m >= 0) {
+        int c = 0;
+        for (int k = s; k < e; ++k)
+            if (fabs((double)va[k]) >= factor * mx) ++c;
+        if (c <= m) m = -1;             // nothing to cap
+        else topk_threshold(va, s, e, factor, mx, m, &th_v, &th_p);
+    }
     for (int k = s; k < e; ++k)
-        if (fabs((double)va[k]) >= factor * mx) sum_new += (double)va[k];
+        if (trunc_keep(va, k, factor, mx, m, th_v, th_p))
+            sum_new += (double)va[k];
     double scale = (sum_new != 0.0 && sum_old != 0.0) ? sum_old / sum_new
                                                       : 1.0;
     int out = ro_out[i];
     for (int k = s; k < e; ++k) {
-        if (fabs((double)va[k]) < factor * mx) continue;
+        if (!trunc_keep(va, k, factor, mx, m, th_v, th_p)) continue;
         ci_out[out] = ci[k];
         va_out[out] = (T)((double)va[k] * scale);
         ++out;
@@ -704,27 +758,29 @@ __global__ __launch_bounds__(AMGX_BLOCK) void truncate_fill(
 
 template <typename T>
 void truncate_rows_gpu(const int* ro, const int* ci, const T* va, int n,
-                       double factor, const int* ro_out, int* ci_out,
-                       T* va_out, int* counts, hipStream_t s) {
+                       double factor, int max_elem, const int* ro_out,
+                       int* ci_out, T* va_out, int* counts, hipStream_t s) {
     hipLaunchKernelGGL((truncate_count<T>), dim3(grid_1d(n)),
-                       dim3(AMGX_BLOCK), 0, s, ro, va, n, factor, counts);
+                       dim3(AMGX_BLOCK), 0, s, ro, va, n, factor, max_elem,
+                       counts);
 }
 
 template <typename T>
 void truncate_fill_gpu(const int* ro, const int* ci, const T* va, int n,
-                       double factor, const int* ro_out, int* ci_out,
-                       T* va_out, hipStream_t s) {
+                       double factor, int max_elem, const int* ro_out,
+                       int* ci_out, T* va_out, hipStream_t s) {
     hipLaunchKernelGGL((truncate_fill<T>), dim3(grid_1d(n)),
-                       dim3(AMGX_BLOCK), 0, s, ro, ci, va, n, factor, ro_out,
+                       dim3(AMGX_BLOCK), 0, s, ro, ci, va, n, factor,
+                       max_elem, ro_out,
                        ci_out, va_out);
 }
 
 #define INSTANTIATE_TRUNC(T)                                                   \
     template void truncate_rows_gpu<T>(const int*, const int*, const T*, int, \
-                                       double, const int*, int*, T*, int*,    \
-                                       hipStream_t);                          \
+                                       double, int, const int*, int*, T*,     \
+                                       int*, hipStream_t);                    \
     template void truncate_fill_gpu<T>(const int*, const int*, const T*, int, \
-                                       double, const int*, int*, T*,          \
+                                       double, int, const int*, int*, T*,     \
                                        hipStream_t);
 
 INSTANTIATE_TRUNC(double)

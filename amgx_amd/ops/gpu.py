@@ -423,16 +423,16 @@ def galerkin_rap(R, A, P):
 
 
 def truncate_rows(P, trunc_factor: float = 0.0, max_elements: int = -1):
-    if not (0.0 < trunc_factor < 1.0) and max_elements < 0:
+    """Drop |p| < factor*rowmax and/or cap rows at the top-max_elements
+    entries by |value| (device kernels; reference src/truncate.cu)."""
+    if not (0.0 < trunc_factor < 1.0):
+        trunc_factor = 0.0
+    if trunc_factor <= 0.0 and max_elements < 0:
         return P
-    if max_elements >= 0:
-        # per-row top-k cap: host pass (setup-time, rarely configured)
-        from . import cpu
-        return cpu.truncate_rows(P.to("cpu"), trunc_factor, max_elements) \
-            .to(P.device)
     from ..matrix import CSRMatrix
     ro, ci, va = _core.truncate_rows(P.row_offsets, P.col_indices,
-                                     P.values, float(trunc_factor))
+                                     P.values, float(trunc_factor),
+                                     int(max_elements))
     return CSRMatrix(ro, ci.contiguous(), va.contiguous(), n_cols=P.n_cols)
 
 

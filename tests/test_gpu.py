@@ -640,6 +640,14 @@ def test_truncate_rows_gpu():
     assert torch.equal(got.row_offsets.cpu(), ref.row_offsets)
     assert torch.equal(got.col_indices.cpu(), ref.col_indices)
     assert torch.allclose(got.values.cpu(), ref.values, atol=1e-13)
+    # device top-k cap (interp_max_elements) vs the host stable-argsort ref
+    for tf, me in ((0.0, 3), (0.2, 4), (0.4, 2)):
+        ref = cpu_ops.truncate_rows(A, trunc_factor=tf, max_elements=me)
+        got = gpu_ops.truncate_rows(to_gpu(A), trunc_factor=tf,
+                                    max_elements=me)
+        assert torch.equal(got.row_offsets.cpu(), ref.row_offsets), (tf, me)
+        assert torch.equal(got.col_indices.cpu(), ref.col_indices), (tf, me)
+        assert torch.allclose(got.values.cpu(), ref.values, atol=1e-13)
 
 
 def test_wf_cycles_gpu():
