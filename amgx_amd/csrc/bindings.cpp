@@ -879,6 +879,42 @@ void ilu0_apply_block(Tensor ro, Tensor ci, Tensor lu, Tensor dinv,
     });
 }
 
+// color-sorted ILU(0) apply (reorder-by-color slabs; cf. dilu_apply_sorted)
+void ilu0_apply_sorted(Tensor ro_s, Tensor ci_s, Tensor lu_s, Tensor diag_s,
+                       Tensor pos, Tensor rows_sorted,
+                       std::vector<int64_t> bounds, Tensor r, Tensor y,
+                       Tensor z, Tensor x, double relax) {
+    int n = (int)(ro_s.numel() - 1);
+    int nc = (int)bounds.size() - 1;
+    y.zero_();
+    z.zero_();
+    DISPATCH_FT2(lu_s, x, "ilu0_apply_sorted", [&] {
+        hipStream_t st = cur_stream();
+        for (int c = 0; c < nc; ++c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::ilu0_fwd_sorted<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>() + s, ci_s.data_ptr<int>(),
+                lu_s.data_ptr<scalar_a>(), pos.data_ptr<int>(),
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                r.data_ptr<scalar_v>(), y.data_ptr<scalar_v>(), n, st);
+        }
+        for (int c = nc - 1; c >= 0; --c) {
+            int64_t s = bounds[c], e = bounds[c + 1];
+            if (e <= s) continue;
+            amgx_hip::ilu0_bwd_sorted<scalar_a, scalar_v>(
+                ro_s.data_ptr<int>() + s, ci_s.data_ptr<int>(),
+                lu_s.data_ptr<scalar_a>(),
+                diag_s.data_ptr<scalar_a>() + s, pos.data_ptr<int>(),
+                rows_sorted.data_ptr<int>() + s, (int)(e - s),
+                y.data_ptr<scalar_v>(), z.data_ptr<scalar_v>(), n, st);
+        }
+        amgx_hip::axpy<scalar_v>(x.data_ptr<scalar_v>(),
+                                 z.data_ptr<scalar_v>(), (scalar_v)relax,
+                                 x.numel(), st);
+    });
+}
+
 void ilu0_apply(Tensor ro, Tensor ci, Tensor lu, Tensor didx, Tensor pos,
                 Tensor rows_sorted, std::vector<int64_t> bounds, Tensor r,
                 Tensor y, Tensor z, Tensor x, double relax) {
@@ -955,6 +991,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("truncate_rows", &truncate_rows);
     m.def("ilu0_setup", &ilu0_setup);
     m.def("ilu0_setup_block", &ilu0_setup_block);
+    m.def("ilu0_apply_sorted", &ilu0_apply_sorted);
     m.def("ilu0_apply_block", &ilu0_apply_block);
     m.def("ilu0_apply", &ilu0_apply);
 }

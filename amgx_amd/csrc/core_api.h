@@ -291,6 +291,15 @@ template <typename TA, typename TV>
 void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos,
                      const TA* lu, const int* didx, const int* rows, int count,
                      const TV* y, TV* z, int n, hipStream_t s);
+// color-sorted ILU(0) sweeps (reorder-by-color slabs)
+template <typename TA, typename TV>
+void ilu0_fwd_sorted(const int* ro_s, const int* ci_s, const TA* lu_s,
+                     const int* pos, const int* rows, int count, const TV* r,
+                     TV* y, int n, hipStream_t s);
+template <typename TA, typename TV>
+void ilu0_bwd_sorted(const int* ro_s, const int* ci_s, const TA* lu_s,
+                     const TA* diag_s, const int* pos, const int* rows,
+                     int count, const TV* y, TV* z, int n, hipStream_t s);
 // block ILU(0): L_ik = A_ik U_kk^{-1}, A_ij -= L_ik U_kj; dinv = inverted
 // pivot diagonal blocks, refreshed per color via ilu0_invert_diag_block
 template <typename T>

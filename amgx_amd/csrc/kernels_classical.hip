@@ -576,6 +576,67 @@ void ilu0_bwd_block_launch(const int* ro, const int* ci, const int* pos,
                        count, y, z, n, b);
 }
 
+// ---------------------------------------------------- color-sorted sweeps
+// Same reorder-by-color slab layout as the DILU sweeps (kernels_solve.hip):
+// lu_s/ci_s are rows_sorted-gathered copies, diag_s the per-slot pivot.
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_fwd_sorted_kernel(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ lu_s, const int* __restrict__ pos,
+    const int* __restrict__ rows, int count, const TV* __restrict__ r,
+    TV* __restrict__ y, int n) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    int pi = pos[i];
+    TV sum = r[i];
+    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+        int j = ci_s[k];
+        if (j < n && pos[j] < pi) sum -= (TV)lu_s[k] * y[j];
+    }
+    y[i] = sum;
+}
+
+template <typename TA, typename TV>
+__global__ __launch_bounds__(AMGX_BLOCK) void ilu0_bwd_sorted_kernel(
+    const int* __restrict__ ro_s, const int* __restrict__ ci_s,
+    const TA* __restrict__ lu_s, const TA* __restrict__ diag_s,
+    const int* __restrict__ pos, const int* __restrict__ rows, int count,
+    const TV* __restrict__ y, TV* __restrict__ z, int n) {
+    int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= count) return;
+    int i = rows[t];
+    int pi = pos[i];
+    TV sum = y[i];
+    for (int k = ro_s[t]; k < ro_s[t + 1]; ++k) {
+        int j = ci_s[k];
+        if (j < n && pos[j] > pi) sum -= (TV)lu_s[k] * z[j];
+    }
+    TV d = (TV)diag_s[t];
+    if (d == TV(0)) d = TV(1);
+    z[i] = sum / d;
+}
+
+template <typename TA, typename TV>
+void ilu0_fwd_sorted(const int* ro_s, const int* ci_s, const TA* lu_s,
+                     const int* pos, const int* rows, int count, const TV* r,
+                     TV* y, int n, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((ilu0_fwd_sorted_kernel<TA, TV>),
+                       dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s, ro_s,
+                       ci_s, lu_s, pos, rows, count, r, y, n);
+}
+
+template <typename TA, typename TV>
+void ilu0_bwd_sorted(const int* ro_s, const int* ci_s, const TA* lu_s,
+                     const TA* diag_s, const int* pos, const int* rows,
+                     int count, const TV* y, TV* z, int n, hipStream_t s) {
+    if (count <= 0) return;
+    hipLaunchKernelGGL((ilu0_bwd_sorted_kernel<TA, TV>),
+                       dim3(grid_1d(count)), dim3(AMGX_BLOCK), 0, s, ro_s,
+                       ci_s, lu_s, diag_s, pos, rows, count, y, z, n);
+}
+
 template <typename T>
 void ilu0_factor_color_launch(const int* ro, const int* ci, const int* pos,
                               const int* didx, const int* rows, int count,
@@ -615,6 +676,12 @@ void ilu0_bwd_launch(const int* ro, const int* ci, const int* pos,
                                               const int*, const int*,          \
                                               const int*, int, T*, int,        \
                                               hipStream_t);                    \
+    template void ilu0_fwd_sorted<T, T>(const int*, const int*, const T*,      \
+                                        const int*, const int*, int,            \
+                                        const T*, T*, int, hipStream_t);        \
+    template void ilu0_bwd_sorted<T, T>(const int*, const int*, const T*,       \
+                                        const T*, const int*, const int*,        \
+                                        int, const T*, T*, int, hipStream_t);   \
     template void ilu0_fwd_launch<T, T>(const int*, const int*, const int*,    \
                                         const T*, const int*, int, const T*,   \
                                         T*, int, hipStream_t);                 \
@@ -642,6 +709,15 @@ template void ilu0_fwd_launch<float, double>(const int*, const int*,
                                              double*, int, hipStream_t);
 template void ilu0_bwd_launch<float, double>(const int*, const int*,
                                              const int*, const float*,
+                                             const int*, const int*, int,
+                                             const double*, double*, int,
+                                             hipStream_t);
+template void ilu0_fwd_sorted<float, double>(const int*, const int*,
+                                             const float*, const int*,
+                                             const int*, int, const double*,
+                                             double*, int, hipStream_t);
+template void ilu0_bwd_sorted<float, double>(const int*, const int*,
+                                             const float*, const float*,
                                              const int*, const int*, int,
                                              const double*, double*, int,
                                              hipStream_t);

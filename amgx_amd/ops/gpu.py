@@ -506,7 +506,19 @@ def ilu0_solve(A, factors, coloring, r, x, relaxation=1.0):
                                coloring.bounds, r.reshape(-1), y, z,
                                x.reshape(-1), float(relaxation))
         return x
-    _core.ilu0_apply(A.row_offsets, A.col_indices, factors, _didx(A), pos,
-                     coloring.rows_sorted, coloring.bounds, r.reshape(-1),
-                     y, z, x.reshape(-1), float(relaxation))
+    # color-sorted slabs (same layout win as the DILU sweeps)
+    ro_s, ci_s, nzpos, perm = _color_sorted_struct(A, coloring)
+    lkey = ("ilu_lu_s", id(coloring), factors._version, factors.data_ptr())
+    cached = A._cache.get(lkey)
+    if cached is None:
+        lu_s = factors.reshape(-1)[nzpos].contiguous()
+        diag_s = factors.reshape(-1)[_didx(A).to(torch.int64)][perm] \
+            .contiguous()
+        A._cache[lkey] = (lu_s, diag_s)
+    else:
+        lu_s, diag_s = cached
+    _core.ilu0_apply_sorted(ro_s, ci_s, lu_s, diag_s, pos,
+                            coloring.rows_sorted, coloring.bounds,
+                            r.reshape(-1), y, z, x.reshape(-1),
+                            float(relaxation))
     return x
