@@ -99,6 +99,10 @@ void bsrmv_b4(const int* ro, const int* ci, const TA* va, const TV* x, TV* y,
               const TV* bvec, double alpha, double beta, double gamma,
               int row_begin, int row_end, hipStream_t s);
 template <typename TA, typename TV>
+void bsrmv_bn(const int* ro, const int* ci, const TA* va, int b, const TV* x,
+              TV* y, const TV* bvec, double alpha, double beta, double gamma,
+              int row_begin, int row_end, hipStream_t s);
+template <typename TA, typename TV>
 void dilu_fwd_b4(const int* ro, const int* ci, const TA* va, const TA* einv,
                  const int* rows, int count, const TV* r, TV* w,
                  hipStream_t s);

@@ -328,6 +328,73 @@ void bsrmv_b4(const int* ro, const int* ci, const TA* va, const TV* x,
                        gamma, row_begin, row_end);
 }
 
+// generic wave-structured block SpMV for b in 2..8: GROUP lanes per row
+// (first B*B active, element (r=t/B, q=t%B)) so va[k*B*B + t] is a
+// coalesced contiguous load per nnz; component sums fold with B-1 wave
+// shuffles (arbitrary-B segments, so plain __shfl reads instead of xor).
+template <typename TA, typename TV, int B, int GROUP>
+__global__ __launch_bounds__(256) void bsrmv_bn_kernel(
+    const int* __restrict__ ro, const int* __restrict__ ci,
+    const TA* __restrict__ va, const TV* __restrict__ x,
+    TV* __restrict__ y, const TV* __restrict__ bvec, double alpha,
+    double beta, double gamma, int row_begin, int row_end) {
+    constexpr int RPW = 64 / GROUP;          // rows per wave
+    int lane = threadIdx.x & 63;
+    int g = lane / GROUP;
+    int t = lane % GROUP;
+    int i = row_begin + blockIdx.x * (4 * RPW) + (threadIdx.x >> 6) * RPW
+            + g;
+    bool active = (t < B * B) && (i < row_end);
+    int r = t / B;
+    int q = t - r * B;
+    double acc = 0.0;
+    if (active) {
+        int k0 = ro[i], k1 = ro[i + 1];
+        for (int k = k0; k < k1; ++k) {
+            double a = (double)va[(long long)k * B * B + t];
+            double xv = (double)x[(long long)ci[k] * B + q];
+            acc += a * xv;
+        }
+    }
+    // fold over q: lanes (laneBase + r*B + 0..B-1); all lanes shuffle
+    int base = lane - t;
+    double s = acc;
+#pragma unroll
+    for (int qq = 1; qq < B; ++qq)
+        s += __shfl(acc, base + r * B + qq, 64);
+    if (active && q == 0) {
+        long long idx = (long long)i * B + r;
+        double out = alpha * s;
+        if (beta != 0.0) out += beta * (double)y[idx];
+        if (gamma != 0.0 && bvec) out += gamma * (double)bvec[idx];
+        y[idx] = (TV)out;
+    }
+}
+
+template <typename TA, typename TV>
+void bsrmv_bn(const int* ro, const int* ci, const TA* va, int b,
+              const TV* x, TV* y, const TV* bvec, double alpha, double beta,
+              double gamma, int row_begin, int row_end, hipStream_t s) {
+    int count = row_end - row_begin;
+    if (count <= 0) return;
+    auto launch = [&](auto kern, int rpw) {
+        int rows_per_wg = 4 * rpw;
+        hipLaunchKernelGGL(kern, dim3((count + rows_per_wg - 1)
+                                      / rows_per_wg), dim3(256), 0, s, ro,
+                           ci, va, x, y, bvec, alpha, beta, gamma,
+                           row_begin, row_end);
+    };
+    switch (b) {
+        case 2: launch(bsrmv_bn_kernel<TA, TV, 2, 4>, 16); break;
+        case 3: launch(bsrmv_bn_kernel<TA, TV, 3, 16>, 4); break;
+        case 5: launch(bsrmv_bn_kernel<TA, TV, 5, 32>, 2); break;
+        case 6: launch(bsrmv_bn_kernel<TA, TV, 6, 64>, 1); break;
+        case 7: launch(bsrmv_bn_kernel<TA, TV, 7, 64>, 1); break;
+        case 8: launch(bsrmv_bn_kernel<TA, TV, 8, 64>, 1); break;
+        default: break;   // caller guards
+    }
+}
+
 // ------------------------------------------------------------ instantiation
 #define INSTANTIATE_MFMA_MIXED(TA, TV)                                        \
     template void dilu_fwd_b4<TA, TV>(const int*, const int*, const TA*,      \
@@ -350,6 +417,9 @@ void bsrmv_b4(const int* ro, const int* ci, const TA* va, const TV* x,
                                              const int*, int, const TV*,      \
                                              TV*, hipStream_t);               \
     template void bsrmv_b4<TA, TV>(const int*, const int*, const TA*,         \
+                                   const TV*, TV*, const TV*, double, double, \
+                                   double, int, int, hipStream_t);            \
+    template void bsrmv_bn<TA, TV>(const int*, const int*, const TA*, int,    \
                                    const TV*, TV*, const TV*, double, double, \
                                    double, int, int, hipStream_t);
 

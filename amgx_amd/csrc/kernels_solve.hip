@@ -121,6 +121,11 @@ void bsrmv(const int* ro, const int* ci, const TA* va, int b, const TV* x,
                          (double)gamma, r0, r1, s);
         return;
     }
+    if (b >= 2 && b <= 8) {   // generic wave kernel, coalesced block loads
+        bsrmv_bn<TA, TV>(ro, ci, va, b, x, y, bvec, (double)alpha,
+                         (double)beta, (double)gamma, r0, r1, s);
+        return;
+    }
     hipLaunchKernelGGL((bsrmv_kernel<TA, TV>), dim3(grid_1d(rows)),
                        dim3(AMGX_BLOCK), 0, s, ro, ci, va, b, x, y, bvec,
                        alpha, beta, gamma, r0, r1);
