@@ -7,7 +7,7 @@ src/cycles/fixed_cycle.cu:59-230 (the recursive cycle engine), src/amg_level.cu.
 from __future__ import annotations
 
 import time
-from typing import List, Optional
+from typing import List
 
 import torch
 

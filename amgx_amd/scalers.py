@@ -11,7 +11,6 @@ from __future__ import annotations
 import numpy as np
 import torch
 
-from . import ops
 
 
 class Scaler:

@@ -171,8 +171,6 @@ bool copy_out(PyObject *arr, void *dst) {
 }
 
 // handle <-> PyObject: the opaque C handle IS a new reference
-template <typename H>
-H wrap(PyObject *obj) { return (H)obj; }
 PyObject *obj(const void *h) { return (PyObject *)h; }
 
 // NOTE: Py_BuildValue needs the GIL, so the helpers take a format +
