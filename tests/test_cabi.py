@@ -200,3 +200,112 @@ int main(){
                        timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "ROUNDTRIP_OK" in r.stdout
+
+
+def test_c_abi_resetup_and_replace(tmp_path):
+    """AMGX_matrix_replace_coefficients + AMGX_solver_resetup through the
+    C ABI (reference include/amgx_c.h:281,603 — the cheap re-setup path)."""
+    _build()
+    src = tmp_path / "rs.c"
+    src.write_text(r'''
+#include <stdio.h>
+#include <math.h>
+#include "amgx_c.h"
+int main(){
+    AMGX_SAFE_CALL(AMGX_initialize());
+    AMGX_config_handle cfg;
+    AMGX_SAFE_CALL(AMGX_config_create(&cfg, "config_version=2, solver=PCG, max_iters=50, monitor_residual=1, tolerance=1e-10"));
+    AMGX_resources_handle r;
+    AMGX_SAFE_CALL(AMGX_resources_create_simple(&r, cfg));
+    AMGX_matrix_handle A; AMGX_vector_handle b, x; AMGX_solver_handle s;
+    AMGX_SAFE_CALL(AMGX_matrix_create(&A, r, AMGX_mode_hDDI));
+    AMGX_SAFE_CALL(AMGX_vector_create(&b, r, AMGX_mode_hDDI));
+    AMGX_SAFE_CALL(AMGX_vector_create(&x, r, AMGX_mode_hDDI));
+    AMGX_SAFE_CALL(AMGX_solver_create(&s, r, AMGX_mode_hDDI, cfg));
+    int n = 6;
+    int ro[7] = {0,2,5,8,11,14,16};
+    int ci[16] = {0,1, 0,1,2, 1,2,3, 2,3,4, 3,4,5, 4,5};
+    double va[16] = {2,-1, -1,2,-1, -1,2,-1, -1,2,-1, -1,2,-1, -1,2};
+    AMGX_SAFE_CALL(AMGX_matrix_upload_all(A, n, 16, 1, 1, ro, ci, va, NULL));
+    double rhs[6] = {1,1,1,1,1,1};
+    AMGX_SAFE_CALL(AMGX_vector_upload(b, n, 1, rhs));
+    AMGX_SAFE_CALL(AMGX_vector_set_zero(x, n, 1));
+    AMGX_SAFE_CALL(AMGX_solver_setup(s, A));
+    AMGX_SAFE_CALL(AMGX_solver_solve(s, b, x));
+    /* scale the values, replace, resetup, solve again */
+    double va2[16];
+    for (int k = 0; k < 16; ++k) va2[k] = 3.0 * va[k];
+    AMGX_SAFE_CALL(AMGX_matrix_replace_coefficients(A, n, 16, va2, NULL));
+    AMGX_SAFE_CALL(AMGX_solver_resetup(s, A));
+    AMGX_SAFE_CALL(AMGX_vector_set_zero(x, n, 1));
+    AMGX_SAFE_CALL(AMGX_solver_solve(s, b, x));
+    double sol[6];
+    AMGX_SAFE_CALL(AMGX_vector_download(x, sol));
+    /* A2 = 3A -> residual of A2 x = b must be small */
+    double nrm = 0;
+    for (int i = 0; i < n; ++i) {
+        double ri = rhs[i];
+        for (int k = ro[i]; k < ro[i+1]; ++k) ri -= va2[k]*sol[ci[k]];
+        nrm += ri*ri;
+    }
+    if (sqrt(nrm) > 1e-8) { printf("BAD %e\n", sqrt(nrm)); return 1; }
+    printf("RESETUP_OK\n");
+    return 0;
+}
+''')
+    exe = tmp_path / "rs"
+    r = subprocess.run(
+        ["cc", "-O2", f"-I{REPO}/include", "-o", str(exe), str(src),
+         f"-L{REPO}/csrc_capi", "-lamgx_amd",
+         f"-Wl,-rpath,{REPO}/csrc_capi", "-lm"],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    r = subprocess.run([str(exe)], cwd=REPO, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0 and "RESETUP_OK" in r.stdout, \
+        r.stdout + r.stderr
+
+
+def test_c_abi_poisson_generator(tmp_path):
+    """AMGX_generate_distributed_poisson_7pt through the C ABI at 1 rank
+    (reference src/amgx_c.cu:4566-4731)."""
+    _build()
+    src = tmp_path / "pg.c"
+    src.write_text(r'''
+#include <stdio.h>
+#include "amgx_c.h"
+int main(){
+    AMGX_SAFE_CALL(AMGX_initialize());
+    AMGX_config_handle cfg;
+    AMGX_SAFE_CALL(AMGX_config_create(&cfg, "config_version=2, solver=PCG, max_iters=200, monitor_residual=1, tolerance=1e-8"));
+    AMGX_resources_handle r;
+    AMGX_SAFE_CALL(AMGX_resources_create_simple(&r, cfg));
+    AMGX_matrix_handle A; AMGX_vector_handle b, x; AMGX_solver_handle s;
+    AMGX_SAFE_CALL(AMGX_matrix_create(&A, r, AMGX_mode_hDDI));
+    AMGX_SAFE_CALL(AMGX_vector_create(&b, r, AMGX_mode_hDDI));
+    AMGX_SAFE_CALL(AMGX_vector_create(&x, r, AMGX_mode_hDDI));
+    AMGX_SAFE_CALL(AMGX_generate_distributed_poisson_7pt(A, b, x, 1, 1, 8, 8, 8, 1, 1, 1));
+    int n, bx, by;
+    AMGX_SAFE_CALL(AMGX_matrix_get_size(A, &n, &bx, &by));
+    if (n != 512) { printf("BAD_N %d\n", n); return 1; }
+    AMGX_SAFE_CALL(AMGX_solver_create(&s, r, AMGX_mode_hDDI, cfg));
+    AMGX_SAFE_CALL(AMGX_solver_setup(s, A));
+    AMGX_SAFE_CALL(AMGX_solver_solve_with_0_initial_guess(s, b, x));
+    AMGX_SOLVE_STATUS st;
+    AMGX_SAFE_CALL(AMGX_solver_get_status(s, &st));
+    if (st != AMGX_SOLVE_SUCCESS) { printf("NOT_CONVERGED\n"); return 1; }
+    printf("POISSON7_OK\n");
+    return 0;
+}
+''')
+    exe = tmp_path / "pg"
+    r = subprocess.run(
+        ["cc", "-O2", f"-I{REPO}/include", "-o", str(exe), str(src),
+         f"-L{REPO}/csrc_capi", "-lamgx_amd",
+         f"-Wl,-rpath,{REPO}/csrc_capi"],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    r = subprocess.run([str(exe)], cwd=REPO, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0 and "POISSON7_OK" in r.stdout, \
+        r.stdout + r.stderr
