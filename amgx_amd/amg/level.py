@@ -67,6 +67,9 @@ class AggregationLevel(AMGLevel):
         super().__init__(A, scope, index)
         self.aggregates = None
         self.num_aggregates = 0
+        # error-scaling state (reference aggregation_amg_level.cu:700-825)
+        self._scale = None
+        self._scale_counter = 0
 
     def create_coarse_vertices(self) -> int:
         from .aggregation import AGG_SELECTOR_REGISTRY
@@ -230,7 +233,61 @@ class AggregationLevel(AMGLevel):
         bc.reshape(-1)[:out.numel()].copy_(out.reshape(-1))
 
     def prolongate_and_apply(self, xc, x):
+        es = int(self.scope.get("error_scaling") or 0)
+        if es in (2, 3):
+            self._prolongate_scaled(xc, x, es)
+            return
         ops.prolongate_agg(x, xc, self.aggregates, self.A.block_dim)
+
+    def _prolongate_scaled(self, xc, x, mode: int):
+        """Scaled coarse-grid correction x += lambda * (smoothed P xc)
+        (reference src/aggregation/aggregation_amg_level.cu:700-825).
+        mode 2 minimizes the residual 2-norm: lambda = <r,Ae>/<Ae,Ae>;
+        mode 3 minimizes the A-norm error (SPD): lambda = <r,e>/<e,Ae>.
+        The fine residual is self.r, computed just before restriction and
+        still current here (x has not changed since). lambda is clamped to
+        0.3 <= |lambda| <= 10 and cached for reuse_scale iterations."""
+        import math
+
+        if self._scale_counter > 0 and self._scale is not None:
+            ef = torch.zeros_like(x)
+            ops.prolongate_agg(ef, xc, self.aggregates, self.A.block_dim)
+            ops.axpy(x, ef, self._scale)
+            self._scale_counter -= 1
+            return
+        ef = torch.zeros_like(x)
+        ops.prolongate_agg(ef, xc, self.aggregates, self.A.block_dim)
+        steps = int(self.scope.get("scaling_smoother_steps") or 2)
+        if self.smoother is not None and steps > 0:
+            # smooth the correction against the pre-restriction residual
+            # (warm start at P xc): approximately solves A e = r
+            self.smoother.sweep(self.r, ef, steps)
+        Aef = torch.zeros_like(ef)
+        ops.spmv(self.A, ef, Aef)
+        mgr = getattr(self.A, "manager", None)
+
+        def gdot(u, v):
+            if mgr is not None:
+                d = ops.dot(u.reshape(-1)[:mgr.owned_size],
+                            v.reshape(-1)[:mgr.owned_size])
+                return float(mgr.global_sum(d))
+            return float(ops.dot(u, v))
+
+        if mode == 2:
+            num, den = gdot(self.r, Aef), gdot(Aef, Aef)
+        else:
+            num, den = gdot(self.r, ef), gdot(ef, Aef)
+        if den == 0.0:
+            alpha = 1.0
+        else:
+            alpha = num / den
+            if abs(alpha) < 0.3:
+                alpha = math.copysign(0.3, alpha)
+            elif abs(alpha) > 10.0:
+                alpha = math.copysign(10.0, alpha)
+        ops.axpy(x, ef, alpha)
+        self._scale = alpha
+        self._scale_counter = int(self.scope.get("reuse_scale") or 0)
 
 
 class ClassicalLevel(AMGLevel):
