@@ -123,9 +123,10 @@ class AggregationLevel(AMGLevel):
 
     def create_coarse_matrix(self) -> CSRMatrix:
         mgr = getattr(self.A, "manager", None)
+        gen = self.scope.get("coarseAgenerator")
         if mgr is None:
             Ac = ops.galerkin_aggregation(self.A, self.aggregates,
-                                          self.num_aggregates)
+                                          self.num_aggregates, generator=gen)
             Ac = self._color_renumber(Ac)
             geom = self.A._cache.get("geometry")
             if geom is not None:
@@ -161,8 +162,9 @@ class AggregationLevel(AMGLevel):
         aggv[:mgr.n_local] = self.aggregates.to(torch.float64) + coarse_start
         mgr.exchange_halo(aggv, block_override=1)
         agg_col = aggv.round().to(torch.int32)
-        Ac_local = ops.galerkin_aggregation(A, self.aggregates, nc_local,
-                                            agg_col, n_global_c)
+        Ac_local = ops.galerkin_aggregation(
+            A, self.aggregates, nc_local, agg_col, n_global_c,
+            generator=self.scope.get("coarseAgenerator"))
         # rebuild distributed structure from global column ids; tensors stay
         # on device (upload_global_csr is torch-native)
         Ac = DistributedManager.upload_global_csr(
@@ -206,7 +208,9 @@ class AggregationLevel(AMGLevel):
             .to(torch.int32)
         self._build_r_structure()     # restriction map follows the new ids
         Ac2 = ops.galerkin_aggregation(self.A, self.aggregates,
-                                       self.num_aggregates)
+                                       self.num_aggregates,
+                                       generator=self.scope.get(
+                                           "coarseAgenerator"))
         # colors in the new numbering are ascending by construction
         self._coarse_coloring = MatrixColoring(
             col.colors[perm].contiguous(), col.num_colors)
@@ -218,7 +222,9 @@ class AggregationLevel(AMGLevel):
                 or getattr(self.A, "manager", None) is not None:
             return None
         Ac = ops.galerkin_aggregation(self.A, self.aggregates,
-                                      self.num_aggregates)
+                                      self.num_aggregates,
+                                      generator=self.scope.get(
+                                          "coarseAgenerator"))
         # aggregates already carry the color renumbering; re-attach the
         # structure-valid coloring so sweeps keep the identity ordering
         coloring = getattr(self, "_coarse_coloring", None)

@@ -173,13 +173,16 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
 
 
 def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
-                         ncols_mod=None):
+                         ncols_mod=None, generator=None):
     """Coarse A for piecewise-constant aggregation P:
     Ac[I,J] = sum_{i in I, j in J} A[i,j].
     Reference: src/aggregation/coarseAgenerators/ (LOW_DEG / THRUST).
-    agg_col/ncols_mod: distributed variant (GLOBAL coarse column ids)."""
+    agg_col/ncols_mod: distributed variant (GLOBAL coarse column ids).
+    generator: None/"LOW_DEG"/"HYBRID" = LDS-hash with one-sort fallback
+    (the default tiering IS the HYBRID role); "THRUST" = force the one-sort
+    sort/reduce generator (reference coarseAgenerators/thrust_...)."""
     return _backend(A).galerkin_aggregation(A, aggregates, num_aggregates,
-                                            agg_col, ncols_mod)
+                                            agg_col, ncols_mod, generator)
 
 
 def restrict_agg(r, aggregates, num_aggregates, block_dim: int = 1,

@@ -345,7 +345,9 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
 
 
 def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
-                         ncols_mod=None):
+                         ncols_mod=None, generator=None):
+    # generator: the host path has a single sort/reduce implementation
+    # (the THRUST analogue); LOW_DEG/HYBRID map onto it unchanged
     from ..matrix import CSRMatrix
     agg = _np(aggregates).astype(np.int64)
     n = A.n_rows

@@ -313,20 +313,21 @@ def size2_matching(A, max_iterations: int = 15, deterministic: bool = True,
 
 
 def galerkin_aggregation(A, aggregates, num_aggregates, agg_col=None,
-                         ncols_mod=None):
+                         ncols_mod=None, generator=None):
     """agg_col/ncols_mod: distributed variant — per-column coarse ids (GLOBAL)
     and the global coarse column count; defaults to the single-process case.
 
     Scalar matrices run the LDS-hash LOW_DEG-style generator
     (kernels_spgemm.hip mode 1: wave-per-coarse-row hash accumulation —
-    reference src/aggregation/coarseAgenerators/low_deg_…); block matrices
-    and hash-capacity overflows use the one-sort rocPRIM generator."""
+    reference src/aggregation/coarseAgenerators/low_deg_…); block matrices,
+    hash-capacity overflows, and generator=="THRUST" use the one-sort
+    rocPRIM generator (the tiered dispatch is the reference HYBRID role)."""
     from ..matrix import CSRMatrix
     if agg_col is None:
         agg_col = aggregates
     if ncols_mod is None:
         ncols_mod = num_aggregates
-    if A.block_dim == 1:
+    if A.block_dim == 1 and generator != "THRUST":
         # aggregate-membership CSR: coarse row -> fine member rows
         agg64 = aggregates.to(torch.int64)
         counts = torch.bincount(agg64, minlength=num_aggregates)
