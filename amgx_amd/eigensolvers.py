@@ -201,10 +201,20 @@ class EigenSolver:
         if self.which == "smallest":
             from .solvers import create_solver
             from .config import ConfigScope as CS
-            inner = create_solver("PCG", CS(None, {
-                "max_iters": 200, "monitor_residual": 1, "tolerance": 1e-10,
-                "convergence": "RELATIVE_INI",
-                "preconditioner": "BLOCK_JACOBI"}), self.res)
+            # inverse iteration: the config's "solver" tree, when present,
+            # configures the inner A^{-1} solve (reference
+            # eigen_configs/INVERSE_FGMRES drives an FGMRES+AMG inner solver
+            # through the same scoped-config mechanism); default = PCG+BJ
+            sub = self.scope.node.get("solver") \
+                if isinstance(self.scope.node, dict) else None
+            if isinstance(sub, dict):
+                inner = create_solver(self.scope.child(sub),
+                                      resources=self.res)
+            else:
+                inner = create_solver("PCG", CS(None, {
+                    "max_iters": 200, "monitor_residual": 1,
+                    "tolerance": 1e-10, "convergence": "RELATIVE_INI",
+                    "preconditioner": "BLOCK_JACOBI"}), self.res)
             inner.setup(A)
             self.op = SolverOperator(A, inner)
         elif self.shift:
