@@ -256,8 +256,16 @@ def aggressive_select(A, S, scope, base: str):
     sub.eliminate_zeros()
     lam = np.asarray(adj.sum(0)).ravel()[c_idx].astype(np.float64)
     cf_sub, nc2 = _mis_select(sub, lam)
+    keep_mask = cf_sub.numpy() >= 0
+    # _mis_select marks degree-0 nodes fine (right for the base pass, where
+    # isolated fine points need no coarse image) — but here the nodes are
+    # the SURVIVING C points: one with no <=2-hop strong C neighbor has no
+    # one to merge with and must STAY coarse (HMIS C sets are 2-ring
+    # independent, so the whole sub-graph is edgeless and dropping isolated
+    # nodes would return nc=0)
+    keep_mask |= np.diff(sub.indptr) == 0
     cf = np.full(A.n_rows, -1, dtype=np.int32)
-    keep = c_idx[cf_sub.numpy() >= 0]
+    keep = c_idx[keep_mask]
     cf[keep] = np.arange(keep.size, dtype=np.int32)
     return torch.from_numpy(cf), int(keep.size)
 
@@ -528,12 +536,19 @@ def _interp_multipass_device(A, S, cf_map, num_coarse, scope,
     diag = torch.zeros(n, dtype=torch.float64, device=dev)
     dmask = rows == ci
     diag[rows[dmask]] = v[dmask]
-    wmask = offd & ~strong
-    weak_lump = torch.zeros(n, dtype=torch.float64, device=dev)
-    weak_lump.index_add_(0, rows[wmask], v[wmask])
-    denom = diag + weak_lump
     sedge = strong & offd & local
     ci_n = torch.clamp(ci, max=n - 1)
+    # isolated fine rows = reference STRONG_FINE: excluded from sum_N
+    iso = torch.zeros(n, dtype=torch.bool, device=dev)
+    iso[:] = True
+    iso[rows[sedge]] = False
+    iso &= cf[:n] < 0
+    excl = local & iso[ci_n]
+    # hypre/reference normalization mass (multipass.cu:1127-1191):
+    # alfa = -sum_N / (sum_C * diag) keeps P row sums ~1
+    nmask = offd & ~excl
+    sum_n = torch.zeros(n, dtype=torch.float64, device=dev)
+    sum_n.index_add_(0, rows[nmask], v[nmask])
     done = cf[:n] >= 0
     c_rows = torch.nonzero(done, as_tuple=True)[0]
 
@@ -563,9 +578,15 @@ def _interp_multipass_device(A, S, cf_map, num_coarse, scope,
             break
         ok_rows = torch.zeros(n, dtype=torch.bool, device=dev)
         ok_rows[rows[e]] = True
-        ok_rows &= denom != 0.0
         if not bool(ok_rows.any()):
             break
+        sum_c = torch.zeros(n, dtype=torch.float64, device=dev)
+        sum_c.index_add_(0, rows[e], v[e])
+        div = sum_c * diag
+        div = torch.where(div.abs() == 0.0, torch.ones_like(div), div)
+        scale = torch.where(
+            ok_rows, -sum_n / div,
+            torch.zeros(n, dtype=torch.float64, device=dev))
         # F (undone -> done edges) and current P as device CSR
         ecnt = torch.bincount(rows[e], minlength=n)
         f_ro = torch.zeros(n + 1, dtype=torch.int32, device=dev)
@@ -574,10 +595,6 @@ def _interp_multipass_device(A, S, cf_map, num_coarse, scope,
                       v[e].contiguous(), n_cols=n)
         P = _coo_to_csr(P_r, P_c, P_v, nc)
         contrib = ops._backend(A).spgemm(F, P)
-        scale = torch.where(
-            ok_rows, -1.0 / torch.where(denom != 0.0, denom,
-                                        torch.ones_like(denom)),
-            torch.zeros(n, dtype=torch.float64, device=dev))
         cdeg = contrib.row_offsets.to(torch.int64)
         c_rows2 = torch.repeat_interleave(
             torch.arange(n, dtype=torch.int64, device=dev),
@@ -615,10 +632,18 @@ def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
     diag = np.zeros(n)
     dmask = rows == ci
     diag[rows[dmask]] = v[dmask]
-    weak_lump = np.bincount(rows[offd & ~strong], weights=v[offd & ~strong],
-                            minlength=n)
-    denom = diag + weak_lump
     sedge = strong & offd & local        # strong edges usable for composing
+    # isolated fine rows (no strong connections) play the reference's
+    # STRONG_FINE role: excluded from the neighborhood sum below
+    iso = np.bincount(rows[sedge], minlength=n) == 0
+    iso &= cf[:n] < 0
+    excl = local & iso[np.minimum(ci, n - 1)]
+    # full neighborhood mass sum_N (hypre/reference normalization: the
+    # total off-diagonal mass is redistributed over the interpolatory set
+    # so P rows sum to ~1 and constants are preserved — reference
+    # multipass.cu:1127-1191 alfa = -sum_N / (sum_C * diag))
+    sum_n = np.bincount(rows[offd & ~excl], weights=v[offd & ~excl],
+                        minlength=n)
     done = cf[:n] >= 0
     c_rows = np.nonzero(done)[0]
     P = sp.csr_matrix((np.ones(c_rows.size), (c_rows, cf[c_rows])),
@@ -633,14 +658,15 @@ def interp_multipass(A, S, cf_map, num_coarse, scope, max_passes: int = 10):
             break
         ok_rows = np.zeros(n, dtype=bool)
         ok_rows[rows[e]] = True
-        ok_rows &= denom != 0.0
         if not ok_rows.any():
             break
+        sum_c = np.bincount(rows[e], weights=v[e], minlength=n)
+        div = sum_c * diag
+        div = np.where(np.abs(div) == 0.0, 1.0, div)   # hypre zero guard
+        alfa = np.where(ok_rows, -sum_n / div, 0.0)
         F = sp.csr_matrix((v[e], (rows[e], ci[e])), shape=(n, n))
         contrib = (F @ P).tocsr()
-        scale = np.where(ok_rows, np.divide(-1.0, denom, out=np.ones(n),
-                                            where=denom != 0.0), 0.0)
-        newP = sp.diags(scale) @ contrib
+        newP = sp.diags(alfa) @ contrib
         P = (P + newP).tocsr()
         done = done | ok_rows
     P.sum_duplicates()
