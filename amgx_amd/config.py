@@ -130,6 +130,9 @@ def _register_core_parameters() -> None:
     P("GS_L1_variant", int, 0, "L1 variant of Gauss-Seidel")
     P("ilu_sparsity_level", int, 0, "ILU(k) level (0 or 1)")
     P("kaczmarz_coloring_needed", int, 1, "accepted for parity")
+    P("cf_smoothing_mode", int, 0,
+      "CF-Jacobi sweep order: 0=CF pre / FC post, 1=opposite "
+      "(reference src/core.cu:416)")
     # --- scalers / misc ---------------------------------------------------------------
     P("scaling", str, "NONE", "system pre-scaling",
       ("NONE", "BINORMALIZATION", "NBINORMALIZATION", "DIAGONAL_SYMMETRIC"))

@@ -2,7 +2,7 @@
 (reference: registerClasses, src/core.cu:596-628)."""
 
 from .base import (Solver, SolveStatus, Convergence, create_solver,
-                   register_solver, SOLVER_REGISTRY)
+                   register_solver, SOLVER_REGISTRY, UserSolver)
 from . import krylov            # noqa: F401  (registers CG/PCG/BiCGStab/FGMRES/IDR)
 from . import smoothers         # noqa: F401  (registers Jacobi/GS/Chebyshev)
 from . import dilu              # noqa: F401  (registers MULTICOLOR_DILU)
@@ -12,4 +12,4 @@ from . import ilu               # noqa: F401  (registers MULTICOLOR_ILU)
 from . import kaczmarz          # noqa: F401  (registers KACZMARZ/CF_JACOBI)
 
 __all__ = ["Solver", "SolveStatus", "Convergence", "create_solver",
-           "register_solver", "SOLVER_REGISTRY"]
+           "register_solver", "SOLVER_REGISTRY", "UserSolver"]

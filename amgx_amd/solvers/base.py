@@ -343,6 +343,36 @@ class Solver:
         return torch.zeros_like(like)
 
 
+@register_solver("USER")
+class UserSolver(Solver):
+    """Application-plugin solver (reference src/solvers/user_solver.cu):
+    each iteration calls a user-registered callback(A, b, x) that updates x
+    in place; convergence is monitored by the normal machinery. Set the
+    callback with ``UserSolver.set_callback(fn)`` (process-global, like the
+    reference's setCallback) or pass one per-instance via
+    ``solver.callback = fn`` before solve. ``register_solver`` remains the
+    richer plugin path for full custom solver classes."""
+    is_smoother = False
+    _global_callback = None
+
+    @classmethod
+    def set_callback(cls, fn):
+        cls._global_callback = fn
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        self.callback = None
+
+    def solve_iteration(self, b, x):
+        cb = self.callback or type(self)._global_callback
+        if cb is None:
+            raise RuntimeError(
+                "USER solver has no callback: call "
+                "amgx_amd.solvers.UserSolver.set_callback(fn) first")
+        cb(self.A, b, x)
+        return False
+
+
 @register_solver("NOSOLVER")
 @register_solver("DUMMY")
 class DummySolver(Solver):
