@@ -68,6 +68,36 @@ def _host_adj(A, level: int = 1):
     return g.indptr, g.indices
 
 
+def _square_graph_matrix(A):
+    """Distance-2 structure as a CSRMatrix on A's device: halo columns
+    dropped, then pattern(A) @ pattern(A) through the backend SpGEMM (the
+    LDS-hash kernel on gfx950 — reference *_2RING schemes square the graph
+    the same way, src/matrix_coloring/min_max_2ring.cu). A's diagonal keeps
+    the 1-ring inside the square, so a coloring of the square is a valid
+    distance-2 coloring of A."""
+    from ..matrix import CSRMatrix
+    n = A.n_rows
+    dev = A.row_offsets.device
+    ci = A.col_indices
+    keep = ci < n
+    if bool(keep.all()):
+        ro, cols = A.row_offsets, A.col_indices
+    else:
+        deg = A.row_offsets.to(torch.int64)
+        rows = torch.repeat_interleave(
+            torch.arange(n, dtype=torch.int64, device=dev),
+            deg[1:] - deg[:-1])
+        kept = torch.bincount(rows[keep], minlength=n)
+        ro = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(kept, 0, out=ro[1:])
+        ro = ro.to(torch.int32)
+        cols = ci[keep].contiguous()
+    P = CSRMatrix(ro, cols,
+                  torch.ones(int(cols.numel()), dtype=torch.float64,
+                             device=dev), n_cols=n)
+    return ops.spgemm(P, P)
+
+
 def _greedy(indptr, indices, order):
     n = indptr.size - 1
     colors = np.full(n, -1, dtype=np.int32)
@@ -90,10 +120,15 @@ def _hash(i: np.ndarray, salt: int) -> np.ndarray:
 
 @register_coloring("MIN_MAX")
 def _min_max(A, scope, level):
+    frac = scope.get("max_uncolored_percentage") if scope is not None \
+        else 0.0
     if level <= 1:
-        frac = scope.get("max_uncolored_percentage") if scope is not None \
-            else 0.0
         return ops.color_matrix(A, max_uncolored_frac=float(frac or 0))
+    if A.row_offsets.is_cuda:
+        # device-resident distance-2: color the squared graph with the same
+        # gfx950 kernel (no host round-trip mid-setup)
+        return ops.color_matrix(_square_graph_matrix(A),
+                                max_uncolored_frac=float(frac or 0))
     return _min_max_rounds(*_host_adj(A, level))
 
 
@@ -132,8 +167,9 @@ def _min_max_rounds(indptr, indices):
 def _parallel_greedy(A, scope, level):
     # Jones-Plassmann with smallest-feasible-color (same algorithm the gfx950
     # kernel runs); host model is sequential over JP rounds
-    if level <= 1 and A.row_offsets.is_cuda:
-        return ops.color_matrix(A, seed=7)
+    if A.row_offsets.is_cuda:
+        return ops.color_matrix(A if level <= 1 else _square_graph_matrix(A),
+                                seed=7)
     indptr, indices = _host_adj(A, level)
     n = indptr.size - 1
     colors = np.full(n, -1, dtype=np.int32)
@@ -204,9 +240,10 @@ def _greedy_recolor(A, scope, level):
 
 @register_coloring("MULTI_HASH")
 def _multi_hash(A, scope, level):
-    if level <= 1 and A.row_offsets.is_cuda:
+    if A.row_offsets.is_cuda:
         max_hash = int(scope.get("max_num_hash")) if scope is not None else 7
-        return ops.color_matrix(A, seed=11, multihash_rounds=max_hash)
+        return ops.color_matrix(A if level <= 1 else _square_graph_matrix(A),
+                                seed=11, multihash_rounds=max_hash)
     indptr, indices = _host_adj(A, level)
     n = indptr.size - 1
     max_hash = int(scope.get("max_num_hash")) if scope is not None else 7
@@ -267,11 +304,17 @@ def _uniform(A, scope, level):
 
 @register_coloring("MIN_MAX_2RING")
 def _min_max_2ring(A, scope, level):
+    if A.row_offsets.is_cuda:
+        return ops.color_matrix(_square_graph_matrix(A))
     return _min_max_rounds(*_host_adj(A, 2))
 
 
 @register_coloring("GREEDY_MIN_MAX_2RING")
 def _greedy_min_max_2ring(A, scope, level):
+    if A.row_offsets.is_cuda:
+        # device role-equivalent: JP smallest-feasible-color on the squared
+        # graph (greedy color counts, distance-2 validity)
+        return ops.color_matrix(_square_graph_matrix(A), seed=3)
     indptr, indices = _host_adj(A, 2)
     return _greedy(indptr, indices, np.arange(indptr.size - 1))
 

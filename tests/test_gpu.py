@@ -942,3 +942,24 @@ print("NCCL_SMOKE_OK")
                                __import__("os").path.abspath(__file__))))
     assert r.returncode == 0 and "NCCL_SMOKE_OK" in r.stdout, \
         r.stdout[-2000:] + r.stderr[-2000:]
+
+
+def test_device_2ring_colorings():
+    """Distance-2 colorings stay device-resident: the *_2RING schemes (and
+    coloring_level=2 of MIN_MAX/PARALLEL_GREEDY/MULTI_HASH) square the
+    graph with the device hash SpGEMM and color it with the gfx950 kernel
+    (reference src/matrix_coloring/min_max_2ring.cu role)."""
+    A = to_gpu(poisson_3d(8, 7, 6))
+    for scheme in ("MIN_MAX_2RING", "GREEDY_MIN_MAX_2RING"):
+        cfg = AMGConfig.from_dict({"solver": {
+            "solver": "MULTICOLOR_GS", "matrix_coloring_scheme": scheme}})
+        col = MatrixColoring.create(A, cfg.root_scope())
+        assert col.colors.is_cuda
+        assert col.validate(A, level=2), scheme
+    for scheme in ("MIN_MAX", "PARALLEL_GREEDY", "MULTI_HASH"):
+        cfg = AMGConfig.from_dict({"solver": {
+            "solver": "MULTICOLOR_GS", "matrix_coloring_scheme": scheme,
+            "coloring_level": 2}})
+        col = MatrixColoring.create(A, cfg.root_scope())
+        assert col.colors.is_cuda
+        assert col.validate(A, level=2), scheme
