@@ -220,22 +220,58 @@ def _serial_greedy_bfs(A, scope, level):
 
 @register_coloring("GREEDY_RECOLOR")
 def _greedy_recolor(A, scope, level):
-    """MIN_MAX first, then one greedy pass in descending-color order that
-    moves every row down to its smallest feasible color."""
+    """MIN_MAX first, then a recolor-down pass in descending-color order
+    that moves every row to its smallest feasible color (reference
+    greedy_recolor.cu). The pass runs one color CLASS at a time: a class is
+    an independent set, so all its rows recolor simultaneously — the same
+    vectorized torch code is the device path (no host round-trip) and the
+    host path."""
     colors, num = COLORING_REGISTRY["MIN_MAX"](A, scope, level)
-    if torch.is_tensor(colors):
-        colors = colors.cpu().numpy().astype(np.int32)
-    indptr, indices = _host_adj(A, level)
-    order = np.argsort(-colors, kind="stable")
-    for i in order:
-        nb = indices[indptr[i]:indptr[i + 1]]
-        used = set(int(colors[j]) for j in nb)
-        c = 0
-        while c in used:
-            c += 1
-        if c < colors[i]:
-            colors[i] = c
-    return colors, int(colors.max()) + 1 if colors.size else 0
+    if not torch.is_tensor(colors):
+        colors = torch.from_numpy(np.ascontiguousarray(colors))
+    colors = colors.to(A.row_offsets.device, torch.int64)
+    n = A.n_rows
+    if n == 0 or num <= 1:
+        return colors.to(torch.int32), max(num, 1 if n else 0)
+    dev = colors.device
+    if level >= 2:
+        if dev.type == "cuda":
+            A2 = _square_graph_matrix(A)
+            ro64 = A2.row_offsets.to(torch.int64)
+            ci = A2.col_indices.to(torch.int64)
+        else:
+            indptr, indices = _host_adj(A, level)
+            ro64 = torch.from_numpy(indptr.astype(np.int64))
+            ci = torch.from_numpy(indices.astype(np.int64))
+    else:
+        ro64 = A.row_offsets.to(torch.int64)
+        ci = A.col_indices.to(torch.int64)
+    rows = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int64, device=dev), ro64[1:] - ro64[:-1])
+    keep = (ci < n) & (ci != rows)
+    er, ec = rows[keep], ci[keep]
+    for c in range(num - 1, 0, -1):
+        in_class = colors == c
+        if not bool(in_class.any()):
+            continue
+        sel = in_class[er]
+        sr, sc = er[sel], ec[sel]
+        # used[i, k] = neighbor of i wears color k (k < c suffices)
+        local = torch.full((n,), -1, dtype=torch.int64, device=dev)
+        cls_rows = torch.nonzero(in_class, as_tuple=True)[0]
+        local[cls_rows] = torch.arange(cls_rows.numel(), dtype=torch.int64,
+                                       device=dev)
+        used = torch.zeros(cls_rows.numel(), c + 1, dtype=torch.bool,
+                           device=dev)
+        nb_col = torch.clamp(colors[sc], max=c)
+        used[local[sr], nb_col] = True
+        # deterministic first-free color (argmin tie order is unspecified)
+        cand = torch.arange(c + 1, dtype=torch.int64, device=dev) \
+            .expand(cls_rows.numel(), c + 1)
+        new_c = torch.where(used, torch.full_like(cand, c), cand) \
+            .min(dim=1).values
+        colors[cls_rows] = torch.minimum(new_c, colors[cls_rows])
+    return colors.to(torch.int32), int(colors.max()) + 1
 
 
 @register_coloring("MULTI_HASH")
@@ -295,11 +331,14 @@ def _round_robin(A, scope, level):
 
 @register_coloring("UNIFORM")
 def _uniform(A, scope, level):
+    # index-pattern coloring, device-resident (reference uniform.cu)
     n = A.n_rows
-    deg = np.diff(A.row_offsets.cpu().numpy())
-    k = int(deg.max()) + 1 if n else 1
-    colors = (np.arange(n) % max(k, 1)).astype(np.int32)
-    return colors, max(k, 1) if n else 0
+    ro = A.row_offsets.to(torch.int64)
+    k = int((ro[1:] - ro[:-1]).max()) + 1 if n else 1
+    k = max(k, 1)
+    colors = (torch.arange(n, dtype=torch.int64,
+                           device=A.row_offsets.device) % k).to(torch.int32)
+    return colors, k if n else 0
 
 
 @register_coloring("MIN_MAX_2RING")
