@@ -7,21 +7,29 @@ distance-2 when ``coloring_level=2``) share a color; multicolor smoothers
 color. Scheme registry mirrors the reference factory names:
 
   MIN_MAX               hash local-maximum independent sets (min_max.cu) —
-                        the default; GPU path is a hand-written gfx950 kernel
+                        the default; GPU path is a hand-written gfx950
+                        kernel; level 2 squares the graph on device
   PARALLEL_GREEDY       Jones-Plassmann greedy, smallest feasible color
-                        (parallel_greedy.cu)
-  SERIAL_GREEDY_BFS     BFS-ordered sequential greedy (serial_greedy_bfs.cu)
-  GREEDY_RECOLOR        MIN_MAX then greedy recolor-down pass
-                        (min_max_2ring/greedy_recolor.cu)
-  MULTI_HASH            per-round hash local-max (multi_hash.cu)
-  ROUND_ROBIN           greedy with rotating first-fit start (round_robin.cu)
+                        (parallel_greedy.cu); device kernel, levels 1/2
+  SERIAL_GREEDY_BFS     BFS-ordered sequential greedy (serial_greedy_bfs.cu;
+                        host by algorithmic nature — BFS order)
+  GREEDY_RECOLOR        MIN_MAX then a recolor-down pass, parallel per
+                        color class (greedy_recolor.cu); device-capable
+  MULTI_HASH            per-round hash local-max (multi_hash.cu); device
+                        kernel, levels 1/2
+  ROUND_ROBIN           greedy with rotating first-fit start (round_robin.cu;
+                        host — the reference device variant is the inexact
+                        i%k pattern, ours keeps validity)
   UNIFORM               index-pattern coloring for banded/structured rows
                         (uniform.cu; validity not guaranteed on general
-                        graphs, exactly like the reference)
-  MIN_MAX_2RING         MIN_MAX on the distance-2 graph (min_max_2ring.cu)
-  GREEDY_MIN_MAX_2RING  greedy on the distance-2 graph
+                        graphs, exactly like the reference); device-resident
+  MIN_MAX_2RING         MIN_MAX on the distance-2 graph (min_max_2ring.cu);
+                        device: hash-SpGEMM square + coloring kernel
+  GREEDY_MIN_MAX_2RING  greedy on the distance-2 graph; device: JP kernel
+                        on the SpGEMM square
   LOCALLY_DOWNWIND      downwind-ordered greedy for Kaczmarz sweeps
-                        (locally_downwind.cu)
+                        (locally_downwind.cu; host — sequential downwind
+                        order)
 
 The attachment precomputes ``rows_sorted`` (row ids stably sorted by color,
 device-resident) and host-side ``bounds`` so every per-color kernel slice is
