@@ -153,6 +153,11 @@ def _register_core_parameters() -> None:
     P("eig_damping_factor", float, 0.85, "pagerank damping factor")
     P("eig_eigenvector", int, 1, "number of eigenvectors")
     P("eig_wanted_count", int, 1, "number of wanted eigenpairs")
+    P("eig_subspace_size", int, -1,
+      "subspace size for block methods (-1 = wanted+2)")
+    P("eig_convergence_check_freq", int, 1, "convergence check period")
+    P("eig_eigenvector_solver", str, "",
+      "eigenvector recovery solver (inverse iteration inner solve)")
 
     # --- remaining reference registry (src/core.cu:307-545) ------------------
     # Wired parameters:

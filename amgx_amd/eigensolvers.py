@@ -149,6 +149,9 @@ class EigenSolver:
         self.which = scope.get("eig_which")
         self.shift = scope.get("eig_shift")
         self.wanted = scope.get("eig_wanted_count")
+        self.subspace = int(scope.get("eig_subspace_size") or -1)
+        self.check_freq = max(int(scope.get("eig_convergence_check_freq")
+                                  or 1), 1)
         self.op: Optional[Operator] = None
         self.status = EigenStatus()
 
@@ -272,16 +275,20 @@ class PowerIteration(EigenSolver):
             ops.scal(v, 1.0 / nv)
             self.op.apply(v, Av)
             lam = self.dot(v, Av)
-            # residual ||Av - lam v||
-            r = Av.clone()
-            ops.axpy(r, v, -lam)
-            rn = self.nrm2(r) / max(abs(lam), 1e-300)
-            st.residuals.append(rn)
             st.iterations = it + 1
+            # eig_convergence_check_freq: skip the residual norm (an extra
+            # vector op + reduction) on off-period iterations (reference
+            # eigensolvers.cu:30)
+            if (it + 1) % self.check_freq == 0 or it + 1 == self.max_iters:
+                r = Av.clone()
+                ops.axpy(r, v, -lam)
+                rn = self.nrm2(r) / max(abs(lam), 1e-300)
+                st.residuals.append(rn)
+                if rn < self.tol:
+                    v, Av = Av.clone(), Av
+                    st.converged = True
+                    break
             v, Av = Av.clone(), Av
-            if rn < self.tol:
-                st.converged = True
-                break
         st.eigenvalues = [self._to_true_eig(lam)]
         nv = self.nrm2(v)
         st.eigenvector = v / nv if nv else v
@@ -403,7 +410,8 @@ class SubspaceIteration(EigenSolver):
     def solve(self, x0=None):
         st = self.status = EigenStatus()
         k = max(self.wanted, 1)
-        p = min(k + 2, self.op.n)
+        p = self.subspace if self.subspace > 0 else k + 2
+        p = min(max(p, k), self.op.n)
         mgr = self._mgr()
         n = mgr.ext_size if mgr is not None else self.op.n
         g = torch.Generator().manual_seed(

@@ -152,3 +152,44 @@ def test_shipped_eigen_configs():
         assert abs(st.eigenvalues[-1 if es.which != "smallest" else 0]
                    - ref) < 1e-3 * abs(ref), (os.path.basename(f),
                                               st.eigenvalues, ref)
+
+
+def test_eig_param_parity_and_knobs():
+    """All reference eigensolver parameters (eigensolvers.cu:20-44) are
+    registered; eig_subspace_size sets the block width of subspace
+    iteration and eig_convergence_check_freq batches the residual check."""
+    from amgx_amd.config import PARAM_REGISTRY
+    for p in ("eig_solver", "eig_max_iters", "eig_tolerance", "eig_which",
+              "eig_shift", "eig_damping_factor", "eig_eigenvector",
+              "eig_wanted_count", "eig_subspace_size",
+              "eig_convergence_check_freq", "eig_eigenvector_solver"):
+        assert p in PARAM_REGISTRY, p
+    import torch
+
+    from amgx_amd.config import AMGConfig
+    from amgx_amd.eigensolvers import create_eigensolver
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    A = poisson_2d(12, 12)
+    cfg = AMGConfig.from_dict({"eig_solver": "SUBSPACE_ITERATION",
+                               "eig_max_iters": 300, "eig_tolerance": 1e-6,
+                               "eig_wanted_count": 2,
+                               "eig_subspace_size": 6})
+    s = create_eigensolver(cfg.root_scope(), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    assert st.converged and len(st.eigenvalues) >= 2
+    import scipy.sparse as sp
+
+    from amgx_amd.matrix import CSRMatrix
+    D = CSRMatrix.from_scipy(sp.diags([20.0] + [1.0 + 0.1 * i for i in range(49)],
+                                      format="csr").tocsr())
+    cfg2 = AMGConfig.from_dict({"eig_solver": "POWER_ITERATION",
+                                "eig_max_iters": 500, "eig_tolerance": 1e-8,
+                                "eig_convergence_check_freq": 5})
+    s2 = create_eigensolver(cfg2.root_scope(), Resources("cpu"))
+    s2.setup(D)
+    st2 = s2.solve()
+    assert st2.converged
+    assert st2.iterations % 5 == 0 or st2.iterations == 500
+    assert abs(st2.eigenvalues[0] - 20.0) < 1e-4
