@@ -687,3 +687,46 @@ def test_square_graph_coloring_path():
     assert A2h.n_cols == A.n_rows
     c2, n2 = ops.color_matrix(A2h)
     assert MatrixColoring(torch.as_tensor(c2), n2).validate(A, level=2)
+
+
+def test_nested_amg_equivalence():
+    """5-level AMG == 3-level AMG whose coarse solver is another 3-level
+    AMG (5 levels total) — bitwise-identical iterates (reference
+    src/tests/nested_amg_equivalence.cu, tol 1e-10; deterministic SIZE_2
+    matching makes the sub-hierarchies identical here)."""
+    import copy
+
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.problems import poisson_3d
+    from amgx_amd.resources import Resources
+
+    def run(cfg_dict):
+        A = poisson_3d(12, 12, 12)
+        cfg = AMGConfig.from_dict(cfg_dict)
+        s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        s.solve(b, x, zero_initial_guess=True)
+        return x
+
+    flat = {"config_version": 2, "solver": {
+        "algorithm": "AGGREGATION", "solver": "AMG",
+        "smoother": "BLOCK_JACOBI", "coarse_solver": "NOSOLVER",
+        "presweeps": 1, "postsweeps": 1, "selector": "SIZE_2",
+        "coarsest_sweeps": 1, "max_iters": 20, "monitor_residual": 1,
+        "max_levels": 5, "tolerance": 1e-6, "norm": "L1", "cycle": "V",
+        "min_coarse_rows": 2}}
+    nested = copy.deepcopy(flat)
+    nested["solver"]["max_levels"] = 3
+    nested["solver"]["coarse_solver"] = {
+        "solver": "AMG", "algorithm": "AGGREGATION",
+        "smoother": "BLOCK_JACOBI", "presweeps": 1, "postsweeps": 1,
+        "selector": "SIZE_2", "max_iters": 1, "coarse_solver": "NOSOLVER",
+        "scope": "lower", "max_levels": 3, "coarsest_sweeps": 1,
+        "cycle": "V", "min_coarse_rows": 2}
+    x1 = run(flat)
+    x2 = run(nested)
+    assert torch.equal(x1, x2), (x1 - x2).abs().max()
