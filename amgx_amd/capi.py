@@ -94,8 +94,11 @@ class _ConfigHandle:
 
 
 class _ResourcesHandle:
-    def __init__(self, res: Resources):
+    def __init__(self, res: Resources, cfg=None):
         self.res = res
+        # the resources-level config governs IO knobs (block_convert,
+        # rhs_from_a) exactly like the reference (src/amgx_c.cu:5016-5019)
+        self.cfg = cfg
 
 
 class _DistributionHandle:
@@ -239,7 +242,7 @@ def AMGX_config_destroy(cfg: _ConfigHandle):
 # ----------------------------------------------------------------- resources
 @_amgx_try
 def AMGX_resources_create_simple(cfg: _ConfigHandle):
-    return RC_OK, _ResourcesHandle(Resources())
+    return RC_OK, _ResourcesHandle(Resources(), cfg)
 
 
 @_amgx_try
@@ -247,7 +250,8 @@ def AMGX_resources_create(cfg: _ConfigHandle, comm=None, device_num: int = 0,
                           devices=None):
     dev = f"cuda:{device_num}" if torch.cuda.is_available() else "cpu"
     distributed = comm is not None
-    return RC_OK, _ResourcesHandle(Resources(dev, distributed=distributed))
+    return RC_OK, _ResourcesHandle(Resources(dev, distributed=distributed),
+                                   cfg)
 
 
 @_amgx_try
@@ -280,7 +284,9 @@ def AMGX_distribution_destroy(dist: _DistributionHandle):
 @_amgx_try
 def AMGX_matrix_create(res: _ResourcesHandle, mode: str):
     _parse_mode(mode)
-    return RC_OK, _MatrixHandle(res.res, mode)
+    h = _MatrixHandle(res.res, mode)
+    h.res_cfg = getattr(res, "cfg", None)
+    return RC_OK, h
 
 
 @_amgx_try
@@ -631,8 +637,21 @@ def AMGX_read_system(m: _MatrixHandle, rhs: _VectorHandle,
     dev = mem if mem == "cpu" else m.res.device
     reader = read_system_any if is_binary_file(path) else read_system
     A, b, x0 = reader(path, device=dev, dtype=matprec)
+    # resources-config IO knobs (reference src/amgx_c.cu:5008-5019):
+    # block_convert=b regroups a scalar system into bxb block-CSR;
+    # rhs_from_a=1 generates b = A*[1..1]^T when the file has no RHS
+    cfg_h = getattr(m, "res_cfg", None)
+    scope = cfg_h.cfg.root_scope() if cfg_h is not None else None
+    bconv = int(scope.get("block_convert") or 0) if scope is not None else 0
+    if bconv > 1 and A.block_dim == 1:
+        A = _block_convert(A, bconv)
     m.A = A
-    n = A.n_rows
+    n = A.n_rows * A.block_dim
+    if b is None and scope is not None \
+            and int(scope.get("rhs_from_a") or 0) == 1:
+        from . import ops as _ops
+        ones = torch.ones(n, dtype=A.dtype, device=A.values.device)
+        b = _ops.spmv(A, ones)
     if rhs is not None:
         rhs.v = b.to(vecprec) if b is not None else \
             torch.ones(n, dtype=vecprec, device=dev)
@@ -642,6 +661,24 @@ def AMGX_read_system(m: _MatrixHandle, rhs: _VectorHandle,
             torch.zeros(n, dtype=vecprec, device=dev)
         sol.n = n
     return RC_OK
+
+
+def _block_convert(A, b: int):
+    """Scalar CSR -> bxb block-CSR by grouping consecutive rows/cols
+    (reference Ahc.convert, src/amgx_c.cu:1303-1317; src/matrix_io).
+    n must be divisible by b; absent entries zero-fill their block."""
+    from .matrix import CSRMatrix
+    if A.n_rows % b:
+        raise ValueError(
+            f"block_convert={b}: {A.n_rows} rows not divisible")
+    bsr = A.to_scipy().tobsr((b, b))
+    bsr.sort_indices()
+    ro = torch.from_numpy(bsr.indptr.astype(np.int32))
+    ci = torch.from_numpy(bsr.indices.astype(np.int32))
+    va = torch.from_numpy(np.ascontiguousarray(bsr.data)).to(A.dtype)
+    dev = A.values.device
+    return CSRMatrix(ro.to(dev), ci.to(dev), va.to(dev),
+                     n_cols=bsr.shape[1] // b, block_dim=b)
 
 
 @_amgx_try

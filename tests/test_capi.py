@@ -394,3 +394,34 @@ def test_upload_all_global_single_process():
     rc, st = C.AMGX_solver_get_status(s)
     assert st == C.AMGX_SOLVE_SUCCESS
     C.AMGX_finalize()
+
+
+def test_read_system_block_convert_and_rhs_from_a():
+    """Resources-config IO knobs (reference src/amgx_c.cu:5008-5019):
+    block_convert=4 regroups a scalar file system into 4x4 block-CSR and
+    rhs_from_a=1 generates b = A*[1..1]^T when the file has no RHS."""
+    from amgx_amd import capi as C
+    from amgx_amd.io.matrix_market import write_matrix_market
+    A = poisson_2d(8, 8)   # 64 rows -> 16 block-4 rows
+    tmp = tempfile.mkdtemp()
+    path = os.path.join(tmp, "m.mtx")
+    write_matrix_market(path, A)
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create(
+        "config_version=2, solver=PCG, max_iters=200, tolerance=1e-8,"
+        " convergence=RELATIVE_INI, monitor_residual=1, block_convert=4,"
+        " rhs_from_a=1")
+    assert rc == C.RC_OK
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    rc, b = C.AMGX_vector_create(res, "hDDI")
+    rc, x = C.AMGX_vector_create(res, "hDDI")
+    assert C.AMGX_read_system(m, b, x, path) == C.RC_OK
+    assert m.A.block_dim == 4 and m.A.n_rows == 16
+    ref = torch.from_numpy(A.to_scipy() @ np.ones(64))
+    assert torch.allclose(b.v, ref)
+    rc, s = C.AMGX_solver_create(res, "hDDI", cfg)
+    assert C.AMGX_solver_setup(s, m) == C.RC_OK
+    assert C.AMGX_solver_solve(s, b, x) == C.RC_OK
+    resid = ref - torch.from_numpy(A.to_scipy() @ x.v.numpy())
+    assert float(resid.norm() / ref.norm()) < 1e-7
