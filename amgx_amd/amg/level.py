@@ -305,6 +305,13 @@ class ClassicalLevel(AMGLevel):
     default_interp = "D1"
 
     def __init__(self, A, scope, index):
+        if A.block_dim != 1:
+            # reference classical path is scalar-only
+            # (classical_amg_level.cu computeAOperator_1x1)
+            raise ValueError(
+                "CLASSICAL AMG supports scalar (1x1) matrices only, got "
+                f"block_dim={A.block_dim}; use algorithm=AGGREGATION for "
+                "block systems")
         super().__init__(A, scope, index)
         self.P = None
         self.R = None

@@ -730,3 +730,22 @@ def test_nested_amg_equivalence():
     x1 = run(flat)
     x2 = run(nested)
     assert torch.equal(x1, x2), (x1 - x2).abs().max()
+
+
+def test_classical_block_matrix_clean_error():
+    """CLASSICAL AMG is scalar-only, like the reference (classical_amg_
+    level.cu computeAOperator_1x1): block matrices get a clear error, not
+    a broadcast failure deep in setup."""
+    import pytest
+    import torch  # noqa: F401
+
+    from amgx_amd import AMGConfig, create_solver
+    from amgx_amd.problems import block_laplacian
+    from amgx_amd.resources import Resources
+    A = block_laplacian(8, 8, block_dim=4)
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "AMG", "algorithm": "CLASSICAL",
+        "smoother": "BLOCK_JACOBI", "max_iters": 1}})
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    with pytest.raises(ValueError, match="scalar"):
+        s.setup(A)
