@@ -963,3 +963,17 @@ def test_device_2ring_colorings():
         col = MatrixColoring.create(A, cfg.root_scope())
         assert col.colors.is_cuda
         assert col.validate(A, level=2), scheme
+
+
+def test_coarse_generators_agree():
+    """The LDS-hash LOW_DEG generator and the one-sort THRUST generator
+    produce the same Galerkin coarse matrix for the same matching
+    (reference coarseAgenerators family contract)."""
+    from amgx_amd.ops import size2_matching
+    A = to_gpu(poisson_3d(12, 12, 12))
+    agg, nagg = size2_matching(A)
+    C1 = ops.galerkin_aggregation(A, agg, nagg)                      # LOW_DEG hash
+    C2 = ops.galerkin_aggregation(A, agg, nagg, generator="THRUST")  # one-sort
+    assert torch.equal(C1.row_offsets, C2.row_offsets)
+    assert torch.equal(C1.col_indices, C2.col_indices)
+    assert torch.allclose(C1.values, C2.values, atol=1e-12)
