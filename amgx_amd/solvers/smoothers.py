@@ -208,9 +208,15 @@ class ChebyshevSolver(_SmootherBase):
         self.dinv = ops.jacobi_dinv(self.A, l1=True)
         if self.est_mode in (0, 1):
             self.lmax = self._power_iteration(16)
-        else:
+            self.lmin = self.lmax / 8.0
+        elif self.est_mode == 2:
+            # preconditioned spectrum assumed compressed to ~[0,1]
+            # (reference cheb_solver.cu:196: lmax=0.9)
+            self.lmax = 0.9
+            self.lmin = self.lmax / 8.0
+        else:   # mode 3: user-provided estimates (cheb_solver.cu:209-211)
             self.lmax = self.scope.get("cheby_max_lambda")
-        self.lmin = self.lmax / 8.0
+            self.lmin = self.scope.get("cheby_min_lambda")
         self._init_cheb()
 
     def _init_cheb(self):

@@ -703,3 +703,28 @@ def test_smoother_random_matrix_stays_finite():
         s.setup(A)
         s.solve(b, x)
         assert torch.isfinite(x).all(), name
+
+
+def test_chebyshev_lambda_modes():
+    """All four reference lambda-estimate modes (cheb_solver.cu:180-213):
+    0/1 power iteration, 2 preconditioned-spectrum assumption (lmax=0.9),
+    3 user-provided bounds."""
+    A = poisson_2d(12, 12)
+    for mode, extra in ((0, {}), (1, {}), (2, {}),
+                        (3, {"cheby_max_lambda": 1.9,
+                             "cheby_min_lambda": 0.1})):
+        cfg = {"solver": "CHEBYSHEV", "max_iters": 10,
+               "chebyshev_lambda_estimate_mode": mode}
+        cfg.update(extra)
+        s = make(cfg)
+        b = torch.ones(A.n_rows, dtype=torch.float64)
+        x = torch.zeros_like(b)
+        s.setup(A)
+        if mode == 2:
+            assert s.lmax == 0.9
+        if mode == 3:
+            assert s.lmax == 1.9 and s.lmin == 0.1
+        r0 = ops.nrm2(ops.residual(A, x, b))
+        s.solve(b, x)
+        r1 = ops.nrm2(ops.residual(A, x, b))
+        assert np.isfinite(r1) and r1 < r0, (mode, r1 / r0)
