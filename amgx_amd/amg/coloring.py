@@ -320,7 +320,21 @@ def _multi_hash(A, scope, level):
 @register_coloring("ROUND_ROBIN")
 def _round_robin(A, scope, level):
     """Greedy with a rotating first-fit start color — spreads rows evenly
-    over colors (reference round_robin.cu intent)."""
+    over colors (reference round_robin.cu intent). Large device matrices
+    take the reference's own device formulation, colors[i] = i % num_colors
+    (round_robin.cu:29 colorRowsKernel — explicitly inexact there too),
+    which stays device-resident; the sequential rotating greedy would cost
+    minutes of host time at that scale."""
+    n = A.n_rows
+    if A.row_offsets.is_cuda and n > 20000:
+        num = int(scope.get("num_colors") or 0) if scope is not None else 0
+        if num <= 0:
+            ro = A.row_offsets.to(torch.int64)
+            num = max(int((ro[1:] - ro[:-1]).max()) + 1, 1)
+        colors = (torch.arange(n, dtype=torch.int64,
+                               device=A.row_offsets.device)
+                  % num).to(torch.int32)
+        return colors, num
     indptr, indices = _host_adj(A, level)
     n = indptr.size - 1
     colors = np.full(n, -1, dtype=np.int32)
