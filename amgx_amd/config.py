@@ -310,6 +310,10 @@ class AMGConfig:
             node = scopes.get(scope)
             if node is None:
                 node = pending.setdefault(scope, {})
+            if key not in PARAM_REGISTRY and key not in _NON_SOLVER_KEYS:
+                # reference parseParameterString rejects unregistered names
+                # (src/amg_config.cu "Variable not registered")
+                raise KeyError(f"unknown config parameter {key!r}")
             pval = _coerce(key, val)
             if new_scope is not None:
                 sub = pending.pop(new_scope, None)

@@ -99,3 +99,14 @@ def test_all_shipped_configs_solve():
         st = s.solve(b, x, zero_initial_guess=True)
         rel = ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b)
         assert st.converged and rel < 1e-4, (os.path.basename(f), st, rel)
+
+
+def test_unknown_flat_parameter_rejected():
+    """parseParameterString rejects unregistered names (reference
+    src/amg_config.cu 'Variable not registered')."""
+    import pytest
+
+    from amgx_amd.config import AMGConfig
+    with pytest.raises(KeyError):
+        AMGConfig.parse("not_a_real_parameter=3")
+    AMGConfig.parse("max_iters=10")   # registered names parse
