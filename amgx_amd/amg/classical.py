@@ -145,6 +145,20 @@ def select_pmis(A, S, scope):
     return ops._backend(A).pmis_select(A, S)
 
 
+@_register(SELECTOR_REGISTRY, "DUMMY")
+def select_dummy(A, S, scope):
+    """Alternating C/F pattern for testing (reference
+    src/classical/selectors/dummy_selector.cu:26: odd rows coarse)."""
+    n = A.n_rows
+    idx = torch.arange(n, device=A.row_offsets.device)
+    cf = torch.full((n,), -1, dtype=torch.int32,
+                    device=A.row_offsets.device)
+    coarse = idx[idx % 2 == 1]
+    cf[coarse] = torch.arange(coarse.numel(), dtype=torch.int32,
+                              device=cf.device)
+    return cf.cpu(), int(coarse.numel())
+
+
 @_register(SELECTOR_REGISTRY, "HMIS")
 def select_hmis(A, S, scope):
     """HMIS = PMIS on the distance-2 strong graph (reference

@@ -749,3 +749,30 @@ def test_classical_block_matrix_clean_error():
     s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
     with pytest.raises(ValueError, match="scalar"):
         s.setup(A)
+
+
+def test_classical_dummy_selector():
+    """DUMMY classical selector: alternating C/F (reference
+    dummy_selector.cu:26) still yields a convergent hierarchy."""
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    A = poisson_2d(16, 16)
+    cfg = AMGConfig.from_dict({"solver": {
+        "preconditioner": {
+            "solver": "AMG", "algorithm": "CLASSICAL", "selector": "DUMMY",
+            "smoother": {"solver": "MULTICOLOR_GS", "symmetric_GS": 1,
+                         "max_iters": 1},
+            "presweeps": 1, "postsweeps": 1, "max_iters": 1,
+            "min_coarse_rows": 8},
+        "solver": "PCG", "max_iters": 200, "monitor_residual": 1,
+        "convergence": "RELATIVE_INI", "tolerance": 1e-8}})
+    s = create_solver(cfg.root_scope(), resources=Resources("cpu"))
+    b = torch.ones(A.n_rows, dtype=torch.float64)
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+    assert st.converged and rel < 1e-7
