@@ -259,9 +259,21 @@ class EigenSolver:
 
 @register_eigensolver("POWER_ITERATION")
 @register_eigensolver("SINGLE_ITERATION")
+@register_eigensolver("PAGERANK")
+@register_eigensolver("INVERSE_ITERATION")
 class PowerIteration(EigenSolver):
     """Reference src/eigensolvers/power_iteration_eigensolver.cu (and
-    single_iteration_eigensolver.cu: one-iteration variant)."""
+    single_iteration_eigensolver.cu). The reference registers PAGERANK and
+    INVERSE_ITERATION to this same single-iteration family
+    (eigensolvers.cu:38-43); here the operator attached at setup (PageRank
+    / shifted-inverse via eig_which=smallest) selects the behavior, and
+    the INVERSE_ITERATION name defaults eig_which to smallest."""
+
+    def __init__(self, scope, resources):
+        super().__init__(scope, resources)
+        if scope.get("eig_solver") == "INVERSE_ITERATION" \
+                and not scope.has("eig_which"):
+            self.which = "smallest"
 
     def solve(self, x0=None):
         st = self.status = EigenStatus()

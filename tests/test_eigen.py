@@ -193,3 +193,25 @@ def test_eig_param_parity_and_knobs():
     assert st2.converged
     assert st2.iterations % 5 == 0 or st2.iterations == 500
     assert abs(st2.eigenvalues[0] - 20.0) < 1e-4
+
+
+def test_inverse_iteration_and_pagerank_aliases():
+    """Reference registers PAGERANK and INVERSE_ITERATION into the
+    single-iteration family (eigensolvers.cu:38-43); INVERSE_ITERATION
+    defaults to the smallest eigenpair."""
+    import scipy.sparse.linalg as spla
+
+    from amgx_amd.config import AMGConfig
+    from amgx_amd.eigensolvers import EIGEN_REGISTRY, create_eigensolver
+    from amgx_amd.problems import poisson_2d
+    from amgx_amd.resources import Resources
+    assert "PAGERANK" in EIGEN_REGISTRY and "INVERSE_ITERATION" in EIGEN_REGISTRY
+    A = poisson_2d(12, 12)
+    cfg = AMGConfig.from_dict({"eig_solver": "INVERSE_ITERATION",
+                               "eig_max_iters": 200, "eig_tolerance": 1e-6})
+    s = create_eigensolver(cfg.root_scope(), Resources("cpu"))
+    s.setup(A)
+    st = s.solve()
+    truth = spla.eigsh(A.to_scipy(), k=1, which="SM",
+                       return_eigenvectors=False)[0]
+    assert st.converged and abs(st.eigenvalues[0] - truth) < 1e-4
