@@ -65,6 +65,7 @@ def select_size8(A, scope):
 
 
 @register_agg_selector("PARALLEL_GREEDY")
+@register_agg_selector("PARALLEL_GREEDY_SELECTOR")
 def select_parallel_greedy(A, scope):
     """Handshaking matching run to exhaustion (reference
     parallel_greedy_selector.cu behaves like an aggressive pairwise pass)."""

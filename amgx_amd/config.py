@@ -55,7 +55,8 @@ def _register_core_parameters() -> None:
     P("tolerance", float, 1e-12, "convergence tolerance")
     P("alt_rel_tolerance", float, -1.0, "alternative relative tolerance")
     P("convergence", str, "ABSOLUTE", "convergence criterion",
-      ("ABSOLUTE", "RELATIVE_INI", "RELATIVE_MAX", "RELATIVE_INI_CORE", "COMBINED_REL_INI_ABS"))
+      ("ABSOLUTE", "RELATIVE_INI", "RELATIVE_MAX", "RELATIVE_INI_CORE",
+       "RELATIVE_MAX_CORE", "COMBINED_REL_INI_ABS"))
     P("norm", str, "L2", "residual norm", ("L1", "L2", "LMAX", "L1_SCALED"))
     P("use_scalar_norm", int, 0, "use scalar norm for block matrices")
     P("monitor_residual", int, 0, "compute residual every iteration")

@@ -66,7 +66,7 @@ class Convergence:
             return nrm <= self.tol
         if self.kind in ("RELATIVE_INI", "RELATIVE_INI_CORE"):
             return nrm <= self.tol * (self.ini_norm if self.ini_norm else 1.0)
-        if self.kind == "RELATIVE_MAX":
+        if self.kind in ("RELATIVE_MAX", "RELATIVE_MAX_CORE"):
             ref = max(self.ini_norm or 0.0, 1e-300)
             return nrm <= self.tol * ref
         if self.kind == "COMBINED_REL_INI_ABS":
