@@ -132,43 +132,10 @@ def _min_max(A, scope, level):
         else 0.0
     if level <= 1:
         return ops.color_matrix(A, max_uncolored_frac=float(frac or 0))
-    if A.row_offsets.is_cuda:
-        # device-resident distance-2: color the squared graph with the same
-        # gfx950 kernel (no host round-trip mid-setup)
-        return ops.color_matrix(_square_graph_matrix(A),
-                                max_uncolored_frac=float(frac or 0))
-    return _min_max_rounds(*_host_adj(A, level))
-
-
-def _min_max_rounds(indptr, indices):
-    """Luby rounds: row is colored c at round c if its hashed weight is a
-    local max/min among uncolored neighbors (reference min_max.cu: two colors
-    per round, max and min)."""
-    n = indptr.size - 1
-    colors = np.full(n, -1, dtype=np.int32)
-    w = _hash(np.arange(n), 1)
-    c = 0
-    guard = 0
-    while (colors < 0).any():
-        un = colors < 0          # frozen snapshot: all decisions this round
-        rows = np.nonzero(un)[0]  # are made against the same uncolored set
-        newc = []
-        for i in rows:
-            nb = indices[indptr[i]:indptr[i + 1]]
-            nb = nb[un[nb]]
-            nbw = w[nb[nb != i]]
-            # _hash is a uint64 bijection, so weights are tie-free
-            if nbw.size == 0 or w[i] > nbw.max():
-                newc.append((i, c))
-            elif w[i] < nbw.min():
-                newc.append((i, c + 1))
-        for i, cc in newc:
-            colors[i] = cc
-        c += 2
-        guard += 1
-        if guard > n + 2:
-            raise RuntimeError("MIN_MAX coloring did not converge")
-    return colors, int(colors.max()) + 1
+    # distance-2: color the squared graph with the same (vectorized /
+    # gfx950-kernel) path — device-resident on GPU, loop-free on host
+    return ops.color_matrix(_square_graph_matrix(A),
+                            max_uncolored_frac=float(frac or 0))
 
 
 @register_coloring("PARALLEL_GREEDY")
@@ -365,9 +332,7 @@ def _uniform(A, scope, level):
 
 @register_coloring("MIN_MAX_2RING")
 def _min_max_2ring(A, scope, level):
-    if A.row_offsets.is_cuda:
-        return ops.color_matrix(_square_graph_matrix(A))
-    return _min_max_rounds(*_host_adj(A, 2))
+    return ops.color_matrix(_square_graph_matrix(A))
 
 
 @register_coloring("GREEDY_MIN_MAX_2RING")
