@@ -105,11 +105,15 @@ class DistributedManager:
         ci_t = torch.as_tensor(np.asarray(cols_global)
                                if not torch.is_tensor(cols_global)
                                else cols_global, dtype=torch.int64).to(device)
+        # explicit width: reshape(0, -1) is ambiguous when a rank's local
+        # matrix is empty (deep distributed coarse levels can leave a rank
+        # with zero rows/nnz)
+        w = block_dim * block_dim
         if torch.is_tensor(vals):
-            va_t = vals.to(dtype).to(device).reshape(ci_t.numel(), -1)
+            va_t = vals.to(dtype).to(device).reshape(int(ci_t.numel()), w)
         else:
             va_t = torch.as_tensor(np.ascontiguousarray(vals)).to(dtype) \
-                .to(device).reshape(ci_t.numel(), -1)
+                .to(device).reshape(int(ci_t.numel()), w)
         own_lo, own_hi = mgr.row_start, mgr.row_start + n_loc
         is_halo = (ci_t < own_lo) | (ci_t >= own_hi)
         halo_cols = torch.unique(ci_t[is_halo])        # sorted global ids

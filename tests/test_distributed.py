@@ -1174,3 +1174,46 @@ def _impl_test_dist_disconnected_partitions(rank, world, tmp):
     st = s.solve(b, x, zero_initial_guess=True)
     rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
     assert st.converged and rel < 1e-7, (st.iterations, rel)
+
+
+def test_dist_empty_rank_upload():
+    _run_dist(test_dist_empty_rank_upload)
+
+
+def _impl_test_dist_empty_rank_upload(rank, world, tmp):
+    """A rank with ZERO local rows/nnz (deep distributed classical coarse
+    levels produce these — the world-8 classical rehearsal crashed on the
+    ambiguous reshape of an empty value tensor). upload_global_csr and the
+    solve path must handle the empty slice."""
+    import numpy as np
+    import scipy.sparse as sp
+    import torch
+
+    from amgx_amd import AMGConfig, create_solver, ops
+    from amgx_amd.distributed.manager import DistributedManager
+    from amgx_amd.resources import Resources
+    n = 40 if rank == 0 else 0
+    T = sp.diags([-1, 2.1, -1], [-1, 0, 1], (40, 40), format="csr")
+    if rank == 0:
+        ro, ci, va = T.indptr, T.indices, T.data
+    else:
+        ro = np.zeros(1, dtype=np.int64)
+        ci = np.zeros(0, dtype=np.int64)
+        va = np.zeros(0)
+    A = DistributedManager.upload_global_csr(ro, ci, va, n,
+                                             0 if rank == 0 else 40, 40)
+    cfg = AMGConfig.from_dict({"solver": {
+        "solver": "PCG", "preconditioner": "BLOCK_JACOBI", "max_iters": 200,
+        "monitor_residual": 1, "convergence": "RELATIVE_INI",
+        "tolerance": 1e-8}})
+    s = create_solver(cfg.root_scope(),
+                      resources=Resources("cpu", distributed=True))
+    b = A.manager.new_ext_vec(torch.float64)
+    b[:A.manager.owned_size] = 1.0
+    x = torch.zeros_like(b)
+    s.setup(A)
+    st = s.solve(b, x, zero_initial_guess=True)
+    assert st.converged, st.iterations
+    if rank == 0:
+        rel = float(ops.nrm2(ops.residual(A, x, b)) / ops.nrm2(b))
+        assert rel < 1e-7
