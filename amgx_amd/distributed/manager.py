@@ -308,12 +308,16 @@ class DistributedManager:
             xf.index_add_(0, idx, recv[i])
 
     # -------------------------------------------------------------- reductions
-    def global_sum(self, v: float) -> float:
-        t = torch.tensor([v], dtype=torch.float64,
+    def global_sum(self, v):
+        """Sum a scalar across ranks; complex dots (hZZI modes) ride the
+        same all_reduce (torch views complex as interleaved reals)."""
+        dt = torch.complex128 if isinstance(v, complex) else torch.float64
+        t = torch.tensor([v], dtype=dt,
                          device=self.device if self.device.type == "cuda"
                          else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
-        return float(t.item())
+        val = t.item()
+        return val if dt is torch.complex128 else float(val)
 
     def global_norm(self, local_nrm: float, kind: str) -> float:
         if kind == "L2":
