@@ -15,6 +15,7 @@ from .. import ops
 from ..config import ConfigScope
 from ..solvers.base import create_solver
 from .level import AMGLevel, create_level
+from ..output import amgx_output
 
 
 class AMGHierarchy:
@@ -80,7 +81,7 @@ class AMGHierarchy:
             torch.cuda.synchronize()
         self.setup_time = time.perf_counter() - t0
         if self.print_grid_stats and self.res.rank == 0:
-            print(self.grid_stats())
+            amgx_output(self.grid_stats() + "\n")
 
     def resetup(self, A, reuse_levels: int):
         """Values-only re-setup: keep the coarsening structure (aggregates /

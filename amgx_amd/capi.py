@@ -191,6 +191,8 @@ def AMGX_reset_signal_handler():
 def AMGX_register_print_callback(cb):
     global _print_callback
     _print_callback = cb
+    from . import output
+    output.print_callback = cb
     return RC_OK
 
 

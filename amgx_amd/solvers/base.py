@@ -18,6 +18,7 @@ import torch
 from .. import ops
 from ..config import ConfigScope
 from ..resources import Resources, default_resources
+from ..output import amgx_output
 
 SOLVER_REGISTRY: Dict[str, Type["Solver"]] = {}
 
@@ -252,9 +253,9 @@ class Solver:
             self.convergence.set_initial(nrm)
             st.residuals.append(nrm)
             if self.print_solve_stats and self.res.rank == 0:
-                print(f"           iter      residual   rate")
-                print(f"           ----------------------------")
-                print(f"            Ini {nrm:14.6e}")
+                amgx_output(f"           iter      residual   rate" "\n")
+                amgx_output(f"           ----------------------------" "\n")
+                amgx_output(f"            Ini {nrm:14.6e}" "\n")
             if self.convergence.converged(nrm) and self.convergence.kind != "RELATIVE_INI":
                 st.status = SolveStatus.SUCCESS
                 st.solve_time = time.perf_counter() - t0
@@ -273,7 +274,7 @@ class Solver:
                 if self.print_solve_stats and self.res.rank == 0:
                     rate = (st.residuals[-1] / st.residuals[-2]
                             if st.residuals[-2] else 0.0)
-                    print(f"           {it:4d} {nrm:14.6e}  {rate:6.4f}")
+                    amgx_output(f"           {it:4d} {nrm:14.6e}  {rate:6.4f}" "\n")
                 if self.convergence.converged(nrm) \
                         and it + 1 >= self.min_iters:
                     st.status = SolveStatus.SUCCESS
@@ -301,18 +302,18 @@ class Solver:
             ini, last = st.residuals[0], st.residuals[-1]
             its = max(st.iterations, 1)
             rate = (last / ini) ** (1.0 / its) if ini > 0 else ini
-            print(f"           ----------------------------")
-            print(f"         Total Iterations: {st.iterations}")
-            print(f"         Avg Convergence Rate: {rate:15.4f}")
-            print(f"         Final Residual: {last:15.6e}")
-            print(f"         Total Reduction in Residual: "
-                  f"{(last / ini if ini > 0 else ini):15.6e}")
+            amgx_output(f"           ----------------------------" "\n")
+            amgx_output(f"         Total Iterations: {st.iterations}" "\n")
+            amgx_output(f"         Avg Convergence Rate: {rate:15.4f}" "\n")
+            amgx_output(f"         Final Residual: {last:15.6e}" "\n")
+            amgx_output(f"         Total Reduction in Residual: "
+                        f"{(last / ini if ini > 0 else ini):15.6e}\n")
             if self.scope.get("obtain_timings"):
-                print(f"         Total Time: {st.setup_time + st.solve_time:10.4f} s")
-                print(f"             setup: {st.setup_time:10.4f} s")
-                print(f"             solve: {st.solve_time:10.4f} s")
-                print(f"             solve(per iteration): "
-                      f"{st.solve_time / its:10.6f} s")
+                amgx_output(f"         Total Time: {st.setup_time + st.solve_time:10.4f} s" "\n")
+                amgx_output(f"             setup: {st.setup_time:10.4f} s" "\n")
+                amgx_output(f"             solve: {st.solve_time:10.4f} s" "\n")
+                amgx_output(f"             solve(per iteration): "
+                            f"{st.solve_time / its:10.6f} s\n")
         return st
 
     # default residual-norm recomputation; Krylov solvers override with their

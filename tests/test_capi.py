@@ -425,3 +425,31 @@ def test_read_system_block_convert_and_rhs_from_a():
     assert C.AMGX_solver_solve(s, b, x) == C.RC_OK
     resid = ref - torch.from_numpy(A.to_scipy() @ x.v.numpy())
     assert float(resid.norm() / ref.norm()) < 1e-7
+
+
+def test_print_callback_captures_all_output():
+    """AMGX_register_print_callback receives EVERY library print —
+    residual table, grid stats, timings (reference src/misc.cu
+    amgx_output indirection)."""
+    from amgx_amd import capi as C
+    captured = []
+    C.AMGX_initialize()
+    C.AMGX_register_print_callback(lambda m: captured.append(m))
+    try:
+        rc, cfg = C.AMGX_config_create(
+            "config_version=2, solver=PCG, max_iters=50, tolerance=1e-6,"
+            " convergence=RELATIVE_INI, monitor_residual=1,"
+            " print_solve_stats=1, obtain_timings=1")
+        rc, res = C.AMGX_resources_create_simple(cfg)
+        rc, m = C.AMGX_matrix_create(res, "hDDI")
+        rc, b = C.AMGX_vector_create(res, "hDDI")
+        rc, x = C.AMGX_vector_create(res, "hDDI")
+        C.AMGX_generate_distributed_poisson_7pt(m, b, x, 1, 1, 6, 6, 6)
+        rc, s = C.AMGX_solver_create(res, "hDDI", cfg)
+        C.AMGX_solver_setup(s, m)
+        C.AMGX_solver_solve(s, b, x)
+    finally:
+        C.AMGX_register_print_callback(None)
+    joined = "".join(captured)
+    assert "Total Iterations" in joined
+    assert "Total Time" in joined
