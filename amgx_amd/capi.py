@@ -969,7 +969,11 @@ def AMGX_solver_calculate_residual_norm(s: _SolverHandle, m: _MatrixHandle,
 
 @_amgx_try
 def AMGX_solver_register_print_callback(s: _SolverHandle, cb):
+    # per-solver output redirection (reference solver-level callback):
+    # consumed by Solver._out in solvers/base.py
     s.print_callback = cb
+    if s.solver is not None:
+        s.solver.print_cb = cb
     return RC_OK
 
 

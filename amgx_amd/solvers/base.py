@@ -105,6 +105,16 @@ class Solver:
 
     solver_name = "?"
     is_smoother = False   # smoothers skip convergence monitoring by default
+    print_cb = None       # per-solver output redirection (C API hook)
+
+    def _out(self, msg: str) -> None:
+        if self.print_cb is not None:
+            try:
+                self.print_cb(msg)
+                return
+            except Exception:
+                pass
+        amgx_output(msg)
 
     def __init__(self, scope: ConfigScope, resources: Resources):
         self.scope = scope
@@ -253,9 +263,9 @@ class Solver:
             self.convergence.set_initial(nrm)
             st.residuals.append(nrm)
             if self.print_solve_stats and self.res.rank == 0:
-                amgx_output(f"           iter      residual   rate" "\n")
-                amgx_output(f"           ----------------------------" "\n")
-                amgx_output(f"            Ini {nrm:14.6e}" "\n")
+                self._out(f"           iter      residual   rate" "\n")
+                self._out(f"           ----------------------------" "\n")
+                self._out(f"            Ini {nrm:14.6e}" "\n")
             if self.convergence.converged(nrm) and self.convergence.kind != "RELATIVE_INI":
                 st.status = SolveStatus.SUCCESS
                 st.solve_time = time.perf_counter() - t0
@@ -274,7 +284,7 @@ class Solver:
                 if self.print_solve_stats and self.res.rank == 0:
                     rate = (st.residuals[-1] / st.residuals[-2]
                             if st.residuals[-2] else 0.0)
-                    amgx_output(f"           {it:4d} {nrm:14.6e}  {rate:6.4f}" "\n")
+                    self._out(f"           {it:4d} {nrm:14.6e}  {rate:6.4f}" "\n")
                 if self.convergence.converged(nrm) \
                         and it + 1 >= self.min_iters:
                     st.status = SolveStatus.SUCCESS
@@ -302,17 +312,17 @@ class Solver:
             ini, last = st.residuals[0], st.residuals[-1]
             its = max(st.iterations, 1)
             rate = (last / ini) ** (1.0 / its) if ini > 0 else ini
-            amgx_output(f"           ----------------------------" "\n")
-            amgx_output(f"         Total Iterations: {st.iterations}" "\n")
-            amgx_output(f"         Avg Convergence Rate: {rate:15.4f}" "\n")
-            amgx_output(f"         Final Residual: {last:15.6e}" "\n")
-            amgx_output(f"         Total Reduction in Residual: "
+            self._out(f"           ----------------------------" "\n")
+            self._out(f"         Total Iterations: {st.iterations}" "\n")
+            self._out(f"         Avg Convergence Rate: {rate:15.4f}" "\n")
+            self._out(f"         Final Residual: {last:15.6e}" "\n")
+            self._out(f"         Total Reduction in Residual: "
                         f"{(last / ini if ini > 0 else ini):15.6e}\n")
             if self.scope.get("obtain_timings"):
-                amgx_output(f"         Total Time: {st.setup_time + st.solve_time:10.4f} s" "\n")
-                amgx_output(f"             setup: {st.setup_time:10.4f} s" "\n")
-                amgx_output(f"             solve: {st.solve_time:10.4f} s" "\n")
-                amgx_output(f"             solve(per iteration): "
+                self._out(f"         Total Time: {st.setup_time + st.solve_time:10.4f} s" "\n")
+                self._out(f"             setup: {st.setup_time:10.4f} s" "\n")
+                self._out(f"             solve: {st.solve_time:10.4f} s" "\n")
+                self._out(f"             solve(per iteration): "
                             f"{st.solve_time / its:10.6f} s\n")
         return st
 
