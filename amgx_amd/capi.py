@@ -581,6 +581,8 @@ def AMGX_solver_setup(s: _SolverHandle, m: _MatrixHandle):
     mem, _, _ = _parse_mode(s.mode)
     res = s.res if mem != "cpu" else Resources("cpu")
     s.solver = create_solver(s.cfg.root_scope(), resources=res)
+    if getattr(s, "print_callback", None) is not None:
+        s.solver.print_cb = s.print_callback   # registered pre-setup
     s.solver.setup(m.A)
     s.matrix = m
     return RC_OK

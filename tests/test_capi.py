@@ -453,3 +453,24 @@ def test_print_callback_captures_all_output():
     joined = "".join(captured)
     assert "Total Iterations" in joined
     assert "Total Time" in joined
+
+
+def test_solver_level_print_callback():
+    """AMGX_solver_register_print_callback redirects THAT solver's output
+    (registration before setup included)."""
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create(
+        "config_version=2, solver=PCG, max_iters=30, tolerance=1e-6,"
+        " convergence=RELATIVE_INI, monitor_residual=1, print_solve_stats=1")
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    rc, b = C.AMGX_vector_create(res, "hDDI")
+    rc, x = C.AMGX_vector_create(res, "hDDI")
+    C.AMGX_generate_distributed_poisson_7pt(m, b, x, 1, 1, 6, 6, 6)
+    rc, s = C.AMGX_solver_create(res, "hDDI", cfg)
+    got = []
+    C.AMGX_solver_register_print_callback(s, lambda msg: got.append(msg))
+    C.AMGX_solver_setup(s, m)
+    C.AMGX_solver_solve(s, b, x)
+    assert any("Total Iterations" in g for g in got)
