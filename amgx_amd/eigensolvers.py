@@ -236,6 +236,15 @@ class EigenSolver:
     def solver_setup(self):
         pass
 
+    def _start_vec(self, x0):
+        """Initial vector: the caller's x0 unless it is (numerically) zero —
+        a zero start annihilates power/Krylov iterations (the C API's
+        freshly-created solution vector is all zeros), fall back to the
+        seeded random start."""
+        if x0 is not None and float(self.nrm2(x0)) > 0.0:
+            return x0.clone()
+        return self._rand_vec()
+
     def _rand_vec(self, seed=42):
         mgr = self._mgr()
         n = mgr.ext_size if mgr is not None else self.op.n
@@ -277,7 +286,7 @@ class PowerIteration(EigenSolver):
 
     def solve(self, x0=None):
         st = self.status = EigenStatus()
-        v = x0.clone() if x0 is not None else self._rand_vec()
+        v = self._start_vec(x0)
         Av = torch.zeros_like(v)
         lam = 0.0
         for it in range(self.max_iters):
@@ -315,7 +324,7 @@ class Lanczos(EigenSolver):
     def solve(self, x0=None):
         st = self.status = EigenStatus()
         m = min(self.max_iters, self.op.n)
-        v = x0.clone() if x0 is not None else self._rand_vec()
+        v = self._start_vec(x0)
         ops.scal(v, 1.0 / self.nrm2(v))
         V = [v]
         alphas, betas = [], []
@@ -376,7 +385,7 @@ class Arnoldi(EigenSolver):
     def solve(self, x0=None):
         st = self.status = EigenStatus()
         m = min(self.max_iters, self.op.n)
-        v = x0.clone() if x0 is not None else self._rand_vec()
+        v = self._start_vec(x0)
         ops.scal(v, 1.0 / self.nrm2(v))
         V = [v]
         H = np.zeros((m + 1, m))
@@ -474,7 +483,7 @@ class LOBPCG(EigenSolver):
 
     def solve(self, x0=None):
         st = self.status = EigenStatus()
-        x = x0.clone() if x0 is not None else self._rand_vec()
+        x = self._start_vec(x0)
         ops.scal(x, 1.0 / self.nrm2(x))
         Ax = torch.zeros_like(x)
         self.op.apply(x, Ax)

@@ -1,3 +1,4 @@
+import os
 import numpy as np
 import pytest
 import torch
@@ -215,3 +216,26 @@ def test_inverse_iteration_and_pagerank_aliases():
     truth = spla.eigsh(A.to_scipy(), k=1, which="SM",
                        return_eigenvectors=False)[0]
     assert st.converged and abs(st.eigenvalues[0] - truth) < 1e-4
+
+
+def test_eigensolver_zero_initial_vector():
+    """A zero x0 (the C API's freshly-created solution vector) must not
+    annihilate the iteration: solvers fall back to the seeded random
+    start. Regression: INVERSE_FGMRES through the C API returned
+    lam=inf."""
+    from amgx_amd import capi as C
+    C.AMGX_initialize()
+    rc, cfg = C.AMGX_config_create_from_file(
+        os.path.join(os.path.dirname(__file__), os.pardir, "configs",
+                     "eigen", "INVERSE_FGMRES"))
+    assert rc == C.RC_OK
+    rc, res = C.AMGX_resources_create_simple(cfg)
+    rc, m = C.AMGX_matrix_create(res, "hDDI")
+    rc, b = C.AMGX_vector_create(res, "hDDI")
+    rc, x = C.AMGX_vector_create(res, "hDDI")
+    C.AMGX_generate_distributed_poisson_7pt(m, b, x, 1, 1, 8, 8, 8)
+    rc, es = C.AMGX_eigensolver_create(res, "hDDI", cfg)
+    assert C.AMGX_eigensolver_setup(es, m) == C.RC_OK
+    assert C.AMGX_eigensolver_solve(es, x) == C.RC_OK
+    st = es.status
+    assert st.converged and abs(st.eigenvalues[0] - 0.3618) < 1e-3
